@@ -1,0 +1,69 @@
+"""flashinfer_amd — an MI355X-native (CDNA4/gfx950) LLM inference kernel
+library with FlashInfer's capabilities: plan/run attention wrappers over
+paged/ragged KV caches, MLA, cascade attention, state merging, fused
+norm/rope/activation/sampling ops, MFMA GEMM (bf16/fp8), fused MoE, and
+RCCL-over-xGMI communication helpers.
+
+Hand-written HIP kernels, one backend per op — no CUDA compatibility layer,
+no CUTLASS/cuDNN/Triton dispatch.
+"""
+
+__version__ = "0.1.0"
+
+from .activation import gelu_and_mul, gelu_tanh_and_mul, silu_and_mul
+from .cascade import merge_state, merge_state_in_place, merge_states
+from .decode import (
+    BatchDecodeWithPagedKVCacheWrapper,
+    CUDAGraphBatchDecodeWithPagedKVCacheWrapper,
+    single_decode_with_kv_cache,
+)
+from .norm import (
+    fused_add_rmsnorm,
+    gemma_fused_add_rmsnorm,
+    gemma_rmsnorm,
+    layernorm,
+    rmsnorm,
+)
+from .page import append_paged_kv_cache, get_batch_indices_positions, get_seq_lens
+from .rope import (
+    apply_llama31_rope,
+    apply_llama31_rope_inplace,
+    apply_llama31_rope_pos_ids,
+    apply_llama31_rope_pos_ids_inplace,
+    apply_rope,
+    apply_rope_inplace,
+    apply_rope_pos_ids,
+    apply_rope_pos_ids_inplace,
+    apply_rope_with_cos_sin_cache,
+    apply_rope_with_cos_sin_cache_inplace,
+)
+
+__all__ = [  # noqa: F405
+    "BatchDecodeWithPagedKVCacheWrapper",
+    "CUDAGraphBatchDecodeWithPagedKVCacheWrapper",
+    "single_decode_with_kv_cache",
+    "merge_state",
+    "merge_state_in_place",
+    "merge_states",
+    "rmsnorm",
+    "gemma_rmsnorm",
+    "fused_add_rmsnorm",
+    "gemma_fused_add_rmsnorm",
+    "layernorm",
+    "silu_and_mul",
+    "gelu_and_mul",
+    "gelu_tanh_and_mul",
+    "append_paged_kv_cache",
+    "get_batch_indices_positions",
+    "get_seq_lens",
+    "apply_rope",
+    "apply_rope_inplace",
+    "apply_rope_pos_ids",
+    "apply_rope_pos_ids_inplace",
+    "apply_llama31_rope",
+    "apply_llama31_rope_inplace",
+    "apply_llama31_rope_pos_ids",
+    "apply_llama31_rope_pos_ids_inplace",
+    "apply_rope_with_cos_sin_cache",
+    "apply_rope_with_cos_sin_cache_inplace",
+]

@@ -1,0 +1,198 @@
+// Batch decode attention over paged KV — the core serving decode path.
+// Functional parity with reference include/flashinfer/attention/decode.cuh
+// (BatchDecodeWithPagedKVCacheKernel:615) + scheduler split-KV, re-designed
+// for CDNA4:
+//   * 1-wave (64-lane) workgroups: lanes split into 16-lane columns x 4 kv
+//     tokens (head_dim 128; 8/64 and 32/2 for 64/256), 16 B loads per lane.
+//   * GQA group (1/2/4/8 q heads per kv head) processed in-register against
+//     one K read — the K/V bytes are the bound, q reuse is free.
+//   * split-KV via host-planned work items (req, chunk); partials are written
+//     f32 to workspace and always reduced by the merge kernel (count==1 is a
+//     plain normalize+cast) — keeps the kernel hipGraph-capturable with a
+//     fixed grid.
+//   * online-softmax state merged across the wave by shfl_xor state exchange.
+// Single-request decode reuses this kernel (batch=1, one/few work items).
+#include "fi/common.hpp"
+#include "fi/page.hpp"
+#include "fi/state.hpp"
+#include "fi/vec.hpp"
+
+namespace fi {
+
+struct DecodeParams {
+  const void* __restrict__ q;  // [n_req, Hq, D]
+  // paged kv raw fields
+  void* k_data;
+  void* v_data;
+  const int32_t* kv_indices;
+  const int32_t* kv_indptr;
+  const int32_t* kv_last_page_len;
+  uint_fastdiv page_size;
+  int num_kv_heads, num_qo_heads, head_dim;
+  int64_t stride_page, stride_n, stride_h;
+  // work items (host plan)
+  const int32_t* work_req;   // [n_items]
+  const int32_t* work_chunk; // [n_items]
+  int n_items;
+  int chunk_size;
+  // outputs (always f32 partials, merged afterwards)
+  float* tmp_v;  // [n_items, Hq, D]
+  float* tmp_s;  // [n_items, Hq]
+  int64_t q_stride_n, q_stride_h;
+  float sm_scale;
+  float logits_soft_cap;  // 0 = disabled
+  int window_left;        // -1 = disabled (sliding window)
+};
+
+// VPL = 8 elems (16B bf16) per lane along head_dim; LPT = head_dim/8 lanes
+// per token; TPW = 64/LPT tokens per wave.
+template <typename T, int HEAD_DIM, int GROUP, bool SOFT_CAP>
+__global__ __launch_bounds__(64) void batch_decode_kernel(DecodeParams p) {
+  constexpr int VPL = 8;
+  constexpr int LPT = HEAD_DIM / VPL;        // lanes per token: 8/16/32
+  constexpr int TPW = kWaveSize / LPT;       // tokens per wave: 8/4/2
+  const int lane = threadIdx.x;
+  const int tsub = lane / LPT;               // which token this lane covers
+  const int dcol = (lane % LPT) * VPL;       // feature offset
+
+  int item = blockIdx.x;
+  int kv_head = blockIdx.y;
+  if (item >= p.n_items) return;
+  int req = p.work_req[item];
+  int chunk = p.work_chunk[item];
+
+  int np = p.kv_indptr[req + 1] - p.kv_indptr[req];
+  int64_t kv_len = np == 0 ? 0 : (int64_t)(np - 1) * p.page_size.d + p.kv_last_page_len[req];
+  int64_t start = (int64_t)chunk * p.chunk_size;
+  int64_t end = start + p.chunk_size;
+  if (end > kv_len) end = kv_len;
+  if (p.window_left >= 0) {
+    int64_t w_start = kv_len - 1 - p.window_left;
+    if (w_start > start) start = w_start;
+  }
+
+  // stage q (pre-scaled) into registers: GROUP heads x VPL slice
+  float qreg[GROUP][VPL];
+  const T* qbase = (const T*)p.q + (int64_t)req * p.q_stride_n;
+  const float scale = p.sm_scale;
+#pragma unroll
+  for (int g = 0; g < GROUP; ++g) {
+    int qh = kv_head * GROUP + g;
+    vec_t<T, VPL> qv;
+    qv.load(qbase + (int64_t)qh * p.q_stride_h + dcol);
+#pragma unroll
+    for (int j = 0; j < VPL; ++j) qreg[g][j] = qv.get(j) * scale;
+  }
+
+  state_t<VPL> st[GROUP];
+#pragma unroll
+  for (int g = 0; g < GROUP; ++g) st[g].init();
+
+  const int32_t* page_ids = p.kv_indices + p.kv_indptr[req];
+  const T* kbase = (const T*)p.k_data;
+  const T* vbase = (const T*)p.v_data;
+
+  for (int64_t pos0 = start; pos0 < end; pos0 += TPW) {
+    int64_t pos = pos0 + tsub;
+    bool valid = pos < end;
+    int64_t ppos = valid ? pos : (end - 1);
+    uint32_t page_iter, entry;
+    p.page_size.divmod((uint32_t)ppos, page_iter, entry);
+    int64_t off = (int64_t)page_ids[page_iter] * p.stride_page +
+                  (int64_t)kv_head * p.stride_h + (int64_t)entry * p.stride_n + dcol;
+    vec_t<T, VPL> kv, vv;
+    kv.load(kbase + off);
+    vv.load(vbase + off);
+    float vf[VPL];
+#pragma unroll
+    for (int j = 0; j < VPL; ++j) vf[j] = vv.get(j);
+
+#pragma unroll
+    for (int g = 0; g < GROUP; ++g) {
+      float s = 0.f;
+#pragma unroll
+      for (int j = 0; j < VPL; ++j) s += qreg[g][j] * kv.get(j);
+      // reduce across the LPT lanes of this token
+#pragma unroll
+      for (int off2 = LPT / 2; off2 > 0; off2 >>= 1) s += __shfl_xor(s, off2, 64);
+      if constexpr (SOFT_CAP) s = p.logits_soft_cap * tanhf(s / p.logits_soft_cap);
+      if (!valid) s = -INFINITY;
+      st[g].push(vf, s);
+    }
+  }
+
+  // merge the TPW per-token states across the wave: lanes with equal (lane%LPT)
+  // hold the same output slice; exchange via shfl_xor at widths LPT, 2*LPT, ...
+#pragma unroll
+  for (int g = 0; g < GROUP; ++g) {
+#pragma unroll
+    for (int w = LPT; w < kWaveSize; w <<= 1) {
+      float m_o = __shfl_xor(st[g].m, w, 64);
+      float d_o = __shfl_xor(st[g].d, w, 64);
+      float o_o[VPL];
+#pragma unroll
+      for (int j = 0; j < VPL; ++j) o_o[j] = __shfl_xor(st[g].o[j], w, 64);
+      st[g].merge(o_o, m_o, d_o);
+    }
+  }
+
+  // lanes of token-sub 0 write the result (normalized partial + lse)
+  if (tsub == 0) {
+#pragma unroll
+    for (int g = 0; g < GROUP; ++g) {
+      int qh = kv_head * GROUP + g;
+      float inv_d = st[g].d > 0.f ? 1.f / st[g].d : 0.f;
+      float* vout = p.tmp_v + ((int64_t)item * p.num_qo_heads + qh) * HEAD_DIM + dcol;
+#pragma unroll
+      for (int j = 0; j < VPL; ++j) vout[j] = st[g].o[j] * inv_d;
+      if (lane == 0) p.tmp_s[(int64_t)item * p.num_qo_heads + qh] = st[g].lse() * 1.4426950408889634f;  // base-2
+    }
+  }
+}
+
+template <typename T>
+hipError_t decode_dispatch(DecodeParams& p, hipStream_t stream) {
+  int group = p.num_qo_heads / p.num_kv_heads;
+  dim3 g(p.n_items, p.num_kv_heads), blk(64);
+  bool sc = p.logits_soft_cap > 0.f;
+#define LAUNCH_D(HD, G, SC) \
+  hipLaunchKernelGGL((batch_decode_kernel<T, HD, G, SC>), g, blk, 0, stream, p)
+#define DISPATCH_G(HD, SC)                              \
+  do {                                                  \
+    switch (group) {                                    \
+      case 1: LAUNCH_D(HD, 1, SC); break;               \
+      case 2: LAUNCH_D(HD, 2, SC); break;               \
+      case 4: LAUNCH_D(HD, 4, SC); break;               \
+      case 8: LAUNCH_D(HD, 8, SC); break;               \
+      case 16: LAUNCH_D(HD, 16, SC); break;             \
+      default: return hipErrorInvalidValue;             \
+    }                                                   \
+  } while (0)
+#define DISPATCH_HD(SC)                                 \
+  do {                                                  \
+    switch (p.head_dim) {                               \
+      case 64: DISPATCH_G(64, SC); break;               \
+      case 128: DISPATCH_G(128, SC); break;             \
+      case 256: DISPATCH_G(256, SC); break;             \
+      default: return hipErrorInvalidValue;             \
+    }                                                   \
+  } while (0)
+  if (sc) DISPATCH_HD(true);
+  else DISPATCH_HD(false);
+#undef DISPATCH_HD
+#undef DISPATCH_G
+#undef LAUNCH_D
+  return hipGetLastError();
+}
+
+}  // namespace fi
+
+extern "C" hipError_t fi_batch_decode(int dtype, fi::DecodeParams* p, hipStream_t stream) {
+  if (p->n_items == 0) return hipSuccess;
+  switch (dtype) {
+    case 0: return fi::decode_dispatch<fi::bf16>(*p, stream);
+    case 1: return fi::decode_dispatch<fi::fp16>(*p, stream);
+    case 2: return fi::decode_dispatch<float>(*p, stream);
+  }
+  return hipErrorInvalidValue;
+}

@@ -1,0 +1,351 @@
+// PyTorch bindings for the MI355X kernel library. Single torch-header TU —
+// all device code lives in the .hip TUs and is reached via extern "C"
+// launchers (fast incremental builds: kernel TUs compile in seconds).
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+namespace fi_ext {
+
+struct RopeParams {
+  const void* q;
+  const void* k;
+  void* q_out;
+  void* k_out;
+  const int32_t* pos_ids;
+  const float* cos_sin_cache;
+  int64_t nnz;
+  int num_qo_heads, num_kv_heads;
+  int head_dim, rot_dim;
+  int64_t q_stride_n, q_stride_h, k_stride_n, k_stride_h;
+  int64_t o_q_stride_n, o_q_stride_h, o_k_stride_n, o_k_stride_h;
+  float rope_rcp_scale;
+  float rope_theta;
+  float smooth_a, smooth_b, rcp_factor;
+  bool interleave;
+};
+
+struct DecodeParams {
+  const void* q;
+  void* k_data;
+  void* v_data;
+  const int32_t* kv_indices;
+  const int32_t* kv_indptr;
+  const int32_t* kv_last_page_len;
+  struct { uint32_t d, m, s, a; } page_size;
+  int num_kv_heads, num_qo_heads, head_dim;
+  int64_t stride_page, stride_n, stride_h;
+  const int32_t* work_req;
+  const int32_t* work_chunk;
+  int n_items;
+  int chunk_size;
+  float* tmp_v;
+  float* tmp_s;
+  int64_t q_stride_n, q_stride_h;
+  float sm_scale;
+  float logits_soft_cap;
+  int window_left;
+};
+
+}  // namespace fi_ext
+
+extern "C" {
+hipError_t fi_norm(int which, int dtype, const void* x, const void* w, const void* b,
+                   void* y, void* residual, int rows, int d, int64_t sx, int64_t sy,
+                   float eps, int weight_bias, hipStream_t stream);
+hipError_t fi_act_and_mul(int which, int dtype, const void* in, void* out, int64_t tokens,
+                          int d, hipStream_t stream);
+hipError_t fi_rope(int dtype, fi_ext::RopeParams* p, hipStream_t stream);
+hipError_t fi_append_paged_kv_cache(int dtype, void* k_data, void* v_data,
+                                    const int32_t* indices, const int32_t* indptr,
+                                    const int32_t* last_page_len, int page_size,
+                                    int num_heads, int head_dim, int64_t stride_page,
+                                    int64_t stride_n, int64_t stride_h, const void* k,
+                                    const void* v, const int32_t* batch_indices,
+                                    const int32_t* positions, int64_t nnz,
+                                    int64_t k_stride_n, int64_t k_stride_h,
+                                    int64_t v_stride_n, int64_t v_stride_h,
+                                    hipStream_t stream);
+hipError_t fi_batch_indices_positions(const int32_t* append_indptr, const int32_t* seq_lens,
+                                      int32_t* batch_indices, int32_t* positions, int batch,
+                                      hipStream_t stream);
+hipError_t fi_merge_states(int in_dtype, int out_dtype, const void* v_in, const float* s_in,
+                           void* v_out, float* s_out, const int32_t* merge_indptr,
+                           int64_t uniform_count, int64_t num_pos, int num_heads,
+                           int head_dim, hipStream_t stream);
+hipError_t fi_merge_state_in_place(int dtype, void* v, float* s, const void* v_other,
+                                   const float* s_other, int64_t num_pos, int num_heads,
+                                   int head_dim, const uint8_t* mask, hipStream_t stream);
+hipError_t fi_batch_decode(int dtype, fi_ext::DecodeParams* p, hipStream_t stream);
+}
+
+namespace {
+
+int dtype_code(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kBFloat16: return 0;
+    case at::kHalf: return 1;
+    case at::kFloat: return 2;
+    default: TORCH_CHECK(false, "unsupported dtype ", t.scalar_type());
+  }
+}
+
+hipStream_t cur_stream(const at::Tensor& t) {
+  return c10::hip::getCurrentHIPStream(t.device().index()).stream();
+}
+
+void check_hip(hipError_t e, const char* what) {
+  TORCH_CHECK(e == hipSuccess, what, " failed: ", hipGetErrorString(e));
+}
+
+// Hacker's Delight magicu — must match fi::uint_fastdiv (fastdiv.hpp).
+void fastdiv_fill(decltype(fi_ext::DecodeParams::page_size)& fd, uint32_t divisor) {
+  fd.d = divisor;
+  fd.a = 0;
+  if (divisor == 1) { fd.m = 0; fd.s = 0; return; }
+  int p = 31;
+  uint32_t nc = (uint32_t)(-1) - ((uint32_t)(-(int32_t)divisor)) % divisor;
+  uint32_t q1 = 0x80000000u / nc, r1 = 0x80000000u - q1 * nc;
+  uint32_t q2 = 0x7FFFFFFFu / divisor, r2 = 0x7FFFFFFFu - q2 * divisor;
+  uint32_t delta;
+  do {
+    p = p + 1;
+    if (r1 >= nc - r1) { q1 = 2 * q1 + 1; r1 = 2 * r1 - nc; }
+    else { q1 = 2 * q1; r1 = 2 * r1; }
+    if (r2 + 1 >= divisor - r2) {
+      if (q2 >= 0x7FFFFFFFu) fd.a = 1;
+      q2 = 2 * q2 + 1;
+      r2 = 2 * r2 + 1 - divisor;
+    } else {
+      if (q2 >= 0x80000000u) fd.a = 1;
+      q2 = 2 * q2;
+      r2 = 2 * r2 + 1;
+    }
+    delta = divisor - 1 - r2;
+  } while (p < 64 && (q1 < delta || (q1 == delta && r1 == 0)));
+  fd.m = q2 + 1;
+  fd.s = p - 32;
+}
+
+// ---------------- norm ----------------
+
+void norm_common(int which, at::Tensor x, at::Tensor w, c10::optional<at::Tensor> b,
+                 c10::optional<at::Tensor> y, c10::optional<at::Tensor> residual,
+                 double eps, bool weight_bias) {
+  TORCH_CHECK(x.is_cuda(), "input must be on GPU");
+  TORCH_CHECK(x.dim() == 2, "expect 2D [rows, hidden]");
+  TORCH_CHECK(x.stride(1) == 1, "innermost dim must be contiguous");
+  int rows = x.size(0), d = x.size(1);
+  const void* bp = b.has_value() ? b->data_ptr() : nullptr;
+  void* yp = y.has_value() ? y->data_ptr() : nullptr;
+  void* rp = residual.has_value() ? residual->data_ptr() : nullptr;
+  int64_t sy = y.has_value() ? y->stride(0) : (residual.has_value() ? residual->stride(0) : 0);
+  check_hip(fi_norm(which, dtype_code(x), x.data_ptr(), w.data_ptr(), bp, yp, rp, rows, d,
+                    x.stride(0), sy, (float)eps, weight_bias, cur_stream(x)),
+            "fi_norm");
+}
+
+void rmsnorm(at::Tensor x, at::Tensor w, at::Tensor out, double eps, bool weight_bias) {
+  norm_common(0, x, w, c10::nullopt, out, c10::nullopt, eps, weight_bias);
+}
+
+void fused_add_rmsnorm(at::Tensor x, at::Tensor residual, at::Tensor w, double eps,
+                       bool weight_bias) {
+  norm_common(1, x, w, c10::nullopt, c10::nullopt, residual, eps, weight_bias);
+}
+
+void layernorm(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> b, at::Tensor out,
+               double eps) {
+  norm_common(2, x, w, b, out, c10::nullopt, eps, false);
+}
+
+// ---------------- activation ----------------
+
+void act_and_mul(int which, at::Tensor x, at::Tensor out) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  int d = x.size(-1) / 2;
+  int64_t tokens = x.numel() / (2 * d);
+  check_hip(fi_act_and_mul(which, dtype_code(x), x.data_ptr(), out.data_ptr(), tokens, d,
+                           cur_stream(x)),
+            "fi_act_and_mul");
+}
+
+// ---------------- rope ----------------
+
+void apply_rope(at::Tensor q, at::Tensor k, at::Tensor q_out, at::Tensor k_out,
+                at::Tensor pos_ids, c10::optional<at::Tensor> cos_sin_cache, int64_t rot_dim,
+                bool interleave, double rope_scale, double rope_theta, double smooth_a,
+                double smooth_b, double rcp_factor) {
+  TORCH_CHECK(q.is_cuda() && k.is_cuda());
+  TORCH_CHECK(q.dim() == 3 && k.dim() == 3, "q/k must be [nnz, heads, head_dim]");
+  TORCH_CHECK(q.stride(2) == 1 && k.stride(2) == 1);
+  TORCH_CHECK(pos_ids.scalar_type() == at::kInt);
+  fi_ext::RopeParams p{};
+  p.q = q.data_ptr();
+  p.k = k.data_ptr();
+  p.q_out = q_out.data_ptr();
+  p.k_out = k_out.data_ptr();
+  p.pos_ids = pos_ids.data_ptr<int32_t>();
+  p.cos_sin_cache = cos_sin_cache.has_value() ? cos_sin_cache->data_ptr<float>() : nullptr;
+  p.nnz = q.size(0);
+  p.num_qo_heads = q.size(1);
+  p.num_kv_heads = k.size(1);
+  p.head_dim = q.size(2);
+  p.rot_dim = (int)rot_dim;
+  p.q_stride_n = q.stride(0);
+  p.q_stride_h = q.stride(1);
+  p.k_stride_n = k.stride(0);
+  p.k_stride_h = k.stride(1);
+  p.o_q_stride_n = q_out.stride(0);
+  p.o_q_stride_h = q_out.stride(1);
+  p.o_k_stride_n = k_out.stride(0);
+  p.o_k_stride_h = k_out.stride(1);
+  p.rope_rcp_scale = (float)(1.0 / rope_scale);
+  p.rope_theta = (float)rope_theta;
+  p.smooth_a = (float)smooth_a;
+  p.smooth_b = (float)smooth_b;
+  p.rcp_factor = (float)rcp_factor;
+  p.interleave = interleave;
+  check_hip(fi_rope(dtype_code(q), &p, cur_stream(q)), "fi_rope");
+}
+
+// ---------------- page ----------------
+
+void append_paged_kv_cache(at::Tensor k, at::Tensor v, at::Tensor batch_indices,
+                           at::Tensor positions, at::Tensor k_cache, at::Tensor v_cache,
+                           at::Tensor indices, at::Tensor indptr, at::Tensor last_page_len,
+                           int64_t layout /*0 NHD, 1 HND*/) {
+  TORCH_CHECK(k.is_cuda() && k.dim() == 3);
+  // k_cache: NHD [pages, page_size, H, D] or HND [pages, H, page_size, D]
+  int page_size = layout == 0 ? k_cache.size(1) : k_cache.size(2);
+  int num_heads = layout == 0 ? k_cache.size(2) : k_cache.size(1);
+  int head_dim = k_cache.size(3);
+  int64_t stride_page = k_cache.stride(0);
+  int64_t stride_n = layout == 0 ? k_cache.stride(1) : k_cache.stride(2);
+  int64_t stride_h = layout == 0 ? k_cache.stride(2) : k_cache.stride(1);
+  check_hip(fi_append_paged_kv_cache(
+                dtype_code(k), k_cache.data_ptr(), v_cache.data_ptr(),
+                indices.data_ptr<int32_t>(), indptr.data_ptr<int32_t>(),
+                last_page_len.data_ptr<int32_t>(), page_size, num_heads, head_dim,
+                stride_page, stride_n, stride_h, k.data_ptr(), v.data_ptr(),
+                batch_indices.data_ptr<int32_t>(), positions.data_ptr<int32_t>(), k.size(0),
+                k.stride(0), k.stride(1), v.stride(0), v.stride(1), cur_stream(k)),
+            "fi_append_paged_kv_cache");
+}
+
+void get_batch_indices_positions(at::Tensor append_indptr, at::Tensor seq_lens,
+                                 at::Tensor batch_indices, at::Tensor positions) {
+  int batch = append_indptr.size(0) - 1;
+  check_hip(fi_batch_indices_positions(append_indptr.data_ptr<int32_t>(),
+                                       seq_lens.data_ptr<int32_t>(),
+                                       batch_indices.data_ptr<int32_t>(),
+                                       positions.data_ptr<int32_t>(), batch,
+                                       cur_stream(append_indptr)),
+            "fi_batch_indices_positions");
+}
+
+// ---------------- merge ----------------
+
+void merge_states(at::Tensor v_in, at::Tensor s_in, at::Tensor v_out,
+                  c10::optional<at::Tensor> s_out, c10::optional<at::Tensor> merge_indptr,
+                  int64_t uniform_count, int64_t num_pos) {
+  int num_heads = v_out.size(-2);
+  int head_dim = v_out.size(-1);
+  check_hip(fi_merge_states(dtype_code(v_in), dtype_code(v_out), v_in.data_ptr(),
+                            s_in.data_ptr<float>(), v_out.data_ptr(),
+                            s_out.has_value() ? s_out->data_ptr<float>() : nullptr,
+                            merge_indptr.has_value() ? merge_indptr->data_ptr<int32_t>()
+                                                     : nullptr,
+                            uniform_count, num_pos, num_heads, head_dim, cur_stream(v_in)),
+            "fi_merge_states");
+}
+
+void merge_state_in_place(at::Tensor v, at::Tensor s, at::Tensor v_other,
+                          at::Tensor s_other, c10::optional<at::Tensor> mask) {
+  int num_heads = v.size(-2), head_dim = v.size(-1);
+  int64_t num_pos = v.numel() / (num_heads * head_dim);
+  check_hip(fi_merge_state_in_place(dtype_code(v), v.data_ptr(), s.data_ptr<float>(),
+                                    v_other.data_ptr(), s_other.data_ptr<float>(), num_pos,
+                                    num_heads, head_dim,
+                                    mask.has_value() ? mask->data_ptr<uint8_t>() : nullptr,
+                                    cur_stream(v)),
+            "fi_merge_state_in_place");
+}
+
+// ---------------- decode ----------------
+
+void batch_decode_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
+                      at::Tensor kv_indices, at::Tensor kv_indptr,
+                      at::Tensor kv_last_page_len, int64_t layout, at::Tensor work_req,
+                      at::Tensor work_chunk, int64_t chunk_size, at::Tensor tmp_v,
+                      at::Tensor tmp_s, double sm_scale, double logits_soft_cap,
+                      int64_t window_left) {
+  TORCH_CHECK(q.is_cuda() && q.dim() == 3, "q must be [batch, num_qo_heads, head_dim]");
+  fi_ext::DecodeParams p{};
+  p.q = q.data_ptr();
+  p.k_data = k_cache.data_ptr();
+  p.v_data = v_cache.data_ptr();
+  p.kv_indices = kv_indices.data_ptr<int32_t>();
+  p.kv_indptr = kv_indptr.data_ptr<int32_t>();
+  p.kv_last_page_len = kv_last_page_len.data_ptr<int32_t>();
+  int page_size = layout == 0 ? k_cache.size(1) : k_cache.size(2);
+  fastdiv_fill(p.page_size, (uint32_t)page_size);
+  p.num_kv_heads = layout == 0 ? k_cache.size(2) : k_cache.size(1);
+  p.num_qo_heads = q.size(1);
+  p.head_dim = q.size(2);
+  p.stride_page = k_cache.stride(0);
+  p.stride_n = layout == 0 ? k_cache.stride(1) : k_cache.stride(2);
+  p.stride_h = layout == 0 ? k_cache.stride(2) : k_cache.stride(1);
+  p.work_req = work_req.data_ptr<int32_t>();
+  p.work_chunk = work_chunk.data_ptr<int32_t>();
+  p.n_items = work_req.size(0);
+  p.chunk_size = (int)chunk_size;
+  p.tmp_v = tmp_v.data_ptr<float>();
+  p.tmp_s = tmp_s.data_ptr<float>();
+  p.q_stride_n = q.stride(0);
+  p.q_stride_h = q.stride(1);
+  p.sm_scale = (float)sm_scale;
+  p.logits_soft_cap = (float)logits_soft_cap;
+  p.window_left = (int)window_left;
+  check_hip(fi_batch_decode(dtype_code(q), &p, cur_stream(q)), "fi_batch_decode");
+}
+
+// fastdiv self-check (host): returns n // d computed via the magic scheme.
+std::vector<int64_t> debug_fastdiv(int64_t d, std::vector<int64_t> ns) {
+  decltype(fi_ext::DecodeParams::page_size) fd;
+  fastdiv_fill(fd, (uint32_t)d);
+  std::vector<int64_t> out;
+  for (auto n64 : ns) {
+    uint32_t n = (uint32_t)n64;
+    uint32_t q;
+    if (fd.d == 1) q = n;
+    else {
+      q = (uint32_t)(((uint64_t)n * fd.m) >> 32);
+      if (fd.a) {
+        uint32_t t = ((n - q) >> 1) + q;
+        q = t >> (fd.s - 1);
+      } else {
+        q = q >> fd.s;
+      }
+    }
+    out.push_back(q);
+  }
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm);
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
+  m.def("layernorm", &layernorm);
+  m.def("act_and_mul", &act_and_mul);
+  m.def("apply_rope", &apply_rope);
+  m.def("append_paged_kv_cache", &append_paged_kv_cache);
+  m.def("get_batch_indices_positions", &get_batch_indices_positions);
+  m.def("merge_states", &merge_states);
+  m.def("merge_state_in_place", &merge_state_in_place);
+  m.def("batch_decode_run", &batch_decode_run);
+  m.def("debug_fastdiv", &debug_fastdiv);
+}

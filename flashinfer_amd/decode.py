@@ -1,0 +1,312 @@
+"""Decode attention: single-request API + batch plan/run wrappers.
+
+Reference parity: flashinfer/decode.py (single_decode_with_kv_cache:515,
+BatchDecodeWithPagedKVCacheWrapper:751 with plan:1280/run:1851,
+CUDAGraphBatchDecodeWithPagedKVCacheWrapper:2334).
+
+MI355X design: one hand-written CDNA4 kernel (csrc/attention/batch_decode.hip)
+covers single and batch decode over paged KV. The CPU planner splits each
+request's KV into chunks sized so the launch fills the 256-CU chip
+(grid = n_items x num_kv_heads 1-wave blocks), partials always go through the
+LSE merge kernel (split-KV merge, reference scheduler.cuh:349 role), and the
+run path is hipGraph-capturable (fixed grids, no allocation, no H2D).
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple, Union
+
+import torch
+
+from ._lib import get_ext
+from .utils import (
+    WorkspaceAllocator,
+    default_sm_scale,
+    layout_code,
+    unpack_paged_kv_cache,
+)
+
+_TARGET_BLOCKS = 2048  # ~8 workgroups per CU on the 256-CU chip
+
+
+def _plan_chunks(kv_lens, num_kv_heads: int, page_size: int):
+    """Pick a split-KV chunk size and emit work items.
+
+    Returns (chunk_size, work_req, work_chunk, merge_indptr) as Python lists.
+    """
+    batch = len(kv_lens)
+    total = sum(kv_lens)
+    max_len = max(kv_lens) if batch else 0
+    items_target = max(batch, _TARGET_BLOCKS // max(1, num_kv_heads))
+    chunk = max(256, page_size, math.ceil(total / max(1, items_target)))
+    # round up to page multiple so chunks don't straddle partially-read pages
+    chunk = (chunk + page_size - 1) // page_size * page_size
+    work_req, work_chunk, merge_indptr = [], [], [0]
+    for b, L in enumerate(kv_lens):
+        n = max(1, math.ceil(L / chunk))
+        for c in range(n):
+            work_req.append(b)
+            work_chunk.append(c)
+        merge_indptr.append(merge_indptr[-1] + n)
+    return chunk, work_req, work_chunk, merge_indptr
+
+
+class BatchDecodeWithPagedKVCacheWrapper:
+    r"""Batch decode attention over a paged KV cache (plan/run API)."""
+
+    def __init__(
+        self,
+        float_workspace_buffer: torch.Tensor,
+        kv_layout: str = "NHD",
+        use_cuda_graph: bool = False,
+        use_tensor_cores: bool = False,
+        paged_kv_indptr_buffer: Optional[torch.Tensor] = None,
+        paged_kv_indices_buffer: Optional[torch.Tensor] = None,
+        paged_kv_last_page_len_buffer: Optional[torch.Tensor] = None,
+        backend: str = "fa2",
+        jit_args=None,
+    ) -> None:
+        self._float_workspace_buffer = float_workspace_buffer
+        self.device = float_workspace_buffer.device
+        self._int_workspace_buffer = torch.empty(
+            8 * 1024 * 1024, dtype=torch.uint8, device=self.device
+        )
+        self._kv_layout = kv_layout
+        self._use_cuda_graph = use_cuda_graph
+        self._use_tensor_cores = use_tensor_cores
+        self._fixed_indptr_buf = paged_kv_indptr_buffer
+        self._fixed_indices_buf = paged_kv_indices_buffer
+        self._fixed_last_page_len_buf = paged_kv_last_page_len_buffer
+        self._plan_info = None
+
+    @property
+    def is_cuda_graph_enabled(self) -> bool:
+        return self._use_cuda_graph
+
+    def reset_workspace_buffer(
+        self, float_workspace_buffer: torch.Tensor, int_workspace_buffer: torch.Tensor
+    ) -> None:
+        self._float_workspace_buffer = float_workspace_buffer
+        self._int_workspace_buffer = int_workspace_buffer
+
+    def plan(
+        self,
+        indptr: torch.Tensor,
+        indices: torch.Tensor,
+        last_page_len: torch.Tensor,
+        num_qo_heads: int,
+        num_kv_heads: int,
+        head_dim: int,
+        page_size: int,
+        pos_encoding_mode: str = "NONE",
+        window_left: int = -1,
+        logits_soft_cap: Optional[float] = None,
+        q_data_type: Optional[torch.dtype] = torch.bfloat16,
+        kv_data_type: Optional[torch.dtype] = None,
+        data_type: Optional[torch.dtype] = None,
+        sm_scale: Optional[float] = None,
+        rope_scale: Optional[float] = None,
+        rope_theta: Optional[float] = None,
+        non_blocking: bool = True,
+        **kwargs,
+    ) -> None:
+        if pos_encoding_mode not in ("NONE",):
+            raise NotImplementedError(
+                "fused pos encodings in decode not supported; apply RoPE beforehand"
+            )
+        batch = indptr.shape[0] - 1
+        indptr_h = indptr.to("cpu", torch.int64)
+        lp_h = last_page_len.to("cpu", torch.int64)
+        np_ = indptr_h[1:] - indptr_h[:-1]
+        kv_lens = (
+            torch.clamp(np_ - 1, min=0) * page_size
+            + torch.where(np_ > 0, lp_h, torch.zeros_like(lp_h))
+        ).tolist()
+
+        chunk, work_req, work_chunk, merge_indptr = _plan_chunks(
+            kv_lens, num_kv_heads, page_size
+        )
+        n_items = len(work_req)
+
+        dev = self.device
+        # device-resident page table (reuse fixed buffers under CUDA graph)
+        def _to_dev(x, buf):
+            xt = x.to(dev, torch.int32, non_blocking=non_blocking)
+            if buf is not None:
+                buf[: xt.numel()].copy_(xt)
+                return buf[: xt.numel()]
+            return xt
+
+        self._indptr_d = _to_dev(indptr, self._fixed_indptr_buf)
+        self._indices_d = _to_dev(indices, self._fixed_indices_buf)
+        self._last_page_len_d = _to_dev(last_page_len, self._fixed_last_page_len_buf)
+
+        meta = torch.tensor(
+            work_req + work_chunk + merge_indptr, dtype=torch.int32
+        ).to(dev, non_blocking=non_blocking)
+        self._work_req_d = meta[:n_items]
+        self._work_chunk_d = meta[n_items : 2 * n_items]
+        self._merge_indptr_d = meta[2 * n_items :]
+
+        alloc = WorkspaceAllocator(self._float_workspace_buffer)
+        self._tmp_v = alloc.alloc(
+            n_items * num_qo_heads * head_dim * 4, torch.float32,
+            (n_items, num_qo_heads, head_dim),
+        )
+        self._tmp_s = alloc.alloc(
+            n_items * num_qo_heads * 4, torch.float32, (n_items, num_qo_heads)
+        )
+
+        self._plan_info = dict(
+            batch=batch, n_items=n_items, chunk=chunk,
+            num_qo_heads=num_qo_heads, num_kv_heads=num_kv_heads,
+            head_dim=head_dim, page_size=page_size,
+            window_left=window_left,
+            logits_soft_cap=float(logits_soft_cap or 0.0),
+            sm_scale=sm_scale if sm_scale is not None else default_sm_scale(head_dim),
+            q_data_type=q_data_type,
+        )
+
+    begin_forward = plan
+
+    def run(
+        self,
+        q: torch.Tensor,
+        paged_kv_cache: Union[torch.Tensor, Tuple[torch.Tensor, torch.Tensor]],
+        q_scale: Optional[float] = None,
+        k_scale: Optional[float] = None,
+        v_scale: Optional[float] = None,
+        out: Optional[torch.Tensor] = None,
+        lse: Optional[torch.Tensor] = None,
+        return_lse: bool = False,
+        **kwargs,
+    ):
+        pi = self._plan_info
+        if pi is None:
+            raise RuntimeError("must call plan() before run()")
+        if q.dim() == 2:
+            q = q.unsqueeze(1)  # [B, D] MQA convenience
+        k_cache, v_cache = unpack_paged_kv_cache(paged_kv_cache, self._kv_layout)
+        sm_scale = pi["sm_scale"]
+        if q_scale is not None:
+            sm_scale *= q_scale
+        if k_scale is not None:
+            sm_scale *= k_scale
+        get_ext().batch_decode_run(
+            q, k_cache, v_cache,
+            self._indices_d, self._indptr_d, self._last_page_len_d,
+            layout_code(self._kv_layout),
+            self._work_req_d, self._work_chunk_d, pi["chunk"],
+            self._tmp_v, self._tmp_s,
+            sm_scale, pi["logits_soft_cap"], pi["window_left"],
+        )
+        if out is None:
+            out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+        if return_lse and lse is None:
+            lse = torch.empty(
+                (pi["batch"], pi["num_qo_heads"]), dtype=torch.float32, device=q.device
+            )
+        get_ext().merge_states(
+            self._tmp_v, self._tmp_s, out, lse, self._merge_indptr_d, 0, pi["batch"]
+        )
+        if v_scale is not None:
+            out = out * v_scale
+        return (out, lse) if return_lse else out
+
+    forward = run
+
+    def end_forward(self) -> None:
+        pass
+
+
+class CUDAGraphBatchDecodeWithPagedKVCacheWrapper(BatchDecodeWithPagedKVCacheWrapper):
+    r"""CUDA-graph (hipGraph) friendly variant: fixed buffers, fixed grids."""
+
+    def __init__(
+        self,
+        workspace_buffer: torch.Tensor,
+        indptr_buffer: torch.Tensor,
+        indices_buffer: torch.Tensor,
+        last_page_len_buffer: torch.Tensor,
+        kv_layout: str = "NHD",
+        use_tensor_cores: bool = False,
+    ) -> None:
+        super().__init__(
+            workspace_buffer, kv_layout, use_cuda_graph=True,
+            use_tensor_cores=use_tensor_cores,
+            paged_kv_indptr_buffer=indptr_buffer,
+            paged_kv_indices_buffer=indices_buffer,
+            paged_kv_last_page_len_buffer=last_page_len_buffer,
+        )
+
+
+def single_decode_with_kv_cache(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    kv_layout: str = "NHD",
+    pos_encoding_mode: str = "NONE",
+    use_tensor_cores: bool = False,
+    q_scale: Optional[float] = None,
+    k_scale: Optional[float] = None,
+    v_scale: Optional[float] = None,
+    window_left: int = -1,
+    logits_soft_cap: Optional[float] = None,
+    sm_scale: Optional[float] = None,
+    rope_scale: Optional[float] = None,
+    rope_theta: Optional[float] = None,
+    return_lse: bool = False,
+):
+    r"""Decode attention for a single request with contiguous KV.
+
+    q: [num_qo_heads, head_dim]; k/v: [kv_len, num_kv_heads, head_dim] (NHD)
+    or [num_kv_heads, kv_len, head_dim] (HND).
+    """
+    if pos_encoding_mode != "NONE":
+        raise NotImplementedError("apply RoPE beforehand")
+    head_dim = q.shape[-1]
+    num_qo_heads = q.shape[0]
+    if kv_layout == "NHD":
+        kv_len, num_kv_heads = k.shape[0], k.shape[1]
+    else:
+        num_kv_heads, kv_len = k.shape[0], k.shape[1]
+    dev = q.device
+    if sm_scale is None:
+        sm_scale = default_sm_scale(head_dim)
+    if q_scale is not None:
+        sm_scale *= q_scale
+    if k_scale is not None:
+        sm_scale *= k_scale
+
+    # view contiguous KV as one giant page
+    k4 = k.unsqueeze(0)
+    v4 = v.unsqueeze(0)
+    items = max(1, min(32, _TARGET_BLOCKS // max(1, num_kv_heads)))
+    chunk = max(256, math.ceil(kv_len / items))
+    n_items = max(1, math.ceil(kv_len / chunk))
+    meta = torch.tensor(
+        [0] * n_items + list(range(n_items)) + [0, n_items], dtype=torch.int32
+    ).to(dev)
+    work_req = meta[:n_items]
+    work_chunk = meta[n_items : 2 * n_items]
+    merge_indptr = meta[2 * n_items :]
+    page_meta = torch.tensor([0, 0, 1, kv_len], dtype=torch.int32).to(dev)
+    indices, indptr, last_page_len = page_meta[:1], page_meta[1:3], page_meta[3:]
+    tmp_v = torch.empty((n_items, num_qo_heads, head_dim), dtype=torch.float32, device=dev)
+    tmp_s = torch.empty((n_items, num_qo_heads), dtype=torch.float32, device=dev)
+    get_ext().batch_decode_run(
+        q.unsqueeze(0), k4, v4, indices, indptr, last_page_len,
+        layout_code(kv_layout), work_req, work_chunk, chunk, tmp_v, tmp_s,
+        sm_scale, float(logits_soft_cap or 0.0), window_left,
+    )
+    out = torch.empty_like(q.unsqueeze(0))
+    lse = (
+        torch.empty((1, num_qo_heads), dtype=torch.float32, device=dev)
+        if return_lse
+        else None
+    )
+    get_ext().merge_states(tmp_v, tmp_s, out, lse, merge_indptr, 0, 1)
+    out = out.squeeze(0)
+    if v_scale is not None:
+        out = out * v_scale
+    return (out, lse.squeeze(0)) if return_lse else out

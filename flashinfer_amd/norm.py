@@ -1,0 +1,75 @@
+"""RMSNorm / LayerNorm ops (reference parity: flashinfer/norm/__init__.py)."""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from ._lib import get_ext
+
+
+def _flat2d(x: torch.Tensor) -> torch.Tensor:
+    return x.reshape(-1, x.shape[-1])
+
+
+def rmsnorm(
+    input: torch.Tensor,
+    weight: torch.Tensor,
+    eps: float = 1e-6,
+    out: Optional[torch.Tensor] = None,
+    enable_pdl: Optional[bool] = None,  # accepted for API parity; no-op on MI355X
+) -> torch.Tensor:
+    r"""Root-mean-square normalization: ``out = input / rms(input) * weight``."""
+    if out is None:
+        out = torch.empty_like(input)
+    get_ext().rmsnorm(_flat2d(input), weight, _flat2d(out), eps, False)
+    return out
+
+
+def gemma_rmsnorm(
+    input: torch.Tensor,
+    weight: torch.Tensor,
+    eps: float = 1e-6,
+    out: Optional[torch.Tensor] = None,
+    enable_pdl: Optional[bool] = None,
+) -> torch.Tensor:
+    r"""Gemma-style RMSNorm: ``out = input / rms(input) * (weight + 1)``."""
+    if out is None:
+        out = torch.empty_like(input)
+    get_ext().rmsnorm(_flat2d(input), weight, _flat2d(out), eps, True)
+    return out
+
+
+def fused_add_rmsnorm(
+    input: torch.Tensor,
+    residual: torch.Tensor,
+    weight: torch.Tensor,
+    eps: float = 1e-6,
+    enable_pdl: Optional[bool] = None,
+) -> None:
+    r"""In-place: ``residual += input; input = rmsnorm(residual) * weight``."""
+    get_ext().fused_add_rmsnorm(_flat2d(input), _flat2d(residual), weight, eps, False)
+
+
+def gemma_fused_add_rmsnorm(
+    input: torch.Tensor,
+    residual: torch.Tensor,
+    weight: torch.Tensor,
+    eps: float = 1e-6,
+    enable_pdl: Optional[bool] = None,
+) -> None:
+    get_ext().fused_add_rmsnorm(_flat2d(input), _flat2d(residual), weight, eps, True)
+
+
+def layernorm(
+    input: torch.Tensor,
+    gemma: torch.Tensor,
+    beta: Optional[torch.Tensor] = None,
+    eps: float = 1e-6,
+    out: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    r"""LayerNorm over the last dimension."""
+    if out is None:
+        out = torch.empty_like(input)
+    get_ext().layernorm(_flat2d(input), gemma, beta, _flat2d(out), eps)
+    return out

@@ -1,0 +1,58 @@
+"""Paged KV-cache ops (reference parity: flashinfer/page.py)."""
+from __future__ import annotations
+
+from typing import Optional, Tuple, Union
+
+import torch
+
+from ._lib import get_ext
+from .utils import layout_code, unpack_paged_kv_cache
+
+
+def get_seq_lens(
+    kv_indptr: torch.Tensor, kv_last_page_len: torch.Tensor, page_size: int
+) -> torch.Tensor:
+    r"""Per-request total KV length from the page table."""
+    np_ = kv_indptr[1:] - kv_indptr[:-1]
+    return torch.clamp(np_ - 1, min=0) * page_size + torch.where(
+        np_ > 0, kv_last_page_len, torch.zeros_like(kv_last_page_len)
+    )
+
+
+def get_batch_indices_positions(
+    append_indptr: torch.Tensor, seq_lens: torch.Tensor, nnz: int
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    r"""Convert per-request append ranges into per-token (batch_idx, position)
+    pairs for :func:`append_paged_kv_cache`."""
+    device = append_indptr.device
+    batch_indices = torch.empty(nnz, dtype=torch.int32, device=device)
+    positions = torch.empty(nnz, dtype=torch.int32, device=device)
+    get_ext().get_batch_indices_positions(
+        append_indptr.to(torch.int32), seq_lens.to(torch.int32), batch_indices, positions
+    )
+    return batch_indices, positions
+
+
+def append_paged_kv_cache(
+    append_key: torch.Tensor,
+    append_value: torch.Tensor,
+    batch_indices: torch.Tensor,
+    positions: torch.Tensor,
+    paged_kv_cache: Union[torch.Tensor, Tuple[torch.Tensor, torch.Tensor]],
+    kv_indices: torch.Tensor,
+    kv_indptr: torch.Tensor,
+    kv_last_page_len: torch.Tensor,
+    kv_layout: str = "NHD",
+) -> None:
+    r"""Scatter ``nnz`` new tokens into the paged cache.
+
+    ``append_key``/``append_value``: [nnz, num_kv_heads, head_dim].
+    """
+    k_cache, v_cache = unpack_paged_kv_cache(paged_kv_cache, kv_layout)
+    get_ext().append_paged_kv_cache(
+        append_key, append_value,
+        batch_indices.to(torch.int32), positions.to(torch.int32),
+        k_cache, v_cache,
+        kv_indices.to(torch.int32), kv_indptr.to(torch.int32),
+        kv_last_page_len.to(torch.int32), layout_code(kv_layout),
+    )

@@ -1,0 +1,89 @@
+"""GPU numerics: norm + activation kernels vs fp32 PyTorch references."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _rmsnorm_ref(x, w, eps, wbias=0.0):
+    xf = x.float()
+    rms = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return xf * rms * (w.float() + wbias)
+
+
+@pytest.mark.parametrize("rows", [1, 17, 999])
+@pytest.mark.parametrize("d", [128, 4096, 4100])
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
+def test_rmsnorm(rows, d, dtype):
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    x = torch.randn(rows, d, dtype=dtype, device="cuda")
+    w = torch.randn(d, dtype=dtype, device="cuda")
+    out = fi.rmsnorm(x, w, eps=1e-6)
+    ref = _rmsnorm_ref(x, w, 1e-6)
+    torch.testing.assert_close(out.float(), ref, atol=2e-2, rtol=2e-2)
+
+
+def test_gemma_rmsnorm():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    x = torch.randn(64, 2048, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(2048, dtype=torch.bfloat16, device="cuda")
+    out = fi.gemma_rmsnorm(x, w, eps=1e-6)
+    ref = _rmsnorm_ref(x, w, 1e-6, wbias=1.0)
+    torch.testing.assert_close(out.float(), ref, atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16])
+def test_fused_add_rmsnorm(dtype):
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    x = torch.randn(33, 1024, dtype=dtype, device="cuda")
+    r = torch.randn(33, 1024, dtype=dtype, device="cuda")
+    w = torch.randn(1024, dtype=dtype, device="cuda")
+    x2, r2 = x.clone(), r.clone()
+    fi.fused_add_rmsnorm(x2, r2, w, eps=1e-6)
+    ref_res = (x.float() + r.float()).to(dtype)
+    ref_out = _rmsnorm_ref(ref_res, w, 1e-6)
+    torch.testing.assert_close(r2.float(), ref_res.float(), atol=1e-2, rtol=1e-2)
+    torch.testing.assert_close(x2.float(), ref_out, atol=2e-2, rtol=2e-2)
+
+
+def test_layernorm():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    x = torch.randn(65, 768, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(768, dtype=torch.bfloat16, device="cuda")
+    b = torch.randn(768, dtype=torch.bfloat16, device="cuda")
+    out = fi.layernorm(x, w, b, eps=1e-5)
+    ref = torch.nn.functional.layer_norm(
+        x.float(), (768,), w.float(), b.float(), 1e-5
+    )
+    torch.testing.assert_close(out.float(), ref, atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.parametrize("act", ["silu", "gelu", "gelu_tanh"])
+@pytest.mark.parametrize("shape", [(7, 256), (1025, 2816 * 2 // 2)])
+def test_act_and_mul(act, shape):
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    rows, d = shape
+    x = torch.randn(rows, 2 * d, dtype=torch.bfloat16, device="cuda")
+    fn = {
+        "silu": fi.silu_and_mul,
+        "gelu": fi.gelu_and_mul,
+        "gelu_tanh": fi.gelu_tanh_and_mul,
+    }[act]
+    out = fn(x)
+    g, u = x.float().chunk(2, dim=-1)
+    ref = {
+        "silu": torch.nn.functional.silu(g),
+        "gelu": torch.nn.functional.gelu(g),
+        "gelu_tanh": torch.nn.functional.gelu(g, approximate="tanh"),
+    }[act] * u
+    torch.testing.assert_close(out.float(), ref, atol=2e-2, rtol=2e-2)
