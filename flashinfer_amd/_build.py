@@ -38,7 +38,7 @@ def _gen_ninja() -> str:
     pyinc = sysconfig.get_paths()["include"]
     common = (
         f"-O3 -std=c++17 -fPIC --offload-arch={OFFLOAD_ARCH} "
-        f"-I{CSRC}/include -DNDEBUG -ffast-math"
+        f"-I{CSRC}/include -DNDEBUG"
     )
     # Kernel TUs never see torch headers; binding TU does.
     bind_flags = (
