@@ -24,6 +24,12 @@ from .norm import (
     layernorm,
     rmsnorm,
 )
+from .prefill import (
+    BatchPrefillWithPagedKVCacheWrapper,
+    BatchPrefillWithRaggedKVCacheWrapper,
+    single_prefill_with_kv_cache,
+)
+from .gemm import mm_bf16
 from .page import append_paged_kv_cache, get_batch_indices_positions, get_seq_lens
 from .rope import (
     apply_llama31_rope,
@@ -39,6 +45,10 @@ from .rope import (
 )
 
 __all__ = [  # noqa: F405
+    "BatchPrefillWithPagedKVCacheWrapper",
+    "BatchPrefillWithRaggedKVCacheWrapper",
+    "single_prefill_with_kv_cache",
+    "mm_bf16",
     "BatchDecodeWithPagedKVCacheWrapper",
     "CUDAGraphBatchDecodeWithPagedKVCacheWrapper",
     "single_decode_with_kv_cache",

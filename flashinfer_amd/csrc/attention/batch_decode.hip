@@ -13,36 +13,13 @@
 //   * online-softmax state merged across the wave by shfl_xor state exchange.
 // Single-request decode reuses this kernel (batch=1, one/few work items).
 #include "fi/common.hpp"
+#include "fi/params.hpp"
 #include "fi/page.hpp"
 #include "fi/state.hpp"
 #include "fi/vec.hpp"
 
 namespace fi {
 
-struct DecodeParams {
-  const void* __restrict__ q;  // [n_req, Hq, D]
-  // paged kv raw fields
-  void* k_data;
-  void* v_data;
-  const int32_t* kv_indices;
-  const int32_t* kv_indptr;
-  const int32_t* kv_last_page_len;
-  uint_fastdiv page_size;
-  int num_kv_heads, num_qo_heads, head_dim;
-  int64_t stride_page, stride_n, stride_h;
-  // work items (host plan)
-  const int32_t* work_req;   // [n_items]
-  const int32_t* work_chunk; // [n_items]
-  int n_items;
-  int chunk_size;
-  // outputs (always f32 partials, merged afterwards)
-  float* tmp_v;  // [n_items, Hq, D]
-  float* tmp_s;  // [n_items, Hq]
-  int64_t q_stride_n, q_stride_h;
-  float sm_scale;
-  float logits_soft_cap;  // 0 = disabled
-  int window_left;        // -1 = disabled (sliding window)
-};
 
 // VPL = 8 elems (16B bf16) per lane along head_dim; LPT = head_dim/8 lanes
 // per token; TPW = 64/LPT tokens per wave.

@@ -1,0 +1,335 @@
+// Batch prefill attention (FA2-class) for gfx950 — the flagship kernel.
+// Functional parity with reference include/flashinfer/attention/prefill.cuh
+// (BatchPrefillWithPagedKVCacheKernel:4134, Ragged:2746, single:2511) but a
+// from-scratch CDNA4 design (guide §B "fused attention prefill" structure):
+//
+//  * 4 waves x 32 packed q-rows = CTA_TILE_Q 128 (GQA rows packed q_pos x
+//    group, reference scheduler packing), KV tile 64.
+//  * swapped QK^T: S^T = mfma32(K_frag, Q_frag) so each lane owns ONE q-row's
+//    logits (col = lane&31) — the online softmax is 16 in-lane values + one
+//    shfl_xor(32) exchange, no cross-lane reduction tree.
+//  * O accumulated TRANSPOSED (O^T = mfma32(V^T_frag, P^T_frag)): the
+//    per-q-row rescale factor is a per-lane scalar.
+//  * K staged row-major [64][D] with XOR swizzle (32-way bank conflict
+//    without it, guide §6 G4); V staged transposed [D][64] with the same
+//    swizzle so PV A-fragments are contiguous 16 B ds reads.
+//  * base-2 softmax domain (v_exp_f32 is exp2 natively): scale2 folds
+//    log2(e); lse output is directly base-2 (library convention).
+//  * causal / sliding-window / logits-soft-cap masking in-register.
+//  * paged (page table) or ragged (contiguous) KV via template.
+// Work items (req, q_tile) are host-planned; grid (n_tiles, num_kv_heads).
+#include "fi/common.hpp"
+#include "fi/params.hpp"
+#include "fi/fastdiv.hpp"
+#include "fi/frag.hpp"
+#include "fi/mfma.hpp"
+#include "fi/vec.hpp"
+
+namespace fi {
+
+
+constexpr int CTA_Q = 128;   // packed q rows per workgroup
+constexpr int KVB = 64;      // kv tile
+constexpr float kLog2e = 1.4426950408889634f;
+
+template <int ROWB>
+__device__ __forceinline__ uint32_t swz_row(uint32_t byte_off) {
+  // XOR swizzle for ROWB-byte rows
+  if constexpr (ROWB == 256) return swz256(byte_off);
+  else if constexpr (ROWB == 128) return swz128(byte_off);
+  else return byte_off;  // 512B rows (D=256): every other row aliases; use 256-swizzle
+}
+
+template <typename T, int HEAD_DIM, bool PAGED, bool CAUSAL>
+__global__ __launch_bounds__(256) void batch_prefill_kernel(PrefillParams p) {
+  constexpr int KCH = HEAD_DIM / 16;  // k-chunks in QK^T
+  constexpr int DT = HEAD_DIM / 32;   // d-tiles in PV / output
+  constexpr int KROWB = HEAD_DIM * 2;        // K tile row bytes
+  constexpr int VROWB = KVB * 2;             // V^T tile row bytes (=128)
+
+  __shared__ T Ks[KVB * HEAD_DIM];
+  __shared__ T Vs[HEAD_DIM * KVB];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int lq = lane & 31;      // this lane's q row within the wave tile
+  const int khalf = (lane >> 5) * 8;
+
+  const int tile = blockIdx.x;
+  const int kv_head = blockIdx.y;
+  if (tile >= p.n_tiles) return;
+  const int req = p.tile_req[tile];
+  const int qstart = p.tile_qstart[tile];  // packed-row offset
+
+  const int qo_begin = p.qo_indptr[req];
+  const int qo_len = p.qo_indptr[req + 1] - qo_begin;
+  const uint32_t group = p.group.d;
+  const int packed_len = qo_len * group;
+
+  int64_t kv_len;
+  const int32_t* page_ids = nullptr;
+  int64_t kv_base = 0;
+  if constexpr (PAGED) {
+    int np = p.kv_indptr[req + 1] - p.kv_indptr[req];
+    kv_len = np == 0 ? 0 : (int64_t)(np - 1) * p.page_size.d + p.kv_last_page_len[req];
+    page_ids = p.kv_indices + p.kv_indptr[req];
+  } else {
+    kv_base = p.kv_indptr[req];
+    kv_len = p.kv_indptr[req + 1] - kv_base;
+  }
+
+  // this wave's q rows: [qstart + wid*32, +32)
+  const int wq0 = qstart + wid * 32;
+  const int my_row = wq0 + lq;          // packed row of this lane
+  uint32_t my_qpos_u, my_g_u;
+  p.group.divmod((uint32_t)(my_row < packed_len ? my_row : 0), my_qpos_u, my_g_u);
+  const int my_qpos = (int)my_qpos_u;
+  const bool row_valid = my_row < packed_len;
+
+  // causal offset: kv position of q row my_qpos's "diagonal"
+  const int64_t diag = kv_len - qo_len;
+
+  // ---- load Q fragments (pre-fold nothing; scale applied post-MFMA) ----
+  using frag = typename mfma_ab_frag<T>::type;
+  frag qf[KCH];
+  {
+    const T* qptr = (const T*)p.q +
+                    (int64_t)(qo_begin + my_qpos) * p.q_stride_n +
+                    (int64_t)(kv_head * group + my_g_u) * p.q_stride_h;
+#pragma unroll
+    for (int c = 0; c < KCH; ++c) {
+      if (row_valid) {
+        qf[c] = *reinterpret_cast<const frag*>(qptr + c * 16 + khalf);
+      } else {
+        qf[c] = frag{};
+      }
+    }
+  }
+
+  // online softmax state (per lane = per q row; m/d in base-2 domain)
+  float m_run = -INFINITY;
+  float d_run = 0.f;
+  floatx16 acc_o[DT];
+#pragma unroll
+  for (int i = 0; i < DT; ++i) acc_o[i] = {};
+
+  const float scale2 = p.sm_scale * kLog2e;
+  const float cap = p.logits_soft_cap;
+  const int wleft = p.window_left;
+
+  // kv range this tile must process (causal upper bound; window lower bound)
+  int64_t kv_hi = kv_len;
+  if constexpr (CAUSAL) {
+    int tile_max_qpos = (qstart + CTA_Q - 1) / (int)group;
+    if (tile_max_qpos >= qo_len) tile_max_qpos = qo_len - 1;
+    int64_t hi = tile_max_qpos + diag + 1;
+    if (hi < kv_hi) kv_hi = hi;
+  }
+  int64_t kv_lo = 0;
+  if (wleft >= 0) {
+    int tile_min_qpos = qstart / (int)group;
+    int64_t lo = tile_min_qpos + diag - wleft;
+    if (lo > 0) kv_lo = lo;
+  }
+
+  const T* kbase = (const T*)p.k_data;
+  const T* vbase = (const T*)p.v_data;
+
+  for (int64_t kv0 = kv_lo; kv0 < kv_hi; kv0 += KVB) {
+    // ---- stage K [KVB][D] (swizzled) and V^T [D][KVB] (swizzled) ----
+    {
+      constexpr int UNITS = KVB * HEAD_DIM / 8;  // 16B units
+      constexpr int ITER = UNITS / 256;
+#pragma unroll
+      for (int it = 0; it < ITER; ++it) {
+        int u = tid + it * 256;
+        int row = u / (HEAD_DIM / 8);       // kv row in tile
+        int chunk = u % (HEAD_DIM / 8);     // 8-elem d chunk
+        int64_t kvpos = kv0 + row;
+        shortx8 kval = {}, vval = {};
+        if (kvpos < kv_len) {
+          int64_t off;
+          if constexpr (PAGED) {
+            uint32_t pg, entry;
+            p.page_size.divmod((uint32_t)kvpos, pg, entry);
+            off = (int64_t)page_ids[pg] * p.kv_stride_page +
+                  (int64_t)kv_head * p.kv_stride_h + (int64_t)entry * p.kv_stride_n +
+                  chunk * 8;
+          } else {
+            off = (kv_base + kvpos) * p.kv_stride_n + (int64_t)kv_head * p.kv_stride_h +
+                  chunk * 8;
+          }
+          kval = *reinterpret_cast<const shortx8*>(kbase + off);
+          vval = *reinterpret_cast<const shortx8*>(vbase + off);
+        }
+        // K: row-major swizzled
+        *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Ks) +
+                                    swz_row<KROWB>(row * KROWB + chunk * 16)) = kval;
+        // V^T: scatter 8 d-elements down a column
+        short vv[8];
+        *reinterpret_cast<shortx8*>(vv) = vval;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int d = chunk * 8 + j;
+          *reinterpret_cast<short*>(reinterpret_cast<char*>(Vs) +
+                                    swz_row<VROWB>(d * VROWB + row * 2)) = vv[j];
+        }
+      }
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int kt = 0; kt < KVB / 32; ++kt) {
+      // ---- S^T = K * Q^T : [32 kv][32 q] ----
+      floatx16 acc_s = {};
+#pragma unroll
+      for (int c = 0; c < KCH; ++c) {
+        frag kfrag = *reinterpret_cast<const frag*>(
+            reinterpret_cast<const char*>(Ks) +
+            swz_row<KROWB>((kt * 32 + lq) * KROWB + (c * 16 + khalf) * 2));
+        acc_s = mfma_ab_frag<T>::mma32(kfrag, qf[c], acc_s);
+      }
+
+      // ---- masking + base-2 logits (16 values, kv row = cd_row(r)) ----
+      float s2[16];
+      const int64_t kvt0 = kv0 + kt * 32;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int64_t kv = kvt0 + mfma32_cd_row(r, lane);
+        float s = acc_s[r] * p.sm_scale;
+        if (cap > 0.f) s = cap * tanhf(s / cap);
+        s *= kLog2e;
+        bool ok = kv < kv_len;
+        if constexpr (CAUSAL) ok &= kv <= my_qpos + diag;
+        if (wleft >= 0) ok &= kv >= my_qpos + diag - wleft;
+        s2[r] = ok ? s : -INFINITY;
+      }
+
+      // ---- online softmax update (per lane; exchange with lane^32) ----
+      float tmax = s2[0];
+#pragma unroll
+      for (int r = 1; r < 16; ++r) tmax = fmaxf(tmax, s2[r]);
+      tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+      float m_new = fmaxf(m_run, tmax);
+      float f = 0.f, psum = 0.f;
+      float pr[16];
+      if (m_new == -INFINITY) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) pr[r] = 0.f;
+        f = 1.f;
+      } else {
+        f = __builtin_exp2f(m_run - m_new);
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          pr[r] = (s2[r] == -INFINITY) ? 0.f : __builtin_exp2f(s2[r] - m_new);
+          psum += pr[r];
+        }
+      }
+      m_run = m_new;
+      d_run = d_run * f + psum;
+#pragma unroll
+      for (int i = 0; i < DT; ++i) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) acc_o[i][r] *= f;
+      }
+
+      // ---- P^T fragments: pack to bf16 pairs, exchange halves ----
+      // own packed words W[j] = (pr[2j], pr[2j+1]) as 2x bf16; partner via
+      // shfl_xor 32. kv order per B-frag (k = (lane>>5)*8 + j).
+      uint32_t W[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        uint32_t lo_ = float_to_bf16(pr[2 * j]);
+        uint32_t hi_ = float_to_bf16(pr[2 * j + 1]);
+        W[j] = lo_ | (hi_ << 16);
+      }
+      uint32_t X[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) X[j] = __shfl_xor((int)W[j], 32, 64);
+      bool hiHalf = (lane >> 5) != 0;
+      // step 0 covers kv kt*32 + [0,16); step 1 covers +[16,32)
+      uint32_t b0[4], b1[4];
+      if (!hiHalf) {
+        b0[0] = W[0]; b0[1] = W[1]; b0[2] = X[0]; b0[3] = X[1];
+        b1[0] = W[4]; b1[1] = W[5]; b1[2] = X[4]; b1[3] = X[5];
+      } else {
+        b0[0] = X[2]; b0[1] = X[3]; b0[2] = W[2]; b0[3] = W[3];
+        b1[0] = X[6]; b1[1] = X[7]; b1[2] = W[6]; b1[3] = W[7];
+      }
+
+      // ---- O^T += V^T * P^T ----
+#pragma unroll
+      for (int i = 0; i < DT; ++i) {
+        frag vf0 = *reinterpret_cast<const frag*>(
+            reinterpret_cast<const char*>(Vs) +
+            swz_row<VROWB>((i * 32 + lq) * VROWB + (kt * 32 + khalf) * 2));
+        frag vf1 = *reinterpret_cast<const frag*>(
+            reinterpret_cast<const char*>(Vs) +
+            swz_row<VROWB>((i * 32 + lq) * VROWB + (kt * 32 + 16 + khalf) * 2));
+        acc_o[i] = mfma_ab_frag<T>::mma32(vf0, *reinterpret_cast<frag*>(b0), acc_o[i]);
+        acc_o[i] = mfma_ab_frag<T>::mma32(vf1, *reinterpret_cast<frag*>(b1), acc_o[i]);
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: normalize and write O (transpose from O^T frags) ----
+  float d_full = d_run + __shfl_xor(d_run, 32, 64);
+  float inv_d = d_full > 0.f ? 1.f / d_full : 0.f;
+  if (row_valid) {
+    T* optr = (T*)p.out + (int64_t)(qo_begin + my_qpos) * p.o_stride_n +
+              (int64_t)(kv_head * group + my_g_u) * p.o_stride_h;
+#pragma unroll
+    for (int i = 0; i < DT; ++i) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int d = i * 32 + mfma32_cd_row(r, lane);
+        optr[d] = from_f32<T>(acc_o[i][r] * inv_d);
+      }
+    }
+    if (p.lse && (lane >> 5) == 0) {
+      float l2 = d_full > 0.f ? m_run + __builtin_log2f(d_full) : -INFINITY;
+      p.lse[(int64_t)(qo_begin + my_qpos) * p.num_qo_heads + kv_head * group + my_g_u] =
+          l2;
+    }
+  }
+}
+
+template <typename T>
+hipError_t prefill_dispatch(PrefillParams& p, bool paged, hipStream_t stream) {
+  dim3 g(p.n_tiles, p.num_kv_heads), blk(256);
+#define LAUNCH_PF(HD, PG, CS) \
+  hipLaunchKernelGGL((batch_prefill_kernel<T, HD, PG, CS>), g, blk, 0, stream, p)
+#define DISPATCH_PC(HD)                                         \
+  do {                                                          \
+    if (paged) {                                                \
+      if (p.causal) LAUNCH_PF(HD, true, true);                  \
+      else LAUNCH_PF(HD, true, false);                          \
+    } else {                                                    \
+      if (p.causal) LAUNCH_PF(HD, false, true);                 \
+      else LAUNCH_PF(HD, false, false);                         \
+    }                                                           \
+  } while (0)
+  switch (p.head_dim) {
+    case 64: DISPATCH_PC(64); break;
+    case 128: DISPATCH_PC(128); break;
+    case 256: DISPATCH_PC(256); break;
+    default: return hipErrorInvalidValue;
+  }
+#undef DISPATCH_PC
+#undef LAUNCH_PF
+  return hipGetLastError();
+}
+
+}  // namespace fi
+
+extern "C" hipError_t fi_batch_prefill(int dtype, fi::PrefillParams* p, int paged,
+                                       hipStream_t stream) {
+  if (p->n_tiles == 0) return hipSuccess;
+  switch (dtype) {
+    case 0: return fi::prefill_dispatch<fi::bf16>(*p, paged, stream);
+    case 1: return fi::prefill_dispatch<fi::fp16>(*p, paged, stream);
+  }
+  return hipErrorInvalidValue;
+}

@@ -6,49 +6,8 @@
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
-namespace fi_ext {
-
-struct RopeParams {
-  const void* q;
-  const void* k;
-  void* q_out;
-  void* k_out;
-  const int32_t* pos_ids;
-  const float* cos_sin_cache;
-  int64_t nnz;
-  int num_qo_heads, num_kv_heads;
-  int head_dim, rot_dim;
-  int64_t q_stride_n, q_stride_h, k_stride_n, k_stride_h;
-  int64_t o_q_stride_n, o_q_stride_h, o_k_stride_n, o_k_stride_h;
-  float rope_rcp_scale;
-  float rope_theta;
-  float smooth_a, smooth_b, rcp_factor;
-  bool interleave;
-};
-
-struct DecodeParams {
-  const void* q;
-  void* k_data;
-  void* v_data;
-  const int32_t* kv_indices;
-  const int32_t* kv_indptr;
-  const int32_t* kv_last_page_len;
-  struct { uint32_t d, m, s, a; } page_size;
-  int num_kv_heads, num_qo_heads, head_dim;
-  int64_t stride_page, stride_n, stride_h;
-  const int32_t* work_req;
-  const int32_t* work_chunk;
-  int n_items;
-  int chunk_size;
-  float* tmp_v;
-  float* tmp_s;
-  int64_t q_stride_n, q_stride_h;
-  float sm_scale;
-  float logits_soft_cap;
-  int window_left;
-};
-
-}  // namespace fi_ext
+#include "fi/params.hpp"
+namespace fi_ext = fi;
 
 extern "C" {
 hipError_t fi_norm(int which, int dtype, const void* x, const void* w, const void* b,
@@ -78,6 +37,11 @@ hipError_t fi_merge_state_in_place(int dtype, void* v, float* s, const void* v_o
                                    const float* s_other, int64_t num_pos, int num_heads,
                                    int head_dim, const uint8_t* mask, hipStream_t stream);
 hipError_t fi_batch_decode(int dtype, fi_ext::DecodeParams* p, hipStream_t stream);
+hipError_t fi_gemm_nt(int dtype, const void* A, const void* B, void* C, int M, int N,
+                      int K, int64_t lda, int64_t ldb, int64_t ldc, float alpha,
+                      hipStream_t stream);
+hipError_t fi_batch_prefill(int dtype, fi_ext::PrefillParams* p, int paged,
+                            hipStream_t stream);
 }
 
 namespace {
@@ -99,34 +63,6 @@ void check_hip(hipError_t e, const char* what) {
   TORCH_CHECK(e == hipSuccess, what, " failed: ", hipGetErrorString(e));
 }
 
-// Hacker's Delight magicu — must match fi::uint_fastdiv (fastdiv.hpp).
-void fastdiv_fill(decltype(fi_ext::DecodeParams::page_size)& fd, uint32_t divisor) {
-  fd.d = divisor;
-  fd.a = 0;
-  if (divisor == 1) { fd.m = 0; fd.s = 0; return; }
-  int p = 31;
-  uint32_t nc = (uint32_t)(-1) - ((uint32_t)(-(int32_t)divisor)) % divisor;
-  uint32_t q1 = 0x80000000u / nc, r1 = 0x80000000u - q1 * nc;
-  uint32_t q2 = 0x7FFFFFFFu / divisor, r2 = 0x7FFFFFFFu - q2 * divisor;
-  uint32_t delta;
-  do {
-    p = p + 1;
-    if (r1 >= nc - r1) { q1 = 2 * q1 + 1; r1 = 2 * r1 - nc; }
-    else { q1 = 2 * q1; r1 = 2 * r1; }
-    if (r2 + 1 >= divisor - r2) {
-      if (q2 >= 0x7FFFFFFFu) fd.a = 1;
-      q2 = 2 * q2 + 1;
-      r2 = 2 * r2 + 1 - divisor;
-    } else {
-      if (q2 >= 0x80000000u) fd.a = 1;
-      q2 = 2 * q2;
-      r2 = 2 * r2 + 1;
-    }
-    delta = divisor - 1 - r2;
-  } while (p < 64 && (q1 < delta || (q1 == delta && r1 == 0)));
-  fd.m = q2 + 1;
-  fd.s = p - 32;
-}
 
 // ---------------- norm ----------------
 
@@ -290,7 +226,7 @@ void batch_decode_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   p.kv_indptr = kv_indptr.data_ptr<int32_t>();
   p.kv_last_page_len = kv_last_page_len.data_ptr<int32_t>();
   int page_size = layout == 0 ? k_cache.size(1) : k_cache.size(2);
-  fastdiv_fill(p.page_size, (uint32_t)page_size);
+  p.page_size = fi::uint_fastdiv((uint32_t)page_size);
   p.num_kv_heads = layout == 0 ? k_cache.size(2) : k_cache.size(1);
   p.num_qo_heads = q.size(1);
   p.head_dim = q.size(2);
@@ -311,26 +247,88 @@ void batch_decode_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   check_hip(fi_batch_decode(dtype_code(q), &p, cur_stream(q)), "fi_batch_decode");
 }
 
+// ---------------- prefill ----------------
+
+// paged: k_cache/v_cache are 4-D cache planes, kv_indptr is a page range.
+// ragged: k_cache/v_cache are [nnz_kv, Hkv, D], kv_indptr is token offsets.
+void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
+                       at::Tensor qo_indptr, c10::optional<at::Tensor> kv_indices,
+                       at::Tensor kv_indptr, c10::optional<at::Tensor> kv_last_page_len,
+                       int64_t layout, at::Tensor tile_req, at::Tensor tile_qstart,
+                       at::Tensor out, c10::optional<at::Tensor> lse, double sm_scale,
+                       double logits_soft_cap, int64_t window_left, bool causal,
+                       bool paged) {
+  TORCH_CHECK(q.is_cuda() && q.dim() == 3, "q must be [nnz, Hq, D]");
+  TORCH_CHECK(q.stride(2) == 1 && out.stride(2) == 1);
+  fi_ext::PrefillParams p{};
+  p.q = q.data_ptr();
+  p.out = out.data_ptr();
+  p.lse = lse.has_value() ? lse->data_ptr<float>() : nullptr;
+  p.qo_indptr = qo_indptr.data_ptr<int32_t>();
+  p.k_data = k_cache.data_ptr();
+  p.v_data = v_cache.data_ptr();
+  p.kv_indices = kv_indices.has_value() ? kv_indices->data_ptr<int32_t>() : nullptr;
+  p.kv_indptr = kv_indptr.data_ptr<int32_t>();
+  p.kv_last_page_len =
+      kv_last_page_len.has_value() ? kv_last_page_len->data_ptr<int32_t>() : nullptr;
+  int num_kv_heads, page_size;
+  if (paged) {
+    TORCH_CHECK(k_cache.dim() == 4);
+    page_size = layout == 0 ? k_cache.size(1) : k_cache.size(2);
+    num_kv_heads = layout == 0 ? k_cache.size(2) : k_cache.size(1);
+    p.kv_stride_page = k_cache.stride(0);
+    p.kv_stride_n = layout == 0 ? k_cache.stride(1) : k_cache.stride(2);
+    p.kv_stride_h = layout == 0 ? k_cache.stride(2) : k_cache.stride(1);
+  } else {
+    TORCH_CHECK(k_cache.dim() == 3);
+    page_size = 1;
+    num_kv_heads = k_cache.size(1);
+    p.kv_stride_page = 0;
+    p.kv_stride_n = k_cache.stride(0);
+    p.kv_stride_h = k_cache.stride(1);
+  }
+  p.page_size = fi::uint_fastdiv((uint32_t)page_size);
+  p.tile_req = tile_req.data_ptr<int32_t>();
+  p.tile_qstart = tile_qstart.data_ptr<int32_t>();
+  p.n_tiles = tile_req.size(0);
+  p.num_qo_heads = q.size(1);
+  p.num_kv_heads = num_kv_heads;
+  p.head_dim = q.size(2);
+  TORCH_CHECK(p.num_qo_heads % num_kv_heads == 0);
+  p.group = fi::uint_fastdiv((uint32_t)(p.num_qo_heads / num_kv_heads));
+  p.q_stride_n = q.stride(0);
+  p.q_stride_h = q.stride(1);
+  p.o_stride_n = out.stride(0);
+  p.o_stride_h = out.stride(1);
+  p.sm_scale = (float)sm_scale;
+  p.logits_soft_cap = (float)logits_soft_cap;
+  p.window_left = (int)window_left;
+  p.causal = causal ? 1 : 0;
+  check_hip(fi_batch_prefill(dtype_code(q), &p, paged ? 1 : 0, cur_stream(q)),
+            "fi_batch_prefill");
+}
+
+// ---------------- gemm ----------------
+
+// C[M,N] = A[M,K] @ B_nt[N,K]^T ; all K-contiguous row-major.
+void gemm_nt(at::Tensor a, at::Tensor b_nt, at::Tensor c, double alpha) {
+  TORCH_CHECK(a.is_cuda() && a.dim() == 2 && a.stride(1) == 1);
+  TORCH_CHECK(b_nt.dim() == 2 && b_nt.stride(1) == 1,
+              "B must be K-contiguous [N, K] (pass weight / b.t())");
+  TORCH_CHECK(c.dim() == 2 && c.stride(1) == 1);
+  int M = a.size(0), K = a.size(1), N = b_nt.size(0);
+  TORCH_CHECK(b_nt.size(1) == K && c.size(0) == M && c.size(1) == N);
+  check_hip(fi_gemm_nt(dtype_code(a), a.data_ptr(), b_nt.data_ptr(), c.data_ptr(), M, N,
+                       K, a.stride(0), b_nt.stride(0), c.stride(0), (float)alpha,
+                       cur_stream(a)),
+            "fi_gemm_nt");
+}
+
 // fastdiv self-check (host): returns n // d computed via the magic scheme.
 std::vector<int64_t> debug_fastdiv(int64_t d, std::vector<int64_t> ns) {
-  decltype(fi_ext::DecodeParams::page_size) fd;
-  fastdiv_fill(fd, (uint32_t)d);
+  fi::uint_fastdiv fd((uint32_t)d);
   std::vector<int64_t> out;
-  for (auto n64 : ns) {
-    uint32_t n = (uint32_t)n64;
-    uint32_t q;
-    if (fd.d == 1) q = n;
-    else {
-      q = (uint32_t)(((uint64_t)n * fd.m) >> 32);
-      if (fd.a) {
-        uint32_t t = ((n - q) >> 1) + q;
-        q = t >> (fd.s - 1);
-      } else {
-        q = q >> fd.s;
-      }
-    }
-    out.push_back(q);
-  }
+  for (auto n64 : ns) out.push_back(fd.div((uint32_t)n64));
   return out;
 }
 
@@ -347,5 +345,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("merge_states", &merge_states);
   m.def("merge_state_in_place", &merge_state_in_place);
   m.def("batch_decode_run", &batch_decode_run);
+  m.def("gemm_nt", &gemm_nt);
+  m.def("batch_prefill_run", &batch_prefill_run);
   m.def("debug_fastdiv", &debug_fastdiv);
 }

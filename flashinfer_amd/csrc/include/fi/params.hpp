@@ -1,0 +1,74 @@
+// Kernel parameter structs shared between the torch binding TU and the HIP
+// kernel TUs (single definition — no mirrored layouts).
+#pragma once
+#include <stdint.h>
+
+#include "fi/fastdiv.hpp"
+
+namespace fi {
+
+struct RopeParams {
+  const void* q;  // [nnz, Hq, D]
+  const void* k;  // [nnz, Hkv, D]
+  void* q_out;
+  void* k_out;
+  const int32_t* pos_ids;      // [nnz]
+  const float* cos_sin_cache;  // [max_pos, rot_dim] or null
+  int64_t nnz;
+  int num_qo_heads, num_kv_heads;
+  int head_dim, rot_dim;
+  int64_t q_stride_n, q_stride_h, k_stride_n, k_stride_h;
+  int64_t o_q_stride_n, o_q_stride_h, o_k_stride_n, o_k_stride_h;
+  float rope_rcp_scale;
+  float rope_theta;
+  float smooth_a, smooth_b, rcp_factor;
+  bool interleave;
+};
+
+struct DecodeParams {
+  const void* q;  // [n_req, Hq, D]
+  void* k_data;
+  void* v_data;
+  const int32_t* kv_indices;
+  const int32_t* kv_indptr;
+  const int32_t* kv_last_page_len;
+  uint_fastdiv page_size;
+  int num_kv_heads, num_qo_heads, head_dim;
+  int64_t stride_page, stride_n, stride_h;
+  const int32_t* work_req;    // [n_items]
+  const int32_t* work_chunk;  // [n_items]
+  int n_items;
+  int chunk_size;
+  float* tmp_v;  // [n_items, Hq, D]
+  float* tmp_s;  // [n_items, Hq]
+  int64_t q_stride_n, q_stride_h;
+  float sm_scale;
+  float logits_soft_cap;  // 0 = disabled
+  int window_left;        // -1 = disabled
+};
+
+struct PrefillParams {
+  const void* q;  // [nnz_q, Hq, D]
+  void* out;      // [nnz_q, Hq, D]
+  float* lse;     // [nnz_q, Hq] base-2, optional
+  const int32_t* qo_indptr;  // [batch+1] token offsets
+  void* k_data;
+  void* v_data;
+  const int32_t* kv_indices;
+  const int32_t* kv_indptr;  // paged: page ranges; ragged: token offsets
+  const int32_t* kv_last_page_len;
+  uint_fastdiv page_size;
+  int64_t kv_stride_page, kv_stride_n, kv_stride_h;
+  const int32_t* tile_req;     // [n_tiles]
+  const int32_t* tile_qstart;  // [n_tiles] packed-row offset in request
+  int n_tiles;
+  int num_qo_heads, num_kv_heads, head_dim;
+  uint_fastdiv group;  // Hq / Hkv
+  int64_t q_stride_n, q_stride_h, o_stride_n, o_stride_h;
+  float sm_scale;
+  float logits_soft_cap;  // 0 disabled
+  int window_left;        // -1 disabled
+  int causal;
+};
+
+}  // namespace fi

@@ -10,29 +10,11 @@
 //  * on-the-fly: freqs from rope_theta with optional llama-3.1 wavelength
 //    scaling, computed per (lane, dim-pair) once and reused across tokens.
 #include "fi/common.hpp"
+#include "fi/params.hpp"
 #include "fi/vec.hpp"
 
 namespace fi {
 
-struct RopeParams {
-  const void* q;     // [nnz, Hq, D]
-  const void* k;     // [nnz, Hkv, D]
-  void* q_out;
-  void* k_out;
-  const int32_t* pos_ids;  // [nnz]
-  const float* cos_sin_cache;  // [max_pos, rot_dim] or null
-  int64_t nnz;
-  int num_qo_heads, num_kv_heads;
-  int head_dim, rot_dim;
-  int64_t q_stride_n, q_stride_h, k_stride_n, k_stride_h;
-  int64_t o_q_stride_n, o_q_stride_h, o_k_stride_n, o_k_stride_h;
-  // on-the-fly params
-  float rope_rcp_scale;   // 1/scale (linear position interpolation)
-  float rope_theta;
-  // llama3.1 smoothing (enabled when smooth_a != 0 or smooth_b != 0)
-  float smooth_a, smooth_b, rcp_factor;
-  bool interleave;
-};
 
 // PVEC consecutive rotation pairs per thread.
 template <typename T, int PVEC, bool kUseCache, bool kInterleave>

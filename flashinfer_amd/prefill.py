@@ -1,0 +1,199 @@
+"""Prefill attention: single API + batch plan/run wrappers over paged/ragged
+KV. Reference parity: flashinfer/prefill.py (single_prefill_with_kv_cache:1186,
+BatchPrefillWithPagedKVCacheWrapper:1538, BatchPrefillWithRaggedKVCacheWrapper:
+3193). One hand-written CDNA4 MFMA kernel (csrc/attention/batch_prefill.hip)
+serves all three — GQA rows packed (q_pos x group), CTA tile 128 rows.
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple, Union
+
+import torch
+
+from ._lib import get_ext
+from .utils import default_sm_scale, layout_code, unpack_paged_kv_cache
+
+_CTA_Q = 128  # must match csrc/attention/batch_prefill.hip
+
+
+def _plan_tiles(qo_lens, group: int):
+    tile_req, tile_qstart = [], []
+    for b, L in enumerate(qo_lens):
+        packed = L * group
+        for start in range(0, max(packed, 1), _CTA_Q):
+            tile_req.append(b)
+            tile_qstart.append(start)
+    return tile_req, tile_qstart
+
+
+class _BatchPrefillBase:
+    def __init__(self, float_workspace_buffer, kv_layout="NHD", use_cuda_graph=False,
+                 backend="fa2", jit_args=None, **kwargs):
+        self._float_workspace_buffer = float_workspace_buffer
+        self.device = float_workspace_buffer.device
+        self._kv_layout = kv_layout
+        self._use_cuda_graph = use_cuda_graph
+        self._plan_info = None
+
+    @property
+    def is_cuda_graph_enabled(self):
+        return self._use_cuda_graph
+
+    def reset_workspace_buffer(self, float_workspace_buffer, int_workspace_buffer=None):
+        self._float_workspace_buffer = float_workspace_buffer
+
+    def _plan_common(self, qo_indptr, num_qo_heads, num_kv_heads, head_dim, causal,
+                     sm_scale, window_left, logits_soft_cap, non_blocking=True):
+        qi = qo_indptr.to("cpu", torch.int64)
+        qo_lens = (qi[1:] - qi[:-1]).tolist()
+        group = num_qo_heads // num_kv_heads
+        tile_req, tile_qstart = _plan_tiles(qo_lens, group)
+        n_tiles = len(tile_req)
+        meta = torch.tensor(tile_req + tile_qstart, dtype=torch.int32).to(
+            self.device, non_blocking=non_blocking
+        )
+        self._tile_req = meta[:n_tiles]
+        self._tile_qstart = meta[n_tiles:]
+        self._qo_indptr_d = qo_indptr.to(self.device, torch.int32,
+                                         non_blocking=non_blocking)
+        self._plan_info = dict(
+            num_qo_heads=num_qo_heads, num_kv_heads=num_kv_heads, head_dim=head_dim,
+            causal=causal, window_left=window_left,
+            logits_soft_cap=float(logits_soft_cap or 0.0),
+            sm_scale=sm_scale if sm_scale is not None else default_sm_scale(head_dim),
+            nnz_q=int(qi[-1]),
+        )
+
+    def _run_common(self, q, k_cache, v_cache, kv_indices, kv_indptr, kv_last_page_len,
+                    paged, out, lse, return_lse, k_scale=None, v_scale=None):
+        pi = self._plan_info
+        if pi is None:
+            raise RuntimeError("must call plan() before run()")
+        if out is None:
+            out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+        if return_lse and lse is None:
+            lse = torch.empty(q.shape[0], q.shape[1], dtype=torch.float32,
+                              device=q.device)
+        sm_scale = pi["sm_scale"]
+        if k_scale is not None:
+            sm_scale *= k_scale
+        get_ext().batch_prefill_run(
+            q, k_cache, v_cache, self._qo_indptr_d, kv_indices, kv_indptr,
+            kv_last_page_len, layout_code(self._kv_layout), self._tile_req,
+            self._tile_qstart, out, lse if return_lse else None, sm_scale,
+            pi["logits_soft_cap"], pi["window_left"], pi["causal"], paged,
+        )
+        if v_scale is not None:
+            out = out * v_scale
+        return (out, lse) if return_lse else out
+
+    def end_forward(self):
+        pass
+
+
+class BatchPrefillWithPagedKVCacheWrapper(_BatchPrefillBase):
+    r"""Batch prefill/append attention over a paged KV cache (plan/run)."""
+
+    def plan(
+        self, qo_indptr, paged_kv_indptr, paged_kv_indices, paged_kv_last_page_len,
+        num_qo_heads, num_kv_heads, head_dim_qk, page_size,
+        head_dim_vo=None, custom_mask=None, packed_custom_mask=None,
+        causal: bool = False, pos_encoding_mode: str = "NONE",
+        sm_scale=None, window_left: int = -1, logits_soft_cap=None,
+        rope_scale=None, rope_theta=None, q_data_type=torch.bfloat16,
+        kv_data_type=None, o_data_type=None, non_blocking: bool = True, **kwargs,
+    ):
+        if pos_encoding_mode != "NONE":
+            raise NotImplementedError("apply RoPE beforehand")
+        if custom_mask is not None or packed_custom_mask is not None:
+            raise NotImplementedError("custom masks arrive in a later drop")
+        self._plan_common(qo_indptr, num_qo_heads, num_kv_heads, head_dim_qk, causal,
+                          sm_scale, window_left, logits_soft_cap, non_blocking)
+        self._kv_indptr_d = paged_kv_indptr.to(self.device, torch.int32,
+                                               non_blocking=non_blocking)
+        self._kv_indices_d = paged_kv_indices.to(self.device, torch.int32,
+                                                 non_blocking=non_blocking)
+        self._kv_last_page_len_d = paged_kv_last_page_len.to(
+            self.device, torch.int32, non_blocking=non_blocking)
+
+    begin_forward = plan
+
+    def run(self, q, paged_kv_cache, *args, k_scale=None, v_scale=None, out=None,
+            lse=None, return_lse: bool = False, **kwargs):
+        k_cache, v_cache = unpack_paged_kv_cache(paged_kv_cache, self._kv_layout)
+        return self._run_common(q, k_cache, v_cache, self._kv_indices_d,
+                                self._kv_indptr_d, self._kv_last_page_len_d, True,
+                                out, lse, return_lse, k_scale, v_scale)
+
+    forward = run
+
+
+class BatchPrefillWithRaggedKVCacheWrapper(_BatchPrefillBase):
+    r"""Batch prefill attention over contiguous (ragged) KV (plan/run)."""
+
+    def plan(
+        self, qo_indptr, kv_indptr, num_qo_heads, num_kv_heads, head_dim_qk,
+        head_dim_vo=None, custom_mask=None, packed_custom_mask=None,
+        causal: bool = False, pos_encoding_mode: str = "NONE",
+        sm_scale=None, window_left: int = -1, logits_soft_cap=None,
+        rope_scale=None, rope_theta=None, q_data_type=torch.bfloat16,
+        kv_data_type=None, o_data_type=None, non_blocking: bool = True, **kwargs,
+    ):
+        if pos_encoding_mode != "NONE":
+            raise NotImplementedError("apply RoPE beforehand")
+        if custom_mask is not None or packed_custom_mask is not None:
+            raise NotImplementedError("custom masks arrive in a later drop")
+        self._plan_common(qo_indptr, num_qo_heads, num_kv_heads, head_dim_qk, causal,
+                          sm_scale, window_left, logits_soft_cap, non_blocking)
+        self._kv_indptr_d = kv_indptr.to(self.device, torch.int32,
+                                         non_blocking=non_blocking)
+
+    begin_forward = plan
+
+    def run(self, q, k, v, *args, k_scale=None, v_scale=None, out=None, lse=None,
+            return_lse: bool = False, **kwargs):
+        return self._run_common(q, k, v, None, self._kv_indptr_d, None, False,
+                                out, lse, return_lse, k_scale, v_scale)
+
+    forward = run
+
+
+def single_prefill_with_kv_cache(
+    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+    custom_mask=None, packed_custom_mask=None, causal: bool = False,
+    kv_layout: str = "NHD", pos_encoding_mode: str = "NONE",
+    use_fp16_qk_reduction: bool = False, sm_scale=None, window_left: int = -1,
+    logits_soft_cap=None, rope_scale=None, rope_theta=None,
+    return_lse: bool = False, backend: str = "auto", **kwargs,
+):
+    r"""Prefill attention for one request.
+
+    q: [qo_len, Hq, D]; k/v: [kv_len, Hkv, D] (NHD) or [Hkv, kv_len, D] (HND).
+    """
+    if pos_encoding_mode != "NONE":
+        raise NotImplementedError("apply RoPE beforehand")
+    if custom_mask is not None or packed_custom_mask is not None:
+        raise NotImplementedError("custom masks arrive in a later drop")
+    if kv_layout == "HND":
+        k = k.transpose(0, 1)
+        v = v.transpose(0, 1)
+    qo_len, Hq, D = q.shape
+    kv_len, Hkv, _ = k.shape
+    dev = q.device
+    group = Hq // Hkv
+    tile_req, tile_qstart = _plan_tiles([qo_len], group)
+    n_tiles = len(tile_req)
+    meta = torch.tensor(
+        tile_req + tile_qstart + [0, qo_len, 0, kv_len], dtype=torch.int32
+    ).to(dev)
+    out = torch.empty_like(q)
+    lse = torch.empty(qo_len, Hq, dtype=torch.float32, device=dev) if return_lse else None
+    get_ext().batch_prefill_run(
+        q, k, v, meta[2 * n_tiles : 2 * n_tiles + 2], None,
+        meta[2 * n_tiles + 2 :], None, 0, meta[:n_tiles],
+        meta[n_tiles : 2 * n_tiles], out, lse,
+        sm_scale if sm_scale is not None else default_sm_scale(D),
+        float(logits_soft_cap or 0.0), window_left, causal, False,
+    )
+    return (out, lse) if return_lse else out

@@ -1,0 +1,149 @@
+"""GPU numerics: prefill attention (MFMA kernel) vs fp32 PyTorch reference."""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def ref_attn(q, k, v, causal=False, sm_scale=None, window_left=-1, soft_cap=0.0):
+    # q [M, Hq, D], k/v [L, Hkv, D] -> [M, Hq, D] fp32
+    M, Hq, D = q.shape
+    L, Hkv, _ = k.shape
+    g = Hq // Hkv
+    qf = q.float().transpose(0, 1)  # [Hq, M, D]
+    kf = k.float().repeat_interleave(g, dim=1).transpose(0, 1)  # [Hq, L, D]
+    vf = v.float().repeat_interleave(g, dim=1).transpose(0, 1)
+    scale = sm_scale if sm_scale is not None else 1 / math.sqrt(D)
+    logits = qf @ kf.transpose(-1, -2) * scale  # [Hq, M, L]
+    if soft_cap > 0:
+        logits = soft_cap * torch.tanh(logits / soft_cap)
+    qpos = torch.arange(M, device=q.device)[:, None]
+    kpos = torch.arange(L, device=q.device)[None, :]
+    diag = L - M
+    if causal:
+        logits = logits.masked_fill((kpos > qpos + diag)[None], float("-inf"))
+    if window_left >= 0:
+        logits = logits.masked_fill((kpos < qpos + diag - window_left)[None],
+                                    float("-inf"))
+    p = torch.softmax(logits, dim=-1)
+    return (p @ vf).transpose(0, 1)
+
+
+@pytest.mark.parametrize("qo_len,kv_len", [(1, 1), (17, 17), (128, 128),
+                                           (333, 500), (1024, 1024), (64, 2048)])
+@pytest.mark.parametrize("Hq,Hkv", [(32, 8), (8, 8), (16, 1)])
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("D", [128])
+def test_single_prefill(qo_len, kv_len, Hq, Hkv, causal, D):
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    q = torch.randn(qo_len, Hq, D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(kv_len, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(kv_len, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    out, lse = fi.single_prefill_with_kv_cache(q, k, v, causal=causal,
+                                               return_lse=True)
+    ref = ref_attn(q, k, v, causal=causal)
+    torch.testing.assert_close(out.float(), ref, atol=3e-2, rtol=3e-2)
+    assert lse.isfinite().all()
+
+
+@pytest.mark.parametrize("D", [64, 256])
+def test_single_prefill_head_dims(D):
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    q = torch.randn(200, 8, D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(200, 2, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(200, 2, D, dtype=torch.bfloat16, device="cuda")
+    out = fi.single_prefill_with_kv_cache(q, k, v, causal=True)
+    ref = ref_attn(q, k, v, causal=True)
+    torch.testing.assert_close(out.float(), ref, atol=3e-2, rtol=3e-2)
+
+
+def test_single_prefill_window_softcap():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    q = torch.randn(256, 8, 128, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(300, 2, 128, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(300, 2, 128, dtype=torch.bfloat16, device="cuda")
+    out = fi.single_prefill_with_kv_cache(q, k, v, causal=True, window_left=64,
+                                          logits_soft_cap=20.0)
+    ref = ref_attn(q, k, v, causal=True, window_left=64, soft_cap=20.0)
+    torch.testing.assert_close(out.float(), ref, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("page_size", [1, 16])
+@pytest.mark.parametrize("causal", [False, True])
+def test_batch_prefill_paged(page_size, causal):
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    Hq, Hkv, D = 32, 8, 128
+    qo_lens = [3, 128, 77, 1]
+    kv_lens = [10, 128, 200, 1]
+    batch = len(qo_lens)
+    nnz_q = sum(qo_lens)
+    qo_indptr = torch.tensor([0] + list(torch.cumsum(torch.tensor(qo_lens), 0)),
+                             dtype=torch.int32, device="cuda")
+    pages_per = [(L + page_size - 1) // page_size for L in kv_lens]
+    kv_indptr = torch.tensor([0] + list(torch.cumsum(torch.tensor(pages_per), 0)),
+                             dtype=torch.int32, device="cuda")
+    npages = int(kv_indptr[-1])
+    kv_indices = torch.randperm(npages, dtype=torch.int32, device="cuda")
+    last_page_len = torch.tensor([(L - 1) % page_size + 1 for L in kv_lens],
+                                 dtype=torch.int32, device="cuda")
+    k_cache = torch.randn(npages, page_size, Hkv, D, dtype=torch.bfloat16,
+                          device="cuda")
+    v_cache = torch.randn(npages, page_size, Hkv, D, dtype=torch.bfloat16,
+                          device="cuda")
+    q = torch.randn(nnz_q, Hq, D, dtype=torch.bfloat16, device="cuda")
+
+    ws = torch.empty(16 * 1024 * 1024, dtype=torch.uint8, device="cuda")
+    wrapper = fi.BatchPrefillWithPagedKVCacheWrapper(ws, "NHD")
+    wrapper.plan(qo_indptr, kv_indptr, kv_indices, last_page_len, Hq, Hkv, D,
+                 page_size, causal=causal)
+    out = wrapper.run(q, (k_cache, v_cache))
+
+    for b in range(batch):
+        L = kv_lens[b]
+        toks_k, toks_v = [], []
+        for pi in range(pages_per[b]):
+            page = int(kv_indices[int(kv_indptr[b]) + pi])
+            n = min(page_size, L - pi * page_size)
+            toks_k.append(k_cache[page, :n])
+            toks_v.append(v_cache[page, :n])
+        kk, vv = torch.cat(toks_k), torch.cat(toks_v)
+        qb = q[int(qo_indptr[b]) : int(qo_indptr[b + 1])]
+        ref = ref_attn(qb, kk, vv, causal=causal)
+        torch.testing.assert_close(
+            out[int(qo_indptr[b]) : int(qo_indptr[b + 1])].float(), ref,
+            atol=3e-2, rtol=3e-2, msg=f"req {b}",
+        )
+
+
+def test_batch_prefill_ragged():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    Hq, Hkv, D = 32, 8, 128
+    qo_lens = [64, 1000, 1]
+    kv_lens = qo_lens
+    qo_indptr = torch.tensor([0] + list(torch.cumsum(torch.tensor(qo_lens), 0)),
+                             dtype=torch.int32, device="cuda")
+    kv_indptr = qo_indptr.clone()
+    nnz = sum(qo_lens)
+    q = torch.randn(nnz, Hq, D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(nnz, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(nnz, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(16 * 1024 * 1024, dtype=torch.uint8, device="cuda")
+    wrapper = fi.BatchPrefillWithRaggedKVCacheWrapper(ws, "NHD")
+    wrapper.plan(qo_indptr, kv_indptr, Hq, Hkv, D, causal=True)
+    out = wrapper.run(q, k, v)
+    for b in range(len(qo_lens)):
+        s, e = int(qo_indptr[b]), int(qo_indptr[b + 1])
+        ref = ref_attn(q[s:e], k[s:e], v[s:e], causal=True)
+        torch.testing.assert_close(out[s:e].float(), ref, atol=3e-2, rtol=3e-2)
