@@ -41,11 +41,10 @@ __device__ __forceinline__ uint32_t swz_row(uint32_t byte_off) {
 }
 
 template <typename T, int HEAD_DIM, bool PAGED, bool CAUSAL>
-__global__ __launch_bounds__(256) void batch_prefill_kernel(PrefillParams p) {
+__global__ __launch_bounds__(256, 2) void batch_prefill_kernel(PrefillParams p) {
   constexpr int KCH = HEAD_DIM / 16;  // k-chunks in QK^T
   constexpr int DT = HEAD_DIM / 32;   // d-tiles in PV / output
-  constexpr int KROWB = HEAD_DIM * 2;        // K tile row bytes
-  constexpr int VROWB = KVB * 2;             // V^T tile row bytes (=128)
+  constexpr int KROWB = HEAD_DIM * 2;        // K/V tile row bytes
 
   __shared__ T Ks[KVB * HEAD_DIM];
   __shared__ T Vs[HEAD_DIM * KVB];
@@ -118,13 +117,18 @@ __global__ __launch_bounds__(256) void batch_prefill_kernel(PrefillParams p) {
   const float cap = p.logits_soft_cap;
   const int wleft = p.window_left;
 
-  // kv range this tile must process (causal upper bound; window lower bound)
+  // kv range this tile must process (causal upper bound; window lower bound);
+  // each wave also skips compute past its own rows' causal bound.
   int64_t kv_hi = kv_len;
+  int64_t wave_kv_hi = kv_len;
   if constexpr (CAUSAL) {
     int tile_max_qpos = (qstart + CTA_Q - 1) / (int)group;
     if (tile_max_qpos >= qo_len) tile_max_qpos = qo_len - 1;
     int64_t hi = tile_max_qpos + diag + 1;
     if (hi < kv_hi) kv_hi = hi;
+    int wave_max_qpos = (wq0 + 31) / (int)group;
+    if (wave_max_qpos >= qo_len) wave_max_qpos = qo_len - 1;
+    wave_kv_hi = wave_max_qpos + diag + 1;
   }
   int64_t kv_lo = 0;
   if (wleft >= 0) {
@@ -163,22 +167,17 @@ __global__ __launch_bounds__(256) void batch_prefill_kernel(PrefillParams p) {
           kval = *reinterpret_cast<const shortx8*>(kbase + off);
           vval = *reinterpret_cast<const shortx8*>(vbase + off);
         }
-        // K: row-major swizzled
+        // K: row-major swizzled (vector frag reads); V: row-major plain
+        // (transposed fragments are built with 8 scalar ds reads in PV)
         *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Ks) +
                                     swz_row<KROWB>(row * KROWB + chunk * 16)) = kval;
-        // V^T: scatter 8 d-elements down a column
-        short vv[8];
-        *reinterpret_cast<shortx8*>(vv) = vval;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int d = chunk * 8 + j;
-          *reinterpret_cast<short*>(reinterpret_cast<char*>(Vs) +
-                                    swz_row<VROWB>(d * VROWB + row * 2)) = vv[j];
-        }
+        *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Vs) +
+                                    (uint32_t)(row * KROWB + chunk * 16)) = vval;
       }
     }
     __syncthreads();
 
+    if (kv0 < wave_kv_hi) {
 #pragma unroll
     for (int kt = 0; kt < KVB / 32; ++kt) {
       // ---- S^T = K * Q^T : [32 kv][32 q] ----
@@ -191,29 +190,28 @@ __global__ __launch_bounds__(256) void batch_prefill_kernel(PrefillParams p) {
         acc_s = mfma_ab_frag<T>::mma32(kfrag, qf[c], acc_s);
       }
 
-      // ---- masking + base-2 logits (16 values, kv row = cd_row(r)) ----
-      float s2[16];
+      // ---- masking + base-2 logits -> p values (in place) ----
+      float pr[16];
       const int64_t kvt0 = kv0 + kt * 32;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         int64_t kv = kvt0 + mfma32_cd_row(r, lane);
-        float s = acc_s[r] * p.sm_scale;
-        if (cap > 0.f) s = cap * tanhf(s / cap);
-        s *= kLog2e;
+        float sv = acc_s[r] * p.sm_scale;
+        if (cap > 0.f) sv = cap * tanhf(sv / cap);
+        sv *= kLog2e;
         bool ok = kv < kv_len;
         if constexpr (CAUSAL) ok &= kv <= my_qpos + diag;
         if (wleft >= 0) ok &= kv >= my_qpos + diag - wleft;
-        s2[r] = ok ? s : -INFINITY;
+        pr[r] = ok ? sv : -INFINITY;
       }
 
       // ---- online softmax update (per lane; exchange with lane^32) ----
-      float tmax = s2[0];
+      float tmax = pr[0];
 #pragma unroll
-      for (int r = 1; r < 16; ++r) tmax = fmaxf(tmax, s2[r]);
+      for (int r = 1; r < 16; ++r) tmax = fmaxf(tmax, pr[r]);
       tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
       float m_new = fmaxf(m_run, tmax);
-      float f = 0.f, psum = 0.f;
-      float pr[16];
+      float f, psum = 0.f;
       if (m_new == -INFINITY) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) pr[r] = 0.f;
@@ -222,7 +220,7 @@ __global__ __launch_bounds__(256) void batch_prefill_kernel(PrefillParams p) {
         f = __builtin_exp2f(m_run - m_new);
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          pr[r] = (s2[r] == -INFINITY) ? 0.f : __builtin_exp2f(s2[r] - m_new);
+          pr[r] = (pr[r] == -INFINITY) ? 0.f : __builtin_exp2f(pr[r] - m_new);
           psum += pr[r];
         }
       }
@@ -234,9 +232,7 @@ __global__ __launch_bounds__(256) void batch_prefill_kernel(PrefillParams p) {
         for (int r = 0; r < 16; ++r) acc_o[i][r] *= f;
       }
 
-      // ---- P^T fragments: pack to bf16 pairs, exchange halves ----
-      // own packed words W[j] = (pr[2j], pr[2j+1]) as 2x bf16; partner via
-      // shfl_xor 32. kv order per B-frag (k = (lane>>5)*8 + j).
+      // ---- P^T fragments: pack to bf16 pairs, lazy half-exchange ----
       uint32_t W[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -244,32 +240,37 @@ __global__ __launch_bounds__(256) void batch_prefill_kernel(PrefillParams p) {
         uint32_t hi_ = float_to_bf16(pr[2 * j + 1]);
         W[j] = lo_ | (hi_ << 16);
       }
+      // NOTE: shfl must run on ALL lanes (uniform control flow) — select after.
       uint32_t X[8];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) X[j] = __shfl_xor((int)W[j], 32, 64);
-      bool hiHalf = (lane >> 5) != 0;
-      // step 0 covers kv kt*32 + [0,16); step 1 covers +[16,32)
+      for (int j = 0; j < 8; ++j) X[j] = (uint32_t)__shfl_xor((int)W[j], 32, 64);
+      const bool hiH = (lane >> 5) != 0;
       uint32_t b0[4], b1[4];
-      if (!hiHalf) {
-        b0[0] = W[0]; b0[1] = W[1]; b0[2] = X[0]; b0[3] = X[1];
-        b1[0] = W[4]; b1[1] = W[5]; b1[2] = X[4]; b1[3] = X[5];
-      } else {
-        b0[0] = X[2]; b0[1] = X[3]; b0[2] = W[2]; b0[3] = W[3];
-        b1[0] = X[6]; b1[1] = X[7]; b1[2] = W[6]; b1[3] = W[7];
-      }
+      b0[0] = hiH ? X[2] : W[0];
+      b0[1] = hiH ? X[3] : W[1];
+      b0[2] = hiH ? W[2] : X[0];
+      b0[3] = hiH ? W[3] : X[1];
+      b1[0] = hiH ? X[6] : W[4];
+      b1[1] = hiH ? X[7] : W[5];
+      b1[2] = hiH ? W[6] : X[4];
+      b1[3] = hiH ? W[7] : X[5];
 
-      // ---- O^T += V^T * P^T ----
+      // ---- O^T += V^T * P^T (V^T fragments via 8 scalar ds reads from the
+      // row-major V tile) ----
 #pragma unroll
       for (int i = 0; i < DT; ++i) {
-        frag vf0 = *reinterpret_cast<const frag*>(
-            reinterpret_cast<const char*>(Vs) +
-            swz_row<VROWB>((i * 32 + lq) * VROWB + (kt * 32 + khalf) * 2));
-        frag vf1 = *reinterpret_cast<const frag*>(
-            reinterpret_cast<const char*>(Vs) +
-            swz_row<VROWB>((i * 32 + lq) * VROWB + (kt * 32 + 16 + khalf) * 2));
-        acc_o[i] = mfma_ab_frag<T>::mma32(vf0, *reinterpret_cast<frag*>(b0), acc_o[i]);
-        acc_o[i] = mfma_ab_frag<T>::mma32(vf1, *reinterpret_cast<frag*>(b1), acc_o[i]);
+        T vw0[8], vw1[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          vw0[j] = Vs[(kt * 32 + khalf + j) * HEAD_DIM + i * 32 + lq];
+          vw1[j] = Vs[(kt * 32 + 16 + khalf + j) * HEAD_DIM + i * 32 + lq];
+        }
+        acc_o[i] = mfma_ab_frag<T>::mma32(*reinterpret_cast<frag*>(vw0),
+                                          *reinterpret_cast<frag*>(b0), acc_o[i]);
+        acc_o[i] = mfma_ab_frag<T>::mma32(*reinterpret_cast<frag*>(vw1),
+                                          *reinterpret_cast<frag*>(b1), acc_o[i]);
       }
+    }
     }
     __syncthreads();
   }

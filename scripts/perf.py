@@ -1,0 +1,92 @@
+"""Quick perf sweep: prefill TFLOPS, decode tok/s+TB/s, GEMM TFLOPS."""
+import sys, pathlib
+sys.path.insert(0, str(pathlib.Path(__file__).resolve().parent.parent))
+import math
+import torch
+import flashinfer_amd as fi
+
+
+def timeit(fn, iters=20, warmup=5):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = torch.cuda.Event(True); t1 = torch.cuda.Event(True)
+    t0.record()
+    for _ in range(iters):
+        fn()
+    t1.record()
+    torch.cuda.synchronize()
+    return t0.elapsed_time(t1) / iters * 1e-3  # seconds
+
+
+def bench_prefill(bs=16, s=1024, Hq=32, Hkv=8, D=128, page=16, causal=True):
+    torch.manual_seed(0)
+    qo_lens = [s] * bs
+    kv_lens = [s] * bs
+    qo_indptr = torch.tensor([0] + list(torch.cumsum(torch.tensor(qo_lens), 0)), dtype=torch.int32, device="cuda")
+    pages_per = [(L + page - 1) // page for L in kv_lens]
+    kv_indptr = torch.tensor([0] + list(torch.cumsum(torch.tensor(pages_per), 0)), dtype=torch.int32, device="cuda")
+    npages = int(kv_indptr[-1])
+    kv_indices = torch.randperm(npages, dtype=torch.int32, device="cuda")
+    last_page = torch.tensor([(L - 1) % page + 1 for L in kv_lens], dtype=torch.int32, device="cuda")
+    k_cache = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v_cache = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    q = torch.randn(sum(qo_lens), Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(256 * 1024 * 1024, dtype=torch.uint8, device="cuda")
+    w = fi.BatchPrefillWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(qo_indptr, kv_indptr, kv_indices, last_page, Hq, Hkv, D, page, causal=causal)
+    out = torch.empty_like(q)
+    t = timeit(lambda: w.run(q, (k_cache, v_cache), out=out))
+    # flops: per request: Hq * qo * kv * D * 2 (QK) * 2 (PV) [causal: ~half]
+    fl = 0
+    for qo, kv in zip(qo_lens, kv_lens):
+        full = 2 * 2 * Hq * D * qo * kv
+        fl += full / 2 + 2 * 2 * Hq * D * qo / 2 if causal else full
+    print(f"prefill bs={bs} s={s} causal={causal}: {t*1e6:.1f} us  {fl/t/1e12:.1f} TFLOPS")
+
+
+def bench_decode(bs=256, kv=32768, Hq=32, Hkv=8, D=128, page=16):
+    torch.manual_seed(0)
+    kv_lens = [kv] * bs
+    pages_per = [(L + page - 1) // page for L in kv_lens]
+    kv_indptr = torch.tensor([0] + list(torch.cumsum(torch.tensor(pages_per), 0)), dtype=torch.int32, device="cuda")
+    npages = int(kv_indptr[-1])
+    kv_indices = torch.randperm(npages, dtype=torch.int32, device="cuda")
+    last_page = torch.tensor([(L - 1) % page + 1 for L in kv_lens], dtype=torch.int32, device="cuda")
+    k_cache = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v_cache = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    q = torch.randn(bs, Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(512 * 1024 * 1024, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(kv_indptr, kv_indices, last_page, Hq, Hkv, D, page, q_data_type=torch.bfloat16)
+    out = torch.empty_like(q)
+    t = timeit(lambda: w.run(q, (k_cache, v_cache), out=out))
+    bytes_kv = bs * kv * Hkv * D * 2 * 2
+    print(f"decode bs={bs} kv={kv}: {t*1e6:.1f} us  {bs/t:.0f} tok/s  {bytes_kv/t/1e12:.2f} TB/s")
+
+
+def bench_gemm(N=4096):
+    torch.manual_seed(0)
+    a = torch.randn(N, N, dtype=torch.bfloat16, device="cuda")
+    b = torch.randn(N, N, dtype=torch.bfloat16, device="cuda").t().contiguous().t()
+    c = torch.empty(N, N, dtype=torch.bfloat16, device="cuda")
+    t = timeit(lambda: fi.mm_bf16(a, b, out=c), iters=10)
+    print(f"gemm {N}^3: {t*1e3:.2f} ms  {2*N**3/t/1e12:.0f} TFLOPS")
+    # torch (hipBLASLt) comparison
+    t2 = timeit(lambda: torch.matmul(a, b, out=c), iters=10)
+    print(f"torch {N}^3: {t2*1e3:.2f} ms  {2*N**3/t2/1e12:.0f} TFLOPS")
+
+
+if __name__ == "__main__":
+    which = sys.argv[1] if len(sys.argv) > 1 else "all"
+    if which in ("all", "prefill"):
+        bench_prefill()
+        bench_prefill(bs=1, s=8192)
+        bench_prefill(bs=16, s=1024, causal=False)
+    if which in ("all", "decode"):
+        bench_decode()
+        bench_decode(bs=16, kv=1024)
+        bench_decode(bs=128, kv=4096)
+    if which in ("all", "gemm"):
+        bench_gemm(4096)
+        bench_gemm(8192)
