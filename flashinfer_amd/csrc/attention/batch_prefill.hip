@@ -28,7 +28,6 @@
 namespace fi {
 
 
-constexpr int CTA_Q = 128;   // packed q rows per workgroup
 constexpr int KVB = 64;      // kv tile
 constexpr float kLog2e = 1.4426950408889634f;
 
@@ -40,8 +39,12 @@ __device__ __forceinline__ uint32_t swz_row(uint32_t byte_off) {
   else return byte_off;  // 512B rows (D=256): every other row aliases; use 256-swizzle
 }
 
-template <typename T, int HEAD_DIM, bool PAGED, bool CAUSAL>
-__global__ __launch_bounds__(256, 2) void batch_prefill_kernel(PrefillParams p) {
+// CTAQ = packed q rows per workgroup (128 -> 4 waves, 256 -> 8 waves; the
+// planner picks by average packed length — bigger tiles amortize K/V staging
+// over more q rows).
+template <typename T, int HEAD_DIM, int CTAQ, bool PAGED, bool CAUSAL>
+__global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParams p) {
+  constexpr int NTHREADS = CTAQ * 2;
   constexpr int KCH = HEAD_DIM / 16;  // k-chunks in QK^T
   constexpr int DT = HEAD_DIM / 32;   // d-tiles in PV / output
   constexpr int KROWB = HEAD_DIM * 2;        // K/V tile row bytes
@@ -122,7 +125,7 @@ __global__ __launch_bounds__(256, 2) void batch_prefill_kernel(PrefillParams p) 
   int64_t kv_hi = kv_len;
   int64_t wave_kv_hi = kv_len;
   if constexpr (CAUSAL) {
-    int tile_max_qpos = (qstart + CTA_Q - 1) / (int)group;
+    int tile_max_qpos = (qstart + CTAQ - 1) / (int)group;
     if (tile_max_qpos >= qo_len) tile_max_qpos = qo_len - 1;
     int64_t hi = tile_max_qpos + diag + 1;
     if (hi < kv_hi) kv_hi = hi;
@@ -144,10 +147,10 @@ __global__ __launch_bounds__(256, 2) void batch_prefill_kernel(PrefillParams p) 
     // ---- stage K [KVB][D] (swizzled) and V^T [D][KVB] (swizzled) ----
     {
       constexpr int UNITS = KVB * HEAD_DIM / 8;  // 16B units
-      constexpr int ITER = UNITS / 256;
+      constexpr int ITER = UNITS / NTHREADS;
 #pragma unroll
       for (int it = 0; it < ITER; ++it) {
-        int u = tid + it * 256;
+        int u = tid + it * NTHREADS;
         int row = u / (HEAD_DIM / 8);       // kv row in tile
         int chunk = u % (HEAD_DIM / 8);     // 8-elem d chunk
         int64_t kvpos = kv0 + row;
@@ -299,9 +302,14 @@ __global__ __launch_bounds__(256, 2) void batch_prefill_kernel(PrefillParams p) 
 
 template <typename T>
 hipError_t prefill_dispatch(PrefillParams& p, bool paged, hipStream_t stream) {
-  dim3 g(p.n_tiles, p.num_kv_heads), blk(256);
-#define LAUNCH_PF(HD, PG, CS) \
-  hipLaunchKernelGGL((batch_prefill_kernel<T, HD, PG, CS>), g, blk, 0, stream, p)
+  dim3 g(p.n_tiles, p.num_kv_heads), blk(p.cta_q * 2);
+#define LAUNCH_PF2(HD, CQ, PG, CS) \
+  hipLaunchKernelGGL((batch_prefill_kernel<T, HD, CQ, PG, CS>), g, blk, 0, stream, p)
+#define LAUNCH_PF(HD, PG, CS)                        \
+  do {                                               \
+    if (p.cta_q == 256) LAUNCH_PF2(HD, 256, PG, CS); \
+    else LAUNCH_PF2(HD, 128, PG, CS);                \
+  } while (0)
 #define DISPATCH_PC(HD)                                         \
   do {                                                          \
     if (paged) {                                                \
@@ -320,6 +328,7 @@ hipError_t prefill_dispatch(PrefillParams& p, bool paged, hipStream_t stream) {
   }
 #undef DISPATCH_PC
 #undef LAUNCH_PF
+#undef LAUNCH_PF2
   return hipGetLastError();
 }
 

@@ -257,7 +257,7 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                        int64_t layout, at::Tensor tile_req, at::Tensor tile_qstart,
                        at::Tensor out, c10::optional<at::Tensor> lse, double sm_scale,
                        double logits_soft_cap, int64_t window_left, bool causal,
-                       bool paged) {
+                       bool paged, int64_t cta_q) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 3, "q must be [nnz, Hq, D]");
   TORCH_CHECK(q.stride(2) == 1 && out.stride(2) == 1);
   fi_ext::PrefillParams p{};
@@ -304,6 +304,7 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   p.logits_soft_cap = (float)logits_soft_cap;
   p.window_left = (int)window_left;
   p.causal = causal ? 1 : 0;
+  p.cta_q = (int)cta_q;
   check_hip(fi_batch_prefill(dtype_code(q), &p, paged ? 1 : 0, cur_stream(q)),
             "fi_batch_prefill");
 }

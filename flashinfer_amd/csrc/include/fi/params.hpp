@@ -69,6 +69,7 @@ struct PrefillParams {
   float logits_soft_cap;  // 0 disabled
   int window_left;        // -1 disabled
   int causal;
+  int cta_q;              // 128 or 256 packed q rows per tile
 };
 
 }  // namespace fi

@@ -14,17 +14,19 @@ import torch
 from ._lib import get_ext
 from .utils import default_sm_scale, layout_code, unpack_paged_kv_cache
 
-_CTA_Q = 128  # must match csrc/attention/batch_prefill.hip
-
-
 def _plan_tiles(qo_lens, group: int):
+    """Pick the q-tile size (128 or 256 packed rows) from the average packed
+    length, then emit (req, qstart) work items (reference scheduler.cuh:616
+    role — CTA_TILE_Q from avg packed qo_len)."""
+    packed = [L * group for L in qo_lens]
+    avg = sum(packed) / max(1, len(packed))
+    cta_q = 256 if avg >= 192 else 128
     tile_req, tile_qstart = [], []
-    for b, L in enumerate(qo_lens):
-        packed = L * group
-        for start in range(0, max(packed, 1), _CTA_Q):
+    for b, pk in enumerate(packed):
+        for start in range(0, max(pk, 1), cta_q):
             tile_req.append(b)
             tile_qstart.append(start)
-    return tile_req, tile_qstart
+    return cta_q, tile_req, tile_qstart
 
 
 class _BatchPrefillBase:
@@ -48,7 +50,7 @@ class _BatchPrefillBase:
         qi = qo_indptr.to("cpu", torch.int64)
         qo_lens = (qi[1:] - qi[:-1]).tolist()
         group = num_qo_heads // num_kv_heads
-        tile_req, tile_qstart = _plan_tiles(qo_lens, group)
+        cta_q, tile_req, tile_qstart = _plan_tiles(qo_lens, group)
         n_tiles = len(tile_req)
         meta = torch.tensor(tile_req + tile_qstart, dtype=torch.int32).to(
             self.device, non_blocking=non_blocking
@@ -62,7 +64,7 @@ class _BatchPrefillBase:
             causal=causal, window_left=window_left,
             logits_soft_cap=float(logits_soft_cap or 0.0),
             sm_scale=sm_scale if sm_scale is not None else default_sm_scale(head_dim),
-            nnz_q=int(qi[-1]),
+            nnz_q=int(qi[-1]), cta_q=cta_q,
         )
 
     def _run_common(self, q, k_cache, v_cache, kv_indices, kv_indptr, kv_last_page_len,
@@ -83,6 +85,7 @@ class _BatchPrefillBase:
             kv_last_page_len, layout_code(self._kv_layout), self._tile_req,
             self._tile_qstart, out, lse if return_lse else None, sm_scale,
             pi["logits_soft_cap"], pi["window_left"], pi["causal"], paged,
+            pi["cta_q"],
         )
         if v_scale is not None:
             out = out * v_scale
@@ -182,7 +185,7 @@ def single_prefill_with_kv_cache(
     kv_len, Hkv, _ = k.shape
     dev = q.device
     group = Hq // Hkv
-    tile_req, tile_qstart = _plan_tiles([qo_len], group)
+    cta_q, tile_req, tile_qstart = _plan_tiles([qo_len], group)
     n_tiles = len(tile_req)
     meta = torch.tensor(
         tile_req + tile_qstart + [0, qo_len, 0, kv_len], dtype=torch.int32
@@ -194,6 +197,6 @@ def single_prefill_with_kv_cache(
         meta[2 * n_tiles + 2 :], None, 0, meta[:n_tiles],
         meta[n_tiles : 2 * n_tiles], out, lse,
         sm_scale if sm_scale is not None else default_sm_scale(D),
-        float(logits_soft_cap or 0.0), window_left, causal, False,
+        float(logits_soft_cap or 0.0), window_left, causal, False, cta_q,
     )
     return (out, lse) if return_lse else out
