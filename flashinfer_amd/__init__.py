@@ -30,6 +30,21 @@ from .prefill import (
     single_prefill_with_kv_cache,
 )
 from .gemm import mm_bf16
+from . import sampling
+from .sampling import (
+    chain_speculative_sampling,
+    min_p_sampling_from_probs,
+    sampling_from_logits,
+    sampling_from_probs,
+    softmax,
+    top_k_mask_logits,
+    top_k_renorm_probs,
+    top_k_sampling_from_probs,
+    top_k_top_p_sampling_from_logits,
+    top_k_top_p_sampling_from_probs,
+    top_p_renorm_probs,
+    top_p_sampling_from_probs,
+)
 from .page import append_paged_kv_cache, get_batch_indices_positions, get_seq_lens
 from .rope import (
     apply_llama31_rope,

@@ -69,26 +69,39 @@ __global__ __launch_bounds__(64) void batch_decode_kernel(DecodeParams p) {
   const T* kbase = (const T*)p.k_data;
   const T* vbase = (const T*)p.v_data;
 
-  for (int64_t pos0 = start; pos0 < end; pos0 += TPW) {
+  // two-deep software pipeline: K/V loads for step i+1 are issued before the
+  // softmax/state update of step i so the ~900-cycle HBM latency hides under
+  // the dependent VALU chain (guide T14 async-split idea, register-staged).
+  auto addr_of = [&](int64_t pos0) -> int64_t {
     int64_t pos = pos0 + tsub;
-    bool valid = pos < end;
-    int64_t ppos = valid ? pos : (end - 1);
+    int64_t ppos = pos < end ? pos : (end - 1);
     uint32_t page_iter, entry;
     p.page_size.divmod((uint32_t)ppos, page_iter, entry);
-    int64_t off = (int64_t)page_ids[page_iter] * p.stride_page +
-                  (int64_t)kv_head * p.stride_h + (int64_t)entry * p.stride_n + dcol;
-    vec_t<T, VPL> kv, vv;
-    kv.load(kbase + off);
-    vv.load(vbase + off);
+    return (int64_t)page_ids[page_iter] * p.stride_page +
+           (int64_t)kv_head * p.stride_h + (int64_t)entry * p.stride_n + dcol;
+  };
+  vec_t<T, VPL> kv_cur, vv_cur, kv_nxt, vv_nxt;
+  if (start < end) {
+    int64_t off = addr_of(start);
+    kv_cur.load(kbase + off);
+    vv_cur.load(vbase + off);
+  }
+  for (int64_t pos0 = start; pos0 < end; pos0 += TPW) {
+    if (pos0 + TPW < end) {
+      int64_t off = addr_of(pos0 + TPW);
+      kv_nxt.load(kbase + off);
+      vv_nxt.load(vbase + off);
+    }
+    bool valid = pos0 + tsub < end;
     float vf[VPL];
 #pragma unroll
-    for (int j = 0; j < VPL; ++j) vf[j] = vv.get(j);
+    for (int j = 0; j < VPL; ++j) vf[j] = vv_cur.get(j);
 
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
       float s = 0.f;
 #pragma unroll
-      for (int j = 0; j < VPL; ++j) s += qreg[g][j] * kv.get(j);
+      for (int j = 0; j < VPL; ++j) s += qreg[g][j] * kv_cur.get(j);
       // reduce across the LPT lanes of this token
 #pragma unroll
       for (int off2 = LPT / 2; off2 > 0; off2 >>= 1) s += __shfl_xor(s, off2, 64);
@@ -96,6 +109,8 @@ __global__ __launch_bounds__(64) void batch_decode_kernel(DecodeParams p) {
       if (!valid) s = -INFINITY;
       st[g].push(vf, s);
     }
+    kv_cur = kv_nxt;
+    vv_cur = vv_nxt;
   }
 
   // merge the TPW per-token states across the wave: lanes with equal (lane%LPT)

@@ -42,6 +42,11 @@ hipError_t fi_gemm_nt(int dtype, const void* A, const void* B, void* C, int M, i
                       hipStream_t stream);
 hipError_t fi_batch_prefill(int dtype, fi_ext::PrefillParams* p, int paged,
                             hipStream_t stream);
+hipError_t fi_softmax(fi_ext::SamplingParams* p, hipStream_t stream);
+hipError_t fi_sampling(int mode, int from_logits, fi_ext::SamplingParams* p,
+                       hipStream_t stream);
+hipError_t fi_renorm(int which, fi_ext::SamplingParams* p, hipStream_t stream);
+hipError_t fi_chain_speculative(fi_ext::SpecParams* sp, hipStream_t stream);
 }
 
 namespace {
@@ -309,6 +314,83 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
             "fi_batch_prefill");
 }
 
+// ---------------- sampling ----------------
+
+fi_ext::SamplingParams sampling_params_common(at::Tensor probs,
+                                              c10::optional<at::Tensor> out_probs,
+                                              c10::optional<at::Tensor> out_ids,
+                                              c10::optional<at::Tensor> uniforms,
+                                              c10::optional<at::Tensor> top_k,
+                                              c10::optional<at::Tensor> top_p,
+                                              c10::optional<at::Tensor> row_indices,
+                                              double scalar_p, int64_t scalar_k,
+                                              double min_p) {
+  TORCH_CHECK(probs.is_cuda() && probs.dim() == 2 && probs.stride(1) == 1);
+  TORCH_CHECK(probs.scalar_type() == at::kFloat, "sampling ops take f32");
+  fi_ext::SamplingParams p{};
+  p.probs = probs.data_ptr<float>();
+  p.out_probs = out_probs.has_value() ? out_probs->data_ptr<float>() : nullptr;
+  p.out_ids = out_ids.has_value() ? out_ids->data_ptr<int32_t>() : nullptr;
+  p.uniforms = uniforms.has_value() ? uniforms->data_ptr<float>() : nullptr;
+  p.top_p = top_p.has_value() ? top_p->data_ptr<float>() : nullptr;
+  p.top_k = top_k.has_value() ? top_k->data_ptr<int32_t>() : nullptr;
+  p.row_indices = row_indices.has_value() ? row_indices->data_ptr<int32_t>() : nullptr;
+  p.scalar_p = (float)scalar_p;
+  p.scalar_k = (int)scalar_k;
+  p.min_p = (float)min_p;
+  p.rows = row_indices.has_value() ? row_indices->size(0)
+           : (out_ids.has_value() ? out_ids->size(0) : probs.size(0));
+  p.vocab = probs.size(1);
+  p.rounds = uniforms.has_value() && uniforms->dim() == 2 ? uniforms->size(1) : 1;
+  p.stride_row = probs.stride(0);
+  p.temperature = 1.f;
+  return p;
+}
+
+void softmax_op(at::Tensor logits, at::Tensor out, double temperature) {
+  auto p = sampling_params_common(logits, out, c10::nullopt, c10::nullopt,
+                                  c10::nullopt, c10::nullopt, c10::nullopt, 0, 0, 0);
+  p.temperature = (float)temperature;
+  check_hip(fi_softmax(&p, cur_stream(logits)), "fi_softmax");
+}
+
+void sampling_op(int64_t mode, bool from_logits, at::Tensor probs, at::Tensor out_ids,
+                 at::Tensor uniforms, c10::optional<at::Tensor> top_k,
+                 c10::optional<at::Tensor> top_p, c10::optional<at::Tensor> row_indices,
+                 double scalar_p, int64_t scalar_k, double min_p) {
+  auto p = sampling_params_common(probs, c10::nullopt, out_ids, uniforms, top_k, top_p,
+                                  row_indices, scalar_p, scalar_k, min_p);
+  check_hip(fi_sampling((int)mode, from_logits ? 1 : 0, &p, cur_stream(probs)),
+            "fi_sampling");
+}
+
+void renorm_op(int64_t which, at::Tensor probs, at::Tensor out,
+               c10::optional<at::Tensor> top_k, c10::optional<at::Tensor> top_p,
+               double scalar_p, int64_t scalar_k) {
+  auto p = sampling_params_common(probs, out, c10::nullopt, c10::nullopt, top_k, top_p,
+                                  c10::nullopt, scalar_p, scalar_k, 0);
+  p.rows = probs.size(0);
+  check_hip(fi_renorm((int)which, &p, cur_stream(probs)), "fi_renorm");
+}
+
+void chain_speculative_op(at::Tensor draft_probs, at::Tensor draft_ids,
+                          at::Tensor target_probs, at::Tensor out_ids,
+                          at::Tensor accepted_num, at::Tensor emitted_num,
+                          at::Tensor uniforms) {
+  fi_ext::SpecParams sp{};
+  sp.draft_probs = draft_probs.data_ptr<float>();
+  sp.draft_ids = draft_ids.data_ptr<int32_t>();
+  sp.target_probs = target_probs.data_ptr<float>();
+  sp.out_ids = out_ids.data_ptr<int32_t>();
+  sp.accepted_num = accepted_num.data_ptr<int32_t>();
+  sp.emitted_num = emitted_num.data_ptr<int32_t>();
+  sp.uniforms = uniforms.data_ptr<float>();
+  sp.B = draft_probs.size(0);
+  sp.n = draft_probs.size(1);
+  sp.vocab = draft_probs.size(2);
+  check_hip(fi_chain_speculative(&sp, cur_stream(draft_probs)), "fi_chain_speculative");
+}
+
 // ---------------- gemm ----------------
 
 // C[M,N] = A[M,K] @ B_nt[N,K]^T ; all K-contiguous row-major.
@@ -348,5 +430,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("batch_decode_run", &batch_decode_run);
   m.def("gemm_nt", &gemm_nt);
   m.def("batch_prefill_run", &batch_prefill_run);
+  m.def("softmax", &softmax_op);
+  m.def("sampling", &sampling_op);
+  m.def("renorm", &renorm_op);
+  m.def("chain_speculative", &chain_speculative_op);
   m.def("debug_fastdiv", &debug_fastdiv);
 }

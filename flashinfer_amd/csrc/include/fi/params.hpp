@@ -72,4 +72,33 @@ struct PrefillParams {
   int cta_q;              // 128 or 256 packed q rows per tile
 };
 
+struct SamplingParams {
+  const float* probs;     // [rows, vocab] (or logits)
+  float* out_probs;       // optional output distribution
+  int32_t* out_ids;       // [rows] sampled token
+  const float* uniforms;  // [rows, rounds]
+  const float* top_p;     // [rows] or null (then scalar_p)
+  const int32_t* top_k;   // [rows] or null (then scalar_k)
+  const int32_t* row_indices;  // optional indirection probs row = row_indices[b]
+  float scalar_p;
+  int scalar_k;
+  float min_p;
+  int rows, vocab, rounds;
+  uint64_t stride_row;
+  float temperature;  // for softmax
+};
+
+
+struct SpecParams {
+  const float* draft_probs;   // [B, n, V]
+  const int32_t* draft_ids;   // [B, n]
+  const float* target_probs;  // [B, n+1, V]
+  int32_t* out_ids;           // [B, n+1]
+  int32_t* accepted_num;      // [B] += accepted count
+  int32_t* emitted_num;       // [B] += emitted draft tokens
+  const float* uniforms;      // [B, n+1+rounds]
+  int B, n, vocab;
+};
+
+
 }  // namespace fi

@@ -26,7 +26,7 @@ from .utils import (
     unpack_paged_kv_cache,
 )
 
-_TARGET_BLOCKS = 2048  # ~8 workgroups per CU on the 256-CU chip
+_TARGET_BLOCKS = 8192  # 1-wave blocks: ~32 per CU keeps enough loads in flight
 
 
 def _plan_chunks(kv_lens, num_kv_heads: int, page_size: int):

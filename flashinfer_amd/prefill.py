@@ -14,18 +14,23 @@ import torch
 from ._lib import get_ext
 from .utils import default_sm_scale, layout_code, unpack_paged_kv_cache
 
-def _plan_tiles(qo_lens, group: int):
+def _plan_tiles(qo_lens, group: int, causal: bool = False):
     """Pick the q-tile size (128 or 256 packed rows) from the average packed
     length, then emit (req, qstart) work items (reference scheduler.cuh:616
-    role — CTA_TILE_Q from avg packed qo_len)."""
+    role — CTA_TILE_Q from avg packed qo_len). For causal attention tiles are
+    emitted LONGEST-FIRST (cost grows with qstart) so the hardware dispatch
+    order doesn't leave the expensive diagonal tiles for the tail."""
     packed = [L * group for L in qo_lens]
     avg = sum(packed) / max(1, len(packed))
     cta_q = 256 if avg >= 192 else 128
-    tile_req, tile_qstart = [], []
+    items = []
     for b, pk in enumerate(packed):
         for start in range(0, max(pk, 1), cta_q):
-            tile_req.append(b)
-            tile_qstart.append(start)
+            items.append((b, start))
+    if causal:
+        items.sort(key=lambda t: -t[1])
+    tile_req = [b for b, _ in items]
+    tile_qstart = [st for _, st in items]
     return cta_q, tile_req, tile_qstart
 
 
@@ -50,7 +55,7 @@ class _BatchPrefillBase:
         qi = qo_indptr.to("cpu", torch.int64)
         qo_lens = (qi[1:] - qi[:-1]).tolist()
         group = num_qo_heads // num_kv_heads
-        cta_q, tile_req, tile_qstart = _plan_tiles(qo_lens, group)
+        cta_q, tile_req, tile_qstart = _plan_tiles(qo_lens, group, causal)
         n_tiles = len(tile_req)
         meta = torch.tensor(tile_req + tile_qstart, dtype=torch.int32).to(
             self.device, non_blocking=non_blocking
@@ -185,7 +190,7 @@ def single_prefill_with_kv_cache(
     kv_len, Hkv, _ = k.shape
     dev = q.device
     group = Hq // Hkv
-    cta_q, tile_req, tile_qstart = _plan_tiles([qo_len], group)
+    cta_q, tile_req, tile_qstart = _plan_tiles([qo_len], group, causal)
     n_tiles = len(tile_req)
     meta = torch.tensor(
         tile_req + tile_qstart + [0, qo_len, 0, kv_len], dtype=torch.int32
