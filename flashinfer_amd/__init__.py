@@ -11,7 +11,15 @@ no CUTLASS/cuDNN/Triton dispatch.
 __version__ = "0.1.0"
 
 from .activation import gelu_and_mul, gelu_tanh_and_mul, silu_and_mul
-from .cascade import merge_state, merge_state_in_place, merge_states
+from .cascade import (
+    MultiLevelCascadeAttentionWrapper,
+    merge_state,
+    merge_state_in_place,
+    merge_states,
+)
+from .sparse import BlockSparseAttentionWrapper
+from . import comm
+from .parallel_attention import ParallelAttention, ring_attention
 from .decode import (
     BatchDecodeWithPagedKVCacheWrapper,
     CUDAGraphBatchDecodeWithPagedKVCacheWrapper,

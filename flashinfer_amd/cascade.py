@@ -39,3 +39,62 @@ def merge_states(v: torch.Tensor, s: torch.Tensor) -> Tuple[torch.Tensor, torch.
         v_out, s_out, None, ns, n,
     )
     return v_out, s_out
+
+
+class MultiLevelCascadeAttentionWrapper:
+    r"""Shared-prefix hierarchical (cascade) attention over a unified paged KV
+    cache (reference parity: flashinfer/cascade.py:226). Level 0 holds KV
+    shared by groups of requests (e.g. a common system prompt), deeper levels
+    hold progressively more request-specific KV; per-level attention states
+    are merged by LSE weight. Causal masking applies to the last level only
+    (the unique suffix)."""
+
+    def __init__(self, num_levels: int, float_workspace_buffer: torch.Tensor,
+                 kv_layout: str = "NHD", use_cuda_graph: bool = False):
+        from .prefill import BatchPrefillWithPagedKVCacheWrapper
+
+        self._num_levels = num_levels
+        self._wrappers = [
+            BatchPrefillWithPagedKVCacheWrapper(float_workspace_buffer, kv_layout)
+            for _ in range(num_levels)
+        ]
+        self._kv_layout = kv_layout
+
+    def reset_workspace_buffer(self, float_workspace_buffer, int_workspace_buffers=None):
+        for w in self._wrappers:
+            w.reset_workspace_buffer(float_workspace_buffer)
+
+    def plan(self, qo_indptr_arr, paged_kv_indptr_arr, paged_kv_indices_arr,
+             paged_kv_last_page_len, num_qo_heads, num_kv_heads, head_dim,
+             page_size, causal: bool = False, pos_encoding_mode: str = "NONE",
+             sm_scale=None, window_left: int = -1, logits_soft_cap=None,
+             q_data_type=None, kv_data_type=None, **kwargs):
+        assert len(qo_indptr_arr) == self._num_levels
+        self._qo_indptr_arr = qo_indptr_arr
+        for lvl, w in enumerate(self._wrappers):
+            is_last = lvl == self._num_levels - 1
+            w.plan(
+                qo_indptr_arr[lvl], paged_kv_indptr_arr[lvl],
+                paged_kv_indices_arr[lvl], paged_kv_last_page_len[lvl],
+                num_qo_heads, num_kv_heads, head_dim, page_size,
+                causal=causal and is_last, pos_encoding_mode=pos_encoding_mode,
+                sm_scale=sm_scale, window_left=window_left if is_last else -1,
+                logits_soft_cap=logits_soft_cap, q_data_type=q_data_type,
+            )
+
+    begin_forward = plan
+
+    def run(self, q: torch.Tensor, paged_kv_cache, out=None, lse=None):
+        v_merged, s_merged = None, None
+        for w in self._wrappers:
+            v, s = w.run(q, paged_kv_cache, return_lse=True)
+            if v_merged is None:
+                v_merged, s_merged = v, s
+            else:
+                merge_state_in_place(v_merged, s_merged, v, s)
+        if out is not None:
+            out.copy_(v_merged)
+            return out
+        return v_merged
+
+    forward = run

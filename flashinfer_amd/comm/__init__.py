@@ -1,0 +1,21 @@
+"""Communication & parallelism over RCCL/xGMI (reference parity:
+flashinfer/comm/). Control plane = torch.distributed; data plane = RCCL
+collectives shaped for the 8xMI355X xGMI topology (7 direct links per GPU:
+prefer all-to-all / one-shot patterns over rings)."""
+
+from .allreduce import AllReduceFusionPattern, allreduce, allreduce_fusion
+from .comm_backend import TorchDistBackend, init_distributed
+from .mapping import Mapping
+from .moe_alltoall import MoeAlltoAll
+from .ulysses import UlyssesCommunicator
+
+__all__ = [
+    "AllReduceFusionPattern",
+    "allreduce",
+    "allreduce_fusion",
+    "TorchDistBackend",
+    "init_distributed",
+    "Mapping",
+    "MoeAlltoAll",
+    "UlyssesCommunicator",
+]

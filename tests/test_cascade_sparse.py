@@ -1,0 +1,103 @@
+"""GPU tests: cascade (shared-prefix) attention and block-sparse attention
+against dense fp32 references."""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def test_cascade_two_level_matches_dense():
+    import flashinfer_amd as fi
+    from flashinfer_amd.cascade import MultiLevelCascadeAttentionWrapper
+
+    torch.manual_seed(0)
+    Hq, Hkv, D, page = 8, 2, 128, 16
+    batch = 4
+    shared_len, unique_len, qo_len = 128, 64, 32
+    # unified page table: shared pages first, then per-request unique pages
+    sp = shared_len // page
+    up = unique_len // page
+    npages = sp + batch * up
+    k_cache = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v_cache = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    nnz = batch * qo_len
+    q = torch.randn(nnz, Hq, D, dtype=torch.bfloat16, device="cuda")
+
+    qo_indptr_top = torch.tensor([0, nnz], dtype=torch.int32, device="cuda")
+    qo_indptr_bot = torch.arange(0, nnz + 1, qo_len, dtype=torch.int32, device="cuda")
+    kv_indptr_top = torch.tensor([0, sp], dtype=torch.int32, device="cuda")
+    kv_indices_top = torch.arange(sp, dtype=torch.int32, device="cuda")
+    kv_indptr_bot = torch.arange(0, batch * up + 1, up, dtype=torch.int32,
+                                 device="cuda")
+    kv_indices_bot = sp + torch.arange(batch * up, dtype=torch.int32, device="cuda")
+    lp_top = torch.tensor([page], dtype=torch.int32, device="cuda")
+    lp_bot = torch.full((batch,), page, dtype=torch.int32, device="cuda")
+
+    ws = torch.empty(32 * 1024 * 1024, dtype=torch.uint8, device="cuda")
+    w = MultiLevelCascadeAttentionWrapper(2, ws, "NHD")
+    w.plan([qo_indptr_top, qo_indptr_bot], [kv_indptr_top, kv_indptr_bot],
+           [kv_indices_top, kv_indices_bot], [lp_top, lp_bot],
+           Hq, Hkv, D, page, causal=True)
+    out = w.run(q, (k_cache, v_cache))
+
+    # dense reference per request: kv = shared + unique, causal over suffix
+    g = Hq // Hkv
+    k_flat = k_cache.float().view(-1, Hkv, D)
+    v_flat = v_cache.float().view(-1, Hkv, D)
+    for b in range(batch):
+        kv_rows = torch.cat([
+            k_flat[:shared_len],
+            k_flat[shared_len + b * unique_len: shared_len + (b + 1) * unique_len],
+        ])
+        v_rows = torch.cat([
+            v_flat[:shared_len],
+            v_flat[shared_len + b * unique_len: shared_len + (b + 1) * unique_len],
+        ])
+        L = shared_len + unique_len
+        qb = q[b * qo_len:(b + 1) * qo_len].float()
+        kf = kv_rows.repeat_interleave(g, dim=1)
+        vf = v_rows.repeat_interleave(g, dim=1)
+        logits = torch.einsum("mhd,lhd->hml", qb, kf) / math.sqrt(D)
+        qpos = torch.arange(qo_len, device="cuda")[:, None]
+        kpos = torch.arange(L, device="cuda")[None, :]
+        # causal on the unique suffix: q i may see shared + unique[0..i+off]
+        mask = kpos > qpos + (L - qo_len)
+        logits = logits.masked_fill(mask[None], float("-inf"))
+        p = torch.softmax(logits, -1)
+        ref = torch.einsum("hml,lhd->mhd", p, vf)
+        torch.testing.assert_close(out[b * qo_len:(b + 1) * qo_len].float(), ref,
+                                   atol=3e-2, rtol=3e-2)
+
+
+def test_block_sparse_matches_dense():
+    import flashinfer_amd as fi
+    from flashinfer_amd.sparse import BlockSparseAttentionWrapper
+
+    torch.manual_seed(0)
+    M, N, R, C = 128, 256, 16, 16
+    Hq, Hkv, D = 4, 4, 64
+    MB, NB = M // R, N // C
+    dense_mask = torch.rand(MB, NB, device="cuda") > 0.5
+    dense_mask[:, 0] = True  # every row attends something
+    indptr = torch.zeros(MB + 1, dtype=torch.int32, device="cuda")
+    indptr[1:] = dense_mask.sum(1).cumsum(0)
+    indices = torch.nonzero(dense_mask)[:, 1].int()
+
+    q = torch.randn(M, Hq, D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(N, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(N, Hkv, D, dtype=torch.bfloat16, device="cuda")
+
+    ws = torch.empty(32 * 1024 * 1024, dtype=torch.uint8, device="cuda")
+    w = BlockSparseAttentionWrapper(ws)
+    w.plan(indptr, indices, M, N, R, C, Hq, Hkv, D)
+    out = w.run(q, k, v)
+
+    # dense reference with -inf outside the block pattern
+    token_mask = dense_mask.repeat_interleave(R, 0).repeat_interleave(C, 1)
+    logits = torch.einsum("mhd,lhd->hml", q.float(), k.float()) / math.sqrt(D)
+    logits = logits.masked_fill(~token_mask[None], float("-inf"))
+    p = torch.softmax(logits, -1)
+    ref = torch.einsum("hml,lhd->mhd", p, v.float())
+    torch.testing.assert_close(out.float(), ref, atol=3e-2, rtol=3e-2)

@@ -1,0 +1,164 @@
+"""Multi-process comm tests on CPU (gloo, world_size=2) + Mapping unit tests.
+The same code paths run over RCCL on the 8xMI355X node."""
+import math
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from flashinfer_amd.comm.mapping import Mapping
+
+
+def test_mapping_tp_pp_cp():
+    m = Mapping(world_size=8, rank=5, tp_size=2, pp_size=2, cp_size=2)
+    # rank = pp*4 + cp*2 + tp: 5 = 1*4 + 0*2 + 1
+    assert m.pp_rank == 1 and m.cp_rank == 0 and m.tp_rank == 1
+    assert m.tp_group == [4, 5]
+    assert m.cp_group == [5, 7]
+    assert m.pp_group == [1, 5]
+    assert m.is_last_pp_rank()
+    assert m.pp_layers(9)[0] == 5  # 5 layers on pp0, 4 on pp1
+
+
+def test_mapping_moe_ep():
+    m = Mapping(world_size=8, rank=3, tp_size=8, moe_ep_size=4)
+    assert m.moe_tp_size == 2 and m.moe_ep_size == 4
+    assert m.moe_ep_rank == 3 and m.moe_tp_rank == 0
+    assert m.moe_ep_group == [0, 1, 2, 3]
+    assert m.ep_experts(8) == [6, 7]
+
+
+def test_mapping_invalid():
+    with pytest.raises(ValueError):
+        Mapping(world_size=8, rank=0, tp_size=3)
+
+
+def _worker(rank, world, port, fn_name):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    torch.manual_seed(0)
+    try:
+        globals()[fn_name](rank, world)
+    finally:
+        dist.destroy_process_group()
+
+
+def _run_mp(fn_name, world=2, port=29511):
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker, args=(r, world, port, fn_name))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(120)
+    for p in procs:
+        assert p.exitcode == 0, f"worker failed with {p.exitcode}"
+
+
+# ---- worker bodies (module-level, spawn-picklable) ----
+
+def _ulysses_body(rank, world):
+    from flashinfer_amd.comm.ulysses import UlyssesCommunicator
+
+    B, S_local, H, D = 2, 4, 8, 16
+    torch.manual_seed(7)  # same full tensor on all ranks
+    full = torch.randn(B, S_local * world, H, D)
+    x = full[:, rank * S_local : (rank + 1) * S_local]
+    comm = UlyssesCommunicator()
+    y = comm.scatter_heads(x)
+    # expect: full sequence, my head slice
+    hl = H // world
+    expected = full[:, :, rank * hl : (rank + 1) * hl]
+    assert torch.equal(y, expected), "scatter_heads mismatch"
+    back = comm.gather_heads(y)
+    assert torch.equal(back, x), "gather_heads not inverse"
+
+
+def _moe_a2a_body(rank, world):
+    from flashinfer_amd.comm.moe_alltoall import MoeAlltoAll
+
+    T, hidden, E, K = 5, 8, 4, 2
+    torch.manual_seed(100 + rank)
+    x = torch.randn(T, hidden)
+    topk_ids = torch.randint(0, E, (T, K))
+    topk_w = torch.softmax(torch.randn(T, K), -1)
+    a2a = MoeAlltoAll(num_experts=E, top_k=K)
+    recv_x, recv_local_exp, state = a2a.dispatch(x, topk_ids)
+    # expert fn: multiply by (global_expert_id + 1)
+    global_exp = recv_local_exp + rank * (E // world)
+    expert_out = recv_x * (global_exp + 1).unsqueeze(1).float()
+    out = a2a.combine(expert_out, topk_w, state)
+    ref = torch.zeros_like(x)
+    for t in range(T):
+        for j in range(K):
+            ref[t] += topk_w[t, j] * x[t] * (topk_ids[t, j] + 1).float()
+    assert torch.allclose(out, ref, atol=1e-5), (out - ref).abs().max()
+
+
+def _ar_fusion_body(rank, world):
+    from flashinfer_amd.comm.allreduce import (
+        AllReduceFusionPattern,
+        allreduce_fusion,
+    )
+
+    torch.manual_seed(50 + rank)
+    x = torch.randn(4, 64)
+    torch.manual_seed(77)  # residual/weight shared
+    res = torch.randn(4, 64)
+    w = torch.randn(64)
+    xs = [None] * world
+    import torch.distributed as dist
+
+    dist.all_gather_object(xs, x)
+    x_sum = sum(xs)
+    ref_res = res + x_sum
+    rms = torch.rsqrt(ref_res.float().pow(2).mean(-1, keepdim=True) + 1e-6)
+    ref_out = (ref_res.float() * rms * w).to(x.dtype)
+    out, new_res = allreduce_fusion(
+        x.clone(), res.clone(), w, 1e-6, AllReduceFusionPattern.kARResidualRMSNorm
+    )
+    assert torch.allclose(new_res, ref_res, atol=1e-5)
+    assert torch.allclose(out, ref_out, atol=1e-4), (out - ref_out).abs().max()
+
+
+def _ring_body(rank, world):
+    from flashinfer_amd.parallel_attention import ring_attention
+
+    M, L_local, Hq, Hkv, D = 6, 5, 4, 2, 16
+    torch.manual_seed(9)  # same everywhere
+    k_full = torch.randn(L_local * world, Hkv, D)
+    v_full = torch.randn(L_local * world, Hkv, D)
+    q_all = torch.randn(world, M, Hq, D)
+    q = q_all[rank]
+    # my kv shard
+    k = k_full[rank * L_local : (rank + 1) * L_local]
+    v = v_full[rank * L_local : (rank + 1) * L_local]
+    out = ring_attention(q, k, v)
+    # reference: full attention
+    g = Hq // Hkv
+    kf = k_full.float().repeat_interleave(g, dim=1)
+    vf = v_full.float().repeat_interleave(g, dim=1)
+    logits = torch.einsum("mhd,lhd->hml", q.float(), kf) / math.sqrt(D)
+    p = torch.softmax(logits, -1)
+    ref = torch.einsum("hml,lhd->mhd", p, vf)
+    assert torch.allclose(out.float(), ref, atol=1e-4), (out.float() - ref).abs().max()
+
+
+def test_ulysses_gloo():
+    _run_mp("_ulysses_body", port=29512)
+
+
+def test_moe_alltoall_gloo():
+    _run_mp("_moe_a2a_body", port=29513)
+
+
+def test_allreduce_fusion_gloo():
+    _run_mp("_ar_fusion_body", port=29514)
+
+
+def test_ring_attention_gloo():
+    _run_mp("_ring_body", port=29515)
