@@ -2,8 +2,13 @@
 // Functional parity with reference include/flashinfer/attention/decode.cuh
 // (BatchDecodeWithPagedKVCacheKernel:615) + scheduler split-KV, re-designed
 // for CDNA4:
-//   * 1-wave (64-lane) workgroups: lanes split into 16-lane columns x 4 kv
-//     tokens (head_dim 128; 8/64 and 32/2 for 64/256), 16 B loads per lane.
+//   * 4-wave workgroups, one (work-item, kv-head) unit per wave (64-thread
+//     workgroups hit the per-CU workgroup cap before the wave cap); within a
+//     wave, lanes split into 16-lane columns x 4 kv tokens (head_dim 128;
+//     8/64 and 32/2 for 64/256), 16 B loads per lane.
+//   * QK dots use packed v_dot2_f32_bf16 (keeps K in bf16 registers, halves
+//     the VALU op count of the convert+FMA form — decode is VALU-bound at
+//     the target batch sizes, not memory-stalled; PMC r01).
 //   * GQA group (1/2/4/8 q heads per kv head) processed in-register against
 //     one K read — the K/V bytes are the bound, q reuse is free.
 //   * split-KV via host-planned work items (req, chunk); partials are written
@@ -23,17 +28,37 @@ namespace fi {
 
 // VPL = 8 elems (16B bf16) per lane along head_dim; LPT = head_dim/8 lanes
 // per token; TPW = 64/LPT tokens per wave.
+typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2_t;
+
+template <typename T>
+__device__ __forceinline__ float qk_dot8(const void* qv, const void* kv, float acc) {
+  if constexpr (__is_same(T, bf16)) {
+    const bf16x2_t* q2 = (const bf16x2_t*)qv;
+    const bf16x2_t* k2 = (const bf16x2_t*)kv;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) acc = __builtin_amdgcn_fdot2_f32_bf16(q2[i], k2[i], acc, false);
+    return acc;
+  } else {
+    const T* q = (const T*)qv;
+    const T* k = (const T*)kv;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) acc += to_f32<T>(q[i]) * to_f32<T>(k[i]);
+    return acc;
+  }
+}
+
 template <typename T, int HEAD_DIM, int GROUP, bool SOFT_CAP>
-__global__ __launch_bounds__(64) void batch_decode_kernel(DecodeParams p) {
+__global__ __launch_bounds__(256, 4) void batch_decode_kernel(DecodeParams p) {
   constexpr int VPL = 8;
   constexpr int LPT = HEAD_DIM / VPL;        // lanes per token: 8/16/32
   constexpr int TPW = kWaveSize / LPT;       // tokens per wave: 8/4/2
-  const int lane = threadIdx.x;
+  const int lane = threadIdx.x & 63;
   const int tsub = lane / LPT;               // which token this lane covers
   const int dcol = (lane % LPT) * VPL;       // feature offset
 
-  int item = blockIdx.x;
-  int kv_head = blockIdx.y;
+  int64_t unit = (int64_t)blockIdx.x * 4 + (threadIdx.x >> 6);
+  int item = (int)(unit / p.num_kv_heads);
+  int kv_head = (int)(unit % p.num_kv_heads);
   if (item >= p.n_items) return;
   int req = p.work_req[item];
   int chunk = p.work_chunk[item];
@@ -48,17 +73,15 @@ __global__ __launch_bounds__(64) void batch_decode_kernel(DecodeParams p) {
     if (w_start > start) start = w_start;
   }
 
-  // stage q (pre-scaled) into registers: GROUP heads x VPL slice
-  float qreg[GROUP][VPL];
+  // stage q packed (dot2 consumes bf16 pairs; sm_scale applied to the
+  // reduced logit instead — 1 op per head per step)
+  vec_t<T, VPL> qreg[GROUP];
   const T* qbase = (const T*)p.q + (int64_t)req * p.q_stride_n;
   const float scale = p.sm_scale;
 #pragma unroll
   for (int g = 0; g < GROUP; ++g) {
     int qh = kv_head * GROUP + g;
-    vec_t<T, VPL> qv;
-    qv.load(qbase + (int64_t)qh * p.q_stride_h + dcol);
-#pragma unroll
-    for (int j = 0; j < VPL; ++j) qreg[g][j] = qv.get(j) * scale;
+    qreg[g].load(qbase + (int64_t)qh * p.q_stride_h + dcol);
   }
 
   state_t<VPL> st[GROUP];
@@ -99,12 +122,11 @@ __global__ __launch_bounds__(64) void batch_decode_kernel(DecodeParams p) {
 
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
-      float s = 0.f;
-#pragma unroll
-      for (int j = 0; j < VPL; ++j) s += qreg[g][j] * kv_cur.get(j);
+      float s = qk_dot8<T>(qreg[g].data, kv_cur.data, 0.f);
       // reduce across the LPT lanes of this token
 #pragma unroll
       for (int off2 = LPT / 2; off2 > 0; off2 >>= 1) s += __shfl_xor(s, off2, 64);
+      s *= scale;
       if constexpr (SOFT_CAP) s = p.logits_soft_cap * tanhf(s / p.logits_soft_cap);
       if (!valid) s = -INFINITY;
       st[g].push(vf, s);
@@ -145,7 +167,8 @@ __global__ __launch_bounds__(64) void batch_decode_kernel(DecodeParams p) {
 template <typename T>
 hipError_t decode_dispatch(DecodeParams& p, hipStream_t stream) {
   int group = p.num_qo_heads / p.num_kv_heads;
-  dim3 g(p.n_items, p.num_kv_heads), blk(64);
+  int64_t units = (int64_t)p.n_items * p.num_kv_heads;
+  dim3 g((uint32_t)((units + 3) / 4)), blk(256);
   bool sc = p.logits_soft_cap > 0.f;
 #define LAUNCH_D(HD, G, SC) \
   hipLaunchKernelGGL((batch_decode_kernel<T, HD, G, SC>), g, blk, 0, stream, p)

@@ -31,6 +31,17 @@ namespace fi {
 constexpr int KVB = 64;      // kv tile
 constexpr float kLog2e = 1.4426950408889634f;
 
+typedef __attribute__((ext_vector_type(4))) __bf16 b16x4;
+
+// gfx950 LDS transpose read: each 16-lane group collectively reads a [4][16]
+// 16-bit row-major subtile (lane addr = subtile_base + (lane&15)*8 bytes) and
+// lane l receives column (l&15) — rows 0..3 in order.
+__device__ __forceinline__ b16x4 ds_read_tr16(uint32_t lds_byte_off) {
+  b16x4 r;
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(r) : "v"(lds_byte_off));
+  return r;
+}
+
 template <int ROWB>
 __device__ __forceinline__ uint32_t swz_row(uint32_t byte_off) {
   // XOR swizzle for ROWB-byte rows
@@ -50,7 +61,12 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
   constexpr int KROWB = HEAD_DIM * 2;        // K/V tile row bytes
 
   __shared__ T Ks[KVB * HEAD_DIM];
-  __shared__ T Vs[HEAD_DIM * KVB];
+  // V is stored in [KVB/4][HEAD_DIM/16] subtiles of [4 kv][16 d] (row-major,
+  // 128 B each) padded to 144 B stride: the shape ds_read_b64_tr_b16 wants
+  // (each 16-lane group reads one subtile and receives it transposed), with
+  // the pad de-aliasing subtile bank positions.
+  constexpr int VTILE_STRIDE = 72;  // elems (144 B)
+  __shared__ T Vs[(KVB / 4) * (HEAD_DIM / 16) * VTILE_STRIDE];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -174,8 +190,10 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
         // (transposed fragments are built with 8 scalar ds reads in PV)
         *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Ks) +
                                     swz_row<KROWB>(row * KROWB + chunk * 16)) = kval;
-        *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Vs) +
-                                    (uint32_t)(row * KROWB + chunk * 16)) = vval;
+        *reinterpret_cast<shortx8*>(
+            reinterpret_cast<char*>(Vs) +
+            ((row >> 2) * (HEAD_DIM / 16) + (chunk >> 1)) * (VTILE_STRIDE * 2) +
+            (row & 3) * 32 + (chunk & 1) * 16) = vval;
       }
     }
     __syncthreads();
@@ -258,20 +276,36 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
       b1[2] = hiH ? W[6] : X[4];
       b1[3] = hiH ? W[7] : X[5];
 
-      // ---- O^T += V^T * P^T (V^T fragments via 8 scalar ds reads from the
-      // row-major V tile) ----
+      // ---- O^T += V^T * P^T: V^T fragments via ds_read_b64_tr_b16 (HW
+      // 4x16 transpose read — guide T10); 4 reads + one wait per d-tile ----
+      {
+        const uint32_t vbase = (uint32_t)(uintptr_t)Vs + (uint32_t)(lane & 15) * 8;
+        const uint32_t tdsel = ((lane >> 4) & 1);
+        const int kvb = kt * 32 + khalf;
 #pragma unroll
-      for (int i = 0; i < DT; ++i) {
-        T vw0[8], vw1[8];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          vw0[j] = Vs[(kt * 32 + khalf + j) * HEAD_DIM + i * 32 + lq];
-          vw1[j] = Vs[(kt * 32 + 16 + khalf + j) * HEAD_DIM + i * 32 + lq];
+        for (int i = 0; i < DT; ++i) {
+          uint32_t a00 = vbase + (((kvb >> 2)) * (HEAD_DIM / 16) + i * 2 + tdsel) *
+                                     (VTILE_STRIDE * 2);
+          b16x4 r00 = ds_read_tr16(a00);
+          b16x4 r01 = ds_read_tr16(a00 + (HEAD_DIM / 16) * (VTILE_STRIDE * 2));
+          uint32_t a10 = a00 + 4 * (HEAD_DIM / 16) * (VTILE_STRIDE * 2);
+          b16x4 r10 = ds_read_tr16(a10);
+          b16x4 r11 = ds_read_tr16(a10 + (HEAD_DIM / 16) * (VTILE_STRIDE * 2));
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+          __builtin_amdgcn_sched_barrier(0);
+          union {
+            b16x4 h[2];
+            frag f;
+          } u0, u1;
+          u0.h[0] = r00;
+          u0.h[1] = r01;
+          u1.h[0] = r10;
+          u1.h[1] = r11;
+          acc_o[i] = mfma_ab_frag<T>::mma32(u0.f, *reinterpret_cast<frag*>(b0),
+                                            acc_o[i]);
+          acc_o[i] = mfma_ab_frag<T>::mma32(u1.f, *reinterpret_cast<frag*>(b1),
+                                            acc_o[i]);
         }
-        acc_o[i] = mfma_ab_frag<T>::mma32(*reinterpret_cast<frag*>(vw0),
-                                          *reinterpret_cast<frag*>(b0), acc_o[i]);
-        acc_o[i] = mfma_ab_frag<T>::mma32(*reinterpret_cast<frag*>(vw1),
-                                          *reinterpret_cast<frag*>(b1), acc_o[i]);
       }
     }
     }

@@ -40,6 +40,9 @@ hipError_t fi_batch_decode(int dtype, fi_ext::DecodeParams* p, hipStream_t strea
 hipError_t fi_gemm_nt(int dtype, const void* A, const void* B, void* C, int M, int N,
                       int K, int64_t lda, int64_t ldb, int64_t ldc, float alpha,
                       hipStream_t stream);
+hipError_t fi_gemm_nt_v2(const void* A, const void* B, void* C, int M, int N, int K,
+                         int64_t lda, int64_t ldb, int64_t ldc, float alpha,
+                         hipStream_t stream);
 hipError_t fi_batch_prefill(int dtype, fi_ext::PrefillParams* p, int paged,
                             hipStream_t stream);
 hipError_t fi_softmax(fi_ext::SamplingParams* p, hipStream_t stream);
@@ -401,6 +404,17 @@ void gemm_nt(at::Tensor a, at::Tensor b_nt, at::Tensor c, double alpha) {
   TORCH_CHECK(c.dim() == 2 && c.stride(1) == 1);
   int M = a.size(0), K = a.size(1), N = b_nt.size(0);
   TORCH_CHECK(b_nt.size(1) == K && c.size(0) == M && c.size(1) == N);
+  // v2 (256^2 tile, global_load_lds pipeline) needs K % 64 == 0 and 16 B
+  // aligned rows; otherwise the general 128^2 kernel handles it.
+  bool v2_ok = dtype_code(a) == 0 && (K % 64 == 0) && (a.stride(0) % 8 == 0) &&
+               (b_nt.stride(0) % 8 == 0) && M > 128;
+  if (v2_ok) {
+    check_hip(fi_gemm_nt_v2(a.data_ptr(), b_nt.data_ptr(), c.data_ptr(), M, N, K,
+                            a.stride(0), b_nt.stride(0), c.stride(0), (float)alpha,
+                            cur_stream(a)),
+              "fi_gemm_nt_v2");
+    return;
+  }
   check_hip(fi_gemm_nt(dtype_code(a), a.data_ptr(), b_nt.data_ptr(), c.data_ptr(), M, N,
                        K, a.stride(0), b_nt.stride(0), c.stride(0), (float)alpha,
                        cur_stream(a)),

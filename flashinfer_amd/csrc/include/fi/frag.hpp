@@ -44,9 +44,11 @@ __device__ __forceinline__ int mfma16_ab_k(int lane) { return (lane >> 4) * 8; }
 __device__ __forceinline__ uint32_t swz128(uint32_t byte_off) {
   return byte_off ^ (((byte_off >> 7) & 7) << 4);
 }
-// For 256-byte rows ([*][128] bf16):
+// For 256-byte rows ([*][128] bf16): 16 16-byte slots per row — XOR with 4
+// row bits for the full-slot spread (8-slot spread leaves rows r and r+8
+// aliased -> 4-way conflicts on 32-row fragment reads).
 __device__ __forceinline__ uint32_t swz256(uint32_t byte_off) {
-  return byte_off ^ (((byte_off >> 8) & 7) << 4);
+  return byte_off ^ (((byte_off >> 8) & 15) << 4);
 }
 
 }  // namespace fi
