@@ -18,6 +18,7 @@ from .cascade import (
     merge_states,
 )
 from .sparse import BlockSparseAttentionWrapper
+from .mla import BatchMLAPagedAttentionWrapper
 from . import comm
 from .parallel_attention import ParallelAttention, ring_attention
 from .decode import (

@@ -1,0 +1,304 @@
+// DeepSeek MLA paged attention (compressed-KV head_dim_ckv=512 + rope
+// head_dim_kpe=64) for gfx950. Functional parity with reference
+// include/flashinfer/attention/mla.cuh (BatchMLAPagedAttentionKernel:976,
+// MLAPlan scheduler.cuh:1580), re-designed for CDNA4:
+//
+//  * MQA structure: every q head attends the same 576-d compressed KV, so
+//    q-"rows" are (q_pos x head) packed; a workgroup tile is 64 rows.
+//  * 8 waves in a 2x4 (q-block x d-slice) grid: QK^T is split across the
+//    576-d axis (each d-slice wave computes a partial S^T over its 144 dims
+//    — no duplicated MFMA), partials are summed through LDS, softmax is
+//    recomputed redundantly per wave (cheap), and PV accumulates O^T over
+//    each wave's own 128-d output slice (64 AGPR per wave).
+//  * the KV tile lives in ONE subtiled [4 kv][16 d] +pad LDS layout that
+//    serves both uses: 16-B vector fragments for QK^T and
+//    ds_read_b64_tr_b16 hardware-transpose fragments for PV.
+//  * split-KV via host-planned (req, row_tile, chunk) work items; partials
+//    go to tmp[pos][max_chunks] slots merged by the shared LSE-merge kernel.
+#include "fi/common.hpp"
+#include "fi/fastdiv.hpp"
+#include "fi/frag.hpp"
+#include "fi/mfma.hpp"
+#include "fi/params.hpp"
+#include "fi/vec.hpp"
+
+namespace fi {
+
+namespace mla {
+
+constexpr int D_CKV = 512, D_KPE = 64, D_QK = 576;
+constexpr int KVB = 32;               // kv tile
+constexpr int ROWS = 64;              // packed rows per workgroup
+constexpr int NTH = 512;              // 8 waves
+constexpr int VTS = 72;               // padded subtile stride (elems)
+constexpr int NTILED = D_QK / 16;     // 36 d-subtiles per 4-kv row group
+constexpr float kLog2e = 1.4426950408889634f;
+
+typedef __attribute__((ext_vector_type(4))) __bf16 b16x4;
+
+__device__ __forceinline__ b16x4 ds_read_tr16(uint32_t lds_byte_off) {
+  b16x4 r;
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(r) : "v"(lds_byte_off));
+  return r;
+}
+
+template <typename T, bool CAUSAL>
+__global__ __launch_bounds__(NTH, 1) void mla_decode_kernel(MlaParams p) {
+  // KV tile in tr-subtiled layout + S-partial exchange buffer
+  __shared__ T KVs[(KVB / 4) * NTILED * VTS];
+  __shared__ float Sx[2][4][32 * 32];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int qb = wid >> 2;   // q block (32 rows each)
+  const int db = wid & 3;    // d slice (144 qk dims; 128 out dims)
+  const int lq = lane & 31;
+  const int khalf = (lane >> 5) * 8;
+
+  const int item = blockIdx.x;
+  if (item >= p.n_items) return;
+  const int req = p.tile_req[item];
+  const int row0 = p.tile_row0[item];
+  const int chunk = p.tile_chunk[item];
+
+  const int qo_begin = p.qo_indptr[req];
+  const int qo_len = p.qo_indptr[req + 1] - qo_begin;
+  const int H = p.num_heads;
+  const int packed_len = qo_len * H;
+
+  int np = p.kv_indptr[req + 1] - p.kv_indptr[req];
+  int64_t kv_len =
+      np == 0 ? 0 : (int64_t)(np - 1) * p.page_size.d + p.kv_last_page_len[req];
+  const int32_t* page_ids = p.kv_indices + p.kv_indptr[req];
+
+  int64_t start = (int64_t)chunk * p.chunk_size;
+  int64_t end = start + p.chunk_size;
+  if (end > kv_len) end = kv_len;
+
+  // this lane's q row (packed row = q_pos * H + head)
+  const int my_row = row0 + qb * 32 + lq;
+  const bool row_valid = my_row < packed_len;
+  uint32_t qpos_u, head_u;
+  p.num_heads_fd.divmod((uint32_t)(row_valid ? my_row : 0), qpos_u, head_u);
+  const int64_t diag = kv_len - qo_len;
+
+  // ---- Q fragments for my (row, d-slice): 9 k-chunks of 16 over 144 dims.
+  // q = [q_nope | q_pe] by d offset.
+  using frag = typename mfma_ab_frag<T>::type;
+  frag qf[9];
+  {
+    const T* qn = (const T*)p.q_nope + (int64_t)(qo_begin + qpos_u) * p.q_nope_stride_n +
+                  (int64_t)head_u * p.q_nope_stride_h;
+    const T* qp = (const T*)p.q_pe + (int64_t)(qo_begin + qpos_u) * p.q_pe_stride_n +
+                  (int64_t)head_u * p.q_pe_stride_h;
+#pragma unroll
+    for (int c = 0; c < 9; ++c) {
+      int d = db * 144 + c * 16 + khalf;
+      if (row_valid) {
+        qf[c] = (d < D_CKV) ? *reinterpret_cast<const frag*>(qn + d)
+                            : *reinterpret_cast<const frag*>(qp + (d - D_CKV));
+      } else {
+        qf[c] = frag{};
+      }
+    }
+  }
+
+  float m_run = -INFINITY, d_run = 0.f;
+  floatx16 acc_o[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) acc_o[i] = {};
+
+  const T* ckv = (const T*)p.ckv_data;
+  const T* kpe = (const T*)p.kpe_data;
+
+  for (int64_t kv0 = start; kv0 < end; kv0 += KVB) {
+    // ---- stage KV tile (subtiled layout; zero-fill OOB rows) ----
+    {
+      constexpr int UNITS = KVB * D_QK / 8;  // 2304
+#pragma unroll
+      for (int it = 0; it < (UNITS + NTH - 1) / NTH; ++it) {
+        int u = tid + it * NTH;
+        if (u < UNITS) {
+          int row = u / (D_QK / 8);
+          int d = (u % (D_QK / 8)) * 8;
+          int64_t kvpos = kv0 + row;
+          shortx8 val = {};
+          if (kvpos < end) {
+            uint32_t pg, entry;
+            p.page_size.divmod((uint32_t)kvpos, pg, entry);
+            if (d < D_CKV) {
+              val = *reinterpret_cast<const shortx8*>(
+                  ckv + (int64_t)page_ids[pg] * p.ckv_stride_page +
+                  (int64_t)entry * p.ckv_stride_n + d);
+            } else {
+              val = *reinterpret_cast<const shortx8*>(
+                  kpe + (int64_t)page_ids[pg] * p.kpe_stride_page +
+                  (int64_t)entry * p.kpe_stride_n + (d - D_CKV));
+            }
+          }
+          *reinterpret_cast<shortx8*>(
+              reinterpret_cast<char*>(KVs) +
+              ((row >> 2) * NTILED + (d >> 4)) * (VTS * 2) + (row & 3) * 32 +
+              (d & 15) * 2) = val;
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- partial S^T over my 144-d slice (9 mfma) ----
+    floatx16 acc_s = {};
+#pragma unroll
+    for (int c = 0; c < 9; ++c) {
+      int d = db * 144 + c * 16 + khalf;
+      // vector A-fragment from the subtiled layout: 8 contiguous d of kv row
+      frag kf = *reinterpret_cast<const frag*>(
+          reinterpret_cast<const char*>(KVs) + ((lq >> 2) * NTILED + (d >> 4)) * (VTS * 2) +
+          (lq & 3) * 32 + (d & 15) * 2);
+      acc_s = mfma_ab_frag<T>::mma32(kf, qf[c], acc_s);
+    }
+    // exchange partials through LDS and sum
+#pragma unroll
+    for (int r = 0; r < 16; ++r)
+      Sx[qb][db][mfma32_cd_row(r, lane) * 32 + lq] = acc_s[r];
+    __syncthreads();
+    float s_full[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int idx = mfma32_cd_row(r, lane) * 32 + lq;
+      s_full[r] = Sx[qb][0][idx] + Sx[qb][1][idx] + Sx[qb][2][idx] + Sx[qb][3][idx];
+    }
+
+    // ---- masking + base-2 softmax (identical across db waves) ----
+    float pr[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int64_t kv = kv0 + mfma32_cd_row(r, lane);
+      float sv = s_full[r] * p.sm_scale * kLog2e;
+      bool ok = kv < end;
+      if constexpr (CAUSAL) ok &= kv <= (int64_t)qpos_u + diag;
+      pr[r] = ok ? sv : -INFINITY;
+    }
+    float tmax = pr[0];
+#pragma unroll
+    for (int r = 1; r < 16; ++r) tmax = fmaxf(tmax, pr[r]);
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    float m_new = fmaxf(m_run, tmax);
+    float f, psum = 0.f;
+    if (m_new == -INFINITY) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) pr[r] = 0.f;
+      f = 1.f;
+    } else {
+      f = __builtin_exp2f(m_run - m_new);
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        pr[r] = (pr[r] == -INFINITY) ? 0.f : __builtin_exp2f(pr[r] - m_new);
+        psum += pr[r];
+      }
+    }
+    m_run = m_new;
+    d_run = d_run * f + psum;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc_o[i][r] *= f;
+    }
+
+    // ---- P^T fragments (pack + half exchange, prefill scheme) ----
+    uint32_t W[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      uint32_t lo_ = float_to_bf16(pr[2 * j]);
+      uint32_t hi_ = float_to_bf16(pr[2 * j + 1]);
+      W[j] = lo_ | (hi_ << 16);
+    }
+    uint32_t X[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) X[j] = (uint32_t)__shfl_xor((int)W[j], 32, 64);
+    const bool hiH = (lane >> 5) != 0;
+    uint32_t b0[4], b1[4];
+    b0[0] = hiH ? X[2] : W[0];
+    b0[1] = hiH ? X[3] : W[1];
+    b0[2] = hiH ? W[2] : X[0];
+    b0[3] = hiH ? W[3] : X[1];
+    b1[0] = hiH ? X[6] : W[4];
+    b1[1] = hiH ? X[7] : W[5];
+    b1[2] = hiH ? W[6] : X[4];
+    b1[3] = hiH ? W[7] : X[5];
+
+    // ---- O^T += ckv^T * P^T over my 128-d output slice (tr reads) ----
+    {
+      const uint32_t vbase = (uint32_t)(uintptr_t)KVs + (uint32_t)(lane & 15) * 8;
+      const uint32_t tdsel = ((lane >> 4) & 1);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        uint32_t a00 =
+            vbase + ((khalf >> 2) * NTILED + db * 8 + i * 2 + tdsel) * (VTS * 2);
+        b16x4 r00 = ds_read_tr16(a00);
+        b16x4 r01 = ds_read_tr16(a00 + NTILED * (VTS * 2));
+        uint32_t a10 = a00 + 4 * NTILED * (VTS * 2);
+        b16x4 r10 = ds_read_tr16(a10);
+        b16x4 r11 = ds_read_tr16(a10 + NTILED * (VTS * 2));
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+        union {
+          b16x4 h[2];
+          frag fr;
+        } u0, u1;
+        u0.h[0] = r00;
+        u0.h[1] = r01;
+        u1.h[0] = r10;
+        u1.h[1] = r11;
+        acc_o[i] =
+            mfma_ab_frag<T>::mma32(u0.fr, *reinterpret_cast<frag*>(b0), acc_o[i]);
+        acc_o[i] =
+            mfma_ab_frag<T>::mma32(u1.fr, *reinterpret_cast<frag*>(b1), acc_o[i]);
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: normalized partial + base-2 lse into tmp slot ----
+  float d_full = d_run + __shfl_xor(d_run, 32, 64);
+  float inv_d = d_full > 0.f ? 1.f / d_full : 0.f;
+  if (row_valid) {
+    int64_t pos = (int64_t)(qo_begin + qpos_u) * H + head_u;
+    float* vout = p.tmp_v + (pos * p.max_chunks + chunk) * D_CKV;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int d = db * 128 + i * 32 + mfma32_cd_row(r, lane);
+        vout[d] = acc_o[i][r] * inv_d;
+      }
+    }
+    if (db == 0 && (lane >> 5) == 0) {
+      float l2 = d_full > 0.f ? m_run + __builtin_log2f(d_full) : -INFINITY;
+      p.tmp_s[pos * p.max_chunks + chunk] = l2;
+    }
+  }
+}
+
+}  // namespace mla
+
+template <typename T>
+hipError_t mla_dispatch(MlaParams& p, hipStream_t stream) {
+  dim3 g(p.n_items), blk(mla::NTH);
+  if (p.causal)
+    hipLaunchKernelGGL((mla::mla_decode_kernel<T, true>), g, blk, 0, stream, p);
+  else
+    hipLaunchKernelGGL((mla::mla_decode_kernel<T, false>), g, blk, 0, stream, p);
+  return hipGetLastError();
+}
+
+}  // namespace fi
+
+extern "C" hipError_t fi_mla_decode(int dtype, fi::MlaParams* p, hipStream_t stream) {
+  if (p->n_items == 0) return hipSuccess;
+  switch (dtype) {
+    case 0: return fi::mla_dispatch<fi::bf16>(*p, stream);
+    case 1: return fi::mla_dispatch<fi::fp16>(*p, stream);
+  }
+  return hipErrorInvalidValue;
+}

@@ -50,6 +50,7 @@ hipError_t fi_sampling(int mode, int from_logits, fi_ext::SamplingParams* p,
                        hipStream_t stream);
 hipError_t fi_renorm(int which, fi_ext::SamplingParams* p, hipStream_t stream);
 hipError_t fi_chain_speculative(fi_ext::SpecParams* sp, hipStream_t stream);
+hipError_t fi_mla_decode(int dtype, fi_ext::MlaParams* p, hipStream_t stream);
 }
 
 namespace {
@@ -394,6 +395,51 @@ void chain_speculative_op(at::Tensor draft_probs, at::Tensor draft_ids,
   check_hip(fi_chain_speculative(&sp, cur_stream(draft_probs)), "fi_chain_speculative");
 }
 
+// ---------------- mla ----------------
+
+void mla_run(at::Tensor q_nope, at::Tensor q_pe, at::Tensor ckv_cache,
+             at::Tensor kpe_cache, at::Tensor qo_indptr, at::Tensor kv_indices,
+             at::Tensor kv_indptr, at::Tensor kv_last_page_len, at::Tensor tile_req,
+             at::Tensor tile_row0, at::Tensor tile_chunk, int64_t chunk_size,
+             int64_t max_chunks, at::Tensor tmp_v, at::Tensor tmp_s, double sm_scale,
+             bool causal) {
+  TORCH_CHECK(q_nope.is_cuda() && q_nope.dim() == 3 && q_nope.size(2) == 512);
+  TORCH_CHECK(q_pe.size(2) == 64);
+  TORCH_CHECK(ckv_cache.dim() == 3 && ckv_cache.size(2) == 512,
+              "ckv cache must be [pages, page_size, 512]");
+  fi_ext::MlaParams p{};
+  p.q_nope = q_nope.data_ptr();
+  p.q_pe = q_pe.data_ptr();
+  p.ckv_data = ckv_cache.data_ptr();
+  p.kpe_data = kpe_cache.data_ptr();
+  p.qo_indptr = qo_indptr.data_ptr<int32_t>();
+  p.kv_indices = kv_indices.data_ptr<int32_t>();
+  p.kv_indptr = kv_indptr.data_ptr<int32_t>();
+  p.kv_last_page_len = kv_last_page_len.data_ptr<int32_t>();
+  p.page_size = fi::uint_fastdiv((uint32_t)ckv_cache.size(1));
+  p.num_heads_fd = fi::uint_fastdiv((uint32_t)q_nope.size(1));
+  p.ckv_stride_page = ckv_cache.stride(0);
+  p.ckv_stride_n = ckv_cache.stride(1);
+  p.kpe_stride_page = kpe_cache.stride(0);
+  p.kpe_stride_n = kpe_cache.stride(1);
+  p.q_nope_stride_n = q_nope.stride(0);
+  p.q_nope_stride_h = q_nope.stride(1);
+  p.q_pe_stride_n = q_pe.stride(0);
+  p.q_pe_stride_h = q_pe.stride(1);
+  p.tile_req = tile_req.data_ptr<int32_t>();
+  p.tile_row0 = tile_row0.data_ptr<int32_t>();
+  p.tile_chunk = tile_chunk.data_ptr<int32_t>();
+  p.n_items = tile_req.size(0);
+  p.chunk_size = (int)chunk_size;
+  p.max_chunks = (int)max_chunks;
+  p.num_heads = q_nope.size(1);
+  p.tmp_v = tmp_v.data_ptr<float>();
+  p.tmp_s = tmp_s.data_ptr<float>();
+  p.sm_scale = (float)sm_scale;
+  p.causal = causal ? 1 : 0;
+  check_hip(fi_mla_decode(dtype_code(q_nope), &p, cur_stream(q_nope)), "fi_mla_decode");
+}
+
 // ---------------- gemm ----------------
 
 // C[M,N] = A[M,K] @ B_nt[N,K]^T ; all K-contiguous row-major.
@@ -448,5 +494,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sampling", &sampling_op);
   m.def("renorm", &renorm_op);
   m.def("chain_speculative", &chain_speculative_op);
+  m.def("mla_run", &mla_run);
   m.def("debug_fastdiv", &debug_fastdiv);
 }

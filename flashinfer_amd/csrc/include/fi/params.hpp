@@ -101,4 +101,31 @@ struct SpecParams {
 };
 
 
+struct MlaParams {
+  const void* q_nope;  // [nnz, H, 512]
+  const void* q_pe;    // [nnz, H, 64]
+  void* ckv_data;      // [pages, page_size, 512]
+  void* kpe_data;      // [pages, page_size, 64]
+  const int32_t* qo_indptr;
+  const int32_t* kv_indices;
+  const int32_t* kv_indptr;
+  const int32_t* kv_last_page_len;
+  uint_fastdiv page_size;
+  uint_fastdiv num_heads_fd;
+  int64_t ckv_stride_page, ckv_stride_n;
+  int64_t kpe_stride_page, kpe_stride_n;
+  int64_t q_nope_stride_n, q_nope_stride_h, q_pe_stride_n, q_pe_stride_h;
+  const int32_t* tile_req;    // [n_items]
+  const int32_t* tile_row0;   // [n_items] packed-row offset
+  const int32_t* tile_chunk;  // [n_items]
+  int n_items;
+  int chunk_size;
+  int max_chunks;
+  int num_heads;
+  float* tmp_v;  // [nnz*H, max_chunks, 512]
+  float* tmp_s;  // [nnz*H, max_chunks]
+  float sm_scale;
+  int causal;
+};
+
 }  // namespace fi

@@ -90,3 +90,30 @@ if __name__ == "__main__":
     if which in ("all", "gemm"):
         bench_gemm(4096)
         bench_gemm(8192)
+    if which in ("all", "mla"):
+        bench_mla()
+        bench_mla(bs=64, kv=4096)
+
+
+def bench_mla(bs=16, kv=1024, H=128, page=32):
+    import math
+    torch.manual_seed(0)
+    pages_per = (kv + page - 1) // page
+    kv_indptr = torch.arange(0, (bs + 1) * pages_per, pages_per, dtype=torch.int32, device="cuda")
+    npages = bs * pages_per
+    kv_indices = torch.randperm(npages, dtype=torch.int32, device="cuda")
+    kv_len_arr = torch.full((bs,), kv, dtype=torch.int32, device="cuda")
+    qo_indptr = torch.arange(0, bs + 1, dtype=torch.int32, device="cuda")
+    ckv = torch.randn(npages, page, 512, dtype=torch.bfloat16, device="cuda")
+    kpe = torch.randn(npages, page, 64, dtype=torch.bfloat16, device="cuda")
+    q_nope = torch.randn(bs, H, 512, dtype=torch.bfloat16, device="cuda")
+    q_pe = torch.randn(bs, H, 64, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(2048 * 1024 * 1024, dtype=torch.uint8, device="cuda")
+    w = fi.BatchMLAPagedAttentionWrapper(ws)
+    sm = 1.0 / math.sqrt(576)
+    w.plan(qo_indptr, kv_indptr, kv_indices, kv_len_arr, H, 512, 64, page, True, sm, torch.bfloat16)
+    out = torch.empty(bs, H, 512, dtype=torch.bfloat16, device="cuda")
+    t = timeit(lambda: w.run(q_nope, q_pe, ckv, kpe, out=out))
+    fl = 2 * bs * H * kv * (576 + 512)
+    bytes_kv = bs * kv * 576 * 2
+    print(f"mla decode bs={bs} kv={kv} H={H}: {t*1e6:.1f} us  {fl/t/1e12:.1f} TFLOPS  {bytes_kv/t/1e12:.2f} TB/s")
