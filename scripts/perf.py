@@ -77,24 +77,6 @@ def bench_gemm(N=4096):
     print(f"torch {N}^3: {t2*1e3:.2f} ms  {2*N**3/t2/1e12:.0f} TFLOPS")
 
 
-if __name__ == "__main__":
-    which = sys.argv[1] if len(sys.argv) > 1 else "all"
-    if which in ("all", "prefill"):
-        bench_prefill()
-        bench_prefill(bs=1, s=8192)
-        bench_prefill(bs=16, s=1024, causal=False)
-    if which in ("all", "decode"):
-        bench_decode()
-        bench_decode(bs=16, kv=1024)
-        bench_decode(bs=128, kv=4096)
-    if which in ("all", "gemm"):
-        bench_gemm(4096)
-        bench_gemm(8192)
-    if which in ("all", "mla"):
-        bench_mla()
-        bench_mla(bs=64, kv=4096)
-
-
 def bench_mla(bs=16, kv=1024, H=128, page=32):
     import math
     torch.manual_seed(0)
@@ -117,3 +99,21 @@ def bench_mla(bs=16, kv=1024, H=128, page=32):
     fl = 2 * bs * H * kv * (576 + 512)
     bytes_kv = bs * kv * 576 * 2
     print(f"mla decode bs={bs} kv={kv} H={H}: {t*1e6:.1f} us  {fl/t/1e12:.1f} TFLOPS  {bytes_kv/t/1e12:.2f} TB/s")
+
+
+if __name__ == "__main__":
+    which = sys.argv[1] if len(sys.argv) > 1 else "all"
+    if which in ("all", "prefill"):
+        bench_prefill()
+        bench_prefill(bs=1, s=8192)
+        bench_prefill(bs=16, s=1024, causal=False)
+    if which in ("all", "decode"):
+        bench_decode()
+        bench_decode(bs=16, kv=1024)
+        bench_decode(bs=128, kv=4096)
+    if which in ("all", "gemm"):
+        bench_gemm(4096)
+        bench_gemm(8192)
+    if which in ("all", "mla"):
+        bench_mla()
+        bench_mla(bs=64, kv=4096)
