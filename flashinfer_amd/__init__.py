@@ -19,6 +19,7 @@ from .cascade import (
 )
 from .sparse import BlockSparseAttentionWrapper
 from .mla import BatchMLAPagedAttentionWrapper
+from .fused_moe import cutlass_fused_moe, dsv3_routing, fused_moe, moe_topk_softmax
 from . import comm
 from .parallel_attention import ParallelAttention, ring_attention
 from .decode import (

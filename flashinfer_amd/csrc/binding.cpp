@@ -51,6 +51,11 @@ hipError_t fi_sampling(int mode, int from_logits, fi_ext::SamplingParams* p,
 hipError_t fi_renorm(int which, fi_ext::SamplingParams* p, hipStream_t stream);
 hipError_t fi_chain_speculative(fi_ext::SpecParams* sp, hipStream_t stream);
 hipError_t fi_mla_decode(int dtype, fi_ext::MlaParams* p, hipStream_t stream);
+hipError_t fi_group_gemm_nt(const void* A, const void* W, void* C,
+                            const int32_t* m_indptr, const int32_t* w_indices,
+                            int num_segments, int max_m_tiles, int N, int K,
+                            int64_t lda, int64_t ldw_n, int64_t ldw_seg, int64_t ldc,
+                            hipStream_t stream);
 }
 
 namespace {
@@ -467,6 +472,23 @@ void gemm_nt(at::Tensor a, at::Tensor b_nt, at::Tensor c, double alpha) {
             "fi_gemm_nt");
 }
 
+// grouped NT GEMM: a [M,K]; w [num_w, N, K]; c [M,N]; m_indptr [S+1] device.
+void group_gemm_nt(at::Tensor a, at::Tensor w, at::Tensor c, at::Tensor m_indptr,
+                   c10::optional<at::Tensor> w_indices, int64_t max_m_tiles) {
+  TORCH_CHECK(a.is_cuda() && a.dim() == 2 && a.stride(1) == 1);
+  TORCH_CHECK(w.dim() == 3 && w.stride(2) == 1, "weights must be [E, N, K] K-contig");
+  TORCH_CHECK(a.scalar_type() == at::kBFloat16);
+  int K = a.size(1), N = w.size(1);
+  TORCH_CHECK(w.size(2) == K && c.size(1) == N);
+  int S = m_indptr.size(0) - 1;
+  check_hip(fi_group_gemm_nt(
+                a.data_ptr(), w.data_ptr(), c.data_ptr(), m_indptr.data_ptr<int32_t>(),
+                w_indices.has_value() ? w_indices->data_ptr<int32_t>() : nullptr, S,
+                (int)max_m_tiles, N, K, a.stride(0), w.stride(1), w.stride(0),
+                c.stride(0), cur_stream(a)),
+            "fi_group_gemm_nt");
+}
+
 // fastdiv self-check (host): returns n // d computed via the magic scheme.
 std::vector<int64_t> debug_fastdiv(int64_t d, std::vector<int64_t> ns) {
   fi::uint_fastdiv fd((uint32_t)d);
@@ -495,5 +517,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("renorm", &renorm_op);
   m.def("chain_speculative", &chain_speculative_op);
   m.def("mla_run", &mla_run);
+  m.def("group_gemm_nt", &group_gemm_nt);
   m.def("debug_fastdiv", &debug_fastdiv);
 }

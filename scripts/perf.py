@@ -101,6 +101,19 @@ def bench_mla(bs=16, kv=1024, H=128, page=32):
     print(f"mla decode bs={bs} kv={kv} H={H}: {t*1e6:.1f} us  {fl/t/1e12:.1f} TFLOPS  {bytes_kv/t/1e12:.2f} TB/s")
 
 
+def bench_moe(T=4096, H=4096, inter=14336, E=8, k=2):
+    from flashinfer_amd.fused_moe import fused_moe, moe_topk_softmax
+    torch.manual_seed(0)
+    x = torch.randn(T, H, dtype=torch.bfloat16, device="cuda") / 4
+    w13 = torch.randn(E, 2 * inter, H, dtype=torch.bfloat16, device="cuda") / 16
+    w2 = torch.randn(E, H, inter, dtype=torch.bfloat16, device="cuda") / 16
+    logits = torch.randn(T, E, device="cuda")
+    weights, ids = moe_topk_softmax(logits, k)
+    t = timeit(lambda: fused_moe(x, w13, w2, weights, ids), iters=10, warmup=3)
+    fl = T * k * 3 * H * inter * 2
+    print(f"fused_moe mixtral T={T}: {t*1e3:.2f} ms  {fl/t/1e12:.0f} TFLOPS  {T/t/1e6:.3f} M tok/s")
+
+
 if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "all"
     if which in ("all", "prefill"):
@@ -117,3 +130,5 @@ if __name__ == "__main__":
     if which in ("all", "mla"):
         bench_mla()
         bench_mla(bs=64, kv=4096)
+    if which in ("all", "moe"):
+        bench_moe()
