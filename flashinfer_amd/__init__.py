@@ -20,6 +20,14 @@ from .cascade import (
 from .sparse import BlockSparseAttentionWrapper
 from .mla import BatchMLAPagedAttentionWrapper
 from .fused_moe import cutlass_fused_moe, dsv3_routing, fused_moe, moe_topk_softmax
+from .fp8_quantization import (
+    bmm_fp8,
+    gemm_fp8_nt_groupwise,
+    group_gemm_fp8_nt_groupwise,
+    mm_fp8,
+    per_block_quant_fp8,
+    per_token_group_quant_fp8,
+)
 from . import comm
 from .parallel_attention import ParallelAttention, ring_attention
 from .decode import (

@@ -56,6 +56,18 @@ hipError_t fi_group_gemm_nt(const void* A, const void* W, void* C,
                             int num_segments, int max_m_tiles, int N, int K,
                             int64_t lda, int64_t ldw_n, int64_t ldw_seg, int64_t ldc,
                             hipStream_t stream);
+hipError_t fi_gemm_fp8_grouped(const void* A, const void* W, void* C,
+                               const int32_t* m_indptr, const int32_t* w_indices,
+                               int num_segments, int max_m_tiles, int N, int K,
+                               int64_t lda, int64_t ldw_n, int64_t ldw_seg, int64_t ldc,
+                               const float* a_scales, const float* b_scales,
+                               float scalar_scale, int64_t a_scale_stride,
+                               hipStream_t stream);
+hipError_t fi_per_group_quant_fp8(int dtype, int trans_scale, const void* x, void* q,
+                                  float* scale, int64_t rows, int K, int64_t stride_row,
+                                  float eps, hipStream_t stream);
+hipError_t fi_scale_quant_fp8(int dtype, const void* x, void* q, const float* inv_scale,
+                              int64_t n, hipStream_t stream);
 }
 
 namespace {
@@ -489,6 +501,44 @@ void group_gemm_nt(at::Tensor a, at::Tensor w, at::Tensor c, at::Tensor m_indptr
             "fi_group_gemm_nt");
 }
 
+// fp8 grouped/groupwise GEMM. a/w are uint8 (fp8 e4m3) tensors.
+void gemm_fp8_grouped(at::Tensor a, at::Tensor w, at::Tensor c, at::Tensor m_indptr,
+                      c10::optional<at::Tensor> w_indices, int64_t max_m_tiles,
+                      c10::optional<at::Tensor> a_scales,
+                      c10::optional<at::Tensor> b_scales, double scalar_scale) {
+  TORCH_CHECK(a.is_cuda() && a.dim() == 2 && a.stride(1) == 1);
+  TORCH_CHECK(w.dim() == 3 && w.stride(2) == 1);
+  TORCH_CHECK(c.scalar_type() == at::kBFloat16);
+  int K = a.size(1), N = w.size(1);
+  int S = m_indptr.size(0) - 1;
+  int64_t a_scale_stride = a_scales.has_value() ? a_scales->stride(0) : 0;
+  check_hip(fi_gemm_fp8_grouped(
+                a.data_ptr(), w.data_ptr(), c.data_ptr(), m_indptr.data_ptr<int32_t>(),
+                w_indices.has_value() ? w_indices->data_ptr<int32_t>() : nullptr, S,
+                (int)max_m_tiles, N, K, a.stride(0), w.stride(1), w.stride(0),
+                c.stride(0),
+                a_scales.has_value() ? a_scales->data_ptr<float>() : nullptr,
+                b_scales.has_value() ? b_scales->data_ptr<float>() : nullptr,
+                (float)scalar_scale, a_scale_stride, cur_stream(a)),
+            "fi_gemm_fp8_grouped");
+}
+
+// per-128-group fp8 quantization: returns (q uint8 [rows,K], scale f32)
+void per_group_quant_fp8(at::Tensor x, at::Tensor q, at::Tensor scale,
+                         bool trans_scale, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.stride(1) == 1);
+  check_hip(fi_per_group_quant_fp8(dtype_code(x), trans_scale ? 1 : 0, x.data_ptr(),
+                                   q.data_ptr(), scale.data_ptr<float>(), x.size(0),
+                                   x.size(1), x.stride(0), (float)eps, cur_stream(x)),
+            "fi_per_group_quant_fp8");
+}
+
+void scale_quant_fp8(at::Tensor x, at::Tensor q, at::Tensor inv_scale) {
+  check_hip(fi_scale_quant_fp8(dtype_code(x), x.data_ptr(), q.data_ptr(),
+                               inv_scale.data_ptr<float>(), x.numel(), cur_stream(x)),
+            "fi_scale_quant_fp8");
+}
+
 // fastdiv self-check (host): returns n // d computed via the magic scheme.
 std::vector<int64_t> debug_fastdiv(int64_t d, std::vector<int64_t> ns) {
   fi::uint_fastdiv fd((uint32_t)d);
@@ -518,5 +568,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("chain_speculative", &chain_speculative_op);
   m.def("mla_run", &mla_run);
   m.def("group_gemm_nt", &group_gemm_nt);
+  m.def("gemm_fp8_grouped", &gemm_fp8_grouped);
+  m.def("per_group_quant_fp8", &per_group_quant_fp8);
+  m.def("scale_quant_fp8", &scale_quant_fp8);
   m.def("debug_fastdiv", &debug_fastdiv);
 }

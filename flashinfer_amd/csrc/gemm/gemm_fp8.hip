@@ -1,0 +1,194 @@
+// FP8 (e4m3) groupwise-scaled NT GEMM for gfx950 — dense, grouped (MoE /
+// segment) and batched in one kernel. Parity targets: reference
+// gemm_fp8_nt_groupwise (1x128 activation scales x 128x128 weight-block
+// scales, gemm/gemm_base.py:7589), group_gemm_fp8_nt_groupwise (:8103),
+// bmm_fp8 (:7387, per-tensor scales).
+//
+// Design: gemm_v2 pipeline shape at BK = 128 fp8 elements (= one K scale
+// group = 128-byte LDS rows, byte-identical swizzle geometry to the bf16
+// BK=64 kernels), v_mfma_f32_32x32x16_fp8_fp8 (8-byte fragments), per-group
+// f32 rescale of the local accumulator into the master accumulator:
+//   C += a_scale[kg, m] * b_scale[kg, n/128] * sum_k A_q B_q.
+// Per-tensor mode (bmm_fp8): scales null, one scalar factor at the end.
+#include "fi/common.hpp"
+#include "fi/frag.hpp"
+#include "fi/mfma.hpp"
+
+namespace fi {
+
+namespace f8gemm {
+
+constexpr int BM = 128, BN = 128, BK = 128;  // BK in fp8 elements (128 B rows)
+constexpr int NTH = 256;                     // 4 waves 2x2, 64x64 per wave
+constexpr int WM = 64, WN = 64;
+
+__device__ __forceinline__ void stage_tile(const uint8_t* __restrict__ gbase,
+                                           int64_t ld, int row0, int rows_end, int k0,
+                                           uint32_t lds_base_bytes, int tid) {
+  // BM x BK x 1B = 16 KB = 1024 16-B units over 256 threads
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int u = tid + i * NTH;
+    uint32_t dst_byte = (uint32_t)u * 16;
+    uint32_t logical = swz128(dst_byte);
+    int row = logical >> 7;
+    int col = logical & 127;
+    int gm = row0 + row;
+    const uint8_t* src =
+        gbase + (int64_t)(gm < rows_end ? gm : rows_end - 1) * ld + k0 + col;
+    __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)src,
+                                     (__attribute__((address_space(3))) void*)(uintptr_t)(
+                                         lds_base_bytes + dst_byte),
+                                     16, 0, 0);
+  }
+}
+
+// OUT_BF16: write bf16, else f32
+template <bool GROUPWISE>
+__global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
+    const uint8_t* __restrict__ A, const uint8_t* __restrict__ W,
+    bf16* __restrict__ C, const int32_t* __restrict__ m_indptr,
+    const int32_t* __restrict__ w_indices, int N, int K, int64_t lda, int64_t ldw_n,
+    int64_t ldw_seg, int64_t ldc, const float* __restrict__ a_scales,  // [K/128, Mtot]
+    const float* __restrict__ b_scales,  // [S, K/128, N/128]
+    float scalar_scale, int64_t a_scale_stride) {
+  __shared__ uint8_t As[2][BM * BK];
+  __shared__ uint8_t Bs[2][BN * BK];
+
+  const int seg = blockIdx.z;
+  const int m0 = m_indptr[seg] + blockIdx.y * BM;
+  const int m_end = m_indptr[seg + 1];
+  if (m0 >= m_end) return;
+  const int bn0 = blockIdx.x * BN;
+  const int widx = w_indices ? w_indices[seg] : seg;
+  const uint8_t* Wb = W + (int64_t)widx * ldw_seg;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = (wid >> 1) * WM;
+  const int wn = (wid & 1) * WN;
+  const int line = lane & 31;
+  const int khalf = (lane >> 5) * 8;
+
+  floatx16 accm[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) accm[i][j] = {};
+
+  const uint32_t as_base = (uint32_t)(uintptr_t)&As[0][0];
+  const uint32_t bs_base = (uint32_t)(uintptr_t)&Bs[0][0];
+  constexpr uint32_t BUF_BYTES = BM * BK;
+
+  int nk = K / BK;
+  stage_tile(A, lda, m0, m_end, 0, as_base, tid);
+  stage_tile(Wb, ldw_n, bn0, N, 0, bs_base, tid);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  int cur = 0;
+  for (int kt = 0; kt < nk; ++kt) {
+    if (kt + 1 < nk) {
+      stage_tile(A, lda, m0, m_end, (kt + 1) * BK, as_base + (cur ^ 1) * BUF_BYTES, tid);
+      stage_tile(Wb, ldw_n, bn0, N, (kt + 1) * BK, bs_base + (cur ^ 1) * BUF_BYTES, tid);
+    }
+    const char* a_lds = (const char*)&As[cur][0];
+    const char* b_lds = (const char*)&Bs[cur][0];
+    floatx16 acc[2][2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j) acc[i][j] = {};
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < BK / 16; ++ks) {
+      int64_t af[2], bfv[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+        af[i] = *reinterpret_cast<const int64_t*>(
+            a_lds + (swz128((wm + i * 32 + line) * BK + ks * 16) + khalf));
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+        bfv[j] = *reinterpret_cast<const int64_t*>(
+            b_lds + (swz128((wn + j * 32 + line) * BK + ks * 16) + khalf));
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = mfma_32x32x16_fp8(af[i], bfv[j], acc[i][j]);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    // rescale local accumulator into master
+    if constexpr (GROUPWISE) {
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        float sa[16];
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int m = m0 + wm + i * 32 + mfma32_cd_row(r, lane);
+          sa[r] = a_scales[(int64_t)kt * a_scale_stride + (m < m_end ? m : m_end - 1)];
+        }
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+          float sb = b_scales[((int64_t)seg * nk + kt) * ((N + 127) / 128) +
+                              (bn0 + wn + j * 32) / 128];
+#pragma unroll
+          for (int r = 0; r < 16; ++r) accm[i][j][r] += acc[i][j][r] * (sa[r] * sb);
+        }
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) accm[i][j][r] += acc[i][j][r];
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  float fs = GROUPWISE ? 1.f : scalar_scale;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int m = m0 + wm + i * 32 + mfma32_cd_row(r, lane);
+        int n = bn0 + wn + j * 32 + mfma32_cd_col(lane);
+        if (m < m_end && n < N)
+          C[(int64_t)m * ldc + n] = from_f32<bf16>(accm[i][j][r] * fs);
+      }
+    }
+  }
+}
+
+}  // namespace f8gemm
+
+}  // namespace fi
+
+extern "C" hipError_t fi_gemm_fp8_grouped(
+    const void* A, const void* W, void* C, const int32_t* m_indptr,
+    const int32_t* w_indices, int num_segments, int max_m_tiles, int N, int K,
+    int64_t lda, int64_t ldw_n, int64_t ldw_seg, int64_t ldc, const float* a_scales,
+    const float* b_scales, float scalar_scale, int64_t a_scale_stride,
+    hipStream_t stream) {
+  if (K % 128 != 0) return hipErrorInvalidValue;
+  dim3 grid((N + fi::f8gemm::BN - 1) / fi::f8gemm::BN, max_m_tiles, num_segments);
+  dim3 blk(fi::f8gemm::NTH);
+  if (a_scales && b_scales) {
+    hipLaunchKernelGGL((fi::f8gemm::gemm_fp8_kernel<true>), grid, blk, 0, stream,
+                       (const uint8_t*)A, (const uint8_t*)W, (fi::bf16*)C, m_indptr,
+                       w_indices, N, K, lda, ldw_n, ldw_seg, ldc, a_scales, b_scales,
+                       scalar_scale, a_scale_stride);
+  } else {
+    hipLaunchKernelGGL((fi::f8gemm::gemm_fp8_kernel<false>), grid, blk, 0, stream,
+                       (const uint8_t*)A, (const uint8_t*)W, (fi::bf16*)C, m_indptr,
+                       w_indices, N, K, lda, ldw_n, ldw_seg, ldc, a_scales, b_scales,
+                       scalar_scale, a_scale_stride);
+  }
+  return hipGetLastError();
+}

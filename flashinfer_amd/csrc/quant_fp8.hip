@@ -1,0 +1,106 @@
+// FP8 (OCP e4m3fn — gfx950 native, NOT the MI300X fnuz variant) quantization
+// kernels. Parity with reference fp8_quantization.py / nv_internal thop quant
+// ops: per-token-group (1x128) activation quant and per-tensor quant.
+#include "fi/common.hpp"
+#include "fi/vec.hpp"
+
+namespace fi {
+
+constexpr float kF8Max = 448.f;  // e4m3fn max
+
+__device__ __forceinline__ uint8_t to_fp8(float x) {
+  return __hip_cvt_float_to_fp8(x, __HIP_SATFINITE, __HIP_E4M3);
+}
+
+// One wave per (row, group of 128): amax -> scale -> quantize.
+// scale layout: MN-major [K/128, M] (transposed=true) or [M, K/128].
+template <typename T, bool TRANS_SCALE>
+__global__ void per_group_quant_kernel(const T* __restrict__ x, uint8_t* __restrict__ q,
+                                       float* __restrict__ scale, int64_t rows, int K,
+                                       int64_t stride_row, float eps) {
+  int groups = K / 128;
+  int64_t total = rows * groups;
+  int64_t wave = (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) >> 6;
+  int lane = threadIdx.x & 63;
+  if (wave >= total) return;
+  int g = (int)(wave % groups);
+  int64_t row = wave / groups;
+  const T* src = x + row * stride_row + g * 128;
+  float v[2];
+  vec_t<T, 2> lv;
+  lv.load(src + lane * 2);
+  v[0] = lv.get(0);
+  v[1] = lv.get(1);
+  float amax = fmaxf(fabsf(v[0]), fabsf(v[1]));
+  amax = wave_reduce_max<kWaveSize>(amax);
+  float s = fmaxf(amax, eps) / kF8Max;
+  float inv = 1.f / s;
+  uint8_t* dst = q + row * K + g * 128 + lane * 2;
+  dst[0] = to_fp8(v[0] * inv);
+  dst[1] = to_fp8(v[1] * inv);
+  if (lane == 0) {
+    if constexpr (TRANS_SCALE) scale[(int64_t)g * rows + row] = s;
+    else scale[row * groups + g] = s;
+  }
+}
+
+template <typename T>
+__global__ void scale_quant_kernel(const T* __restrict__ x, uint8_t* __restrict__ q,
+                                   const float* __restrict__ inv_scale, int64_t n) {
+  float is = *inv_scale;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    q[i] = to_fp8(to_f32<T>(x[i]) * is);
+  }
+}
+
+}  // namespace fi
+
+extern "C" hipError_t fi_per_group_quant_fp8(int dtype, int trans_scale, const void* x,
+                                             void* q, float* scale, int64_t rows, int K,
+                                             int64_t stride_row, float eps,
+                                             hipStream_t stream) {
+  if (K % 128 != 0) return hipErrorInvalidValue;
+  int64_t waves = rows * (K / 128);
+  int64_t blocks = (waves * 64 + 255) / 256;
+  dim3 g((uint32_t)(blocks < 0x7FFFFFFF ? blocks : 0x7FFFFFFF)), blk(256);
+#define LQ(T, TS)                                                                  \
+  hipLaunchKernelGGL((fi::per_group_quant_kernel<T, TS>), g, blk, 0, stream,        \
+                     (const T*)x, (uint8_t*)q, scale, rows, K, stride_row, eps)
+  switch (dtype * 2 + trans_scale) {
+    case 0: LQ(fi::bf16, false); break;
+    case 1: LQ(fi::bf16, true); break;
+    case 2: LQ(fi::fp16, false); break;
+    case 3: LQ(fi::fp16, true); break;
+    case 4: LQ(float, false); break;
+    case 5: LQ(float, true); break;
+    default: return hipErrorInvalidValue;
+  }
+#undef LQ
+  return hipGetLastError();
+}
+
+extern "C" hipError_t fi_scale_quant_fp8(int dtype, const void* x, void* q,
+                                         const float* inv_scale, int64_t n,
+                                         hipStream_t stream) {
+  int grid = (int)((n + 255) / 256);
+  if (grid > 2048) grid = 2048;
+  dim3 g(grid), blk(256);
+  switch (dtype) {
+    case 0:
+      hipLaunchKernelGGL((fi::scale_quant_kernel<fi::bf16>), g, blk, 0, stream,
+                         (const fi::bf16*)x, (uint8_t*)q, inv_scale, n);
+      break;
+    case 1:
+      hipLaunchKernelGGL((fi::scale_quant_kernel<fi::fp16>), g, blk, 0, stream,
+                         (const fi::fp16*)x, (uint8_t*)q, inv_scale, n);
+      break;
+    case 2:
+      hipLaunchKernelGGL((fi::scale_quant_kernel<float>), g, blk, 0, stream,
+                         (const float*)x, (uint8_t*)q, inv_scale, n);
+      break;
+    default:
+      return hipErrorInvalidValue;
+  }
+  return hipGetLastError();
+}
