@@ -20,6 +20,8 @@ from .cascade import (
 from .sparse import BlockSparseAttentionWrapper
 from .mla import BatchMLAPagedAttentionWrapper
 from .fused_moe import cutlass_fused_moe, dsv3_routing, fused_moe, moe_topk_softmax
+from .topk import top_k
+from .quantization import packbits, segment_packbits
 from .fp8_quantization import (
     bmm_fp8,
     gemm_fp8_nt_groupwise,

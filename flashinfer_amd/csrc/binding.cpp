@@ -68,6 +68,12 @@ hipError_t fi_per_group_quant_fp8(int dtype, int trans_scale, const void* x, voi
                                   float eps, hipStream_t stream);
 hipError_t fi_scale_quant_fp8(int dtype, const void* x, void* q, const float* inv_scale,
                               int64_t n, hipStream_t stream);
+hipError_t fi_topk(const float* x, float* out_v, int32_t* out_i, int rows, int d, int k,
+                   int64_t stride, hipStream_t stream);
+hipError_t fi_packbits(const uint8_t* x, uint8_t* y, int64_t n, hipStream_t stream);
+hipError_t fi_segment_packbits(const uint8_t* x, uint8_t* y, const int32_t* x_indptr,
+                               const int32_t* y_indptr, int num_segments,
+                               hipStream_t stream);
 }
 
 namespace {
@@ -539,6 +545,29 @@ void scale_quant_fp8(at::Tensor x, at::Tensor q, at::Tensor inv_scale) {
             "fi_scale_quant_fp8");
 }
 
+void topk_op(at::Tensor x, at::Tensor out_v, at::Tensor out_i, int64_t k) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.scalar_type() == at::kFloat);
+  check_hip(fi_topk(x.data_ptr<float>(), out_v.data_ptr<float>(),
+                    out_i.data_ptr<int32_t>(), x.size(0), x.size(1), (int)k,
+                    x.stride(0), cur_stream(x)),
+            "fi_topk");
+}
+
+void packbits_op(at::Tensor x, at::Tensor y) {
+  check_hip(fi_packbits(x.data_ptr<uint8_t>(), y.data_ptr<uint8_t>(), x.numel(),
+                        cur_stream(x)),
+            "fi_packbits");
+}
+
+void segment_packbits_op(at::Tensor x, at::Tensor y, at::Tensor x_indptr,
+                         at::Tensor y_indptr) {
+  check_hip(fi_segment_packbits(x.data_ptr<uint8_t>(), y.data_ptr<uint8_t>(),
+                                x_indptr.data_ptr<int32_t>(),
+                                y_indptr.data_ptr<int32_t>(), x_indptr.size(0) - 1,
+                                cur_stream(x)),
+            "fi_segment_packbits");
+}
+
 // fastdiv self-check (host): returns n // d computed via the magic scheme.
 std::vector<int64_t> debug_fastdiv(int64_t d, std::vector<int64_t> ns) {
   fi::uint_fastdiv fd((uint32_t)d);
@@ -571,5 +600,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_fp8_grouped", &gemm_fp8_grouped);
   m.def("per_group_quant_fp8", &per_group_quant_fp8);
   m.def("scale_quant_fp8", &scale_quant_fp8);
+  m.def("topk", &topk_op);
+  m.def("packbits", &packbits_op);
+  m.def("segment_packbits", &segment_packbits_op);
   m.def("debug_fastdiv", &debug_fastdiv);
 }
