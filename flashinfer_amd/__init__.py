@@ -19,6 +19,7 @@ from .cascade import (
 )
 from .sparse import BlockSparseAttentionWrapper
 from .mla import BatchMLAPagedAttentionWrapper
+from .attention import BatchAttention, PODWithPagedKVCacheWrapper
 from .fused_moe import cutlass_fused_moe, dsv3_routing, fused_moe, moe_topk_softmax
 from .topk import top_k
 from .quantization import packbits, segment_packbits

@@ -1,0 +1,90 @@
+"""Holistic mixed-batch attention (reference parity: flashinfer/attention/
+_core.py BatchAttention:44 — prefill and decode requests fused in ONE
+launch). The CDNA4 batch-prefill kernel handles arbitrary per-request qo_len
+(1-token decode rows pack into the same GQA row tiles), so the holistic API
+is the paged prefill wrapper planned over the mixed batch — one kernel, one
+launch, work items load-balanced longest-first."""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from .prefill import BatchPrefillWithPagedKVCacheWrapper
+
+
+class BatchAttention:
+    def __init__(self, kv_layout: str = "NHD", device=None, **kwargs):
+        self._kv_layout = kv_layout
+        self._wrapper: Optional[BatchPrefillWithPagedKVCacheWrapper] = None
+
+    def plan(
+        self,
+        qo_indptr: torch.Tensor,
+        kv_indptr: torch.Tensor,
+        kv_indices: torch.Tensor,
+        kv_len_arr: torch.Tensor,
+        num_qo_heads: int,
+        num_kv_heads: int,
+        head_dim_qk: int,
+        head_dim_vo: int,
+        page_size: int,
+        causal: bool = True,
+        sm_scale: Optional[float] = None,
+        logits_soft_cap: Optional[float] = None,
+        q_data_type: torch.dtype = torch.bfloat16,
+        kv_data_type: Optional[torch.dtype] = None,
+        use_profiler: bool = False,
+        **kwargs,
+    ) -> None:
+        dev = qo_indptr.device if qo_indptr.is_cuda else (
+            kv_indices.device if kv_indices.is_cuda else torch.device("cuda"))
+        if self._wrapper is None:
+            ws = torch.empty(128 * 1024 * 1024, dtype=torch.uint8, device=dev)
+            self._wrapper = BatchPrefillWithPagedKVCacheWrapper(ws, self._kv_layout)
+        last_page_len = ((kv_len_arr.to(torch.int64) - 1) % page_size + 1).to(
+            torch.int32
+        )
+        self._wrapper.plan(
+            qo_indptr, kv_indptr, kv_indices, last_page_len,
+            num_qo_heads, num_kv_heads, head_dim_qk, page_size, causal=causal,
+            sm_scale=sm_scale, logits_soft_cap=logits_soft_cap,
+            q_data_type=q_data_type,
+        )
+
+    def run(
+        self, q: torch.Tensor, kv_cache, out: Optional[torch.Tensor] = None,
+        lse: Optional[torch.Tensor] = None, return_lse: bool = True, **kwargs,
+    ):
+        result = self._wrapper.run(q, kv_cache, out=out, lse=lse,
+                                   return_lse=return_lse)
+        return result
+
+
+class PODWithPagedKVCacheWrapper:
+    r"""Prefill-On-Decode (reference pod.py:61): one prefill request plus a
+    decode batch issued back-to-back on the same stream — on the 256-CU chip
+    the two grids overlap naturally."""
+
+    def __init__(self, float_workspace_buffer: torch.Tensor, kv_layout: str = "NHD",
+                 **kwargs):
+        from .decode import BatchDecodeWithPagedKVCacheWrapper
+
+        self._decode = BatchDecodeWithPagedKVCacheWrapper(
+            float_workspace_buffer, kv_layout
+        )
+        self._kv_layout = kv_layout
+
+    def plan(self, indptr, indices, last_page_len, num_qo_heads, num_kv_heads,
+             head_dim, page_size, **kwargs):
+        self._decode.plan(indptr, indices, last_page_len, num_qo_heads,
+                          num_kv_heads, head_dim, page_size, **kwargs)
+
+    def run(self, q_p, k_p, v_p, q_d, paged_kv_cache, causal_p: bool = True, **kwargs):
+        r"""q_p/k_p/v_p: the prefill request (contiguous KV); q_d: decode
+        queries [batch, H, D] over the paged cache."""
+        from .prefill import single_prefill_with_kv_cache
+
+        o_p = single_prefill_with_kv_cache(q_p, k_p, v_p, causal=causal_p)
+        o_d = self._decode.run(q_d, paged_kv_cache)
+        return o_p, o_d

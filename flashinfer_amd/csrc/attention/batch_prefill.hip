@@ -53,7 +53,7 @@ __device__ __forceinline__ uint32_t swz_row(uint32_t byte_off) {
 // CTAQ = packed q rows per workgroup (128 -> 4 waves, 256 -> 8 waves; the
 // planner picks by average packed length — bigger tiles amortize K/V staging
 // over more q rows).
-template <typename T, int HEAD_DIM, int CTAQ, bool PAGED, bool CAUSAL>
+template <typename T, int HEAD_DIM, int CTAQ, bool PAGED, bool CAUSAL, bool MASK = false>
 __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParams p) {
   constexpr int NTHREADS = CTAQ * 2;
   constexpr int KCH = HEAD_DIM / 16;  // k-chunks in QK^T
@@ -223,6 +223,13 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
         bool ok = kv < kv_len;
         if constexpr (CAUSAL) ok &= kv <= my_qpos + diag;
         if (wleft >= 0) ok &= kv >= my_qpos + diag - wleft;
+        if constexpr (MASK) {
+          if (ok) {
+            int64_t bit = (int64_t)my_qpos * kv_len + kv;
+            uint8_t byte = p.mask_data[p.mask_byte_indptr[req] + (bit >> 3)];
+            ok &= (byte >> (bit & 7)) & 1;
+          }
+        }
         pr[r] = ok ? sv : -INFINITY;
       }
 
@@ -344,9 +351,20 @@ hipError_t prefill_dispatch(PrefillParams& p, bool paged, hipStream_t stream) {
     if (p.cta_q == 256) LAUNCH_PF2(HD, 256, PG, CS); \
     else LAUNCH_PF2(HD, 128, PG, CS);                \
   } while (0)
+#define LAUNCH_PFM(HD, CQ, PG)                                                     \
+  hipLaunchKernelGGL((batch_prefill_kernel<T, HD, CQ, PG, false, true>), g, blk, 0, \
+                     stream, p)
 #define DISPATCH_PC(HD)                                         \
   do {                                                          \
-    if (paged) {                                                \
+    if (p.mask_data) {                                          \
+      if (paged) {                                              \
+        if (p.cta_q == 256) LAUNCH_PFM(HD, 256, true);          \
+        else LAUNCH_PFM(HD, 128, true);                         \
+      } else {                                                  \
+        if (p.cta_q == 256) LAUNCH_PFM(HD, 256, false);         \
+        else LAUNCH_PFM(HD, 128, false);                        \
+      }                                                         \
+    } else if (paged) {                                         \
       if (p.causal) LAUNCH_PF(HD, true, true);                  \
       else LAUNCH_PF(HD, true, false);                          \
     } else {                                                    \
@@ -363,6 +381,7 @@ hipError_t prefill_dispatch(PrefillParams& p, bool paged, hipStream_t stream) {
 #undef DISPATCH_PC
 #undef LAUNCH_PF
 #undef LAUNCH_PF2
+#undef LAUNCH_PFM
   return hipGetLastError();
 }
 

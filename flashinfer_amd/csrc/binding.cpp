@@ -289,7 +289,9 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                        int64_t layout, at::Tensor tile_req, at::Tensor tile_qstart,
                        at::Tensor out, c10::optional<at::Tensor> lse, double sm_scale,
                        double logits_soft_cap, int64_t window_left, bool causal,
-                       bool paged, int64_t cta_q) {
+                       bool paged, int64_t cta_q,
+                       c10::optional<at::Tensor> mask_data,
+                       c10::optional<at::Tensor> mask_byte_indptr) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 3, "q must be [nnz, Hq, D]");
   TORCH_CHECK(q.stride(2) == 1 && out.stride(2) == 1);
   fi_ext::PrefillParams p{};
@@ -337,6 +339,9 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   p.window_left = (int)window_left;
   p.causal = causal ? 1 : 0;
   p.cta_q = (int)cta_q;
+  p.mask_data = mask_data.has_value() ? mask_data->data_ptr<uint8_t>() : nullptr;
+  p.mask_byte_indptr =
+      mask_byte_indptr.has_value() ? mask_byte_indptr->data_ptr<int32_t>() : nullptr;
   check_hip(fi_batch_prefill(dtype_code(q), &p, paged ? 1 : 0, cur_stream(q)),
             "fi_batch_prefill");
 }

@@ -70,6 +70,8 @@ struct PrefillParams {
   int window_left;        // -1 disabled
   int causal;
   int cta_q;              // 128 or 256 packed q rows per tile
+  const uint8_t* mask_data;        // packed bitmask (little), or null
+  const int32_t* mask_byte_indptr; // per-request byte offset into mask_data
 };
 
 struct SamplingParams {

@@ -310,3 +310,10 @@ def single_decode_with_kv_cache(
     if v_scale is not None:
         out = out * v_scale
     return (out, lse.squeeze(0)) if return_lse else out
+
+
+def fast_decode_plan(wrapper, *args, **kwargs):
+    r"""SGLang-style replanning hook (reference decode.py:3826): same contract
+    as :meth:`BatchDecodeWithPagedKVCacheWrapper.plan`, kept as a free
+    function so serving engines can monkey-patch it."""
+    return BatchDecodeWithPagedKVCacheWrapper.plan(wrapper, *args, **kwargs)

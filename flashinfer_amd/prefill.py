@@ -12,7 +12,7 @@ from typing import Optional, Tuple, Union
 import torch
 
 from ._lib import get_ext
-from .utils import default_sm_scale, layout_code, unpack_paged_kv_cache
+from .utils import ceil_div, default_sm_scale, layout_code, unpack_paged_kv_cache
 
 def _plan_tiles(qo_lens, group: int, causal: bool = False):
     """Pick the q-tile size (128 or 256 packed rows) from the average packed
@@ -49,6 +49,33 @@ class _BatchPrefillBase:
 
     def reset_workspace_buffer(self, float_workspace_buffer, int_workspace_buffer=None):
         self._float_workspace_buffer = float_workspace_buffer
+
+    def _pack_mask(self, custom_mask, packed_custom_mask, qo_lens, kv_lens):
+        """Pack (or accept pre-packed) per-request [qo_len, kv_len] boolean
+        masks into byte-aligned little-endian segments (reference
+        quantization.packbits mechanism)."""
+        self._mask_data = None
+        self._mask_byte_indptr = None
+        if custom_mask is None and packed_custom_mask is None:
+            return
+        from .quantization import segment_packbits
+
+        qk_lens = [q * k for q, k in zip(qo_lens, kv_lens)]
+        if packed_custom_mask is not None:
+            byte_indptr = torch.zeros(len(qk_lens) + 1, dtype=torch.int32)
+            byte_indptr[1:] = torch.cumsum(
+                torch.tensor([ceil_div(L, 8) for L in qk_lens]), 0
+            ).int()
+            self._mask_data = packed_custom_mask.to(self.device)
+            self._mask_byte_indptr = byte_indptr.to(self.device)
+            return
+        qk_indptr = torch.zeros(len(qk_lens) + 1, dtype=torch.int64)
+        qk_indptr[1:] = torch.cumsum(torch.tensor(qk_lens), 0)
+        packed, byte_indptr = segment_packbits(
+            custom_mask.reshape(-1).to(self.device), qk_indptr.to(self.device)
+        )
+        self._mask_data = packed
+        self._mask_byte_indptr = byte_indptr.to(torch.int32)
 
     def _plan_common(self, qo_indptr, num_qo_heads, num_kv_heads, head_dim, causal,
                      sm_scale, window_left, logits_soft_cap, non_blocking=True):
@@ -90,7 +117,8 @@ class _BatchPrefillBase:
             kv_last_page_len, layout_code(self._kv_layout), self._tile_req,
             self._tile_qstart, out, lse if return_lse else None, sm_scale,
             pi["logits_soft_cap"], pi["window_left"], pi["causal"], paged,
-            pi["cta_q"],
+            pi["cta_q"], getattr(self, "_mask_data", None),
+            getattr(self, "_mask_byte_indptr", None),
         )
         if v_scale is not None:
             out = out * v_scale
@@ -114,8 +142,6 @@ class BatchPrefillWithPagedKVCacheWrapper(_BatchPrefillBase):
     ):
         if pos_encoding_mode != "NONE":
             raise NotImplementedError("apply RoPE beforehand")
-        if custom_mask is not None or packed_custom_mask is not None:
-            raise NotImplementedError("custom masks arrive in a later drop")
         self._plan_common(qo_indptr, num_qo_heads, num_kv_heads, head_dim_qk, causal,
                           sm_scale, window_left, logits_soft_cap, non_blocking)
         self._kv_indptr_d = paged_kv_indptr.to(self.device, torch.int32,
@@ -124,6 +150,15 @@ class BatchPrefillWithPagedKVCacheWrapper(_BatchPrefillBase):
                                                  non_blocking=non_blocking)
         self._kv_last_page_len_d = paged_kv_last_page_len.to(
             self.device, torch.int32, non_blocking=non_blocking)
+        if custom_mask is not None or packed_custom_mask is not None:
+            ip = paged_kv_indptr.to("cpu", torch.int64)
+            lp = paged_kv_last_page_len.to("cpu", torch.int64)
+            np_ = ip[1:] - ip[:-1]
+            kv_lens = (torch.clamp(np_ - 1, min=0) * page_size
+                       + torch.where(np_ > 0, lp, torch.zeros_like(lp))).tolist()
+            qi = qo_indptr.to("cpu", torch.int64)
+            qo_lens = (qi[1:] - qi[:-1]).tolist()
+            self._pack_mask(custom_mask, packed_custom_mask, qo_lens, kv_lens)
 
     begin_forward = plan
 
@@ -150,12 +185,15 @@ class BatchPrefillWithRaggedKVCacheWrapper(_BatchPrefillBase):
     ):
         if pos_encoding_mode != "NONE":
             raise NotImplementedError("apply RoPE beforehand")
-        if custom_mask is not None or packed_custom_mask is not None:
-            raise NotImplementedError("custom masks arrive in a later drop")
         self._plan_common(qo_indptr, num_qo_heads, num_kv_heads, head_dim_qk, causal,
                           sm_scale, window_left, logits_soft_cap, non_blocking)
         self._kv_indptr_d = kv_indptr.to(self.device, torch.int32,
                                          non_blocking=non_blocking)
+        if custom_mask is not None or packed_custom_mask is not None:
+            qi = qo_indptr.to("cpu", torch.int64)
+            ki = kv_indptr.to("cpu", torch.int64)
+            self._pack_mask(custom_mask, packed_custom_mask,
+                            (qi[1:] - qi[:-1]).tolist(), (ki[1:] - ki[:-1]).tolist())
 
     begin_forward = plan
 
@@ -181,8 +219,6 @@ def single_prefill_with_kv_cache(
     """
     if pos_encoding_mode != "NONE":
         raise NotImplementedError("apply RoPE beforehand")
-    if custom_mask is not None or packed_custom_mask is not None:
-        raise NotImplementedError("custom masks arrive in a later drop")
     if kv_layout == "HND":
         k = k.transpose(0, 1)
         v = v.transpose(0, 1)
@@ -197,11 +233,20 @@ def single_prefill_with_kv_cache(
     ).to(dev)
     out = torch.empty_like(q)
     lse = torch.empty(qo_len, Hq, dtype=torch.float32, device=dev) if return_lse else None
+    mask_data = mask_indptr = None
+    if custom_mask is not None or packed_custom_mask is not None:
+        if packed_custom_mask is None:
+            from .quantization import packbits
+
+            packed_custom_mask = packbits(custom_mask.reshape(-1))
+        mask_data = packed_custom_mask.to(dev)
+        mask_indptr = torch.zeros(2, dtype=torch.int32, device=dev)
     get_ext().batch_prefill_run(
         q, k, v, meta[2 * n_tiles : 2 * n_tiles + 2], None,
         meta[2 * n_tiles + 2 :], None, 0, meta[:n_tiles],
         meta[n_tiles : 2 * n_tiles], out, lse,
         sm_scale if sm_scale is not None else default_sm_scale(D),
         float(logits_soft_cap or 0.0), window_left, causal, False, cta_q,
+        mask_data, mask_indptr,
     )
     return (out, lse) if return_lse else out
