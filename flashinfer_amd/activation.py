@@ -6,6 +6,8 @@ from typing import Optional
 import torch
 
 from ._lib import get_ext
+from .api_logging import flashinfer_api
+from .fi_trace import fi_trace
 
 
 def _act(which: int, input: torch.Tensor, out: Optional[torch.Tensor]) -> torch.Tensor:
@@ -18,6 +20,8 @@ def _act(which: int, input: torch.Tensor, out: Optional[torch.Tensor]) -> torch.
     return out
 
 
+@flashinfer_api
+@fi_trace
 def silu_and_mul(
     input: torch.Tensor, out: Optional[torch.Tensor] = None,
     enable_pdl: Optional[bool] = None,

@@ -101,3 +101,15 @@ class MoeAlltoAll:
                                            device=state["order"].device)
         contrib = back[inv].view(T, K, -1)
         return (contrib * topk_weights.unsqueeze(-1).to(contrib.dtype)).sum(dim=1)
+
+    # ---- workspace checkpoint/restore (reference trtllm_moe_alltoall.py
+    # :1052/:1072 contract — engines snapshot comm state across hipGraph
+    # re-capture; the RCCL data plane is stateless, so this captures only the
+    # config) ----
+    def checkpoint_prepare(self) -> dict:
+        return dict(num_experts=self.num_experts, top_k=self.top_k,
+                    world=self.world, rank=self.rank)
+
+    def checkpoint_restore(self, state: dict) -> None:
+        assert state["num_experts"] == self.num_experts
+        assert state["world"] == self.world

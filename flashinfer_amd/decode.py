@@ -19,6 +19,8 @@ from typing import Optional, Tuple, Union
 import torch
 
 from ._lib import get_ext
+from .api_logging import flashinfer_api
+from .fi_trace import fi_trace
 from .utils import (
     WorkspaceAllocator,
     default_sm_scale,
@@ -240,6 +242,8 @@ class CUDAGraphBatchDecodeWithPagedKVCacheWrapper(BatchDecodeWithPagedKVCacheWra
         )
 
 
+@flashinfer_api
+@fi_trace
 def single_decode_with_kv_cache(
     q: torch.Tensor,
     k: torch.Tensor,

@@ -6,6 +6,8 @@ from typing import Optional
 import torch
 
 from ._lib import get_ext
+from .api_logging import flashinfer_api
+from .fi_trace import fi_trace
 
 
 def _as_nt(b: torch.Tensor) -> torch.Tensor:
@@ -19,6 +21,8 @@ def _as_nt(b: torch.Tensor) -> torch.Tensor:
     )
 
 
+@flashinfer_api
+@fi_trace
 def mm_bf16(
     a: torch.Tensor, b: torch.Tensor, out: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:

@@ -6,12 +6,16 @@ from typing import Optional
 import torch
 
 from ._lib import get_ext
+from .api_logging import flashinfer_api
+from .fi_trace import fi_trace
 
 
 def _flat2d(x: torch.Tensor) -> torch.Tensor:
     return x.reshape(-1, x.shape[-1])
 
 
+@flashinfer_api
+@fi_trace
 def rmsnorm(
     input: torch.Tensor,
     weight: torch.Tensor,
@@ -40,6 +44,8 @@ def gemma_rmsnorm(
     return out
 
 
+@flashinfer_api
+@fi_trace
 def fused_add_rmsnorm(
     input: torch.Tensor,
     residual: torch.Tensor,

@@ -12,6 +12,8 @@ from typing import Optional, Tuple, Union
 import torch
 
 from ._lib import get_ext
+from .api_logging import flashinfer_api
+from .fi_trace import fi_trace
 from .utils import ceil_div, default_sm_scale, layout_code, unpack_paged_kv_cache
 
 def _plan_tiles(qo_lens, group: int, causal: bool = False):
@@ -205,6 +207,8 @@ class BatchPrefillWithRaggedKVCacheWrapper(_BatchPrefillBase):
     forward = run
 
 
+@flashinfer_api
+@fi_trace
 def single_prefill_with_kv_cache(
     q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     custom_mask=None, packed_custom_mask=None, causal: bool = False,

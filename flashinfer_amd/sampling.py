@@ -10,6 +10,8 @@ from typing import Optional, Tuple, Union
 import torch
 
 from ._lib import get_ext
+from .api_logging import flashinfer_api
+from .fi_trace import fi_trace
 
 _ROUNDS = 32  # max rejection rounds
 
@@ -66,6 +68,8 @@ def _sample(mode, from_logits, probs, indices, generator, top_k=None, top_p=None
     return out
 
 
+@flashinfer_api
+@fi_trace
 def sampling_from_probs(
     probs: torch.Tensor, indices: Optional[torch.Tensor] = None,
     deterministic: bool = True, generator: Optional[torch.Generator] = None,
