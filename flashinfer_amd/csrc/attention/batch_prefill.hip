@@ -159,44 +159,62 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
   const T* kbase = (const T*)p.k_data;
   const T* vbase = (const T*)p.v_data;
 
-  for (int64_t kv0 = kv_lo; kv0 < kv_hi; kv0 += KVB) {
-    // ---- stage K [KVB][D] (swizzled) and V^T [D][KVB] (swizzled) ----
-    {
-      constexpr int UNITS = KVB * HEAD_DIM / 8;  // 16B units
-      constexpr int ITER = UNITS / NTHREADS;
+  // ---- async-STAGE split (guide T14): global loads for tile n+1 are issued
+  // BEFORE tile n's compute (HBM latency hides under the MFMA phases); the
+  // LDS writes land between the two barriers after compute. ----
+  constexpr int S_UNITS = KVB * HEAD_DIM / 8;  // 16B units
+  constexpr int S_ITER = S_UNITS / NTHREADS;
+  shortx8 kreg[S_ITER], vreg[S_ITER];
+
+  auto stage_load = [&](int64_t kv0) {
 #pragma unroll
-      for (int it = 0; it < ITER; ++it) {
-        int u = tid + it * NTHREADS;
-        int row = u / (HEAD_DIM / 8);       // kv row in tile
-        int chunk = u % (HEAD_DIM / 8);     // 8-elem d chunk
-        int64_t kvpos = kv0 + row;
-        shortx8 kval = {}, vval = {};
-        if (kvpos < kv_len) {
-          int64_t off;
-          if constexpr (PAGED) {
-            uint32_t pg, entry;
-            p.page_size.divmod((uint32_t)kvpos, pg, entry);
-            off = (int64_t)page_ids[pg] * p.kv_stride_page +
-                  (int64_t)kv_head * p.kv_stride_h + (int64_t)entry * p.kv_stride_n +
-                  chunk * 8;
-          } else {
-            off = (kv_base + kvpos) * p.kv_stride_n + (int64_t)kv_head * p.kv_stride_h +
-                  chunk * 8;
-          }
-          kval = *reinterpret_cast<const shortx8*>(kbase + off);
-          vval = *reinterpret_cast<const shortx8*>(vbase + off);
+    for (int it = 0; it < S_ITER; ++it) {
+      int u = tid + it * NTHREADS;
+      int row = u / (HEAD_DIM / 8);
+      int chunk = u % (HEAD_DIM / 8);
+      int64_t kvpos = kv0 + row;
+      kreg[it] = {};
+      vreg[it] = {};
+      if (kvpos < kv_len) {
+        int64_t off;
+        if constexpr (PAGED) {
+          uint32_t pg, entry;
+          p.page_size.divmod((uint32_t)kvpos, pg, entry);
+          off = (int64_t)page_ids[pg] * p.kv_stride_page +
+                (int64_t)kv_head * p.kv_stride_h + (int64_t)entry * p.kv_stride_n +
+                chunk * 8;
+        } else {
+          off = (kv_base + kvpos) * p.kv_stride_n + (int64_t)kv_head * p.kv_stride_h +
+                chunk * 8;
         }
-        // K: row-major swizzled (vector frag reads); V: row-major plain
-        // (transposed fragments are built with 8 scalar ds reads in PV)
-        *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Ks) +
-                                    swz_row<KROWB>(row * KROWB + chunk * 16)) = kval;
-        *reinterpret_cast<shortx8*>(
-            reinterpret_cast<char*>(Vs) +
-            ((row >> 2) * (HEAD_DIM / 16) + (chunk >> 1)) * (VTILE_STRIDE * 2) +
-            (row & 3) * 32 + (chunk & 1) * 16) = vval;
+        kreg[it] = *reinterpret_cast<const shortx8*>(kbase + off);
+        vreg[it] = *reinterpret_cast<const shortx8*>(vbase + off);
       }
     }
-    __syncthreads();
+  };
+  auto stage_write = [&]() {
+#pragma unroll
+    for (int it = 0; it < S_ITER; ++it) {
+      int u = tid + it * NTHREADS;
+      int row = u / (HEAD_DIM / 8);
+      int chunk = u % (HEAD_DIM / 8);
+      // K: row-major swizzled (vector frag reads); V: tr-read subtiled
+      *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Ks) +
+                                  swz_row<KROWB>(row * KROWB + chunk * 16)) = kreg[it];
+      *reinterpret_cast<shortx8*>(
+          reinterpret_cast<char*>(Vs) +
+          ((row >> 2) * (HEAD_DIM / 16) + (chunk >> 1)) * (VTILE_STRIDE * 2) +
+          (row & 3) * 32 + (chunk & 1) * 16) = vreg[it];
+    }
+  };
+
+  stage_load(kv_lo);
+  stage_write();
+  __syncthreads();
+
+  for (int64_t kv0 = kv_lo; kv0 < kv_hi; kv0 += KVB) {
+    const bool have_next = kv0 + KVB < kv_hi;
+    if (have_next) stage_load(kv0 + KVB);
 
     if (kv0 < wave_kv_hi) {
 #pragma unroll
@@ -316,7 +334,9 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
       }
     }
     }
-    __syncthreads();
+    __syncthreads();          // all waves done reading this tile's LDS
+    if (have_next) stage_write();
+    __syncthreads();          // next tile staged
   }
 
   // ---- epilogue: normalize and write O (transpose from O^T frags) ----
