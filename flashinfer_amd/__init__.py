@@ -20,6 +20,7 @@ from .cascade import (
 from .sparse import BlockSparseAttentionWrapper
 from .mla import BatchMLAPagedAttentionWrapper
 from .attention import BatchAttention, PODWithPagedKVCacheWrapper
+from .mamba import selective_state_update, ssu_checkpoint, ssu_rollback
 from .fused_moe import cutlass_fused_moe, dsv3_routing, fused_moe, moe_topk_softmax
 from .topk import top_k
 from .quantization import packbits, segment_packbits

@@ -221,6 +221,7 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
     for (int kt = 0; kt < KVB / 32; ++kt) {
       // ---- S^T = K * Q^T : [32 kv][32 q] ----
       floatx16 acc_s = {};
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int c = 0; c < KCH; ++c) {
         frag kfrag = *reinterpret_cast<const frag*>(
@@ -228,6 +229,7 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
             swz_row<KROWB>((kt * 32 + lq) * KROWB + (c * 16 + khalf) * 2));
         acc_s = mfma_ab_frag<T>::mma32(kfrag, qf[c], acc_s);
       }
+      __builtin_amdgcn_s_setprio(0);
 
       // ---- masking + base-2 logits -> p values (in place) ----
       float pr[16];
@@ -256,26 +258,32 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
 #pragma unroll
       for (int r = 1; r < 16; ++r) tmax = fmaxf(tmax, pr[r]);
       tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
-      float m_new = fmaxf(m_run, tmax);
+      // defer-max (guide T13): skip the O-wide rescale while the running max
+      // grows by < 8 (base-2) — p values stay bounded by 2^8, f32 accum is
+      // fine. __all keeps the wave branch-uniform.
+      bool defer = m_run != -INFINITY && __all(tmax - m_run <= 8.f);
+      float m_new = defer ? m_run : fmaxf(m_run, tmax);
       float f, psum = 0.f;
       if (m_new == -INFINITY) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) pr[r] = 0.f;
         f = 1.f;
       } else {
-        f = __builtin_exp2f(m_run - m_new);
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           pr[r] = (pr[r] == -INFINITY) ? 0.f : __builtin_exp2f(pr[r] - m_new);
           psum += pr[r];
         }
+        f = defer ? 1.f : __builtin_exp2f(m_run - m_new);
       }
-      m_run = m_new;
       d_run = d_run * f + psum;
+      if (!defer) {
+        m_run = m_new;
 #pragma unroll
-      for (int i = 0; i < DT; ++i) {
+        for (int i = 0; i < DT; ++i) {
 #pragma unroll
-        for (int r = 0; r < 16; ++r) acc_o[i][r] *= f;
+          for (int r = 0; r < 16; ++r) acc_o[i][r] *= f;
+        }
       }
 
       // ---- P^T fragments: pack to bf16 pairs, lazy half-exchange ----

@@ -74,6 +74,8 @@ hipError_t fi_packbits(const uint8_t* x, uint8_t* y, int64_t n, hipStream_t stre
 hipError_t fi_segment_packbits(const uint8_t* x, uint8_t* y, const int32_t* x_indptr,
                                const int32_t* y_indptr, int num_segments,
                                hipStream_t stream);
+hipError_t fi_selective_state_update(int dtype, fi_ext::SSUParams* p,
+                                     hipStream_t stream);
 }
 
 namespace {
@@ -573,6 +575,34 @@ void segment_packbits_op(at::Tensor x, at::Tensor y, at::Tensor x_indptr,
             "fi_segment_packbits");
 }
 
+void selective_state_update(at::Tensor state, at::Tensor x, at::Tensor dt,
+                            at::Tensor A, at::Tensor B, at::Tensor C,
+                            c10::optional<at::Tensor> D, c10::optional<at::Tensor> z,
+                            c10::optional<at::Tensor> dt_bias, at::Tensor out,
+                            bool dt_softplus) {
+  TORCH_CHECK(state.is_cuda() && state.dim() == 4 && state.is_contiguous());
+  fi_ext::SSUParams p{};
+  p.state = state.data_ptr();
+  p.x = x.data_ptr();
+  p.dt = dt.data_ptr();
+  p.A = A.data_ptr();
+  p.Bm = B.data_ptr();
+  p.Cm = C.data_ptr();
+  p.D = D.has_value() ? D->data_ptr() : nullptr;
+  p.z = z.has_value() ? z->data_ptr() : nullptr;
+  p.dt_bias = dt_bias.has_value() ? dt_bias->data_ptr() : nullptr;
+  p.out = out.data_ptr();
+  p.batch = state.size(0);
+  p.nheads = state.size(1);
+  p.headdim = state.size(2);
+  p.dstate = state.size(3);
+  p.ngroups = B.size(1);
+  p.dt_softplus = dt_softplus ? 1 : 0;
+  p.state_f32 = state.scalar_type() == at::kFloat && x.scalar_type() != at::kFloat;
+  check_hip(fi_selective_state_update(dtype_code(x), &p, cur_stream(x)),
+            "fi_selective_state_update");
+}
+
 // fastdiv self-check (host): returns n // d computed via the magic scheme.
 std::vector<int64_t> debug_fastdiv(int64_t d, std::vector<int64_t> ns) {
   fi::uint_fastdiv fd((uint32_t)d);
@@ -608,5 +638,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("topk", &topk_op);
   m.def("packbits", &packbits_op);
   m.def("segment_packbits", &segment_packbits_op);
+  m.def("selective_state_update", &selective_state_update);
   m.def("debug_fastdiv", &debug_fastdiv);
 }

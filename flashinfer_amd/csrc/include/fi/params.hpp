@@ -130,4 +130,21 @@ struct MlaParams {
   int causal;
 };
 
+struct SSUParams {
+  void* state;        // [B, H, P, S] (f32 or T)
+  const void* x;      // [B, H, P]
+  const void* dt;     // [B, H]
+  const void* A;      // [H]
+  const void* Bm;     // [B, G, S]
+  const void* Cm;     // [B, G, S]
+  const void* D;      // [H] or null
+  const void* z;      // [B, H, P] or null
+  const void* dt_bias;  // [H] or null
+  void* out;          // [B, H, P]
+  int batch, nheads, headdim, dstate, ngroups;
+  int dt_softplus;
+  int state_f32;  // state stored as f32 (else T)
+};
+
+
 }  // namespace fi
