@@ -34,7 +34,7 @@ def test_selective_state_update(state_f32, with_z):
     ref_state = state0.float() * dA[..., None, None] + (
         dtf[..., None] * x.float()
     )[..., None] * Bg[:, :, None, :]
-    ref_y = torch.einsum("bhps,bhs->bhp", ref_state, Cg) + D.float() * x.float()
+    ref_y = torch.einsum("bhps,bhs->bhp", ref_state, Cg) + D.float()[:, None] * x.float()
     if with_z:
         ref_y = ref_y * torch.nn.functional.silu(z.float())
     torch.testing.assert_close(state.float(), ref_state, atol=5e-2, rtol=5e-2)
