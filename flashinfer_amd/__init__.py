@@ -51,7 +51,7 @@ from .prefill import (
     BatchPrefillWithRaggedKVCacheWrapper,
     single_prefill_with_kv_cache,
 )
-from .gemm import mm_bf16
+from .gemm import SegmentGEMMWrapper, mm_bf16
 from . import sampling
 from .sampling import (
     chain_speculative_sampling,

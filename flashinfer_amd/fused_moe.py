@@ -68,29 +68,50 @@ def _permute(x: torch.Tensor, topk_ids: torch.Tensor, num_experts: int):
 
 def fused_moe(
     x: torch.Tensor,
-    w13: torch.Tensor,  # [E, 2*inter, hidden]  (gate | up rows)
+    w13: torch.Tensor,  # [E, 2*inter, hidden]  (gate | up rows; bf16 or fp8)
     w2: torch.Tensor,   # [E, hidden, inter]
     topk_weights: torch.Tensor,  # [T, k] f32
     topk_ids: torch.Tensor,      # [T, k] int
     activation: str = "silu",
     out: Optional[torch.Tensor] = None,
+    w13_scale: Optional[torch.Tensor] = None,  # [E, H/128, 2I/128] (fp8 path)
+    w2_scale: Optional[torch.Tensor] = None,   # [E, I/128, H/128]
 ) -> torch.Tensor:
+    r"""bf16 path: grouped bf16 MFMA GEMMs. fp8 path (weights in e4m3 +
+    128x128 block scales): activations are per-128-group quantized on the
+    fly and the grouped fp8 MFMA GEMM applies groupwise rescaling (the
+    reference's trtllm_fp8_block_scale_moe contract)."""
     T, H = x.shape
     E, I2, Hw = w13.shape
     assert Hw == H, "w13 must be [E, 2*inter, hidden]"
     inter = I2 // 2
     k = topk_ids.shape[1]
     ext = get_ext()
+    fp8 = w13.dtype == torch.float8_e4m3fn
 
     a_perm, m_indptr, order, token_of_copy = _permute(x, topk_ids, E)
     R = a_perm.shape[0]
     max_m_tiles = ceil_div(R, 128) + 1
 
-    h1 = torch.empty(R, I2, dtype=x.dtype, device=x.device)
-    ext.group_gemm_nt(a_perm, w13, h1, m_indptr, None, max_m_tiles)
+    h1 = torch.empty(R, I2, dtype=torch.bfloat16, device=x.device)
+    if fp8:
+        from .fp8_quantization import per_token_group_quant_fp8
+
+        a_q, a_s = per_token_group_quant_fp8(a_perm, transpose_scale=True)
+        ext.gemm_fp8_grouped(a_q.view(torch.uint8), w13.view(torch.uint8), h1,
+                             m_indptr, None, max_m_tiles, a_s.contiguous(),
+                             w13_scale.contiguous(), 1.0)
+    else:
+        ext.group_gemm_nt(a_perm, w13, h1, m_indptr, None, max_m_tiles)
     act = {"silu": silu_and_mul, "gelu": gelu_and_mul}[activation](h1)
-    h2 = torch.empty(R, H, dtype=x.dtype, device=x.device)
-    ext.group_gemm_nt(act, w2, h2, m_indptr, None, max_m_tiles)
+    h2 = torch.empty(R, H, dtype=torch.bfloat16, device=x.device)
+    if fp8:
+        act_q, act_s = per_token_group_quant_fp8(act, transpose_scale=True)
+        ext.gemm_fp8_grouped(act_q.view(torch.uint8), w2.view(torch.uint8), h2,
+                             m_indptr, None, max_m_tiles, act_s.contiguous(),
+                             w2_scale.contiguous(), 1.0)
+    else:
+        ext.group_gemm_nt(act, w2, h2, m_indptr, None, max_m_tiles)
 
     # finalize: out[token] = sum_j weight[t, j] * h2[row of (t, j)]
     inv = torch.empty_like(order)
@@ -119,3 +140,30 @@ def cutlass_fused_moe(
         input, fc1_expert_weights, fc2_expert_weights,
         token_final_scales, token_selected_experts,
     ).to(output_dtype)
+
+
+def trtllm_fp8_block_scale_moe(
+    routing_logits: torch.Tensor,
+    routing_bias: Optional[torch.Tensor],
+    hidden_states: torch.Tensor,
+    gemm1_weights: torch.Tensor,       # [E, 2*inter, hidden] fp8
+    gemm1_weights_scale: torch.Tensor, # [E, hidden/128, 2*inter/128]
+    gemm2_weights: torch.Tensor,       # [E, hidden, inter] fp8
+    gemm2_weights_scale: torch.Tensor, # [E, inter/128, hidden/128]
+    num_experts: int, top_k: int,
+    n_group: Optional[int] = None, topk_group: Optional[int] = None,
+    intermediate_size: Optional[int] = None,
+    routed_scaling_factor: float = 1.0,
+    routing_method_type: int = 0,
+    **kwargs,
+) -> torch.Tensor:
+    r"""Reference-compatible fp8 block-scale MoE (flashinfer
+    trtllm_fp8_block_scale_moe): DSv3 grouped routing when n_group is set,
+    renormalized top-k softmax otherwise."""
+    if n_group:
+        w, ids = dsv3_routing(routing_logits, top_k, n_group, topk_group,
+                              routed_scaling_factor, bias=routing_bias)
+    else:
+        w, ids = moe_topk_softmax(routing_logits, top_k)
+    return fused_moe(hidden_states, gemm1_weights, gemm2_weights, w, ids,
+                     w13_scale=gemm1_weights_scale, w2_scale=gemm2_weights_scale)

@@ -33,3 +33,49 @@ def mm_bf16(
         out = torch.empty(M, N, dtype=a.dtype, device=a.device)
     get_ext().gemm_nt(a, _as_nt(b), out, 1.0)
     return out
+
+
+class SegmentGEMMWrapper:
+    r"""Segment (grouped) GEMM for LoRA-style per-request weights (reference
+    parity: flashinfer/gemm/gemm_base.py SegmentGEMMWrapper:2352). Each
+    segment of x multiplies its own weight matrix (optionally indirected via
+    weight_indices)."""
+
+    def __init__(self, float_workspace_buffer: torch.Tensor, backend: str = "auto"):
+        self._ws = float_workspace_buffer
+        self.device = float_workspace_buffer.device
+
+    def reset_workspace_buffer(self, float_workspace_buffer, int_workspace_buffer=None):
+        self._ws = float_workspace_buffer
+
+    def run(
+        self,
+        x: torch.Tensor,          # [M, K]
+        weights: torch.Tensor,    # [num_w, N, K] (column-major per weight)
+        batch_size: int,
+        weight_column_major: bool = True,
+        seg_lens: Optional[torch.Tensor] = None,
+        seg_indptr: Optional[torch.Tensor] = None,
+        weight_indices: Optional[torch.Tensor] = None,
+        out: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        if not weight_column_major:
+            raise ValueError("weights must be [num_w, N, K] (column-major)")
+        if seg_indptr is None:
+            if seg_lens is None:
+                raise ValueError("need seg_lens or seg_indptr")
+            seg_indptr = torch.zeros(batch_size + 1, dtype=torch.int32,
+                                     device=x.device)
+            seg_indptr[1:] = seg_lens.to(x.device, torch.int64).cumsum(0).int()
+        M, N = x.shape[0], weights.shape[1]
+        if out is None:
+            out = torch.zeros(M, N, dtype=x.dtype, device=x.device)
+        max_m_tiles = (M + 127) // 128 + 1
+        get_ext().group_gemm_nt(
+            x, weights, out, seg_indptr.to(torch.int32),
+            weight_indices.to(torch.int32) if weight_indices is not None else None,
+            max_m_tiles,
+        )
+        return out
+
+    forward = run

@@ -64,3 +64,46 @@ def test_dsv3_routing_properties():
         assert groups[t].unique().numel() <= tkg
     torch.testing.assert_close(w.sum(-1), torch.full((T,), 2.5, device="cuda"),
                                atol=1e-4, rtol=1e-4)
+
+
+def test_fused_moe_fp8_matches_dequant():
+    from flashinfer_amd.fused_moe import fused_moe, moe_topk_softmax
+    from flashinfer_amd.fp8_quantization import per_block_quant_fp8
+
+    torch.manual_seed(0)
+    T, E, k, H, inter = 128, 4, 2, 256, 256
+    x = torch.randn(T, H, dtype=torch.bfloat16, device="cuda") / 4
+    w13 = torch.randn(E, 2 * inter, H, dtype=torch.bfloat16, device="cuda") / 8
+    w2 = torch.randn(E, H, inter, dtype=torch.bfloat16, device="cuda") / 8
+    w13_q, w13_s = per_block_quant_fp8(w13)
+    w2_q, w2_s = per_block_quant_fp8(w2)
+    logits = torch.randn(T, E, device="cuda")
+    weights, ids = moe_topk_softmax(logits, k)
+    out_fp8 = fused_moe(x, w13_q, w2_q, weights, ids,
+                        w13_scale=w13_s, w2_scale=w2_s)
+    out_bf16 = fused_moe(x, w13, w2, weights, ids)
+    # fp8 ~2 decimal digits; compare against the bf16 path loosely
+    err = (out_fp8.float() - out_bf16.float()).abs().mean() / out_bf16.float().abs().mean().clamp(min=1e-6)
+    assert err < 0.05, f"relative error {err}"
+
+
+def test_segment_gemm_wrapper():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    ws = torch.empty(8 * 1024 * 1024, dtype=torch.uint8, device="cuda")
+    seg = fi.SegmentGEMMWrapper(ws)
+    K, N = 128, 256
+    seg_lens = torch.tensor([17, 0, 40])
+    M = int(seg_lens.sum())
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda") / 4
+    weights = torch.randn(5, N, K, dtype=torch.bfloat16, device="cuda") / 4
+    w_idx = torch.tensor([4, 0, 2], dtype=torch.int32, device="cuda")
+    y = seg.run(x, weights, 3, seg_lens=seg_lens, weight_indices=w_idx)
+    off = 0
+    for i, L in enumerate(seg_lens.tolist()):
+        if L:
+            ref = x[off:off + L].float() @ weights[int(w_idx[i])].float().t()
+            torch.testing.assert_close(y[off:off + L].float(), ref,
+                                       atol=5e-2, rtol=5e-2)
+        off += L
