@@ -114,6 +114,37 @@ def bench_moe(T=4096, H=4096, inter=14336, E=8, k=2):
     print(f"fused_moe mixtral T={T}: {t*1e3:.2f} ms  {fl/t/1e12:.0f} TFLOPS  {T/t/1e6:.3f} M tok/s")
 
 
+
+
+def bench_moe_fp8(T=4096, H=4096, inter=14336, E=8, k=2):
+    from flashinfer_amd.fused_moe import fused_moe, moe_topk_softmax
+    from flashinfer_amd.fp8_quantization import per_block_quant_fp8
+    torch.manual_seed(0)
+    x = torch.randn(T, H, dtype=torch.bfloat16, device="cuda") / 4
+    w13 = torch.randn(E, 2 * inter, H, dtype=torch.bfloat16, device="cuda") / 16
+    w2 = torch.randn(E, H, inter, dtype=torch.bfloat16, device="cuda") / 16
+    w13_q, w13_s = per_block_quant_fp8(w13)
+    w2_q, w2_s = per_block_quant_fp8(w2)
+    del w13, w2
+    logits = torch.randn(T, E, device="cuda")
+    weights, ids = moe_topk_softmax(logits, k)
+    t = timeit(lambda: fused_moe(x, w13_q, w2_q, weights, ids, w13_scale=w13_s, w2_scale=w2_s), iters=10, warmup=3)
+    fl = T * k * 3 * H * inter * 2
+    print(f"fused_moe fp8 mixtral T={T}: {t*1e3:.2f} ms  {fl/t/1e12:.0f} TFLOPS  {T/t/1e6:.3f} M tok/s")
+
+
+def bench_fp8_gemm(M=4096, N=4096, K=4096):
+    from flashinfer_amd.fp8_quantization import (gemm_fp8_nt_groupwise, per_block_quant_fp8, per_token_group_quant_fp8)
+    torch.manual_seed(0)
+    a = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    b = torch.randn(N, K, dtype=torch.bfloat16, device="cuda")
+    a_q, a_s = per_token_group_quant_fp8(a, transpose_scale=True)
+    b_q, b_s = per_block_quant_fp8(b)
+    out = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+    t = timeit(lambda: gemm_fp8_nt_groupwise(a_q, b_q, a_s, b_s, out=out), iters=10)
+    print(f"fp8 groupwise gemm {M}x{N}x{K}: {t*1e3:.2f} ms  {2*M*N*K/t/1e12:.0f} TFLOPS")
+
+
 if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "all"
     if which in ("all", "prefill"):
@@ -132,3 +163,6 @@ if __name__ == "__main__":
         bench_mla(bs=64, kv=4096)
     if which in ("all", "moe"):
         bench_moe()
+        bench_moe_fp8()
+    if which in ("all", "fp8"):
+        bench_fp8_gemm()
