@@ -19,15 +19,18 @@ namespace fi {
 namespace f8gemm {
 
 constexpr int BM = 128, BN = 128, BK = 128;  // BK in fp8 elements (128 B rows)
-constexpr int NTH = 256;                     // 4 waves 2x2, 64x64 per wave
-constexpr int WM = 64, WN = 64;
+// 8 waves of 64x32 output each (2x4 wave grid): the groupwise variant keeps
+// TWO accumulators (per-K-group local + scaled master) — small per-wave
+// tiles keep the pair inside the register budget (no spill).
+constexpr int NTH = 512;
+constexpr int WM = 64, WN = 32;
 
 __device__ __forceinline__ void stage_tile(const uint8_t* __restrict__ gbase,
                                            int64_t ld, int row0, int rows_end, int k0,
                                            uint32_t lds_base_bytes, int tid) {
-  // BM x BK x 1B = 16 KB = 1024 16-B units over 256 threads
+  // BM x BK x 1B = 16 KB = 1024 16-B units over 512 threads
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
+  for (int i = 0; i < 2; ++i) {
     int u = tid + i * NTH;
     uint32_t dst_byte = (uint32_t)u * 16;
     uint32_t logical = swz128(dst_byte);
@@ -66,16 +69,14 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
-  const int wm = (wid >> 1) * WM;
-  const int wn = (wid & 1) * WN;
+  const int wm = (wid >> 2) * WM;
+  const int wn = (wid & 3) * WN;
   const int line = lane & 31;
   const int khalf = (lane >> 5) * 8;
 
-  floatx16 accm[2][2];
+  floatx16 accm[2];
 #pragma unroll
-  for (int i = 0; i < 2; ++i)
-#pragma unroll
-    for (int j = 0; j < 2; ++j) accm[i][j] = {};
+  for (int i = 0; i < 2; ++i) accm[i] = {};
 
   const uint32_t as_base = (uint32_t)(uintptr_t)&As[0][0];
   const uint32_t bs_base = (uint32_t)(uintptr_t)&Bs[0][0];
@@ -95,55 +96,40 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
     }
     const char* a_lds = (const char*)&As[cur][0];
     const char* b_lds = (const char*)&Bs[cur][0];
-    floatx16 acc[2][2];
+    floatx16 acc[2];
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
-#pragma unroll
-      for (int j = 0; j < 2; ++j) acc[i][j] = {};
+    for (int i = 0; i < 2; ++i) acc[i] = {};
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < BK / 16; ++ks) {
-      int64_t af[2], bfv[2];
+      int64_t af[2], bfv;
 #pragma unroll
       for (int i = 0; i < 2; ++i)
         af[i] = *reinterpret_cast<const int64_t*>(
             a_lds + (swz128((wm + i * 32 + line) * BK + ks * 16) + khalf));
+      bfv = *reinterpret_cast<const int64_t*>(
+          b_lds + (swz128((wn + line) * BK + ks * 16) + khalf));
 #pragma unroll
-      for (int j = 0; j < 2; ++j)
-        bfv[j] = *reinterpret_cast<const int64_t*>(
-            b_lds + (swz128((wn + j * 32 + line) * BK + ks * 16) + khalf));
-#pragma unroll
-      for (int i = 0; i < 2; ++i)
-#pragma unroll
-        for (int j = 0; j < 2; ++j)
-          acc[i][j] = mfma_32x32x16_fp8(af[i], bfv[j], acc[i][j]);
+      for (int i = 0; i < 2; ++i) acc[i] = mfma_32x32x16_fp8(af[i], bfv, acc[i]);
     }
     __builtin_amdgcn_s_setprio(0);
     // rescale local accumulator into master
     if constexpr (GROUPWISE) {
+      float sb = b_scales[((int64_t)seg * nk + kt) * ((N + 127) / 128) + (bn0 + wn) / 128];
 #pragma unroll
       for (int i = 0; i < 2; ++i) {
-        float sa[16];
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           int m = m0 + wm + i * 32 + mfma32_cd_row(r, lane);
-          sa[r] = a_scales[(int64_t)kt * a_scale_stride + (m < m_end ? m : m_end - 1)];
-        }
-#pragma unroll
-        for (int j = 0; j < 2; ++j) {
-          float sb = b_scales[((int64_t)seg * nk + kt) * ((N + 127) / 128) +
-                              (bn0 + wn + j * 32) / 128];
-#pragma unroll
-          for (int r = 0; r < 16; ++r) accm[i][j][r] += acc[i][j][r] * (sa[r] * sb);
+          float sa = a_scales[(int64_t)kt * a_scale_stride + (m < m_end ? m : m_end - 1)];
+          accm[i][r] += acc[i][r] * (sa * sb);
         }
       }
     } else {
 #pragma unroll
       for (int i = 0; i < 2; ++i)
 #pragma unroll
-        for (int j = 0; j < 2; ++j)
-#pragma unroll
-          for (int r = 0; r < 16; ++r) accm[i][j][r] += acc[i][j][r];
+        for (int r = 0; r < 16; ++r) accm[i][r] += acc[i][r];
     }
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
@@ -154,14 +140,11 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int m = m0 + wm + i * 32 + mfma32_cd_row(r, lane);
-        int n = bn0 + wn + j * 32 + mfma32_cd_col(lane);
-        if (m < m_end && n < N)
-          C[(int64_t)m * ldc + n] = from_f32<bf16>(accm[i][j][r] * fs);
-      }
+    for (int r = 0; r < 16; ++r) {
+      int m = m0 + wm + i * 32 + mfma32_cd_row(r, lane);
+      int n = bn0 + wn + mfma32_cd_col(lane);
+      if (m < m_end && n < N)
+        C[(int64_t)m * ldc + n] = from_f32<bf16>(accm[i][r] * fs);
     }
   }
 }
