@@ -93,3 +93,15 @@ def test_decode_hipgraph_capture():
     graph.replay()
     torch.cuda.synchronize()
     torch.testing.assert_close(out, eager2)
+
+
+def test_cu_mask_streams():
+    import flashinfer_amd as fi
+
+    streams, counts = fi.split_device_cu_streams(torch.device("cuda:0"), [192, 32])
+    assert counts == [192, 32]
+    x = torch.randn(1024, 1024, device="cuda")
+    with torch.cuda.stream(streams[1]):
+        y = x @ x
+    torch.cuda.synchronize()
+    assert y.isfinite().all()
