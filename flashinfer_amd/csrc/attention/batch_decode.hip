@@ -84,9 +84,18 @@ __global__ __launch_bounds__(256, 4) void batch_decode_kernel(DecodeParams p) {
     qreg[g].load(qbase + (int64_t)qh * p.q_stride_h + dcol);
   }
 
-  state_t<VPL> st[GROUP];
+  // base-2-domain online softmax with the single-FMA update:
+  //   s <= m:  p = exp2(s - m); d += p; o += p*v          (8 FMA)
+  //   s >  m:  r = exp2(m - s); d = d*r + 1; o = fma(o, r, v); m = s
+  float m_run[GROUP], d_run[GROUP], o_acc[GROUP][VPL];
 #pragma unroll
-  for (int g = 0; g < GROUP; ++g) st[g].init();
+  for (int g = 0; g < GROUP; ++g) {
+    m_run[g] = -INFINITY;
+    d_run[g] = 0.f;
+#pragma unroll
+    for (int j = 0; j < VPL; ++j) o_acc[g][j] = 0.f;
+  }
+  const float scale2 = scale * 1.4426950408889634f;
 
   const int32_t* page_ids = p.kv_indices + p.kv_indptr[req];
   const T* kbase = (const T*)p.k_data;
@@ -126,40 +135,67 @@ __global__ __launch_bounds__(256, 4) void batch_decode_kernel(DecodeParams p) {
       // reduce across the LPT lanes of this token
 #pragma unroll
       for (int off2 = LPT / 2; off2 > 0; off2 >>= 1) s += __shfl_xor(s, off2, 64);
-      s *= scale;
-      if constexpr (SOFT_CAP) s = p.logits_soft_cap * tanhf(s / p.logits_soft_cap);
-      if (!valid) s = -INFINITY;
-      st[g].push(vf, s);
+      float s2;
+      if constexpr (SOFT_CAP) {
+        s2 = p.logits_soft_cap * tanhf(s * scale / p.logits_soft_cap) *
+             1.4426950408889634f;
+      } else {
+        s2 = s * scale2;
+      }
+      if (valid) {
+        if (s2 <= m_run[g]) {
+          float pv = __builtin_exp2f(s2 - m_run[g]);
+          d_run[g] += pv;
+#pragma unroll
+          for (int j = 0; j < VPL; ++j) o_acc[g][j] += pv * vf[j];
+        } else {
+          float r = __builtin_exp2f(m_run[g] - s2);  // exp2(-inf)=0 first time
+          d_run[g] = d_run[g] * r + 1.f;
+#pragma unroll
+          for (int j = 0; j < VPL; ++j) o_acc[g][j] = __builtin_fmaf(o_acc[g][j], r, vf[j]);
+          m_run[g] = s2;
+        }
+      }
     }
     kv_cur = kv_nxt;
     vv_cur = vv_nxt;
   }
 
   // merge the TPW per-token states across the wave: lanes with equal (lane%LPT)
-  // hold the same output slice; exchange via shfl_xor at widths LPT, 2*LPT, ...
+  // hold the same output slice; exchange via shfl_xor (base-2 merge)
 #pragma unroll
   for (int g = 0; g < GROUP; ++g) {
 #pragma unroll
     for (int w = LPT; w < kWaveSize; w <<= 1) {
-      float m_o = __shfl_xor(st[g].m, w, 64);
-      float d_o = __shfl_xor(st[g].d, w, 64);
+      float m_o = __shfl_xor(m_run[g], w, 64);
+      float d_o = __shfl_xor(d_run[g], w, 64);
       float o_o[VPL];
 #pragma unroll
-      for (int j = 0; j < VPL; ++j) o_o[j] = __shfl_xor(st[g].o[j], w, 64);
-      st[g].merge(o_o, m_o, d_o);
+      for (int j = 0; j < VPL; ++j) o_o[j] = __shfl_xor(o_acc[g][j], w, 64);
+      float m_new = fmaxf(m_run[g], m_o);
+      if (m_new != -INFINITY) {
+        float s1 = __builtin_exp2f(m_run[g] - m_new);
+        float s2x = __builtin_exp2f(m_o - m_new);
+        d_run[g] = d_run[g] * s1 + d_o * s2x;
+#pragma unroll
+        for (int j = 0; j < VPL; ++j) o_acc[g][j] = o_acc[g][j] * s1 + o_o[j] * s2x;
+        m_run[g] = m_new;
+      }
     }
   }
 
-  // lanes of token-sub 0 write the result (normalized partial + lse)
+  // lanes of token-sub 0 write the result (normalized partial + base-2 lse)
   if (tsub == 0) {
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
       int qh = kv_head * GROUP + g;
-      float inv_d = st[g].d > 0.f ? 1.f / st[g].d : 0.f;
+      float inv_d = d_run[g] > 0.f ? 1.f / d_run[g] : 0.f;
       float* vout = p.tmp_v + ((int64_t)item * p.num_qo_heads + qh) * HEAD_DIM + dcol;
 #pragma unroll
-      for (int j = 0; j < VPL; ++j) vout[j] = st[g].o[j] * inv_d;
-      if (lane == 0) p.tmp_s[(int64_t)item * p.num_qo_heads + qh] = st[g].lse() * 1.4426950408889634f;  // base-2
+      for (int j = 0; j < VPL; ++j) vout[j] = o_acc[g][j] * inv_d;
+      if (lane == 0)
+        p.tmp_s[(int64_t)item * p.num_qo_heads + qh] =
+            d_run[g] > 0.f ? m_run[g] + __builtin_log2f(d_run[g]) : -INFINITY;
     }
   }
 }
