@@ -232,25 +232,43 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
       __builtin_amdgcn_s_setprio(0);
 
       // ---- masking + base-2 logits -> p values (in place) ----
+      // interior fast path: when the whole 32-kv tile is in-bounds, inside
+      // every causal/window bound for this wave's rows, and unmasked, the
+      // per-element predicate work (the VALU hotspot) is skipped.
       float pr[16];
       const int64_t kvt0 = kv0 + kt * 32;
+      bool tile_full = (kvt0 + 32 <= kv_len) && !MASK && cap <= 0.f;
+      if constexpr (CAUSAL) {
+        int wave_min_qpos = wq0 / (int)group;
+        tile_full &= (kvt0 + 32 <= wave_min_qpos + diag + 1);
+      }
+      if (wleft >= 0) {
+        int wave_max_qpos2 = (wq0 + 31) / (int)group;
+        tile_full &= (kvt0 >= wave_max_qpos2 + diag - wleft);
+      }
+      if (tile_full) {
+        const float sc2 = p.sm_scale * kLog2e;
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int64_t kv = kvt0 + mfma32_cd_row(r, lane);
-        float sv = acc_s[r] * p.sm_scale;
-        if (cap > 0.f) sv = cap * tanhf(sv / cap);
-        sv *= kLog2e;
-        bool ok = kv < kv_len;
-        if constexpr (CAUSAL) ok &= kv <= my_qpos + diag;
-        if (wleft >= 0) ok &= kv >= my_qpos + diag - wleft;
-        if constexpr (MASK) {
-          if (ok) {
-            int64_t bit = (int64_t)my_qpos * kv_len + kv;
-            uint8_t byte = p.mask_data[p.mask_byte_indptr[req] + (bit >> 3)];
-            ok &= (byte >> (bit & 7)) & 1;
+        for (int r = 0; r < 16; ++r) pr[r] = acc_s[r] * sc2;
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int64_t kv = kvt0 + mfma32_cd_row(r, lane);
+          float sv = acc_s[r] * p.sm_scale;
+          if (cap > 0.f) sv = cap * tanhf(sv / cap);
+          sv *= kLog2e;
+          bool ok = kv < kv_len;
+          if constexpr (CAUSAL) ok &= kv <= my_qpos + diag;
+          if (wleft >= 0) ok &= kv >= my_qpos + diag - wleft;
+          if constexpr (MASK) {
+            if (ok) {
+              int64_t bit = (int64_t)my_qpos * kv_len + kv;
+              uint8_t byte = p.mask_data[p.mask_byte_indptr[req] + (bit >> 3)];
+              ok &= (byte >> (bit & 7)) & 1;
+            }
           }
+          pr[r] = ok ? sv : -INFINITY;
         }
-        pr[r] = ok ? sv : -INFINITY;
       }
 
       // ---- online softmax update (per lane; exchange with lane^32) ----
