@@ -308,9 +308,10 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
       uint32_t W[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        uint32_t lo_ = float_to_bf16(pr[2 * j]);
-        uint32_t hi_ = float_to_bf16(pr[2 * j + 1]);
-        W[j] = lo_ | (hi_ << 16);
+        // hardware packed convert (no builtin on gfx950 — inline asm)
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2"
+            : "=v"(W[j])
+            : "v"(pr[2 * j]), "v"(pr[2 * j + 1]));
       }
       // NOTE: shfl must run on ALL lanes (uniform control flow) — select after.
       uint32_t X[8];

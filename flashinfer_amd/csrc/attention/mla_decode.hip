@@ -209,9 +209,9 @@ __global__ __launch_bounds__(NTH, 1) void mla_decode_kernel(MlaParams p) {
     uint32_t W[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      uint32_t lo_ = float_to_bf16(pr[2 * j]);
-      uint32_t hi_ = float_to_bf16(pr[2 * j + 1]);
-      W[j] = lo_ | (hi_ << 16);
+      asm("v_cvt_pk_bf16_f32 %0, %1, %2"
+          : "=v"(W[j])
+          : "v"(pr[2 * j]), "v"(pr[2 * j + 1]));
     }
     uint32_t X[8];
 #pragma unroll
