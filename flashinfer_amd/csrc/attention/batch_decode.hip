@@ -96,6 +96,14 @@ __global__ __launch_bounds__(256, 4) void batch_decode_kernel(DecodeParams p) {
     for (int j = 0; j < VPL; ++j) o_acc[g][j] = 0.f;
   }
   const float scale2 = scale * 1.4426950408889634f;
+  float alibi_slope2[GROUP];
+#pragma unroll
+  for (int g = 0; g < GROUP; ++g) {
+    alibi_slope2[g] = p.alibi ? __builtin_exp2f(-8.f * (kv_head * GROUP + g + 1) /
+                                                p.num_qo_heads) *
+                                    1.4426950408889634f
+                              : 0.f;
+  }
 
   const int32_t* page_ids = p.kv_indices + p.kv_indptr[req];
   const T* kbase = (const T*)p.k_data;
@@ -142,6 +150,7 @@ __global__ __launch_bounds__(256, 4) void batch_decode_kernel(DecodeParams p) {
       } else {
         s2 = s * scale2;
       }
+      if (p.alibi) s2 -= alibi_slope2[g] * (float)(kv_len - 1 - (pos0 + tsub));
       if (valid) {
         if (s2 <= m_run[g]) {
           float pv = __builtin_exp2f(s2 - m_run[g]);

@@ -237,7 +237,7 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
       // per-element predicate work (the VALU hotspot) is skipped.
       float pr[16];
       const int64_t kvt0 = kv0 + kt * 32;
-      bool tile_full = (kvt0 + 32 <= kv_len) && !MASK && cap <= 0.f;
+      bool tile_full = (kvt0 + 32 <= kv_len) && !MASK && cap <= 0.f && !p.alibi;
       if constexpr (CAUSAL) {
         int wave_min_qpos = wq0 / (int)group;
         tile_full &= (kvt0 + 32 <= wave_min_qpos + diag + 1);
@@ -251,11 +251,18 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
 #pragma unroll
         for (int r = 0; r < 16; ++r) pr[r] = acc_s[r] * sc2;
       } else {
+        // ALiBi slope for this lane's qo head: 2^(-8*(h+1)/Hq)
+        float slope = 0.f;
+        if (p.alibi) {
+          int qh = (int)(kv_head * group + my_g_u);
+          slope = __builtin_exp2f(-8.f * (qh + 1) / p.num_qo_heads);
+        }
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           int64_t kv = kvt0 + mfma32_cd_row(r, lane);
           float sv = acc_s[r] * p.sm_scale;
           if (cap > 0.f) sv = cap * tanhf(sv / cap);
+          if (p.alibi) sv -= slope * (float)(my_qpos + diag - kv);
           sv *= kLog2e;
           bool ok = kv < kv_len;
           if constexpr (CAUSAL) ok &= kv <= my_qpos + diag;

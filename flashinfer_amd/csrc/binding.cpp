@@ -250,7 +250,7 @@ void batch_decode_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                       at::Tensor kv_last_page_len, int64_t layout, at::Tensor work_req,
                       at::Tensor work_chunk, int64_t chunk_size, at::Tensor tmp_v,
                       at::Tensor tmp_s, double sm_scale, double logits_soft_cap,
-                      int64_t window_left) {
+                      int64_t window_left, bool alibi) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 3, "q must be [batch, num_qo_heads, head_dim]");
   fi_ext::DecodeParams p{};
   p.q = q.data_ptr();
@@ -278,6 +278,7 @@ void batch_decode_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   p.sm_scale = (float)sm_scale;
   p.logits_soft_cap = (float)logits_soft_cap;
   p.window_left = (int)window_left;
+  p.alibi = alibi ? 1 : 0;
   check_hip(fi_batch_decode(dtype_code(q), &p, cur_stream(q)), "fi_batch_decode");
 }
 
@@ -293,7 +294,7 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                        double logits_soft_cap, int64_t window_left, bool causal,
                        bool paged, int64_t cta_q,
                        c10::optional<at::Tensor> mask_data,
-                       c10::optional<at::Tensor> mask_byte_indptr) {
+                       c10::optional<at::Tensor> mask_byte_indptr, bool alibi) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 3, "q must be [nnz, Hq, D]");
   TORCH_CHECK(q.stride(2) == 1 && out.stride(2) == 1);
   fi_ext::PrefillParams p{};
@@ -344,6 +345,7 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   p.mask_data = mask_data.has_value() ? mask_data->data_ptr<uint8_t>() : nullptr;
   p.mask_byte_indptr =
       mask_byte_indptr.has_value() ? mask_byte_indptr->data_ptr<int32_t>() : nullptr;
+  p.alibi = alibi ? 1 : 0;
   check_hip(fi_batch_prefill(dtype_code(q), &p, paged ? 1 : 0, cur_stream(q)),
             "fi_batch_prefill");
 }

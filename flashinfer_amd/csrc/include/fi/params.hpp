@@ -45,6 +45,7 @@ struct DecodeParams {
   float sm_scale;
   float logits_soft_cap;  // 0 = disabled
   int window_left;        // -1 = disabled
+  int alibi;              // ALiBi position bias
 };
 
 struct PrefillParams {
@@ -72,6 +73,7 @@ struct PrefillParams {
   int cta_q;              // 128 or 256 packed q rows per tile
   const uint8_t* mask_data;        // packed bitmask (little), or null
   const int32_t* mask_byte_indptr; // per-request byte offset into mask_data
+  int alibi;                       // ALiBi position bias (slope by qo head)
 };
 
 struct SamplingParams {

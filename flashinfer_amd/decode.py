@@ -31,16 +31,24 @@ from .utils import (
 _TARGET_BLOCKS = 8192  # 1-wave blocks: ~32 per CU keeps enough loads in flight
 
 
-def _plan_chunks(kv_lens, num_kv_heads: int, page_size: int):
+def _plan_chunks(kv_lens, num_kv_heads: int, page_size: int,
+                 fixed_split_size=None, disable_split_kv=False):
     """Pick a split-KV chunk size and emit work items.
 
+    fixed_split_size / disable_split_kv give batch-invariant deterministic
+    partitioning (reference decode.py:1354 plan args).
     Returns (chunk_size, work_req, work_chunk, merge_indptr) as Python lists.
     """
     batch = len(kv_lens)
     total = sum(kv_lens)
     max_len = max(kv_lens) if batch else 0
-    items_target = max(batch, _TARGET_BLOCKS // max(1, num_kv_heads))
-    chunk = max(256, page_size, math.ceil(total / max(1, items_target)))
+    if disable_split_kv:
+        chunk = max(max_len, page_size)
+    elif fixed_split_size is not None:
+        chunk = int(fixed_split_size)
+    else:
+        items_target = max(batch, _TARGET_BLOCKS // max(1, num_kv_heads))
+        chunk = max(256, page_size, math.ceil(total / max(1, items_target)))
     # round up to page multiple so chunks don't straddle partially-read pages
     chunk = (chunk + page_size - 1) // page_size * page_size
     work_req, work_chunk, merge_indptr = [], [], [0]
@@ -110,11 +118,13 @@ class BatchDecodeWithPagedKVCacheWrapper:
         rope_scale: Optional[float] = None,
         rope_theta: Optional[float] = None,
         non_blocking: bool = True,
+        fixed_split_size: Optional[int] = None,
+        disable_split_kv: bool = False,
         **kwargs,
     ) -> None:
-        if pos_encoding_mode not in ("NONE",):
+        if pos_encoding_mode not in ("NONE", "ALIBI"):
             raise NotImplementedError(
-                "fused pos encodings in decode not supported; apply RoPE beforehand"
+                "pos_encoding_mode must be NONE or ALIBI; apply RoPE beforehand"
             )
         batch = indptr.shape[0] - 1
         indptr_h = indptr.to("cpu", torch.int64)
@@ -126,7 +136,7 @@ class BatchDecodeWithPagedKVCacheWrapper:
         ).tolist()
 
         chunk, work_req, work_chunk, merge_indptr = _plan_chunks(
-            kv_lens, num_kv_heads, page_size
+            kv_lens, num_kv_heads, page_size, fixed_split_size, disable_split_kv
         )
         n_items = len(work_req)
 
@@ -166,7 +176,7 @@ class BatchDecodeWithPagedKVCacheWrapper:
             window_left=window_left,
             logits_soft_cap=float(logits_soft_cap or 0.0),
             sm_scale=sm_scale if sm_scale is not None else default_sm_scale(head_dim),
-            q_data_type=q_data_type,
+            q_data_type=q_data_type, alibi=pos_encoding_mode == "ALIBI",
         )
 
     begin_forward = plan
@@ -181,6 +191,7 @@ class BatchDecodeWithPagedKVCacheWrapper:
         out: Optional[torch.Tensor] = None,
         lse: Optional[torch.Tensor] = None,
         return_lse: bool = False,
+        sinks: Optional[torch.Tensor] = None,
         **kwargs,
     ):
         pi = self._plan_info
@@ -200,17 +211,25 @@ class BatchDecodeWithPagedKVCacheWrapper:
             layout_code(self._kv_layout),
             self._work_req_d, self._work_chunk_d, pi["chunk"],
             self._tmp_v, self._tmp_s,
-            sm_scale, pi["logits_soft_cap"], pi["window_left"],
+            sm_scale, pi["logits_soft_cap"], pi["window_left"], pi["alibi"],
         )
         if out is None:
             out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
-        if return_lse and lse is None:
+        if (return_lse or sinks is not None) and lse is None:
             lse = torch.empty(
                 (pi["batch"], pi["num_qo_heads"]), dtype=torch.float32, device=q.device
             )
         get_ext().merge_states(
             self._tmp_v, self._tmp_s, out, lse, self._merge_indptr_d, 0, pi["batch"]
         )
+        if sinks is not None:
+            # attention sink: a per-head virtual logit enters the softmax
+            # denominator only: out *= 1 / (1 + e^sink / 2^lse)
+            import math as _m
+
+            w = 1.0 / (1.0 + torch.exp2(
+                sinks.float()[None, :] * _m.log2(_m.e) - lse))
+            out = (out.float() * w[..., None]).to(out.dtype)
         if v_scale is not None:
             out = out * v_scale
         return (out, lse) if return_lse else out
@@ -266,7 +285,7 @@ def single_decode_with_kv_cache(
     q: [num_qo_heads, head_dim]; k/v: [kv_len, num_kv_heads, head_dim] (NHD)
     or [num_kv_heads, kv_len, head_dim] (HND).
     """
-    if pos_encoding_mode != "NONE":
+    if pos_encoding_mode not in ("NONE", "ALIBI"):
         raise NotImplementedError("apply RoPE beforehand")
     head_dim = q.shape[-1]
     num_qo_heads = q.shape[0]
@@ -302,6 +321,7 @@ def single_decode_with_kv_cache(
         q.unsqueeze(0), k4, v4, indices, indptr, last_page_len,
         layout_code(kv_layout), work_req, work_chunk, chunk, tmp_v, tmp_s,
         sm_scale, float(logits_soft_cap or 0.0), window_left,
+        pos_encoding_mode == "ALIBI",
     )
     out = torch.empty_like(q.unsqueeze(0))
     lse = (

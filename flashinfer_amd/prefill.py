@@ -80,7 +80,8 @@ class _BatchPrefillBase:
         self._mask_byte_indptr = byte_indptr.to(torch.int32)
 
     def _plan_common(self, qo_indptr, num_qo_heads, num_kv_heads, head_dim, causal,
-                     sm_scale, window_left, logits_soft_cap, non_blocking=True):
+                     sm_scale, window_left, logits_soft_cap, non_blocking=True,
+                     alibi=False):
         qi = qo_indptr.to("cpu", torch.int64)
         qo_lens = (qi[1:] - qi[:-1]).tolist()
         group = num_qo_heads // num_kv_heads
@@ -98,7 +99,7 @@ class _BatchPrefillBase:
             causal=causal, window_left=window_left,
             logits_soft_cap=float(logits_soft_cap or 0.0),
             sm_scale=sm_scale if sm_scale is not None else default_sm_scale(head_dim),
-            nnz_q=int(qi[-1]), cta_q=cta_q,
+            nnz_q=int(qi[-1]), cta_q=cta_q, alibi=alibi,
         )
 
     def _run_common(self, q, k_cache, v_cache, kv_indices, kv_indptr, kv_last_page_len,
@@ -120,7 +121,7 @@ class _BatchPrefillBase:
             self._tile_qstart, out, lse if return_lse else None, sm_scale,
             pi["logits_soft_cap"], pi["window_left"], pi["causal"], paged,
             pi["cta_q"], getattr(self, "_mask_data", None),
-            getattr(self, "_mask_byte_indptr", None),
+            getattr(self, "_mask_byte_indptr", None), pi.get("alibi", False),
         )
         if v_scale is not None:
             out = out * v_scale
@@ -142,10 +143,11 @@ class BatchPrefillWithPagedKVCacheWrapper(_BatchPrefillBase):
         rope_scale=None, rope_theta=None, q_data_type=torch.bfloat16,
         kv_data_type=None, o_data_type=None, non_blocking: bool = True, **kwargs,
     ):
-        if pos_encoding_mode != "NONE":
+        if pos_encoding_mode not in ("NONE", "ALIBI"):
             raise NotImplementedError("apply RoPE beforehand")
         self._plan_common(qo_indptr, num_qo_heads, num_kv_heads, head_dim_qk, causal,
-                          sm_scale, window_left, logits_soft_cap, non_blocking)
+                          sm_scale, window_left, logits_soft_cap, non_blocking,
+                          alibi=pos_encoding_mode == "ALIBI")
         self._kv_indptr_d = paged_kv_indptr.to(self.device, torch.int32,
                                                non_blocking=non_blocking)
         self._kv_indices_d = paged_kv_indices.to(self.device, torch.int32,
@@ -185,10 +187,11 @@ class BatchPrefillWithRaggedKVCacheWrapper(_BatchPrefillBase):
         rope_scale=None, rope_theta=None, q_data_type=torch.bfloat16,
         kv_data_type=None, o_data_type=None, non_blocking: bool = True, **kwargs,
     ):
-        if pos_encoding_mode != "NONE":
+        if pos_encoding_mode not in ("NONE", "ALIBI"):
             raise NotImplementedError("apply RoPE beforehand")
         self._plan_common(qo_indptr, num_qo_heads, num_kv_heads, head_dim_qk, causal,
-                          sm_scale, window_left, logits_soft_cap, non_blocking)
+                          sm_scale, window_left, logits_soft_cap, non_blocking,
+                          alibi=pos_encoding_mode == "ALIBI")
         self._kv_indptr_d = kv_indptr.to(self.device, torch.int32,
                                          non_blocking=non_blocking)
         if custom_mask is not None or packed_custom_mask is not None:
@@ -221,7 +224,7 @@ def single_prefill_with_kv_cache(
 
     q: [qo_len, Hq, D]; k/v: [kv_len, Hkv, D] (NHD) or [Hkv, kv_len, D] (HND).
     """
-    if pos_encoding_mode != "NONE":
+    if pos_encoding_mode not in ("NONE", "ALIBI"):
         raise NotImplementedError("apply RoPE beforehand")
     if kv_layout == "HND":
         k = k.transpose(0, 1)
@@ -251,6 +254,6 @@ def single_prefill_with_kv_cache(
         meta[n_tiles : 2 * n_tiles], out, lse,
         sm_scale if sm_scale is not None else default_sm_scale(D),
         float(logits_soft_cap or 0.0), window_left, causal, False, cta_q,
-        mask_data, mask_indptr,
+        mask_data, mask_indptr, pos_encoding_mode == "ALIBI",
     )
     return (out, lse) if return_lse else out
