@@ -76,6 +76,11 @@ hipError_t fi_segment_packbits(const uint8_t* x, uint8_t* y, const int32_t* x_in
                                hipStream_t stream);
 hipError_t fi_selective_state_update(int dtype, fi_ext::SSUParams* p,
                                      hipStream_t stream);
+hipError_t fi_gather_rows(int dtype, const void* src, void* dst, const int32_t* row_map,
+                          int64_t rows, int cols, hipStream_t stream);
+hipError_t fi_moe_finalize(int dtype, const void* h, void* out, const int32_t* pos,
+                           const float* w, int64_t tokens, int topk, int cols,
+                           hipStream_t stream);
 }
 
 namespace {
@@ -605,6 +610,20 @@ void selective_state_update(at::Tensor state, at::Tensor x, at::Tensor dt,
             "fi_selective_state_update");
 }
 
+void gather_rows(at::Tensor src, at::Tensor dst, at::Tensor row_map) {
+  check_hip(fi_gather_rows(dtype_code(src), src.data_ptr(), dst.data_ptr(),
+                           row_map.data_ptr<int32_t>(), dst.size(0), src.size(1),
+                           cur_stream(src)),
+            "fi_gather_rows");
+}
+
+void moe_finalize(at::Tensor h, at::Tensor out, at::Tensor pos, at::Tensor w) {
+  check_hip(fi_moe_finalize(dtype_code(h), h.data_ptr(), out.data_ptr(),
+                            pos.data_ptr<int32_t>(), w.data_ptr<float>(), out.size(0),
+                            pos.size(1), h.size(1), cur_stream(h)),
+            "fi_moe_finalize");
+}
+
 // fastdiv self-check (host): returns n // d computed via the magic scheme.
 std::vector<int64_t> debug_fastdiv(int64_t d, std::vector<int64_t> ns) {
   fi::uint_fastdiv fd((uint32_t)d);
@@ -641,5 +660,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("packbits", &packbits_op);
   m.def("segment_packbits", &segment_packbits_op);
   m.def("selective_state_update", &selective_state_update);
+  m.def("gather_rows", &gather_rows);
+  m.def("moe_finalize", &moe_finalize);
   m.def("debug_fastdiv", &debug_fastdiv);
 }

@@ -61,8 +61,9 @@ def _permute(x: torch.Tensor, topk_ids: torch.Tensor, num_experts: int):
     counts = torch.bincount(flat, minlength=num_experts)
     m_indptr = torch.zeros(num_experts + 1, dtype=torch.int32, device=x.device)
     m_indptr[1:] = counts.cumsum(0).to(torch.int32)
-    token_of_copy = order // k  # source token per permuted row
-    a_perm = x.index_select(0, token_of_copy)
+    token_of_copy = (order // k).to(torch.int32)  # source token per permuted row
+    a_perm = torch.empty(T * k, x.shape[1], dtype=x.dtype, device=x.device)
+    get_ext().gather_rows(x, a_perm, token_of_copy)
     return a_perm, m_indptr, order, token_of_copy
 
 
@@ -116,11 +117,9 @@ def fused_moe(
     # finalize: out[token] = sum_j weight[t, j] * h2[row of (t, j)]
     inv = torch.empty_like(order)
     inv[order] = torch.arange(R, device=x.device)
-    contrib = h2.index_select(0, inv).view(T, k, H)
-    res = (contrib.float() * topk_weights.float().unsqueeze(-1)).sum(1).to(x.dtype)
-    if out is not None:
-        out.copy_(res)
-        return out
+    res = out if out is not None else torch.empty(T, H, dtype=x.dtype, device=x.device)
+    get_ext().moe_finalize(h2, res, inv.view(T, k).to(torch.int32),
+                           topk_weights.float().contiguous())
     return res
 
 
