@@ -47,8 +47,9 @@ __device__ __forceinline__ float qk_dot8(const void* qv, const void* kv, float a
   }
 }
 
-template <typename T, int HEAD_DIM, int GROUP, bool SOFT_CAP>
+template <typename T, typename TKV, int HEAD_DIM, int GROUP, bool SOFT_CAP>
 __global__ __launch_bounds__(256, 4) void batch_decode_kernel(DecodeParams p) {
+  constexpr bool kSameT = __is_same(T, TKV);
   constexpr int VPL = 8;
   constexpr int LPT = HEAD_DIM / VPL;        // lanes per token: 8/16/32
   constexpr int TPW = kWaveSize / LPT;       // tokens per wave: 8/4/2
@@ -74,14 +75,21 @@ __global__ __launch_bounds__(256, 4) void batch_decode_kernel(DecodeParams p) {
   }
 
   // stage q packed (dot2 consumes bf16 pairs; sm_scale applied to the
-  // reduced logit instead — 1 op per head per step)
+  // reduced logit instead — 1 op per head per step). With a quantized KV
+  // cache (TKV != T, e.g. fp8 e4m3) q is kept as f32 and the dot runs on
+  // converted values.
   vec_t<T, VPL> qreg[GROUP];
+  float qf32[kSameT ? 1 : GROUP][kSameT ? 1 : VPL];
   const T* qbase = (const T*)p.q + (int64_t)req * p.q_stride_n;
   const float scale = p.sm_scale;
 #pragma unroll
   for (int g = 0; g < GROUP; ++g) {
     int qh = kv_head * GROUP + g;
     qreg[g].load(qbase + (int64_t)qh * p.q_stride_h + dcol);
+    if constexpr (!kSameT) {
+#pragma unroll
+      for (int j = 0; j < VPL; ++j) qf32[g][j] = qreg[g].get(j);
+    }
   }
 
   // base-2-domain online softmax with the single-FMA update:
@@ -106,8 +114,8 @@ __global__ __launch_bounds__(256, 4) void batch_decode_kernel(DecodeParams p) {
   }
 
   const int32_t* page_ids = p.kv_indices + p.kv_indptr[req];
-  const T* kbase = (const T*)p.k_data;
-  const T* vbase = (const T*)p.v_data;
+  const TKV* kbase = (const TKV*)p.k_data;
+  const TKV* vbase = (const TKV*)p.v_data;
 
   // two-deep software pipeline: K/V loads for step i+1 are issued before the
   // softmax/state update of step i so the ~900-cycle HBM latency hides under
@@ -120,7 +128,7 @@ __global__ __launch_bounds__(256, 4) void batch_decode_kernel(DecodeParams p) {
     return (int64_t)page_ids[page_iter] * p.stride_page +
            (int64_t)kv_head * p.stride_h + (int64_t)entry * p.stride_n + dcol;
   };
-  vec_t<T, VPL> kv_cur, vv_cur, kv_nxt, vv_nxt;
+  vec_t<TKV, VPL> kv_cur, vv_cur, kv_nxt, vv_nxt;
   if (start < end) {
     int64_t off = addr_of(start);
     kv_cur.load(kbase + off);
@@ -139,7 +147,14 @@ __global__ __launch_bounds__(256, 4) void batch_decode_kernel(DecodeParams p) {
 
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
-      float s = qk_dot8<T>(qreg[g].data, kv_cur.data, 0.f);
+      float s;
+      if constexpr (kSameT) {
+        s = qk_dot8<T>(qreg[g].data, kv_cur.data, 0.f);
+      } else {
+        s = 0.f;
+#pragma unroll
+        for (int j = 0; j < VPL; ++j) s += qf32[g][j] * kv_cur.get(j);
+      }
       // reduce across the LPT lanes of this token
 #pragma unroll
       for (int off2 = LPT / 2; off2 > 0; off2 >>= 1) s += __shfl_xor(s, off2, 64);
@@ -209,14 +224,14 @@ __global__ __launch_bounds__(256, 4) void batch_decode_kernel(DecodeParams p) {
   }
 }
 
-template <typename T>
+template <typename T, typename TKV>
 hipError_t decode_dispatch(DecodeParams& p, hipStream_t stream) {
   int group = p.num_qo_heads / p.num_kv_heads;
   int64_t units = (int64_t)p.n_items * p.num_kv_heads;
   dim3 g((uint32_t)((units + 3) / 4)), blk(256);
   bool sc = p.logits_soft_cap > 0.f;
 #define LAUNCH_D(HD, G, SC) \
-  hipLaunchKernelGGL((batch_decode_kernel<T, HD, G, SC>), g, blk, 0, stream, p)
+  hipLaunchKernelGGL((batch_decode_kernel<T, TKV, HD, G, SC>), g, blk, 0, stream, p)
 #define DISPATCH_G(HD, SC)                              \
   do {                                                  \
     switch (group) {                                    \
@@ -247,12 +262,20 @@ hipError_t decode_dispatch(DecodeParams& p, hipStream_t stream) {
 
 }  // namespace fi
 
-extern "C" hipError_t fi_batch_decode(int dtype, fi::DecodeParams* p, hipStream_t stream) {
+// dtype = q dtype (0 bf16, 1 f16, 2 f32); kv_dtype additionally 3 = fp8 e4m3
+extern "C" hipError_t fi_batch_decode(int dtype, int kv_dtype, fi::DecodeParams* p,
+                                      hipStream_t stream) {
   if (p->n_items == 0) return hipSuccess;
-  switch (dtype) {
-    case 0: return fi::decode_dispatch<fi::bf16>(*p, stream);
-    case 1: return fi::decode_dispatch<fi::fp16>(*p, stream);
-    case 2: return fi::decode_dispatch<float>(*p, stream);
+  if (kv_dtype == dtype) {
+    switch (dtype) {
+      case 0: return fi::decode_dispatch<fi::bf16, fi::bf16>(*p, stream);
+      case 1: return fi::decode_dispatch<fi::fp16, fi::fp16>(*p, stream);
+      case 2: return fi::decode_dispatch<float, float>(*p, stream);
+    }
+  } else if (kv_dtype == 3 && dtype == 0) {
+    return fi::decode_dispatch<fi::bf16, fi::fp8_e4m3>(*p, stream);
+  } else if (kv_dtype == 3 && dtype == 1) {
+    return fi::decode_dispatch<fi::fp16, fi::fp8_e4m3>(*p, stream);
   }
   return hipErrorInvalidValue;
 }

@@ -16,8 +16,9 @@ hipError_t fi_norm(int which, int dtype, const void* x, const void* w, const voi
 hipError_t fi_act_and_mul(int which, int dtype, const void* in, void* out, int64_t tokens,
                           int d, hipStream_t stream);
 hipError_t fi_rope(int dtype, fi_ext::RopeParams* p, hipStream_t stream);
-hipError_t fi_append_paged_kv_cache(int dtype, void* k_data, void* v_data,
-                                    const int32_t* indices, const int32_t* indptr,
+hipError_t fi_append_paged_kv_cache(int dtype, int cache_dtype, void* k_data,
+                                    void* v_data, const int32_t* indices,
+                                    const int32_t* indptr,
                                     const int32_t* last_page_len, int page_size,
                                     int num_heads, int head_dim, int64_t stride_page,
                                     int64_t stride_n, int64_t stride_h, const void* k,
@@ -25,7 +26,7 @@ hipError_t fi_append_paged_kv_cache(int dtype, void* k_data, void* v_data,
                                     const int32_t* positions, int64_t nnz,
                                     int64_t k_stride_n, int64_t k_stride_h,
                                     int64_t v_stride_n, int64_t v_stride_h,
-                                    hipStream_t stream);
+                                    float k_scale, float v_scale, hipStream_t stream);
 hipError_t fi_batch_indices_positions(const int32_t* append_indptr, const int32_t* seq_lens,
                                       int32_t* batch_indices, int32_t* positions, int batch,
                                       hipStream_t stream);
@@ -36,7 +37,8 @@ hipError_t fi_merge_states(int in_dtype, int out_dtype, const void* v_in, const 
 hipError_t fi_merge_state_in_place(int dtype, void* v, float* s, const void* v_other,
                                    const float* s_other, int64_t num_pos, int num_heads,
                                    int head_dim, const uint8_t* mask, hipStream_t stream);
-hipError_t fi_batch_decode(int dtype, fi_ext::DecodeParams* p, hipStream_t stream);
+hipError_t fi_batch_decode(int dtype, int kv_dtype, fi_ext::DecodeParams* p,
+                           hipStream_t stream);
 hipError_t fi_gemm_nt(int dtype, const void* A, const void* B, void* C, int M, int N,
                       int K, int64_t lda, int64_t ldb, int64_t ldc, float alpha,
                       hipStream_t stream);
@@ -90,6 +92,8 @@ int dtype_code(const at::Tensor& t) {
     case at::kBFloat16: return 0;
     case at::kHalf: return 1;
     case at::kFloat: return 2;
+    case at::ScalarType::Float8_e4m3fn: return 3;
+    case at::kByte: return 3;  // fp8 cache passed as uint8 view
     default: TORCH_CHECK(false, "unsupported dtype ", t.scalar_type());
   }
 }
@@ -190,7 +194,8 @@ void apply_rope(at::Tensor q, at::Tensor k, at::Tensor q_out, at::Tensor k_out,
 void append_paged_kv_cache(at::Tensor k, at::Tensor v, at::Tensor batch_indices,
                            at::Tensor positions, at::Tensor k_cache, at::Tensor v_cache,
                            at::Tensor indices, at::Tensor indptr, at::Tensor last_page_len,
-                           int64_t layout /*0 NHD, 1 HND*/) {
+                           int64_t layout /*0 NHD, 1 HND*/, double k_scale,
+                           double v_scale) {
   TORCH_CHECK(k.is_cuda() && k.dim() == 3);
   // k_cache: NHD [pages, page_size, H, D] or HND [pages, H, page_size, D]
   int page_size = layout == 0 ? k_cache.size(1) : k_cache.size(2);
@@ -200,12 +205,14 @@ void append_paged_kv_cache(at::Tensor k, at::Tensor v, at::Tensor batch_indices,
   int64_t stride_n = layout == 0 ? k_cache.stride(1) : k_cache.stride(2);
   int64_t stride_h = layout == 0 ? k_cache.stride(2) : k_cache.stride(1);
   check_hip(fi_append_paged_kv_cache(
-                dtype_code(k), k_cache.data_ptr(), v_cache.data_ptr(),
-                indices.data_ptr<int32_t>(), indptr.data_ptr<int32_t>(),
-                last_page_len.data_ptr<int32_t>(), page_size, num_heads, head_dim,
-                stride_page, stride_n, stride_h, k.data_ptr(), v.data_ptr(),
-                batch_indices.data_ptr<int32_t>(), positions.data_ptr<int32_t>(), k.size(0),
-                k.stride(0), k.stride(1), v.stride(0), v.stride(1), cur_stream(k)),
+                dtype_code(k), dtype_code(k_cache), k_cache.data_ptr(),
+                v_cache.data_ptr(), indices.data_ptr<int32_t>(),
+                indptr.data_ptr<int32_t>(), last_page_len.data_ptr<int32_t>(), page_size,
+                num_heads, head_dim, stride_page, stride_n, stride_h, k.data_ptr(),
+                v.data_ptr(), batch_indices.data_ptr<int32_t>(),
+                positions.data_ptr<int32_t>(), k.size(0), k.stride(0), k.stride(1),
+                v.stride(0), v.stride(1), (float)k_scale, (float)v_scale,
+                cur_stream(k)),
             "fi_append_paged_kv_cache");
 }
 
@@ -284,7 +291,8 @@ void batch_decode_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   p.logits_soft_cap = (float)logits_soft_cap;
   p.window_left = (int)window_left;
   p.alibi = alibi ? 1 : 0;
-  check_hip(fi_batch_decode(dtype_code(q), &p, cur_stream(q)), "fi_batch_decode");
+  check_hip(fi_batch_decode(dtype_code(q), dtype_code(k_cache), &p, cur_stream(q)),
+            "fi_batch_decode");
 }
 
 // ---------------- prefill ----------------

@@ -55,6 +55,8 @@ template <>
 __device__ __forceinline__ float to_f32<fp16>(fp16 x) { return __half2float(x); }
 template <>
 __device__ __forceinline__ float to_f32<float>(float x) { return x; }
+template <>
+__device__ __forceinline__ float to_f32<fp8_e4m3>(fp8_e4m3 x) { return (float)x; }
 
 template <typename T>
 __device__ __forceinline__ T from_f32(float x);

@@ -43,10 +43,14 @@ def append_paged_kv_cache(
     kv_indptr: torch.Tensor,
     kv_last_page_len: torch.Tensor,
     kv_layout: str = "NHD",
+    k_scale: float = 1.0,
+    v_scale: float = 1.0,
 ) -> None:
     r"""Scatter ``nnz`` new tokens into the paged cache.
 
-    ``append_key``/``append_value``: [nnz, num_kv_heads, head_dim].
+    ``append_key``/``append_value``: [nnz, num_kv_heads, head_dim]. If the
+    cache is fp8 (e4m3) and the appended values are bf16/fp16, they are
+    quantized on the fly with the per-tensor scales.
     """
     k_cache, v_cache = unpack_paged_kv_cache(paged_kv_cache, kv_layout)
     get_ext().append_paged_kv_cache(
@@ -55,4 +59,5 @@ def append_paged_kv_cache(
         k_cache, v_cache,
         kv_indices.to(torch.int32), kv_indptr.to(torch.int32),
         kv_last_page_len.to(torch.int32), layout_code(kv_layout),
+        k_scale, v_scale,
     )

@@ -122,3 +122,57 @@ def test_decode_batch_invariance():
     # fixed split size -> identical partition -> bitwise identical results
     assert torch.equal(both[0], solo0[0])
     assert torch.equal(both[1], solo1[0])
+
+
+def test_decode_fp8_kv_cache():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    Hq, Hkv, D, kv, page, bs = 32, 8, 128, 1024, 16, 4
+    pp = kv // page
+    indptr = torch.arange(0, (bs + 1) * pp, pp, dtype=torch.int32, device="cuda")
+    npages = bs * pp
+    indices = torch.arange(npages, dtype=torch.int32, device="cuda")
+    last = torch.full((bs,), page, dtype=torch.int32, device="cuda")
+    k_scale, v_scale = 0.05, 0.04
+    kc8 = (torch.randn(npages, page, Hkv, D, device="cuda") * 8).to(
+        torch.float8_e4m3fn)
+    vc8 = (torch.randn(npages, page, Hkv, D, device="cuda") * 8).to(
+        torch.float8_e4m3fn)
+    q = torch.randn(bs, Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(128 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(indptr, indices, last, Hq, Hkv, D, page, q_data_type=torch.bfloat16,
+           kv_data_type=torch.float8_e4m3fn)
+    out = w.run(q, (kc8, vc8), k_scale=k_scale, v_scale=v_scale)
+    # reference on dequantized cache
+    from tests.test_decode import sdpa_ref
+
+    for b in range(bs):
+        rows_k = (kc8.float() * k_scale).view(-1, Hkv, D)[b * kv:(b + 1) * kv]
+        rows_v = (vc8.float() * v_scale).view(-1, Hkv, D)[b * kv:(b + 1) * kv]
+        ref = sdpa_ref(q[b], rows_k.bfloat16(), rows_v.bfloat16())
+        torch.testing.assert_close(out[b].float(), ref, atol=5e-2, rtol=5e-2)
+
+
+def test_append_fp8_quantize():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    H, D, page = 2, 64, 16
+    nnz = 40
+    npages = 4
+    indptr = torch.tensor([0, 4], dtype=torch.int32, device="cuda")
+    indices = torch.arange(npages, dtype=torch.int32, device="cuda")
+    last = torch.tensor([nnz - 3 * page], dtype=torch.int32, device="cuda")
+    kc = torch.zeros(npages, page, H, D, dtype=torch.float8_e4m3fn, device="cuda")
+    vc = torch.zeros(npages, page, H, D, dtype=torch.float8_e4m3fn, device="cuda")
+    k_new = torch.randn(nnz, H, D, dtype=torch.bfloat16, device="cuda")
+    v_new = torch.randn(nnz, H, D, dtype=torch.bfloat16, device="cuda")
+    append_indptr = torch.tensor([0, nnz], dtype=torch.int32, device="cuda")
+    seq_lens = torch.tensor([nnz], dtype=torch.int32, device="cuda")
+    bi, pos = fi.get_batch_indices_positions(append_indptr, seq_lens, nnz)
+    fi.append_paged_kv_cache(k_new, v_new, bi, pos, (kc, vc), indices, indptr,
+                             last, "NHD", k_scale=0.1, v_scale=0.2)
+    got_k = kc.view(-1, H, D).float()[:nnz] * 0.1
+    torch.testing.assert_close(got_k, k_new.float(), atol=0.05, rtol=0.1)
