@@ -8,6 +8,7 @@ from .comm_backend import TorchDistBackend, init_distributed
 from .mapping import Mapping
 from .moe_alltoall import MoeAlltoAll
 from .ulysses import UlyssesCommunicator
+from .dcp import dcp_gather_o, dcp_scatter_q
 
 __all__ = [
     "AllReduceFusionPattern",
@@ -18,4 +19,6 @@ __all__ = [
     "Mapping",
     "MoeAlltoAll",
     "UlyssesCommunicator",
+    "dcp_scatter_q",
+    "dcp_gather_o",
 ]

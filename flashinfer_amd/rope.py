@@ -169,6 +169,28 @@ def apply_rope_with_cos_sin_cache(
     return q_out, k_out
 
 
+def apply_rope_quantize_append(
+    q, k, v, pos_ids, paged_kv_cache, kv_indices, kv_indptr, kv_last_page_len,
+    batch_indices, positions, cos_sin_cache=None, rotary_dim=None,
+    interleave=False, rope_theta: float = 1e4, k_scale: float = 1.0,
+    v_scale: float = 1.0, kv_layout: str = "NHD",
+):
+    r"""RoPE -> (fp8) quantize -> paged append (reference
+    RopeQuantizeAppendPagedKVCacheKernel pos_enc.cuh:808 role; composed from
+    the rope and quantizing-append kernels here — two launches)."""
+    from .page import append_paged_kv_cache
+
+    q_out, k_out = _run_rope(
+        q, k, torch.empty_like(q), torch.empty_like(k), pos_ids,
+        cos_sin_cache=cos_sin_cache, rotary_dim=rotary_dim,
+        interleave=interleave, rope_theta=rope_theta,
+    )
+    append_paged_kv_cache(k_out, v, batch_indices, positions, paged_kv_cache,
+                          kv_indices, kv_indptr, kv_last_page_len, kv_layout,
+                          k_scale=k_scale, v_scale=v_scale)
+    return q_out
+
+
 def apply_rope_with_cos_sin_cache_inplace(
     positions: torch.Tensor, query: torch.Tensor, key: torch.Tensor,
     head_size: int, cos_sin_cache: torch.Tensor, is_neox: bool = True,
