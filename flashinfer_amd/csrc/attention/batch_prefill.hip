@@ -53,9 +53,14 @@ __device__ __forceinline__ uint32_t swz_row(uint32_t byte_off) {
 // CTAQ = packed q rows per workgroup (128 -> 4 waves, 256 -> 8 waves; the
 // planner picks by average packed length — bigger tiles amortize K/V staging
 // over more q rows).
-template <typename T, int HEAD_DIM, int CTAQ, bool PAGED, bool CAUSAL, bool MASK = false>
+template <typename T, typename TKV, int HEAD_DIM, int CTAQ, bool PAGED, bool CAUSAL,
+          bool MASK = false>
 __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParams p) {
   constexpr int NTHREADS = CTAQ * 2;
+  // fp8 (e4m3) KV caches are dequantized to bf16 during the LDS staging
+  // write (reference prefill.cuh:1150 repack_fp8_tile_to_bf16 design): the
+  // HBM bytes halve, the MFMA pipeline stays bf16.
+  constexpr bool kF8KV = !__is_same(T, TKV);
   constexpr int KCH = HEAD_DIM / 16;  // k-chunks in QK^T
   constexpr int DT = HEAD_DIM / 32;   // d-tiles in PV / output
   constexpr int KROWB = HEAD_DIM * 2;        // K/V tile row bytes
@@ -156,15 +161,15 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
     if (lo > 0) kv_lo = lo;
   }
 
-  const T* kbase = (const T*)p.k_data;
-  const T* vbase = (const T*)p.v_data;
+  const TKV* kbase = (const TKV*)p.k_data;
+  const TKV* vbase = (const TKV*)p.v_data;
 
   // ---- async-STAGE split (guide T14): global loads for tile n+1 are issued
   // BEFORE tile n's compute (HBM latency hides under the MFMA phases); the
   // LDS writes land between the two barriers after compute. ----
-  constexpr int S_UNITS = KVB * HEAD_DIM / 8;  // 16B units
+  constexpr int S_UNITS = KVB * HEAD_DIM / 8;  // 8-elem units
   constexpr int S_ITER = S_UNITS / NTHREADS;
-  shortx8 kreg[S_ITER], vreg[S_ITER];
+  vec_t<TKV, 8> kreg[S_ITER], vreg[S_ITER];
 
   auto stage_load = [&](int64_t kv0) {
 #pragma unroll
@@ -173,8 +178,8 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
       int row = u / (HEAD_DIM / 8);
       int chunk = u % (HEAD_DIM / 8);
       int64_t kvpos = kv0 + row;
-      kreg[it] = {};
-      vreg[it] = {};
+      kreg[it].fill(0.f);
+      vreg[it].fill(0.f);
       if (kvpos < kv_len) {
         int64_t off;
         if constexpr (PAGED) {
@@ -187,8 +192,8 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
           off = (kv_base + kvpos) * p.kv_stride_n + (int64_t)kv_head * p.kv_stride_h +
                 chunk * 8;
         }
-        kreg[it] = *reinterpret_cast<const shortx8*>(kbase + off);
-        vreg[it] = *reinterpret_cast<const shortx8*>(vbase + off);
+        kreg[it].load(kbase + off);
+        vreg[it].load(vbase + off);
       }
     }
   };
@@ -198,13 +203,25 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
       int u = tid + it * NTHREADS;
       int row = u / (HEAD_DIM / 8);
       int chunk = u % (HEAD_DIM / 8);
+      shortx8 kw, vw;
+      if constexpr (kF8KV) {
+        // dequantize fp8 -> bf16 on the staging write
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          reinterpret_cast<T*>(&kw)[j] = from_f32<T>(kreg[it].get(j) * p.k_descale);
+          reinterpret_cast<T*>(&vw)[j] = from_f32<T>(vreg[it].get(j) * p.v_descale);
+        }
+      } else {
+        kw = *reinterpret_cast<const shortx8*>(kreg[it].data);
+        vw = *reinterpret_cast<const shortx8*>(vreg[it].data);
+      }
       // K: row-major swizzled (vector frag reads); V: tr-read subtiled
       *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Ks) +
-                                  swz_row<KROWB>(row * KROWB + chunk * 16)) = kreg[it];
+                                  swz_row<KROWB>(row * KROWB + chunk * 16)) = kw;
       *reinterpret_cast<shortx8*>(
           reinterpret_cast<char*>(Vs) +
           ((row >> 2) * (HEAD_DIM / 16) + (chunk >> 1)) * (VTILE_STRIDE * 2) +
-          (row & 3) * 32 + (chunk & 1) * 16) = vreg[it];
+          (row & 3) * 32 + (chunk & 1) * 16) = vw;
     }
   };
 
@@ -395,19 +412,19 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
   }
 }
 
-template <typename T>
+template <typename T, typename TKV>
 hipError_t prefill_dispatch(PrefillParams& p, bool paged, hipStream_t stream) {
   dim3 g(p.n_tiles, p.num_kv_heads), blk(p.cta_q * 2);
 #define LAUNCH_PF2(HD, CQ, PG, CS) \
-  hipLaunchKernelGGL((batch_prefill_kernel<T, HD, CQ, PG, CS>), g, blk, 0, stream, p)
+  hipLaunchKernelGGL((batch_prefill_kernel<T, TKV, HD, CQ, PG, CS>), g, blk, 0, stream, p)
 #define LAUNCH_PF(HD, PG, CS)                        \
   do {                                               \
     if (p.cta_q == 256) LAUNCH_PF2(HD, 256, PG, CS); \
     else LAUNCH_PF2(HD, 128, PG, CS);                \
   } while (0)
-#define LAUNCH_PFM(HD, CQ, PG)                                                     \
-  hipLaunchKernelGGL((batch_prefill_kernel<T, HD, CQ, PG, false, true>), g, blk, 0, \
-                     stream, p)
+#define LAUNCH_PFM(HD, CQ, PG)                                                       \
+  hipLaunchKernelGGL((batch_prefill_kernel<T, TKV, HD, CQ, PG, false, true>), g, blk, \
+                     0, stream, p)
 #define DISPATCH_PC(HD)                                         \
   do {                                                          \
     if (p.mask_data) {                                          \
@@ -441,12 +458,16 @@ hipError_t prefill_dispatch(PrefillParams& p, bool paged, hipStream_t stream) {
 
 }  // namespace fi
 
-extern "C" hipError_t fi_batch_prefill(int dtype, fi::PrefillParams* p, int paged,
-                                       hipStream_t stream) {
+extern "C" hipError_t fi_batch_prefill(int dtype, int kv_dtype, fi::PrefillParams* p,
+                                       int paged, hipStream_t stream) {
   if (p->n_tiles == 0) return hipSuccess;
-  switch (dtype) {
-    case 0: return fi::prefill_dispatch<fi::bf16>(*p, paged, stream);
-    case 1: return fi::prefill_dispatch<fi::fp16>(*p, paged, stream);
+  if (kv_dtype == dtype) {
+    switch (dtype) {
+      case 0: return fi::prefill_dispatch<fi::bf16, fi::bf16>(*p, paged, stream);
+      case 1: return fi::prefill_dispatch<fi::fp16, fi::fp16>(*p, paged, stream);
+    }
+  } else if (kv_dtype == 3 && dtype == 0) {
+    return fi::prefill_dispatch<fi::bf16, fi::fp8_e4m3>(*p, paged, stream);
   }
   return hipErrorInvalidValue;
 }

@@ -42,8 +42,9 @@ __device__ __forceinline__ b16x4 ds_read_tr16(uint32_t lds_byte_off) {
   return r;
 }
 
-template <typename T, bool CAUSAL>
+template <typename T, typename TKV, bool CAUSAL>
 __global__ __launch_bounds__(NTH, 1) void mla_decode_kernel(MlaParams p) {
+  constexpr bool kF8KV = !__is_same(T, TKV);
   // KV tile in tr-subtiled layout + S-partial exchange buffer
   __shared__ T KVs[(KVB / 4) * NTILED * VTS];
   __shared__ float Sx[2][4][32 * 32];
@@ -109,8 +110,8 @@ __global__ __launch_bounds__(NTH, 1) void mla_decode_kernel(MlaParams p) {
 #pragma unroll
   for (int i = 0; i < 4; ++i) acc_o[i] = {};
 
-  const T* ckv = (const T*)p.ckv_data;
-  const T* kpe = (const T*)p.kpe_data;
+  const TKV* ckv = (const TKV*)p.ckv_data;
+  const TKV* kpe = (const TKV*)p.kpe_data;
 
   for (int64_t kv0 = start; kv0 < end; kv0 += KVB) {
     // ---- stage KV tile (subtiled layout; zero-fill OOB rows) ----
@@ -123,19 +124,31 @@ __global__ __launch_bounds__(NTH, 1) void mla_decode_kernel(MlaParams p) {
           int row = u / (D_QK / 8);
           int d = (u % (D_QK / 8)) * 8;
           int64_t kvpos = kv0 + row;
-          shortx8 val = {};
+          vec_t<TKV, 8> raw;
+          raw.fill(0.f);
+          float descale = 1.f;
           if (kvpos < end) {
             uint32_t pg, entry;
             p.page_size.divmod((uint32_t)kvpos, pg, entry);
             if (d < D_CKV) {
-              val = *reinterpret_cast<const shortx8*>(
-                  ckv + (int64_t)page_ids[pg] * p.ckv_stride_page +
-                  (int64_t)entry * p.ckv_stride_n + d);
+              raw.load(ckv + (int64_t)page_ids[pg] * p.ckv_stride_page +
+                       (int64_t)entry * p.ckv_stride_n + d);
+              descale = p.ckv_descale;
             } else {
-              val = *reinterpret_cast<const shortx8*>(
-                  kpe + (int64_t)page_ids[pg] * p.kpe_stride_page +
-                  (int64_t)entry * p.kpe_stride_n + (d - D_CKV));
+              raw.load(kpe + (int64_t)page_ids[pg] * p.kpe_stride_page +
+                       (int64_t)entry * p.kpe_stride_n + (d - D_CKV));
+              descale = p.kpe_descale;
             }
+          }
+          shortx8 val;
+          if constexpr (kF8KV) {
+            // fp8 -> bf16 dequant on the staging write (reference
+            // repack_fp8_tile_to_bf16 design)
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              reinterpret_cast<T*>(&val)[j] = from_f32<T>(raw.get(j) * descale);
+          } else {
+            val = *reinterpret_cast<const shortx8*>(raw.data);
           }
           *reinterpret_cast<shortx8*>(
               reinterpret_cast<char*>(KVs) +
@@ -282,23 +295,28 @@ __global__ __launch_bounds__(NTH, 1) void mla_decode_kernel(MlaParams p) {
 
 }  // namespace mla
 
-template <typename T>
+template <typename T, typename TKV>
 hipError_t mla_dispatch(MlaParams& p, hipStream_t stream) {
   dim3 g(p.n_items), blk(mla::NTH);
   if (p.causal)
-    hipLaunchKernelGGL((mla::mla_decode_kernel<T, true>), g, blk, 0, stream, p);
+    hipLaunchKernelGGL((mla::mla_decode_kernel<T, TKV, true>), g, blk, 0, stream, p);
   else
-    hipLaunchKernelGGL((mla::mla_decode_kernel<T, false>), g, blk, 0, stream, p);
+    hipLaunchKernelGGL((mla::mla_decode_kernel<T, TKV, false>), g, blk, 0, stream, p);
   return hipGetLastError();
 }
 
 }  // namespace fi
 
-extern "C" hipError_t fi_mla_decode(int dtype, fi::MlaParams* p, hipStream_t stream) {
+extern "C" hipError_t fi_mla_decode(int dtype, int kv_dtype, fi::MlaParams* p,
+                                    hipStream_t stream) {
   if (p->n_items == 0) return hipSuccess;
-  switch (dtype) {
-    case 0: return fi::mla_dispatch<fi::bf16>(*p, stream);
-    case 1: return fi::mla_dispatch<fi::fp16>(*p, stream);
+  if (kv_dtype == dtype) {
+    switch (dtype) {
+      case 0: return fi::mla_dispatch<fi::bf16, fi::bf16>(*p, stream);
+      case 1: return fi::mla_dispatch<fi::fp16, fi::fp16>(*p, stream);
+    }
+  } else if (kv_dtype == 3 && dtype == 0) {
+    return fi::mla_dispatch<fi::bf16, fi::fp8_e4m3>(*p, stream);
   }
   return hipErrorInvalidValue;
 }

@@ -45,14 +45,15 @@ hipError_t fi_gemm_nt(int dtype, const void* A, const void* B, void* C, int M, i
 hipError_t fi_gemm_nt_v2(const void* A, const void* B, void* C, int M, int N, int K,
                          int64_t lda, int64_t ldb, int64_t ldc, float alpha,
                          hipStream_t stream);
-hipError_t fi_batch_prefill(int dtype, fi_ext::PrefillParams* p, int paged,
-                            hipStream_t stream);
+hipError_t fi_batch_prefill(int dtype, int kv_dtype, fi_ext::PrefillParams* p,
+                            int paged, hipStream_t stream);
 hipError_t fi_softmax(fi_ext::SamplingParams* p, hipStream_t stream);
 hipError_t fi_sampling(int mode, int from_logits, fi_ext::SamplingParams* p,
                        hipStream_t stream);
 hipError_t fi_renorm(int which, fi_ext::SamplingParams* p, hipStream_t stream);
 hipError_t fi_chain_speculative(fi_ext::SpecParams* sp, hipStream_t stream);
-hipError_t fi_mla_decode(int dtype, fi_ext::MlaParams* p, hipStream_t stream);
+hipError_t fi_mla_decode(int dtype, int kv_dtype, fi_ext::MlaParams* p,
+                         hipStream_t stream);
 hipError_t fi_group_gemm_nt(const void* A, const void* W, void* C,
                             const int32_t* m_indptr, const int32_t* w_indices,
                             int num_segments, int max_m_tiles, int N, int K,
@@ -307,7 +308,8 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                        double logits_soft_cap, int64_t window_left, bool causal,
                        bool paged, int64_t cta_q,
                        c10::optional<at::Tensor> mask_data,
-                       c10::optional<at::Tensor> mask_byte_indptr, bool alibi) {
+                       c10::optional<at::Tensor> mask_byte_indptr, bool alibi,
+                       double k_descale, double v_descale) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 3, "q must be [nnz, Hq, D]");
   TORCH_CHECK(q.stride(2) == 1 && out.stride(2) == 1);
   fi_ext::PrefillParams p{};
@@ -359,7 +361,10 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   p.mask_byte_indptr =
       mask_byte_indptr.has_value() ? mask_byte_indptr->data_ptr<int32_t>() : nullptr;
   p.alibi = alibi ? 1 : 0;
-  check_hip(fi_batch_prefill(dtype_code(q), &p, paged ? 1 : 0, cur_stream(q)),
+  p.k_descale = (float)k_descale;
+  p.v_descale = (float)v_descale;
+  check_hip(fi_batch_prefill(dtype_code(q), dtype_code(k_cache), &p, paged ? 1 : 0,
+                             cur_stream(q)),
             "fi_batch_prefill");
 }
 
@@ -447,7 +452,7 @@ void mla_run(at::Tensor q_nope, at::Tensor q_pe, at::Tensor ckv_cache,
              at::Tensor kv_indptr, at::Tensor kv_last_page_len, at::Tensor tile_req,
              at::Tensor tile_row0, at::Tensor tile_chunk, int64_t chunk_size,
              int64_t max_chunks, at::Tensor tmp_v, at::Tensor tmp_s, double sm_scale,
-             bool causal) {
+             bool causal, double ckv_descale, double kpe_descale) {
   TORCH_CHECK(q_nope.is_cuda() && q_nope.dim() == 3 && q_nope.size(2) == 512);
   TORCH_CHECK(q_pe.size(2) == 64);
   TORCH_CHECK(ckv_cache.dim() == 3 && ckv_cache.size(2) == 512,
@@ -482,7 +487,11 @@ void mla_run(at::Tensor q_nope, at::Tensor q_pe, at::Tensor ckv_cache,
   p.tmp_s = tmp_s.data_ptr<float>();
   p.sm_scale = (float)sm_scale;
   p.causal = causal ? 1 : 0;
-  check_hip(fi_mla_decode(dtype_code(q_nope), &p, cur_stream(q_nope)), "fi_mla_decode");
+  p.ckv_descale = (float)ckv_descale;
+  p.kpe_descale = (float)kpe_descale;
+  check_hip(fi_mla_decode(dtype_code(q_nope), dtype_code(ckv_cache), &p,
+                          cur_stream(q_nope)),
+            "fi_mla_decode");
 }
 
 // ---------------- gemm ----------------

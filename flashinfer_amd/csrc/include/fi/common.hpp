@@ -66,6 +66,8 @@ template <>
 __device__ __forceinline__ fp16 from_f32<fp16>(float x) { return __float2half(x); }
 template <>
 __device__ __forceinline__ float from_f32<float>(float x) { return x; }
+template <>
+__device__ __forceinline__ fp8_e4m3 from_f32<fp8_e4m3>(float x) { return fp8_e4m3(x); }
 
 __host__ __device__ __forceinline__ constexpr int ceil_div(int a, int b) {
   return (a + b - 1) / b;

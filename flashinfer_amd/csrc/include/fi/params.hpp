@@ -74,6 +74,7 @@ struct PrefillParams {
   const uint8_t* mask_data;        // packed bitmask (little), or null
   const int32_t* mask_byte_indptr; // per-request byte offset into mask_data
   int alibi;                       // ALiBi position bias (slope by qo head)
+  float k_descale, v_descale;      // fp8-KV dequant factors (staging)
 };
 
 struct SamplingParams {
@@ -130,6 +131,7 @@ struct MlaParams {
   float* tmp_s;  // [nnz*H, max_chunks]
   float sm_scale;
   int causal;
+  float ckv_descale, kpe_descale;  // fp8-KV dequant factors (staging)
 };
 
 struct SSUParams {

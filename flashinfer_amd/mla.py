@@ -111,6 +111,8 @@ class BatchMLAPagedAttentionWrapper:
         out: Optional[torch.Tensor] = None,
         lse: Optional[torch.Tensor] = None,
         return_lse: bool = False,
+        ckv_scale: float = 1.0,
+        kpe_scale: float = 1.0,
         **kwargs,
     ):
         pi = self._plan_info
@@ -124,6 +126,7 @@ class BatchMLAPagedAttentionWrapper:
             self._kv_last_page_len_d, self._tile_req, self._tile_row0,
             self._tile_chunk, pi["chunk"], pi["max_chunks"],
             self._tmp_v, self._tmp_s, pi["sm_scale"], pi["causal"],
+            ckv_scale, kpe_scale,
         )
         nnz, H = q_nope.shape[0], pi["num_heads"]
         if out is None:

@@ -113,7 +113,13 @@ class _BatchPrefillBase:
             lse = torch.empty(q.shape[0], q.shape[1], dtype=torch.float32,
                               device=q.device)
         sm_scale = pi["sm_scale"]
-        if k_scale is not None:
+        kv_fp8 = k_cache.dtype in (torch.float8_e4m3fn, torch.uint8)
+        k_descale = v_descale = 1.0
+        if kv_fp8:
+            # fp8 KV: scales are applied in the dequantizing LDS staging
+            k_descale = k_scale if k_scale is not None else 1.0
+            v_descale = v_scale if v_scale is not None else 1.0
+        elif k_scale is not None:
             sm_scale *= k_scale
         get_ext().batch_prefill_run(
             q, k_cache, v_cache, self._qo_indptr_d, kv_indices, kv_indptr,
@@ -122,8 +128,9 @@ class _BatchPrefillBase:
             pi["logits_soft_cap"], pi["window_left"], pi["causal"], paged,
             pi["cta_q"], getattr(self, "_mask_data", None),
             getattr(self, "_mask_byte_indptr", None), pi.get("alibi", False),
+            k_descale, v_descale,
         )
-        if v_scale is not None:
+        if not kv_fp8 and v_scale is not None:
             out = out * v_scale
         return (out, lse) if return_lse else out
 
@@ -254,6 +261,6 @@ def single_prefill_with_kv_cache(
         meta[n_tiles : 2 * n_tiles], out, lse,
         sm_scale if sm_scale is not None else default_sm_scale(D),
         float(logits_soft_cap or 0.0), window_left, causal, False, cta_q,
-        mask_data, mask_indptr, pos_encoding_mode == "ALIBI",
+        mask_data, mask_indptr, pos_encoding_mode == "ALIBI", 1.0, 1.0,
     )
     return (out, lse) if return_lse else out

@@ -176,3 +176,58 @@ def test_append_fp8_quantize():
                              last, "NHD", k_scale=0.1, v_scale=0.2)
     got_k = kc.view(-1, H, D).float()[:nnz] * 0.1
     torch.testing.assert_close(got_k, k_new.float(), atol=0.05, rtol=0.1)
+
+
+def test_prefill_fp8_kv_cache():
+    import flashinfer_amd as fi
+    from tests.test_prefill import ref_attn
+
+    torch.manual_seed(0)
+    Hq, Hkv, D, page = 8, 2, 128, 16
+    qo = kv = 256
+    pp = kv // page
+    qo_indptr = torch.tensor([0, qo], dtype=torch.int32, device="cuda")
+    kv_indptr = torch.tensor([0, pp], dtype=torch.int32, device="cuda")
+    kv_indices = torch.arange(pp, dtype=torch.int32, device="cuda")
+    last = torch.tensor([page], dtype=torch.int32, device="cuda")
+    ks, vs = 0.07, 0.06
+    kc8 = (torch.randn(pp, page, Hkv, D, device="cuda") * 8).to(torch.float8_e4m3fn)
+    vc8 = (torch.randn(pp, page, Hkv, D, device="cuda") * 8).to(torch.float8_e4m3fn)
+    q = torch.randn(qo, Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(32 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchPrefillWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(qo_indptr, kv_indptr, kv_indices, last, Hq, Hkv, D, page, causal=True,
+           kv_data_type=torch.float8_e4m3fn)
+    out = w.run(q, (kc8, vc8), k_scale=ks, v_scale=vs)
+    kd = (kc8.float() * ks).view(kv, Hkv, D).bfloat16()
+    vd = (vc8.float() * vs).view(kv, Hkv, D).bfloat16()
+    ref = ref_attn(q, kd, vd, causal=True)
+    torch.testing.assert_close(out.float(), ref, atol=5e-2, rtol=5e-2)
+
+
+def test_mla_fp8_kv_cache():
+    import flashinfer_amd as fi
+    from tests.test_mla import mla_ref
+
+    torch.manual_seed(0)
+    H, page, kv = 32, 32, 256
+    pp = kv // page
+    sm = 1.0 / math.sqrt(576)
+    kv_indptr = torch.tensor([0, pp], dtype=torch.int32, device="cuda")
+    kv_indices = torch.arange(pp, dtype=torch.int32, device="cuda")
+    kv_len_arr = torch.tensor([kv], dtype=torch.int32, device="cuda")
+    qo_indptr = torch.tensor([0, 1], dtype=torch.int32, device="cuda")
+    cs, ps = 0.05, 0.05
+    ckv8 = (torch.randn(pp, page, 512, device="cuda") * 8).to(torch.float8_e4m3fn)
+    kpe8 = (torch.randn(pp, page, 64, device="cuda") * 8).to(torch.float8_e4m3fn)
+    qn = torch.randn(1, H, 512, dtype=torch.bfloat16, device="cuda") / 4
+    qp = torch.randn(1, H, 64, dtype=torch.bfloat16, device="cuda") / 4
+    ws = torch.empty(256 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchMLAPagedAttentionWrapper(ws)
+    w.plan(qo_indptr, kv_indptr, kv_indices, kv_len_arr, H, 512, 64, page, True,
+           sm, torch.bfloat16, kv_data_type=torch.float8_e4m3fn)
+    out = w.run(qn, qp, ckv8, kpe8, ckv_scale=cs, kpe_scale=ps)
+    ckv_d = (ckv8.float() * cs).view(kv, 512).bfloat16()
+    kpe_d = (kpe8.float() * ps).view(kv, 64).bfloat16()
+    ref = mla_ref(qn, qp, ckv_d, kpe_d, sm, True, 1, kv)
+    torch.testing.assert_close(out.float(), ref, atol=5e-2, rtol=5e-2)
