@@ -84,6 +84,10 @@ hipError_t fi_gather_rows(int dtype, const void* src, void* dst, const int32_t* 
 hipError_t fi_moe_finalize(int dtype, const void* h, void* out, const int32_t* pos,
                            const float* w, int64_t tokens, int topk, int cols,
                            hipStream_t stream);
+hipError_t fi_gdn_decode(int dtype, int state_f32, void* state, const void* q,
+                         const void* k, const void* v, const float* g,
+                         const float* beta, void* out, int B, int H, int Dk, int Dv,
+                         hipStream_t stream);
 }
 
 namespace {
@@ -641,6 +645,19 @@ void moe_finalize(at::Tensor h, at::Tensor out, at::Tensor pos, at::Tensor w) {
             "fi_moe_finalize");
 }
 
+void gdn_decode(at::Tensor state, at::Tensor q, at::Tensor k, at::Tensor v,
+                at::Tensor g, at::Tensor beta, at::Tensor out) {
+  TORCH_CHECK(state.is_cuda() && state.dim() == 4 && state.is_contiguous());
+  int B = state.size(0), H = state.size(1), Dk = state.size(2), Dv = state.size(3);
+  bool state_f32 =
+      state.scalar_type() == at::kFloat && q.scalar_type() != at::kFloat;
+  check_hip(fi_gdn_decode(dtype_code(q), state_f32, state.data_ptr(), q.data_ptr(),
+                          k.data_ptr(), v.data_ptr(), g.data_ptr<float>(),
+                          beta.data_ptr<float>(), out.data_ptr(), B, H, Dk, Dv,
+                          cur_stream(q)),
+            "fi_gdn_decode");
+}
+
 // fastdiv self-check (host): returns n // d computed via the magic scheme.
 std::vector<int64_t> debug_fastdiv(int64_t d, std::vector<int64_t> ns) {
   fi::uint_fastdiv fd((uint32_t)d);
@@ -679,5 +696,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("selective_state_update", &selective_state_update);
   m.def("gather_rows", &gather_rows);
   m.def("moe_finalize", &moe_finalize);
+  m.def("gdn_decode", &gdn_decode);
   m.def("debug_fastdiv", &debug_fastdiv);
 }
