@@ -22,7 +22,7 @@ from .mla import BatchMLAPagedAttentionWrapper
 from .attention import BatchAttention, PODWithPagedKVCacheWrapper
 from .mamba import selective_state_update, ssu_checkpoint, ssu_rollback
 from .green_ctx import split_device_cu_streams, split_device_green_ctx
-from .gdn import gdn_fused_decode_step
+from .gdn import fused_kda_decode, gdn_fused_decode_step
 from .fused_moe import cutlass_fused_moe, dsv3_routing, fused_moe, moe_topk_softmax
 from .topk import top_k
 from .quantization import packbits, segment_packbits

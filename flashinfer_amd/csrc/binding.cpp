@@ -84,8 +84,8 @@ hipError_t fi_gather_rows(int dtype, const void* src, void* dst, const int32_t* 
 hipError_t fi_moe_finalize(int dtype, const void* h, void* out, const int32_t* pos,
                            const float* w, int64_t tokens, int topk, int cols,
                            hipStream_t stream);
-hipError_t fi_gdn_decode(int dtype, int state_f32, void* state, const void* q,
-                         const void* k, const void* v, const float* g,
+hipError_t fi_gdn_decode(int dtype, int state_f32, int per_channel_gate, void* state,
+                         const void* q, const void* k, const void* v, const float* g,
                          const float* beta, void* out, int B, int H, int Dk, int Dv,
                          hipStream_t stream);
 }
@@ -651,10 +651,11 @@ void gdn_decode(at::Tensor state, at::Tensor q, at::Tensor k, at::Tensor v,
   int B = state.size(0), H = state.size(1), Dk = state.size(2), Dv = state.size(3);
   bool state_f32 =
       state.scalar_type() == at::kFloat && q.scalar_type() != at::kFloat;
-  check_hip(fi_gdn_decode(dtype_code(q), state_f32, state.data_ptr(), q.data_ptr(),
-                          k.data_ptr(), v.data_ptr(), g.data_ptr<float>(),
-                          beta.data_ptr<float>(), out.data_ptr(), B, H, Dk, Dv,
-                          cur_stream(q)),
+  bool per_channel = g.dim() == 3;
+  check_hip(fi_gdn_decode(dtype_code(q), state_f32, per_channel, state.data_ptr(),
+                          q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                          g.data_ptr<float>(), beta.data_ptr<float>(), out.data_ptr(),
+                          B, H, Dk, Dv, cur_stream(q)),
             "fi_gdn_decode");
 }
 

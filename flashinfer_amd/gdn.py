@@ -15,14 +15,22 @@ def gdn_fused_decode_step(
     q: torch.Tensor,      # [B, H, Dk]
     k: torch.Tensor,      # [B, H, Dk]
     v: torch.Tensor,      # [B, H, Dv]
-    g: torch.Tensor,      # [B, H] f32 gate decay (e.g. exp(a * dt))
+    g: torch.Tensor,      # [B, H] f32 gate decay, or [B, H, Dk] per-channel (KDA)
     beta: torch.Tensor,   # [B, H] f32 delta-rule step size
     out: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     r"""One gated-delta-rule step (updates state in place, returns o = q^T S):
-    ``S = g*S + k (x) (beta * (v - k^T (g*S)));  o = q^T S``."""
+    ``S = g*S + k (x) (beta * (v - k^T (g*S)));  o = q^T S``.
+    A 3-D gate enables the KDA-style per-channel (diagonal) decay."""
     if out is None:
         out = torch.empty_like(v)
     get_ext().gdn_decode(state, q.contiguous(), k.contiguous(), v.contiguous(),
                          g.float().contiguous(), beta.float().contiguous(), out)
     return out
+
+
+def fused_kda_decode(state, q, k, v, g_channel, beta, out=None):
+    r"""Kimi Delta Attention decode step (reference flashinfer/kda
+    fused_kda_decode role): delta rule with per-channel gate [B, H, Dk]."""
+    assert g_channel.dim() == 3
+    return gdn_fused_decode_step(state, q, k, v, g_channel, beta, out)

@@ -33,3 +33,25 @@ def test_gdn_decode_step(state_f32):
     for _ in range(8):
         gdn_fused_decode_step(S, q, k, v, g, beta)
     assert S.float().isfinite().all()
+
+
+def test_kda_per_channel_gate():
+    from flashinfer_amd.gdn import fused_kda_decode
+
+    torch.manual_seed(0)
+    B, H, Dk, Dv = 2, 4, 64, 128
+    S = torch.randn(B, H, Dk, Dv, device="cuda") / 8
+    S0 = S.clone()
+    q = torch.randn(B, H, Dk, device="cuda").bfloat16()
+    k = torch.randn(B, H, Dk, device="cuda").bfloat16()
+    v = (torch.randn(B, H, Dv, device="cuda") / 4).bfloat16()
+    g = torch.rand(B, H, Dk, device="cuda") * 0.9
+    beta = torch.rand(B, H, device="cuda")
+    o = fused_kda_decode(S, q, k, v, g, beta)
+    Sf = S0 * g[..., None]
+    kv = torch.einsum("bhi,bhij->bhj", k.float(), Sf)
+    delta = beta[..., None] * (v.float() - kv)
+    S_new = Sf + torch.einsum("bhi,bhj->bhij", k.float(), delta)
+    o_ref = torch.einsum("bhi,bhij->bhj", q.float(), S_new)
+    torch.testing.assert_close(S, S_new, atol=5e-2, rtol=5e-2)
+    torch.testing.assert_close(o.float(), o_ref, atol=5e-2, rtol=5e-2)
