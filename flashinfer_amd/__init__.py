@@ -17,12 +17,20 @@ from .cascade import (
     merge_state_in_place,
     merge_states,
 )
-from .sparse import BlockSparseAttentionWrapper
+from .sparse import (
+    BlockSparseAttentionWrapper,
+    VariableBlockSparseAttentionWrapper,
+)
 from .mla import BatchMLAPagedAttentionWrapper
 from .attention import BatchAttention, PODWithPagedKVCacheWrapper
 from .mamba import selective_state_update, ssu_checkpoint, ssu_rollback
 from .green_ctx import split_device_cu_streams, split_device_green_ctx
-from .gdn import fused_kda_decode, gdn_fused_decode_step
+from .gdn import (
+    chunk_gated_delta_rule,
+    chunk_kda,
+    fused_kda_decode,
+    gdn_fused_decode_step,
+)
 from .fused_moe import cutlass_fused_moe, dsv3_routing, fused_moe, moe_topk_softmax
 from .topk import top_k
 from .quantization import packbits, segment_packbits

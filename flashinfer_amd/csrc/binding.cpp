@@ -88,6 +88,11 @@ hipError_t fi_gdn_decode(int dtype, int state_f32, int per_channel_gate, void* s
                          const void* q, const void* k, const void* v, const float* g,
                          const float* beta, void* out, int B, int H, int Dk, int Dv,
                          hipStream_t stream);
+hipError_t fi_gdn_chunk(int dtype, int per_channel_gate, const void* q, const void* k,
+                        const void* v, const float* gate, const float* beta, void* out,
+                        const int32_t* cu_seqlens, const float* init_state,
+                        float* final_state, float scale, int num_seqs, int H, int D,
+                        hipStream_t stream);
 }
 
 namespace {
@@ -659,6 +664,24 @@ void gdn_decode(at::Tensor state, at::Tensor q, at::Tensor k, at::Tensor v,
             "fi_gdn_decode");
 }
 
+void gdn_chunk(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor gate,
+               at::Tensor beta, at::Tensor out, at::Tensor cu_seqlens,
+               std::optional<at::Tensor> init_state,
+               std::optional<at::Tensor> final_state, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.dim() == 3 && q.is_contiguous());
+  TORCH_CHECK(cu_seqlens.scalar_type() == at::kInt);
+  int H = q.size(1), D = q.size(2);
+  int num_seqs = cu_seqlens.size(0) - 1;
+  bool per_channel = gate.dim() == 3;
+  check_hip(fi_gdn_chunk(dtype_code(q), per_channel, q.data_ptr(), k.data_ptr(),
+                         v.data_ptr(), gate.data_ptr<float>(), beta.data_ptr<float>(),
+                         out.data_ptr(), cu_seqlens.data_ptr<int32_t>(),
+                         init_state ? init_state->data_ptr<float>() : nullptr,
+                         final_state ? final_state->data_ptr<float>() : nullptr,
+                         (float)scale, num_seqs, H, D, cur_stream(q)),
+            "fi_gdn_chunk");
+}
+
 // fastdiv self-check (host): returns n // d computed via the magic scheme.
 std::vector<int64_t> debug_fastdiv(int64_t d, std::vector<int64_t> ns) {
   fi::uint_fastdiv fd((uint32_t)d);
@@ -698,5 +721,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_rows", &gather_rows);
   m.def("moe_finalize", &moe_finalize);
   m.def("gdn_decode", &gdn_decode);
+  m.def("gdn_chunk", &gdn_chunk);
   m.def("debug_fastdiv", &debug_fastdiv);
 }

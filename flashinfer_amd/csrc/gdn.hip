@@ -59,7 +59,149 @@ __global__ void gdn_decode_kernel(TS* __restrict__ state, const T* __restrict__ 
   }
 }
 
+// ---------------------------------------------------------------------------
+// Chunked GDN/KDA prefill (reference parity: flashinfer/gdn_kernels/blackwell/
+// gdn_prefill.py chunk_gated_delta_rule_sm100:137 semantics, validated against
+// tests/gdn/reference_delta_rule.py blockwise_delta_rule:856). Per-token
+// recurrence over each ragged sequence:
+//   S   <- a_t * S                      (gate decay; scalar or per-channel)
+//   u_t <- beta_t * (v_t - k_t^T S)
+//   S   <- S + k_t (x) u_t
+//   o_t <- scale * q_t^T S
+// One block per (seq, head); the [D, D] f32 state lives in registers, sliced
+// as thread t owning column (t % D) rows [(t/D)*RPT, ...+RPT). Tokens are
+// staged through LDS in chunks of CH so the global loads amortize; the two
+// cross-row reductions per token combine GPC=256/D partials through LDS.
+template <typename T, int D, bool PER_CHANNEL_GATE>
+__global__ __launch_bounds__(256, 2) void gdn_chunk_kernel(
+    const T* __restrict__ q, const T* __restrict__ k, const T* __restrict__ v,
+    const float* __restrict__ gate, const float* __restrict__ beta,
+    T* __restrict__ out, const int32_t* __restrict__ cu_seqlens,
+    const float* __restrict__ init_state, float* __restrict__ final_state,
+    float scale, int H) {
+  constexpr int NT = 256;
+  constexpr int GPC = NT / D;       // thread groups per column
+  constexpr int RPT = D / GPC;      // state rows per thread
+  constexpr int CH = 16;            // tokens staged per LDS chunk
+  __shared__ float kq_s[CH][2 * D];                       // k then q
+  __shared__ float v_s[CH][D];
+  __shared__ float a_s[CH][PER_CHANNEL_GATE ? D : 1];
+  __shared__ float b_s[CH];
+  __shared__ float part[2][GPC][D];                       // kS / o partials
+  __shared__ float delta_s[D];
+
+  const int seq = blockIdx.x / H, h = blockIdx.x % H;
+  const int64_t s0 = cu_seqlens[seq], s1 = cu_seqlens[seq + 1];
+  const int len = (int)(s1 - s0);
+  const int tid = threadIdx.x;
+  const int col = tid % D, grp = tid / D;
+  const int r0 = grp * RPT;
+
+  float S[RPT];
+  const int64_t state_off = ((int64_t)seq * H + h) * D * D;
+#pragma unroll
+  for (int i = 0; i < RPT; ++i)
+    S[i] = init_state ? init_state[state_off + (int64_t)(r0 + i) * D + col] : 0.f;
+  if (len == 0) {  // zero-length sequence: state passes through untouched
+    if (final_state)
+#pragma unroll
+      for (int i = 0; i < RPT; ++i)
+        final_state[state_off + (int64_t)(r0 + i) * D + col] = S[i];
+    return;
+  }
+
+  for (int base = 0; base < len; base += CH) {
+    const int nt = min(CH, len - base);
+    // stage the chunk: k/q/v rows for this head, gate/beta scalars
+    for (int e = tid; e < nt * D; e += NT) {
+      int t = e / D, d = e % D;
+      int64_t row = (s0 + base + t) * (int64_t)H + h;
+      kq_s[t][d] = to_f32<T>(k[row * D + d]);
+      kq_s[t][D + d] = scale * to_f32<T>(q[row * D + d]);
+      v_s[t][d] = to_f32<T>(v[row * D + d]);
+      if constexpr (PER_CHANNEL_GATE) a_s[t][d] = gate[row * D + d];
+    }
+    for (int t = tid; t < nt; t += NT) {
+      int64_t row = (s0 + base + t) * (int64_t)H + h;
+      if constexpr (!PER_CHANNEL_GATE) a_s[t][0] = gate[row];
+      b_s[t] = beta[row];
+    }
+    __syncthreads();
+
+    for (int t = 0; t < nt; ++t) {
+      // decay + k^T S partial over this thread's rows
+      float acc = 0.f;
+#pragma unroll
+      for (int i = 0; i < RPT; ++i) {
+        float a = PER_CHANNEL_GATE ? a_s[t][r0 + i] : a_s[t][0];
+        S[i] *= a;
+        acc = __builtin_fmaf(kq_s[t][r0 + i], S[i], acc);
+      }
+      part[0][grp][col] = acc;
+      __syncthreads();
+      if (tid < D) {
+        float ks = 0.f;
+#pragma unroll
+        for (int g2 = 0; g2 < GPC; ++g2) ks += part[0][g2][tid];
+        delta_s[tid] = b_s[t] * (v_s[t][tid] - ks);
+      }
+      __syncthreads();
+      // rank-1 update + q^T S partial
+      const float dlt = delta_s[col];
+      float oacc = 0.f;
+#pragma unroll
+      for (int i = 0; i < RPT; ++i) {
+        S[i] = __builtin_fmaf(kq_s[t][r0 + i], dlt, S[i]);
+        oacc = __builtin_fmaf(kq_s[t][D + r0 + i], S[i], oacc);
+      }
+      part[1][grp][col] = oacc;
+      __syncthreads();
+      if (tid < D) {
+        float o = 0.f;
+#pragma unroll
+        for (int g2 = 0; g2 < GPC; ++g2) o += part[1][g2][tid];
+        out[((s0 + base + t) * (int64_t)H + h) * D + tid] = from_f32<T>(o);
+      }
+      __syncthreads();
+    }
+  }
+  if (final_state)
+#pragma unroll
+    for (int i = 0; i < RPT; ++i)
+      final_state[state_off + (int64_t)(r0 + i) * D + col] = S[i];
+}
+
 }  // namespace fi
+
+extern "C" hipError_t fi_gdn_chunk(int dtype, int per_channel_gate, const void* q,
+                                   const void* k, const void* v, const float* gate,
+                                   const float* beta, void* out,
+                                   const int32_t* cu_seqlens, const float* init_state,
+                                   float* final_state, float scale, int num_seqs,
+                                   int H, int D, hipStream_t stream) {
+  dim3 grid((uint32_t)num_seqs * H), blk(256);
+#define LGC(T, D, PC)                                                            \
+  hipLaunchKernelGGL((fi::gdn_chunk_kernel<T, D, PC>), grid, blk, 0, stream,     \
+                     (const T*)q, (const T*)k, (const T*)v, gate, beta, (T*)out, \
+                     cu_seqlens, init_state, final_state, scale, H)
+#define LGC2(T)                                                       \
+  do {                                                                \
+    if (D == 128 && per_channel_gate) LGC(T, 128, true);              \
+    else if (D == 128) LGC(T, 128, false);                            \
+    else if (D == 64 && per_channel_gate) LGC(T, 64, true);           \
+    else if (D == 64) LGC(T, 64, false);                              \
+    else return hipErrorInvalidValue;                                 \
+  } while (0)
+  switch (dtype) {
+    case 0: LGC2(fi::bf16); break;
+    case 1: LGC2(fi::fp16); break;
+    case 2: LGC2(float); break;
+    default: return hipErrorInvalidValue;
+  }
+#undef LGC2
+#undef LGC
+  return hipGetLastError();
+}
 
 extern "C" hipError_t fi_gdn_decode(int dtype, int state_f32, int per_channel_gate,
                                     void* state, const void* q, const void* k,

@@ -34,3 +34,52 @@ def fused_kda_decode(state, q, k, v, g_channel, beta, out=None):
     fused_kda_decode role): delta rule with per-channel gate [B, H, Dk]."""
     assert g_channel.dim() == 3
     return gdn_fused_decode_step(state, q, k, v, g_channel, beta, out)
+
+
+def chunk_gated_delta_rule(
+    q: torch.Tensor,          # [total_tokens, H, D]
+    k: torch.Tensor,          # [total_tokens, H, D]
+    v: torch.Tensor,          # [total_tokens, H, D]
+    gate: torch.Tensor,       # [total_tokens, H] decay in (0,1], or [.., H, D] (KDA)
+    beta: torch.Tensor,       # [total_tokens, H]
+    cu_seqlens: torch.Tensor,  # [num_seqs + 1] int32
+    scale: Optional[float] = None,
+    initial_state: Optional[torch.Tensor] = None,   # [num_seqs, H, D, D] f32
+    output_final_state: bool = False,
+    output: Optional[torch.Tensor] = None,
+):
+    r"""Gated delta-rule prefill over ragged sequences (reference parity:
+    flashinfer/gdn_kernels/blackwell/gdn_prefill.py
+    chunk_gated_delta_rule_sm100:137; math validated against the reference's
+    tests/gdn/reference_delta_rule.py blockwise_delta_rule:856). Per token:
+    ``S = a*S; u = beta*(v - k^T S); S += k (x) u; o = scale * q^T S``.
+    State layout is [num_seqs, H, D_k, D_v] (same as ``gdn_fused_decode_step``;
+    the reference's output_state is the [.., D_v, D_k] transpose).
+    Returns ``out`` or ``(out, final_state)``. A 3-D gate selects the
+    KDA-style per-channel decay."""
+    total, H, D = q.shape
+    if scale is None:
+        scale = D ** -0.5
+    if output is None:
+        output = torch.empty_like(v)
+    num_seqs = cu_seqlens.numel() - 1
+    final_state = None
+    if output_final_state:
+        final_state = torch.empty(num_seqs, H, D, D, dtype=torch.float32,
+                                  device=q.device)
+    get_ext().gdn_chunk(
+        q.contiguous(), k.contiguous(), v.contiguous(),
+        gate.float().contiguous(), beta.float().contiguous(), output,
+        cu_seqlens.to(q.device, torch.int32),
+        initial_state.float().contiguous() if initial_state is not None else None,
+        final_state, float(scale),
+    )
+    return (output, final_state) if output_final_state else output
+
+
+def chunk_kda(q, k, v, g_channel, beta, cu_seqlens, scale=None,
+              initial_state=None, output_final_state=False, output=None):
+    r"""KDA chunked prefill: delta rule with per-channel gate [T, H, D]."""
+    assert g_channel.dim() == 3
+    return chunk_gated_delta_rule(q, k, v, g_channel, beta, cu_seqlens, scale,
+                                  initial_state, output_final_state, output)

@@ -101,3 +101,41 @@ def test_block_sparse_matches_dense():
     p = torch.softmax(logits, -1)
     ref = torch.einsum("hml,lhd->mhd", p, v.float())
     torch.testing.assert_close(out.float(), ref, atol=3e-2, rtol=3e-2)
+
+
+def test_variable_block_sparse_matches_dense():
+    from flashinfer_amd.sparse import VariableBlockSparseAttentionWrapper
+
+    torch.manual_seed(1)
+    Hkv, G, D = 2, 2, 128
+    Hq = Hkv * G
+    MB, NB = 3, 4
+    # per-head variable block sizes summing to the same seq lens
+    row_sz = torch.tensor([[32, 64, 32], [16, 96, 16]], dtype=torch.int32)
+    col_sz = torch.tensor([[64, 32, 16, 16], [32, 32, 32, 32]], dtype=torch.int32)
+    qo_len, kv_len = 128, 128
+    assert (row_sz.sum(1) == qo_len).all() and (col_sz.sum(1) == kv_len).all()
+    mask_map = torch.rand(Hkv, MB, NB) > 0.4
+    mask_map[:, :, 0] = True  # every row block attends something
+
+    q = torch.randn(Hq, qo_len, D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(Hkv, kv_len, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(Hkv, kv_len, D, dtype=torch.bfloat16, device="cuda")
+
+    ws = torch.empty(64 * 1024 * 1024, dtype=torch.uint8, device="cuda")
+    w = VariableBlockSparseAttentionWrapper(ws)
+    w.plan(mask_map, row_sz, col_sz, Hq, Hkv, D)
+    out, lse = w.run(q, k, v, return_lse=True)
+    assert out.shape == (Hq, qo_len, D) and lse.shape == (Hq, qo_len)
+
+    # dense fp32 reference: expand the per-head block mask to token level
+    for h in range(Hkv):
+        tok_mask = mask_map[h].repeat_interleave(
+            row_sz[h].long(), 0).repeat_interleave(col_sz[h].long(), 1).cuda()
+        for gi in range(G):
+            qh = q[h * G + gi].float()
+            logits = qh @ k[h].float().t() / math.sqrt(D)
+            logits = logits.masked_fill(~tok_mask, float("-inf"))
+            ref = torch.softmax(logits, -1) @ v[h].float()
+            torch.testing.assert_close(out[h * G + gi].float(), ref,
+                                       atol=3e-2, rtol=3e-2)
