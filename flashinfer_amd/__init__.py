@@ -23,7 +23,12 @@ from .sparse import (
 )
 from .mla import BatchMLAPagedAttentionWrapper
 from .attention import BatchAttention, PODWithPagedKVCacheWrapper
-from .mamba import selective_state_update, ssu_checkpoint, ssu_rollback
+from .mamba import (
+    mamba_chunk_scan_combined,
+    selective_state_update,
+    ssu_checkpoint,
+    ssu_rollback,
+)
 from .green_ctx import split_device_cu_streams, split_device_green_ctx
 from .gdn import (
     chunk_gated_delta_rule,

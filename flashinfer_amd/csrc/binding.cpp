@@ -88,6 +88,12 @@ hipError_t fi_gdn_decode(int dtype, int state_f32, int per_channel_gate, void* s
                          const void* q, const void* k, const void* v, const float* g,
                          const float* beta, void* out, int B, int H, int Dk, int Dv,
                          hipStream_t stream);
+hipError_t fi_ssd_scan(int dtype, const void* x, const float* dt, const float* A,
+                       const void* Bm, const void* Cm, const float* D, const void* z,
+                       const float* dt_bias, const float* init_states,
+                       float* final_states, void* out, int batch, int L, int H, int G,
+                       int P, int dstate, int dt_softplus, float dt_min, float dt_max,
+                       int d_has_hdim, hipStream_t stream);
 hipError_t fi_gdn_chunk(int dtype, int per_channel_gate, const void* q, const void* k,
                         const void* v, const float* gate, const float* beta, void* out,
                         const int32_t* cu_seqlens, const float* init_state,
@@ -664,6 +670,29 @@ void gdn_decode(at::Tensor state, at::Tensor q, at::Tensor k, at::Tensor v,
             "fi_gdn_decode");
 }
 
+void ssd_scan(at::Tensor x, at::Tensor dt, at::Tensor A, at::Tensor B, at::Tensor C,
+              std::optional<at::Tensor> D, std::optional<at::Tensor> z,
+              std::optional<at::Tensor> dt_bias,
+              std::optional<at::Tensor> init_states,
+              std::optional<at::Tensor> final_states, at::Tensor out,
+              bool dt_softplus, double dt_min, double dt_max) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.is_contiguous());
+  TORCH_CHECK(A.scalar_type() == at::kFloat && dt.scalar_type() == at::kFloat);
+  int batch = x.size(0), L = x.size(1), H = x.size(2), P = x.size(3);
+  int G = B.size(2), dstate = B.size(3);
+  bool d_has_hdim = D && D->dim() == 2;
+  check_hip(
+      fi_ssd_scan(dtype_code(x), x.data_ptr(), dt.data_ptr<float>(),
+                  A.data_ptr<float>(), B.data_ptr(), C.data_ptr(),
+                  D ? D->data_ptr<float>() : nullptr, z ? z->data_ptr() : nullptr,
+                  dt_bias ? dt_bias->data_ptr<float>() : nullptr,
+                  init_states ? init_states->data_ptr<float>() : nullptr,
+                  final_states ? final_states->data_ptr<float>() : nullptr,
+                  out.data_ptr(), batch, L, H, G, P, dstate, dt_softplus,
+                  (float)dt_min, (float)dt_max, d_has_hdim, cur_stream(x)),
+      "fi_ssd_scan");
+}
+
 void gdn_chunk(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor gate,
                at::Tensor beta, at::Tensor out, at::Tensor cu_seqlens,
                std::optional<at::Tensor> init_state,
@@ -722,5 +751,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_finalize", &moe_finalize);
   m.def("gdn_decode", &gdn_decode);
   m.def("gdn_chunk", &gdn_chunk);
+  m.def("ssd_scan", &ssd_scan);
   m.def("debug_fastdiv", &debug_fastdiv);
 }
