@@ -24,6 +24,7 @@ from .sparse import (
 from .mla import BatchMLAPagedAttentionWrapper
 from .attention import BatchAttention, PODWithPagedKVCacheWrapper
 from .mamba import (
+    cake_selective_state_update,
     mamba_chunk_scan_combined,
     selective_state_update,
     ssu_checkpoint,

@@ -88,3 +88,11 @@ def mamba_chunk_scan_combined(
         final_states, out, dt_softplus, float(dt_limit[0]), float(dt_limit[1]),
     )
     return (out, final_states) if return_final_states else out
+
+
+def cake_selective_state_update(*args, **kwargs):
+    r"""Alias of :func:`selective_state_update` (reference flashinfer/mamba/
+    cake_selective_state_update.py:12 routes to the same op with
+    backend="cake"; here there is a single CDNA4 backend)."""
+    kwargs.pop("backend", None)
+    return selective_state_update(*args, **kwargs)
