@@ -57,10 +57,13 @@ from .decode import (
 )
 from .norm import (
     fused_add_rmsnorm,
+    fused_add_rmsnorm_quant,
+    fused_rmsnorm_silu,
     gemma_fused_add_rmsnorm,
     gemma_rmsnorm,
     layernorm,
     rmsnorm,
+    rmsnorm_quant,
 )
 from .prefill import (
     BatchPrefillWithPagedKVCacheWrapper,

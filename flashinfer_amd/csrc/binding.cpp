@@ -10,6 +10,9 @@
 namespace fi_ext = fi;
 
 extern "C" {
+hipError_t fi_norm_quant(int add, int dtype, void* x, void* residual, const void* w,
+                         void* out, const float* scale, int rows, int d, float eps,
+                         hipStream_t stream);
 hipError_t fi_norm(int which, int dtype, const void* x, const void* w, const void* b,
                    void* y, void* residual, int rows, int d, int64_t sx, int64_t sy,
                    float eps, int weight_bias, hipStream_t stream);
@@ -153,6 +156,22 @@ void fused_add_rmsnorm(at::Tensor x, at::Tensor residual, at::Tensor w, double e
 void layernorm(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> b, at::Tensor out,
                double eps) {
   norm_common(2, x, w, b, out, c10::nullopt, eps, false);
+}
+
+void rmsnorm_silu(at::Tensor x, at::Tensor w, at::Tensor out, double eps) {
+  norm_common(3, x, w, c10::nullopt, out, c10::nullopt, eps, false);
+}
+
+void rmsnorm_quant(at::Tensor x, c10::optional<at::Tensor> residual, at::Tensor w,
+                   at::Tensor out, at::Tensor scale, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  TORCH_CHECK(out.scalar_type() == at::kFloat8_e4m3fn && out.is_contiguous());
+  TORCH_CHECK(scale.scalar_type() == at::kFloat && scale.is_cuda());
+  check_hip(fi_norm_quant(residual.has_value(), dtype_code(x), x.data_ptr(),
+                          residual ? residual->data_ptr() : nullptr, w.data_ptr(),
+                          out.data_ptr(), scale.data_ptr<float>(), x.size(0),
+                          x.size(1), (float)eps, cur_stream(x)),
+            "fi_norm_quant");
 }
 
 // ---------------- activation ----------------
@@ -723,6 +742,8 @@ std::vector<int64_t> debug_fastdiv(int64_t d, std::vector<int64_t> ns) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm);
+  m.def("rmsnorm_silu", &rmsnorm_silu);
+  m.def("rmsnorm_quant", &rmsnorm_quant);
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
   m.def("layernorm", &layernorm);
   m.def("act_and_mul", &act_and_mul);

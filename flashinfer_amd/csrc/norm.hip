@@ -30,8 +30,9 @@ __device__ __forceinline__ float block_reduce_sum(float x, float* smem) {
 }
 
 // ---------------- RMSNorm ----------------
-// y = x / rms(x) * (w + wbias)
-template <typename T, int VEC, bool kWeightBias>
+// y = x / rms(x) * (w + wbias)    (optionally y = silu(...) — reference
+// flashinfer/norm/__init__.py fused_rmsnorm_silu:689)
+template <typename T, int VEC, bool kWeightBias, bool kSilu = false>
 __global__ void rmsnorm_kernel(const T* __restrict__ x, const T* __restrict__ w,
                                T* __restrict__ y, int rows, int d, int64_t stride_x,
                                int64_t stride_y, float eps) {
@@ -58,9 +59,65 @@ __global__ void rmsnorm_kernel(const T* __restrict__ x, const T* __restrict__ w,
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
         float wj = kWeightBias ? wv.get(j) + 1.f : wv.get(j);
-        out.set(j, v.get(j) * rrms * wj);
+        float r = v.get(j) * rrms * wj;
+        if constexpr (kSilu) r *= 1.f / (1.f + __builtin_expf(-r));
+        out.set(j, r);
       }
       out.store(yr + i);
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------- RMSNorm + fp8 quantize ----------------
+// Reference flashinfer/norm/__init__.py rmsnorm_quant:214 /
+// fused_add_rmsnorm_quant:323:  out = (rmsnorm(x or x+residual) * w) / scale
+// cast to fp8 e4m3; with kAdd the residual is updated in place first.
+template <typename T, int VEC, bool kAdd>
+__global__ void rmsnorm_quant_kernel(T* __restrict__ x, T* __restrict__ residual,
+                                     const T* __restrict__ w,
+                                     fp8_e4m3* __restrict__ out,
+                                     const float* __restrict__ scale, int rows, int d,
+                                     float eps) {
+  __shared__ float smem[kNormWaves + 1];
+  const float inv_scale = 1.f / *scale;
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    T* xr = x + (int64_t)row * d;
+    T* rr = kAdd ? residual + (int64_t)row * d : nullptr;
+    fp8_e4m3* orow = out + (int64_t)row * d;
+    float ss = 0.f;
+    for (int i = threadIdx.x * VEC; i < d; i += kNormThreads * VEC) {
+      vec_t<T, VEC> vx;
+      vx.load(xr + i);
+      if constexpr (kAdd) {
+        vec_t<T, VEC> vr;
+        vr.load(rr + i);
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+          float f = vx.get(j) + vr.get(j);
+          vr.set(j, f);
+          ss += f * f;
+        }
+        vr.store(rr + i);
+      } else {
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+          float f = vx.get(j);
+          ss += f * f;
+        }
+      }
+    }
+    ss = block_reduce_sum<kNormWaves>(ss, smem);
+    float rrms = rsqrtf(ss / d + eps);
+    for (int i = threadIdx.x * VEC; i < d; i += kNormThreads * VEC) {
+      vec_t<T, VEC> v, wv;
+      v.load(kAdd ? rr + i : xr + i);
+      wv.load(w + i);
+      vec_t<fp8_e4m3, VEC> q;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        q.set(j, v.get(j) * rrms * wv.get(j) * inv_scale);
+      q.store(orow + i);
     }
     __syncthreads();
   }
@@ -185,14 +242,50 @@ hipError_t norm_launch(int which, const void* x, const void* w, const void* b, v
       if (vec8) hipLaunchKernelGGL((layernorm_kernel<T, 8>), g, blk, 0, stream, X, W, B, Y, rows, d, sx, sy, eps);
       else hipLaunchKernelGGL((layernorm_kernel<T, 1>), g, blk, 0, stream, X, W, B, Y, rows, d, sx, sy, eps);
       break;
+    case 3:  // fused rmsnorm + silu
+      if (vec8) hipLaunchKernelGGL((rmsnorm_kernel<T, 8, false, true>), g, blk, 0, stream, X, W, Y, rows, d, sx, sy, eps);
+      else hipLaunchKernelGGL((rmsnorm_kernel<T, 1, false, true>), g, blk, 0, stream, X, W, Y, rows, d, sx, sy, eps);
+      break;
   }
 #undef LAUNCH_N
   return hipGetLastError();
 }
 
+template <typename T>
+hipError_t norm_quant_launch(bool add, void* x, void* residual, const void* w, void* out,
+                             const float* scale, int rows, int d, float eps,
+                             hipStream_t stream) {
+  int grid = rows < 2048 ? rows : 2048;
+  dim3 g(grid), blk(kNormThreads);
+  bool vec8 = (d % 8 == 0);
+  auto X = (T*)x;
+  auto R = (T*)residual;
+  auto W = (const T*)w;
+  auto O = (fp8_e4m3*)out;
+  if (add) {
+    if (vec8) hipLaunchKernelGGL((rmsnorm_quant_kernel<T, 8, true>), g, blk, 0, stream, X, R, W, O, scale, rows, d, eps);
+    else hipLaunchKernelGGL((rmsnorm_quant_kernel<T, 1, true>), g, blk, 0, stream, X, R, W, O, scale, rows, d, eps);
+  } else {
+    if (vec8) hipLaunchKernelGGL((rmsnorm_quant_kernel<T, 8, false>), g, blk, 0, stream, X, R, W, O, scale, rows, d, eps);
+    else hipLaunchKernelGGL((rmsnorm_quant_kernel<T, 1, false>), g, blk, 0, stream, X, R, W, O, scale, rows, d, eps);
+  }
+  return hipGetLastError();
+}
+
 }  // namespace fi
 
-// which: 0 rmsnorm, 1 fused_add_rmsnorm, 2 layernorm
+extern "C" hipError_t fi_norm_quant(int add, int dtype, void* x, void* residual,
+                                    const void* w, void* out, const float* scale,
+                                    int rows, int d, float eps, hipStream_t stream) {
+  switch (dtype) {
+    case 0: return fi::norm_quant_launch<fi::bf16>(add, x, residual, w, out, scale, rows, d, eps, stream);
+    case 1: return fi::norm_quant_launch<fi::fp16>(add, x, residual, w, out, scale, rows, d, eps, stream);
+    case 2: return fi::norm_quant_launch<float>(add, x, residual, w, out, scale, rows, d, eps, stream);
+  }
+  return hipErrorInvalidValue;
+}
+
+// which: 0 rmsnorm, 1 fused_add_rmsnorm, 2 layernorm, 3 rmsnorm+silu
 // dtype: 0 bf16, 1 fp16, 2 fp32
 extern "C" hipError_t fi_norm(int which, int dtype, const void* x, const void* w,
                               const void* b, void* y, void* residual, int rows, int d,

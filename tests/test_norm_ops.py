@@ -87,3 +87,49 @@ def test_act_and_mul(act, shape):
         "gelu_tanh": torch.nn.functional.gelu(g, approximate="tanh"),
     }[act] * u
     torch.testing.assert_close(out.float(), ref, atol=2e-2, rtol=2e-2)
+
+
+def test_rmsnorm_quant():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    x = torch.randn(64, 4096, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(4096, dtype=torch.bfloat16, device="cuda")
+    scale = torch.tensor([0.5], dtype=torch.float32, device="cuda")
+    out = torch.empty(64, 4096, dtype=torch.float8_e4m3fn, device="cuda")
+    fi.rmsnorm_quant(out, x, w, scale)
+    xf = x.float()
+    ref = xf / torch.sqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-6) * w.float()
+    ref = (ref / scale).to(torch.float8_e4m3fn).float()
+    torch.testing.assert_close(out.float(), ref, atol=0.2, rtol=0.2)
+
+
+def test_fused_add_rmsnorm_quant():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(1)
+    x = torch.randn(32, 2048, dtype=torch.bfloat16, device="cuda")
+    r = torch.randn(32, 2048, dtype=torch.bfloat16, device="cuda")
+    r0 = r.clone()
+    w = torch.randn(2048, dtype=torch.bfloat16, device="cuda")
+    out = torch.empty(32, 2048, dtype=torch.float8_e4m3fn, device="cuda")
+    fi.fused_add_rmsnorm_quant(out, x, r, w, 2.0)
+    rf = (x.float() + r0.float())
+    torch.testing.assert_close(r.float(), rf, atol=2e-2, rtol=2e-2)
+    rf = r.float()  # quantize path uses the bf16-rounded residual
+    ref = rf / torch.sqrt(rf.pow(2).mean(-1, keepdim=True) + 1e-6) * w.float()
+    ref = (ref / 2.0).to(torch.float8_e4m3fn).float()
+    torch.testing.assert_close(out.float(), ref, atol=0.2, rtol=0.2)
+
+
+def test_fused_rmsnorm_silu():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(2)
+    x = torch.randn(16, 1024, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(1024, dtype=torch.bfloat16, device="cuda")
+    out = fi.fused_rmsnorm_silu(x, w)
+    xf = x.float()
+    ref = xf / torch.sqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-6) * w.float()
+    ref = ref * torch.sigmoid(ref)
+    torch.testing.assert_close(out.float(), ref, atol=2e-2, rtol=2e-2)
