@@ -133,3 +133,5 @@ __all__ = [  # noqa: F405
     "apply_rope_with_cos_sin_cache",
     "apply_rope_with_cos_sin_cache_inplace",
 ]
+
+from . import profiler

@@ -23,6 +23,7 @@
 #include "fi/fastdiv.hpp"
 #include "fi/frag.hpp"
 #include "fi/mfma.hpp"
+#include "fi/profiler.hpp"
 #include "fi/vec.hpp"
 
 namespace fi {
@@ -225,10 +226,12 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
     }
   };
 
+  prof_event(p.prof_buf, 0, ProfType::kBegin);  // event 0: whole tile
   stage_load(kv_lo);
   stage_write();
   __syncthreads();
 
+  prof_event(p.prof_buf, 1, ProfType::kBegin);  // event 1: kv mainloop
   for (int64_t kv0 = kv_lo; kv0 < kv_hi; kv0 += KVB) {
     const bool have_next = kv0 + KVB < kv_hi;
     if (have_next) stage_load(kv0 + KVB);
@@ -389,6 +392,7 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
     if (have_next) stage_write();
     __syncthreads();          // next tile staged
   }
+  prof_event(p.prof_buf, 1, ProfType::kEnd);
 
   // ---- epilogue: normalize and write O (transpose from O^T frags) ----
   float d_full = d_run + __shfl_xor(d_run, 32, 64);
@@ -410,6 +414,7 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
           l2;
     }
   }
+  prof_event(p.prof_buf, 0, ProfType::kEnd);
 }
 
 template <typename T, typename TKV>

@@ -343,7 +343,8 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                        bool paged, int64_t cta_q,
                        c10::optional<at::Tensor> mask_data,
                        c10::optional<at::Tensor> mask_byte_indptr, bool alibi,
-                       double k_descale, double v_descale) {
+                       double k_descale, double v_descale,
+                       c10::optional<at::Tensor> prof_buf = c10::nullopt) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 3, "q must be [nnz, Hq, D]");
   TORCH_CHECK(q.stride(2) == 1 && out.stride(2) == 1);
   fi_ext::PrefillParams p{};
@@ -397,6 +398,10 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   p.alibi = alibi ? 1 : 0;
   p.k_descale = (float)k_descale;
   p.v_descale = (float)v_descale;
+  if (prof_buf.has_value()) {
+    TORCH_CHECK(prof_buf->scalar_type() == at::kUInt64 && prof_buf->is_cuda());
+    p.prof_buf = (unsigned long long*)prof_buf->data_ptr();
+  }
   check_hip(fi_batch_prefill(dtype_code(q), dtype_code(k_cache), &p, paged ? 1 : 0,
                              cur_stream(q)),
             "fi_batch_prefill");

@@ -75,6 +75,7 @@ struct PrefillParams {
   const int32_t* mask_byte_indptr; // per-request byte offset into mask_data
   int alibi;                       // ALiBi position bias (slope by qo head)
   float k_descale, v_descale;      // fp8-KV dequant factors (staging)
+  unsigned long long* prof_buf;    // intra-kernel event buffer (optional)
 };
 
 struct SamplingParams {

@@ -103,7 +103,8 @@ class _BatchPrefillBase:
         )
 
     def _run_common(self, q, k_cache, v_cache, kv_indices, kv_indptr, kv_last_page_len,
-                    paged, out, lse, return_lse, k_scale=None, v_scale=None):
+                    paged, out, lse, return_lse, k_scale=None, v_scale=None,
+                    profiler_buffer=None):
         pi = self._plan_info
         if pi is None:
             raise RuntimeError("must call plan() before run()")
@@ -128,7 +129,7 @@ class _BatchPrefillBase:
             pi["logits_soft_cap"], pi["window_left"], pi["causal"], paged,
             pi["cta_q"], getattr(self, "_mask_data", None),
             getattr(self, "_mask_byte_indptr", None), pi.get("alibi", False),
-            k_descale, v_descale,
+            k_descale, v_descale, profiler_buffer,
         )
         if not kv_fp8 and v_scale is not None:
             out = out * v_scale
@@ -174,11 +175,12 @@ class BatchPrefillWithPagedKVCacheWrapper(_BatchPrefillBase):
     begin_forward = plan
 
     def run(self, q, paged_kv_cache, *args, k_scale=None, v_scale=None, out=None,
-            lse=None, return_lse: bool = False, **kwargs):
+            lse=None, return_lse: bool = False, profiler_buffer=None, **kwargs):
         k_cache, v_cache = unpack_paged_kv_cache(paged_kv_cache, self._kv_layout)
         return self._run_common(q, k_cache, v_cache, self._kv_indices_d,
                                 self._kv_indptr_d, self._kv_last_page_len_d, True,
-                                out, lse, return_lse, k_scale, v_scale)
+                                out, lse, return_lse, k_scale, v_scale,
+                                profiler_buffer)
 
     forward = run
 
@@ -210,9 +212,10 @@ class BatchPrefillWithRaggedKVCacheWrapper(_BatchPrefillBase):
     begin_forward = plan
 
     def run(self, q, k, v, *args, k_scale=None, v_scale=None, out=None, lse=None,
-            return_lse: bool = False, **kwargs):
+            return_lse: bool = False, profiler_buffer=None, **kwargs):
         return self._run_common(q, k, v, None, self._kv_indptr_d, None, False,
-                                out, lse, return_lse, k_scale, v_scale)
+                                out, lse, return_lse, k_scale, v_scale,
+                                profiler_buffer)
 
     forward = run
 
@@ -261,6 +264,6 @@ def single_prefill_with_kv_cache(
         meta[n_tiles : 2 * n_tiles], out, lse,
         sm_scale if sm_scale is not None else default_sm_scale(D),
         float(logits_soft_cap or 0.0), window_left, causal, False, cta_q,
-        mask_data, mask_indptr, pos_encoding_mode == "ALIBI", 1.0, 1.0,
+        mask_data, mask_indptr, pos_encoding_mode == "ALIBI", 1.0, 1.0, None,
     )
     return (out, lse) if return_lse else out

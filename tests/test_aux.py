@@ -4,6 +4,9 @@ import os
 import subprocess
 import sys
 
+import pytest
+import torch
+
 
 def test_cli_show_config_and_module_status():
     env = dict(os.environ)
@@ -62,3 +65,39 @@ def test_autotuner_cache_selection():
     )
     out = r.run([1, 2, 3])  # not tuning: tactic 0
     assert out == [1, 2, 3] and calls == [0]
+
+
+@pytest.mark.gpu
+def test_intra_kernel_profiler():
+    """Prefill with a profiler buffer emits decodable begin/end event pairs."""
+    import flashinfer_amd as fi
+    from flashinfer_amd import profiler
+
+    torch.manual_seed(0)
+    q = torch.randn(256, 8, 128, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(512, 2, 128, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(512, 2, 128, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(16 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchPrefillWithRaggedKVCacheWrapper(ws, "NHD")
+    qo = torch.tensor([0, 256], dtype=torch.int32, device="cuda")
+    kvi = torch.tensor([0, 512], dtype=torch.int32, device="cuda")
+    w.plan(qo, kvi, 8, 2, 128, causal=True)
+    buf = profiler.make_profiler_buffer(1 << 14)
+    out = w.run(q, k, v, profiler_buffer=buf)
+    assert out.isfinite().all()
+    events = profiler.decode_events(buf)
+    assert events, "no profiler events recorded"
+    # every block must emit tile begin (0) and end, kv_mainloop begin/end
+    by_block = {}
+    for blk, eidx, etype, ts in events:
+        by_block.setdefault(blk, []).append((eidx, etype))
+    for blk, evs in by_block.items():
+        assert (0, 0) in evs and (0, 1) in evs and (1, 0) in evs and (1, 1) in evs
+    # run without buffer still fine (null-pointer fast path)
+    out2 = w.run(q, k, v)
+    assert torch.equal(out, out2)
+    import tempfile, os, json
+    fn = os.path.join(tempfile.mkdtemp(), "trace.json")
+    profiler.export_to_chrome_trace(buf, ["tile", "kv_mainloop"], fn)
+    tr = json.load(open(fn))
+    assert len(tr["traceEvents"]) >= 4
