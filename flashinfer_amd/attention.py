@@ -62,9 +62,11 @@ class BatchAttention:
 
 
 class PODWithPagedKVCacheWrapper:
-    r"""Prefill-On-Decode (reference pod.py:61): one prefill request plus a
-    decode batch issued back-to-back on the same stream — on the 256-CU chip
-    the two grids overlap naturally."""
+    r"""Prefill-On-Decode (reference pod.py:61): one prefill request runs
+    CONCURRENTLY with a decode batch. The reference fuses both into one
+    persistent kernel; on the 256-CU / 8-XCD chip the same effect falls out
+    of two HIP streams — the MFMA-bound prefill grid and the HBM-bound
+    decode grid co-schedule across CUs, so decode hides under prefill."""
 
     def __init__(self, float_workspace_buffer: torch.Tensor, kv_layout: str = "NHD",
                  **kwargs):
@@ -74,6 +76,9 @@ class PODWithPagedKVCacheWrapper:
             float_workspace_buffer, kv_layout
         )
         self._kv_layout = kv_layout
+        self._aux_stream = torch.cuda.Stream(
+            device=float_workspace_buffer.device
+        ) if float_workspace_buffer.is_cuda else None
 
     def plan(self, indptr, indices, last_page_len, num_qo_heads, num_kv_heads,
              head_dim, page_size, **kwargs):
@@ -82,9 +87,14 @@ class PODWithPagedKVCacheWrapper:
 
     def run(self, q_p, k_p, v_p, q_d, paged_kv_cache, causal_p: bool = True, **kwargs):
         r"""q_p/k_p/v_p: the prefill request (contiguous KV); q_d: decode
-        queries [batch, H, D] over the paged cache."""
+        queries [batch, H, D] over the paged cache. The decode batch is
+        issued on a second stream so both grids occupy the chip at once."""
         from .prefill import single_prefill_with_kv_cache
 
+        main = torch.cuda.current_stream(q_p.device)
+        self._aux_stream.wait_stream(main)
+        with torch.cuda.stream(self._aux_stream):
+            o_d = self._decode.run(q_d, paged_kv_cache)
         o_p = single_prefill_with_kv_cache(q_p, k_p, v_p, causal=causal_p)
-        o_d = self._decode.run(q_d, paged_kv_cache)
+        main.wait_stream(self._aux_stream)
         return o_p, o_d

@@ -101,3 +101,37 @@ def test_intra_kernel_profiler():
     profiler.export_to_chrome_trace(buf, ["tile", "kv_mainloop"], fn)
     tr = json.load(open(fn))
     assert len(tr["traceEvents"]) >= 4
+
+
+@pytest.mark.gpu
+def test_pod_wrapper_concurrent_streams():
+    """POD outputs must match the same prefill/decode run separately."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    Hq, Hkv, D, page = 8, 2, 128, 16
+    q_p = torch.randn(256, Hq, D, dtype=torch.bfloat16, device="cuda")
+    k_p = torch.randn(256, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v_p = torch.randn(256, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    batch, pages_per = 8, 16
+    npages = batch * pages_per
+    kc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    indptr = torch.arange(0, npages + 1, pages_per, dtype=torch.int32, device="cuda")
+    indices = torch.arange(npages, dtype=torch.int32, device="cuda")
+    lpl = torch.full((batch,), page, dtype=torch.int32, device="cuda")
+    q_d = torch.randn(batch, Hq, D, dtype=torch.bfloat16, device="cuda")
+
+    ws = torch.empty(32 << 20, dtype=torch.uint8, device="cuda")
+    pod = fi.PODWithPagedKVCacheWrapper(ws, "NHD")
+    pod.plan(indptr, indices, lpl, Hq, Hkv, D, page, q_data_type=torch.bfloat16)
+    o_p, o_d = pod.run(q_p, k_p, v_p, q_d, (kc, vc))
+    torch.cuda.synchronize()
+
+    ref_p = fi.single_prefill_with_kv_cache(q_p, k_p, v_p, causal=True)
+    ws2 = torch.empty(32 << 20, dtype=torch.uint8, device="cuda")
+    dec = fi.BatchDecodeWithPagedKVCacheWrapper(ws2, "NHD")
+    dec.plan(indptr, indices, lpl, Hq, Hkv, D, page, q_data_type=torch.bfloat16)
+    ref_d = dec.run(q_d, (kc, vc))
+    torch.testing.assert_close(o_p.float(), ref_p.float())
+    torch.testing.assert_close(o_d.float(), ref_d.float())
