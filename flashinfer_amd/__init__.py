@@ -38,7 +38,12 @@ from .gdn import (
     gdn_fused_decode_step,
 )
 from .fused_moe import cutlass_fused_moe, dsv3_routing, fused_moe, moe_topk_softmax
-from .topk import top_k
+from .topk import (
+    TopKTieBreak,
+    top_k,
+    top_k_page_table_transform,
+    top_k_ragged_transform,
+)
 from .quantization import packbits, segment_packbits
 from .fp8_quantization import (
     bmm_fp8,

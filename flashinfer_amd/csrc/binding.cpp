@@ -74,8 +74,12 @@ hipError_t fi_per_group_quant_fp8(int dtype, int trans_scale, const void* x, voi
                                   float eps, hipStream_t stream);
 hipError_t fi_scale_quant_fp8(int dtype, const void* x, void* q, const float* inv_scale,
                               int64_t n, hipStream_t stream);
-hipError_t fi_topk(const float* x, float* out_v, int32_t* out_i, int rows, int d, int k,
-                   int64_t stride, hipStream_t stream);
+hipError_t fi_topk(const float* x, float* out_v, int32_t* out_i,
+                   const int32_t* lengths, const int32_t* row_starts,
+                   const int32_t* offsets, const int32_t* page_table,
+                   const int32_t* row_to_batch, const int32_t* page_table_row_starts,
+                   int pt_cols, int page_size, int mode, int rows, int d, int k,
+                   int64_t stride, int tie_break, hipStream_t stream);
 hipError_t fi_packbits(const uint8_t* x, uint8_t* y, int64_t n, hipStream_t stream);
 hipError_t fi_segment_packbits(const uint8_t* x, uint8_t* y, const int32_t* x_indptr,
                                const int32_t* y_indptr, int num_segments,
@@ -615,12 +619,25 @@ void scale_quant_fp8(at::Tensor x, at::Tensor q, at::Tensor inv_scale) {
             "fi_scale_quant_fp8");
 }
 
-void topk_op(at::Tensor x, at::Tensor out_v, at::Tensor out_i, int64_t k) {
+void topk_op(at::Tensor x, c10::optional<at::Tensor> out_v, at::Tensor out_i,
+             int64_t k, c10::optional<at::Tensor> lengths,
+             c10::optional<at::Tensor> row_starts, c10::optional<at::Tensor> offsets,
+             c10::optional<at::Tensor> page_table,
+             c10::optional<at::Tensor> row_to_batch,
+             c10::optional<at::Tensor> page_table_row_starts, int64_t page_size,
+             int64_t mode, int64_t tie_break) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.scalar_type() == at::kFloat);
-  check_hip(fi_topk(x.data_ptr<float>(), out_v.data_ptr<float>(),
-                    out_i.data_ptr<int32_t>(), x.size(0), x.size(1), (int)k,
-                    x.stride(0), cur_stream(x)),
+#define IPTR(t) (t.has_value() ? t->data_ptr<int32_t>() : nullptr)
+  check_hip(fi_topk(x.data_ptr<float>(),
+                    out_v.has_value() ? out_v->data_ptr<float>() : nullptr,
+                    out_i.data_ptr<int32_t>(), IPTR(lengths), IPTR(row_starts),
+                    IPTR(offsets), IPTR(page_table), IPTR(row_to_batch),
+                    IPTR(page_table_row_starts),
+                    page_table.has_value() ? (int)page_table->size(1) : 0,
+                    (int)page_size, (int)mode, x.size(0), x.size(1), (int)k,
+                    x.stride(0), (int)tie_break, cur_stream(x)),
             "fi_topk");
+#undef IPTR
 }
 
 void packbits_op(at::Tensor x, at::Tensor y) {

@@ -35,48 +35,120 @@ __device__ __forceinline__ float tk_u2f(uint32_t u) {
   return __uint_as_float((u & 0x80000000u) ? (u & 0x7FFFFFFFu) : ~u);
 }
 
-// one block per row; unordered top-k (values + indices)
+// Index transform applied to selected local indices (reference
+// flashinfer/topk.py top_k_page_table_transform:676 /
+// top_k_ragged_transform:873 — DSA sparse-attention index translation):
+//   mode 0: out = i
+//   mode 1: out = i + offsets[row]
+//   mode 2: out = page_table[batch, pt_start + i/page_size]*page_size + i%page_size
+struct TopkTransform {
+  const int32_t* offsets;               // mode 1
+  const int32_t* page_table;            // mode 2 [B, pt_cols]
+  const int32_t* row_to_batch;          // mode 2, optional
+  const int32_t* page_table_row_starts; // mode 2, optional
+  int pt_cols, page_size, mode;
+};
+
+__device__ __forceinline__ int32_t tk_transform(const TopkTransform& t, int row,
+                                                int i) {
+  if (t.mode == 1) return i + t.offsets[row];
+  if (t.mode == 2) {
+    int b = t.row_to_batch ? t.row_to_batch[row] : row;
+    int ps = t.page_table_row_starts ? t.page_table_row_starts[row] : 0;
+    int32_t page = t.page_table[(int64_t)b * t.pt_cols + ps + i / t.page_size];
+    return page * t.page_size + i % t.page_size;
+  }
+  return i;
+}
+
+// one block per row; unordered top-k (values + indices). lengths/row_starts
+// give a ragged window [row_starts[row], +lengths[row]) within the row;
+// positions beyond min(k, len) pad as (-inf, -1). tie_break 1/2 bounds which
+// boundary-value indices are selected (smaller / larger preferred) via a
+// second binary search over the index.
 __global__ void topk_kernel(const float* __restrict__ x, float* __restrict__ out_v,
-                            int32_t* __restrict__ out_i, int rows, int d, int k,
-                            int64_t stride) {
+                            int32_t* __restrict__ out_i, const int32_t* __restrict__ lengths,
+                            const int32_t* __restrict__ row_starts, TopkTransform tr,
+                            int rows, int d, int k, int64_t stride, int tie_break) {
   __shared__ float smem[TW + 1];
   __shared__ int s_cnt;
   for (int row = blockIdx.x; row < rows; row += gridDim.x) {
-    const float* xr = x + (int64_t)row * stride;
-    // threshold tau: largest value with |{v >= tau}| >= k
+    const int base = row_starts ? row_starts[row] : 0;
+    const int len = lengths ? min(lengths[row], d - base) : d;
+    const int keff = min(k, len);
+    const float* xr = x + (int64_t)row * stride + base;
+    // pad the tail first
+    for (int p = keff + threadIdx.x; p < k; p += TB) {
+      if (out_v) out_v[(int64_t)row * k + p] = -INFINITY;
+      out_i[(int64_t)row * k + p] = -1;
+    }
+    if (keff == 0) { __syncthreads(); continue; }
+    // threshold tau: largest value with |{v >= tau}| >= keff
     uint32_t lo = 0, hi = 0xFFFFFFFFu;
-    for (int it = 0; it < 24 && lo < hi; ++it) {
-      uint32_t mid = lo + ((hi - lo) >> 1);
+    for (int it = 0; it < 33 && lo < hi; ++it) {
+      uint32_t mid = lo + ((hi - lo + 1) >> 1);
       float tau = tk_u2f(mid);
       float cnt = 0.f;
-      for (int i = threadIdx.x; i < d; i += TB)
+      for (int i = threadIdx.x; i < len; i += TB)
         if (xr[i] >= tau) cnt += 1.f;
       float g = tk_block_sum(cnt, smem);
-      if (g >= (float)k) lo = mid;
+      if (g >= (float)keff) lo = mid;
       else hi = mid - 1;
     }
     float tau = tk_u2f(lo);
-    // compact: strictly-greater first, then fill ties
+    // count strictly greater -> how many ties to take
+    float cg = 0.f;
+    for (int i = threadIdx.x; i < len; i += TB)
+      if (xr[i] > tau) cg += 1.f;
+    int n_greater = (int)tk_block_sum(cg, smem);
+    int n_ties = keff - n_greater;
+    // tie index bound: smallest B with |{i < B : x[i]==tau}| >= n_ties
+    // (tie_break 1), or mirrored from the top (tie_break 2)
+    int tie_lo = 0, tie_hi = len;  // selected ties: [tie_lo, tie_hi)
+    if (tie_break == 1) {
+      int blo = 0, bhi = len;
+      while (blo < bhi) {
+        int mid = blo + (bhi - blo) / 2;
+        float c = 0.f;
+        for (int i = threadIdx.x; i < mid; i += TB)
+          if (xr[i] == tau) c += 1.f;
+        if ((int)tk_block_sum(c, smem) >= n_ties) bhi = mid;
+        else blo = mid + 1;
+      }
+      tie_hi = blo;
+    } else if (tie_break == 2) {
+      int blo = 0, bhi = len;
+      while (blo < bhi) {
+        int mid = blo + (bhi - blo + 1) / 2;
+        float c = 0.f;
+        for (int i = threadIdx.x + mid; i < len; i += TB)
+          if (xr[i] == tau) c += 1.f;
+        if ((int)tk_block_sum(c, smem) >= n_ties) blo = mid;
+        else bhi = mid - 1;
+      }
+      tie_lo = blo;
+    }
+    // compact: strictly-greater first, then the bounded ties
     if (threadIdx.x == 0) s_cnt = 0;
     __syncthreads();
-    for (int i = threadIdx.x; i < d; i += TB) {
+    for (int i = threadIdx.x; i < len; i += TB) {
       float v = xr[i];
       if (v > tau) {
         int pos = atomicAdd(&s_cnt, 1);
-        if (pos < k) {
-          out_v[(int64_t)row * k + pos] = v;
-          out_i[(int64_t)row * k + pos] = i;
+        if (pos < keff) {
+          if (out_v) out_v[(int64_t)row * k + pos] = v;
+          out_i[(int64_t)row * k + pos] = tk_transform(tr, row, i);
         }
       }
     }
     __syncthreads();
-    for (int i = threadIdx.x; i < d; i += TB) {
+    for (int i = threadIdx.x; i < len; i += TB) {
       float v = xr[i];
-      if (v == tau) {
+      if (v == tau && i >= tie_lo && i < tie_hi) {
         int pos = atomicAdd(&s_cnt, 1);
-        if (pos < k) {
-          out_v[(int64_t)row * k + pos] = v;
-          out_i[(int64_t)row * k + pos] = i;
+        if (pos < keff) {
+          if (out_v) out_v[(int64_t)row * k + pos] = v;
+          out_i[(int64_t)row * k + pos] = tk_transform(tr, row, i);
         }
       }
     }
@@ -123,11 +195,18 @@ __global__ void segment_packbits_kernel(const uint8_t* __restrict__ x,
 
 }  // namespace fi
 
-extern "C" hipError_t fi_topk(const float* x, float* out_v, int32_t* out_i, int rows,
-                              int d, int k, int64_t stride, hipStream_t stream) {
+extern "C" hipError_t fi_topk(const float* x, float* out_v, int32_t* out_i,
+                              const int32_t* lengths, const int32_t* row_starts,
+                              const int32_t* offsets, const int32_t* page_table,
+                              const int32_t* row_to_batch,
+                              const int32_t* page_table_row_starts, int pt_cols,
+                              int page_size, int mode, int rows, int d, int k,
+                              int64_t stride, int tie_break, hipStream_t stream) {
   int grid = rows < 1024 ? rows : 1024;
+  fi::TopkTransform tr{offsets, page_table, row_to_batch, page_table_row_starts,
+                       pt_cols, page_size, mode};
   hipLaunchKernelGGL(fi::topk_kernel, dim3(grid), dim3(fi::TB), 0, stream, x, out_v,
-                     out_i, rows, d, k, stride);
+                     out_i, lengths, row_starts, tr, rows, d, k, stride, tie_break);
   return hipGetLastError();
 }
 
