@@ -86,7 +86,7 @@ __global__ void topk_kernel(const float* __restrict__ x, float* __restrict__ out
     // threshold tau: largest value with |{v >= tau}| >= keff
     uint32_t lo = 0, hi = 0xFFFFFFFFu;
     for (int it = 0; it < 33 && lo < hi; ++it) {
-      uint32_t mid = lo + ((hi - lo + 1) >> 1);
+      uint32_t mid = hi - ((hi - lo) >> 1);  // ceil midpoint, overflow-safe
       float tau = tk_u2f(mid);
       float cnt = 0.f;
       for (int i = threadIdx.x; i < len; i += TB)
