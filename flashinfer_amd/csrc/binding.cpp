@@ -95,6 +95,18 @@ hipError_t fi_gdn_decode(int dtype, int state_f32, int per_channel_gate, void* s
                          const void* q, const void* k, const void* v, const float* g,
                          const float* beta, void* out, int B, int H, int Dk, int Dv,
                          hipStream_t stream);
+hipError_t fi_mhc_post(int dtype, const void* x, const void* residual,
+                       const float* post_mix, const float* comb_mix, void* out,
+                       int64_t tokens, int H, hipStream_t stream);
+hipError_t fi_mhc_pre(int dtype, int prenorm, const float* dot_mix,
+                      const float* sqrsum, const void* residual, const float* scale,
+                      const float* base, float* post_mix, float* comb_mix,
+                      void* layer_input, int64_t tokens, int H, int num_splits,
+                      float inv_k, float rms_eps, float pre_eps, float sink_eps,
+                      float post_mult, int sink_repeat, hipStream_t stream);
+hipError_t fi_concat_mla_k(int dtype, void* k, const void* k_nope, const void* k_rope,
+                           int64_t tokens, int Hk, int nope, int rope,
+                           hipStream_t stream);
 hipError_t fi_ssd_scan(int dtype, const void* x, const float* dt, const float* A,
                        const void* Bm, const void* Cm, const float* D, const void* z,
                        const float* dt_bias, const float* init_states,
@@ -711,6 +723,48 @@ void gdn_decode(at::Tensor state, at::Tensor q, at::Tensor k, at::Tensor v,
             "fi_gdn_decode");
 }
 
+void mhc_post(at::Tensor x, at::Tensor residual, at::Tensor post_mix,
+              at::Tensor comb_mix, at::Tensor out) {
+  TORCH_CHECK(x.is_cuda() && residual.dim() == 3 && residual.size(1) == 4);
+  int64_t tokens = residual.size(0);
+  int H = residual.size(2);
+  check_hip(fi_mhc_post(dtype_code(x), x.data_ptr(), residual.data_ptr(),
+                        post_mix.data_ptr<float>(), comb_mix.data_ptr<float>(),
+                        out.data_ptr(), tokens, H, cur_stream(x)),
+            "fi_mhc_post");
+}
+
+void mhc_pre(at::Tensor dot_mix, c10::optional<at::Tensor> sqrsum,
+             at::Tensor residual, at::Tensor scale, at::Tensor base,
+             at::Tensor post_mix, at::Tensor comb_mix, at::Tensor layer_input,
+             int64_t num_splits, double inv_k, double rms_eps, double pre_eps,
+             double sink_eps, double post_mult, int64_t sink_repeat) {
+  TORCH_CHECK(residual.is_cuda() && residual.dim() == 3 && residual.size(1) == 4);
+  int64_t tokens = residual.size(0);
+  int H = residual.size(2);
+  check_hip(fi_mhc_pre(dtype_code(residual), !sqrsum.has_value(),
+                       dot_mix.data_ptr<float>(),
+                       sqrsum ? sqrsum->data_ptr<float>() : nullptr,
+                       residual.data_ptr(), scale.data_ptr<float>(),
+                       base.data_ptr<float>(), post_mix.data_ptr<float>(),
+                       comb_mix.data_ptr<float>(), layer_input.data_ptr(), tokens, H,
+                       (int)num_splits, (float)inv_k, (float)rms_eps, (float)pre_eps,
+                       (float)sink_eps, (float)post_mult, (int)sink_repeat,
+                       cur_stream(residual)),
+            "fi_mhc_pre");
+}
+
+void concat_mla_k(at::Tensor k, at::Tensor k_nope, at::Tensor k_rope) {
+  TORCH_CHECK(k.is_cuda() && k.dim() == 3 && k_nope.dim() == 3 && k_rope.dim() == 3);
+  int64_t tokens = k.size(0);
+  int Hk = k.size(1), nope = k_nope.size(2), rope = k_rope.size(2);
+  TORCH_CHECK(k.size(2) == nope + rope && k_rope.size(1) == 1);
+  check_hip(fi_concat_mla_k(dtype_code(k), k.data_ptr(), k_nope.data_ptr(),
+                            k_rope.data_ptr(), tokens, Hk, nope, rope,
+                            cur_stream(k)),
+            "fi_concat_mla_k");
+}
+
 void ssd_scan(at::Tensor x, at::Tensor dt, at::Tensor A, at::Tensor B, at::Tensor C,
               std::optional<at::Tensor> D, std::optional<at::Tensor> z,
               std::optional<at::Tensor> dt_bias,
@@ -795,5 +849,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gdn_decode", &gdn_decode);
   m.def("gdn_chunk", &gdn_chunk);
   m.def("ssd_scan", &ssd_scan);
+  m.def("mhc_post", &mhc_post);
+  m.def("mhc_pre", &mhc_pre);
+  m.def("concat_mla_k", &concat_mla_k);
   m.def("debug_fastdiv", &debug_fastdiv);
 }
