@@ -142,6 +142,7 @@ __all__ = [  # noqa: F405
 from . import profiler
 from . import concat_ops
 from . import logits_processor
+from . import msa_ops
 from . import dsv3_ops
 from .mhc import mhc_post, mhc_pre_big_fuse, mhc_pre_big_fuse_with_prenorm
 from .concat_ops import concat_mla_k
