@@ -214,9 +214,9 @@ extern "C" hipError_t fi_mhc_post(int dtype, const void* x, const void* residual
                      (const T*)residual, post_mix, comb_mix, (T*)out, tokens, H)
   bool v8 = (H % 8 == 0);
   switch (dtype) {
-    case 0: v8 ? LMP(fi::bf16, 8) : LMP(fi::bf16, 1); break;
-    case 1: v8 ? LMP(fi::fp16, 8) : LMP(fi::fp16, 1); break;
-    case 2: v8 ? LMP(float, 4) : LMP(float, 1); break;
+    case 0: if (v8) LMP(fi::bf16, 8); else LMP(fi::bf16, 1); break;
+    case 1: if (v8) LMP(fi::fp16, 8); else LMP(fi::fp16, 1); break;
+    case 2: if (v8) LMP(float, 4); else LMP(float, 1); break;
     default: return hipErrorInvalidValue;
   }
 #undef LMP
@@ -246,9 +246,9 @@ extern "C" hipError_t fi_mhc_pre(int dtype, int prenorm, const float* dot_mix,
   } while (0)
   bool v8 = (H % 8 == 0);
   switch (dtype) {
-    case 0: v8 ? LMQ2(fi::bf16, 8) : LMQ2(fi::bf16, 1); break;
-    case 1: v8 ? LMQ2(fi::fp16, 8) : LMQ2(fi::fp16, 1); break;
-    case 2: v8 ? LMQ2(float, 4) : LMQ2(float, 1); break;
+    case 0: if (v8) LMQ2(fi::bf16, 8); else LMQ2(fi::bf16, 1); break;
+    case 1: if (v8) LMQ2(fi::fp16, 8); else LMQ2(fi::fp16, 1); break;
+    case 2: if (v8) LMQ2(float, 4); else LMQ2(float, 1); break;
     default: return hipErrorInvalidValue;
   }
 #undef LMQ2
@@ -267,9 +267,9 @@ extern "C" hipError_t fi_concat_mla_k(int dtype, void* k, const void* k_nope,
                      (const T*)k_nope, (const T*)k_rope, tokens, Hk, nope, rope)
   bool v8 = (nope % 8 == 0) && (rope % 8 == 0);
   switch (dtype) {
-    case 0: v8 ? LCK(fi::bf16, 8) : LCK(fi::bf16, 1); break;
-    case 1: v8 ? LCK(fi::fp16, 8) : LCK(fi::fp16, 1); break;
-    case 3: v8 ? LCK(fi::fp8_e4m3, 8) : LCK(fi::fp8_e4m3, 1); break;
+    case 0: if (v8) LCK(fi::bf16, 8); else LCK(fi::bf16, 1); break;
+    case 1: if (v8) LCK(fi::fp16, 8); else LCK(fi::fp16, 1); break;
+    case 3: if (v8) LCK(fi::fp8_e4m3, 8); else LCK(fi::fp8_e4m3, 1); break;
     default: return hipErrorInvalidValue;
   }
 #undef LCK
