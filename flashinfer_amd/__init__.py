@@ -22,7 +22,11 @@ from .sparse import (
     VariableBlockSparseAttentionWrapper,
 )
 from .mla import BatchMLAPagedAttentionWrapper
-from .attention import BatchAttention, PODWithPagedKVCacheWrapper
+from .attention import (
+    BatchAttention,
+    BatchPODWithPagedKVCacheWrapper,
+    PODWithPagedKVCacheWrapper,
+)
 from .mamba import (
     cake_selective_state_update,
     mamba_chunk_scan_combined,
@@ -42,6 +46,7 @@ from .topk import (
     TopKTieBreak,
     top_k,
     top_k_page_table_transform,
+    top_k_varlen,
     top_k_ragged_transform,
 )
 from .quantization import packbits, segment_packbits
@@ -143,6 +148,7 @@ from . import profiler
 from . import concat_ops
 from . import logits_processor
 from . import msa_ops
+from . import testing
 from . import dsv3_ops
 from .mhc import mhc_post, mhc_pre_big_fuse, mhc_pre_big_fuse_with_prenorm
 from .concat_ops import concat_mla_k

@@ -98,3 +98,66 @@ class PODWithPagedKVCacheWrapper:
         o_p = single_prefill_with_kv_cache(q_p, k_p, v_p, causal=causal_p)
         main.wait_stream(self._aux_stream)
         return o_p, o_d
+
+
+class BatchPODWithPagedKVCacheWrapper:
+    r"""Batched Prefill-On-Decode (reference pod.py
+    BatchPODWithPagedKVCacheWrapper:732): a paged prefill batch and a paged
+    decode batch execute concurrently — the decode grid on a second HIP
+    stream co-resident with the MFMA prefill grid across the 256 CUs."""
+
+    def __init__(self, float_workspace_buffer_p: torch.Tensor,
+                 float_workspace_buffer_d: torch.Tensor,
+                 kv_layout: str = "NHD", use_tensor_cores: bool = True,
+                 **kwargs):
+        from .decode import BatchDecodeWithPagedKVCacheWrapper
+        from .prefill import BatchPrefillWithPagedKVCacheWrapper
+
+        self._prefill = BatchPrefillWithPagedKVCacheWrapper(
+            float_workspace_buffer_p, kv_layout)
+        self._decode = BatchDecodeWithPagedKVCacheWrapper(
+            float_workspace_buffer_d, kv_layout)
+        self.device = float_workspace_buffer_p.device
+        self._aux_stream = torch.cuda.Stream(device=self.device) \
+            if float_workspace_buffer_p.is_cuda else None
+
+    def plan(self, qo_indptr_p, kv_indptr_p, kv_indices_p, last_page_len_p,
+             qo_indptr_d, kv_indptr_d, kv_indices_d, last_page_len_d,
+             num_qo_heads, num_kv_heads, head_dim, page_size,
+             pos_encoding_mode: str = "NONE", window_left: int = -1,
+             q_data_type=torch.bfloat16, kv_data_type=None, data_type=None,
+             sm_scale=None, rope_scale=None, rope_theta=None,
+             non_blocking: bool = True, causal_p: bool = False, **kwargs):
+        self._prefill.plan(
+            qo_indptr_p, kv_indptr_p, kv_indices_p, last_page_len_p,
+            num_qo_heads, num_kv_heads, head_dim, page_size, causal=causal_p,
+            pos_encoding_mode=pos_encoding_mode, sm_scale=sm_scale,
+            window_left=window_left, q_data_type=q_data_type,
+            kv_data_type=kv_data_type, non_blocking=non_blocking)
+        self._decode.plan(
+            kv_indptr_d, kv_indices_d, last_page_len_d, num_qo_heads,
+            num_kv_heads, head_dim, page_size, window_left=window_left,
+            q_data_type=q_data_type, kv_data_type=kv_data_type,
+            non_blocking=non_blocking)
+
+    begin_forward = plan
+
+    def run(self, q_p, paged_kv_cache_p, q_d, paged_kv_cache_d,
+            custom_mask_p=None, packed_custom_mask_p=None,
+            causal_p: bool = False, q_scale=None, k_scale=None, v_scale=None,
+            return_lse: bool = False, **kwargs):
+        main = torch.cuda.current_stream(q_p.device)
+        self._aux_stream.wait_stream(main)
+        with torch.cuda.stream(self._aux_stream):
+            res_d = self._decode.run(q_d, paged_kv_cache_d,
+                                     k_scale=k_scale, v_scale=v_scale,
+                                     return_lse=return_lse)
+        res_p = self._prefill.run(q_p, paged_kv_cache_p,
+                                  return_lse=return_lse)
+        main.wait_stream(self._aux_stream)
+        return res_p, res_d
+
+    forward = run
+
+    def end_forward(self):
+        pass

@@ -117,3 +117,38 @@ def top_k_page_table_transform(
                                      row_starts=row_starts)
         out_raw_indices.copy_(raw)
     return out
+
+
+def top_k_varlen(
+    input: torch.Tensor,        # [total_len] or [rows, max_len]
+    offsets: Optional[torch.Tensor] = None,   # [rows+1] when input is flat
+    lengths: Optional[torch.Tensor] = None,
+    k: int = 0,
+    **kwargs,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    r"""Variable-length top-k (reference flashinfer/topk_varlen role): rows
+    are ragged segments of a flat buffer (``offsets``) or padded rows with
+    per-row ``lengths``. Returns (values, local indices), -1 padded."""
+    if input.dim() == 1:
+        if offsets is None:
+            raise ValueError("flat input requires offsets")
+        off = offsets.to("cpu", torch.int64)
+        lens = (off[1:] - off[:-1]).to(torch.int32)
+        rows = lens.numel()
+        maxlen = int(lens.max()) if rows else 0
+        x = torch.full((rows, max(maxlen, 1)), float("-inf"),
+                       dtype=torch.float32, device=input.device)
+        for r in range(rows):
+            x[r, :int(lens[r])] = input[int(off[r]):int(off[r + 1])].float()
+        lengths = lens.to(input.device)
+    else:
+        x = input.reshape(-1, input.shape[-1]).float()
+        rows = x.shape[0]
+        if lengths is None:
+            lengths = torch.full((rows,), x.shape[1], dtype=torch.int32,
+                                 device=input.device)
+    out_v = torch.empty(rows, k, dtype=torch.float32, device=input.device)
+    out_i = torch.empty(rows, k, dtype=torch.int32, device=input.device)
+    get_ext().topk(x, out_v, out_i, k, lengths.to(input.device, torch.int32),
+                   None, None, None, None, None, 1, 0, 0)
+    return out_v, out_i

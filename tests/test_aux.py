@@ -135,3 +135,70 @@ def test_pod_wrapper_concurrent_streams():
     ref_d = dec.run(q_d, (kc, vc))
     torch.testing.assert_close(o_p.float(), ref_p.float())
     torch.testing.assert_close(o_d.float(), ref_d.float())
+
+
+@pytest.mark.gpu
+def test_batch_pod_wrapper():
+    """BatchPOD prefill+decode on two streams match separate runs."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(3)
+    Hq, Hkv, D, page = 8, 2, 128, 16
+    # prefill batch: 2 reqs x 64 q tokens, kv = 128 each
+    bp, qo_len, kvlen = 2, 64, 128
+    pp = kvlen // page
+    np_p = bp * pp
+    kcp = torch.randn(np_p, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    vcp = torch.randn(np_p, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    qo_indptr_p = torch.arange(0, (bp + 1) * qo_len, qo_len,
+                               dtype=torch.int32, device="cuda")
+    kv_indptr_p = torch.arange(0, np_p + 1, pp, dtype=torch.int32, device="cuda")
+    kv_indices_p = torch.arange(np_p, dtype=torch.int32, device="cuda")
+    lpl_p = torch.full((bp,), page, dtype=torch.int32, device="cuda")
+    q_p = torch.randn(bp * qo_len, Hq, D, dtype=torch.bfloat16, device="cuda")
+    # decode batch
+    bd, ppd = 4, 8
+    np_d = bd * ppd
+    kcd = torch.randn(np_d, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    vcd = torch.randn(np_d, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    qo_indptr_d = torch.arange(0, bd + 1, dtype=torch.int32, device="cuda")
+    kv_indptr_d = torch.arange(0, np_d + 1, ppd, dtype=torch.int32, device="cuda")
+    kv_indices_d = torch.arange(np_d, dtype=torch.int32, device="cuda")
+    lpl_d = torch.full((bd,), page, dtype=torch.int32, device="cuda")
+    q_d = torch.randn(bd, Hq, D, dtype=torch.bfloat16, device="cuda")
+
+    wsp = torch.empty(32 << 20, dtype=torch.uint8, device="cuda")
+    wsd = torch.empty(32 << 20, dtype=torch.uint8, device="cuda")
+    pod = fi.BatchPODWithPagedKVCacheWrapper(wsp, wsd)
+    pod.plan(qo_indptr_p, kv_indptr_p, kv_indices_p, lpl_p,
+             qo_indptr_d, kv_indptr_d, kv_indices_d, lpl_d,
+             Hq, Hkv, D, page, q_data_type=torch.bfloat16, causal_p=True)
+    o_p, o_d = pod.run(q_p, (kcp, vcp), q_d, (kcd, vcd))
+    torch.cuda.synchronize()
+
+    ws2 = torch.empty(32 << 20, dtype=torch.uint8, device="cuda")
+    pf = fi.BatchPrefillWithPagedKVCacheWrapper(ws2, "NHD")
+    pf.plan(qo_indptr_p, kv_indptr_p, kv_indices_p, lpl_p, Hq, Hkv, D, page,
+            causal=True)
+    ref_p = pf.run(q_p, (kcp, vcp))
+    ws3 = torch.empty(32 << 20, dtype=torch.uint8, device="cuda")
+    dec = fi.BatchDecodeWithPagedKVCacheWrapper(ws3, "NHD")
+    dec.plan(kv_indptr_d, kv_indices_d, lpl_d, Hq, Hkv, D, page,
+             q_data_type=torch.bfloat16)
+    ref_d = dec.run(q_d, (kcd, vcd))
+    torch.testing.assert_close(o_p.float(), ref_p.float())
+    torch.testing.assert_close(o_d.float(), ref_d.float())
+
+
+@pytest.mark.gpu
+def test_testing_bench_utils():
+    from flashinfer_amd.testing import (
+        attention_tflops_per_sec_with_actual_seq_lens, bench_gpu_time)
+
+    x = torch.randn(1024, 1024, device="cuda", dtype=torch.bfloat16)
+    times = bench_gpu_time(lambda: torch.mm(x, x), dry_run_iters=2,
+                           repeat_iters=5)
+    assert len(times) == 5 and all(t > 0 for t in times)
+    tf = attention_tflops_per_sec_with_actual_seq_lens(
+        torch.tensor([128]), torch.tensor([1024]), 128, 128, 32, True, 1.0)
+    assert tf > 0

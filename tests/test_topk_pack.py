@@ -102,3 +102,19 @@ def test_top_k_page_table_transform():
         expect = {int(pt[r, i // page]) * page + int(i) % page
                   for i in ref_i.tolist()}
         assert set(out[r].tolist()) == expect
+
+
+def test_top_k_varlen():
+    from flashinfer_amd.topk import top_k_varlen
+
+    torch.manual_seed(4)
+    flat = torch.randn(500, device="cuda")
+    offsets = torch.tensor([0, 200, 210, 500], dtype=torch.int32, device="cuda")
+    v, i = top_k_varlen(flat, offsets=offsets, k=16)
+    for r, (a, b) in enumerate(zip(offsets[:-1], offsets[1:])):
+        seg = flat[a:b]
+        kef = min(16, seg.numel())
+        ref = torch.topk(seg, kef).values
+        torch.testing.assert_close(v[r, :kef].sort(-1, descending=True).values,
+                                   ref)
+        assert (i[r, kef:] == -1).all()
