@@ -26,33 +26,45 @@
 namespace fi {
 
 
-// VPL = 8 elems (16B bf16) per lane along head_dim; LPT = head_dim/8 lanes
-// per token; TPW = 64/LPT tokens per wave.
+// VPL elems per lane along head_dim (8 = 16B, or 32B for the small-group
+// wide path); LPT = head_dim/VPL lanes per token; TPW = 64/LPT tokens/wave.
+// The wide path (VPL=16, HEAD_DIM>=128 && GROUP<=4) halves the per-token
+// shfl/softmax/address overhead per loaded byte — decode is VALU-bound at
+// the target batch sizes (PMC r01) — at the cost of ~2x the accumulator
+// registers, so its occupancy target drops to 2-3 waves/SIMD.
 typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2_t;
 
-template <typename T>
-__device__ __forceinline__ float qk_dot8(const void* qv, const void* kv, float acc) {
+template <typename T, int N>
+__device__ __forceinline__ float qk_dot(const void* qv, const void* kv, float acc) {
   if constexpr (__is_same(T, bf16)) {
     const bf16x2_t* q2 = (const bf16x2_t*)qv;
     const bf16x2_t* k2 = (const bf16x2_t*)kv;
 #pragma unroll
-    for (int i = 0; i < 4; ++i) acc = __builtin_amdgcn_fdot2_f32_bf16(q2[i], k2[i], acc, false);
+    for (int i = 0; i < N / 2; ++i)
+      acc = __builtin_amdgcn_fdot2_f32_bf16(q2[i], k2[i], acc, false);
     return acc;
   } else {
     const T* q = (const T*)qv;
     const T* k = (const T*)kv;
 #pragma unroll
-    for (int i = 0; i < 8; ++i) acc += to_f32<T>(q[i]) * to_f32<T>(k[i]);
+    for (int i = 0; i < N; ++i) acc += to_f32<T>(q[i]) * to_f32<T>(k[i]);
     return acc;
   }
 }
 
+template <int HEAD_DIM, int GROUP>
+struct decode_traits {
+  static constexpr int vpl = (HEAD_DIM >= 128 && GROUP <= 4) ? 16 : 8;
+  static constexpr int occ = vpl == 16 ? (GROUP >= 4 ? 2 : 3) : 4;
+};
+
 template <typename T, typename TKV, int HEAD_DIM, int GROUP, bool SOFT_CAP>
-__global__ __launch_bounds__(256, 4) void batch_decode_kernel(DecodeParams p) {
+__global__ __launch_bounds__(
+    256, (decode_traits<HEAD_DIM, GROUP>::occ)) void batch_decode_kernel(DecodeParams p) {
   constexpr bool kSameT = __is_same(T, TKV);
-  constexpr int VPL = 8;
-  constexpr int LPT = HEAD_DIM / VPL;        // lanes per token: 8/16/32
-  constexpr int TPW = kWaveSize / LPT;       // tokens per wave: 8/4/2
+  constexpr int VPL = decode_traits<HEAD_DIM, GROUP>::vpl;
+  constexpr int LPT = HEAD_DIM / VPL;        // lanes per token
+  constexpr int TPW = kWaveSize / LPT;       // tokens per wave
   const int lane = threadIdx.x & 63;
   const int tsub = lane / LPT;               // which token this lane covers
   const int dcol = (lane % LPT) * VPL;       // feature offset
@@ -149,7 +161,7 @@ __global__ __launch_bounds__(256, 4) void batch_decode_kernel(DecodeParams p) {
     for (int g = 0; g < GROUP; ++g) {
       float s;
       if constexpr (kSameT) {
-        s = qk_dot8<T>(qreg[g].data, kv_cur.data, 0.f);
+        s = qk_dot<T, VPL>(qreg[g].data, kv_cur.data, 0.f);
       } else {
         s = 0.f;
 #pragma unroll

@@ -28,7 +28,10 @@ struct vec_t {
 template <typename T, int N>
 __device__ __forceinline__ void load_bytes(T* dst, const T* src) {
   constexpr int bytes = sizeof(T) * N;
-  if constexpr (bytes == 16) {
+  if constexpr (bytes == 32) {
+    reinterpret_cast<uintx4*>(dst)[0] = reinterpret_cast<const uintx4*>(src)[0];
+    reinterpret_cast<uintx4*>(dst)[1] = reinterpret_cast<const uintx4*>(src)[1];
+  } else if constexpr (bytes == 16) {
     *reinterpret_cast<uintx4*>(dst) = *reinterpret_cast<const uintx4*>(src);
   } else if constexpr (bytes == 8) {
     *reinterpret_cast<uint64_t*>(dst) = *reinterpret_cast<const uint64_t*>(src);
