@@ -149,6 +149,7 @@ from . import concat_ops
 from . import logits_processor
 from . import msa_ops
 from . import testing
+from .xqa import xqa, xqa_mla
 from . import dsv3_ops
 from .mhc import mhc_post, mhc_pre_big_fuse, mhc_pre_big_fuse_with_prenorm
 from .concat_ops import concat_mla_k
