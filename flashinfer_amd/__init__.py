@@ -150,6 +150,9 @@ from . import logits_processor
 from . import msa_ops
 from . import testing
 from .xqa import xqa, xqa_mla
+from . import deep_gemm
+from . import grouped_mm
+from .grouped_mm import grouped_mm_bf16, grouped_mm_fp8
 from . import dsv3_ops
 from .mhc import mhc_post, mhc_pre_big_fuse, mhc_pre_big_fuse_with_prenorm
 from .concat_ops import concat_mla_k

@@ -107,3 +107,66 @@ def test_segment_gemm_wrapper():
             torch.testing.assert_close(y[off:off + L].float(), ref,
                                        atol=5e-2, rtol=5e-2)
         off += L
+
+
+@pytest.mark.gpu
+def test_grouped_mm_bf16():
+    from flashinfer_amd.grouped_mm import grouped_mm_bf16
+
+    torch.manual_seed(0)
+    E, K, N = 3, 256, 128
+    ms = [128, 0, 200]
+    m_indptr = torch.tensor([0, 128, 128, 328], dtype=torch.int32, device="cuda")
+    a = torch.randn(328, K, dtype=torch.bfloat16, device="cuda")
+    b = torch.randn(E, N, K, dtype=torch.bfloat16, device="cuda") / 8
+    out = grouped_mm_bf16(a, b, m_indptr)
+    for g in range(E):
+        s, e = int(m_indptr[g]), int(m_indptr[g + 1])
+        if s == e:
+            continue
+        ref = a[s:e].float() @ b[g].float().t()
+        torch.testing.assert_close(out[s:e].float(), ref, atol=2e-1, rtol=2e-2)
+
+
+@pytest.mark.gpu
+def test_deep_gemm_contiguous_and_masked():
+    import flashinfer_amd as fi
+    from flashinfer_amd.deep_gemm import (
+        m_grouped_fp8_gemm_nt_contiguous, m_grouped_fp8_gemm_nt_masked)
+
+    torch.manual_seed(1)
+    E, K, N = 2, 256, 128
+    b, sfb = fi.per_block_quant_fp8(torch.randn(E, N, K, device="cuda") / 4)
+    # contiguous: rows 0..127 expert 0, 128..255 expert 1
+    M = 256
+    af = torch.randn(M, K, device="cuda") / 4
+    a, sfa = fi.per_token_group_quant_fp8(af.bfloat16())  # [M, K/128]
+    m_indices = torch.repeat_interleave(
+        torch.arange(E, device="cuda"), M // E).int()
+    d = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+    m_grouped_fp8_gemm_nt_contiguous((a, sfa), (b, sfb), d, m_indices)
+    for g in range(E):
+        s, e = g * (M // E), (g + 1) * (M // E)
+        bsc = sfb[g].repeat_interleave(128, 0).repeat_interleave(128, 1)  # [K, N]
+        b_deq = b[g].float() * bsc.t()[:, :K]
+        ref = (a[s:e].float() * sfa[s:e].repeat_interleave(128, 1)) @ \
+              b_deq.t()
+        torch.testing.assert_close(d[s:e].float(), ref, atol=2e-1, rtol=5e-2)
+
+    # masked
+    m_max = 64
+    a2f = torch.randn(E, m_max, K, device="cuda") / 4
+    a2, sfa2_flat = fi.per_token_group_quant_fp8(
+        a2f.reshape(E * m_max, K).bfloat16())
+    a2 = a2.view(E, m_max, K)
+    sfa2 = sfa2_flat.view(E, m_max, -1)
+    masked_m = torch.tensor([40, 10], dtype=torch.int32, device="cuda")
+    d2 = torch.zeros(E, m_max, N, dtype=torch.bfloat16, device="cuda")
+    m_grouped_fp8_gemm_nt_masked((a2, sfa2), (b, sfb), d2, masked_m, 40)
+    for g in range(E):
+        mv = int(masked_m[g])
+        bsc = sfb[g].repeat_interleave(128, 0).repeat_interleave(128, 1)
+        b_deq = b[g].float() * bsc.t()[:, :K]
+        ref = (a2[g, :mv].float() *
+               sfa2[g, :mv].repeat_interleave(128, 1)) @ b_deq.t()
+        torch.testing.assert_close(d2[g, :mv].float(), ref, atol=2e-1, rtol=5e-2)
