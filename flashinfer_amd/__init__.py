@@ -151,6 +151,7 @@ from . import msa_ops
 from . import testing
 from .xqa import xqa, xqa_mla
 from . import deep_gemm
+from . import moe_ep
 from . import grouped_mm
 from .grouped_mm import grouped_mm_bf16, grouped_mm_fp8
 from . import dsv3_ops

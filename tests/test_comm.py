@@ -162,3 +162,62 @@ def test_allreduce_fusion_gloo():
 
 def test_ring_attention_gloo():
     _run_mp("_ring_body", port=29515)
+
+
+def test_eplb_rebalance():
+    from flashinfer_amd.moe_ep import eplb_rebalance
+
+    load = torch.tensor([100.0, 10, 10, 10, 10, 10, 10, 10])
+    phy2log, replicas = eplb_rebalance(load, num_ranks=4, num_slots_per_rank=3)
+    assert phy2log.shape == (4, 3)
+    placed = phy2log[phy2log >= 0]
+    # every expert placed at least once; the hottest gets the extra replicas
+    assert set(placed.tolist()) == set(range(8))
+    assert replicas.sum() == 12 and replicas[0] == replicas.max()
+    # replica loads balance: max rank load close to mean
+    rank_load = torch.zeros(4)
+    for r in range(4):
+        for s in range(3):
+            e = int(phy2log[r, s])
+            if e >= 0:
+                rank_load[r] += float(load[e] / replicas[e])
+    assert rank_load.max() <= rank_load.mean() * 1.6
+
+
+def test_moe_ep_errors_and_enums():
+    from flashinfer_amd import moe_ep
+
+    assert moe_ep.supports_fault_tolerance()
+    with pytest.raises(moe_ep.MoEEpTransportError):
+        raise moe_ep.MoEEpTransportError("dispatch", 3, "timeout")
+    assert moe_ep.EpAlgorithm("split") == moe_ep.EpAlgorithm.SPLIT
+
+
+def _ep_layer_body(rank, world):
+    import torch.distributed as dist
+
+    from flashinfer_amd.moe_ep import MoeEp
+
+    E, K, T, H = 4, 2, 6, 8
+    ep = MoeEp(E, K, enable_fault_tolerance=True)
+    torch.manual_seed(42)  # same logits everywhere
+    logits = torch.randn(T, E)
+    w, ids = ep.route(logits)
+    x = torch.randn(T, H) + rank
+    recv_x, local_exp, state = ep.dispatch(x, ids)
+    assert (local_exp >= 0).all() and (local_exp < ep.experts_per_rank).all()
+    y = recv_x * 2.0  # stand-in expert: f(x) = 2x regardless of expert
+    out = ep.combine(y, w, state)
+    torch.testing.assert_close(out, 2.0 * x, atol=1e-5, rtol=1e-5)
+
+    # fault tolerance: mask the other rank; all routes go local
+    ep.mask_rank(1 - rank)
+    w2, ids2 = ep.route(logits)
+    own = (ids2 // ep.experts_per_rank == rank)
+    assert own.all()
+    ep.clear_faults()
+    assert bool(ep.alive_mask().all())
+
+
+def test_moe_ep_layer_gloo():
+    _run_mp("_ep_layer_body", world=2, port=29531)
