@@ -72,6 +72,7 @@ from .norm import (
     gemma_fused_add_rmsnorm,
     gemma_rmsnorm,
     layernorm,
+    layernorm_quant,
     rmsnorm,
     rmsnorm_quant,
 )

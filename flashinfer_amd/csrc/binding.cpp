@@ -13,6 +13,9 @@ extern "C" {
 hipError_t fi_norm_quant(int add, int dtype, void* x, void* residual, const void* w,
                          void* out, const float* scale, int rows, int d, float eps,
                          hipStream_t stream);
+hipError_t fi_layernorm_quant(int dtype, const void* x, const void* w, const void* b,
+                              void* out, const float* scale, int rows, int d,
+                              float eps, hipStream_t stream);
 hipError_t fi_norm(int which, int dtype, const void* x, const void* w, const void* b,
                    void* y, void* residual, int rows, int d, int64_t sx, int64_t sy,
                    float eps, int weight_bias, hipStream_t stream);
@@ -176,6 +179,17 @@ void layernorm(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> b, at::Tens
 
 void rmsnorm_silu(at::Tensor x, at::Tensor w, at::Tensor out, double eps) {
   norm_common(3, x, w, c10::nullopt, out, c10::nullopt, eps, false);
+}
+
+void layernorm_quant(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> b,
+                     at::Tensor out, at::Tensor scale, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  TORCH_CHECK(out.scalar_type() == at::kFloat8_e4m3fn && out.is_contiguous());
+  check_hip(fi_layernorm_quant(dtype_code(x), x.data_ptr(), w.data_ptr(),
+                               b ? b->data_ptr() : nullptr, out.data_ptr(),
+                               scale.data_ptr<float>(), x.size(0), x.size(1),
+                               (float)eps, cur_stream(x)),
+            "fi_layernorm_quant");
 }
 
 void rmsnorm_quant(at::Tensor x, c10::optional<at::Tensor> residual, at::Tensor w,
@@ -820,6 +834,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm);
   m.def("rmsnorm_silu", &rmsnorm_silu);
   m.def("rmsnorm_quant", &rmsnorm_quant);
+  m.def("layernorm_quant", &layernorm_quant);
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
   m.def("layernorm", &layernorm);
   m.def("act_and_mul", &act_and_mul);

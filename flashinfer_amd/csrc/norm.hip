@@ -163,6 +163,53 @@ __global__ void fused_add_rmsnorm_kernel(T* __restrict__ x, T* __restrict__ resi
   }
 }
 
+// ---------------- LayerNorm + fp8 quantize ----------------
+// out = ((x - mean) / sqrt(var + eps) * w (+ b)) / scale -> fp8 e4m3
+// (reference flashinfer/norm/__init__.py layernorm_quant role)
+template <typename T, int VEC>
+__global__ void layernorm_quant_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                       const T* __restrict__ b,
+                                       fp8_e4m3* __restrict__ out,
+                                       const float* __restrict__ scale, int rows,
+                                       int d, float eps) {
+  __shared__ float smem[kNormWaves + 1];
+  const float inv_scale = 1.f / *scale;
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* xr = x + (int64_t)row * d;
+    fp8_e4m3* orow = out + (int64_t)row * d;
+    float sum = 0.f, sq = 0.f;
+    for (int i = threadIdx.x * VEC; i < d; i += kNormThreads * VEC) {
+      vec_t<T, VEC> v;
+      v.load(xr + i);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float f = v.get(j);
+        sum += f;
+        sq += f * f;
+      }
+    }
+    sum = block_reduce_sum<kNormWaves>(sum, smem);
+    sq = block_reduce_sum<kNormWaves>(sq, smem);
+    float mean = sum / d;
+    float rstd = rsqrtf(sq / d - mean * mean + eps);
+    for (int i = threadIdx.x * VEC; i < d; i += kNormThreads * VEC) {
+      vec_t<T, VEC> v, wv, bv;
+      v.load(xr + i);
+      wv.load(w + i);
+      if (b) bv.load(b + i);
+      vec_t<fp8_e4m3, VEC> q;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float r = (v.get(j) - mean) * rstd * wv.get(j);
+        if (b) r += bv.get(j);
+        q.set(j, r * inv_scale);
+      }
+      q.store(orow + i);
+    }
+    __syncthreads();
+  }
+}
+
 // ---------------- LayerNorm ----------------
 template <typename T, int VEC>
 __global__ void layernorm_kernel(const T* __restrict__ x, const T* __restrict__ w,
@@ -273,6 +320,35 @@ hipError_t norm_quant_launch(bool add, void* x, void* residual, const void* w, v
 }
 
 }  // namespace fi
+
+namespace fi {
+template <typename T>
+hipError_t ln_quant_launch(const void* x, const void* w, const void* b, void* out,
+                           const float* scale, int rows, int d, float eps,
+                           hipStream_t stream) {
+  int grid = rows < 2048 ? rows : 2048;
+  dim3 g(grid), blk(kNormThreads);
+  if (d % 8 == 0)
+    hipLaunchKernelGGL((layernorm_quant_kernel<T, 8>), g, blk, 0, stream, (const T*)x,
+                       (const T*)w, (const T*)b, (fp8_e4m3*)out, scale, rows, d, eps);
+  else
+    hipLaunchKernelGGL((layernorm_quant_kernel<T, 1>), g, blk, 0, stream, (const T*)x,
+                       (const T*)w, (const T*)b, (fp8_e4m3*)out, scale, rows, d, eps);
+  return hipGetLastError();
+}
+}  // namespace fi
+
+extern "C" hipError_t fi_layernorm_quant(int dtype, const void* x, const void* w,
+                                         const void* b, void* out, const float* scale,
+                                         int rows, int d, float eps,
+                                         hipStream_t stream) {
+  switch (dtype) {
+    case 0: return fi::ln_quant_launch<fi::bf16>(x, w, b, out, scale, rows, d, eps, stream);
+    case 1: return fi::ln_quant_launch<fi::fp16>(x, w, b, out, scale, rows, d, eps, stream);
+    case 2: return fi::ln_quant_launch<float>(x, w, b, out, scale, rows, d, eps, stream);
+  }
+  return hipErrorInvalidValue;
+}
 
 extern "C" hipError_t fi_norm_quant(int add, int dtype, void* x, void* residual,
                                     const void* w, void* out, const float* scale,

@@ -142,3 +142,22 @@ def fused_rmsnorm_silu(
     assert out.dtype == input.dtype, "out dtype must match input on CDNA4 path"
     get_ext().rmsnorm_silu(_flat2d(input), weight, _flat2d(out), eps)
     return out
+
+
+@flashinfer_api
+@fi_trace
+def layernorm_quant(
+    out: torch.Tensor,
+    input: torch.Tensor,
+    weight: torch.Tensor,
+    scale,
+    bias: Optional[torch.Tensor] = None,
+    eps: float = 1e-6,
+    enable_pdl: Optional[bool] = None,
+) -> None:
+    r"""Fused LayerNorm + fp8 quantization (reference parity:
+    flashinfer/norm/__init__.py layernorm_quant role):
+    ``out = (((input - mean) / sqrt(var + eps)) * weight (+ bias)
+    / scale).to(fp8_e4m3)``."""
+    get_ext().layernorm_quant(_flat2d(input), weight, bias, _flat2d(out),
+                              _scale_tensor(scale, input.device), eps)

@@ -133,3 +133,19 @@ def test_fused_rmsnorm_silu():
     ref = xf / torch.sqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-6) * w.float()
     ref = ref * torch.sigmoid(ref)
     torch.testing.assert_close(out.float(), ref, atol=2e-2, rtol=2e-2)
+
+
+def test_layernorm_quant():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(3)
+    x = torch.randn(32, 1024, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(1024, dtype=torch.bfloat16, device="cuda")
+    b = torch.randn(1024, dtype=torch.bfloat16, device="cuda")
+    out = torch.empty(32, 1024, dtype=torch.float8_e4m3fn, device="cuda")
+    fi.layernorm_quant(out, x, w, 0.5, bias=b)
+    xf = x.float()
+    ref = (xf - xf.mean(-1, keepdim=True)) / torch.sqrt(
+        xf.var(-1, unbiased=False, keepdim=True) + 1e-6) * w.float() + b.float()
+    ref = (ref / 0.5).to(torch.float8_e4m3fn).float()
+    torch.testing.assert_close(out.float(), ref, atol=0.25, rtol=0.25)
