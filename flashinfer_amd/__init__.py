@@ -153,6 +153,9 @@ from . import testing
 from .xqa import xqa, xqa_mla
 from . import deep_gemm
 from . import moe_ep
+from . import trace_apply
+from . import collect_env
+trace_apply._enable_apply_from_env()
 from . import grouped_mm
 from .grouped_mm import grouped_mm_bf16, grouped_mm_fp8
 from . import dsv3_ops

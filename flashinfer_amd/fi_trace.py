@@ -26,22 +26,26 @@ def _sig(x):
 
 
 def fi_trace(fn):
-    if not _DIR:
-        return fn
-
     @functools.wraps(fn)
     def wrapper(*args, **kwargs):
-        rec = {
-            "api": fn.__qualname__,
-            "args": [_sig(a) for a in args],
-            "kwargs": {k: _sig(v) for k, v in kwargs.items()},
-        }
-        key = json.dumps(rec, sort_keys=True, default=str)
-        if key not in _seen:
-            _seen.add(key)
-            Path(_DIR).mkdir(parents=True, exist_ok=True)
-            with open(Path(_DIR) / f"{fn.__name__}.jsonl", "a") as f:
-                f.write(key + "\n")
+        # trace_apply hook: a registered substitute takes over this call
+        from . import trace_apply as _ta
+
+        sub = _ta._dispatch(fn.__name__) if _ta.is_enabled() else None
+        if sub is not None:
+            return sub(*args, **kwargs)
+        if _DIR:
+            rec = {
+                "api": fn.__qualname__,
+                "args": [_sig(a) for a in args],
+                "kwargs": {k: _sig(v) for k, v in kwargs.items()},
+            }
+            key = json.dumps(rec, sort_keys=True, default=str)
+            if key not in _seen:
+                _seen.add(key)
+                Path(_DIR).mkdir(parents=True, exist_ok=True)
+                with open(Path(_DIR) / f"{fn.__name__}.jsonl", "a") as f:
+                    f.write(key + "\n")
         return fn(*args, **kwargs)
 
     return wrapper

@@ -202,3 +202,34 @@ def test_testing_bench_utils():
     tf = attention_tflops_per_sec_with_actual_seq_lens(
         torch.tensor([128]), torch.tensor([1024]), 128, 128, 32, True, 1.0)
     assert tf > 0
+
+
+def test_trace_apply_substitution():
+    import flashinfer_amd as fi
+    from flashinfer_amd import trace_apply
+
+    calls = []
+
+    def fake_rmsnorm(input, weight, *a, **kw):
+        calls.append(input.shape)
+        return input * 0 + 7
+
+    try:
+        trace_apply.enable_apply({"rmsnorm": fake_rmsnorm})
+        assert trace_apply.is_enabled()
+        x = torch.randn(2, 8)
+        w = torch.randn(8)
+        out = fi.rmsnorm(x, w)
+        assert calls and (out == 7).all()
+        st = trace_apply.stats()
+        assert st[("rmsnorm", "hit")] >= 1
+    finally:
+        trace_apply.disable_apply()
+    assert not trace_apply.is_enabled()
+
+
+def test_collect_env_runs():
+    from flashinfer_amd.collect_env import collect_env
+
+    info = collect_env()
+    assert "torch" in info and "extension_loaded" in info
