@@ -239,6 +239,10 @@ __global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParam
     if (kv0 < wave_kv_hi) {
 #pragma unroll
     for (int kt = 0; kt < KVB / 32; ++kt) {
+      // causal: skip a 32-kv chunk every one of this wave's rows masks out
+      // (the diagonal KVB often has only its first half visible) —
+      // wave_kv_hi is wave-uniform, so the branch is divergence-free
+      if (CAUSAL && kv0 + kt * 32 >= wave_kv_hi) continue;
       // ---- S^T = K * Q^T : [32 kv][32 q] ----
       floatx16 acc_s = {};
       __builtin_amdgcn_s_setprio(1);
