@@ -52,17 +52,22 @@ __device__ __forceinline__ float qk_dot(const void* qv, const void* kv, float ac
   }
 }
 
-template <int HEAD_DIM, int GROUP>
+template <typename T, typename TKV, int HEAD_DIM, int GROUP>
 struct decode_traits {
-  static constexpr int vpl = (HEAD_DIM >= 128 && GROUP <= 4) ? 16 : 8;
-  static constexpr int occ = vpl == 16 ? (GROUP >= 4 ? 2 : 3) : 4;
+  // the wide path needs the packed bf16 dot2 pipeline; f16/f32/fp8-KV paths
+  // carry f32 temporaries that would spill at its register budget
+  static constexpr bool wide =
+      __is_same(T, bf16) && __is_same(TKV, bf16) && HEAD_DIM >= 128 && GROUP <= 4;
+  static constexpr int vpl = wide ? 16 : 8;
+  static constexpr int occ = wide ? (GROUP >= 4 ? 2 : 3) : 4;
 };
 
 template <typename T, typename TKV, int HEAD_DIM, int GROUP, bool SOFT_CAP>
 __global__ __launch_bounds__(
-    256, (decode_traits<HEAD_DIM, GROUP>::occ)) void batch_decode_kernel(DecodeParams p) {
+    256,
+    (decode_traits<T, TKV, HEAD_DIM, GROUP>::occ)) void batch_decode_kernel(DecodeParams p) {
   constexpr bool kSameT = __is_same(T, TKV);
-  constexpr int VPL = decode_traits<HEAD_DIM, GROUP>::vpl;
+  constexpr int VPL = decode_traits<T, TKV, HEAD_DIM, GROUP>::vpl;
   constexpr int LPT = HEAD_DIM / VPL;        // lanes per token
   constexpr int TPW = kWaveSize / LPT;       // tokens per wave
   const int lane = threadIdx.x & 63;
@@ -129,9 +134,8 @@ __global__ __launch_bounds__(
   const TKV* kbase = (const TKV*)p.k_data;
   const TKV* vbase = (const TKV*)p.v_data;
 
-  // two-deep software pipeline: K/V loads for step i+1 are issued before the
-  // softmax/state update of step i so the ~900-cycle HBM latency hides under
-  // the dependent VALU chain (guide T14 async-split idea, register-staged).
+  // software-pipelined K/V loads hide the ~1us HBM latency under the
+  // dependent VALU chain (guide T14 async-split idea, register-staged).
   auto addr_of = [&](int64_t pos0) -> int64_t {
     int64_t pos = pos0 + tsub;
     int64_t ppos = pos < end ? pos : (end - 1);
@@ -140,18 +144,14 @@ __global__ __launch_bounds__(
     return (int64_t)page_ids[page_iter] * p.stride_page +
            (int64_t)kv_head * p.stride_h + (int64_t)entry * p.stride_n + dcol;
   };
-  vec_t<TKV, VPL> kv_cur, vv_cur, kv_nxt, vv_nxt;
-  if (start < end) {
-    int64_t off = addr_of(start);
-    kv_cur.load(kbase + off);
-    vv_cur.load(vbase + off);
-  }
-  for (int64_t pos0 = start; pos0 < end; pos0 += TPW) {
-    if (pos0 + TPW < end) {
-      int64_t off = addr_of(pos0 + TPW);
-      kv_nxt.load(kbase + off);
-      vv_nxt.load(vbase + off);
-    }
+  // STAGES-deep register ring (static indices via the unrolled sub-loop):
+  // at 2-3 waves/SIMD the wide path has to keep >= 3 iterations of K/V in
+  // flight to cover the ~1us HBM latency (PMC r01: MemUnitStalled ~0,
+  // VALUBusy ~30% at 2-deep — latency-, not bandwidth- or VALU-bound).
+  constexpr int STAGES = VPL == 16 ? (GROUP >= 4 ? 4 : 3) : 2;
+  vec_t<TKV, VPL> kvb[STAGES], vvb[STAGES];
+  auto process = [&](const vec_t<TKV, VPL>& kv_cur, const vec_t<TKV, VPL>& vv_cur,
+                     int64_t pos0) {
     bool valid = pos0 + tsub < end;
     float vf[VPL];
 #pragma unroll
@@ -193,8 +193,47 @@ __global__ __launch_bounds__(
         }
       }
     }
-    kv_cur = kv_nxt;
-    vv_cur = vv_nxt;
+  };
+  if constexpr (STAGES == 2) {
+    // classic copy-forward double buffer (lowest register pressure)
+    if (start < end) {
+      int64_t off = addr_of(start);
+      kvb[0].load(kbase + off);
+      vvb[0].load(vbase + off);
+    }
+    for (int64_t pos0 = start; pos0 < end; pos0 += TPW) {
+      if (pos0 + TPW < end) {
+        int64_t off = addr_of(pos0 + TPW);
+        kvb[1].load(kbase + off);
+        vvb[1].load(vbase + off);
+      }
+      process(kvb[0], vvb[0], pos0);
+      kvb[0] = kvb[1];
+      vvb[0] = vvb[1];
+    }
+  } else {
+#pragma unroll
+    for (int s = 0; s < STAGES - 1; ++s) {
+      if (start + s * TPW < end) {
+        int64_t off = addr_of(start + s * TPW);
+        kvb[s].load(kbase + off);
+        vvb[s].load(vbase + off);
+      }
+    }
+    for (int64_t pos0 = start; pos0 < end; pos0 += (int64_t)TPW * STAGES) {
+#pragma unroll
+      for (int s = 0; s < STAGES; ++s) {
+        int64_t cur = pos0 + s * TPW;
+        if (cur >= end) break;
+        int64_t pf = cur + (int64_t)(STAGES - 1) * TPW;
+        if (pf < end) {
+          int64_t off = addr_of(pf);
+          kvb[(s + STAGES - 1) % STAGES].load(kbase + off);
+          vvb[(s + STAGES - 1) % STAGES].load(vbase + off);
+        }
+        process(kvb[s], vvb[s], cur);
+      }
+    }
   }
 
   // merge the TPW per-token states across the wave: lanes with equal (lane%LPT)
