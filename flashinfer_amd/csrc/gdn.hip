@@ -68,11 +68,13 @@ __global__ void gdn_decode_kernel(TS* __restrict__ state, const T* __restrict__ 
 //   u_t <- beta_t * (v_t - k_t^T S)
 //   S   <- S + k_t (x) u_t
 //   o_t <- scale * q_t^T S
-// One block per (seq, head); the [D, D] f32 state lives in registers, sliced
-// as thread t owning column (t % D) rows [(t/D)*RPT, ...+RPT). Tokens are
-// staged through LDS in chunks of CH so the global loads amortize; the two
-// cross-row reductions per token combine GPC=256/D partials through LDS.
-template <typename T, int D, bool PER_CHANNEL_GATE>
+// One block per (seq, head, column-slice); the [D, D/SPLIT] f32 state slice
+// lives in registers — every step of the recurrence is column-local (the
+// k^T S and q^T S reductions run over ROWS), so Dv splits across SPLIT
+// independent blocks for small batch*heads. Thread t owns column
+// (slice + t % CB), rows [(t/CB)*RPT, +RPT). Tokens stage through LDS in
+// chunks of CH; the cross-row reductions combine GPC partials through LDS.
+template <typename T, int D, bool PER_CHANNEL_GATE, int SPLIT = 1>
 __global__ __launch_bounds__(256, 2) void gdn_chunk_kernel(
     const T* __restrict__ q, const T* __restrict__ k, const T* __restrict__ v,
     const float* __restrict__ gate, const float* __restrict__ beta,
@@ -80,21 +82,24 @@ __global__ __launch_bounds__(256, 2) void gdn_chunk_kernel(
     const float* __restrict__ init_state, float* __restrict__ final_state,
     float scale, int H) {
   constexpr int NT = 256;
-  constexpr int GPC = NT / D;       // thread groups per column
+  constexpr int CB = D / SPLIT;     // columns per block
+  constexpr int GPC = NT / CB;      // thread groups per column
   constexpr int RPT = D / GPC;      // state rows per thread
   constexpr int CH = 16;            // tokens staged per LDS chunk
   __shared__ float kq_s[CH][2 * D];                       // k then q
-  __shared__ float v_s[CH][D];
+  __shared__ float v_s[CH][CB];
   __shared__ float a_s[CH][PER_CHANNEL_GATE ? D : 1];
   __shared__ float b_s[CH];
-  __shared__ float part[2][GPC][D];                       // kS / o partials
-  __shared__ float delta_s[D];
+  __shared__ float part[2][GPC][CB];                      // kS / o partials
+  __shared__ float delta_s[CB];
 
-  const int seq = blockIdx.x / H, h = blockIdx.x % H;
+  const int seq = blockIdx.x / (H * SPLIT);
+  const int h = (blockIdx.x / SPLIT) % H;
+  const int c0 = (blockIdx.x % SPLIT) * CB;               // column slice base
   const int64_t s0 = cu_seqlens[seq], s1 = cu_seqlens[seq + 1];
   const int len = (int)(s1 - s0);
   const int tid = threadIdx.x;
-  const int col = tid % D, grp = tid / D;
+  const int col = c0 + tid % CB, grp = tid / CB;
   const int r0 = grp * RPT;
 
   float S[RPT];
@@ -118,7 +123,7 @@ __global__ __launch_bounds__(256, 2) void gdn_chunk_kernel(
       int64_t row = (s0 + base + t) * (int64_t)H + h;
       kq_s[t][d] = to_f32<T>(k[row * D + d]);
       kq_s[t][D + d] = scale * to_f32<T>(q[row * D + d]);
-      v_s[t][d] = to_f32<T>(v[row * D + d]);
+      if (d >= c0 && d < c0 + CB) v_s[t][d - c0] = to_f32<T>(v[row * D + d]);
       if constexpr (PER_CHANNEL_GATE) a_s[t][d] = gate[row * D + d];
     }
     for (int t = tid; t < nt; t += NT) {
@@ -137,9 +142,9 @@ __global__ __launch_bounds__(256, 2) void gdn_chunk_kernel(
         S[i] *= a;
         acc = __builtin_fmaf(kq_s[t][r0 + i], S[i], acc);
       }
-      part[0][grp][col] = acc;
+      part[0][grp][col - c0] = acc;
       __syncthreads();
-      if (tid < D) {
+      if (tid < CB) {
         float ks = 0.f;
 #pragma unroll
         for (int g2 = 0; g2 < GPC; ++g2) ks += part[0][g2][tid];
@@ -147,20 +152,20 @@ __global__ __launch_bounds__(256, 2) void gdn_chunk_kernel(
       }
       __syncthreads();
       // rank-1 update + q^T S partial
-      const float dlt = delta_s[col];
+      const float dlt = delta_s[col - c0];
       float oacc = 0.f;
 #pragma unroll
       for (int i = 0; i < RPT; ++i) {
         S[i] = __builtin_fmaf(kq_s[t][r0 + i], dlt, S[i]);
         oacc = __builtin_fmaf(kq_s[t][D + r0 + i], S[i], oacc);
       }
-      part[1][grp][col] = oacc;
+      part[1][grp][col - c0] = oacc;
       __syncthreads();
-      if (tid < D) {
+      if (tid < CB) {
         float o = 0.f;
 #pragma unroll
         for (int g2 = 0; g2 < GPC; ++g2) o += part[1][g2][tid];
-        out[((s0 + base + t) * (int64_t)H + h) * D + tid] = from_f32<T>(o);
+        out[((s0 + base + t) * (int64_t)H + h) * D + c0 + tid] = from_f32<T>(o);
       }
       __syncthreads();
     }
@@ -179,17 +184,28 @@ extern "C" hipError_t fi_gdn_chunk(int dtype, int per_channel_gate, const void* 
                                    const int32_t* cu_seqlens, const float* init_state,
                                    float* final_state, float scale, int num_seqs,
                                    int H, int D, hipStream_t stream) {
-  dim3 grid((uint32_t)num_seqs * H), blk(256);
-#define LGC(T, D, PC)                                                            \
-  hipLaunchKernelGGL((fi::gdn_chunk_kernel<T, D, PC>), grid, blk, 0, stream,     \
+  // split the Dv columns across blocks until the 256-CU chip has >= 2
+  // blocks/CU to schedule (each block is sequential over the sequence)
+  int64_t bh = (int64_t)num_seqs * H;
+  int split = 1;
+  while (split < 4 && bh * split * 2 < 512 && D / (split * 2) >= 32) split *= 2;
+  dim3 grid((uint32_t)(bh * split)), blk(256);
+#define LGC(T, D, PC, SP)                                                        \
+  hipLaunchKernelGGL((fi::gdn_chunk_kernel<T, D, PC, SP>), grid, blk, 0, stream, \
                      (const T*)q, (const T*)k, (const T*)v, gate, beta, (T*)out, \
                      cu_seqlens, init_state, final_state, scale, H)
+#define LGC_SP(T, D, PC)                                              \
+  do {                                                                \
+    if (split == 4) LGC(T, D, PC, 4);                                 \
+    else if (split == 2) LGC(T, D, PC, 2);                            \
+    else LGC(T, D, PC, 1);                                            \
+  } while (0)
 #define LGC2(T)                                                       \
   do {                                                                \
-    if (D == 128 && per_channel_gate) LGC(T, 128, true);              \
-    else if (D == 128) LGC(T, 128, false);                            \
-    else if (D == 64 && per_channel_gate) LGC(T, 64, true);           \
-    else if (D == 64) LGC(T, 64, false);                              \
+    if (D == 128 && per_channel_gate) LGC_SP(T, 128, true);           \
+    else if (D == 128) LGC_SP(T, 128, false);                         \
+    else if (D == 64 && per_channel_gate) LGC_SP(T, 64, true);        \
+    else if (D == 64) LGC_SP(T, 64, false);                           \
     else return hipErrorInvalidValue;                                 \
   } while (0)
   switch (dtype) {
@@ -199,6 +215,7 @@ extern "C" hipError_t fi_gdn_chunk(int dtype, int per_channel_gate, const void* 
     default: return hipErrorInvalidValue;
   }
 #undef LGC2
+#undef LGC_SP
 #undef LGC
   return hipGetLastError();
 }
