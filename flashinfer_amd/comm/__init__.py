@@ -9,6 +9,8 @@ from .mapping import Mapping
 from .moe_alltoall import MoeAlltoAll
 from .ulysses import UlyssesCommunicator
 from .dcp import dcp_gather_o, dcp_scatter_q
+from .all_gather_matmul import all_gather_matmul
+from .quantized_allreduce import quantized_all_reduce
 
 __all__ = [
     "AllReduceFusionPattern",

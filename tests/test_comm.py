@@ -221,3 +221,40 @@ def _ep_layer_body(rank, world):
 
 def test_moe_ep_layer_gloo():
     _run_mp("_ep_layer_body", world=2, port=29531)
+
+
+def _agmm_body(rank, world):
+    from flashinfer_amd.comm.all_gather_matmul import all_gather_matmul
+
+    torch.manual_seed(10 + rank)
+    m, K, N = 4, 16, 8
+    inp = torch.randn(m, K)
+    w_shared = torch.ones(K, N) * 0.5  # same on all ranks
+    out = all_gather_matmul(inp, w_shared)
+    # reference
+    import torch.distributed as dist
+    gathered = [torch.empty_like(inp) for _ in range(world)]
+    dist.all_gather(gathered, inp)
+    ref = torch.cat(gathered) @ w_shared
+    torch.testing.assert_close(out, ref, atol=1e-5, rtol=1e-5)
+
+
+def test_all_gather_matmul_gloo():
+    _run_mp("_agmm_body", world=2, port=29541)
+
+
+def _qar_body(rank, world):
+    from flashinfer_amd.comm.quantized_allreduce import quantized_all_reduce
+
+    torch.manual_seed(20 + rank)
+    x = torch.randn(4, 256 * world)  # numel divides world * scale_group(256)
+    out = quantized_all_reduce(x, scale_group=256)
+    import torch.distributed as dist
+    ref = x.clone()
+    dist.all_reduce(ref)
+    # fp8 payloads: ~2^-3 relative error bound per element group
+    torch.testing.assert_close(out, ref, atol=0.25, rtol=0.1)
+
+
+def test_quantized_all_reduce_gloo():
+    _run_mp("_qar_body", world=2, port=29551)
