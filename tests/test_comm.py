@@ -258,3 +258,50 @@ def _qar_body(rank, world):
 
 def test_quantized_all_reduce_gloo():
     _run_mp("_qar_body", world=2, port=29551)
+
+
+@pytest.mark.gpu
+def test_hip_ipc_two_process_one_gpu():
+    """Real hipIpc handle exchange: two processes share one GPU buffer."""
+    import torch.multiprocessing as mp2
+
+    ctx = mp2.get_context("spawn")
+    q01 = ctx.Queue()
+    q10 = ctx.Queue()
+    procs = [ctx.Process(target=_ipc_worker, args=(r, q01, q10))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(120)
+    for p in procs:
+        assert p.exitcode == 0, f"ipc worker failed with {p.exitcode}"
+
+
+def _ipc_worker(rank, q01, q10):
+    import ctypes
+
+    import torch
+
+    from flashinfer_amd.comm.hip_ipc import (HipRTLibrary, hipIpcMemHandle_t)
+
+    torch.cuda.init()
+    rt = HipRTLibrary()
+    rt.hipSetDevice(0)
+    if rank == 0:
+        ptr = rt.hipMalloc(4096)
+        rt.hipMemset(ptr, 0x5A, 4096)
+        rt.hipDeviceSynchronize()
+        q01.put(rt.hipIpcGetMemHandle(ptr).bytes_())
+        assert q10.get(timeout=60) == "ok"
+        rt.hipFree(ptr)
+    else:
+        handle = hipIpcMemHandle_t.from_bytes(q01.get(timeout=60))
+        ptr = rt.hipIpcOpenMemHandle(handle)
+        host = (ctypes.c_byte * 4096)()
+        # hipMemcpyDeviceToHost = 2
+        rt._check(rt.lib.hipMemcpy(ctypes.byref(host), ctypes.c_void_p(ptr),
+                                   4096, 2), "hipMemcpy")
+        assert all((b & 0xFF) == 0x5A for b in bytes(host))
+        rt.hipIpcCloseMemHandle(ptr)
+        q10.put("ok")
