@@ -34,6 +34,8 @@ class HipRTLibrary:
         self.lib = ctypes.CDLL(path)
         self.lib.hipMalloc.argtypes = [ctypes.POINTER(ctypes.c_void_p),
                                        ctypes.c_size_t]
+        self.lib.hipExtMallocWithFlags.argtypes = [
+            ctypes.POINTER(ctypes.c_void_p), ctypes.c_size_t, ctypes.c_uint]
         self.lib.hipFree.argtypes = [ctypes.c_void_p]
         self.lib.hipMemset.argtypes = [ctypes.c_void_p, ctypes.c_int,
                                        ctypes.c_size_t]
@@ -59,6 +61,14 @@ class HipRTLibrary:
     def hipMalloc(self, nbytes: int) -> int:
         p = ctypes.c_void_p()
         self._check(self.lib.hipMalloc(ctypes.byref(p), nbytes), "hipMalloc")
+        return p.value
+
+    def hipMallocFineGrained(self, nbytes: int) -> int:
+        r"""System-coherent (fine-grained) allocation — required for buffers
+        that peer processes read mid-kernel (hipDeviceMallocFinegrained)."""
+        p = ctypes.c_void_p()
+        self._check(self.lib.hipExtMallocWithFlags(ctypes.byref(p), nbytes, 0x1),
+                    "hipExtMallocWithFlags")
         return p.value
 
     def hipFree(self, ptr: int):
@@ -100,7 +110,8 @@ def hip_rt() -> HipRTLibrary:
     return _rt
 
 
-def create_shared_buffer(nbytes: int, group=None) -> List[int]:
+def create_shared_buffer(nbytes: int, group=None,
+                         fine_grained: bool = True) -> List[int]:
     r"""Allocate ``nbytes`` on this rank's GPU, exchange hipIpc handles over
     the (gloo/RCCL) group, and open every peer's buffer: returns one device
     pointer per rank (own rank's is the local allocation). The building
@@ -108,7 +119,7 @@ def create_shared_buffer(nbytes: int, group=None) -> List[int]:
     import torch.distributed as dist
 
     rt = hip_rt()
-    ptr = rt.hipMalloc(nbytes)
+    ptr = rt.hipMallocFineGrained(nbytes) if fine_grained else rt.hipMalloc(nbytes)
     handle = rt.hipIpcGetMemHandle(ptr).bytes_()
     world = dist.get_world_size(group)
     rank = dist.get_rank(group)

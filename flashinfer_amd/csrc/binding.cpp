@@ -98,6 +98,11 @@ hipError_t fi_gdn_decode(int dtype, int state_f32, int per_channel_gate, void* s
                          const void* q, const void* k, const void* v, const float* g,
                          const float* beta, void* out, int B, int H, int Dk, int Dv,
                          hipStream_t stream);
+hipError_t fi_one_shot_ar(int dtype, fi::ARParams* p, void* out, int64_t numel,
+                          hipStream_t stream);
+hipError_t fi_one_shot_ar_rmsnorm(int dtype, fi::ARParams* p, void* out,
+                                  void* residual, const void* weight, int rows,
+                                  int d, float eps, hipStream_t stream);
 hipError_t fi_mhc_post(int dtype, const void* x, const void* residual,
                        const float* post_mix, const float* comb_mix, void* out,
                        int64_t tokens, int H, hipStream_t stream);
@@ -737,6 +742,52 @@ void gdn_decode(at::Tensor state, at::Tensor q, at::Tensor k, at::Tensor v,
             "fi_gdn_decode");
 }
 
+void ipc_memcpy_to(int64_t dst_ptr, at::Tensor src) {
+  TORCH_CHECK(src.is_cuda() && src.is_contiguous());
+  check_hip(hipMemcpyAsync((void*)dst_ptr, src.data_ptr(),
+                           src.numel() * src.element_size(),
+                           hipMemcpyDeviceToDevice, cur_stream(src)),
+            "hipMemcpyAsync");
+}
+
+static fi::ARParams make_ar_params(const std::vector<int64_t>& bufs, int64_t rank,
+                                   int64_t seq, at::Tensor error_flag,
+                                   int64_t spin_limit, int64_t data_off) {
+  fi::ARParams p{};
+  TORCH_CHECK((int)bufs.size() <= fi::kMaxRanks);
+  for (size_t i = 0; i < bufs.size(); ++i) p.bufs[i] = (unsigned long long)bufs[i];
+  p.data_off = (unsigned long long)data_off;
+  p.world = (int)bufs.size();
+  p.rank = (int)rank;
+  p.seq = (unsigned long long)seq;
+  p.spin_limit = (unsigned long long)spin_limit;
+  p.error_flag = error_flag.data_ptr<int32_t>();
+  return p;
+}
+
+void one_shot_all_reduce(at::Tensor out, std::vector<int64_t> bufs, int64_t rank,
+                         int64_t seq, at::Tensor error_flag, int64_t spin_limit,
+                         int64_t data_off) {
+  fi::ARParams p = make_ar_params(bufs, rank, seq, error_flag, spin_limit, data_off);
+  check_hip(fi_one_shot_ar(dtype_code(out), &p, out.data_ptr(), out.numel(),
+                           cur_stream(out)),
+            "fi_one_shot_ar");
+}
+
+void one_shot_all_reduce_rmsnorm(at::Tensor out, c10::optional<at::Tensor> residual,
+                                 at::Tensor weight, std::vector<int64_t> bufs,
+                                 int64_t rank, int64_t seq, at::Tensor error_flag,
+                                 int64_t spin_limit, int64_t data_off, double eps) {
+  fi::ARParams p = make_ar_params(bufs, rank, seq, error_flag, spin_limit, data_off);
+  int d = out.size(-1);
+  int rows = out.numel() / d;
+  check_hip(fi_one_shot_ar_rmsnorm(dtype_code(out), &p, out.data_ptr(),
+                                   residual ? residual->data_ptr() : nullptr,
+                                   weight.data_ptr(), rows, d, (float)eps,
+                                   cur_stream(out)),
+            "fi_one_shot_ar_rmsnorm");
+}
+
 void mhc_post(at::Tensor x, at::Tensor residual, at::Tensor post_mix,
               at::Tensor comb_mix, at::Tensor out) {
   TORCH_CHECK(x.is_cuda() && residual.dim() == 3 && residual.size(1) == 4);
@@ -867,5 +918,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mhc_post", &mhc_post);
   m.def("mhc_pre", &mhc_pre);
   m.def("concat_mla_k", &concat_mla_k);
+  m.def("ipc_memcpy_to", &ipc_memcpy_to);
+  m.def("one_shot_all_reduce", &one_shot_all_reduce);
+  m.def("one_shot_all_reduce_rmsnorm", &one_shot_all_reduce_rmsnorm);
   m.def("debug_fastdiv", &debug_fastdiv);
 }

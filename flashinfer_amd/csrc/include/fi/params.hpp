@@ -78,6 +78,18 @@ struct PrefillParams {
   unsigned long long* prof_buf;    // intra-kernel event buffer (optional)
 };
 
+
+// one-shot hipIpc allreduce (csrc/comm_ar.hip)
+constexpr int kMaxRanks = 8;
+struct ARParams {
+  unsigned long long bufs[kMaxRanks];  // device pointers (rank-local mapping)
+  unsigned long long data_off;  // double-buffer slot offset (seq parity)
+  int world, rank;
+  unsigned long long seq;
+  unsigned long long spin_limit;
+  int* error_flag;  // device int: set to 1 on spin timeout
+};
+
 struct SamplingParams {
   const float* probs;     // [rows, vocab] (or logits)
   float* out_probs;       // optional output distribution
