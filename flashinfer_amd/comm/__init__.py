@@ -12,6 +12,7 @@ from .dcp import dcp_gather_o, dcp_scatter_q
 from .all_gather_matmul import all_gather_matmul
 from .quantized_allreduce import quantized_all_reduce
 from .hip_ipc import create_shared_buffer, free_shared_buffer
+from .custom_ar import CustomAllReduce
 
 __all__ = [
     "AllReduceFusionPattern",
