@@ -69,6 +69,7 @@ from .norm import (
     fused_add_rmsnorm,
     fused_add_rmsnorm_quant,
     fused_rmsnorm_silu,
+    fused_qk_rmsnorm_rope,
     gemma_fused_add_rmsnorm,
     gemma_rmsnorm,
     layernorm,

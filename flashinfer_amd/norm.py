@@ -1,6 +1,8 @@
 """RMSNorm / LayerNorm ops (reference parity: flashinfer/norm/__init__.py)."""
 from __future__ import annotations
 
+import functools
+
 from typing import Optional
 
 import torch
@@ -161,3 +163,97 @@ def layernorm_quant(
     / scale).to(fp8_e4m3)``."""
     get_ext().layernorm_quant(_flat2d(input), weight, bias, _flat2d(out),
                               _scale_tensor(scale, input.device), eps)
+
+
+@functools.lru_cache(maxsize=8)
+def _rope3d_cache(ppf: int, pph: int, ppw: int, head_dim: int, base: float,
+                  device_str: str):
+    """Interleaved 3D (frame/height/width) rope tables, WAN-style split:
+    h_dim = w_dim = 2*(head_dim//6), t_dim = remainder."""
+    device = torch.device(device_str)
+    h_dim = w_dim = 2 * (head_dim // 6)
+    t_dim = head_dim - h_dim - w_dim
+
+    def table(dim, length):
+        inv = 1.0 / (base ** (torch.arange(0, dim, 2, device=device,
+                                           dtype=torch.float64) / dim))
+        pos = torch.arange(length, device=device, dtype=torch.float64)
+        fr = torch.einsum("i,j->ij", pos, inv)
+        cos = torch.repeat_interleave(torch.cos(fr), 2, dim=-1)
+        sin = torch.repeat_interleave(torch.sin(fr), 2, dim=-1)
+        return cos, sin
+
+    mx = max(ppf, pph, ppw)
+    tc, ts = table(t_dim, mx)
+    hc, hs = table(h_dim, mx)
+    wc, ws = table(w_dim, mx)
+    S = ppf * pph * ppw
+    tok = torch.arange(S, device=device)
+    pt, ph, pw = tok // (pph * ppw), (tok // ppw) % pph, tok % ppw
+    cos = torch.cat([tc[pt], hc[ph], wc[pw]], dim=-1).float()
+    sin = torch.cat([ts[pt], hs[ph], ws[pw]], dim=-1).float()
+    return cos, sin  # [S, head_dim]
+
+
+def fused_qk_rmsnorm_rope(
+    qkv: torch.Tensor,           # [B, S, (Hq+Hk+Hv)*D]
+    q_weight: torch.Tensor,      # [D] (or [Hq*D] per-hidden)
+    k_weight: torch.Tensor,
+    *,
+    ppf: int, pph: int, ppw: int,
+    num_frame_channels: int = 0, num_height_channels: int = 0,
+    num_width_channels: int = 0,
+    num_heads_q: int = 0, num_heads_k: int = 0, num_heads_v: int = 0,
+    head_dim: int = 128,
+    eps: float = 1e-6,
+    base: float = 10000.0,
+    interleave: bool = True,
+    factor: float = 1.0, low: float = 0.0, high: float = 0.0,
+    attention_factor: float = 1.0,
+    is_qk_norm: bool = True,
+    output_fp8: bool = False, output_quant_scale: float = 1.0,
+    v_quant_scale: float = 1.0,
+    q_out=None, k_out=None, v_out=None,
+):
+    r"""Video-DiT QKV epilogue (reference parity: flashinfer/norm/__init__.py
+    fused_qk_rmsnorm_rope:1608; math per the reference's own
+    tests/norm/test_fused_qk_rmsnorm_rope.py reference_qk_norm_rope:213):
+    per-head RMSNorm on q/k then interleaved 3D (frame, height, width) RoPE
+    over the ppf x pph x ppw token grid. NeoX element mapping is kernel-
+    specific on the reference side and not supported here."""
+    if not interleave:
+        raise NotImplementedError("neox element mapping not supported")
+    B, S, _ = qkv.shape
+    D = head_dim
+    Hq, Hk, Hv = num_heads_q, num_heads_k, num_heads_v
+    q, k, v = qkv.split([Hq * D, Hk * D, Hv * D], dim=-1)
+    q = q.reshape(B, S, Hq, D)
+    k = k.reshape(B, S, Hk, D)
+    v = v.reshape(B, S, Hv, D)
+    if is_qk_norm:
+        qf = q.float()
+        q = (qf * torch.rsqrt(qf.pow(2).mean(-1, keepdim=True) + eps)
+             * q_weight.float().view(-1)[-D:]).to(qkv.dtype)
+        kf = k.float()
+        k = (kf * torch.rsqrt(kf.pow(2).mean(-1, keepdim=True) + eps)
+             * k_weight.float().view(-1)[-D:]).to(qkv.dtype)
+    cos, sin = _rope3d_cache(ppf, pph, ppw, D, base, str(qkv.device))
+
+    def rope(x):
+        x1, x2 = x.float().unflatten(-1, (-1, 2)).unbind(-1)
+        c = cos[None, :, None, 0::2]
+        s = sin[None, :, None, 1::2]
+        out = torch.empty_like(x, dtype=torch.float32)
+        out[..., 0::2] = x1 * c - x2 * s
+        out[..., 1::2] = x1 * s + x2 * c
+        return (out * attention_factor).to(x.dtype)
+
+    q, k = rope(q), rope(k)
+    if output_fp8:
+        q = (q.float() / output_quant_scale).to(torch.float8_e4m3fn)
+        k = (k.float() / output_quant_scale).to(torch.float8_e4m3fn)
+        v = (v.float() / v_quant_scale).to(torch.float8_e4m3fn)
+    for dst, src in ((q_out, q), (k_out, k), (v_out, v)):
+        if dst is not None:
+            dst.copy_(src)
+    return q, k, v

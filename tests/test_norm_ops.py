@@ -149,3 +149,53 @@ def test_layernorm_quant():
         xf.var(-1, unbiased=False, keepdim=True) + 1e-6) * w.float() + b.float()
     ref = (ref / 0.5).to(torch.float8_e4m3fn).float()
     torch.testing.assert_close(out.float(), ref, atol=0.25, rtol=0.25)
+
+
+def test_fused_qk_rmsnorm_rope_3d():
+    """WAN-style qk-norm + interleaved 3D rope vs the documented reference
+    math (reference tests/norm/test_fused_qk_rmsnorm_rope.py)."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(5)
+    B, ppf, pph, ppw = 2, 3, 4, 5
+    S = ppf * pph * ppw
+    H, D = 4, 128
+    qkv = torch.randn(B, S, 3 * H * D, dtype=torch.bfloat16, device="cuda")
+    qw = torch.randn(D, dtype=torch.bfloat16, device="cuda")
+    kw = torch.randn(D, dtype=torch.bfloat16, device="cuda")
+    q, k, v = fi.fused_qk_rmsnorm_rope(
+        qkv, qw, kw, ppf=ppf, pph=pph, ppw=ppw,
+        num_heads_q=H, num_heads_k=H, num_heads_v=H, head_dim=D)
+    assert q.shape == (B, S, H, D)
+
+    # reference math
+    h_dim = w_dim = 2 * (D // 6)
+    t_dim = D - h_dim - w_dim
+
+    def table(dim, length):
+        inv = 1.0 / (10000.0 ** (torch.arange(0, dim, 2, device="cuda",
+                                              dtype=torch.float64) / dim))
+        pos = torch.arange(length, device="cuda", dtype=torch.float64)
+        fr = torch.einsum("i,j->ij", pos, inv)
+        return (torch.repeat_interleave(torch.cos(fr), 2, -1),
+                torch.repeat_interleave(torch.sin(fr), 2, -1))
+
+    mx = max(ppf, pph, ppw)
+    tc, ts = table(t_dim, mx)
+    hc, hs = table(h_dim, mx)
+    wc, ws = table(w_dim, mx)
+    tok = torch.arange(S, device="cuda")
+    pt, ph, pw = tok // (pph * ppw), (tok // ppw) % pph, tok % ppw
+    cos = torch.cat([tc[pt], hc[ph], wc[pw]], -1).float()
+    sin = torch.cat([ts[pt], hs[ph], ws[pw]], -1).float()
+
+    qkv3 = qkv.view(B, S, 3, H, D)
+    qn = qkv3[:, :, 0].float()
+    qn = qn * torch.rsqrt(qn.pow(2).mean(-1, keepdim=True) + 1e-6) * qw.float()
+    qn = qn.to(torch.bfloat16).float()
+    x1, x2 = qn.unflatten(-1, (-1, 2)).unbind(-1)
+    ref_q = torch.empty_like(qn)
+    ref_q[..., 0::2] = x1 * cos[None, :, None, 0::2] - x2 * sin[None, :, None, 1::2]
+    ref_q[..., 1::2] = x1 * sin[None, :, None, 1::2] + x2 * cos[None, :, None, 0::2]
+    torch.testing.assert_close(q.float(), ref_q, atol=3e-2, rtol=3e-2)
+    torch.testing.assert_close(v.float(), qkv3[:, :, 2].float())
