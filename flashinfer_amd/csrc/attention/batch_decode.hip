@@ -4,11 +4,12 @@
 // for CDNA4:
 //   * 4-wave workgroups, one (work-item, kv-head) unit per wave (64-thread
 //     workgroups hit the per-CU workgroup cap before the wave cap); within a
-//     wave, lanes split into 16-lane columns x 4 kv tokens (head_dim 128;
-//     8/64 and 32/2 for 64/256), 16 B loads per lane.
+//     wave, lanes split into head_dim/VPL-lane columns x tokens. The bf16
+//     GROUP<=4 path loads 32 B/lane (VPL=16) behind a 4-stage register ring
+//     — PMC r01 showed it HBM-LATENCY-bound (MemUnitStalled ~0, VALUBusy
+//     ~30%) at its 2-wave occupancy, so ~3 iterations stay in flight.
 //   * QK dots use packed v_dot2_f32_bf16 (keeps K in bf16 registers, halves
-//     the VALU op count of the convert+FMA form — decode is VALU-bound at
-//     the target batch sizes, not memory-stalled; PMC r01).
+//     the VALU op count of the convert+FMA form).
 //   * GQA group (1/2/4/8 q heads per kv head) processed in-register against
 //     one K read — the K/V bytes are the bound, q reuse is free.
 //   * split-KV via host-planned work items (req, chunk); partials are written
