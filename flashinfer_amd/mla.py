@@ -95,6 +95,10 @@ class BatchMLAPagedAttentionWrapper:
         self._tmp_s = alloc.alloc(
             total_rows * max_chunks * 4, torch.float32, (total_rows * max_chunks, 1)
         )
+        # slots the kernel never writes (rows with < max_chunks chunks) must
+        # merge as empty; the kernel always (re)writes every valid slot, so
+        # one fill at plan time covers every run
+        self._tmp_s.fill_(float("-inf"))
         self._plan_info = dict(
             num_heads=num_heads, chunk=chunk, max_chunks=max_chunks,
             causal=causal, sm_scale=sm_scale, total_rows=total_rows, nnz=nnz,
@@ -118,8 +122,6 @@ class BatchMLAPagedAttentionWrapper:
         pi = self._plan_info
         if pi is None:
             raise RuntimeError("must call plan() before run()")
-        # unused tmp slots must merge as empty
-        self._tmp_s.fill_(float("-inf"))
         get_ext().mla_run(
             q_nope, q_pe, ckv_cache, kpe_cache,
             self._qo_indptr_d, self._kv_indices_d, self._kv_indptr_d,
