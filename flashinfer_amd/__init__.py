@@ -162,3 +162,41 @@ from .grouped_mm import grouped_mm_bf16, grouped_mm_fp8
 from . import dsv3_ops
 from .mhc import mhc_post, mhc_pre_big_fuse, mhc_pre_big_fuse_with_prenorm
 from .concat_ops import concat_mla_k
+from . import jit
+from . import autotuner
+from .autotuner import autotune
+from .decode import fast_decode_plan
+from .attention import BatchAttentionWithAttentionSinkWrapper
+from .cascade import (
+    BatchDecodeWithSharedPrefixPagedKVCacheWrapper,
+    BatchPrefillWithSharedPrefixPagedKVCacheWrapper,
+)
+from .mla import BatchDecodeMlaWithPagedKVCacheWrapper
+from .gemm import bmm_bf16
+from .prefill import single_prefill_with_kv_cache_return_lse
+from .page import append_paged_mla_kv_cache
+from .gdn import gdn_fused_decode_step_supported
+from . import kda
+from .kda import (
+    RecurrentKDAPrefillWorkspace,
+    RecurrentKDAPrefillWrapper,
+    packed_kda_decode,
+    recurrent_kda,
+)
+from .fused_moe import (
+    ActivationType,
+    RoutingMethodType,
+    is_gated_activation,
+    trtllm_bf16_moe,
+    trtllm_bf16_routed_moe,
+    trtllm_fp8_block_scale_routed_moe,
+    trtllm_fp8_per_tensor_scale_moe,
+    trtllm_fp8_per_tensor_scale_routed_moe,
+)
+from .utils import next_positive_power_of_2
+from .norm import (
+    fused_dit_gate_residual_layernorm_gamma_beta,
+    fused_dit_gate_residual_layernorm_scale_shift,
+    fused_dit_residual_layernorm_scale_shift,
+)
+from .fused_moe import trtllm_fp8_block_scale_moe

@@ -161,3 +161,31 @@ class BatchPODWithPagedKVCacheWrapper:
 
     def end_forward(self):
         pass
+
+
+class BatchAttentionWithAttentionSinkWrapper(BatchAttention):
+    r"""Holistic mixed-batch attention with attention sinks (reference
+    flashinfer/attention BatchAttentionWithAttentionSinkWrapper role): a
+    per-head virtual logit joins the softmax denominator only —
+    ``out *= 1 / (1 + 2^(sink*log2(e) - lse))`` applied on the kernel's
+    base-2 LSE."""
+
+    def __init__(self, *args, sink: Optional[torch.Tensor] = None, **kwargs):
+        super().__init__(*args, **kwargs)
+        self._sink = sink
+
+    def plan(self, *args, sink: Optional[torch.Tensor] = None, **kwargs):
+        if sink is not None:
+            self._sink = sink
+        kwargs.pop("use_profiler", None)
+        return super().plan(*args, **kwargs)
+
+    def run(self, q, kv_cache, out=None, lse=None, return_lse: bool = True,
+            sinks: Optional[torch.Tensor] = None, **kwargs):
+        s = sinks if sinks is not None else self._sink
+        o, l2 = super().run(q, kv_cache, out=out, lse=lse, return_lse=True)
+        if s is not None:
+            log2e = 1.4426950408889634
+            corr = 1.0 / (1.0 + torch.exp2(s.float()[None, :] * log2e - l2))
+            o = (o.float() * corr[..., None]).to(o.dtype)
+        return (o, l2) if return_lse else o

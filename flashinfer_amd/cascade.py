@@ -98,3 +98,69 @@ class MultiLevelCascadeAttentionWrapper:
         return v_merged
 
     forward = run
+
+
+class BatchDecodeWithSharedPrefixPagedKVCacheWrapper:
+    r"""Two-level shared-prefix decode (reference cascade.py older API):
+    the shared prefix runs as a single ragged prefill over all queries, the
+    per-request suffix as paged decode, and the two states LSE-merge."""
+
+    def __init__(self, float_workspace_buffer: torch.Tensor,
+                 kv_layout: str = "NHD"):
+        from .decode import BatchDecodeWithPagedKVCacheWrapper
+
+        self._decode = BatchDecodeWithPagedKVCacheWrapper(
+            float_workspace_buffer, kv_layout)
+
+    def plan(self, indptr, indices, last_page_len, num_qo_heads, num_kv_heads,
+             head_dim, page_size, **kwargs):
+        kwargs.setdefault("q_data_type", None)
+        qdt = kwargs.pop("q_data_type") or kwargs.pop("data_type", None)
+        self._decode.plan(indptr, indices, last_page_len, num_qo_heads,
+                          num_kv_heads, head_dim, page_size,
+                          q_data_type=qdt or __import__("torch").bfloat16)
+
+    begin_forward = plan
+
+    def forward(self, q, k_shared, v_shared, unique_kv_cache, **kwargs):
+        from .prefill import single_prefill_with_kv_cache
+
+        o_s, lse_s = single_prefill_with_kv_cache(
+            q, k_shared, v_shared, causal=False, return_lse=True)
+        o_u, lse_u = self._decode.run(q, unique_kv_cache, return_lse=True)
+        merge_state_in_place(o_s, lse_s, o_u, lse_u)
+        return o_s
+
+    run = forward
+
+
+class BatchPrefillWithSharedPrefixPagedKVCacheWrapper:
+    r"""Two-level shared-prefix prefill (reference cascade.py older API)."""
+
+    def __init__(self, float_workspace_buffer: torch.Tensor,
+                 kv_layout: str = "NHD"):
+        from .prefill import BatchPrefillWithPagedKVCacheWrapper
+
+        self._prefill = BatchPrefillWithPagedKVCacheWrapper(
+            float_workspace_buffer, kv_layout)
+
+    def plan(self, qo_indptr, paged_kv_indptr, paged_kv_indices,
+             paged_kv_last_page_len, num_qo_heads, num_kv_heads, head_dim,
+             page_size, causal: bool = True, **kwargs):
+        self._prefill.plan(qo_indptr, paged_kv_indptr, paged_kv_indices,
+                           paged_kv_last_page_len, num_qo_heads, num_kv_heads,
+                           head_dim, page_size, causal=causal)
+
+    begin_forward = plan
+
+    def forward(self, q, k_shared, v_shared, unique_kv_cache,
+                causal: bool = True, **kwargs):
+        from .prefill import single_prefill_with_kv_cache
+
+        o_s, lse_s = single_prefill_with_kv_cache(
+            q, k_shared, v_shared, causal=False, return_lse=True)
+        o_u, lse_u = self._prefill.run(q, unique_kv_cache, return_lse=True)
+        merge_state_in_place(o_s, lse_s, o_u, lse_u)
+        return o_s
+
+    run = forward

@@ -9,6 +9,8 @@ noAuxTcKernels).
 """
 from __future__ import annotations
 
+import enum
+
 from typing import Optional, Tuple
 
 import torch
@@ -182,3 +184,118 @@ def trtllm_fp8_block_scale_moe(
         w, ids = moe_topk_softmax(routing_logits, top_k)
     return fused_moe(hidden_states, gemm1_weights, gemm2_weights, w, ids,
                      w13_scale=gemm1_weights_scale, w2_scale=gemm2_weights_scale)
+
+
+class RoutingMethodType(enum.IntEnum):
+    """Reference trtllm routing-method enum (fused_moe/core.py role)."""
+    Default = 0            # softmax -> top-k
+    Renormalize = 1        # top-k -> softmax over the k
+    DeepSeekV3 = 2         # grouped sigmoid (dsv3_routing)
+    Llama4 = 3             # top-1 sigmoid
+    RenormalizeNaive = 4
+    TopK = 5
+
+
+class ActivationType(enum.IntEnum):
+    Swiglu = 0
+    Geglu = 1
+
+
+def is_gated_activation(act) -> bool:
+    return act in (ActivationType.Swiglu, ActivationType.Geglu, "silu", "gelu")
+
+
+def _route(router_logits, top_k, method: int, routed_scaling_factor=1.0,
+           n_group=1, topk_group=1, routing_bias=None):
+    logits = router_logits.float()
+    if method == RoutingMethodType.DeepSeekV3:
+        return dsv3_routing(logits, top_k, n_group, topk_group,
+                            routed_scaling_factor, bias=routing_bias)
+    if method == RoutingMethodType.Llama4:
+        w, ids = torch.topk(torch.sigmoid(logits), 1, dim=-1)
+        return w * routed_scaling_factor, ids.int()
+    if method in (RoutingMethodType.Renormalize,
+                  RoutingMethodType.RenormalizeNaive):
+        vals, ids = torch.topk(logits, top_k, dim=-1)
+        return torch.softmax(vals, -1) * routed_scaling_factor, ids.int()
+    w, ids = torch.topk(torch.softmax(logits, -1), top_k, dim=-1)
+    return w * routed_scaling_factor, ids.int()
+
+
+def trtllm_bf16_moe(
+    routing_logits: torch.Tensor, hidden_states: torch.Tensor,
+    gemm1_weights: torch.Tensor, gemm2_weights: torch.Tensor,
+    top_k: int, routing_method_type: int = RoutingMethodType.Renormalize,
+    routed_scaling_factor: float = 1.0, n_group: int = 1, topk_group: int = 1,
+    routing_bias=None, **kwargs,
+) -> torch.Tensor:
+    r"""Routed bf16 MoE (reference trtllm_bf16_moe contract): routing +
+    grouped GEMM pipeline in one call."""
+    w, ids = _route(routing_logits, top_k, routing_method_type,
+                    routed_scaling_factor, n_group, topk_group, routing_bias)
+    return fused_moe(hidden_states, gemm1_weights, gemm2_weights, w, ids)
+
+
+def trtllm_bf16_routed_moe(hidden_states, topk_ids, topk_weights,
+                           gemm1_weights, gemm2_weights, **kwargs):
+    r"""Pre-routed variant: caller supplies (ids, weights)."""
+    return fused_moe(hidden_states, gemm1_weights, gemm2_weights,
+                     topk_weights, topk_ids)
+
+
+def trtllm_fp8_per_tensor_scale_moe(
+    routing_logits: torch.Tensor, routing_bias, hidden_states: torch.Tensor,
+    gemm1_weights: torch.Tensor, output1_scales_scalar,
+    gemm2_weights: torch.Tensor, output2_scales_scalar, num_experts: int,
+    top_k: int, n_group: int = 1, topk_group: int = 1,
+    intermediate_size: int = 0, local_expert_offset: int = 0,
+    local_num_experts: int = 0, routed_scaling_factor: float = 1.0,
+    routing_method_type: int = RoutingMethodType.Renormalize, **kwargs,
+) -> torch.Tensor:
+    r"""Routed fp8 MoE with per-tensor weight scales folded into the
+    groupwise pipeline's block scales."""
+    w, ids = _route(routing_logits, top_k, routing_method_type,
+                    routed_scaling_factor, n_group, topk_group, routing_bias)
+    E = gemm1_weights.shape[0]
+    N1, H = gemm1_weights.shape[1], gemm1_weights.shape[2]
+    N2 = gemm2_weights.shape[1]
+    dev = hidden_states.device
+
+    def scales(n, k, s):
+        t = torch.as_tensor(s, dtype=torch.float32, device=dev).reshape(-1)
+        base = torch.ones(E, (k + 127) // 128, (n + 127) // 128,
+                          dtype=torch.float32, device=dev)
+        return base * t.view(-1, 1, 1)
+
+    return fused_moe(
+        hidden_states, gemm1_weights, gemm2_weights, w, ids,
+        w13_scale=scales(N1, H, output1_scales_scalar),
+        w2_scale=scales(N2, N1 // 2, output2_scales_scalar))
+
+
+def trtllm_fp8_per_tensor_scale_routed_moe(hidden_states, topk_ids,
+                                           topk_weights, gemm1_weights,
+                                           s1, gemm2_weights, s2, **kwargs):
+    E = gemm1_weights.shape[0]
+    N1, H = gemm1_weights.shape[1], gemm1_weights.shape[2]
+    N2 = gemm2_weights.shape[1]
+    dev = hidden_states.device
+
+    def scales(n, k, s):
+        t = torch.as_tensor(s, dtype=torch.float32, device=dev).reshape(-1)
+        base = torch.ones(E, (k + 127) // 128, (n + 127) // 128,
+                          dtype=torch.float32, device=dev)
+        return base * t.view(-1, 1, 1)
+
+    return fused_moe(hidden_states, gemm1_weights, gemm2_weights,
+                     topk_weights, topk_ids,
+                     w13_scale=scales(N1, H, s1), w2_scale=scales(N2, N1 // 2, s2))
+
+
+def trtllm_fp8_block_scale_routed_moe(hidden_states, topk_ids, topk_weights,
+                                      gemm1_weights, w13_scale, gemm2_weights,
+                                      w2_scale, **kwargs):
+    r"""Pre-routed variant of trtllm_fp8_block_scale_moe."""
+    return fused_moe(hidden_states, gemm1_weights, gemm2_weights,
+                     topk_weights, topk_ids, w13_scale=w13_scale,
+                     w2_scale=w2_scale)

@@ -83,3 +83,9 @@ def chunk_kda(q, k, v, g_channel, beta, cu_seqlens, scale=None,
     assert g_channel.dim() == 3
     return chunk_gated_delta_rule(q, k, v, g_channel, beta, cu_seqlens, scale,
                                   initial_state, output_final_state, output)
+
+
+def gdn_fused_decode_step_supported(*args, **kwargs) -> bool:
+    r"""Capability probe (reference gdn_kernels role): the CDNA4 decode-step
+    kernel covers every (dtype, state dtype, gate) combination we ship."""
+    return True

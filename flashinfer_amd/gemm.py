@@ -91,3 +91,13 @@ class SegmentGEMMWrapper:
         return out
 
     forward = run
+
+
+def bmm_bf16(
+    a: torch.Tensor, b: torch.Tensor, out: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    r"""Batched ``out[i] = a[i] @ b[i]`` — a plain library GEMM, routed to
+    hipBLASLt's strided-batched path (reference gemm_base.py bmm_bf16 role)."""
+    if out is None:
+        return torch.bmm(a, b)
+    return torch.bmm(a, b, out=out)

@@ -147,3 +147,8 @@ class BatchMLAPagedAttentionWrapper:
 
     def end_forward(self):
         pass
+
+
+# reference decode.py:2406 exposes the MLA decode wrapper under this name;
+# the plan/run contract here follows BatchMLAPagedAttentionWrapper
+BatchDecodeMlaWithPagedKVCacheWrapper = BatchMLAPagedAttentionWrapper

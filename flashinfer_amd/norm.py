@@ -257,3 +257,71 @@ def fused_qk_rmsnorm_rope(
         if dst is not None:
             dst.copy_(src)
     return q, k, v
+
+
+def _dit_tail(res_f32, out_dtype, residual_out, norm_out, gamma=None,
+              beta=None, scale=None, shift=None, eps=1e-6):
+    H = res_f32.shape[-1]
+    n = torch.nn.functional.layer_norm(
+        res_f32, [H],
+        gamma.float() if gamma is not None else None,
+        beta.float() if beta is not None else None, eps)
+    if scale is not None:
+        n = n * (1 + scale.float()) + shift.float()
+    r = res_f32.to(out_dtype)
+    n = n.to(out_dtype)
+    if residual_out is not None:
+        residual_out.copy_(r)
+        r = residual_out
+    if norm_out is not None:
+        norm_out.copy_(n)
+        n = norm_out
+    return r, n
+
+
+def fused_dit_gate_residual_layernorm_gamma_beta(
+    input: torch.Tensor, residual: torch.Tensor, gate: torch.Tensor,
+    gamma: torch.Tensor, beta: torch.Tensor, *, gate_bias=None,
+    epsilon: float = 1e-6, input_global_scaling_factor=None,
+    residual_out=None, norm_out=None, sf_out=None,
+):
+    r"""DiT adaLN epilogue (reference flashinfer/norm/__init__.py
+    fused_dit_gate_residual_layernorm_gamma_beta:1138):
+    ``residual += input * gate; norm = LN(residual, gamma, beta)``."""
+    g = gate.float() + (gate_bias.float() if gate_bias is not None else 0.0)
+    res = residual.float() + input.float() * g
+    return _dit_tail(res, input.dtype, residual_out, norm_out,
+                     gamma=gamma, beta=beta, eps=epsilon)
+
+
+def fused_dit_gate_residual_layernorm_scale_shift(
+    input: torch.Tensor, residual: torch.Tensor, gate: torch.Tensor,
+    scale: torch.Tensor, shift: torch.Tensor, *, gate_bias=None,
+    scale_bias=None, shift_bias=None, epsilon: float = 1e-6,
+    input_global_scaling_factor=None, residual_out=None, norm_out=None,
+    sf_out=None,
+):
+    r"""``residual += input * gate; norm = LN(residual) * (1+scale) + shift``
+    (reference :1267)."""
+    g = gate.float() + (gate_bias.float() if gate_bias is not None else 0.0)
+    sc = scale.float() + (scale_bias.float() if scale_bias is not None else 0.0)
+    sh = shift.float() + (shift_bias.float() if shift_bias is not None else 0.0)
+    res = residual.float() + input.float() * g
+    return _dit_tail(res, input.dtype, residual_out, norm_out,
+                     scale=sc, shift=sh, eps=epsilon)
+
+
+def fused_dit_residual_layernorm_scale_shift(
+    input: torch.Tensor, scale: torch.Tensor, shift: torch.Tensor,
+    residual: Optional[torch.Tensor] = None, *, scale_bias=None,
+    shift_bias=None, epsilon: float = 1e-6,
+    input_global_scaling_factor=None, residual_out=None, norm_out=None,
+    sf_out=None,
+):
+    r"""``residual += input; norm = LN(residual) * (1+scale) + shift``
+    (reference :1389)."""
+    sc = scale.float() + (scale_bias.float() if scale_bias is not None else 0.0)
+    sh = shift.float() + (shift_bias.float() if shift_bias is not None else 0.0)
+    res = input.float() if residual is None else residual.float() + input.float()
+    return _dit_tail(res, input.dtype, residual_out, norm_out,
+                     scale=sc, shift=sh, eps=epsilon)

@@ -61,3 +61,28 @@ def append_paged_kv_cache(
         kv_last_page_len.to(torch.int32), layout_code(kv_layout),
         k_scale, v_scale,
     )
+
+
+def append_paged_mla_kv_cache(
+    append_ckv: torch.Tensor,       # [nnz, 512]
+    append_kpe: torch.Tensor,       # [nnz, 64]
+    batch_indices: torch.Tensor,
+    positions: torch.Tensor,
+    ckv_cache: torch.Tensor,        # [pages, page_size, 512]
+    kpe_cache: torch.Tensor,        # [pages, page_size, 64]
+    kv_indices: torch.Tensor,
+    kv_indptr: torch.Tensor,
+    kv_last_page_len: torch.Tensor,
+) -> None:
+    r"""Append compressed-KV + rope-K rows into the paged MLA caches
+    (reference parity: flashinfer/page.py MLA append /
+    include/flashinfer/page.cuh AppendPagedKVMlaCacheKernel:870). Pure
+    gather/scatter — one indexed copy per cache."""
+    page_size = ckv_cache.shape[1]
+    bi = batch_indices.long()
+    pos = positions.long()
+    page_iter = kv_indptr.long()[bi] + pos // page_size
+    page_ids = kv_indices.long()[page_iter]
+    entry = pos % page_size
+    ckv_cache[page_ids, entry] = append_ckv.to(ckv_cache.dtype)
+    kpe_cache[page_ids, entry] = append_kpe.to(kpe_cache.dtype)

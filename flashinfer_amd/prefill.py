@@ -267,3 +267,10 @@ def single_prefill_with_kv_cache(
         mask_data, mask_indptr, pos_encoding_mode == "ALIBI", 1.0, 1.0, None,
     )
     return (out, lse) if return_lse else out
+
+
+def single_prefill_with_kv_cache_return_lse(*args, **kwargs):
+    r"""Reference-name alias: ``single_prefill_with_kv_cache(...,
+    return_lse=True)``."""
+    kwargs["return_lse"] = True
+    return single_prefill_with_kv_cache(*args, **kwargs)

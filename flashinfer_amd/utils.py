@@ -79,3 +79,10 @@ def ceil_div(a: int, b: int) -> int:
 
 def round_up(a: int, b: int) -> int:
     return ceil_div(a, b) * b
+
+
+def next_positive_power_of_2(x: int) -> int:
+    """Smallest power of two >= max(x, 1) (reference utils role)."""
+    if x < 1:
+        return 1
+    return 1 << (x - 1).bit_length()
