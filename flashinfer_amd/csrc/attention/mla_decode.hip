@@ -175,19 +175,18 @@ __global__ __launch_bounds__(NTH, 1) void mla_decode_kernel(MlaParams p) {
     for (int r = 0; r < 16; ++r)
       Sx[qb][db][mfma32_cd_row(r, lane) * 32 + lq] = acc_s[r];
     __syncthreads();
-    float s_full[16];
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      int idx = mfma32_cd_row(r, lane) * 32 + lq;
-      s_full[r] = Sx[qb][0][idx] + Sx[qb][1][idx] + Sx[qb][2][idx] + Sx[qb][3][idx];
-    }
 
-    // ---- masking + base-2 softmax (identical across db waves) ----
+    // ---- sum partials + masking + base-2 softmax (identical across db
+    // waves; fused so the 16 s_full temporaries never materialize) ----
     float pr[16];
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
-      int64_t kv = kv0 + mfma32_cd_row(r, lane);
-      float sv = s_full[r] * p.sm_scale * kLog2e;
+      int row = mfma32_cd_row(r, lane);
+      int idx = row * 32 + lq;
+      float sv = (Sx[qb][0][idx] + Sx[qb][1][idx] + Sx[qb][2][idx] +
+                  Sx[qb][3][idx]) *
+                 p.sm_scale * kLog2e;
+      int64_t kv = kv0 + row;
       bool ok = kv < end;
       if constexpr (CAUSAL) ok &= kv <= (int64_t)qpos_u + diag;
       pr[r] = ok ? sv : -INFINITY;
