@@ -56,7 +56,7 @@ __device__ __forceinline__ uint32_t swz_row(uint32_t byte_off) {
 // over more q rows).
 template <typename T, typename TKV, int HEAD_DIM, int CTAQ, bool PAGED, bool CAUSAL,
           bool MASK = false>
-__global__ __launch_bounds__(CTAQ * 2, 2) void batch_prefill_kernel(PrefillParams p) {
+__global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_prefill_kernel(PrefillParams p) {
   constexpr int NTHREADS = CTAQ * 2;
   // fp8 (e4m3) KV caches are dequantized to bf16 during the LDS staging
   // write (reference prefill.cuh:1150 repack_fp8_tile_to_bf16 design): the
