@@ -60,7 +60,10 @@ struct decode_traits {
   static constexpr bool wide =
       __is_same(T, bf16) && __is_same(TKV, bf16) && HEAD_DIM >= 128 && GROUP <= 4;
   static constexpr int vpl = wide ? 16 : 8;
-  static constexpr int occ = wide ? (GROUP >= 4 ? 2 : 3) : 4;
+  // narrow path: the GROUP-8/16 accumulators (o_acc[GROUP][8]) blow the
+  // 128-VGPR budget of 4 waves/SIMD — drop the target instead of spilling
+  static constexpr int occ = wide ? (GROUP >= 4 ? 2 : 3)
+                                  : (GROUP >= 16 ? 2 : (GROUP >= 8 ? 3 : 4));
 };
 
 template <typename T, typename TKV, int HEAD_DIM, int GROUP, bool SOFT_CAP>
