@@ -105,3 +105,80 @@ def test_cu_mask_streams():
         y = x @ x
     torch.cuda.synchronize()
     assert y.isfinite().all()
+
+
+def test_prefill_hipgraph_capture():
+    """Prefill run() captures into a hipGraph (fixed grid, no alloc)."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    Hq, Hkv, D, page = 8, 2, 128, 16
+    bs, s = 4, 256
+    pages_per = s // page
+    npages = bs * pages_per
+    kc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    q = torch.randn(bs * s, Hq, D, dtype=torch.bfloat16, device="cuda")
+    out = torch.empty_like(q)
+    qo_indptr = torch.arange(0, (bs + 1) * s, s, dtype=torch.int32, device="cuda")
+    kv_indptr = torch.arange(0, npages + 1, pages_per, dtype=torch.int32,
+                             device="cuda")
+    kv_indices = torch.arange(npages, dtype=torch.int32, device="cuda")
+    lpl = torch.full((bs,), page, dtype=torch.int32, device="cuda")
+    ws = torch.empty(32 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchPrefillWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(qo_indptr, kv_indptr, kv_indices, lpl, Hq, Hkv, D, page, causal=True)
+    w.run(q, (kc, vc), out=out)  # warm
+    torch.cuda.synchronize()
+    ref = out.clone()
+
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        w.run(q, (kc, vc), out=out)
+    out.zero_()
+    g.replay()
+    torch.cuda.synchronize()
+    torch.testing.assert_close(out.float(), ref.float())
+    # replay with mutated inputs picks up the new data
+    q.normal_()
+    w.run(q, (kc, vc), out=ref)
+    torch.cuda.synchronize()
+    ref2 = ref.clone()
+    out.zero_()
+    g.replay()
+    torch.cuda.synchronize()
+    torch.testing.assert_close(out.float(), ref2.float())
+
+
+def test_mla_hipgraph_capture():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(1)
+    B, H, page = 4, 16, 16
+    kv = 256
+    pages_per = kv // page
+    npages = B * pages_per
+    ckv = torch.randn(npages, page, 512, dtype=torch.bfloat16, device="cuda") / 4
+    kpe = torch.randn(npages, page, 64, dtype=torch.bfloat16, device="cuda") / 4
+    qn = torch.randn(B, H, 512, dtype=torch.bfloat16, device="cuda") / 4
+    qp = torch.randn(B, H, 64, dtype=torch.bfloat16, device="cuda") / 4
+    qo_indptr = torch.arange(0, B + 1, dtype=torch.int32, device="cuda")
+    kv_indptr = torch.arange(0, npages + 1, pages_per, dtype=torch.int32,
+                             device="cuda")
+    kv_indices = torch.arange(npages, dtype=torch.int32, device="cuda")
+    kvlen = torch.full((B,), kv, dtype=torch.int32, device="cuda")
+    ws = torch.empty(128 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchMLAPagedAttentionWrapper(ws)
+    w.plan(qo_indptr, kv_indptr, kv_indices, kvlen, H, 512, 64, page,
+           causal=False, sm_scale=576 ** -0.5)
+    out = torch.empty(B, H, 512, dtype=torch.bfloat16, device="cuda")
+    w.run(qn, qp, ckv, kpe, out=out)
+    torch.cuda.synchronize()
+    ref = out.clone()
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        w.run(qn, qp, ckv, kpe, out=out)
+    out.zero_()
+    g.replay()
+    torch.cuda.synchronize()
+    torch.testing.assert_close(out.float(), ref.float())
