@@ -41,18 +41,28 @@ class BlockSparseAttentionWrapper:
         non_blocking: bool = True,
         **kwargs,
     ):
-        if mask is not None or packed_mask is not None:
-            raise NotImplementedError("per-element masks arrive in a later drop")
         if M % R != 0 or N % C != 0:
             raise ValueError("M/N must be divisible by block sizes R/C")
         n_blockrows = M // R
         qo_indptr = torch.arange(0, M + 1, R, dtype=torch.int32)
         last_page_len = torch.full((n_blockrows,), C, dtype=torch.int32)
         self._shape = (M, N, R, C, num_qo_heads, num_kv_heads, head_dim)
+        custom_mask = None
+        if mask is not None and packed_mask is None:
+            # (nnz, R, C) per selected block -> per-request [R, cnt*C] rows
+            # (reference sparse.py convert_bsr_mask_layout:170 layout)
+            ip = indptr.to("cpu", torch.int64)
+            mk = mask.to(torch.bool)
+            parts = [
+                mk[int(ip[i]):int(ip[i + 1])].transpose(0, 1).reshape(-1)
+                for i in range(n_blockrows)
+            ]
+            custom_mask = torch.cat(parts)
         self._wrapper.plan(
             qo_indptr, indptr, indices, last_page_len,
             num_qo_heads, num_kv_heads, head_dim, C,
             causal=False, q_data_type=q_data_type,
+            custom_mask=custom_mask, packed_custom_mask=packed_mask,
         )
 
     begin_forward = plan

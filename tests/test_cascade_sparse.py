@@ -139,3 +139,40 @@ def test_variable_block_sparse_matches_dense():
             ref = torch.softmax(logits, -1) @ v[h].float()
             torch.testing.assert_close(out[h * G + gi].float(), ref,
                                        atol=3e-2, rtol=3e-2)
+
+
+def test_block_sparse_with_element_mask():
+    from flashinfer_amd.sparse import BlockSparseAttentionWrapper
+
+    torch.manual_seed(7)
+    M, N, R, C = 64, 128, 16, 16
+    Hq, Hkv, D = 2, 2, 64
+    MB, NB = M // R, N // C
+    dense_mask = torch.rand(MB, NB, device="cuda") > 0.4
+    dense_mask[:, 0] = True
+    indptr = torch.zeros(MB + 1, dtype=torch.int32, device="cuda")
+    indptr[1:] = dense_mask.sum(1).cumsum(0)
+    indices = torch.nonzero(dense_mask)[:, 1].int()
+    nnz = int(indptr[-1])
+    elem = torch.rand(nnz, R, C, device="cuda") > 0.3
+    elem[:, :, 0] = True  # keep every row non-empty
+
+    q = torch.randn(M, Hq, D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(N, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(N, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(32 << 20, dtype=torch.uint8, device="cuda")
+    w = BlockSparseAttentionWrapper(ws)
+    w.plan(indptr, indices, M, N, R, C, Hq, Hkv, D, mask=elem)
+    out = w.run(q, k, v)
+
+    # dense reference: block mask AND element mask
+    token_mask = torch.zeros(M, N, dtype=torch.bool, device="cuda")
+    for br in range(MB):
+        for j in range(int(indptr[br]), int(indptr[br + 1])):
+            bc = int(indices[j])
+            token_mask[br * R:(br + 1) * R, bc * C:(bc + 1) * C] = elem[j]
+    logits = torch.einsum("mhd,lhd->hml", q.float(), k.float()) / math.sqrt(D)
+    logits = logits.masked_fill(~token_mask[None], float("-inf"))
+    p = torch.softmax(logits, -1)
+    ref = torch.einsum("hml,lhd->mhd", p, v.float())
+    torch.testing.assert_close(out.float(), ref, atol=3e-2, rtol=3e-2)
