@@ -66,13 +66,16 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
   constexpr int DT = HEAD_DIM / 32;   // d-tiles in PV / output
   constexpr int KROWB = HEAD_DIM * 2;        // K/V tile row bytes
 
-  __shared__ T Ks[KVB * HEAD_DIM];
+  // DOUBLE-BUFFERED K/V stage: compute reads buf while the next tile's
+  // registers write buf^1 — ONE barrier per KV tile instead of two (PMC r01:
+  // the kernel is barrier-, not MFMA- or bandwidth-bound).
+  __shared__ T Ks[2][KVB * HEAD_DIM];
   // V is stored in [KVB/4][HEAD_DIM/16] subtiles of [4 kv][16 d] (row-major,
   // 128 B each) padded to 144 B stride: the shape ds_read_b64_tr_b16 wants
   // (each 16-lane group reads one subtile and receives it transposed), with
   // the pad de-aliasing subtile bank positions.
   constexpr int VTILE_STRIDE = 72;  // elems (144 B)
-  __shared__ T Vs[(KVB / 4) * (HEAD_DIM / 16) * VTILE_STRIDE];
+  __shared__ T Vs[2][(KVB / 4) * (HEAD_DIM / 16) * VTILE_STRIDE];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -198,7 +201,7 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
       }
     }
   };
-  auto stage_write = [&]() {
+  auto stage_write = [&](int wb) {
 #pragma unroll
     for (int it = 0; it < S_ITER; ++it) {
       int u = tid + it * NTHREADS;
@@ -217,10 +220,10 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
         vw = *reinterpret_cast<const shortx8*>(vreg[it].data);
       }
       // K: row-major swizzled (vector frag reads); V: tr-read subtiled
-      *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Ks) +
+      *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Ks[wb]) +
                                   swz_row<KROWB>(row * KROWB + chunk * 16)) = kw;
       *reinterpret_cast<shortx8*>(
-          reinterpret_cast<char*>(Vs) +
+          reinterpret_cast<char*>(Vs[wb]) +
           ((row >> 2) * (HEAD_DIM / 16) + (chunk >> 1)) * (VTILE_STRIDE * 2) +
           (row & 3) * 32 + (chunk & 1) * 16) = vw;
     }
@@ -228,10 +231,11 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
 
   prof_event(p.prof_buf, 0, ProfType::kBegin);  // event 0: whole tile
   stage_load(kv_lo);
-  stage_write();
+  stage_write(0);
   __syncthreads();
 
   prof_event(p.prof_buf, 1, ProfType::kBegin);  // event 1: kv mainloop
+  int buf = 0;
   for (int64_t kv0 = kv_lo; kv0 < kv_hi; kv0 += KVB) {
     const bool have_next = kv0 + KVB < kv_hi;
     if (have_next) stage_load(kv0 + KVB);
@@ -249,7 +253,7 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
 #pragma unroll
       for (int c = 0; c < KCH; ++c) {
         frag kfrag = *reinterpret_cast<const frag*>(
-            reinterpret_cast<const char*>(Ks) +
+            reinterpret_cast<const char*>(Ks[buf]) +
             swz_row<KROWB>((kt * 32 + lq) * KROWB + (c * 16 + khalf) * 2));
         acc_s = mfma_ab_frag<T>::mma32(kfrag, qf[c], acc_s);
       }
@@ -362,7 +366,7 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
       // ---- O^T += V^T * P^T: V^T fragments via ds_read_b64_tr_b16 (HW
       // 4x16 transpose read — guide T10); 4 reads + one wait per d-tile ----
       {
-        const uint32_t vbase = (uint32_t)(uintptr_t)Vs + (uint32_t)(lane & 15) * 8;
+        const uint32_t vbase = (uint32_t)(uintptr_t)Vs[buf] + (uint32_t)(lane & 15) * 8;
         const uint32_t tdsel = ((lane >> 4) & 1);
         const int kvb = kt * 32 + khalf;
 #pragma unroll
@@ -392,9 +396,10 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
       }
     }
     }
-    __syncthreads();          // all waves done reading this tile's LDS
-    if (have_next) stage_write();
-    __syncthreads();          // next tile staged
+    if (have_next) stage_write(buf ^ 1);
+    __syncthreads();  // one barrier: this tile's reads AND the next tile's
+                      // writes (to the other buffer) are both complete
+    buf ^= 1;
   }
   prof_event(p.prof_buf, 1, ProfType::kEnd);
 
