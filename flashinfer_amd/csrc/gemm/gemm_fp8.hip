@@ -57,6 +57,7 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
     float scalar_scale, int64_t a_scale_stride) {
   __shared__ uint8_t As[2][BM * BK];
   __shared__ uint8_t Bs[2][BN * BK];
+  __shared__ float ascale_s[2][BM];  // per-row activation scales for this kt
 
   const int seg = blockIdx.z;
   const int m0 = m_indptr[seg] + blockIdx.y * BM;
@@ -85,6 +86,10 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
   int nk = K / BK;
   stage_tile(A, lda, m0, m_end, 0, as_base, tid);
   stage_tile(Wb, ldw_n, bn0, N, 0, bs_base, tid);
+  if (GROUPWISE && tid < BM) {
+    int m = m0 + tid;
+    ascale_s[0][tid] = a_scales[m < m_end ? m : m_end - 1];
+  }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
@@ -93,24 +98,39 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
     if (kt + 1 < nk) {
       stage_tile(A, lda, m0, m_end, (kt + 1) * BK, as_base + (cur ^ 1) * BUF_BYTES, tid);
       stage_tile(Wb, ldw_n, bn0, N, (kt + 1) * BK, bs_base + (cur ^ 1) * BUF_BYTES, tid);
+      if (GROUPWISE && tid < BM) {
+        int m = m0 + tid;
+        ascale_s[cur ^ 1][tid] =
+            a_scales[(int64_t)(kt + 1) * a_scale_stride + (m < m_end ? m : m_end - 1)];
+      }
     }
     const char* a_lds = (const char*)&As[cur][0];
     const char* b_lds = (const char*)&Bs[cur][0];
     floatx16 acc[2];
 #pragma unroll
     for (int i = 0; i < 2; ++i) acc[i] = {};
+    // 32-byte lane fragments for the large-K (x64) fp8 MFMA: two 16-B LDS
+    // reads (the swizzle permutes 16-B blocks within a 128-B row)
+    union frag32 {
+      intx8 v;
+      uint4 q[2];
+    };
+    auto ld32 = [&](const char* base, int row, int kb) {
+      frag32 u;
+      u.q[0] = *reinterpret_cast<const uint4*>(base + swz128(row * BK + kb));
+      u.q[1] = *reinterpret_cast<const uint4*>(base + swz128(row * BK + kb + 16));
+      return u.v;
+    };
+    const int kh32 = (lane >> 5) * 32;
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int ks = 0; ks < BK / 16; ++ks) {
-      int64_t af[2], bfv;
+    for (int g2 = 0; g2 < 2; ++g2) {  // two K=64 groups per BK=128 tile
+      int kb = g2 * 64 + kh32;
+      intx8 bfrag = ld32(b_lds, wn + line, kb);
 #pragma unroll
       for (int i = 0; i < 2; ++i)
-        af[i] = *reinterpret_cast<const int64_t*>(
-            a_lds + (swz128((wm + i * 32 + line) * BK + ks * 16) + khalf));
-      bfv = *reinterpret_cast<const int64_t*>(
-          b_lds + (swz128((wn + line) * BK + ks * 16) + khalf));
-#pragma unroll
-      for (int i = 0; i < 2; ++i) acc[i] = mfma_32x32x16_fp8(af[i], bfv, acc[i]);
+        acc[i] = mfma_32x32x64_fp8(ld32(a_lds, wm + i * 32 + line, kb), bfrag,
+                                   acc[i]);
     }
     __builtin_amdgcn_s_setprio(0);
     // rescale local accumulator into master
@@ -120,8 +140,7 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
       for (int i = 0; i < 2; ++i) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          int m = m0 + wm + i * 32 + mfma32_cd_row(r, lane);
-          float sa = a_scales[(int64_t)kt * a_scale_stride + (m < m_end ? m : m_end - 1)];
+          float sa = ascale_s[cur][wm + i * 32 + mfma32_cd_row(r, lane)];
           accm[i][r] += acc[i][r] * (sa * sb);
         }
       }
