@@ -16,3 +16,16 @@ def test_llama_serving_example():
     out = llama_serving.main(batch=2, prompt_len=32, gen_tokens=4, layers=2,
                              vocab=1000)
     assert out.shape == (2, 4)
+
+
+@pytest.mark.gpu
+def test_deepseek_mla_moe_example():
+    import importlib.util
+    from pathlib import Path
+
+    path = Path(__file__).resolve().parent.parent / "examples" / \
+        "deepseek_mla_moe_serving.py"
+    spec = importlib.util.spec_from_file_location("dsv3_example", path)
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    mod.main(batch=4, ctx_len=128, steps=2, layers=1)
