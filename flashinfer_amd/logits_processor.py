@@ -46,6 +46,41 @@ class LegalizationError(ValueError):
     pass
 
 
+class Op:
+    """Typed low-level operator (reference op.py Op:23): consumes and
+    produces :class:`TaggedTensor` with declared IN/OUT types."""
+
+    IN: Optional[TensorType] = None
+    OUT: Optional[TensorType] = None
+
+    def __init__(self):
+        if self.IN is None or self.OUT is None:
+            raise ValueError(
+                f"Operator {type(self).__name__} must define IN and OUT "
+                "tensor types")
+
+    def __call__(self, tensor: TaggedTensor, **kwargs) -> TaggedTensor:
+        raise NotImplementedError
+
+    def _validate_input_type(self, tensor: TaggedTensor) -> TensorType:
+        if tensor.type != self.IN:
+            raise ValueError(
+                f"Operator {type(self).__name__} cannot accept input type "
+                f"{tensor.type}. Expected: {self.IN}")
+        return self.OUT
+
+    def __repr__(self):
+        return f"{type(self).__name__}({self.IN} -> {self.OUT})"
+
+
+class ParameterizedOp(Op):
+    """Op with bound default parameters (reference op.py role)."""
+
+    def __init__(self, **params):
+        super().__init__()
+        self.params = params
+
+
 class LogitsProcessor:
     """One pipeline stage. ``in_types`` lists accepted input types;
     ``out_type(in_type)`` gives the produced type."""
