@@ -99,17 +99,24 @@ __global__ __launch_bounds__(
   // reduced logit instead — 1 op per head per step). With a quantized KV
   // cache (TKV != T, e.g. fp8 e4m3) q is kept as f32 and the dot runs on
   // converted values.
-  vec_t<T, VPL> qreg[GROUP];
-  float qf32[kSameT ? 1 : GROUP][kSameT ? 1 : VPL];
-  const T* qbase = (const T*)p.q + (int64_t)req * p.q_stride_n;
+  // GROUP>=8 narrow path: keeping GROUP q fragments resident is what pushes
+  // the instantiation into loop-carried scratch spill (measured 104 B/lane,
+  // dominating tiny-batch latency) — re-read q from L1 per step instead.
+  constexpr bool kQReg = GROUP < 8 || decode_traits<T, TKV, HEAD_DIM, GROUP>::wide;
+  constexpr int QR = kQReg ? GROUP : 1;
+  vec_t<T, VPL> qreg[QR];
+  float qf32[kSameT ? 1 : QR][kSameT ? 1 : VPL];
+  const T* qbase = (const T*)p.q + (int64_t)req * p.q_stride_n +
+                   (int64_t)(kv_head * GROUP) * p.q_stride_h + dcol;
   const float scale = p.sm_scale;
+  if constexpr (kQReg) {
 #pragma unroll
-  for (int g = 0; g < GROUP; ++g) {
-    int qh = kv_head * GROUP + g;
-    qreg[g].load(qbase + (int64_t)qh * p.q_stride_h + dcol);
-    if constexpr (!kSameT) {
+    for (int g = 0; g < GROUP; ++g) {
+      qreg[g].load(qbase + (int64_t)g * p.q_stride_h);
+      if constexpr (!kSameT) {
 #pragma unroll
-      for (int j = 0; j < VPL; ++j) qf32[g][j] = qreg[g].get(j);
+        for (int j = 0; j < VPL; ++j) qf32[g][j] = qreg[g].get(j);
+      }
     }
   }
 
@@ -125,14 +132,6 @@ __global__ __launch_bounds__(
     for (int j = 0; j < VPL; ++j) o_acc[g][j] = 0.f;
   }
   const float scale2 = scale * 1.4426950408889634f;
-  float alibi_slope2[GROUP];
-#pragma unroll
-  for (int g = 0; g < GROUP; ++g) {
-    alibi_slope2[g] = p.alibi ? __builtin_exp2f(-8.f * (kv_head * GROUP + g + 1) /
-                                                p.num_qo_heads) *
-                                    1.4426950408889634f
-                              : 0.f;
-  }
 
   const int32_t* page_ids = p.kv_indices + p.kv_indptr[req];
   const TKV* kbase = (const TKV*)p.k_data;
@@ -165,11 +164,25 @@ __global__ __launch_bounds__(
     for (int g = 0; g < GROUP; ++g) {
       float s;
       if constexpr (kSameT) {
-        s = qk_dot<T, VPL>(qreg[g].data, kv_cur.data, 0.f);
+        if constexpr (kQReg) {
+          s = qk_dot<T, VPL>(qreg[g].data, kv_cur.data, 0.f);
+        } else {
+          vec_t<T, VPL> qv;  // L1-resident: same 16 B every iteration
+          qv.load(qbase + (int64_t)g * p.q_stride_h);
+          s = qk_dot<T, VPL>(qv.data, kv_cur.data, 0.f);
+        }
       } else {
-        s = 0.f;
+        if constexpr (kQReg) {
+          s = 0.f;
 #pragma unroll
-        for (int j = 0; j < VPL; ++j) s += qf32[g][j] * kv_cur.get(j);
+          for (int j = 0; j < VPL; ++j) s += qf32[g][j] * kv_cur.get(j);
+        } else {
+          vec_t<T, VPL> qv;
+          qv.load(qbase + (int64_t)g * p.q_stride_h);
+          s = 0.f;
+#pragma unroll
+          for (int j = 0; j < VPL; ++j) s += qv.get(j) * kv_cur.get(j);
+        }
       }
       // reduce across the LPT lanes of this token
 #pragma unroll
@@ -181,7 +194,14 @@ __global__ __launch_bounds__(
       } else {
         s2 = s * scale2;
       }
-      if (p.alibi) s2 -= alibi_slope2[g] * (float)(kv_len - 1 - (pos0 + tsub));
+      if (p.alibi) {
+        // recomputed on demand: keeping GROUP slopes resident costs the
+        // registers that push the GROUP-8/16 instantiations into scratch
+        float slope2 = __builtin_exp2f(-8.f * (kv_head * GROUP + g + 1) /
+                                       p.num_qo_heads) *
+                       1.4426950408889634f;
+        s2 -= slope2 * (float)(kv_len - 1 - (pos0 + tsub));
+      }
       if (valid) {
         if (s2 <= m_run[g]) {
           float pv = __builtin_exp2f(s2 - m_run[g]);
