@@ -129,6 +129,35 @@ class BatchDecodeWithPagedKVCacheWrapper:
             raise NotImplementedError(
                 "pos_encoding_mode must be NONE or ALIBI; apply RoPE beforehand"
             )
+        # tensor-core decode (reference decode.py:1697 "use_tensor_cores"
+        # prefill-module reuse): at GQA group >= 8 the vector kernel is
+        # VALU-bound (8 dot/softmax chains per K read) while the MFMA prefill
+        # kernel amortizes the whole group into one 32x32 tile row — measured
+        # 1.4-1.9x on MI355X (profiles/README r01). Auto-routed; fp8-KV keeps
+        # the vector kernel's native-fp8 loads.
+        group = num_qo_heads // max(1, num_kv_heads)
+        kv_dt = kv_data_type or q_data_type
+        self._tc = (self._use_tensor_cores or group >= 8) and kv_dt == q_data_type
+        if self._tc:
+            from .prefill import BatchPrefillWithPagedKVCacheWrapper
+
+            if getattr(self, "_tc_wrapper", None) is None:
+                self._tc_wrapper = BatchPrefillWithPagedKVCacheWrapper(
+                    self._float_workspace_buffer, self._kv_layout)
+            batch = indptr.shape[0] - 1
+            qo_indptr = torch.arange(0, batch + 1, dtype=torch.int32)
+            self._tc_wrapper.plan(
+                qo_indptr, indptr, indices, last_page_len, num_qo_heads,
+                num_kv_heads, head_dim, page_size, causal=False,
+                pos_encoding_mode=pos_encoding_mode, sm_scale=sm_scale,
+                window_left=window_left, logits_soft_cap=logits_soft_cap,
+                q_data_type=q_data_type, kv_data_type=kv_data_type,
+                non_blocking=non_blocking)
+            self._plan_info = dict(
+                batch=batch, num_qo_heads=num_qo_heads,
+                sm_scale=sm_scale if sm_scale is not None
+                else default_sm_scale(head_dim))
+            return
         batch = indptr.shape[0] - 1
         indptr_h = indptr.to("cpu", torch.int64)
         lp_h = last_page_len.to("cpu", torch.int64)
@@ -202,6 +231,24 @@ class BatchDecodeWithPagedKVCacheWrapper:
             raise RuntimeError("must call plan() before run()")
         if q.dim() == 2:
             q = q.unsqueeze(1)  # [B, D] MQA convenience
+        if getattr(self, "_tc", False):
+            need_lse = return_lse or sinks is not None
+            # q_scale and k_scale both act on the logits: fold into the
+            # inner wrapper's k_scale (it multiplies sm_scale at run time)
+            ks = None
+            if q_scale is not None or k_scale is not None:
+                ks = (q_scale or 1.0) * (k_scale or 1.0)
+            res = self._tc_wrapper.run(
+                q, paged_kv_cache, k_scale=ks, v_scale=v_scale, out=out,
+                lse=lse, return_lse=need_lse)
+            out, lse = res if need_lse else (res, None)
+            if sinks is not None:
+                import math as _m
+
+                w = 1.0 / (1.0 + torch.exp2(
+                    sinks.float()[None, :] * _m.log2(_m.e) - lse))
+                out = (out.float() * w[..., None]).to(out.dtype)
+            return (out, lse) if return_lse else out
         k_cache, v_cache = unpack_paged_kv_cache(paged_kv_cache, self._kv_layout)
         sm_scale = pi["sm_scale"]
         if q_scale is not None:
