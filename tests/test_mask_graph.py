@@ -182,3 +182,35 @@ def test_mla_hipgraph_capture():
     g.replay()
     torch.cuda.synchronize()
     torch.testing.assert_close(out.float(), ref.float())
+
+
+def test_tensor_core_decode_graph_capture():
+    """GQA-8 decode (auto tensor-core route) captures into a hipGraph."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(4)
+    Hq, Hkv, D, page = 64, 8, 128, 16
+    bs, pages_per = 8, 8
+    npages = bs * pages_per
+    kc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    indptr = torch.arange(0, npages + 1, pages_per, dtype=torch.int32,
+                          device="cuda")
+    indices = torch.arange(npages, dtype=torch.int32, device="cuda")
+    lpl = torch.full((bs,), page, dtype=torch.int32, device="cuda")
+    q = torch.randn(bs, Hq, D, dtype=torch.bfloat16, device="cuda")
+    out = torch.empty_like(q)
+    ws = torch.empty(64 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(indptr, indices, lpl, Hq, Hkv, D, page, q_data_type=torch.bfloat16)
+    assert w._tc  # the GQA-8 auto route must be active
+    w.run(q, (kc, vc), out=out)
+    torch.cuda.synchronize()
+    ref = out.clone()
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        w.run(q, (kc, vc), out=out)
+    out.zero_()
+    g.replay()
+    torch.cuda.synchronize()
+    torch.testing.assert_close(out.float(), ref.float())
