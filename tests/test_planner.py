@@ -63,3 +63,72 @@ def test_fastdiv_magic():
     for d in [1, 2, 3, 5, 7, 15, 16, 17, 64, 100, 128, 255, 257, 32768, 65535]:
         got = ext.debug_fastdiv(d, vals)
         assert got == [v // d for v in vals], f"divisor {d}"
+
+
+# ---- property-based invariants (hypothesis) ----
+try:
+    from hypothesis import given, settings, strategies as st
+
+    HAVE_HYP = True
+except ImportError:  # pragma: no cover
+    HAVE_HYP = False
+
+
+if HAVE_HYP:
+    @given(
+        kv_lens=st.lists(st.integers(min_value=0, max_value=100000),
+                         min_size=1, max_size=64),
+        num_kv_heads=st.sampled_from([1, 2, 4, 8, 16]),
+        page_size=st.sampled_from([1, 16, 32, 64]),
+    )
+    @settings(max_examples=200, deadline=None)
+    def test_plan_chunks_invariants(kv_lens, num_kv_heads, page_size):
+        from flashinfer_amd.decode import _plan_chunks
+
+        chunk, work_req, work_chunk, merge_indptr = _plan_chunks(
+            kv_lens, num_kv_heads, page_size)
+        assert chunk % page_size == 0 and chunk >= page_size
+        assert len(work_req) == len(work_chunk) == merge_indptr[-1]
+        assert merge_indptr[0] == 0
+        # every request gets at least one item; chunks tile the kv length
+        for b, L in enumerate(kv_lens):
+            items = [c for r, c in zip(work_req, work_chunk) if r == b]
+            assert items == list(range(len(items)))
+            assert len(items) == max(1, -(-L // chunk))
+            assert merge_indptr[b + 1] - merge_indptr[b] == len(items)
+
+    @given(
+        qo_lens=st.lists(st.integers(min_value=0, max_value=5000),
+                         min_size=1, max_size=32),
+        group=st.sampled_from([1, 2, 4, 8]),
+        causal=st.booleans(),
+    )
+    @settings(max_examples=200, deadline=None)
+    def test_plan_tiles_invariants(qo_lens, group, causal):
+        from flashinfer_amd.prefill import _plan_tiles
+
+        cta_q, tile_req, tile_qstart = _plan_tiles(qo_lens, group, causal)
+        assert cta_q in (128, 256)
+        assert len(tile_req) == len(tile_qstart)
+        # tiles exactly cover every request's packed rows
+        from collections import defaultdict
+
+        starts = defaultdict(list)
+        for r, s in zip(tile_req, tile_qstart):
+            starts[r].append(s)
+        for b, L in enumerate(qo_lens):
+            pk = max(L * group, 1)
+            expect = list(range(0, pk, cta_q))
+            assert sorted(starts[b]) == expect
+
+    @given(n=st.integers(min_value=1, max_value=2**31 - 1),
+           d=st.integers(min_value=1, max_value=2**31 - 1))
+    @settings(max_examples=300, deadline=None)
+    def test_fastdiv_magic_matches_python(n, d):
+        import flashinfer_amd as fi
+        from flashinfer_amd._lib import has_ext, get_ext
+
+        if not has_ext():
+            return
+        res = get_ext().debug_fastdiv(d, [n])
+        assert res[0] == n // d
