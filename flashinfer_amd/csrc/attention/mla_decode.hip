@@ -276,13 +276,17 @@ __global__ __launch_bounds__(NTH, 1) void mla_decode_kernel(MlaParams p) {
   float inv_d = d_full > 0.f ? 1.f / d_full : 0.f;
   if (row_valid) {
     int64_t pos = (int64_t)(qo_begin + qpos_u) * H + head_u;
-    float* vout = p.tmp_v + (pos * p.max_chunks + chunk) * D_CKV;
+    // bf16 partials: the split-KV tmp round-trip (~66 MB at the baseline
+    // bs=16 shape) is the latency-bound kernel's dominant traffic — the
+    // chunk partials are already normalized, so bf16 storage costs ~1e-3
+    // relative error while halving the tmp bytes
+    T* vout = (T*)p.tmp_v + (pos * p.max_chunks + chunk) * D_CKV;
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         int d = db * 128 + i * 32 + mfma32_cd_row(r, lane);
-        vout[d] = acc_o[i][r] * inv_d;
+        vout[d] = from_f32<T>(acc_o[i][r] * inv_d);
       }
     }
     if (db == 0 && (lane >> 5) == 0) {

@@ -353,7 +353,7 @@ void batch_decode_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   p.work_chunk = work_chunk.data_ptr<int32_t>();
   p.n_items = work_req.size(0);
   p.chunk_size = (int)chunk_size;
-  p.tmp_v = tmp_v.data_ptr<float>();
+  p.tmp_v = (float*)tmp_v.data_ptr();  // stored in the q dtype (see mla.py)
   p.tmp_s = tmp_s.data_ptr<float>();
   p.q_stride_n = q.stride(0);
   p.q_stride_h = q.stride(1);
@@ -557,7 +557,7 @@ void mla_run(at::Tensor q_nope, at::Tensor q_pe, at::Tensor ckv_cache,
   p.chunk_size = (int)chunk_size;
   p.max_chunks = (int)max_chunks;
   p.num_heads = q_nope.size(1);
-  p.tmp_v = tmp_v.data_ptr<float>();
+  p.tmp_v = (float*)tmp_v.data_ptr();  // stored in the q dtype (see mla.py)
   p.tmp_s = tmp_s.data_ptr<float>();
   p.sm_scale = (float)sm_scale;
   p.causal = causal ? 1 : 0;

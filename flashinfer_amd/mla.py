@@ -88,8 +88,10 @@ class BatchMLAPagedAttentionWrapper:
         self._kv_last_page_len_d = lpl.to(self.device, non_blocking=True)
 
         alloc = WorkspaceAllocator(self._float_workspace_buffer)
+        # partials stored in the q dtype (bf16/f16): halves the split-KV
+        # round-trip traffic that bounds the small-batch kernel
         self._tmp_v = alloc.alloc(
-            total_rows * max_chunks * 512 * 4, torch.float32,
+            total_rows * max_chunks * 512 * 2, q_data_type,
             (total_rows * max_chunks, 1, 512),
         )
         self._tmp_s = alloc.alloc(
