@@ -108,3 +108,25 @@ def test_batch_decode(page_size, kv_layout, Hq, Hkv, D):
         logits = torch.einsum("hd,lhd->hl", q[b].float(), kf) / math.sqrt(D)
         ref_lse = torch.logsumexp(logits, -1) / math.log(2)
         torch.testing.assert_close(lse[b], ref_lse, atol=2e-2, rtol=2e-2)
+
+
+def test_decode_with_empty_request():
+    """A request with zero KV pages must produce zero output, not NaN."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    Hq, Hkv, D, page = 8, 2, 128, 16
+    # req0: 2 pages, req1: EMPTY, req2: 1 page
+    indptr = torch.tensor([0, 2, 2, 3], dtype=torch.int32, device="cuda")
+    indices = torch.arange(3, dtype=torch.int32, device="cuda")
+    lpl = torch.tensor([page, 0, 5], dtype=torch.int32, device="cuda")
+    kc = torch.randn(3, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(3, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    q = torch.randn(3, Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(16 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(indptr, indices, lpl, Hq, Hkv, D, page, q_data_type=torch.bfloat16)
+    out, lse = w.run(q, (kc, vc), return_lse=True)
+    assert out[1].abs().max() == 0           # empty request -> zeros
+    assert torch.isinf(lse[1]).all() and (lse[1] < 0).all()
+    assert out[0].isfinite().all() and out[2].isfinite().all()
