@@ -34,6 +34,7 @@ namespace fi {
 // the target batch sizes (PMC r01) — at the cost of ~2x the accumulator
 // registers, so its occupancy target drops to 2-3 waves/SIMD.
 typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2_t;
+typedef __attribute__((ext_vector_type(2))) _Float16 f16x2_t;
 
 template <typename T, int N>
 __device__ __forceinline__ float qk_dot(const void* qv, const void* kv, float acc) {
@@ -43,6 +44,13 @@ __device__ __forceinline__ float qk_dot(const void* qv, const void* kv, float ac
 #pragma unroll
     for (int i = 0; i < N / 2; ++i)
       acc = __builtin_amdgcn_fdot2_f32_bf16(q2[i], k2[i], acc, false);
+    return acc;
+  } else if constexpr (__is_same(T, fp16)) {
+    const f16x2_t* q2 = (const f16x2_t*)qv;
+    const f16x2_t* k2 = (const f16x2_t*)kv;
+#pragma unroll
+    for (int i = 0; i < N / 2; ++i)
+      acc = __builtin_amdgcn_fdot2(q2[i], k2[i], acc, false);
     return acc;
   } else {
     const T* q = (const T*)qv;
@@ -55,10 +63,10 @@ __device__ __forceinline__ float qk_dot(const void* qv, const void* kv, float ac
 
 template <typename T, typename TKV, int HEAD_DIM, int GROUP>
 struct decode_traits {
-  // the wide path needs the packed bf16 dot2 pipeline; f16/f32/fp8-KV paths
-  // carry f32 temporaries that would spill at its register budget
-  static constexpr bool wide =
-      __is_same(T, bf16) && __is_same(TKV, bf16) && HEAD_DIM >= 128 && GROUP <= 4;
+  // the wide path needs a packed dot2 pipeline (bf16 or f16); f32/fp8-KV
+  // paths carry f32 temporaries that would spill at its register budget
+  static constexpr bool wide = (__is_same(T, bf16) || __is_same(T, fp16)) &&
+                               __is_same(T, TKV) && HEAD_DIM >= 128 && GROUP <= 4;
   static constexpr int vpl = wide ? 16 : 8;
   // narrow path: the GROUP-8/16 accumulators (o_acc[GROUP][8]) blow the
   // 128-VGPR budget of 4 waves/SIMD — drop the target instead of spilling
