@@ -155,6 +155,9 @@ if __name__ == "__main__":
         bench_decode()
         bench_decode(bs=16, kv=1024)
         bench_decode(bs=128, kv=4096)
+        # GQA-8 (Llama-70B class): auto tensor-core route
+        bench_decode(bs=256, kv=8192, Hq=64, Hkv=8)
+        bench_decode(bs=16, kv=1024, Hq=64, Hkv=8)
     if which in ("all", "gemm"):
         bench_gemm(4096)
         bench_gemm(8192)
