@@ -160,6 +160,7 @@ trace_apply._enable_apply_from_env()
 from . import grouped_mm
 from .grouped_mm import grouped_mm_bf16, grouped_mm_fp8
 from . import dsv3_ops
+from . import diffusion_ops
 from .mhc import mhc_post, mhc_pre_big_fuse, mhc_pre_big_fuse_with_prenorm
 from .concat_ops import concat_mla_k
 from . import jit
