@@ -130,3 +130,35 @@ def test_decode_with_empty_request():
     assert out[1].abs().max() == 0           # empty request -> zeros
     assert torch.isinf(lse[1]).all() and (lse[1] < 0).all()
     assert out[0].isfinite().all() and out[2].isfinite().all()
+
+
+@pytest.mark.parametrize("Hq,Hkv", [(8, 8), (16, 2)])
+def test_decode_head_dim_256(Hq, Hkv):
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    D, page, bs, kv = 256, 16, 4, 333
+    pages_per = (kv + page - 1) // page
+    npages = bs * pages_per
+    indptr = torch.arange(0, npages + 1, pages_per, dtype=torch.int32,
+                          device="cuda")
+    indices = torch.arange(npages, dtype=torch.int32, device="cuda")
+    lpl = torch.full((bs,), (kv - 1) % page + 1, dtype=torch.int32,
+                     device="cuda")
+    kc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    q = torch.randn(bs, Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(64 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(indptr, indices, lpl, Hq, Hkv, D, page, q_data_type=torch.bfloat16)
+    out = w.run(q, (kc, vc))
+    import math
+    G = Hq // Hkv
+    for b in range(bs):
+        kb = kc.view(-1, Hkv, D)[b * pages_per * page:][:kv].float()
+        vb = vc.view(-1, Hkv, D)[b * pages_per * page:][:kv].float()
+        logits = torch.einsum("hd,khd->hk", q[b].float(),
+                              kb.repeat_interleave(G, 1)) / math.sqrt(D)
+        ref = torch.einsum("hk,khd->hd", torch.softmax(logits, -1),
+                           vb.repeat_interleave(G, 1))
+        torch.testing.assert_close(out[b].float(), ref, atol=3e-2, rtol=3e-2)
