@@ -103,6 +103,11 @@ hipError_t fi_one_shot_ar(int dtype, fi::ARParams* p, void* out, int64_t numel,
 hipError_t fi_one_shot_ar_rmsnorm(int dtype, fi::ARParams* p, void* out,
                                   void* residual, const void* weight, int rows,
                                   int d, float eps, hipStream_t stream);
+hipError_t fi_qk_rope(int dtype, const void* qkv, const void* qw, const void* kw,
+                      const float* cos_t, const float* sin_t, void* q_out,
+                      void* k_out, void* v_out, int64_t tokens, int S, int Hq,
+                      int Hk, int Hv, int D, float eps, float attn_factor,
+                      int qk_norm, hipStream_t stream);
 hipError_t fi_mhc_post(int dtype, const void* x, const void* residual,
                        const float* post_mix, const float* comb_mix, void* out,
                        int64_t tokens, int H, hipStream_t stream);
@@ -788,6 +793,22 @@ void one_shot_all_reduce_rmsnorm(at::Tensor out, c10::optional<at::Tensor> resid
             "fi_one_shot_ar_rmsnorm");
 }
 
+void qk_rope(at::Tensor qkv, at::Tensor qw, at::Tensor kw, at::Tensor cos_t,
+             at::Tensor sin_t, at::Tensor q_out, at::Tensor k_out,
+             at::Tensor v_out, int64_t S, int64_t Hq, int64_t Hk, int64_t Hv,
+             int64_t D, double eps, double attn_factor, bool qk_norm) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous());
+  TORCH_CHECK(cos_t.scalar_type() == at::kFloat && sin_t.scalar_type() == at::kFloat);
+  int64_t tokens = qkv.numel() / ((Hq + Hk + Hv) * D);
+  check_hip(fi_qk_rope(dtype_code(qkv), qkv.data_ptr(), qw.data_ptr(),
+                       kw.data_ptr(), cos_t.data_ptr<float>(),
+                       sin_t.data_ptr<float>(), q_out.data_ptr(), k_out.data_ptr(),
+                       v_out.data_ptr(), tokens, (int)S, (int)Hq, (int)Hk, (int)Hv,
+                       (int)D, (float)eps, (float)attn_factor, qk_norm,
+                       cur_stream(qkv)),
+            "fi_qk_rope");
+}
+
 void mhc_post(at::Tensor x, at::Tensor residual, at::Tensor post_mix,
               at::Tensor comb_mix, at::Tensor out) {
   TORCH_CHECK(x.is_cuda() && residual.dim() == 3 && residual.size(1) == 4);
@@ -916,6 +937,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gdn_chunk", &gdn_chunk);
   m.def("ssd_scan", &ssd_scan);
   m.def("mhc_post", &mhc_post);
+  m.def("qk_rope", &qk_rope);
   m.def("mhc_pre", &mhc_pre);
   m.def("concat_mla_k", &concat_mla_k);
   m.def("ipc_memcpy_to", &ipc_memcpy_to);

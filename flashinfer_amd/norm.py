@@ -226,6 +226,25 @@ def fused_qk_rmsnorm_rope(
     B, S, _ = qkv.shape
     D = head_dim
     Hq, Hk, Hv = num_heads_q, num_heads_k, num_heads_v
+    if (D == 128 and qkv.dtype in (torch.bfloat16, torch.float16)
+            and not output_fp8):
+        # fused CDNA4 kernel (csrc/qk_rope.hip): one wave per (token, head)
+        from ._lib import get_ext
+
+        cos, sin = _rope3d_cache(ppf, pph, ppw, D, base, str(qkv.device))
+        q = q_out if q_out is not None else torch.empty(
+            B, S, Hq, D, dtype=qkv.dtype, device=qkv.device)
+        k = k_out if k_out is not None else torch.empty(
+            B, S, Hk, D, dtype=qkv.dtype, device=qkv.device)
+        v = v_out if v_out is not None else torch.empty(
+            B, S, Hv, D, dtype=qkv.dtype, device=qkv.device)
+        get_ext().qk_rope(qkv.contiguous(), q_weight.reshape(-1)[-D:].contiguous(),
+                          k_weight.reshape(-1)[-D:].contiguous(),
+                          cos.contiguous(), sin.contiguous(),
+                          q.view(B * S, Hq, D), k.view(B * S, Hk, D),
+                          v.view(B * S, Hv, D), S, Hq, Hk, Hv, D, eps,
+                          attention_factor, is_qk_norm)
+        return q, k, v
     q, k, v = qkv.split([Hq * D, Hk * D, Hv * D], dim=-1)
     q = q.reshape(B, S, Hq, D)
     k = k.reshape(B, S, Hk, D)
