@@ -59,7 +59,13 @@ class CustomAllReduce:
         get_ext().ipc_memcpy_to(self.bufs[self.rank] + off, x.contiguous())
         return off
 
-    def _check_error(self):
+    def check_errors(self):
+        """On-demand timeout check (synchronizes). The per-call path never
+        syncs — the kernel's bounded spin writes the device flag and aborts,
+        so a dead peer corrupts one output but cannot hang the GPU; callers
+        poll this at natural sync points (end of step / before a checkpoint).
+        Advisor r01: a per-call .item() serialized every TP collective and
+        broke hipGraph capture."""
         if int(self.error_flag.item()) != 0:
             raise RuntimeError(
                 "one-shot allreduce spin timeout: a peer never arrived")
@@ -72,7 +78,6 @@ class CustomAllReduce:
         get_ext().one_shot_all_reduce(out.view(-1), self.bufs, self.rank,
                                       self.seq, self.error_flag,
                                       self.spin_limit, off)
-        self._check_error()
         return out
 
     def all_reduce_rmsnorm(self, x: torch.Tensor, residual: Optional[torch.Tensor],
@@ -86,7 +91,6 @@ class CustomAllReduce:
         get_ext().one_shot_all_reduce_rmsnorm(
             out, residual, weight, self.bufs, self.rank, self.seq,
             self.error_flag, self.spin_limit, off, eps)
-        self._check_error()
         return out
 
     def close(self):

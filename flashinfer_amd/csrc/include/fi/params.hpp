@@ -46,6 +46,11 @@ struct DecodeParams {
   float logits_soft_cap;  // 0 = disabled
   int window_left;        // -1 = disabled
   int alibi;              // ALiBi position bias
+  // fused whole-request path (no split/merge): direct output in q dtype
+  void* o;                // [batch, Hq, D]
+  float* lse;             // [batch, Hq] base-2, optional
+  int64_t o_stride_n, o_stride_h;
+  int batch;
 };
 
 struct PrefillParams {

@@ -342,7 +342,18 @@ __global__ void chain_spec_kernel(SpecParams sp) {
     }
   }
   if (threadIdx.x == 0) {
-    atomicAdd(sp.accepted_num + b, pos);
+    // accepted_num keeps running the accept test past the first rejection
+    // (reference sampling.cuh:1906-1920 contract: accepted >= emitted);
+    // emitted_num counts only the prefix actually emitted.
+    int accepted = pos;
+    for (int i = pos; i < sp.n; ++i) {
+      int t = sp.draft_ids[(uint64_t)b * sp.n + i];
+      float pd = sp.draft_probs[((uint64_t)b * sp.n + i) * V + t];
+      float pt = sp.target_probs[((uint64_t)b * (sp.n + 1) + i) * V + t];
+      float u = sp.uniforms[(uint64_t)b * (sp.n + 1) + i];
+      if (u * pd <= pt) ++accepted;
+    }
+    atomicAdd(sp.accepted_num + b, accepted);
     atomicAdd(sp.emitted_num + b, pos);
   }
   // sample one more token: from residual at `pos` (rejected) or from the

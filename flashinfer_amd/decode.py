@@ -30,6 +30,12 @@ from .utils import (
 
 _TARGET_BLOCKS = 8192  # 1-wave blocks: ~32 per CU keeps enough loads in flight
 
+# Whole-request fused kernel (one workgroup per (req, kv_head), in-LDS wave
+# merge, single launch): wins while the per-wave KV slice stays latency-sized;
+# past this the split path's finer-grain work items win back. Measured
+# crossover on MI355X (profiles/README r02 fused-decode entry).
+_FUSED_MAX_KV = 2048
+
 
 def _plan_chunks(kv_lens, num_kv_heads: int, page_size: int,
                  fixed_split_size=None, disable_split_kv=False):
@@ -72,7 +78,7 @@ class BatchDecodeWithPagedKVCacheWrapper:
         float_workspace_buffer: torch.Tensor,
         kv_layout: str = "NHD",
         use_cuda_graph: bool = False,
-        use_tensor_cores: bool = False,
+        use_tensor_cores: Optional[bool] = None,
         paged_kv_indptr_buffer: Optional[torch.Tensor] = None,
         paged_kv_indices_buffer: Optional[torch.Tensor] = None,
         paged_kv_last_page_len_buffer: Optional[torch.Tensor] = None,
@@ -129,15 +135,46 @@ class BatchDecodeWithPagedKVCacheWrapper:
             raise NotImplementedError(
                 "pos_encoding_mode must be NONE or ALIBI; apply RoPE beforehand"
             )
-        # tensor-core decode (reference decode.py:1697 "use_tensor_cores"
-        # prefill-module reuse): at GQA group >= 8 the vector kernel is
-        # VALU-bound (8 dot/softmax chains per K read) while the MFMA prefill
-        # kernel amortizes the whole group into one 32x32 tile row — measured
-        # 1.4-1.9x on MI355X (profiles/README r01). Auto-routed; fp8-KV keeps
-        # the vector kernel's native-fp8 loads.
+        # Kernel routing (3 decode shapes, auto-picked unless the caller pins
+        # one — use_tensor_cores: None=auto, True=force prefill-MFMA path,
+        # False=never (advisor r01: explicit False must be respected)):
+        #   1. fused whole-request kernel — short KV: one workgroup per
+        #      (req, kv_head), in-LDS merge, single launch (latency shape);
+        #   2. tensor-core (prefill-MFMA) path — GQA group >= 8 at long KV:
+        #      the whole group costs one 32x32 MFMA tile row instead of 8
+        #      VALU dot chains (reference decode.py:1697 design, measured
+        #      1.4-1.9x on MI355X, profiles/README r01);
+        #   3. split vector kernel + LSE merge — everything else.
         group = num_qo_heads // max(1, num_kv_heads)
         kv_dt = kv_data_type or q_data_type
-        self._tc = (self._use_tensor_cores or group >= 8) and kv_dt == q_data_type
+        batch = indptr.shape[0] - 1
+        indptr_h = indptr.to("cpu", torch.int64)
+        lp_h = last_page_len.to("cpu", torch.int64)
+        np_ = indptr_h[1:] - indptr_h[:-1]
+        kv_lens = (
+            torch.clamp(np_ - 1, min=0) * page_size
+            + torch.where(np_ > 0, lp_h, torch.zeros_like(lp_h))
+        ).tolist()
+        max_len = max(kv_lens) if kv_lens else 0
+
+        fused_ok = (
+            head_dim in (64, 128, 256)
+            and group in (1, 2, 4, 8)
+            and not (head_dim == 256 and group == 8)
+            and max_len <= _FUSED_MAX_KV
+            and fixed_split_size is None
+            and self._use_tensor_cores is not True
+        )
+        self._tc = (
+            not fused_ok
+            and kv_dt == q_data_type
+            and (
+                self._use_tensor_cores is True
+                or (self._use_tensor_cores is None and group >= 8
+                    and fixed_split_size is None and not disable_split_kv)
+            )
+        )
+        self._fused = fused_ok and not self._tc
         if self._tc:
             from .prefill import BatchPrefillWithPagedKVCacheWrapper
 
@@ -158,14 +195,31 @@ class BatchDecodeWithPagedKVCacheWrapper:
                 sm_scale=sm_scale if sm_scale is not None
                 else default_sm_scale(head_dim))
             return
-        batch = indptr.shape[0] - 1
-        indptr_h = indptr.to("cpu", torch.int64)
-        lp_h = last_page_len.to("cpu", torch.int64)
-        np_ = indptr_h[1:] - indptr_h[:-1]
-        kv_lens = (
-            torch.clamp(np_ - 1, min=0) * page_size
-            + torch.where(np_ > 0, lp_h, torch.zeros_like(lp_h))
-        ).tolist()
+
+        if self._fused:
+            dev = self.device
+
+            def _to_dev_f(x, buf):
+                xt = x.to(dev, torch.int32, non_blocking=non_blocking)
+                if buf is not None:
+                    buf[: xt.numel()].copy_(xt)
+                    return buf[: xt.numel()]
+                return xt
+
+            self._indptr_d = _to_dev_f(indptr, self._fixed_indptr_buf)
+            self._indices_d = _to_dev_f(indices, self._fixed_indices_buf)
+            self._last_page_len_d = _to_dev_f(
+                last_page_len, self._fixed_last_page_len_buf)
+            self._plan_info = dict(
+                batch=batch, num_qo_heads=num_qo_heads,
+                num_kv_heads=num_kv_heads, head_dim=head_dim,
+                page_size=page_size, window_left=window_left,
+                logits_soft_cap=float(logits_soft_cap or 0.0),
+                sm_scale=sm_scale if sm_scale is not None
+                else default_sm_scale(head_dim),
+                q_data_type=q_data_type, alibi=pos_encoding_mode == "ALIBI",
+            )
+            return
 
         chunk, work_req, work_chunk, merge_indptr = _plan_chunks(
             kv_lens, num_kv_heads, page_size, fixed_split_size, disable_split_kv
@@ -255,6 +309,28 @@ class BatchDecodeWithPagedKVCacheWrapper:
             sm_scale *= q_scale
         if k_scale is not None:
             sm_scale *= k_scale
+        if getattr(self, "_fused", False):
+            if out is None:
+                out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+            need_lse = return_lse or sinks is not None
+            if need_lse and lse is None:
+                lse = torch.empty((pi["batch"], pi["num_qo_heads"]),
+                                  dtype=torch.float32, device=q.device)
+            get_ext().batch_decode_fused_run(
+                q, k_cache, v_cache,
+                self._indices_d, self._indptr_d, self._last_page_len_d,
+                layout_code(self._kv_layout), out, lse if need_lse else None,
+                sm_scale, pi["logits_soft_cap"], pi["window_left"], pi["alibi"],
+            )
+            if sinks is not None:
+                import math as _m
+
+                w = 1.0 / (1.0 + torch.exp2(
+                    sinks.float()[None, :] * _m.log2(_m.e) - lse))
+                out = (out.float() * w[..., None]).to(out.dtype)
+            if v_scale is not None:
+                out = out * v_scale
+            return (out, lse) if return_lse else out
         get_ext().batch_decode_run(
             q, k_cache, v_cache,
             self._indices_d, self._indptr_d, self._last_page_len_d,

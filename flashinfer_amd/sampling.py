@@ -41,12 +41,16 @@ def softmax(
     temperature: Optional[Union[torch.Tensor, float]] = None,
     enable_pdl: Optional[bool] = None,
 ) -> torch.Tensor:
-    r"""Online safe softmax with temperature scaling."""
-    out = torch.empty_like(logits)
+    r"""Online safe softmax with temperature scaling.
+
+    Returns float32 for any input dtype (reference softmax upcasts; the
+    kernel computes and writes f32 — advisor r01 fix)."""
     if isinstance(temperature, torch.Tensor):
-        logits = logits / temperature[:, None].clamp(min=1e-10)
+        logits = logits.float() / temperature[:, None].clamp(min=1e-10)
         temperature = 1.0
-    get_ext().softmax(logits.contiguous().float(), out,
+    logits = logits.contiguous().float()
+    out = torch.empty_like(logits)
+    get_ext().softmax(logits, out,
                       1.0 if temperature is None else float(temperature))
     return out
 

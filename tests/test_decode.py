@@ -162,3 +162,156 @@ def test_decode_head_dim_256(Hq, Hkv):
         ref = torch.einsum("hk,khd->hd", torch.softmax(logits, -1),
                            vb.repeat_interleave(G, 1))
         torch.testing.assert_close(out[b].float(), ref, atol=3e-2, rtol=3e-2)
+
+
+# ---------------- fused whole-request decode (short-kv shape) ----------------
+
+def _paged(bs, kv_lens, Hkv, D, page, dtype=torch.bfloat16):
+    pages_per = [(L + page - 1) // page for L in kv_lens]
+    indptr = torch.tensor([0] + list(torch.cumsum(torch.tensor(pages_per), 0)),
+                          dtype=torch.int32, device="cuda")
+    npages = int(indptr[-1])
+    indices = torch.randperm(npages, dtype=torch.int32, device="cuda")
+    lpl = torch.tensor([(L - 1) % page + 1 if L else 0 for L in kv_lens],
+                       dtype=torch.int32, device="cuda")
+    kc = torch.randn(npages, page, Hkv, D, dtype=dtype, device="cuda")
+    vc = torch.randn(npages, page, Hkv, D, dtype=dtype, device="cuda")
+    return indptr, indices, lpl, kc, vc
+
+
+def _gather(indptr, indices, kv_lens, kc, vc, b, page):
+    L = kv_lens[b]
+    toks_k, toks_v = [], []
+    for p in range((L + page - 1) // page):
+        pg = int(indices[int(indptr[b]) + p])
+        n = min(page, L - p * page)
+        toks_k.append(kc[pg, :n])
+        toks_v.append(vc[pg, :n])
+    return torch.cat(toks_k, 0), torch.cat(toks_v, 0)
+
+
+@pytest.mark.parametrize("Hq,Hkv,D", [(64, 8, 128), (32, 8, 128), (8, 8, 64),
+                                      (8, 2, 128), (16, 4, 256)])
+def test_decode_fused_routing_and_numerics(Hq, Hkv, D):
+    """Short-kv plans must route to the fused whole-request kernel and match
+    the fp32 reference (covers the BASELINE bs=16/kv=1024 GQA-8 config)."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    page = 16
+    kv_lens = [1024, 1, 777, 1024, 16, 300] + [1024] * 10
+    bs = len(kv_lens)
+    indptr, indices, lpl, kc, vc = _paged(bs, kv_lens, Hkv, D, page)
+    q = torch.randn(bs, Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(64 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(indptr, indices, lpl, Hq, Hkv, D, page, q_data_type=torch.bfloat16)
+    assert w._fused, "short-kv plan must pick the fused kernel"
+    out, lse = w.run(q, (kc, vc), return_lse=True)
+    for b in [0, 1, 2, 4, 5]:
+        kk, vv = _gather(indptr, indices, kv_lens, kc, vc, b, page)
+        ref = sdpa_ref(q[b], kk, vv)
+        torch.testing.assert_close(out[b].float(), ref, atol=3e-2, rtol=3e-2)
+        g = Hq // Hkv
+        kf = kk.float().repeat_interleave(g, dim=1)
+        logits = torch.einsum("hd,lhd->hl", q[b].float(), kf) / math.sqrt(D)
+        ref_lse = torch.logsumexp(logits, -1) / math.log(2)
+        torch.testing.assert_close(lse[b], ref_lse, atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.parametrize("soft_cap,window", [(30.0, -1), (0.0, 127), (30.0, 127)])
+def test_decode_fused_softcap_window(soft_cap, window):
+    import flashinfer_amd as fi
+
+    torch.manual_seed(1)
+    Hq, Hkv, D, page = 32, 8, 128, 16
+    kv_lens = [513, 1024, 64]
+    bs = len(kv_lens)
+    indptr, indices, lpl, kc, vc = _paged(bs, kv_lens, Hkv, D, page)
+    q = torch.randn(bs, Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(64 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(indptr, indices, lpl, Hq, Hkv, D, page,
+           logits_soft_cap=soft_cap or None, window_left=window,
+           q_data_type=torch.bfloat16)
+    assert w._fused
+    out = w.run(q, (kc, vc))
+    for b in range(bs):
+        kk, vv = _gather(indptr, indices, kv_lens, kc, vc, b, page)
+        ref = sdpa_ref(q[b], kk, vv, soft_cap=soft_cap, window_left=window)
+        torch.testing.assert_close(out[b].float(), ref, atol=3e-2, rtol=3e-2)
+
+
+def test_decode_fused_batch_invariant():
+    """The fused kernel's per-wave KV partition depends only on the request's
+    own kv_len -> a request's output is bitwise identical regardless of the
+    other requests in the batch (reference batch-invariant FA2 contract)."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(2)
+    Hq, Hkv, D, page = 32, 8, 128, 16
+    kv_lens = [1024, 300, 2000]
+    indptr, indices, lpl, kc, vc = _paged(3, kv_lens, Hkv, D, page)
+    q = torch.randn(3, Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(64 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(indptr, indices, lpl, Hq, Hkv, D, page, q_data_type=torch.bfloat16)
+    assert w._fused
+    full = w.run(q, (kc, vc))
+    # rerun -> bitwise identical
+    again = w.run(q, (kc, vc))
+    assert torch.equal(full, again)
+    # request 0 alone -> bitwise identical to its row in the full batch
+    w2 = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w2.plan(indptr[:2], indices, lpl[:1], Hq, Hkv, D, page,
+            q_data_type=torch.bfloat16)
+    assert w2._fused
+    solo = w2.run(q[:1], (kc, vc))
+    assert torch.equal(solo[0], full[0])
+
+
+def test_decode_fused_fp8_kv():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(3)
+    Hq, Hkv, D, page = 32, 8, 128, 16
+    kv_lens = [1024, 511]
+    indptr, indices, lpl, kc, vc = _paged(2, kv_lens, Hkv, D, page)
+    kc8 = kc.clamp(-8, 8).to(torch.float8_e4m3fn)
+    vc8 = vc.clamp(-8, 8).to(torch.float8_e4m3fn)
+    q = torch.randn(2, Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(64 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(indptr, indices, lpl, Hq, Hkv, D, page,
+           q_data_type=torch.bfloat16, kv_data_type=torch.float8_e4m3fn)
+    assert w._fused
+    out = w.run(q, (kc8, vc8))
+    for b in range(2):
+        kk, vv = _gather(indptr, indices, kv_lens,
+                         kc8.to(torch.bfloat16), vc8.to(torch.bfloat16), b, page)
+        ref = sdpa_ref(q[b], kk, vv)
+        torch.testing.assert_close(out[b].float(), ref, atol=6e-2, rtol=6e-2)
+
+
+def test_decode_tensor_cores_opt_out():
+    """Explicit use_tensor_cores=False must never route to the prefill-MFMA
+    path, and explicit True must (advisor r01 contract fix)."""
+    import flashinfer_amd as fi
+
+    Hq, Hkv, D, page = 64, 8, 128, 16
+    kv_lens = [8192] * 4  # long kv: auto would pick tc at group 8
+    indptr, indices, lpl, kc, vc = _paged(4, kv_lens, Hkv, D, page)
+    ws = torch.empty(64 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD", use_tensor_cores=False)
+    w.plan(indptr, indices, lpl, Hq, Hkv, D, page, q_data_type=torch.bfloat16)
+    assert not w._tc and not w._fused
+    w2 = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD", use_tensor_cores=True)
+    w2.plan(indptr, indices, lpl, Hq, Hkv, D, page, q_data_type=torch.bfloat16)
+    assert w2._tc
+    w3 = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w3.plan(indptr, indices, lpl, Hq, Hkv, D, page, q_data_type=torch.bfloat16)
+    assert w3._tc  # auto at group 8 / long kv
+    q = torch.randn(4, Hq, D, dtype=torch.bfloat16, device="cuda")
+    o1 = w.run(q, (kc, vc))
+    o2 = w2.run(q, (kc, vc))
+    torch.testing.assert_close(o1.float(), o2.float(), atol=2e-2, rtol=2e-2)
