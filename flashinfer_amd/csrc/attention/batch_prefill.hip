@@ -53,29 +53,33 @@ __device__ __forceinline__ uint32_t swz_row(uint32_t byte_off) {
 
 // CTAQ = packed q rows per workgroup (128 -> 4 waves, 256 -> 8 waves; the
 // planner picks by average packed length — bigger tiles amortize K/V staging
-// over more q rows).
-template <typename T, typename TKV, int HEAD_DIM, int CTAQ, bool PAGED, bool CAUSAL,
-          bool MASK = false>
-__global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_prefill_kernel(PrefillParams p) {
+// over more q rows). HDQK/HDVO may differ (DeepSeek MHA 192/128, reference
+// benchmarks/samples/sample_testlist_output.csv:5 config).
+template <typename T, typename TKV, int HDQK, int HDVO, int CTAQ, bool PAGED,
+          bool CAUSAL, bool MASK = false>
+__global__ __launch_bounds__(CTAQ * 2, ((HDQK >= 192 || HDVO >= 256) ? 1 : 2)) void batch_prefill_kernel(PrefillParams p) {
   constexpr int NTHREADS = CTAQ * 2;
   // fp8 (e4m3) KV caches are dequantized to bf16 during the LDS staging
   // write (reference prefill.cuh:1150 repack_fp8_tile_to_bf16 design): the
   // HBM bytes halve, the MFMA pipeline stays bf16.
   constexpr bool kF8KV = !__is_same(T, TKV);
-  constexpr int KCH = HEAD_DIM / 16;  // k-chunks in QK^T
-  constexpr int DT = HEAD_DIM / 32;   // d-tiles in PV / output
-  constexpr int KROWB = HEAD_DIM * 2;        // K/V tile row bytes
+  constexpr int KCH = HDQK / 16;  // k-chunks in QK^T
+  constexpr int DT = HDVO / 32;   // d-tiles in PV / output
+  // K tile row stride: power-of-two dims keep the XOR swizzle; 192 (384 B
+  // rows) pads to 400 B — row*400 mod 256 has period 16 rows, so 32-row
+  // fragment reads see at worst 2-way bank conflicts with no XOR needed.
+  constexpr int KROWB = HDQK == 192 ? 400 : HDQK * 2;
 
   // DOUBLE-BUFFERED K/V stage: compute reads buf while the next tile's
   // registers write buf^1 — ONE barrier per KV tile instead of two (PMC r01:
   // the kernel is barrier-, not MFMA- or bandwidth-bound).
-  __shared__ T Ks[2][KVB * HEAD_DIM];
-  // V is stored in [KVB/4][HEAD_DIM/16] subtiles of [4 kv][16 d] (row-major,
+  __shared__ T Ks[2][KVB * KROWB / 2];
+  // V is stored in [KVB/4][HDVO/16] subtiles of [4 kv][16 d] (row-major,
   // 128 B each) padded to 144 B stride: the shape ds_read_b64_tr_b16 wants
   // (each 16-lane group reads one subtile and receives it transposed), with
   // the pad de-aliasing subtile bank positions.
   constexpr int VTILE_STRIDE = 72;  // elems (144 B)
-  __shared__ T Vs[2][(KVB / 4) * (HEAD_DIM / 16) * VTILE_STRIDE];
+  __shared__ T Vs[2][(KVB / 4) * (HDVO / 16) * VTILE_STRIDE];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -117,6 +121,20 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
   // causal offset: kv position of q row my_qpos's "diagonal"
   const int64_t diag = kv_len - qo_len;
 
+  // split-KV: this tile covers only [chunk*kv_chunk, +kv_chunk) of the KV
+  // (bounds-masking clamps to kv_valid_hi; causal/ALiBi geometry keeps the
+  // true kv_len). n_chunks is derived per request from the plan's global
+  // chunk size (reference scheduler.cuh:101 binary-searched kv_chunk_size).
+  int chunk = 0, n_chunks = 1;
+  int64_t kv_valid_hi = kv_len;
+  if (p.tile_kv_chunk) {
+    chunk = p.tile_kv_chunk[tile];
+    n_chunks = (int)((kv_len + p.kv_chunk - 1) / p.kv_chunk);
+    if (n_chunks < 1) n_chunks = 1;
+    int64_t chi = ((int64_t)chunk + 1) * p.kv_chunk;
+    if (chi < kv_valid_hi) kv_valid_hi = chi;
+  }
+
   // ---- load Q fragments (pre-fold nothing; scale applied post-MFMA) ----
   using frag = typename mfma_ab_frag<T>::type;
   frag qf[KCH];
@@ -145,10 +163,11 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
   const float cap = p.logits_soft_cap;
   const int wleft = p.window_left;
 
-  // kv range this tile must process (causal upper bound; window lower bound);
-  // each wave also skips compute past its own rows' causal bound.
-  int64_t kv_hi = kv_len;
-  int64_t wave_kv_hi = kv_len;
+  // kv range this tile must process (causal upper bound; window lower bound;
+  // split-KV chunk bounds); each wave also skips compute past its own rows'
+  // causal bound.
+  int64_t kv_hi = kv_valid_hi;
+  int64_t wave_kv_hi = kv_valid_hi;
   if constexpr (CAUSAL) {
     int tile_max_qpos = (qstart + CTAQ - 1) / (int)group;
     if (tile_max_qpos >= qo_len) tile_max_qpos = qo_len - 1;
@@ -159,10 +178,11 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
     wave_kv_hi = wave_max_qpos + diag + 1;
   }
   int64_t kv_lo = 0;
+  if (p.tile_kv_chunk) kv_lo = (int64_t)chunk * p.kv_chunk;
   if (wleft >= 0) {
     int tile_min_qpos = qstart / (int)group;
     int64_t lo = tile_min_qpos + diag - wleft;
-    if (lo > 0) kv_lo = lo;
+    if (lo > kv_lo) kv_lo = lo;
   }
 
   const TKV* kbase = (const TKV*)p.k_data;
@@ -171,60 +191,117 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
   // ---- async-STAGE split (guide T14): global loads for tile n+1 are issued
   // BEFORE tile n's compute (HBM latency hides under the MFMA phases); the
   // LDS writes land between the two barriers after compute. ----
-  constexpr int S_UNITS = KVB * HEAD_DIM / 8;  // 8-elem units
-  constexpr int S_ITER = S_UNITS / NTHREADS;
-  vec_t<TKV, 8> kreg[S_ITER], vreg[S_ITER];
+  constexpr int KS_ITER = KVB * HDQK / 8 / NTHREADS;
+  constexpr int VS_ITER = KVB * HDVO / 8 / NTHREADS;
+  vec_t<TKV, 8> kreg[KS_ITER], vreg[VS_ITER];
 
   auto stage_load = [&](int64_t kv0) {
+    if constexpr (HDQK == HDVO) {
+      // square dims: one page lookup serves both K and V (k/v caches share a
+      // layout, so the k strides address both — the tuned flagship path)
 #pragma unroll
-    for (int it = 0; it < S_ITER; ++it) {
-      int u = tid + it * NTHREADS;
-      int row = u / (HEAD_DIM / 8);
-      int chunk = u % (HEAD_DIM / 8);
-      int64_t kvpos = kv0 + row;
-      kreg[it].fill(0.f);
-      vreg[it].fill(0.f);
-      if (kvpos < kv_len) {
-        int64_t off;
-        if constexpr (PAGED) {
-          uint32_t pg, entry;
-          p.page_size.divmod((uint32_t)kvpos, pg, entry);
-          off = (int64_t)page_ids[pg] * p.kv_stride_page +
-                (int64_t)kv_head * p.kv_stride_h + (int64_t)entry * p.kv_stride_n +
-                chunk * 8;
-        } else {
-          off = (kv_base + kvpos) * p.kv_stride_n + (int64_t)kv_head * p.kv_stride_h +
-                chunk * 8;
+      for (int it = 0; it < KS_ITER; ++it) {
+        int u = tid + it * NTHREADS;
+        int row = u / (HDQK / 8);
+        int chunk8 = u % (HDQK / 8);
+        int64_t kvpos = kv0 + row;
+        kreg[it].fill(0.f);
+        vreg[it].fill(0.f);
+        if (kvpos < kv_valid_hi) {
+          int64_t off;
+          if constexpr (PAGED) {
+            uint32_t pg, entry;
+            p.page_size.divmod((uint32_t)kvpos, pg, entry);
+            off = (int64_t)page_ids[pg] * p.kv_stride_page +
+                  (int64_t)kv_head * p.kv_stride_h + (int64_t)entry * p.kv_stride_n +
+                  chunk8 * 8;
+          } else {
+            off = (kv_base + kvpos) * p.kv_stride_n + (int64_t)kv_head * p.kv_stride_h +
+                  chunk8 * 8;
+          }
+          kreg[it].load(kbase + off);
+          vreg[it].load(vbase + off);
         }
-        kreg[it].load(kbase + off);
-        vreg[it].load(vbase + off);
+      }
+    } else {
+#pragma unroll
+      for (int it = 0; it < KS_ITER; ++it) {
+        int u = tid + it * NTHREADS;
+        int row = u / (HDQK / 8);
+        int chunk8 = u % (HDQK / 8);
+        int64_t kvpos = kv0 + row;
+        kreg[it].fill(0.f);
+        if (kvpos < kv_valid_hi) {
+          int64_t off;
+          if constexpr (PAGED) {
+            uint32_t pg, entry;
+            p.page_size.divmod((uint32_t)kvpos, pg, entry);
+            off = (int64_t)page_ids[pg] * p.kv_stride_page +
+                  (int64_t)kv_head * p.kv_stride_h + (int64_t)entry * p.kv_stride_n;
+          } else {
+            off = (kv_base + kvpos) * p.kv_stride_n + (int64_t)kv_head * p.kv_stride_h;
+          }
+          kreg[it].load(kbase + off + chunk8 * 8);
+        }
+      }
+#pragma unroll
+      for (int it = 0; it < VS_ITER; ++it) {
+        int u = tid + it * NTHREADS;
+        int row = u / (HDVO / 8);
+        int chunk8 = u % (HDVO / 8);
+        int64_t kvpos = kv0 + row;
+        vreg[it].fill(0.f);
+        if (kvpos < kv_valid_hi) {
+          int64_t off;
+          if constexpr (PAGED) {
+            uint32_t pg, entry;
+            p.page_size.divmod((uint32_t)kvpos, pg, entry);
+            off = (int64_t)page_ids[pg] * p.v_stride_page +
+                  (int64_t)kv_head * p.v_stride_h + (int64_t)entry * p.v_stride_n;
+          } else {
+            off = (kv_base + kvpos) * p.v_stride_n + (int64_t)kv_head * p.v_stride_h;
+          }
+          vreg[it].load(vbase + off + chunk8 * 8);
+        }
       }
     }
   };
   auto stage_write = [&](int wb) {
 #pragma unroll
-    for (int it = 0; it < S_ITER; ++it) {
+    for (int it = 0; it < KS_ITER; ++it) {
       int u = tid + it * NTHREADS;
-      int row = u / (HEAD_DIM / 8);
-      int chunk = u % (HEAD_DIM / 8);
-      shortx8 kw, vw;
+      int row = u / (HDQK / 8);
+      int chunk = u % (HDQK / 8);
+      shortx8 kw;
       if constexpr (kF8KV) {
         // dequantize fp8 -> bf16 on the staging write
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
+        for (int j = 0; j < 8; ++j)
           reinterpret_cast<T*>(&kw)[j] = from_f32<T>(kreg[it].get(j) * p.k_descale);
-          reinterpret_cast<T*>(&vw)[j] = from_f32<T>(vreg[it].get(j) * p.v_descale);
-        }
       } else {
         kw = *reinterpret_cast<const shortx8*>(kreg[it].data);
-        vw = *reinterpret_cast<const shortx8*>(vreg[it].data);
       }
-      // K: row-major swizzled (vector frag reads); V: tr-read subtiled
+      // K: row-major swizzled (vector frag reads)
       *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Ks[wb]) +
                                   swz_row<KROWB>(row * KROWB + chunk * 16)) = kw;
+    }
+#pragma unroll
+    for (int it = 0; it < VS_ITER; ++it) {
+      int u = tid + it * NTHREADS;
+      int row = u / (HDVO / 8);
+      int chunk = u % (HDVO / 8);
+      shortx8 vw;
+      if constexpr (kF8KV) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          reinterpret_cast<T*>(&vw)[j] = from_f32<T>(vreg[it].get(j) * p.v_descale);
+      } else {
+        vw = *reinterpret_cast<const shortx8*>(vreg[it].data);
+      }
+      // V: tr-read subtiled
       *reinterpret_cast<shortx8*>(
           reinterpret_cast<char*>(Vs[wb]) +
-          ((row >> 2) * (HEAD_DIM / 16) + (chunk >> 1)) * (VTILE_STRIDE * 2) +
+          ((row >> 2) * (HDVO / 16) + (chunk >> 1)) * (VTILE_STRIDE * 2) +
           (row & 3) * 32 + (chunk & 1) * 16) = vw;
     }
   };
@@ -265,7 +342,7 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
       // per-element predicate work (the VALU hotspot) is skipped.
       float pr[16];
       const int64_t kvt0 = kv0 + kt * 32;
-      bool tile_full = (kvt0 + 32 <= kv_len) && !MASK && cap <= 0.f && !p.alibi;
+      bool tile_full = (kvt0 + 32 <= kv_valid_hi) && !MASK && cap <= 0.f && !p.alibi;
       if constexpr (CAUSAL) {
         int wave_min_qpos = wq0 / (int)group;
         tile_full &= (kvt0 + 32 <= wave_min_qpos + diag + 1);
@@ -292,7 +369,7 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
           if (cap > 0.f) sv = cap * tanhf(sv / cap);
           if (p.alibi) sv -= slope * (float)(my_qpos + diag - kv);
           sv *= kLog2e;
-          bool ok = kv < kv_len;
+          bool ok = kv < kv_valid_hi;
           if constexpr (CAUSAL) ok &= kv <= my_qpos + diag;
           if (wleft >= 0) ok &= kv >= my_qpos + diag - wleft;
           if constexpr (MASK) {
@@ -371,13 +448,13 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
         const int kvb = kt * 32 + khalf;
 #pragma unroll
         for (int i = 0; i < DT; ++i) {
-          uint32_t a00 = vbase + (((kvb >> 2)) * (HEAD_DIM / 16) + i * 2 + tdsel) *
+          uint32_t a00 = vbase + (((kvb >> 2)) * (HDVO / 16) + i * 2 + tdsel) *
                                      (VTILE_STRIDE * 2);
           b16x4 r00 = ds_read_tr16(a00);
-          b16x4 r01 = ds_read_tr16(a00 + (HEAD_DIM / 16) * (VTILE_STRIDE * 2));
-          uint32_t a10 = a00 + 4 * (HEAD_DIM / 16) * (VTILE_STRIDE * 2);
+          b16x4 r01 = ds_read_tr16(a00 + (HDVO / 16) * (VTILE_STRIDE * 2));
+          uint32_t a10 = a00 + 4 * (HDVO / 16) * (VTILE_STRIDE * 2);
           b16x4 r10 = ds_read_tr16(a10);
-          b16x4 r11 = ds_read_tr16(a10 + (HEAD_DIM / 16) * (VTILE_STRIDE * 2));
+          b16x4 r11 = ds_read_tr16(a10 + (HDVO / 16) * (VTILE_STRIDE * 2));
           asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
           __builtin_amdgcn_sched_barrier(0);
           union {
@@ -407,20 +484,38 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
   float d_full = d_run + __shfl_xor(d_run, 32, 64);
   float inv_d = d_full > 0.f ? 1.f / d_full : 0.f;
   if (row_valid) {
-    T* optr = (T*)p.out + (int64_t)(qo_begin + my_qpos) * p.o_stride_n +
-              (int64_t)(kv_head * group + my_g_u) * p.o_stride_h;
+    const int qh = (int)(kv_head * group + my_g_u);
+    if (p.tile_kv_chunk) {
+      // split-KV: normalized f32 partial + base-2 lse into the merge slots
+      const int slot = p.req_slot_base[req] + my_qpos * n_chunks + chunk;
+      float* vout = p.tmp_v + ((int64_t)slot * p.num_qo_heads + qh) * HDVO;
 #pragma unroll
-    for (int i = 0; i < DT; ++i) {
+      for (int i = 0; i < DT; ++i) {
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int d = i * 32 + mfma32_cd_row(r, lane);
-        optr[d] = from_f32<T>(acc_o[i][r] * inv_d);
+        for (int r = 0; r < 16; ++r) {
+          int d = i * 32 + mfma32_cd_row(r, lane);
+          vout[d] = acc_o[i][r] * inv_d;
+        }
       }
-    }
-    if (p.lse && (lane >> 5) == 0) {
-      float l2 = d_full > 0.f ? m_run + __builtin_log2f(d_full) : -INFINITY;
-      p.lse[(int64_t)(qo_begin + my_qpos) * p.num_qo_heads + kv_head * group + my_g_u] =
-          l2;
+      if ((lane >> 5) == 0) {
+        p.tmp_s[(int64_t)slot * p.num_qo_heads + qh] =
+            d_full > 0.f ? m_run + __builtin_log2f(d_full) : -INFINITY;
+      }
+    } else {
+      T* optr = (T*)p.out + (int64_t)(qo_begin + my_qpos) * p.o_stride_n +
+                (int64_t)qh * p.o_stride_h;
+#pragma unroll
+      for (int i = 0; i < DT; ++i) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int d = i * 32 + mfma32_cd_row(r, lane);
+          optr[d] = from_f32<T>(acc_o[i][r] * inv_d);
+        }
+      }
+      if (p.lse && (lane >> 5) == 0) {
+        float l2 = d_full > 0.f ? m_run + __builtin_log2f(d_full) : -INFINITY;
+        p.lse[(int64_t)(qo_begin + my_qpos) * p.num_qo_heads + qh] = l2;
+      }
     }
   }
   prof_event(p.prof_buf, 0, ProfType::kEnd);
@@ -429,39 +524,48 @@ __global__ __launch_bounds__(CTAQ * 2, (HEAD_DIM >= 256 ? 1 : 2)) void batch_pre
 template <typename T, typename TKV>
 hipError_t prefill_dispatch(PrefillParams& p, bool paged, hipStream_t stream) {
   dim3 g(p.n_tiles, p.num_kv_heads), blk(p.cta_q * 2);
-#define LAUNCH_PF2(HD, CQ, PG, CS) \
-  hipLaunchKernelGGL((batch_prefill_kernel<T, TKV, HD, CQ, PG, CS>), g, blk, 0, stream, p)
-#define LAUNCH_PF(HD, PG, CS)                        \
-  do {                                               \
-    if (p.cta_q == 256) LAUNCH_PF2(HD, 256, PG, CS); \
-    else LAUNCH_PF2(HD, 128, PG, CS);                \
-  } while (0)
-#define LAUNCH_PFM(HD, CQ, PG)                                                       \
-  hipLaunchKernelGGL((batch_prefill_kernel<T, TKV, HD, CQ, PG, false, true>), g, blk, \
+#define LAUNCH_PF2(HD, HV, CQ, PG, CS)                                            \
+  hipLaunchKernelGGL((batch_prefill_kernel<T, TKV, HD, HV, CQ, PG, CS>), g, blk, \
                      0, stream, p)
-#define DISPATCH_PC(HD)                                         \
+#define LAUNCH_PF(HD, HV, PG, CS)                        \
+  do {                                                   \
+    if (p.cta_q == 256) LAUNCH_PF2(HD, HV, 256, PG, CS); \
+    else LAUNCH_PF2(HD, HV, 128, PG, CS);                \
+  } while (0)
+#define LAUNCH_PFM(HD, HV, CQ, PG)                                                \
+  hipLaunchKernelGGL((batch_prefill_kernel<T, TKV, HD, HV, CQ, PG, false, true>), \
+                     g, blk, 0, stream, p)
+#define DISPATCH_PC(HD, HV)                                     \
   do {                                                          \
     if (p.mask_data) {                                          \
       if (paged) {                                              \
-        if (p.cta_q == 256) LAUNCH_PFM(HD, 256, true);          \
-        else LAUNCH_PFM(HD, 128, true);                         \
+        if (p.cta_q == 256) LAUNCH_PFM(HD, HV, 256, true);      \
+        else LAUNCH_PFM(HD, HV, 128, true);                     \
       } else {                                                  \
-        if (p.cta_q == 256) LAUNCH_PFM(HD, 256, false);         \
-        else LAUNCH_PFM(HD, 128, false);                        \
+        if (p.cta_q == 256) LAUNCH_PFM(HD, HV, 256, false);     \
+        else LAUNCH_PFM(HD, HV, 128, false);                    \
       }                                                         \
     } else if (paged) {                                         \
-      if (p.causal) LAUNCH_PF(HD, true, true);                  \
-      else LAUNCH_PF(HD, true, false);                          \
+      if (p.causal) LAUNCH_PF(HD, HV, true, true);              \
+      else LAUNCH_PF(HD, HV, true, false);                      \
     } else {                                                    \
-      if (p.causal) LAUNCH_PF(HD, false, true);                 \
-      else LAUNCH_PF(HD, false, false);                         \
+      if (p.causal) LAUNCH_PF(HD, HV, false, true);             \
+      else LAUNCH_PF(HD, HV, false, false);                     \
     }                                                           \
   } while (0)
-  switch (p.head_dim) {
-    case 64: DISPATCH_PC(64); break;
-    case 128: DISPATCH_PC(128); break;
-    case 256: DISPATCH_PC(256); break;
-    default: return hipErrorInvalidValue;
+  int hv = p.head_dim_vo > 0 ? p.head_dim_vo : p.head_dim;
+  if (p.head_dim == hv) {
+    switch (p.head_dim) {
+      case 64: DISPATCH_PC(64, 64); break;
+      case 128: DISPATCH_PC(128, 128); break;
+      case 256: DISPATCH_PC(256, 256); break;
+      default: return hipErrorInvalidValue;
+    }
+  } else if (p.head_dim == 192 && hv == 128) {
+    // DeepSeek MHA prefill (q/k 128 nope + 64 rope, v 128)
+    DISPATCH_PC(192, 128);
+  } else {
+    return hipErrorInvalidValue;
   }
 #undef DISPATCH_PC
 #undef LAUNCH_PF

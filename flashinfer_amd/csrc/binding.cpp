@@ -425,7 +425,12 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                        c10::optional<at::Tensor> mask_data,
                        c10::optional<at::Tensor> mask_byte_indptr, bool alibi,
                        double k_descale, double v_descale,
-                       c10::optional<at::Tensor> prof_buf = c10::nullopt) {
+                       c10::optional<at::Tensor> prof_buf = c10::nullopt,
+                       c10::optional<at::Tensor> tile_kv_chunk = c10::nullopt,
+                       int64_t kv_chunk = 0,
+                       c10::optional<at::Tensor> req_slot_base = c10::nullopt,
+                       c10::optional<at::Tensor> tmp_v = c10::nullopt,
+                       c10::optional<at::Tensor> tmp_s = c10::nullopt) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 3, "q must be [nnz, Hq, D]");
   TORCH_CHECK(q.stride(2) == 1 && out.stride(2) == 1);
   fi_ext::PrefillParams p{};
@@ -447,6 +452,10 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
     p.kv_stride_page = k_cache.stride(0);
     p.kv_stride_n = layout == 0 ? k_cache.stride(1) : k_cache.stride(2);
     p.kv_stride_h = layout == 0 ? k_cache.stride(2) : k_cache.stride(1);
+    p.v_stride_page = v_cache.stride(0);
+    p.v_stride_n = layout == 0 ? v_cache.stride(1) : v_cache.stride(2);
+    p.v_stride_h = layout == 0 ? v_cache.stride(2) : v_cache.stride(1);
+    p.head_dim_vo = v_cache.size(3);
   } else {
     TORCH_CHECK(k_cache.dim() == 3);
     page_size = 1;
@@ -454,6 +463,10 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
     p.kv_stride_page = 0;
     p.kv_stride_n = k_cache.stride(0);
     p.kv_stride_h = k_cache.stride(1);
+    p.v_stride_page = 0;
+    p.v_stride_n = v_cache.stride(0);
+    p.v_stride_h = v_cache.stride(1);
+    p.head_dim_vo = v_cache.size(2);
   }
   p.page_size = fi::uint_fastdiv((uint32_t)page_size);
   p.tile_req = tile_req.data_ptr<int32_t>();
@@ -482,6 +495,15 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   if (prof_buf.has_value()) {
     TORCH_CHECK(prof_buf->scalar_type() == at::kUInt64 && prof_buf->is_cuda());
     p.prof_buf = (unsigned long long*)prof_buf->data_ptr();
+  }
+  if (tile_kv_chunk.has_value()) {
+    TORCH_CHECK(req_slot_base.has_value() && tmp_v.has_value() && tmp_s.has_value(),
+                "split-KV prefill needs req_slot_base/tmp_v/tmp_s");
+    p.tile_kv_chunk = tile_kv_chunk->data_ptr<int32_t>();
+    p.kv_chunk = (int)kv_chunk;
+    p.req_slot_base = req_slot_base->data_ptr<int32_t>();
+    p.tmp_v = (float*)tmp_v->data_ptr();
+    p.tmp_s = tmp_s->data_ptr<float>();
   }
   check_hip(fi_batch_prefill(dtype_code(q), dtype_code(k_cache), &p, paged ? 1 : 0,
                              cur_stream(q)),

@@ -81,6 +81,19 @@ struct PrefillParams {
   int alibi;                       // ALiBi position bias (slope by qo head)
   float k_descale, v_descale;      // fp8-KV dequant factors (staging)
   unsigned long long* prof_buf;    // intra-kernel event buffer (optional)
+  // head_dim_qk != head_dim_vo (DeepSeek MHA 192/128): V gets its own strides
+  int head_dim_vo;                 // = head_dim when square
+  int64_t v_stride_page, v_stride_n, v_stride_h;
+  // split-KV (reference scheduler.cuh:545 PrefillSplitQOKVIndptr role):
+  // when tile_kv_chunk != null each tile covers one kv chunk and writes
+  // normalized f32 partials + base-2 lse to (tmp_v, tmp_s) at
+  // slot = req_slot_base[req] + qpos * n_chunks(req) + chunk; the host then
+  // runs the LSE merge kernel over per-token slot ranges.
+  const int32_t* tile_kv_chunk;    // [n_tiles] chunk index, null = no split
+  int kv_chunk;                    // tokens per kv chunk
+  const int32_t* req_slot_base;    // [batch]
+  float* tmp_v;                    // [n_slots, Hq, head_dim_vo]
+  float* tmp_s;                    // [n_slots, Hq]
 };
 
 

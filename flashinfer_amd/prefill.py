@@ -36,6 +36,63 @@ def _plan_tiles(qo_lens, group: int, causal: bool = False):
     return cta_q, tile_req, tile_qstart
 
 
+# split-KV engages when the (q_tile x kv_head) grid underfills the 256-CU
+# chip (reference scheduler.cuh:545 PrefillSplitQOKVIndptr + :101 binary
+# search role): a bs=1 long-kv short-q prefill otherwise runs on a handful
+# of workgroups.
+_SPLIT_TARGET_UNITS = 512
+_SPLIT_MIN_CHUNK = 256  # tokens (multiple of the 64-token kv tile)
+
+
+def _plan_tiles_split(qo_lens, kv_lens, group, causal, num_kv_heads,
+                      num_qo_heads, head_dim_vo, ws_bytes):
+    """Emit (req, qstart[, kv_chunk]) work items, splitting KV when the base
+    grid underfills the chip. Returns a dict with the tile arrays and (when
+    split) the merge bookkeeping."""
+    cta_q, tile_req, tile_qstart = _plan_tiles(qo_lens, group, causal)
+    units = len(tile_req) * num_kv_heads
+    max_kv = max(kv_lens) if kv_lens else 0
+    if units >= _SPLIT_TARGET_UNITS or max_kv < 2 * _SPLIT_MIN_CHUNK:
+        return dict(cta_q=cta_q, tile_req=tile_req, tile_qstart=tile_qstart,
+                    split=False)
+    want = min(-(-_SPLIT_TARGET_UNITS // max(1, units)),
+               -(-max_kv // _SPLIT_MIN_CHUNK))
+    kv_chunk = -(-max_kv // want)
+    kv_chunk = -(-kv_chunk // _SPLIT_MIN_CHUNK) * _SPLIT_MIN_CHUNK
+    # bound the partial buffers by the float workspace
+    def slots_for(ck):
+        return sum(q * max(1, -(-k // ck)) for q, k in zip(qo_lens, kv_lens))
+    bytes_per_slot = num_qo_heads * (head_dim_vo + 1) * 4
+    while slots_for(kv_chunk) * bytes_per_slot > ws_bytes * 0.9:
+        kv_chunk *= 2
+        if kv_chunk >= max_kv:
+            return dict(cta_q=cta_q, tile_req=tile_req,
+                        tile_qstart=tile_qstart, split=False)
+    n_chunks = [max(1, -(-k // kv_chunk)) for k in kv_lens]
+    slot_base, acc = [], 0
+    for q, nc in zip(qo_lens, n_chunks):
+        slot_base.append(acc)
+        acc += q * nc
+    n_slots = acc
+    # every (qtile, chunk) pair is emitted — chunks fully beyond a causal
+    # bound early-out in-kernel and write empty partials, which the LSE
+    # merge treats as zero weight (no uninitialized slots by construction)
+    s_req, s_qstart, s_chunk = [], [], []
+    for b, st in zip(tile_req, tile_qstart):
+        for c in range(n_chunks[b]):
+            s_req.append(b)
+            s_qstart.append(st)
+            s_chunk.append(c)
+    merge_indptr = [0]
+    for q, nc in zip(qo_lens, n_chunks):
+        for _ in range(q):
+            merge_indptr.append(merge_indptr[-1] + nc)
+    return dict(cta_q=cta_q, tile_req=s_req, tile_qstart=s_qstart,
+                tile_kv_chunk=s_chunk, kv_chunk=kv_chunk,
+                slot_base=slot_base, n_slots=n_slots,
+                merge_indptr=merge_indptr, split=True)
+
+
 class _BatchPrefillBase:
     def __init__(self, float_workspace_buffer, kv_layout="NHD", use_cuda_graph=False,
                  backend="fa2", jit_args=None, **kwargs):
@@ -81,21 +138,53 @@ class _BatchPrefillBase:
 
     def _plan_common(self, qo_indptr, num_qo_heads, num_kv_heads, head_dim, causal,
                      sm_scale, window_left, logits_soft_cap, non_blocking=True,
-                     alibi=False):
+                     alibi=False, kv_lens=None, head_dim_vo=None):
         qi = qo_indptr.to("cpu", torch.int64)
         qo_lens = (qi[1:] - qi[:-1]).tolist()
         group = num_qo_heads // num_kv_heads
-        cta_q, tile_req, tile_qstart = _plan_tiles(qo_lens, group, causal)
-        n_tiles = len(tile_req)
-        meta = torch.tensor(tile_req + tile_qstart, dtype=torch.int32).to(
-            self.device, non_blocking=non_blocking
+        head_dim_vo = head_dim_vo or head_dim
+        plan = _plan_tiles_split(
+            qo_lens, kv_lens if kv_lens is not None else [0] * len(qo_lens),
+            group, causal, num_kv_heads, num_qo_heads, head_dim_vo,
+            self._float_workspace_buffer.numel()
+            * self._float_workspace_buffer.element_size(),
         )
-        self._tile_req = meta[:n_tiles]
-        self._tile_qstart = meta[n_tiles:]
+        cta_q = plan["cta_q"]
+        tile_req, tile_qstart = plan["tile_req"], plan["tile_qstart"]
+        n_tiles = len(tile_req)
+        self._split = plan["split"]
+        if self._split:
+            meta = torch.tensor(
+                tile_req + tile_qstart + plan["tile_kv_chunk"] + plan["slot_base"]
+                + plan["merge_indptr"], dtype=torch.int32,
+            ).to(self.device, non_blocking=non_blocking)
+            self._tile_req = meta[:n_tiles]
+            self._tile_qstart = meta[n_tiles : 2 * n_tiles]
+            self._tile_kv_chunk = meta[2 * n_tiles : 3 * n_tiles]
+            nb = len(plan["slot_base"])
+            self._req_slot_base = meta[3 * n_tiles : 3 * n_tiles + nb]
+            self._merge_indptr = meta[3 * n_tiles + nb :]
+            self._kv_chunk = plan["kv_chunk"]
+            from .utils import WorkspaceAllocator
+
+            alloc = WorkspaceAllocator(self._float_workspace_buffer)
+            ns = plan["n_slots"]
+            self._tmp_v = alloc.alloc(
+                ns * num_qo_heads * head_dim_vo * 4, torch.float32,
+                (ns, num_qo_heads, head_dim_vo))
+            self._tmp_s = alloc.alloc(
+                ns * num_qo_heads * 4, torch.float32, (ns, num_qo_heads))
+        else:
+            meta = torch.tensor(tile_req + tile_qstart, dtype=torch.int32).to(
+                self.device, non_blocking=non_blocking
+            )
+            self._tile_req = meta[:n_tiles]
+            self._tile_qstart = meta[n_tiles:]
         self._qo_indptr_d = qo_indptr.to(self.device, torch.int32,
                                          non_blocking=non_blocking)
         self._plan_info = dict(
             num_qo_heads=num_qo_heads, num_kv_heads=num_kv_heads, head_dim=head_dim,
+            head_dim_vo=head_dim_vo,
             causal=causal, window_left=window_left,
             logits_soft_cap=float(logits_soft_cap or 0.0),
             sm_scale=sm_scale if sm_scale is not None else default_sm_scale(head_dim),
@@ -108,9 +197,12 @@ class _BatchPrefillBase:
         pi = self._plan_info
         if pi is None:
             raise RuntimeError("must call plan() before run()")
+        vo = pi.get("head_dim_vo") or q.shape[2]
         if out is None:
-            out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
-        if return_lse and lse is None:
+            out = torch.empty(q.shape[0], q.shape[1], vo, dtype=q.dtype,
+                              device=q.device)
+        split = getattr(self, "_split", False)
+        if (return_lse or split) and lse is None:
             lse = torch.empty(q.shape[0], q.shape[1], dtype=torch.float32,
                               device=q.device)
         sm_scale = pi["sm_scale"]
@@ -130,7 +222,17 @@ class _BatchPrefillBase:
             pi["cta_q"], getattr(self, "_mask_data", None),
             getattr(self, "_mask_byte_indptr", None), pi.get("alibi", False),
             k_descale, v_descale, profiler_buffer,
+            self._tile_kv_chunk if split else None,
+            self._kv_chunk if split else 0,
+            self._req_slot_base if split else None,
+            self._tmp_v if split else None,
+            self._tmp_s if split else None,
         )
+        if split:
+            # LSE merge of the per-chunk partials (cascade merge kernel)
+            get_ext().merge_states(
+                self._tmp_v, self._tmp_s, out, lse, self._merge_indptr, 0,
+                q.shape[0])
         if not kv_fp8 and v_scale is not None:
             out = out * v_scale
         return (out, lse) if return_lse else out
@@ -153,9 +255,15 @@ class BatchPrefillWithPagedKVCacheWrapper(_BatchPrefillBase):
     ):
         if pos_encoding_mode not in ("NONE", "ALIBI"):
             raise NotImplementedError("apply RoPE beforehand")
+        ip = paged_kv_indptr.to("cpu", torch.int64)
+        lp = paged_kv_last_page_len.to("cpu", torch.int64)
+        np_ = ip[1:] - ip[:-1]
+        kv_lens = (torch.clamp(np_ - 1, min=0) * page_size
+                   + torch.where(np_ > 0, lp, torch.zeros_like(lp))).tolist()
         self._plan_common(qo_indptr, num_qo_heads, num_kv_heads, head_dim_qk, causal,
                           sm_scale, window_left, logits_soft_cap, non_blocking,
-                          alibi=pos_encoding_mode == "ALIBI")
+                          alibi=pos_encoding_mode == "ALIBI", kv_lens=kv_lens,
+                          head_dim_vo=head_dim_vo)
         self._kv_indptr_d = paged_kv_indptr.to(self.device, torch.int32,
                                                non_blocking=non_blocking)
         self._kv_indices_d = paged_kv_indices.to(self.device, torch.int32,
@@ -163,11 +271,6 @@ class BatchPrefillWithPagedKVCacheWrapper(_BatchPrefillBase):
         self._kv_last_page_len_d = paged_kv_last_page_len.to(
             self.device, torch.int32, non_blocking=non_blocking)
         if custom_mask is not None or packed_custom_mask is not None:
-            ip = paged_kv_indptr.to("cpu", torch.int64)
-            lp = paged_kv_last_page_len.to("cpu", torch.int64)
-            np_ = ip[1:] - ip[:-1]
-            kv_lens = (torch.clamp(np_ - 1, min=0) * page_size
-                       + torch.where(np_ > 0, lp, torch.zeros_like(lp))).tolist()
             qi = qo_indptr.to("cpu", torch.int64)
             qo_lens = (qi[1:] - qi[:-1]).tolist()
             self._pack_mask(custom_mask, packed_custom_mask, qo_lens, kv_lens)
@@ -198,9 +301,12 @@ class BatchPrefillWithRaggedKVCacheWrapper(_BatchPrefillBase):
     ):
         if pos_encoding_mode not in ("NONE", "ALIBI"):
             raise NotImplementedError("apply RoPE beforehand")
+        ki = kv_indptr.to("cpu", torch.int64)
+        kv_lens = (ki[1:] - ki[:-1]).tolist()
         self._plan_common(qo_indptr, num_qo_heads, num_kv_heads, head_dim_qk, causal,
                           sm_scale, window_left, logits_soft_cap, non_blocking,
-                          alibi=pos_encoding_mode == "ALIBI")
+                          alibi=pos_encoding_mode == "ALIBI", kv_lens=kv_lens,
+                          head_dim_vo=head_dim_vo)
         self._kv_indptr_d = kv_indptr.to(self.device, torch.int32,
                                          non_blocking=non_blocking)
         if custom_mask is not None or packed_custom_mask is not None:
@@ -241,6 +347,7 @@ def single_prefill_with_kv_cache(
         v = v.transpose(0, 1)
     qo_len, Hq, D = q.shape
     kv_len, Hkv, _ = k.shape
+    Dvo = v.shape[-1]
     dev = q.device
     group = Hq // Hkv
     cta_q, tile_req, tile_qstart = _plan_tiles([qo_len], group, causal)
@@ -248,7 +355,7 @@ def single_prefill_with_kv_cache(
     meta = torch.tensor(
         tile_req + tile_qstart + [0, qo_len, 0, kv_len], dtype=torch.int32
     ).to(dev)
-    out = torch.empty_like(q)
+    out = torch.empty(qo_len, Hq, Dvo, dtype=q.dtype, device=dev)
     lse = torch.empty(qo_len, Hq, dtype=torch.float32, device=dev) if return_lse else None
     mask_data = mask_indptr = None
     if custom_mask is not None or packed_custom_mask is not None:
@@ -265,6 +372,7 @@ def single_prefill_with_kv_cache(
         sm_scale if sm_scale is not None else default_sm_scale(D),
         float(logits_soft_cap or 0.0), window_left, causal, False, cta_q,
         mask_data, mask_indptr, pos_encoding_mode == "ALIBI", 1.0, 1.0, None,
+        None, 0, None, None, None,
     )
     return (out, lse) if return_lse else out
 
