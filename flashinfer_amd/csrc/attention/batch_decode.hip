@@ -324,16 +324,22 @@ __global__ __launch_bounds__(
 // shared function: this shape runs at launch_bounds(512,1) (256-VGPR budget,
 // GROUP-8 keeps q resident and a 4-deep ring with no spill) while the split
 // kernel is tuned at (256, 2-4); folding them would regress the tuned path.
-template <typename T, typename TKV, int HEAD_DIM, int GROUP, bool SOFT_CAP>
+template <typename T, typename TKV, int HEAD_DIM, int GROUP, bool SOFT_CAP,
+          bool WIDE8 = false>
 __global__ __launch_bounds__(512, 1) void decode_fused_kernel(DecodeParams p) {
   constexpr bool kSameT = __is_same(T, TKV);
+  // GROUP-8 wide (WIDE8): tried — VPL=16 would halve the per-token
+  // shfl-reduce chain, but o_acc[8][16] + the ring spills 191 VGPRs even at
+  // launch_bounds(512,1); never instantiated (GQA-8 small-batch decode is
+  // the MFMA decode kernel's job instead).
   constexpr bool kWide = (__is_same(T, bf16) || __is_same(T, fp16)) && kSameT &&
-                         HEAD_DIM >= 128 && GROUP <= 4;
+                         HEAD_DIM >= 128 && (GROUP <= 4 || (WIDE8 && GROUP == 8));
   constexpr int VPL = kWide ? 16 : 8;
   constexpr int LPT = HEAD_DIM / VPL;
   constexpr int TPW = kWaveSize / LPT;
   constexpr int WAVES = 8;
-  constexpr int STAGES = 4;
+  constexpr int STAGES = (kWide && GROUP >= 8) ? 2 : 4;
+  constexpr bool kQReg = !(kWide && GROUP >= 8);
   __shared__ float lds_o[WAVES][GROUP][HEAD_DIM];
   __shared__ float lds_m[WAVES][GROUP];
   __shared__ float lds_d[WAVES][GROUP];
@@ -360,16 +366,19 @@ __global__ __launch_bounds__(512, 1) void decode_fused_kernel(DecodeParams p) {
   int64_t end = ws + per_wave;
   if (end > kv_len) end = kv_len;
 
-  vec_t<T, VPL> qreg[GROUP];
-  float qf32[kSameT ? 1 : GROUP][kSameT ? 1 : VPL];
+  constexpr int QR = kQReg ? GROUP : 1;
+  vec_t<T, VPL> qreg[QR];
+  float qf32[kSameT ? 1 : QR][kSameT ? 1 : VPL];
   const T* qbase = (const T*)p.q + (int64_t)req * p.q_stride_n +
                    (int64_t)(kv_head * GROUP) * p.q_stride_h + dcol;
+  if constexpr (kQReg) {
 #pragma unroll
-  for (int g = 0; g < GROUP; ++g) {
-    qreg[g].load(qbase + (int64_t)g * p.q_stride_h);
-    if constexpr (!kSameT) {
+    for (int g = 0; g < GROUP; ++g) {
+      qreg[g].load(qbase + (int64_t)g * p.q_stride_h);
+      if constexpr (!kSameT) {
 #pragma unroll
-      for (int j = 0; j < VPL; ++j) qf32[g][j] = qreg[g].get(j);
+        for (int j = 0; j < VPL; ++j) qf32[g][j] = qreg[g].get(j);
+      }
     }
   }
   const float scale = p.sm_scale;
@@ -406,7 +415,13 @@ __global__ __launch_bounds__(512, 1) void decode_fused_kernel(DecodeParams p) {
     for (int g = 0; g < GROUP; ++g) {
       float s;
       if constexpr (kSameT) {
-        s = qk_dot<T, VPL>(qreg[g].data, kv_cur.data, 0.f);
+        if constexpr (kQReg) {
+          s = qk_dot<T, VPL>(qreg[g].data, kv_cur.data, 0.f);
+        } else {
+          vec_t<T, VPL> qv;  // L1-resident: same bytes every iteration
+          qv.load(qbase + (int64_t)g * p.q_stride_h);
+          s = qk_dot<T, VPL>(qv.data, kv_cur.data, 0.f);
+        }
       } else {
         s = 0.f;
 #pragma unroll

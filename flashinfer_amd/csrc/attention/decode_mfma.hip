@@ -1,0 +1,365 @@
+// MFMA fused decode — the GQA small-batch latency shape (GROUP >= 8).
+//
+// Role parity: reference decode.cuh:615 + the "tensor-core decode" design of
+// flashinfer/decode.py:1697, but a dedicated CDNA4 kernel instead of reusing
+// the 128-row prefill tile: at bs=16 the prefill reuse runs 128-row MFMA
+// tiles that are 94% padding AND underfills the grid; the vector kernel burns
+// 8 dot+shfl chains per K read (ds_bpermute-bound at GQA-8, measured 78 us at
+// the BASELINE bs=16/kv=1024 config). Here:
+//
+//  * one workgroup per (request, kv_head); 4 waves each own a contiguous
+//    quarter of the request's KV — NO __syncthreads in the main loop (each
+//    wave stages into its own LDS region; LDS ops of one wave complete in
+//    issue order, so single-buffered stage + register ring is race-free).
+//  * per wave: 32-token KV tiles, S^T = mfma32(K_frag, Q_frag) with the GQA
+//    group (8/16/32 q heads) as the 32-column q dimension — one MFMA pass
+//    amortizes the whole group; decode q rows are all at the same position so
+//    bounds/window/soft-cap/ALiBi masks are lane-uniform or per-lane scalars.
+//  * same softmax-in-registers dance as the prefill kernel (lane owns one q
+//    column, defer-max, v_cvt_pk_bf16_f32 P-pack, lazy half-exchange,
+//    ds_read_b64_tr_b16 V^T fragments).
+//  * epilogue: the 4 per-wave online-softmax states merge through LDS (one
+//    barrier in the whole kernel) and the output is written directly in the
+//    q dtype — one launch, no f32 round-trip, no merge kernel.
+#include "fi/common.hpp"
+#include "fi/params.hpp"
+#include "fi/fastdiv.hpp"
+#include "fi/frag.hpp"
+#include "fi/mfma.hpp"
+#include "fi/vec.hpp"
+
+namespace fi {
+
+namespace {
+constexpr int KVB = 32;  // kv tokens per tile (one 32x32 S^T MFMA pass)
+constexpr int WAVES = 4;
+constexpr float kLog2e = 1.4426950408889634f;
+
+typedef __attribute__((ext_vector_type(4))) __bf16 b16x4;
+
+__device__ __forceinline__ b16x4 ds_read_tr16(uint32_t lds_byte_off) {
+  b16x4 r;
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(r) : "v"(lds_byte_off));
+  return r;
+}
+}  // namespace
+
+template <typename T, int HEAD_DIM, int GROUP>
+__global__ __launch_bounds__(WAVES * 64, 1) void decode_mfma_kernel(DecodeParams p) {
+  constexpr int KCH = HEAD_DIM / 16;   // k-chunks in QK^T
+  constexpr int DT = HEAD_DIM / 32;    // d-tiles in PV / output
+  constexpr int KROWB = HEAD_DIM * 2;  // K LDS row bytes
+  constexpr int QROWS = GROUP < 32 ? GROUP : 32;
+  constexpr int VTILE_STRIDE = 72;     // elems (144 B) — tr-read subtile pad
+
+  // per-wave LDS regions (single-buffered; in-wave LDS ordering makes the
+  // read-then-overwrite safe) + the cross-wave merge arrays
+  __shared__ T Ks[WAVES][KVB * HEAD_DIM];
+  __shared__ T Vs[WAVES][(KVB / 4) * (HEAD_DIM / 16) * VTILE_STRIDE];
+  __shared__ float lds_o[WAVES][QROWS][HEAD_DIM];
+  __shared__ float lds_m[WAVES][QROWS];
+  __shared__ float lds_d[WAVES][QROWS];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int lq = lane & 31;      // q column this lane owns in S^T
+  const int khalf = (lane >> 5) * 8;
+
+  const int req = blockIdx.x;
+  const int kv_head = blockIdx.y;
+
+  int np = p.kv_indptr[req + 1] - p.kv_indptr[req];
+  int64_t kv_len =
+      np == 0 ? 0 : (int64_t)(np - 1) * p.page_size.d + p.kv_last_page_len[req];
+  int64_t kv_lo = 0;
+  if (p.window_left >= 0) {
+    int64_t w = kv_len - 1 - p.window_left;
+    if (w > 0) kv_lo = w;
+  }
+  // contiguous per-wave slice, 32-token aligned
+  int64_t total = kv_len - kv_lo;
+  int64_t per_wave = ((total + WAVES * KVB - 1) / (WAVES * KVB)) * KVB;
+  int64_t ws_ = kv_lo + (int64_t)wave * per_wave;
+  int64_t we_ = ws_ + per_wave;
+  if (we_ > kv_len) we_ = kv_len;
+
+  const int32_t* page_ids = p.kv_indices + p.kv_indptr[req];
+  const T* kbase = (const T*)p.k_data;
+  const T* vbase = (const T*)p.v_data;
+
+  // ---- Q fragments (B-operand): lane l holds Q[q = l&31][k = khalf + c*16] —
+  // decode q rows are the GQA group of this kv head ----
+  using frag = typename mfma_ab_frag<T>::type;
+  frag qf[KCH];
+  {
+    const bool qvalid = lq < QROWS;
+    const T* qptr = (const T*)p.q + (int64_t)req * p.q_stride_n +
+                    (int64_t)(kv_head * GROUP + lq) * p.q_stride_h;
+#pragma unroll
+    for (int c = 0; c < KCH; ++c) {
+      if (qvalid) qf[c] = *reinterpret_cast<const frag*>(qptr + c * 16 + khalf);
+      else qf[c] = frag{};
+    }
+  }
+
+  float m_run = -INFINITY, d_run = 0.f;
+  floatx16 acc_o[DT];
+#pragma unroll
+  for (int i = 0; i < DT; ++i) acc_o[i] = {};
+
+  const float scale2 = p.sm_scale * kLog2e;
+  const float cap = p.logits_soft_cap;
+  // ALiBi: slope per q head (this lane's column); bias = -slope*(pos - kv)
+  float slope = 0.f;
+  if (p.alibi)
+    slope = __builtin_exp2f(-8.f * (kv_head * GROUP + lq + 1) / p.num_qo_heads);
+
+  // ---- register-staged K/V tile loads (8-elem units; 64 lanes cover a
+  // 32 x HEAD_DIM tile in KVB*HEAD_DIM/8/64 iterations) ----
+  constexpr int S_ITER = KVB * HEAD_DIM / 8 / 64;
+  vec_t<T, 8> kreg[S_ITER], vreg[S_ITER];
+  auto stage_load = [&](int64_t kv0) {
+#pragma unroll
+    for (int it = 0; it < S_ITER; ++it) {
+      int u = lane + it * 64;
+      int row = u / (HEAD_DIM / 8);
+      int chunk8 = u % (HEAD_DIM / 8);
+      int64_t kvpos = kv0 + row;
+      kreg[it].fill(0.f);
+      vreg[it].fill(0.f);
+      if (kvpos < we_) {
+        uint32_t pg, entry;
+        p.page_size.divmod((uint32_t)kvpos, pg, entry);
+        int64_t off = (int64_t)page_ids[pg] * p.stride_page +
+                      (int64_t)kv_head * p.stride_h + (int64_t)entry * p.stride_n +
+                      chunk8 * 8;
+        kreg[it].load(kbase + off);
+        vreg[it].load(vbase + off);
+      }
+    }
+  };
+  auto stage_write = [&]() {
+#pragma unroll
+    for (int it = 0; it < S_ITER; ++it) {
+      int u = lane + it * 64;
+      int row = u / (HEAD_DIM / 8);
+      int chunk8 = u % (HEAD_DIM / 8);
+      *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Ks[wave]) +
+                                  swz256(row * KROWB + chunk8 * 16)) =
+          *reinterpret_cast<const shortx8*>(kreg[it].data);
+      *reinterpret_cast<shortx8*>(
+          reinterpret_cast<char*>(Vs[wave]) +
+          ((row >> 2) * (HEAD_DIM / 16) + (chunk8 >> 1)) * (VTILE_STRIDE * 2) +
+          (row & 3) * 32 + (chunk8 & 1) * 16) =
+          *reinterpret_cast<const shortx8*>(vreg[it].data);
+    }
+  };
+
+  if (ws_ < we_) {
+    stage_load(ws_);
+    stage_write();
+  }
+  for (int64_t kv0 = ws_; kv0 < we_; kv0 += KVB) {
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // stage visible
+    const bool have_next = kv0 + KVB < we_;
+    if (have_next) stage_load(kv0 + KVB);  // global loads fly under compute
+
+    // ---- S^T = K * Q^T ----
+    floatx16 acc_s = {};
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int c = 0; c < KCH; ++c) {
+      frag kfrag = *reinterpret_cast<const frag*>(
+          reinterpret_cast<const char*>(Ks[wave]) +
+          swz256(lq * KROWB + (c * 16 + khalf) * 2));
+      acc_s = mfma_ab_frag<T>::mma32(kfrag, qf[c], acc_s);
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- mask + base-2 logits (decode: all q rows at position kv_len-1,
+    // so the bounds are the same for every row; soft-cap/ALiBi are cheap
+    // per-element ops on the 16 in-lane values) ----
+    float pr[16];
+    if (kv0 + KVB <= we_ && cap <= 0.f && !p.alibi) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) pr[r] = acc_s[r] * scale2;
+    } else {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int64_t kv = kv0 + mfma32_cd_row(r, lane);
+        float sv = acc_s[r] * p.sm_scale;
+        if (cap > 0.f) sv = cap * tanhf(sv / cap);
+        if (p.alibi) sv -= slope * (float)(kv_len - 1 - kv);
+        sv *= kLog2e;
+        pr[r] = (kv < we_) ? sv : -INFINITY;
+      }
+    }
+
+    // ---- online softmax (defer-max; exchange with lane^32) ----
+    float tmax = pr[0];
+#pragma unroll
+    for (int r = 1; r < 16; ++r) tmax = fmaxf(tmax, pr[r]);
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    bool defer = m_run != -INFINITY && __all(tmax - m_run <= 8.f);
+    float m_new = defer ? m_run : fmaxf(m_run, tmax);
+    float f, psum = 0.f;
+    if (m_new == -INFINITY) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) pr[r] = 0.f;
+      f = 1.f;
+    } else {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        pr[r] = (pr[r] == -INFINITY) ? 0.f : __builtin_exp2f(pr[r] - m_new);
+        psum += pr[r];
+      }
+      f = defer ? 1.f : __builtin_exp2f(m_run - m_new);
+    }
+    d_run = d_run * f + psum;
+    if (!defer) {
+      m_run = m_new;
+#pragma unroll
+      for (int i = 0; i < DT; ++i) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) acc_o[i][r] *= f;
+      }
+    }
+
+    // ---- P^T fragments: pack + lazy half-exchange (prefill idiom) ----
+    uint32_t W[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      asm("v_cvt_pk_bf16_f32 %0, %1, %2"
+          : "=v"(W[j])
+          : "v"(pr[2 * j]), "v"(pr[2 * j + 1]));
+    }
+    uint32_t X[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) X[j] = (uint32_t)__shfl_xor((int)W[j], 32, 64);
+    const bool hiH = (lane >> 5) != 0;
+    uint32_t b0[4], b1[4];
+    b0[0] = hiH ? X[2] : W[0];
+    b0[1] = hiH ? X[3] : W[1];
+    b0[2] = hiH ? W[2] : X[0];
+    b0[3] = hiH ? W[3] : X[1];
+    b1[0] = hiH ? X[6] : W[4];
+    b1[1] = hiH ? X[7] : W[5];
+    b1[2] = hiH ? W[6] : X[4];
+    b1[3] = hiH ? W[7] : X[5];
+
+    // ---- O^T += V^T * P^T via HW transpose reads ----
+    {
+      const uint32_t vb = (uint32_t)(uintptr_t)Vs[wave] + (uint32_t)(lane & 15) * 8;
+      const uint32_t tdsel = ((lane >> 4) & 1);
+      const int kvb = khalf;  // 0 or 8: kv sub-rows of this half
+#pragma unroll
+      for (int i = 0; i < DT; ++i) {
+        uint32_t a00 = vb + (((kvb >> 2)) * (HEAD_DIM / 16) + i * 2 + tdsel) *
+                               (VTILE_STRIDE * 2);
+        b16x4 r00 = ds_read_tr16(a00);
+        b16x4 r01 = ds_read_tr16(a00 + (HEAD_DIM / 16) * (VTILE_STRIDE * 2));
+        uint32_t a10 = a00 + 4 * (HEAD_DIM / 16) * (VTILE_STRIDE * 2);
+        b16x4 r10 = ds_read_tr16(a10);
+        b16x4 r11 = ds_read_tr16(a10 + (HEAD_DIM / 16) * (VTILE_STRIDE * 2));
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+        union {
+          b16x4 h[2];
+          frag f_;
+        } u0, u1;
+        u0.h[0] = r00;
+        u0.h[1] = r01;
+        u1.h[0] = r10;
+        u1.h[1] = r11;
+        acc_o[i] = mfma_ab_frag<T>::mma32(u0.f_, *reinterpret_cast<frag*>(b0),
+                                          acc_o[i]);
+        acc_o[i] = mfma_ab_frag<T>::mma32(u1.f_, *reinterpret_cast<frag*>(b1),
+                                          acc_o[i]);
+      }
+    }
+    if (have_next) stage_write();
+  }
+
+  // ---- per-wave state -> LDS merge arrays ----
+  float d_full = d_run + __shfl_xor(d_run, 32, 64);
+  if (lq < QROWS) {
+#pragma unroll
+    for (int i = 0; i < DT; ++i) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int d = i * 32 + mfma32_cd_row(r, lane);
+        lds_o[wave][lq][d] = acc_o[i][r];
+      }
+    }
+    if ((lane >> 5) == 0) {
+      lds_m[wave][lq] = m_run;
+      lds_d[wave][lq] = d_full;
+    }
+  }
+  __syncthreads();
+
+  // ---- cross-wave merge + direct store ----
+  T* obase = (T*)p.o + (int64_t)req * p.o_stride_n;
+  for (int idx = tid; idx < QROWS * HEAD_DIM; idx += WAVES * 64) {
+    int g = idx / HEAD_DIM;
+    int d = idx % HEAD_DIM;
+    float m_star = -INFINITY;
+#pragma unroll
+    for (int w = 0; w < WAVES; ++w) m_star = fmaxf(m_star, lds_m[w][g]);
+    float d_sum = 0.f, o_sum = 0.f;
+    if (m_star != -INFINITY) {
+#pragma unroll
+      for (int w = 0; w < WAVES; ++w) {
+        float sw = __builtin_exp2f(lds_m[w][g] - m_star);
+        d_sum += lds_d[w][g] * sw;
+        o_sum += lds_o[w][g][d] * sw;
+      }
+    }
+    int qh = kv_head * GROUP + g;
+    float inv_d = d_sum > 0.f ? 1.f / d_sum : 0.f;
+    obase[(int64_t)qh * p.o_stride_h + d] = from_f32<T>(o_sum * inv_d);
+    if (d == 0 && p.lse)
+      p.lse[(int64_t)req * p.num_qo_heads + qh] =
+          d_sum > 0.f ? m_star + __builtin_log2f(d_sum) : -INFINITY;
+  }
+}
+
+template <typename T>
+hipError_t decode_mfma_dispatch(DecodeParams& p, hipStream_t stream) {
+  int group = p.num_qo_heads / p.num_kv_heads;
+  dim3 g((uint32_t)p.batch, (uint32_t)p.num_kv_heads), blk(WAVES * 64);
+#define LAUNCH_M(HD, G) \
+  hipLaunchKernelGGL((decode_mfma_kernel<T, HD, G>), g, blk, 0, stream, p)
+  if (p.head_dim == 128) {
+    switch (group) {
+      case 8: LAUNCH_M(128, 8); break;
+      case 16: LAUNCH_M(128, 16); break;
+      case 32: LAUNCH_M(128, 32); break;
+      default: return hipErrorInvalidValue;
+    }
+  } else if (p.head_dim == 64) {
+    switch (group) {
+      case 8: LAUNCH_M(64, 8); break;
+      case 16: LAUNCH_M(64, 16); break;
+      case 32: LAUNCH_M(64, 32); break;
+      default: return hipErrorInvalidValue;
+    }
+  } else {
+    return hipErrorInvalidValue;
+  }
+#undef LAUNCH_M
+  return hipGetLastError();
+}
+
+}  // namespace fi
+
+extern "C" hipError_t fi_decode_mfma(int dtype, fi::DecodeParams* p,
+                                     hipStream_t stream) {
+  if (p->batch == 0) return hipSuccess;
+  switch (dtype) {
+    case 0: return fi::decode_mfma_dispatch<fi::bf16>(*p, stream);
+    case 1: return fi::decode_mfma_dispatch<fi::fp16>(*p, stream);
+  }
+  return hipErrorInvalidValue;
+}

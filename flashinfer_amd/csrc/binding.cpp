@@ -47,6 +47,7 @@ hipError_t fi_batch_decode(int dtype, int kv_dtype, fi_ext::DecodeParams* p,
                            hipStream_t stream);
 hipError_t fi_batch_decode_fused(int dtype, int kv_dtype, fi_ext::DecodeParams* p,
                                  hipStream_t stream);
+hipError_t fi_decode_mfma(int dtype, fi_ext::DecodeParams* p, hipStream_t stream);
 hipError_t fi_gemm_nt(int dtype, const void* A, const void* B, void* C, int M, int N,
                       int K, int64_t lda, int64_t ldb, int64_t ldc, float alpha,
                       hipStream_t stream);
@@ -409,6 +410,45 @@ void batch_decode_fused_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache
   p.batch = q.size(0);
   check_hip(fi_batch_decode_fused(dtype_code(q), dtype_code(k_cache), &p, cur_stream(q)),
             "fi_batch_decode_fused");
+}
+
+// MFMA fused decode (GQA group >= 8): same argument surface as the fused
+// vector run, different kernel family (csrc/attention/decode_mfma.hip).
+void batch_decode_mfma_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
+                           at::Tensor kv_indices, at::Tensor kv_indptr,
+                           at::Tensor kv_last_page_len, int64_t layout,
+                           at::Tensor out, c10::optional<at::Tensor> lse,
+                           double sm_scale, double logits_soft_cap,
+                           int64_t window_left, bool alibi) {
+  TORCH_CHECK(q.is_cuda() && q.dim() == 3, "q must be [batch, num_qo_heads, head_dim]");
+  TORCH_CHECK(q.stride(2) == 1 && out.stride(2) == 1);
+  fi_ext::DecodeParams p{};
+  p.q = q.data_ptr();
+  p.k_data = k_cache.data_ptr();
+  p.v_data = v_cache.data_ptr();
+  p.kv_indices = kv_indices.data_ptr<int32_t>();
+  p.kv_indptr = kv_indptr.data_ptr<int32_t>();
+  p.kv_last_page_len = kv_last_page_len.data_ptr<int32_t>();
+  int page_size = layout == 0 ? k_cache.size(1) : k_cache.size(2);
+  p.page_size = fi::uint_fastdiv((uint32_t)page_size);
+  p.num_kv_heads = layout == 0 ? k_cache.size(2) : k_cache.size(1);
+  p.num_qo_heads = q.size(1);
+  p.head_dim = q.size(2);
+  p.stride_page = k_cache.stride(0);
+  p.stride_n = layout == 0 ? k_cache.stride(1) : k_cache.stride(2);
+  p.stride_h = layout == 0 ? k_cache.stride(2) : k_cache.stride(1);
+  p.q_stride_n = q.stride(0);
+  p.q_stride_h = q.stride(1);
+  p.sm_scale = (float)sm_scale;
+  p.logits_soft_cap = (float)logits_soft_cap;
+  p.window_left = (int)window_left;
+  p.alibi = alibi ? 1 : 0;
+  p.o = out.data_ptr();
+  p.lse = lse.has_value() ? lse->data_ptr<float>() : nullptr;
+  p.o_stride_n = out.stride(0);
+  p.o_stride_h = out.stride(1);
+  p.batch = q.size(0);
+  check_hip(fi_decode_mfma(dtype_code(q), &p, cur_stream(q)), "fi_decode_mfma");
 }
 
 // ---------------- prefill ----------------
@@ -980,6 +1020,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("merge_state_in_place", &merge_state_in_place);
   m.def("batch_decode_run", &batch_decode_run);
   m.def("batch_decode_fused_run", &batch_decode_fused_run);
+  m.def("batch_decode_mfma_run", &batch_decode_mfma_run);
   m.def("gemm_nt", &gemm_nt);
   m.def("batch_prefill_run", &batch_prefill_run);
   m.def("softmax", &softmax_op);

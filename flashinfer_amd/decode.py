@@ -36,6 +36,13 @@ _TARGET_BLOCKS = 8192  # 1-wave blocks: ~32 per CU keeps enough loads in flight
 # crossover on MI355X (profiles/README r02 fused-decode entry).
 _FUSED_MAX_KV = 2048
 
+# MFMA fused decode (GQA group >= 8): one workgroup per (req, kv_head), the
+# whole group as the 32-col q dimension of 32x32 MFMA tiles, 4 barrier-free
+# waves splitting the KV, in-LDS merge. The small-batch/short-kv GQA shape
+# where both the vector kernel (8 dot+shfl chains per K read) and the
+# prefill-reuse path (94%-padded 128-row tiles) lose.
+_MFMA_MAX_KV = 4096
+
 
 def _plan_chunks(kv_lens, num_kv_heads: int, page_size: int,
                  fixed_split_size=None, disable_split_kv=False):
@@ -159,14 +166,23 @@ class BatchDecodeWithPagedKVCacheWrapper:
 
         fused_ok = (
             head_dim in (64, 128, 256)
-            and group in (1, 2, 4, 8)
-            and not (head_dim == 256 and group == 8)
+            and group in (1, 2, 4)  # GQA >= 8 goes to the MFMA decode kernel
             and max_len <= _FUSED_MAX_KV
             and fixed_split_size is None
             and self._use_tensor_cores is not True
         )
+        mfma_ok = (
+            head_dim in (64, 128)
+            and group in (8, 16, 32)
+            and kv_dt == q_data_type
+            and q_data_type in (torch.bfloat16, torch.float16)
+            and max_len <= _MFMA_MAX_KV
+            and fixed_split_size is None
+            and self._use_tensor_cores is not True
+            and self._use_tensor_cores is not False
+        )
         self._tc = (
-            not fused_ok
+            not fused_ok and not mfma_ok
             and kv_dt == q_data_type
             and (
                 self._use_tensor_cores is True
@@ -174,7 +190,8 @@ class BatchDecodeWithPagedKVCacheWrapper:
                     and fixed_split_size is None and not disable_split_kv)
             )
         )
-        self._fused = fused_ok and not self._tc
+        self._fused = (fused_ok or mfma_ok) and not self._tc
+        self._fused_mfma = mfma_ok and not self._tc
         if self._tc:
             from .prefill import BatchPrefillWithPagedKVCacheWrapper
 
@@ -316,7 +333,10 @@ class BatchDecodeWithPagedKVCacheWrapper:
             if need_lse and lse is None:
                 lse = torch.empty((pi["batch"], pi["num_qo_heads"]),
                                   dtype=torch.float32, device=q.device)
-            get_ext().batch_decode_fused_run(
+            run_fn = (get_ext().batch_decode_mfma_run
+                      if getattr(self, "_fused_mfma", False)
+                      else get_ext().batch_decode_fused_run)
+            run_fn(
                 q, k_cache, v_cache,
                 self._indices_d, self._indptr_d, self._last_page_len_d,
                 layout_code(self._kv_layout), out, lse if need_lse else None,

@@ -145,12 +145,89 @@ def bench_fp8_gemm(M=4096, N=4096, K=4096):
     print(f"fp8 groupwise gemm {M}x{N}x{K}: {t*1e3:.2f} ms  {2*M*N*K/t/1e12:.0f} TFLOPS")
 
 
+def bench_prefill_ragged_192(bs=16, s=1024, Hq=128, Hkv=128, causal=True):
+    """DeepSeek MHA ragged prefill, hd_qk=192/hd_vo=128 (BASELINE.md ragged
+    row; B200 fa2 = 213.7 TF)."""
+    torch.manual_seed(0)
+    qo_indptr = torch.arange(0, (bs + 1) * s, s, dtype=torch.int32, device="cuda")
+    q = torch.randn(bs * s, Hq, 192, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(bs * s, Hkv, 192, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(bs * s, Hkv, 128, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(256 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchPrefillWithRaggedKVCacheWrapper(ws, "NHD")
+    w.plan(qo_indptr, qo_indptr.clone(), Hq, Hkv, 192, head_dim_vo=128,
+           causal=causal, q_data_type=torch.bfloat16)
+    out = torch.empty(bs * s, Hq, 128, dtype=torch.bfloat16, device="cuda")
+    t = timeit(lambda: w.run(q, k, v, out=out))
+    fl = bs * Hq * s * s * 2 * (192 + 128)
+    if causal:
+        fl /= 2
+    print(f"prefill ragged 192/128 bs={bs} s={s} H={Hq} causal={causal}: "
+          f"{t*1e6:.1f} us  {fl/t/1e12:.1f} TFLOPS")
+
+
+def bench_prefill_splitkv(qo=16, kv=65536, Hq=32, Hkv=8, D=128):
+    """bs=1 short-q/long-kv prefill: split-KV must fill the chip."""
+    torch.manual_seed(0)
+    qo_indptr = torch.tensor([0, qo], dtype=torch.int32, device="cuda")
+    kv_indptr = torch.tensor([0, kv], dtype=torch.int32, device="cuda")
+    q = torch.randn(qo, Hq, D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(kv, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(kv, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(512 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchPrefillWithRaggedKVCacheWrapper(ws, "NHD")
+    w.plan(qo_indptr, kv_indptr, Hq, Hkv, D, causal=False,
+           q_data_type=torch.bfloat16)
+    out = torch.empty_like(q)
+    t = timeit(lambda: w.run(q, k, v, out=out))
+    bytes_kv = kv * Hkv * D * 2 * 2
+    print(f"prefill splitkv qo={qo} kv={kv} (split={w._split}): {t*1e6:.1f} us  "
+          f"{bytes_kv/t/1e12:.2f} TB/s")
+
+
+def bench_decode_modes(bs=16, kv=1024, Hq=64, Hkv=8, D=128, page=16):
+    """A/B the three decode shapes (fused / split-vector / tensor-core) at one
+    config — the BASELINE small-batch decode latency row."""
+    torch.manual_seed(0)
+    pages_per = (kv + page - 1) // page
+    kv_indptr = torch.arange(0, (bs + 1) * pages_per, pages_per,
+                             dtype=torch.int32, device="cuda")
+    npages = bs * pages_per
+    kv_indices = torch.randperm(npages, dtype=torch.int32, device="cuda")
+    last_page = torch.full((bs,), (kv - 1) % page + 1, dtype=torch.int32,
+                           device="cuda")
+    k_cache = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v_cache = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    q = torch.randn(bs, Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(512 << 20, dtype=torch.uint8, device="cuda")
+    bytes_kv = bs * kv * Hkv * D * 2 * 2
+    out = torch.empty_like(q)
+    for mode, kwargs in (("auto", {}), ("vector", dict(use_tensor_cores=False)),
+                         ("tc", dict(use_tensor_cores=True))):
+        from flashinfer_amd import decode as _dec
+        saved = _dec._FUSED_MAX_KV
+        if mode == "vector":
+            _dec._FUSED_MAX_KV = 0  # force the split vector path
+        w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD", **kwargs)
+        w.plan(kv_indptr, kv_indices, last_page, Hq, Hkv, D, page,
+               q_data_type=torch.bfloat16)
+        _dec._FUSED_MAX_KV = saved
+        t = timeit(lambda: w.run(q, (k_cache, v_cache), out=out))
+        tag = "fused" if getattr(w, "_fused", False) else (
+            "tc" if getattr(w, "_tc", False) else "vector")
+        print(f"decode_modes bs={bs} kv={kv} GQA{Hq}/{Hkv} [{mode}->{tag}]: "
+              f"{t*1e6:.1f} us  {bytes_kv/t/1e12:.2f} TB/s")
+
+
+
 if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "all"
     if which in ("all", "prefill"):
         bench_prefill()
         bench_prefill(bs=1, s=8192)
         bench_prefill(bs=16, s=1024, causal=False)
+        bench_prefill_ragged_192()
+        bench_prefill_splitkv()
     if which in ("all", "decode"):
         bench_decode()
         bench_decode(bs=16, kv=1024)
@@ -158,6 +235,12 @@ if __name__ == "__main__":
         # GQA-8 (Llama-70B class): auto tensor-core route
         bench_decode(bs=256, kv=8192, Hq=64, Hkv=8)
         bench_decode(bs=16, kv=1024, Hq=64, Hkv=8)
+    if which in ("all", "modes", "decode"):
+        bench_decode_modes()                       # BASELINE small-batch row
+        bench_decode_modes(bs=16, kv=1024, Hq=32, Hkv=8)
+        bench_decode_modes(bs=64, kv=1024, Hq=64, Hkv=8)
+        bench_decode_modes(bs=256, kv=2048, Hq=64, Hkv=8)
+        bench_decode_modes(bs=32, kv=512, Hq=32, Hkv=8)
     if which in ("all", "gemm"):
         bench_gemm(4096)
         bench_gemm(8192)

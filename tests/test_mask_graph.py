@@ -201,9 +201,9 @@ def test_tensor_core_decode_graph_capture():
     q = torch.randn(bs, Hq, D, dtype=torch.bfloat16, device="cuda")
     out = torch.empty_like(q)
     ws = torch.empty(64 << 20, dtype=torch.uint8, device="cuda")
-    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD", use_tensor_cores=True)
     w.plan(indptr, indices, lpl, Hq, Hkv, D, page, q_data_type=torch.bfloat16)
-    assert w._tc  # the GQA-8 auto route must be active
+    assert w._tc  # explicit opt-in must pin the tensor-core route
     w.run(q, (kc, vc), out=out)
     torch.cuda.synchronize()
     ref = out.clone()
@@ -214,3 +214,35 @@ def test_tensor_core_decode_graph_capture():
     g.replay()
     torch.cuda.synchronize()
     torch.testing.assert_close(out.float(), ref.float())
+
+
+def test_fused_decode_graph_capture():
+    """Short-kv decode (fused whole-request route) captures into a hipGraph."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(5)
+    Hq, Hkv, D, page = 64, 8, 128, 16
+    bs, pages_per = 8, 8
+    npages = bs * pages_per
+    kc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    indptr = torch.arange(0, npages + 1, pages_per, dtype=torch.int32,
+                          device="cuda")
+    indices = torch.arange(npages, dtype=torch.int32, device="cuda")
+    lpl = torch.full((bs,), page, dtype=torch.int32, device="cuda")
+    q = torch.randn(bs, Hq, D, dtype=torch.bfloat16, device="cuda")
+    out = torch.empty_like(q)
+    ws = torch.empty(64 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(indptr, indices, lpl, Hq, Hkv, D, page, q_data_type=torch.bfloat16)
+    assert w._fused
+    w.run(q, (kc, vc), out=out)
+    torch.cuda.synchronize()
+    ref = out.clone()
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        w.run(q, (kc, vc), out=out)
+    out.zero_()
+    g.replay()
+    torch.cuda.synchronize()
+    torch.testing.assert_close(out, ref)

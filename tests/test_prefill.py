@@ -177,3 +177,129 @@ def test_prefill_with_empty_request():
     logits = logits.masked_fill(mask[None], float("-inf"))
     ref = torch.einsum("hml,lhd->mhd", torch.softmax(logits, -1), vv)
     torch.testing.assert_close(out[32:].float(), ref, atol=3e-2, rtol=3e-2)
+
+
+# ------------- head_dim_qk 192 / head_dim_vo 128 (DeepSeek MHA) -------------
+
+def ref_attn_vo(q, k, v, causal=False, sm_scale=None):
+    """Reference with head_dim_vo != head_dim_qk: v [L, Hkv, Dvo]."""
+    M, Hq, D = q.shape
+    L, Hkv, _ = k.shape
+    g = Hq // Hkv
+    qf = q.float().transpose(0, 1)
+    kf = k.float().repeat_interleave(g, dim=1).transpose(0, 1)
+    vf = v.float().repeat_interleave(g, dim=1).transpose(0, 1)
+    scale = sm_scale if sm_scale is not None else 1 / math.sqrt(D)
+    logits = qf @ kf.transpose(-1, -2) * scale
+    if causal:
+        qpos = torch.arange(M, device=q.device)[:, None]
+        kpos = torch.arange(L, device=q.device)[None, :]
+        logits = logits.masked_fill((kpos > qpos + (L - M))[None], float("-inf"))
+    p = torch.softmax(logits, dim=-1)
+    return (p @ vf).transpose(0, 1)
+
+
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("Hq,Hkv", [(16, 16), (16, 4)])
+def test_ragged_prefill_hd192_128(causal, Hq, Hkv):
+    """DeepSeek MHA shape (192 qk / 128 vo) on the ragged path — the
+    BASELINE.md ragged-prefill row config (reference
+    benchmarks/samples/sample_testlist_output.csv:5)."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    qo_lens = [128, 333, 64]
+    kv_lens = [128, 333, 517]
+    nnz_q, nnz_kv = sum(qo_lens), sum(kv_lens)
+    q = torch.randn(nnz_q, Hq, 192, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(nnz_kv, Hkv, 192, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(nnz_kv, Hkv, 128, dtype=torch.bfloat16, device="cuda")
+    qo_indptr = torch.tensor([0] + list(torch.cumsum(torch.tensor(qo_lens), 0)),
+                             dtype=torch.int32, device="cuda")
+    kv_indptr = torch.tensor([0] + list(torch.cumsum(torch.tensor(kv_lens), 0)),
+                             dtype=torch.int32, device="cuda")
+    ws = torch.empty(128 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchPrefillWithRaggedKVCacheWrapper(ws, "NHD")
+    w.plan(qo_indptr, kv_indptr, Hq, Hkv, 192, head_dim_vo=128, causal=causal,
+           q_data_type=torch.bfloat16)
+    out = w.run(q, k, v)
+    assert out.shape == (nnz_q, Hq, 128)
+    for b in range(3):
+        qs, qe = int(qo_indptr[b]), int(qo_indptr[b + 1])
+        ks, ke = int(kv_indptr[b]), int(kv_indptr[b + 1])
+        ref = ref_attn_vo(q[qs:qe], k[ks:ke], v[ks:ke], causal=causal)
+        torch.testing.assert_close(out[qs:qe].float(), ref, atol=3e-2, rtol=3e-2)
+
+
+# ---------------------------- prefill split-KV ----------------------------
+
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("qo_len,kv_len", [(16, 8192), (1, 16384), (128, 4096)])
+def test_prefill_split_kv_ragged(causal, qo_len, kv_len):
+    """Short-q/long-kv prefill must split KV across workgroups and merge
+    (reference scheduler.cuh:545 role) — and match the fp32 reference."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(0)
+    Hq, Hkv, D = 8, 8, 128
+    q = torch.randn(qo_len, Hq, D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(kv_len, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(kv_len, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    qo_indptr = torch.tensor([0, qo_len], dtype=torch.int32, device="cuda")
+    kv_indptr = torch.tensor([0, kv_len], dtype=torch.int32, device="cuda")
+    ws = torch.empty(256 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchPrefillWithRaggedKVCacheWrapper(ws, "NHD")
+    w.plan(qo_indptr, kv_indptr, Hq, Hkv, D, causal=causal,
+           q_data_type=torch.bfloat16)
+    assert w._split, "short-q/long-kv plan must split KV"
+    out, lse = w.run(q, k, v, return_lse=True)
+    ref = ref_attn(q, k, v, causal=causal)
+    torch.testing.assert_close(out.float(), ref, atol=3e-2, rtol=3e-2)
+    # lse sanity (non-causal row 0 vs torch)
+    kf = k.float()
+    logits = torch.einsum("hd,lhd->hl", q[0].float(), kf) / math.sqrt(D)
+    if causal:
+        kpos = torch.arange(kv_len, device=q.device)
+        logits = logits.masked_fill((kpos > (kv_len - qo_len))[None],
+                                    float("-inf"))
+    ref_lse = torch.logsumexp(logits, -1) / math.log(2)
+    torch.testing.assert_close(lse[0], ref_lse, atol=2e-2, rtol=2e-2)
+
+
+def test_prefill_split_kv_paged():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(1)
+    Hq, Hkv, D, page = 32, 8, 128, 16
+    qo_lens = [4, 2]
+    kv_lens = [6000, 3111]
+    pages_per = [(L + page - 1) // page for L in kv_lens]
+    qo_indptr = torch.tensor([0, 4, 6], dtype=torch.int32, device="cuda")
+    kv_indptr = torch.tensor([0] + list(torch.cumsum(torch.tensor(pages_per), 0)),
+                             dtype=torch.int32, device="cuda")
+    npages = int(kv_indptr[-1])
+    kv_indices = torch.randperm(npages, dtype=torch.int32, device="cuda")
+    lpl = torch.tensor([(L - 1) % page + 1 for L in kv_lens], dtype=torch.int32,
+                       device="cuda")
+    kc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    q = torch.randn(6, Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(256 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchPrefillWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(qo_indptr, kv_indptr, kv_indices, lpl, Hq, Hkv, D, page,
+           causal=True, q_data_type=torch.bfloat16)
+    assert w._split
+    out = w.run(q, (kc, vc))
+    for b in range(2):
+        qs, qe = int(qo_indptr[b]), int(qo_indptr[b + 1])
+        toks_k, toks_v = [], []
+        base = [0, pages_per[0]][b]
+        for p_ in range(pages_per[b]):
+            pg = int(kv_indices[base + p_])
+            n = min(page, kv_lens[b] - p_ * page)
+            toks_k.append(kc[pg, :n])
+            toks_v.append(vc[pg, :n])
+        kk = torch.cat(toks_k, 0)
+        vv = torch.cat(toks_v, 0)
+        ref = ref_attn(q[qs:qe], kk, vv, causal=True)
+        torch.testing.assert_close(out[qs:qe].float(), ref, atol=3e-2, rtol=3e-2)
