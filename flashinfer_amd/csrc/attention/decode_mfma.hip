@@ -32,7 +32,10 @@ namespace fi {
 
 namespace {
 constexpr int KVB = 32;  // kv tokens per tile (one 32x32 S^T MFMA pass)
-constexpr int WAVES = 4;
+// 8 waves/WG = 2 waves per SIMD at occupancy 1: co-resident waves hide the
+// MFMA/transcendental/DS latency a lone wave leaves exposed (4-wave variant
+// measured 46 us vs this shape at the BASELINE config)
+constexpr int WAVES = 8;
 constexpr float kLog2e = 1.4426950408889634f;
 
 typedef __attribute__((ext_vector_type(4))) __bf16 b16x4;
@@ -53,12 +56,22 @@ __global__ __launch_bounds__(WAVES * 64, 1) void decode_mfma_kernel(DecodeParams
   constexpr int VTILE_STRIDE = 72;     // elems (144 B) — tr-read subtile pad
 
   // per-wave LDS regions (single-buffered; in-wave LDS ordering makes the
-  // read-then-overwrite safe) + the cross-wave merge arrays
+  // read-then-overwrite safe). The cross-wave merge state is aliased into
+  // each wave's OWN stage region after its loop ends (the one barrier
+  // separates the phases) — 8 waves of K+V stage already fill the 160 KB
+  // budget at HEAD_DIM 128.
   __shared__ T Ks[WAVES][KVB * HEAD_DIM];
   __shared__ T Vs[WAVES][(KVB / 4) * (HEAD_DIM / 16) * VTILE_STRIDE];
-  __shared__ float lds_o[WAVES][QROWS][HEAD_DIM];
   __shared__ float lds_m[WAVES][QROWS];
   __shared__ float lds_d[WAVES][QROWS];
+  // merge row (g) of wave w: first 16 q rows live in Ks[w], the rest in Vs[w]
+  auto merge_row = [&](int w, int g) -> float* {
+    return g < 16 ? reinterpret_cast<float*>(&Ks[w][0]) + g * HEAD_DIM
+                  : reinterpret_cast<float*>(&Vs[w][0]) + (g - 16) * HEAD_DIM;
+  };
+  static_assert(QROWS <= 16 || (KVB / 4) * (HEAD_DIM / 16) * VTILE_STRIDE * 2 >=
+                                   (QROWS - 16) * HEAD_DIM * 4,
+                "merge rows exceed the aliased stage LDS");
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -89,19 +102,12 @@ __global__ __launch_bounds__(WAVES * 64, 1) void decode_mfma_kernel(DecodeParams
   const T* vbase = (const T*)p.v_data;
 
   // ---- Q fragments (B-operand): lane l holds Q[q = l&31][k = khalf + c*16] —
-  // decode q rows are the GQA group of this kv head ----
+  // decode q rows are the GQA group of this kv head. Q is re-read from L1
+  // every tile (2 KB, resident) — the freed 32 VGPRs buy the depth-2 KV ring.
   using frag = typename mfma_ab_frag<T>::type;
-  frag qf[KCH];
-  {
-    const bool qvalid = lq < QROWS;
-    const T* qptr = (const T*)p.q + (int64_t)req * p.q_stride_n +
-                    (int64_t)(kv_head * GROUP + lq) * p.q_stride_h;
-#pragma unroll
-    for (int c = 0; c < KCH; ++c) {
-      if (qvalid) qf[c] = *reinterpret_cast<const frag*>(qptr + c * 16 + khalf);
-      else qf[c] = frag{};
-    }
-  }
+  const bool qvalid = lq < QROWS;
+  const T* qptr = (const T*)p.q + (int64_t)req * p.q_stride_n +
+                  (int64_t)(kv_head * GROUP + lq) * p.q_stride_h;
 
   float m_run = -INFINITY, d_run = 0.f;
   floatx16 acc_o[DT];
@@ -115,31 +121,34 @@ __global__ __launch_bounds__(WAVES * 64, 1) void decode_mfma_kernel(DecodeParams
   if (p.alibi)
     slope = __builtin_exp2f(-8.f * (kv_head * GROUP + lq + 1) / p.num_qo_heads);
 
-  // ---- register-staged K/V tile loads (8-elem units; 64 lanes cover a
-  // 32 x HEAD_DIM tile in KVB*HEAD_DIM/8/64 iterations) ----
+  // ---- register-staged K/V tile loads (one tile in regs while the staged
+  // tile computes; with 2 waves/SIMD the partner wave fills the vmcnt gaps —
+  // a 2-deep ring measured slower: the extra 64 VGPRs cost more than the
+  // depth bought) ----
   constexpr int S_ITER = KVB * HEAD_DIM / 8 / 64;
-  vec_t<T, 8> kreg[S_ITER], vreg[S_ITER];
-  auto stage_load = [&](int64_t kv0) {
+  vec_t<T, 8> kregA[S_ITER], vregA[S_ITER];
+  auto stage_load = [&](int64_t kv0, vec_t<T, 8>(&kr)[S_ITER],
+                        vec_t<T, 8>(&vr)[S_ITER]) {
 #pragma unroll
     for (int it = 0; it < S_ITER; ++it) {
       int u = lane + it * 64;
       int row = u / (HEAD_DIM / 8);
       int chunk8 = u % (HEAD_DIM / 8);
       int64_t kvpos = kv0 + row;
-      kreg[it].fill(0.f);
-      vreg[it].fill(0.f);
+      kr[it].fill(0.f);
+      vr[it].fill(0.f);
       if (kvpos < we_) {
         uint32_t pg, entry;
         p.page_size.divmod((uint32_t)kvpos, pg, entry);
         int64_t off = (int64_t)page_ids[pg] * p.stride_page +
                       (int64_t)kv_head * p.stride_h + (int64_t)entry * p.stride_n +
                       chunk8 * 8;
-        kreg[it].load(kbase + off);
-        vreg[it].load(vbase + off);
+        kr[it].load(kbase + off);
+        vr[it].load(vbase + off);
       }
     }
   };
-  auto stage_write = [&]() {
+  auto stage_write = [&](vec_t<T, 8>(&kr)[S_ITER], vec_t<T, 8>(&vr)[S_ITER]) {
 #pragma unroll
     for (int it = 0; it < S_ITER; ++it) {
       int u = lane + it * 64;
@@ -147,33 +156,27 @@ __global__ __launch_bounds__(WAVES * 64, 1) void decode_mfma_kernel(DecodeParams
       int chunk8 = u % (HEAD_DIM / 8);
       *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Ks[wave]) +
                                   swz256(row * KROWB + chunk8 * 16)) =
-          *reinterpret_cast<const shortx8*>(kreg[it].data);
+          *reinterpret_cast<const shortx8*>(kr[it].data);
       *reinterpret_cast<shortx8*>(
           reinterpret_cast<char*>(Vs[wave]) +
           ((row >> 2) * (HEAD_DIM / 16) + (chunk8 >> 1)) * (VTILE_STRIDE * 2) +
           (row & 3) * 32 + (chunk8 & 1) * 16) =
-          *reinterpret_cast<const shortx8*>(vreg[it].data);
+          *reinterpret_cast<const shortx8*>(vr[it].data);
     }
   };
 
-  if (ws_ < we_) {
-    stage_load(ws_);
-    stage_write();
-  }
-  for (int64_t kv0 = ws_; kv0 < we_; kv0 += KVB) {
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // stage visible
-    const bool have_next = kv0 + KVB < we_;
-    if (have_next) stage_load(kv0 + KVB);  // global loads fly under compute
-
-    // ---- S^T = K * Q^T ----
+  auto process_tile = [&](int64_t kv0) {
+    // ---- S^T = K * Q^T (Q frags re-read from L1 each tile) ----
     floatx16 acc_s = {};
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int c = 0; c < KCH; ++c) {
+      frag qf = qvalid ? *reinterpret_cast<const frag*>(qptr + c * 16 + khalf)
+                       : frag{};
       frag kfrag = *reinterpret_cast<const frag*>(
           reinterpret_cast<const char*>(Ks[wave]) +
           swz256(lq * KROWB + (c * 16 + khalf) * 2));
-      acc_s = mfma_ab_frag<T>::mma32(kfrag, qf[c], acc_s);
+      acc_s = mfma_ab_frag<T>::mma32(kfrag, qf, acc_s);
     }
     __builtin_amdgcn_s_setprio(0);
 
@@ -278,18 +281,35 @@ __global__ __launch_bounds__(WAVES * 64, 1) void decode_mfma_kernel(DecodeParams
                                           acc_o[i]);
       }
     }
-    if (have_next) stage_write();
+  };
+
+  // ---- main loop: wait stage -> issue tile n+1's global loads -> compute
+  // tile n from LDS -> ds_write tile n+1 (safe: in-wave LDS ops complete in
+  // issue order, so tile n's reads land before the overwrite) ----
+  if (ws_ < we_) {
+    stage_load(ws_, kregA, vregA);
+    stage_write(kregA, vregA);
+  }
+  for (int64_t kv0 = ws_; kv0 < we_; kv0 += KVB) {
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    const bool have_next = kv0 + KVB < we_;
+    if (have_next) stage_load(kv0 + KVB, kregA, vregA);
+    process_tile(kv0);
+    if (have_next) stage_write(kregA, vregA);
   }
 
-  // ---- per-wave state -> LDS merge arrays ----
+  // ---- per-wave state -> merge rows (aliased into this wave's stage LDS;
+  // every wave has finished reading its stage before writing here, and the
+  // barrier below orders the cross-wave reads) ----
   float d_full = d_run + __shfl_xor(d_run, 32, 64);
   if (lq < QROWS) {
+    float* orow = merge_row(wave, lq);
 #pragma unroll
     for (int i = 0; i < DT; ++i) {
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         int d = i * 32 + mfma32_cd_row(r, lane);
-        lds_o[wave][lq][d] = acc_o[i][r];
+        orow[d] = acc_o[i][r];
       }
     }
     if ((lane >> 5) == 0) {
@@ -313,7 +333,7 @@ __global__ __launch_bounds__(WAVES * 64, 1) void decode_mfma_kernel(DecodeParams
       for (int w = 0; w < WAVES; ++w) {
         float sw = __builtin_exp2f(lds_m[w][g] - m_star);
         d_sum += lds_d[w][g] * sw;
-        o_sum += lds_o[w][g][d] * sw;
+        o_sum += merge_row(w, g)[d] * sw;
       }
     }
     int qh = kv_head * GROUP + g;

@@ -213,8 +213,9 @@ def bench_decode_modes(bs=16, kv=1024, Hq=64, Hkv=8, D=128, page=16):
                q_data_type=torch.bfloat16)
         _dec._FUSED_MAX_KV = saved
         t = timeit(lambda: w.run(q, (k_cache, v_cache), out=out))
-        tag = "fused" if getattr(w, "_fused", False) else (
-            "tc" if getattr(w, "_tc", False) else "vector")
+        tag = ("mfma" if getattr(w, "_fused_mfma", False) else
+               "fused" if getattr(w, "_fused", False) else
+               "tc" if getattr(w, "_tc", False) else "vector")
         print(f"decode_modes bs={bs} kv={kv} GQA{Hq}/{Hkv} [{mode}->{tag}]: "
               f"{t*1e6:.1f} us  {bytes_kv/t/1e12:.2f} TB/s")
 
