@@ -62,6 +62,7 @@ __global__ __launch_bounds__(WAVES * 64, 1) void decode_mfma_kernel(DecodeParams
   // budget at HEAD_DIM 128.
   __shared__ T Ks[WAVES][KVB * HEAD_DIM];
   __shared__ T Vs[WAVES][(KVB / 4) * (HEAD_DIM / 16) * VTILE_STRIDE];
+  __shared__ T Qs[32 * HEAD_DIM];  // group q rows, staged once (swizzled)
   __shared__ float lds_m[WAVES][QROWS];
   __shared__ float lds_d[WAVES][QROWS];
   // merge row (g) of wave w: first 16 q rows live in Ks[w], the rest in Vs[w]
@@ -101,13 +102,31 @@ __global__ __launch_bounds__(WAVES * 64, 1) void decode_mfma_kernel(DecodeParams
   const T* kbase = (const T*)p.k_data;
   const T* vbase = (const T*)p.v_data;
 
-  // ---- Q fragments (B-operand): lane l holds Q[q = l&31][k = khalf + c*16] —
-  // decode q rows are the GQA group of this kv head. Q is re-read from L1
-  // every tile (2 KB, resident) — the freed 32 VGPRs buy the depth-2 KV ring.
+  // ---- Q staged ONCE into LDS (B-operand fragments re-read per tile via
+  // ds_read: lgkm-tracked, so they never wait on the in-flight HBM staging
+  // loads the way a per-tile GLOBAL re-read would — vmcnt is an in-order
+  // counter — and they cost no resident VGPRs the way qf registers would
+  // (measured: global re-read 33 us, register qf 108-132 B/lane spill) ----
   using frag = typename mfma_ab_frag<T>::type;
-  const bool qvalid = lq < QROWS;
-  const T* qptr = (const T*)p.q + (int64_t)req * p.q_stride_n +
-                  (int64_t)(kv_head * GROUP + lq) * p.q_stride_h;
+  {
+    constexpr int QUNITS = 32 * HEAD_DIM / 8;
+    for (int u = tid; u < QUNITS; u += WAVES * 64) {
+      int row = u / (HEAD_DIM / 8);
+      int chunk8 = u % (HEAD_DIM / 8);
+      vec_t<T, 8> qv;
+      if (row < QROWS) {
+        qv.load((const T*)p.q + (int64_t)req * p.q_stride_n +
+                (int64_t)(kv_head * GROUP + row) * p.q_stride_h + chunk8 * 8);
+      } else {
+        qv.fill(0.f);
+      }
+      *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Qs) +
+                                  (KROWB == 256 ? swz256(row * KROWB + chunk8 * 16)
+                                                : swz128(row * KROWB + chunk8 * 16))) =
+          *reinterpret_cast<const shortx8*>(qv.data);
+    }
+    __syncthreads();
+  }
 
   float m_run = -INFINITY, d_run = 0.f;
   floatx16 acc_o[DT];
@@ -122,60 +141,60 @@ __global__ __launch_bounds__(WAVES * 64, 1) void decode_mfma_kernel(DecodeParams
     slope = __builtin_exp2f(-8.f * (kv_head * GROUP + lq + 1) / p.num_qo_heads);
 
   // ---- register-staged K/V tile loads (one tile in regs while the staged
-  // tile computes; with 2 waves/SIMD the partner wave fills the vmcnt gaps —
-  // a 2-deep ring measured slower: the extra 64 VGPRs cost more than the
-  // depth bought) ----
+  // tile computes; 2 waves/SIMD co-residency fills the vmcnt gaps) ----
   constexpr int S_ITER = KVB * HEAD_DIM / 8 / 64;
   vec_t<T, 8> kregA[S_ITER], vregA[S_ITER];
-  auto stage_load = [&](int64_t kv0, vec_t<T, 8>(&kr)[S_ITER],
-                        vec_t<T, 8>(&vr)[S_ITER]) {
+  auto stage_load = [&](int64_t kv0) {
 #pragma unroll
     for (int it = 0; it < S_ITER; ++it) {
       int u = lane + it * 64;
       int row = u / (HEAD_DIM / 8);
       int chunk8 = u % (HEAD_DIM / 8);
       int64_t kvpos = kv0 + row;
-      kr[it].fill(0.f);
-      vr[it].fill(0.f);
+      kregA[it].fill(0.f);
+      vregA[it].fill(0.f);
       if (kvpos < we_) {
         uint32_t pg, entry;
         p.page_size.divmod((uint32_t)kvpos, pg, entry);
         int64_t off = (int64_t)page_ids[pg] * p.stride_page +
                       (int64_t)kv_head * p.stride_h + (int64_t)entry * p.stride_n +
                       chunk8 * 8;
-        kr[it].load(kbase + off);
-        vr[it].load(vbase + off);
+        kregA[it].load(kbase + off);
+        vregA[it].load(vbase + off);
       }
     }
   };
-  auto stage_write = [&](vec_t<T, 8>(&kr)[S_ITER], vec_t<T, 8>(&vr)[S_ITER]) {
+  auto stage_write = [&]() {
 #pragma unroll
     for (int it = 0; it < S_ITER; ++it) {
       int u = lane + it * 64;
       int row = u / (HEAD_DIM / 8);
       int chunk8 = u % (HEAD_DIM / 8);
       *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Ks[wave]) +
-                                  swz256(row * KROWB + chunk8 * 16)) =
-          *reinterpret_cast<const shortx8*>(kr[it].data);
+                                  (KROWB == 256 ? swz256(row * KROWB + chunk8 * 16)
+                                                : swz128(row * KROWB + chunk8 * 16))) =
+          *reinterpret_cast<const shortx8*>(kregA[it].data);
       *reinterpret_cast<shortx8*>(
           reinterpret_cast<char*>(Vs[wave]) +
           ((row >> 2) * (HEAD_DIM / 16) + (chunk8 >> 1)) * (VTILE_STRIDE * 2) +
           (row & 3) * 32 + (chunk8 & 1) * 16) =
-          *reinterpret_cast<const shortx8*>(vr[it].data);
+          *reinterpret_cast<const shortx8*>(vregA[it].data);
     }
   };
 
   auto process_tile = [&](int64_t kv0) {
-    // ---- S^T = K * Q^T (Q frags re-read from L1 each tile) ----
+    // ---- S^T = K * Q^T (both operand frags from LDS) ----
     floatx16 acc_s = {};
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int c = 0; c < KCH; ++c) {
-      frag qf = qvalid ? *reinterpret_cast<const frag*>(qptr + c * 16 + khalf)
-                       : frag{};
+      auto swz = [&](uint32_t x) { return KROWB == 256 ? swz256(x) : swz128(x); };
+      frag qf = *reinterpret_cast<const frag*>(
+          reinterpret_cast<const char*>(Qs) +
+          swz(lq * KROWB + (c * 16 + khalf) * 2));
       frag kfrag = *reinterpret_cast<const frag*>(
           reinterpret_cast<const char*>(Ks[wave]) +
-          swz256(lq * KROWB + (c * 16 + khalf) * 2));
+          swz(lq * KROWB + (c * 16 + khalf) * 2));
       acc_s = mfma_ab_frag<T>::mma32(kfrag, qf, acc_s);
     }
     __builtin_amdgcn_s_setprio(0);
@@ -283,19 +302,18 @@ __global__ __launch_bounds__(WAVES * 64, 1) void decode_mfma_kernel(DecodeParams
     }
   };
 
-  // ---- main loop: wait stage -> issue tile n+1's global loads -> compute
-  // tile n from LDS -> ds_write tile n+1 (safe: in-wave LDS ops complete in
-  // issue order, so tile n's reads land before the overwrite) ----
+  // ---- main loop: wait stage(n) -> issue loads(n+1) -> compute(n) ->
+  // ds_write(n+1) (in-wave LDS ordering makes the single buffer safe) ----
   if (ws_ < we_) {
-    stage_load(ws_, kregA, vregA);
-    stage_write(kregA, vregA);
+    stage_load(ws_);
+    stage_write();
   }
   for (int64_t kv0 = ws_; kv0 < we_; kv0 += KVB) {
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     const bool have_next = kv0 + KVB < we_;
-    if (have_next) stage_load(kv0 + KVB, kregA, vregA);
+    if (have_next) stage_load(kv0 + KVB);
     process_tile(kv0);
-    if (have_next) stage_write(kregA, vregA);
+    if (have_next) stage_write();
   }
 
   // ---- per-wave state -> merge rows (aliased into this wave's stage LDS;
