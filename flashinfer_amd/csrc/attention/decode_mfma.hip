@@ -91,12 +91,23 @@ __global__ __launch_bounds__(WAVES * 64, 1) void decode_mfma_kernel(DecodeParams
     int64_t w = kv_len - 1 - p.window_left;
     if (w > 0) kv_lo = w;
   }
-  // contiguous per-wave slice, 32-token aligned
+  // cross-WG split (gridDim.z): z-chunk of the KV first, then the
+  // contiguous per-wave slice within it — both 32-token aligned
+  const int split = p.split > 1 ? p.split : 1;
   int64_t total = kv_len - kv_lo;
+  int64_t z_lo = kv_lo, z_hi = kv_len;
+  if (split > 1) {
+    int64_t per_z = ((total + (int64_t)split * KVB - 1) / ((int64_t)split * KVB)) * KVB;
+    z_lo = kv_lo + (int64_t)blockIdx.z * per_z;
+    z_hi = z_lo + per_z;
+    if (z_hi > kv_len) z_hi = kv_len;
+    if (z_lo > kv_len) z_lo = kv_len;
+    total = z_hi - z_lo;
+  }
   int64_t per_wave = ((total + WAVES * KVB - 1) / (WAVES * KVB)) * KVB;
-  int64_t ws_ = kv_lo + (int64_t)wave * per_wave;
+  int64_t ws_ = z_lo + (int64_t)wave * per_wave;
   int64_t we_ = ws_ + per_wave;
-  if (we_ > kv_len) we_ = kv_len;
+  if (we_ > z_hi) we_ = z_hi;
 
   const int32_t* page_ids = p.kv_indices + p.kv_indptr[req];
   const T* kbase = (const T*)p.k_data;
@@ -337,7 +348,11 @@ __global__ __launch_bounds__(WAVES * 64, 1) void decode_mfma_kernel(DecodeParams
   }
   __syncthreads();
 
-  // ---- cross-wave merge + direct store ----
+  // ---- cross-wave merge; split==1 stores directly, else the per-z
+  // NORMALIZED partial + base-2 lse goes to the merge-kernel workspace
+  // (an in-kernel atomic last-WG merge was tried: the device-scope
+  // threadfence it needs write-backs L2 across the 8 XCDs and measured 4x
+  // slower than a second tiny launch — profiles/README r02) ----
   T* obase = (T*)p.o + (int64_t)req * p.o_stride_n;
   for (int idx = tid; idx < QROWS * HEAD_DIM; idx += WAVES * 64) {
     int g = idx / HEAD_DIM;
@@ -356,17 +371,28 @@ __global__ __launch_bounds__(WAVES * 64, 1) void decode_mfma_kernel(DecodeParams
     }
     int qh = kv_head * GROUP + g;
     float inv_d = d_sum > 0.f ? 1.f / d_sum : 0.f;
-    obase[(int64_t)qh * p.o_stride_h + d] = from_f32<T>(o_sum * inv_d);
-    if (d == 0 && p.lse)
-      p.lse[(int64_t)req * p.num_qo_heads + qh] =
-          d_sum > 0.f ? m_star + __builtin_log2f(d_sum) : -INFINITY;
+    if (split == 1) {
+      obase[(int64_t)qh * p.o_stride_h + d] = from_f32<T>(o_sum * inv_d);
+      if (d == 0 && p.lse)
+        p.lse[(int64_t)req * p.num_qo_heads + qh] =
+            d_sum > 0.f ? m_star + __builtin_log2f(d_sum) : -INFINITY;
+    } else {
+      // merge_states layout: item = req*split + z over ALL Hq heads
+      int64_t item = (int64_t)req * split + blockIdx.z;
+      p.tmp_v[(item * p.num_qo_heads + qh) * HEAD_DIM + d] = o_sum * inv_d;
+      if (d == 0)
+        p.tmp_s[item * p.num_qo_heads + qh] =
+            d_sum > 0.f ? m_star + __builtin_log2f(d_sum) : -INFINITY;
+    }
   }
 }
 
 template <typename T>
 hipError_t decode_mfma_dispatch(DecodeParams& p, hipStream_t stream) {
   int group = p.num_qo_heads / p.num_kv_heads;
-  dim3 g((uint32_t)p.batch, (uint32_t)p.num_kv_heads), blk(WAVES * 64);
+  dim3 g((uint32_t)p.batch, (uint32_t)p.num_kv_heads,
+         (uint32_t)(p.split > 1 ? p.split : 1));
+  dim3 blk(WAVES * 64);
 #define LAUNCH_M(HD, G) \
   hipLaunchKernelGGL((decode_mfma_kernel<T, HD, G>), g, blk, 0, stream, p)
   if (p.head_dim == 128) {

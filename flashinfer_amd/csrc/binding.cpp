@@ -419,7 +419,10 @@ void batch_decode_mfma_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                            at::Tensor kv_last_page_len, int64_t layout,
                            at::Tensor out, c10::optional<at::Tensor> lse,
                            double sm_scale, double logits_soft_cap,
-                           int64_t window_left, bool alibi) {
+                           int64_t window_left, bool alibi, int64_t split,
+                           c10::optional<at::Tensor> tmp_v,
+                           c10::optional<at::Tensor> tmp_s,
+                           c10::optional<at::Tensor> counters) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 3, "q must be [batch, num_qo_heads, head_dim]");
   TORCH_CHECK(q.stride(2) == 1 && out.stride(2) == 1);
   fi_ext::DecodeParams p{};
@@ -448,6 +451,13 @@ void batch_decode_mfma_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   p.o_stride_n = out.stride(0);
   p.o_stride_h = out.stride(1);
   p.batch = q.size(0);
+  p.split = (int)split;
+  if (split > 1) {
+    TORCH_CHECK(tmp_v.has_value() && tmp_s.has_value(),
+                "split mfma decode needs tmp_v/tmp_s");
+    p.tmp_v = (float*)tmp_v->data_ptr();
+    p.tmp_s = tmp_s->data_ptr<float>();
+  }
   check_hip(fi_decode_mfma(dtype_code(q), &p, cur_stream(q)), "fi_decode_mfma");
 }
 

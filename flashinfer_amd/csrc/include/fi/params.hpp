@@ -51,6 +51,10 @@ struct DecodeParams {
   float* lse;             // [batch, Hq] base-2, optional
   int64_t o_stride_n, o_stride_h;
   int batch;
+  // MFMA decode cross-WG split (gridDim.z = split): partials + monotonic
+  // arrival counters (modulo split — no per-launch zeroing, hipGraph-safe)
+  int split;
+  uint32_t* counters;     // [batch * num_kv_heads]
 };
 
 struct PrefillParams {

@@ -227,6 +227,25 @@ class BatchDecodeWithPagedKVCacheWrapper:
             self._indices_d = _to_dev_f(indices, self._fixed_indices_buf)
             self._last_page_len_d = _to_dev_f(
                 last_page_len, self._fixed_last_page_len_buf)
+            # cross-WG split (mfma path): fill the 256-CU chip when
+            # batch x kv_heads alone cannot (bs=16 x 8 = 128 WGs = half idle)
+            self._mfma_split = 1
+            if self._fused_mfma and not disable_split_kv:
+                units = batch * num_kv_heads
+                want = min(-(-512 // max(1, units)), max(1, max_len // 256), 8)
+                if want > 1:
+                    self._mfma_split = int(want)
+                    alloc = WorkspaceAllocator(self._float_workspace_buffer)
+                    ni = batch * self._mfma_split
+                    self._mfma_tmp_v = alloc.alloc(
+                        ni * num_qo_heads * head_dim * 4, torch.float32,
+                        (ni, num_qo_heads, head_dim))
+                    self._mfma_tmp_s = alloc.alloc(
+                        ni * num_qo_heads * 4, torch.float32,
+                        (ni, num_qo_heads))
+                    self._mfma_merge_indptr = torch.arange(
+                        0, (batch + 1) * self._mfma_split, self._mfma_split,
+                        dtype=torch.int32).to(dev, non_blocking=non_blocking)
             self._plan_info = dict(
                 batch=batch, num_qo_heads=num_qo_heads,
                 num_kv_heads=num_kv_heads, head_dim=head_dim,
@@ -333,15 +352,32 @@ class BatchDecodeWithPagedKVCacheWrapper:
             if need_lse and lse is None:
                 lse = torch.empty((pi["batch"], pi["num_qo_heads"]),
                                   dtype=torch.float32, device=q.device)
-            run_fn = (get_ext().batch_decode_mfma_run
-                      if getattr(self, "_fused_mfma", False)
-                      else get_ext().batch_decode_fused_run)
-            run_fn(
-                q, k_cache, v_cache,
-                self._indices_d, self._indptr_d, self._last_page_len_d,
-                layout_code(self._kv_layout), out, lse if need_lse else None,
-                sm_scale, pi["logits_soft_cap"], pi["window_left"], pi["alibi"],
-            )
+            if getattr(self, "_fused_mfma", False):
+                sp = getattr(self, "_mfma_split", 1)
+                get_ext().batch_decode_mfma_run(
+                    q, k_cache, v_cache,
+                    self._indices_d, self._indptr_d, self._last_page_len_d,
+                    layout_code(self._kv_layout), out,
+                    lse if need_lse else None,
+                    sm_scale, pi["logits_soft_cap"], pi["window_left"],
+                    pi["alibi"], sp,
+                    self._mfma_tmp_v if sp > 1 else None,
+                    self._mfma_tmp_s if sp > 1 else None,
+                    None,
+                )
+                if sp > 1:
+                    get_ext().merge_states(
+                        self._mfma_tmp_v, self._mfma_tmp_s, out, lse,
+                        self._mfma_merge_indptr, 0, pi["batch"])
+            else:
+                get_ext().batch_decode_fused_run(
+                    q, k_cache, v_cache,
+                    self._indices_d, self._indptr_d, self._last_page_len_d,
+                    layout_code(self._kv_layout), out,
+                    lse if need_lse else None,
+                    sm_scale, pi["logits_soft_cap"], pi["window_left"],
+                    pi["alibi"],
+                )
             if sinks is not None:
                 import math as _m
 
