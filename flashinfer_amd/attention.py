@@ -1,9 +1,15 @@
 """Holistic mixed-batch attention (reference parity: flashinfer/attention/
-_core.py BatchAttention:44 — prefill and decode requests fused in ONE
-launch). The CDNA4 batch-prefill kernel handles arbitrary per-request qo_len
-(1-token decode rows pack into the same GQA row tiles), so the holistic API
-is the paged prefill wrapper planned over the mixed batch — one kernel, one
-launch, work items load-balanced longest-first."""
+_core.py BatchAttention:44, persistent kernel include/flashinfer/attention/
+persistent.cuh + TwoStageHolisticPlan scheduler.cuh:1243).
+
+MI355X design (csrc/attention/batch_attention.hip): ONE persistent kernel
+(a workgroup per CU) drains an atomic work queue of tagged items — 256-row
+prefill MFMA tiles and 32-row MFMA decode items (the GQA group as the q
+dimension, in-item LDS merge, direct output: no reduction stage). The host
+orders items most-expensive-first; the ticket gives dynamic load balance.
+Falls back to the batch-prefill wrapper for shapes outside the persistent
+kernel's coverage (hd 192/256, fp8 KV, custom masks, GQA group != 8 decode
+rows become prefill tiles)."""
 from __future__ import annotations
 
 from typing import Optional, Tuple
@@ -17,6 +23,7 @@ class BatchAttention:
     def __init__(self, kv_layout: str = "NHD", device=None, **kwargs):
         self._kv_layout = kv_layout
         self._wrapper: Optional[BatchPrefillWithPagedKVCacheWrapper] = None
+        self._persistent = False
 
     def plan(
         self,
@@ -37,28 +44,91 @@ class BatchAttention:
         use_profiler: bool = False,
         **kwargs,
     ) -> None:
+        from .utils import default_sm_scale
+
         dev = qo_indptr.device if qo_indptr.is_cuda else (
             kv_indices.device if kv_indices.is_cuda else torch.device("cuda"))
-        if self._wrapper is None:
-            ws = torch.empty(128 * 1024 * 1024, dtype=torch.uint8, device=dev)
-            self._wrapper = BatchPrefillWithPagedKVCacheWrapper(ws, self._kv_layout)
-        last_page_len = ((kv_len_arr.to(torch.int64) - 1) % page_size + 1).to(
-            torch.int32
+        group = num_qo_heads // max(1, num_kv_heads)
+        kv_dt = kv_data_type or q_data_type
+        self._persistent = (
+            head_dim_qk == head_dim_vo
+            and head_dim_qk in (64, 128)
+            and q_data_type in (torch.bfloat16, torch.float16)
+            and kv_dt == q_data_type
         )
-        self._wrapper.plan(
-            qo_indptr, kv_indptr, kv_indices, last_page_len,
-            num_qo_heads, num_kv_heads, head_dim_qk, page_size, causal=causal,
-            sm_scale=sm_scale, logits_soft_cap=logits_soft_cap,
-            q_data_type=q_data_type,
-        )
+        if not self._persistent:
+            if self._wrapper is None:
+                ws = torch.empty(128 << 20, dtype=torch.uint8, device=dev)
+                self._wrapper = BatchPrefillWithPagedKVCacheWrapper(
+                    ws, self._kv_layout)
+            last_page_len = ((kv_len_arr.to(torch.int64) - 1) % page_size + 1
+                             ).to(torch.int32)
+            self._wrapper.plan(
+                qo_indptr, kv_indptr, kv_indices, last_page_len,
+                num_qo_heads, num_kv_heads, head_dim_qk, page_size,
+                causal=causal, sm_scale=sm_scale,
+                logits_soft_cap=logits_soft_cap, q_data_type=q_data_type)
+            return
+        qi = qo_indptr.to("cpu", torch.int64)
+        qo_lens = (qi[1:] - qi[:-1]).tolist()
+        kv_lens = kv_len_arr.to("cpu", torch.int64).tolist()
+        self._group_dec = 8 if group == 8 else 0
+        # tagged work items, most-expensive-first (persistent queue order)
+        items = []  # (cost, kind, req, a, b)
+        for r, (ql, kl) in enumerate(zip(qo_lens, kv_lens)):
+            if ql == 1 and self._group_dec:
+                for h in range(num_kv_heads):
+                    items.append((kl * 32, 1, r, h, int(qi[r])))
+            else:
+                for qstart in range(0, max(1, ql * group), 256):
+                    for h in range(num_kv_heads):
+                        items.append((kl * 256, 0, r, qstart, h))
+        items.sort(key=lambda t: -t[0])
+        self._items = torch.tensor([it[1:] for it in items],
+                                   dtype=torch.int32).to(dev).reshape(-1, 4)
+        self._queue_head = torch.zeros(1, dtype=torch.int32, device=dev)
+        self._qo_indptr_d = qo_indptr.to(dev, torch.int32)
+        self._kv_indptr_d = kv_indptr.to(dev, torch.int32)
+        self._kv_indices_d = kv_indices.to(dev, torch.int32)
+        self._last_page_len_d = (
+            (kv_len_arr.to(torch.int64) - 1) % page_size + 1
+        ).to(dev, torch.int32)
+        n_cu = torch.cuda.get_device_properties(dev).multi_processor_count
+        self._n_wgs = min(len(items), n_cu)
+        self._pi = dict(
+            nnz_q=int(qi[-1]), num_qo_heads=num_qo_heads, causal=causal,
+            head_dim_vo=head_dim_vo,
+            sm_scale=sm_scale if sm_scale is not None
+            else default_sm_scale(head_dim_qk),
+            logits_soft_cap=float(logits_soft_cap or 0.0))
 
     def run(
         self, q: torch.Tensor, kv_cache, out: Optional[torch.Tensor] = None,
         lse: Optional[torch.Tensor] = None, return_lse: bool = True, **kwargs,
     ):
-        result = self._wrapper.run(q, kv_cache, out=out, lse=lse,
-                                   return_lse=return_lse)
-        return result
+        if not self._persistent:
+            return self._wrapper.run(q, kv_cache, out=out, lse=lse,
+                                     return_lse=return_lse)
+        from ._lib import get_ext
+        from .utils import layout_code, unpack_paged_kv_cache
+
+        pi = self._pi
+        k_cache, v_cache = unpack_paged_kv_cache(kv_cache, self._kv_layout)
+        if out is None:
+            out = torch.empty(q.shape[0], q.shape[1], pi["head_dim_vo"],
+                              dtype=q.dtype, device=q.device)
+        if return_lse and lse is None:
+            lse = torch.empty(q.shape[0], q.shape[1], dtype=torch.float32,
+                              device=q.device)
+        self._queue_head.zero_()
+        get_ext().batch_attention_run(
+            q, k_cache, v_cache, self._qo_indptr_d, self._kv_indices_d,
+            self._kv_indptr_d, self._last_page_len_d,
+            layout_code(self._kv_layout), self._items, self._queue_head,
+            out, lse if return_lse else None, pi["sm_scale"],
+            pi["logits_soft_cap"], -1, pi["causal"], self._group_dec,
+            self._n_wgs)
+        return (out, lse) if return_lse else out
 
 
 class PODWithPagedKVCacheWrapper:

@@ -48,6 +48,17 @@ hipError_t fi_batch_decode(int dtype, int kv_dtype, fi_ext::DecodeParams* p,
 hipError_t fi_batch_decode_fused(int dtype, int kv_dtype, fi_ext::DecodeParams* p,
                                  hipStream_t stream);
 hipError_t fi_decode_mfma(int dtype, fi_ext::DecodeParams* p, hipStream_t stream);
+namespace fi_ext2 {
+struct HolisticParams {
+  fi_ext::PrefillParams pf;
+  fi_ext::DecodeParams dec;
+  const int32_t* items;
+  int n_items;
+  uint32_t* queue_head;
+};
+}  // namespace fi_ext2
+hipError_t fi_batch_attention(int dtype, fi_ext2::HolisticParams* h, int group_dec,
+                              int causal, int n_wgs, hipStream_t stream);
 hipError_t fi_gemm_nt(int dtype, const void* A, const void* B, void* C, int M, int N,
                       int K, int64_t lda, int64_t ldb, int64_t ldc, float alpha,
                       hipStream_t stream);
@@ -459,6 +470,86 @@ void batch_decode_mfma_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
     p.tmp_s = tmp_s->data_ptr<float>();
   }
   check_hip(fi_decode_mfma(dtype_code(q), &p, cur_stream(q)), "fi_decode_mfma");
+}
+
+// Persistent holistic BatchAttention: one launch over a tagged work queue of
+// prefill tiles + decode items (csrc/attention/batch_attention.hip).
+void batch_attention_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
+                         at::Tensor qo_indptr, at::Tensor kv_indices,
+                         at::Tensor kv_indptr, at::Tensor kv_last_page_len,
+                         int64_t layout, at::Tensor items, at::Tensor queue_head,
+                         at::Tensor out, c10::optional<at::Tensor> lse,
+                         double sm_scale, double logits_soft_cap,
+                         int64_t window_left, bool causal, int64_t group_dec,
+                         int64_t n_wgs) {
+  TORCH_CHECK(q.is_cuda() && q.dim() == 3, "q must be [nnz, Hq, D]");
+  TORCH_CHECK(q.stride(2) == 1 && out.stride(2) == 1);
+  fi_ext2::HolisticParams h{};
+  // prefill side
+  fi_ext::PrefillParams& pf = h.pf;
+  pf.q = q.data_ptr();
+  pf.out = out.data_ptr();
+  pf.lse = lse.has_value() ? lse->data_ptr<float>() : nullptr;
+  pf.qo_indptr = qo_indptr.data_ptr<int32_t>();
+  pf.k_data = k_cache.data_ptr();
+  pf.v_data = v_cache.data_ptr();
+  pf.kv_indices = kv_indices.data_ptr<int32_t>();
+  pf.kv_indptr = kv_indptr.data_ptr<int32_t>();
+  pf.kv_last_page_len = kv_last_page_len.data_ptr<int32_t>();
+  int page_size = layout == 0 ? k_cache.size(1) : k_cache.size(2);
+  int num_kv_heads = layout == 0 ? k_cache.size(2) : k_cache.size(1);
+  pf.page_size = fi::uint_fastdiv((uint32_t)page_size);
+  pf.kv_stride_page = k_cache.stride(0);
+  pf.kv_stride_n = layout == 0 ? k_cache.stride(1) : k_cache.stride(2);
+  pf.kv_stride_h = layout == 0 ? k_cache.stride(2) : k_cache.stride(1);
+  pf.v_stride_page = v_cache.stride(0);
+  pf.v_stride_n = layout == 0 ? v_cache.stride(1) : v_cache.stride(2);
+  pf.v_stride_h = layout == 0 ? v_cache.stride(2) : v_cache.stride(1);
+  pf.head_dim_vo = v_cache.size(3);
+  pf.num_qo_heads = q.size(1);
+  pf.num_kv_heads = num_kv_heads;
+  pf.head_dim = q.size(2);
+  pf.group = fi::uint_fastdiv((uint32_t)(pf.num_qo_heads / num_kv_heads));
+  pf.q_stride_n = q.stride(0);
+  pf.q_stride_h = q.stride(1);
+  pf.o_stride_n = out.stride(0);
+  pf.o_stride_h = out.stride(1);
+  pf.sm_scale = (float)sm_scale;
+  pf.logits_soft_cap = (float)logits_soft_cap;
+  pf.window_left = (int)window_left;
+  pf.causal = causal ? 1 : 0;
+  pf.cta_q = 256;
+  // decode side (shares the page table; q rows indexed by qo_indptr[req])
+  fi_ext::DecodeParams& d = h.dec;
+  d.k_data = k_cache.data_ptr();
+  d.v_data = v_cache.data_ptr();
+  d.kv_indices = kv_indices.data_ptr<int32_t>();
+  d.kv_indptr = kv_indptr.data_ptr<int32_t>();
+  d.kv_last_page_len = kv_last_page_len.data_ptr<int32_t>();
+  d.page_size = fi::uint_fastdiv((uint32_t)page_size);
+  d.num_kv_heads = num_kv_heads;
+  d.num_qo_heads = q.size(1);
+  d.head_dim = q.size(2);
+  d.stride_page = pf.kv_stride_page;
+  d.stride_n = pf.kv_stride_n;
+  d.stride_h = pf.kv_stride_h;
+  d.q = q.data_ptr();
+  d.q_stride_n = q.stride(0);
+  d.q_stride_h = q.stride(1);
+  d.o = out.data_ptr();
+  d.lse = pf.lse;
+  d.o_stride_n = out.stride(0);
+  d.o_stride_h = out.stride(1);
+  d.sm_scale = (float)sm_scale;
+  d.logits_soft_cap = (float)logits_soft_cap;
+  d.window_left = (int)window_left;
+  d.split = 1;
+  h.items = items.data_ptr<int32_t>();
+  h.n_items = items.size(0);
+  h.queue_head = (uint32_t*)queue_head.data_ptr();
+  check_hip(fi_batch_attention(dtype_code(q), &h, (int)group_dec, causal ? 1 : 0,
+                               (int)n_wgs, cur_stream(q)),
+            "fi_batch_attention");
 }
 
 // ---------------- prefill ----------------
@@ -1031,6 +1122,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("batch_decode_run", &batch_decode_run);
   m.def("batch_decode_fused_run", &batch_decode_fused_run);
   m.def("batch_decode_mfma_run", &batch_decode_mfma_run);
+  m.def("batch_attention_run", &batch_attention_run);
   m.def("gemm_nt", &gemm_nt);
   m.def("batch_prefill_run", &batch_prefill_run);
   m.def("softmax", &softmax_op);

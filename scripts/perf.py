@@ -221,6 +221,57 @@ def bench_decode_modes(bs=16, kv=1024, Hq=64, Hkv=8, D=128, page=16):
 
 
 
+def bench_holistic(n_prefill=8, n_decode=120, s_prefill=1024, kv_decode=1024,
+                   Hq=64, Hkv=8, D=128, page=16):
+    """Persistent holistic BatchAttention vs the two-wrapper composition on a
+    mixed prefill+decode batch (VERDICT #2 acceptance measurement)."""
+    torch.manual_seed(0)
+    qo_lens = [s_prefill] * n_prefill + [1] * n_decode
+    kv_lens = [s_prefill] * n_prefill + [kv_decode] * n_decode
+    pages_per = [(L + page - 1) // page for L in kv_lens]
+    qo_indptr = torch.tensor([0] + list(torch.cumsum(torch.tensor(qo_lens), 0)), dtype=torch.int32, device="cuda")
+    kv_indptr = torch.tensor([0] + list(torch.cumsum(torch.tensor(pages_per), 0)), dtype=torch.int32, device="cuda")
+    npages = int(kv_indptr[-1])
+    kv_indices = torch.randperm(npages, dtype=torch.int32, device="cuda")
+    kv_len_arr = torch.tensor(kv_lens, dtype=torch.int32, device="cuda")
+    kc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    nnz = sum(qo_lens)
+    q = torch.randn(nnz, Hq, D, dtype=torch.bfloat16, device="cuda")
+    out = torch.empty_like(q)
+    # holistic single launch
+    w = fi.BatchAttention("NHD")
+    w.plan(qo_indptr, kv_indptr, kv_indices, kv_len_arr, Hq, Hkv, D, D, page,
+           causal=True, q_data_type=torch.bfloat16)
+    t_h = timeit(lambda: w.run(q, (kc, vc), out=out, return_lse=False))
+    # two-wrapper composition (prefill wrapper for prefills + decode wrapper)
+    lp = ((kv_len_arr.to(torch.int64) - 1) % page + 1).to(torch.int32)
+    wp = fi.BatchPrefillWithPagedKVCacheWrapper(
+        torch.empty(256 << 20, dtype=torch.uint8, device="cuda"), "NHD")
+    wp.plan(qo_indptr[: n_prefill + 1], kv_indptr[: n_prefill + 1], kv_indices,
+            lp[:n_prefill], Hq, Hkv, D, page, causal=True,
+            q_data_type=torch.bfloat16)
+    wd = fi.BatchDecodeWithPagedKVCacheWrapper(
+        torch.empty(256 << 20, dtype=torch.uint8, device="cuda"), "NHD")
+    dec_indptr = (kv_indptr[n_prefill:] - kv_indptr[n_prefill]).contiguous()
+    wd.plan(dec_indptr, kv_indices[int(kv_indptr[n_prefill]):], lp[n_prefill:],
+            Hq, Hkv, D, page, q_data_type=torch.bfloat16)
+    q_pf = q[: qo_indptr[n_prefill]]
+    q_dec = q[qo_indptr[n_prefill]:]
+    o_pf = out[: qo_indptr[n_prefill]]
+    o_dec = out[qo_indptr[n_prefill]:]
+    def two():
+        wp.run(q_pf, (kc, vc), out=o_pf)
+        wd.run(q_dec, (kc, vc), out=o_dec)
+    t_2 = timeit(two)
+    fl = sum(2 * 2 * Hq * D * q_ * k_ / (2 if q_ > 1 else 1)
+             for q_, k_ in zip(qo_lens, kv_lens))
+    print(f"holistic {n_prefill}pf/{n_decode}dec: one-launch {t_h*1e6:.0f} us "
+          f"({fl/t_h/1e12:.1f} TF)  two-wrapper {t_2*1e6:.0f} us  "
+          f"speedup {t_2/t_h:.2f}x")
+
+
+
 if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "all"
     if which in ("all", "prefill"):
@@ -248,6 +299,10 @@ if __name__ == "__main__":
     if which in ("all", "mla"):
         bench_mla()
         bench_mla(bs=64, kv=4096)
+    if which in ("all", "holistic"):
+        bench_holistic()
+        bench_holistic(n_prefill=2, n_decode=200, kv_decode=2048)
+        bench_holistic(n_prefill=16, n_decode=16)
     if which in ("all", "moe"):
         bench_moe()
         bench_moe_fp8()
