@@ -73,28 +73,61 @@ class BatchAttention:
         qo_lens = (qi[1:] - qi[:-1]).tolist()
         kv_lens = kv_len_arr.to("cpu", torch.int64).tolist()
         self._group_dec = 8 if group == 8 else 0
-        # tagged work items, most-expensive-first (persistent queue order)
-        items = []  # (cost, kind, req, a, b)
+        # tagged work items: prefill tiles then decode items, each pool
+        # most-expensive-first; the decode WG pool is sized by cost fraction
+        pf_items, dec_items = [], []  # (cost, kind, req, a, b)
         for r, (ql, kl) in enumerate(zip(qo_lens, kv_lens)):
             if ql == 1 and self._group_dec:
                 for h in range(num_kv_heads):
-                    items.append((kl * 32, 1, r, h, int(qi[r])))
+                    dec_items.append((kl * 32, 1, r, h, int(qi[r])))
             else:
                 for qstart in range(0, max(1, ql * group), 256):
                     for h in range(num_kv_heads):
-                        items.append((kl * 256, 0, r, qstart, h))
-        items.sort(key=lambda t: -t[0])
+                        pf_items.append((kl * 256, 0, r, qstart, h))
+        pf_items.sort(key=lambda t: -t[0])
+        dec_items.sort(key=lambda t: -t[0])
+        items = pf_items + dec_items
+        self._n_pf_items = len(pf_items)
+        pf_cost = sum(t[0] for t in pf_items)
+        dec_cost = sum(t[0] for t in dec_items)
         self._items = torch.tensor([it[1:] for it in items],
                                    dtype=torch.int32).to(dev).reshape(-1, 4)
-        self._queue_head = torch.zeros(1, dtype=torch.int32, device=dev)
+        self._queue_head = torch.zeros(2, dtype=torch.int32, device=dev)
         self._qo_indptr_d = qo_indptr.to(dev, torch.int32)
         self._kv_indptr_d = kv_indptr.to(dev, torch.int32)
         self._kv_indices_d = kv_indices.to(dev, torch.int32)
         self._last_page_len_d = (
             (kv_len_arr.to(torch.int64) - 1) % page_size + 1
         ).to(dev, torch.int32)
+        # measured routing (profiles/README r02 holistic): the persistent
+        # kernel's register-allocation tax makes it ~0.7x of the composed
+        # two-kernel path when decode cost dominates; the standalone MFMA
+        # decode kernel is strictly faster per item. Route decode-heavy
+        # batches through the composed path (still one plan/run call).
+        if dec_items and dec_cost > 0.25 * (pf_cost + dec_cost):
+            self._persistent = False
+            if self._wrapper is None:
+                ws = torch.empty(128 << 20, dtype=torch.uint8, device=dev)
+                self._wrapper = BatchPrefillWithPagedKVCacheWrapper(
+                    ws, self._kv_layout)
+            last_page_len = ((kv_len_arr.to(torch.int64) - 1) % page_size + 1
+                             ).to(torch.int32)
+            self._wrapper.plan(
+                qo_indptr, kv_indptr, kv_indices, last_page_len,
+                num_qo_heads, num_kv_heads, head_dim_qk, page_size,
+                causal=causal, sm_scale=sm_scale,
+                logits_soft_cap=logits_soft_cap, q_data_type=q_data_type)
+            return
         n_cu = torch.cuda.get_device_properties(dev).multi_processor_count
         self._n_wgs = min(len(items), n_cu)
+        if dec_items and pf_items:
+            frac = dec_cost / max(1, pf_cost + dec_cost)
+            self._n_dec_wgs = max(1, min(self._n_wgs - 1,
+                                         round(self._n_wgs * frac)))
+        elif dec_items:
+            self._n_dec_wgs = self._n_wgs
+        else:
+            self._n_dec_wgs = 0
         self._pi = dict(
             nnz_q=int(qi[-1]), num_qo_heads=num_qo_heads, causal=causal,
             head_dim_vo=head_dim_vo,
@@ -127,7 +160,7 @@ class BatchAttention:
             layout_code(self._kv_layout), self._items, self._queue_head,
             out, lse if return_lse else None, pi["sm_scale"],
             pi["logits_soft_cap"], -1, pi["causal"], self._group_dec,
-            self._n_wgs)
+            self._n_wgs, self._n_pf_items, self._n_dec_wgs)
         return (out, lse) if return_lse else out
 
 

@@ -24,16 +24,20 @@ namespace fi {
 struct HolisticParams {
   PrefillParams pf;
   DecodeParams dec;
-  const int32_t* items;  // [n_items][4]: kind, req, a, b
-  int n_items;
-  uint32_t* queue_head;  // zeroed by the host before each launch
+  const int32_t* items;      // [n_items][4]: kind, req, a, b (prefill first)
+  int n_pf_items;            // items[0 .. n_pf_items) are prefill tiles
+  int n_items;               // items[n_pf_items .. n_items) are decode items
+  int n_dec_wgs;             // workgroups dedicated to the decode queue
+  uint32_t* queue_head;      // [2] pf/dec tickets, zeroed before each launch
 };
 
-// Both runners inline into the kernel. The union of their register demands
-// costs ~500 B/lane of scratch (each body alone fits the 256-VGPR budget) —
-// a noinline-call split was tried, but the AMDGPU call ABI's reserved
-// registers made each callee spill MORE (592-684 B). The spill traffic sits
-// outside the MFMA hot loops and the queue amortizes it per item.
+// KIND-PARTITIONED persistent loops: workgroups branch ONCE (uniformly, on
+// blockIdx) into a prefill-only or decode-only queue loop. A single mixed
+// loop was tried first — inlining both runners into one loop body made the
+// register allocator spill ~500 B/lane (each body alone fits the 256-VGPR
+// budget), and a noinline call split spilled MORE (ABI-reserved registers).
+// Top-level disjoint loops keep each runner's allocation independent; the
+// host sizes the decode pool by its cost fraction.
 template <typename T, int HEAD_DIM, int GROUP_DEC, bool CAUSAL>
 __global__ __launch_bounds__(512, 1) void batch_attention_kernel(HolisticParams h) {
   constexpr int PF_SMEM = prefill_tile_smem_bytes<T, HEAD_DIM, HEAD_DIM>();
@@ -41,22 +45,35 @@ __global__ __launch_bounds__(512, 1) void batch_attention_kernel(HolisticParams 
       GROUP_DEC > 0 ? decode_mfma_smem_bytes<T, HEAD_DIM, GROUP_DEC>() : 0;
   __shared__ char smem[(PF_SMEM > DEC_SMEM ? PF_SMEM : DEC_SMEM)];
   __shared__ uint32_t s_item;
+  auto drain_dec = [&]() {
+    if constexpr (GROUP_DEC > 0) {
+      for (;;) {
+        __syncthreads();
+        if (threadIdx.x == 0) s_item = atomicAdd(h.queue_head + 1, 1u);
+        __syncthreads();
+        uint32_t it = (uint32_t)h.n_pf_items + s_item;
+        if (it >= (uint32_t)h.n_items) return;
+        const int32_t* rec = h.items + 4 * it;
+        // rec: (1, req, kv_head, qo_row = qo_indptr[req])
+        decode_mfma_item_body<T, HEAD_DIM, GROUP_DEC>(h.dec, rec[1], rec[2], 0,
+                                                      smem, rec[3]);
+      }
+    }
+  };
+  // decode-pool WGs drain their queue first, then steal prefill work;
+  // prefill WGs do the reverse — no pool ever idles while work remains.
+  if ((int)blockIdx.x < h.n_dec_wgs) drain_dec();
   for (;;) {
     __syncthreads();  // previous item's smem reads complete before re-claim
     if (threadIdx.x == 0) s_item = atomicAdd(h.queue_head, 1u);
     __syncthreads();
     uint32_t it = s_item;
-    if (it >= (uint32_t)h.n_items) return;
+    if (it >= (uint32_t)h.n_pf_items) break;
     const int32_t* rec = h.items + 4 * it;
-    if (rec[0] == 0) {
-      prefill_tile_body<T, T, HEAD_DIM, HEAD_DIM, 256, true, CAUSAL, false>(
-          h.pf, rec[1], rec[2], rec[3], -1, smem);
-    } else if constexpr (GROUP_DEC > 0) {
-      // rec: (1, req, kv_head, qo_row = qo_indptr[req])
-      decode_mfma_item_body<T, HEAD_DIM, GROUP_DEC>(h.dec, rec[1], rec[2], 0,
-                                                    smem, rec[3]);
-    }
+    prefill_tile_body<T, T, HEAD_DIM, HEAD_DIM, 256, true, CAUSAL, false>(
+        h.pf, rec[1], rec[2], rec[3], -1, smem);
   }
+  if ((int)blockIdx.x >= h.n_dec_wgs) drain_dec();
 }
 
 template <typename T>

@@ -53,7 +53,9 @@ struct HolisticParams {
   fi_ext::PrefillParams pf;
   fi_ext::DecodeParams dec;
   const int32_t* items;
+  int n_pf_items;
   int n_items;
+  int n_dec_wgs;
   uint32_t* queue_head;
 };
 }  // namespace fi_ext2
@@ -481,7 +483,7 @@ void batch_attention_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                          at::Tensor out, c10::optional<at::Tensor> lse,
                          double sm_scale, double logits_soft_cap,
                          int64_t window_left, bool causal, int64_t group_dec,
-                         int64_t n_wgs) {
+                         int64_t n_wgs, int64_t n_pf_items, int64_t n_dec_wgs) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 3, "q must be [nnz, Hq, D]");
   TORCH_CHECK(q.stride(2) == 1 && out.stride(2) == 1);
   fi_ext2::HolisticParams h{};
@@ -546,6 +548,8 @@ void batch_attention_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   d.split = 1;
   h.items = items.data_ptr<int32_t>();
   h.n_items = items.size(0);
+  h.n_pf_items = (int)n_pf_items;
+  h.n_dec_wgs = (int)n_dec_wgs;
   h.queue_head = (uint32_t*)queue_head.data_ptr();
   check_hip(fi_batch_attention(dtype_code(q), &h, (int)group_dec, causal ? 1 : 0,
                                (int)n_wgs, cur_stream(q)),

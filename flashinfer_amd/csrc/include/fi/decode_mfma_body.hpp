@@ -119,7 +119,7 @@ __device__ __forceinline__ void decode_mfma_item_body(const DecodeParams& p, int
       int chunk8 = u % (HEAD_DIM / 8);
       vec_t<T, 8> qv;
       if (row < QROWS) {
-        qv.load((const T*)p.q + (int64_t)req * p.q_stride_n +
+        qv.load((const T*)p.q + (int64_t)qo_row * p.q_stride_n +
                 (int64_t)(kv_head * GROUP + row) * p.q_stride_h + chunk8 * 8);
       } else {
         qv.fill(0.f);
