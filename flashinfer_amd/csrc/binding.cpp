@@ -107,6 +107,20 @@ hipError_t fi_selective_state_update(int dtype, fi_ext::SSUParams* p,
                                      hipStream_t stream);
 hipError_t fi_gather_rows(int dtype, const void* src, void* dst, const int32_t* row_map,
                           int64_t rows, int cols, hipStream_t stream);
+hipError_t fi_moe_topk_softmax(const float* logits, float* weights, int32_t* ids,
+                               int T, int E, int k, int renorm, hipStream_t stream);
+hipError_t fi_dsv3_routing(const float* logits, const float* bias, float* weights,
+                           int32_t* ids, int T, int E, int k, int n_group,
+                           int topk_group, float scale, hipStream_t stream);
+hipError_t fi_moe_build_permute(const int32_t* ids, int32_t* counts,
+                                int32_t* m_indptr, int32_t* cursor,
+                                int32_t* token_of_copy, int32_t* inv, int n,
+                                int k, int E, hipStream_t stream);
+hipError_t fi_gather_quant(int dtype, const void* src, const int32_t* token_of_copy,
+                           uint8_t* dst, float* scale, int R, int K,
+                           hipStream_t stream);
+hipError_t fi_silu_mul_quant(int dtype, const void* h, uint8_t* dst, float* scale,
+                             int R, int I, int gelu, hipStream_t stream);
 hipError_t fi_moe_finalize(int dtype, const void* h, void* out, const int32_t* pos,
                            const float* w, int64_t tokens, int topk, int cols,
                            hipStream_t stream);
@@ -1110,6 +1124,59 @@ std::vector<int64_t> debug_fastdiv(int64_t d, std::vector<int64_t> ns) {
 
 }  // namespace
 
+
+void moe_topk_softmax_run(at::Tensor logits, at::Tensor weights, at::Tensor ids,
+                          bool renorm) {
+  TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == at::kFloat);
+  check_hip(fi_moe_topk_softmax(logits.data_ptr<float>(), weights.data_ptr<float>(),
+                                ids.data_ptr<int32_t>(), logits.size(0),
+                                logits.size(1), ids.size(1), renorm ? 1 : 0,
+                                cur_stream(logits)),
+            "fi_moe_topk_softmax");
+}
+
+void dsv3_routing_run(at::Tensor logits, c10::optional<at::Tensor> bias,
+                      at::Tensor weights, at::Tensor ids, int64_t n_group,
+                      int64_t topk_group, double scale) {
+  TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == at::kFloat);
+  check_hip(fi_dsv3_routing(logits.data_ptr<float>(),
+                            bias.has_value() ? bias->data_ptr<float>() : nullptr,
+                            weights.data_ptr<float>(), ids.data_ptr<int32_t>(),
+                            logits.size(0), logits.size(1), ids.size(1),
+                            (int)n_group, (int)topk_group, (float)scale,
+                            cur_stream(logits)),
+            "fi_dsv3_routing");
+}
+
+void moe_build_permute_run(at::Tensor ids, at::Tensor counts, at::Tensor m_indptr,
+                           at::Tensor cursor, at::Tensor token_of_copy,
+                           at::Tensor inv) {
+  int n = ids.numel(), k = ids.size(-1), E = counts.size(0);
+  check_hip(fi_moe_build_permute(ids.data_ptr<int32_t>(), counts.data_ptr<int32_t>(),
+                                 m_indptr.data_ptr<int32_t>(),
+                                 cursor.data_ptr<int32_t>(),
+                                 token_of_copy.data_ptr<int32_t>(),
+                                 inv.data_ptr<int32_t>(), n, k, E,
+                                 cur_stream(ids)),
+            "fi_moe_build_permute");
+}
+
+void gather_quant_run(at::Tensor src, at::Tensor token_of_copy, at::Tensor dst,
+                      at::Tensor scale) {
+  check_hip(fi_gather_quant(dtype_code(src), src.data_ptr(),
+                            token_of_copy.data_ptr<int32_t>(),
+                            dst.data_ptr<uint8_t>(), scale.data_ptr<float>(),
+                            dst.size(0), dst.size(1), cur_stream(src)),
+            "fi_gather_quant");
+}
+
+void silu_mul_quant_run(at::Tensor h, at::Tensor dst, at::Tensor scale, bool gelu) {
+  check_hip(fi_silu_mul_quant(dtype_code(h), h.data_ptr(), dst.data_ptr<uint8_t>(),
+                              scale.data_ptr<float>(), dst.size(0), dst.size(1),
+                              gelu ? 1 : 0, cur_stream(h)),
+            "fi_silu_mul_quant");
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm);
   m.def("rmsnorm_silu", &rmsnorm_silu);
@@ -1122,6 +1189,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("append_paged_kv_cache", &append_paged_kv_cache);
   m.def("get_batch_indices_positions", &get_batch_indices_positions);
   m.def("merge_states", &merge_states);
+  m.def("moe_topk_softmax_run", &moe_topk_softmax_run);
+  m.def("dsv3_routing_run", &dsv3_routing_run);
+  m.def("moe_build_permute_run", &moe_build_permute_run);
+  m.def("gather_quant_run", &gather_quant_run);
+  m.def("silu_mul_quant_run", &silu_mul_quant_run);
   m.def("merge_state_in_place", &merge_state_in_place);
   m.def("batch_decode_run", &batch_decode_run);
   m.def("batch_decode_fused_run", &batch_decode_fused_run);

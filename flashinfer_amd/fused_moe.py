@@ -23,7 +23,18 @@ from .utils import ceil_div
 def moe_topk_softmax(
     router_logits: torch.Tensor, top_k: int, renormalize: bool = True
 ) -> Tuple[torch.Tensor, torch.Tensor]:
-    """Standard Mixtral-style routing: softmax then top-k (+ renorm)."""
+    """Standard Mixtral-style routing: softmax then top-k (+ renorm).
+
+    One fused wave-per-token kernel on GPU (reference
+    trtllm_gen_routing topk-softmax role); torch fallback on CPU."""
+    T, E = router_logits.shape
+    if router_logits.is_cuda and E <= 256 and top_k <= 16:
+        logits = router_logits.float().contiguous()
+        weights = torch.empty(T, top_k, dtype=torch.float32,
+                              device=logits.device)
+        ids = torch.empty(T, top_k, dtype=torch.int32, device=logits.device)
+        get_ext().moe_topk_softmax_run(logits, weights, ids, renormalize)
+        return weights, ids
     probs = torch.softmax(router_logits.float(), dim=-1)
     weights, ids = torch.topk(probs, top_k, dim=-1)
     if renormalize:
@@ -37,7 +48,21 @@ def dsv3_routing(
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """DeepSeek-V3 no-aux-loss routing: sigmoid scores (+bias for selection),
     group-limited top-k (keep topk_group groups by their top-2 sum), weights
-    from the unbiased scores, normalized and scaled."""
+    from the unbiased scores, normalized and scaled.
+
+    Fused wave-per-token kernel on GPU (reference
+    trtllm_fused_moe_routing_deepseek.cu / noAuxTcKernels role)."""
+    T, E = router_logits.shape
+    if (router_logits.is_cuda and E <= 256 and top_k <= 16
+            and n_group <= 32 and E // n_group <= 32):
+        logits = router_logits.float().contiguous()
+        weights = torch.empty(T, top_k, dtype=torch.float32,
+                              device=logits.device)
+        ids = torch.empty(T, top_k, dtype=torch.int32, device=logits.device)
+        get_ext().dsv3_routing_run(
+            logits, bias.float().contiguous() if bias is not None else None,
+            weights, ids, n_group, topk_group, routed_scaling_factor)
+        return weights, ids
     scores = torch.sigmoid(router_logits.float())
     sel = scores + bias.float() if bias is not None else scores
     T, E = scores.shape
@@ -72,17 +97,32 @@ def _grouped_nt_bf16(a, w, m_indptr, max_m_tiles):
     return out
 
 
-def _permute(x: torch.Tensor, topk_ids: torch.Tensor, num_experts: int):
+def _build_permute(topk_ids: torch.Tensor, num_experts: int):
+    """hist -> scan -> atomic scatter kernels (replaces the torch
+    argsort/bincount/cumsum chain — reference moe_kernels.h expand stage).
+    Returns (m_indptr, token_of_copy, inv) where inv[t, j] is the permuted
+    row of copy (t, j). Atomic order within an expert is arbitrary, but all
+    downstream ops are row-local and the finalize gathers by inv, so outputs
+    are bitwise deterministic."""
     T, k = topk_ids.shape
-    flat = topk_ids.reshape(-1).long()
-    order = torch.argsort(flat, stable=True)
-    counts = torch.bincount(flat, minlength=num_experts)
-    m_indptr = torch.zeros(num_experts + 1, dtype=torch.int32, device=x.device)
-    m_indptr[1:] = counts.cumsum(0).to(torch.int32)
-    token_of_copy = (order // k).to(torch.int32)  # source token per permuted row
-    a_perm = torch.empty(T * k, x.shape[1], dtype=x.dtype, device=x.device)
+    dev = topk_ids.device
+    ids = topk_ids.to(torch.int32).contiguous()
+    counts = torch.zeros(num_experts, dtype=torch.int32, device=dev)
+    cursor = torch.empty(num_experts, dtype=torch.int32, device=dev)
+    m_indptr = torch.empty(num_experts + 1, dtype=torch.int32, device=dev)
+    token_of_copy = torch.empty(T * k, dtype=torch.int32, device=dev)
+    inv = torch.empty(T * k, dtype=torch.int32, device=dev)
+    get_ext().moe_build_permute_run(ids, counts, m_indptr, cursor,
+                                    token_of_copy, inv)
+    return m_indptr, token_of_copy, inv
+
+
+def _permute(x: torch.Tensor, topk_ids: torch.Tensor, num_experts: int):
+    m_indptr, token_of_copy, inv = _build_permute(topk_ids, num_experts)
+    a_perm = torch.empty(topk_ids.numel(), x.shape[1], dtype=x.dtype,
+                         device=x.device)
     get_ext().gather_rows(x, a_perm, token_of_copy)
-    return a_perm, m_indptr, order, token_of_copy
+    return a_perm, m_indptr, inv, token_of_copy
 
 
 def fused_moe(
@@ -108,35 +148,37 @@ def fused_moe(
     ext = get_ext()
     fp8 = w13.dtype == torch.float8_e4m3fn
 
-    a_perm, m_indptr, order, token_of_copy = _permute(x, topk_ids, E)
-    R = a_perm.shape[0]
+    R = T * k
     max_m_tiles = ceil_div(R, 128) + 1
 
     if fp8:
-        from .fp8_quantization import per_token_group_quant_fp8
-
+        # fused data movers: gather+quant feeds GEMM1, silu_mul+quant feeds
+        # GEMM2 — the bf16 intermediate round-trips are gone
+        m_indptr, token_of_copy, inv = _build_permute(topk_ids, E)
+        a_q = torch.empty(R, H, dtype=torch.uint8, device=x.device)
+        a_s = torch.empty(H // 128, R, dtype=torch.float32, device=x.device)
+        ext.gather_quant_run(x, token_of_copy, a_q, a_s)
         h1 = torch.empty(R, I2, dtype=torch.bfloat16, device=x.device)
-        a_q, a_s = per_token_group_quant_fp8(a_perm, transpose_scale=True)
-        ext.gemm_fp8_grouped(a_q.view(torch.uint8), w13.view(torch.uint8), h1,
-                             m_indptr, None, max_m_tiles, a_s.contiguous(),
+        ext.gemm_fp8_grouped(a_q, w13.view(torch.uint8), h1,
+                             m_indptr, None, max_m_tiles, a_s,
                              w13_scale.contiguous(), 1.0)
-    else:
-        h1 = _grouped_nt_bf16(a_perm, w13, m_indptr, max_m_tiles)
-    act = {"silu": silu_and_mul, "gelu": gelu_and_mul}[activation](h1)
-    if fp8:
+        act_q = torch.empty(R, inter, dtype=torch.uint8, device=x.device)
+        act_s = torch.empty(inter // 128, R, dtype=torch.float32,
+                            device=x.device)
+        ext.silu_mul_quant_run(h1, act_q, act_s, activation == "gelu")
         h2 = torch.empty(R, H, dtype=torch.bfloat16, device=x.device)
-        act_q, act_s = per_token_group_quant_fp8(act, transpose_scale=True)
-        ext.gemm_fp8_grouped(act_q.view(torch.uint8), w2.view(torch.uint8), h2,
-                             m_indptr, None, max_m_tiles, act_s.contiguous(),
+        ext.gemm_fp8_grouped(act_q, w2.view(torch.uint8), h2,
+                             m_indptr, None, max_m_tiles, act_s,
                              w2_scale.contiguous(), 1.0)
     else:
+        a_perm, m_indptr, inv, token_of_copy = _permute(x, topk_ids, E)
+        h1 = _grouped_nt_bf16(a_perm, w13, m_indptr, max_m_tiles)
+        act = {"silu": silu_and_mul, "gelu": gelu_and_mul}[activation](h1)
         h2 = _grouped_nt_bf16(act, w2, m_indptr, max_m_tiles)
 
     # finalize: out[token] = sum_j weight[t, j] * h2[row of (t, j)]
-    inv = torch.empty_like(order)
-    inv[order] = torch.arange(R, device=x.device)
     res = out if out is not None else torch.empty(T, H, dtype=x.dtype, device=x.device)
-    get_ext().moe_finalize(h2, res, inv.view(T, k).to(torch.int32),
+    get_ext().moe_finalize(h2, res, inv.view(T, k),
                            topk_weights.float().contiguous())
     return res
 

@@ -170,3 +170,92 @@ def test_deep_gemm_contiguous_and_masked():
         ref = (a2[g, :mv].float() *
                sfa2[g, :mv].repeat_interleave(128, 1)) @ b_deq.t()
         torch.testing.assert_close(d2[g, :mv].float(), ref, atol=2e-1, rtol=5e-2)
+
+
+@pytest.mark.gpu
+def test_moe_routing_kernels_vs_torch():
+    """Fused routing kernels vs the torch reference formulas."""
+    import flashinfer_amd.fused_moe as fm
+
+    torch.manual_seed(0)
+    for T, E, k in [(64, 8, 2), (128, 64, 8), (33, 256, 8)]:
+        logits = torch.randn(T, E, device="cuda")
+        w_k, i_k = fm.moe_topk_softmax(logits, k)
+        # torch reference
+        probs = torch.softmax(logits.float(), dim=-1)
+        w_r, i_r = torch.topk(probs, k, dim=-1)
+        w_r = w_r / w_r.sum(-1, keepdim=True).clamp(min=1e-20)
+        assert torch.equal(torch.sort(i_k, -1).values,
+                           torch.sort(i_r.int(), -1).values)
+        torch.testing.assert_close(w_k.sort(-1).values, w_r.sort(-1).values,
+                                   atol=1e-5, rtol=1e-5)
+
+
+@pytest.mark.gpu
+def test_dsv3_routing_kernel_vs_torch():
+    import flashinfer_amd.fused_moe as fm
+
+    torch.manual_seed(1)
+    T, E, k, n_group, topk_group = 77, 256, 8, 8, 4
+    logits = torch.randn(T, E, device="cuda")
+    bias = torch.randn(E, device="cuda") * 0.1
+    w_k, i_k = fm.dsv3_routing(logits, k, n_group, topk_group, 2.5, bias)
+    # force the torch path by moving to CPU
+    w_r, i_r = fm.dsv3_routing(logits.cpu(), k, n_group, topk_group, 2.5,
+                               bias.cpu())
+    assert torch.equal(torch.sort(i_k.cpu(), -1).values,
+                       torch.sort(i_r.int(), -1).values)
+    torch.testing.assert_close(w_k.cpu().sort(-1).values,
+                               w_r.sort(-1).values, atol=1e-5, rtol=1e-5)
+
+
+@pytest.mark.gpu
+def test_moe_build_permute_kernel():
+    from flashinfer_amd.fused_moe import _build_permute
+
+    torch.manual_seed(2)
+    T, k, E = 512, 2, 8
+    ids = torch.randint(0, E, (T, k), dtype=torch.int32, device="cuda")
+    m_indptr, token_of_copy, inv = _build_permute(ids, E)
+    flat = ids.reshape(-1).long()
+    counts = torch.bincount(flat, minlength=E)
+    ref_indptr = torch.zeros(E + 1, dtype=torch.int64, device="cuda")
+    ref_indptr[1:] = counts.cumsum(0)
+    assert torch.equal(m_indptr.long(), ref_indptr)
+    # every permuted row's expert matches its segment; inv is a bijection
+    e_of_pos = flat[torch.argsort(inv.long())]  # expert of copy at pos p
+    for e in range(E):
+        seg = e_of_pos[int(ref_indptr[e]):int(ref_indptr[e + 1])]
+        assert (seg == e).all()
+    assert torch.equal(inv.long().sort().values,
+                       torch.arange(T * k, device="cuda"))
+    # token_of_copy consistency
+    tok = (torch.arange(T * k, device="cuda") // k)
+    assert torch.equal(token_of_copy.long()[inv.long()], tok)
+
+
+@pytest.mark.gpu
+def test_silu_mul_quant_and_gather_quant():
+    from flashinfer_amd._lib import get_ext
+
+    torch.manual_seed(3)
+    R, I = 64, 256
+    h = torch.randn(R, 2 * I, dtype=torch.bfloat16, device="cuda")
+    q = torch.empty(R, I, dtype=torch.uint8, device="cuda")
+    s = torch.empty(I // 128, R, dtype=torch.float32, device="cuda")
+    get_ext().silu_mul_quant_run(h, q, s, False)
+    gate, up = h.float().chunk(2, dim=-1)
+    ref = torch.nn.functional.silu(gate) * up
+    deq = q.view(torch.float8_e4m3fn).float() * \
+        s.t().repeat_interleave(128, dim=1)
+    torch.testing.assert_close(deq, ref, atol=0.08, rtol=0.08)
+
+    x = torch.randn(16, 256, dtype=torch.bfloat16, device="cuda")
+    toc = torch.randint(0, 16, (32,), dtype=torch.int32, device="cuda")
+    gq = torch.empty(32, 256, dtype=torch.uint8, device="cuda")
+    gs = torch.empty(2, 32, dtype=torch.float32, device="cuda")
+    get_ext().gather_quant_run(x, toc, gq, gs)
+    ref2 = x.float()[toc.long()]
+    deq2 = gq.view(torch.float8_e4m3fn).float() * \
+        gs.t().repeat_interleave(128, dim=1)
+    torch.testing.assert_close(deq2, ref2, atol=0.08, rtol=0.08)
