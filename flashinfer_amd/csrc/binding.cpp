@@ -87,7 +87,7 @@ hipError_t fi_gemm_fp8_grouped(const void* A, const void* W, void* C,
                                int64_t lda, int64_t ldw_n, int64_t ldw_seg, int64_t ldc,
                                const float* a_scales, const float* b_scales,
                                float scalar_scale, int64_t a_scale_stride,
-                               hipStream_t stream);
+                               int flat_tiles, hipStream_t stream);
 hipError_t fi_per_group_quant_fp8(int dtype, int trans_scale, const void* x, void* q,
                                   float* scale, int64_t rows, int K, int64_t stride_row,
                                   float eps, hipStream_t stream);
@@ -115,7 +115,7 @@ hipError_t fi_dsv3_routing(const float* logits, const float* bias, float* weight
 hipError_t fi_moe_build_permute(const int32_t* ids, int32_t* counts,
                                 int32_t* m_indptr, int32_t* cursor,
                                 int32_t* token_of_copy, int32_t* inv, int n,
-                                int k, int E, hipStream_t stream);
+                                int k, int E, int align, hipStream_t stream);
 hipError_t fi_gather_quant(int dtype, const void* src, const int32_t* token_of_copy,
                            uint8_t* dst, float* scale, int R, int K,
                            hipStream_t stream);
@@ -843,7 +843,8 @@ void group_gemm_nt(at::Tensor a, at::Tensor w, at::Tensor c, at::Tensor m_indptr
 void gemm_fp8_grouped(at::Tensor a, at::Tensor w, at::Tensor c, at::Tensor m_indptr,
                       c10::optional<at::Tensor> w_indices, int64_t max_m_tiles,
                       c10::optional<at::Tensor> a_scales,
-                      c10::optional<at::Tensor> b_scales, double scalar_scale) {
+                      c10::optional<at::Tensor> b_scales, double scalar_scale,
+                      int64_t flat_tiles = 0) {
   TORCH_CHECK(a.is_cuda() && a.dim() == 2 && a.stride(1) == 1);
   TORCH_CHECK(w.dim() == 3 && w.stride(2) == 1);
   TORCH_CHECK(c.scalar_type() == at::kBFloat16);
@@ -857,7 +858,8 @@ void gemm_fp8_grouped(at::Tensor a, at::Tensor w, at::Tensor c, at::Tensor m_ind
                 c.stride(0),
                 a_scales.has_value() ? a_scales->data_ptr<float>() : nullptr,
                 b_scales.has_value() ? b_scales->data_ptr<float>() : nullptr,
-                (float)scalar_scale, a_scale_stride, cur_stream(a)),
+                (float)scalar_scale, a_scale_stride, (int)flat_tiles,
+                cur_stream(a)),
             "fi_gemm_fp8_grouped");
 }
 
@@ -1150,13 +1152,13 @@ void dsv3_routing_run(at::Tensor logits, c10::optional<at::Tensor> bias,
 
 void moe_build_permute_run(at::Tensor ids, at::Tensor counts, at::Tensor m_indptr,
                            at::Tensor cursor, at::Tensor token_of_copy,
-                           at::Tensor inv) {
+                           at::Tensor inv, int64_t align) {
   int n = ids.numel(), k = ids.size(-1), E = counts.size(0);
   check_hip(fi_moe_build_permute(ids.data_ptr<int32_t>(), counts.data_ptr<int32_t>(),
                                  m_indptr.data_ptr<int32_t>(),
                                  cursor.data_ptr<int32_t>(),
                                  token_of_copy.data_ptr<int32_t>(),
-                                 inv.data_ptr<int32_t>(), n, k, E,
+                                 inv.data_ptr<int32_t>(), n, k, E, (int)align,
                                  cur_stream(ids)),
             "fi_moe_build_permute");
 }

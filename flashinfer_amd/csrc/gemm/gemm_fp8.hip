@@ -33,7 +33,12 @@ __device__ __forceinline__ void stage_tile(const uint8_t* __restrict__ gbase,
   for (int i = 0; i < 2; ++i) {
     int u = tid + i * NTH;
     uint32_t dst_byte = (uint32_t)u * 16;
-    uint32_t logical = swz128(dst_byte);
+    // swz256 over ROW PAIRS (two 128-B rows share a 256-B bank row): a
+    // 128-B-row swz128 leaves rows r/r+8 aliased -> 4-way conflicts on
+    // 32-row fragment reads (PMC: LDSBankConflict 17.7); the 4-bit spread
+    // de-conflicts all 32 rows. XOR is an involution, so pre-swizzling the
+    // GLOBAL address keeps the DMA destination linear (guide s09 pattern).
+    uint32_t logical = swz256(dst_byte);
     int row = logical >> 7;
     int col = logical & 127;
     int gm = row0 + row;
@@ -54,14 +59,31 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
     const int32_t* __restrict__ w_indices, int N, int K, int64_t lda, int64_t ldw_n,
     int64_t ldw_seg, int64_t ldc, const float* __restrict__ a_scales,  // [K/128, Mtot]
     const float* __restrict__ b_scales,  // [S, K/128, N/128]
-    float scalar_scale, int64_t a_scale_stride) {
+    float scalar_scale, int64_t a_scale_stride, int flat_segments) {
   __shared__ uint8_t As[2][BM * BK];
   __shared__ uint8_t Bs[2][BN * BK];
   __shared__ float ascale_s[2][BM];  // per-row activation scales for this kt
 
-  const int seg = blockIdx.z;
-  const int m0 = m_indptr[seg] + blockIdx.y * BM;
-  const int m_end = m_indptr[seg + 1];
+  // FLAT grid (num_segments passed > 0): blockIdx.y is a GLOBAL M tile over
+  // 128-aligned segments; binary-search the owning segment. Legacy grid
+  // (num_segments == 0): blockIdx.z is the segment, y the tile within —
+  // kept for unaligned segment GEMM callers (LoRA / plain groupwise).
+  int seg, m0, m_end;
+  if (flat_segments > 0) {
+    m0 = blockIdx.y * BM;
+    int lo = 0, hi = flat_segments - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (m_indptr[mid] <= m0) lo = mid;
+      else hi = mid - 1;
+    }
+    seg = lo;
+    m_end = m_indptr[seg + 1];
+  } else {
+    seg = blockIdx.z;
+    m0 = m_indptr[seg] + blockIdx.y * BM;
+    m_end = m_indptr[seg + 1];
+  }
   if (m0 >= m_end) return;
   const int bn0 = blockIdx.x * BN;
   const int widx = w_indices ? w_indices[seg] : seg;
@@ -70,10 +92,13 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
-  const int wm = (wid >> 2) * WM;
-  const int wn = (wid & 3) * WN;
+  // C^T wave tiling (32 m x 64 n per wave): the accumulator's LANE COLUMN is
+  // the M row, so the per-row activation scale is a per-lane SCALAR — the
+  // straight-C layout needed 32 ds_read_b32 of ascale_s per tile per wave
+  // (the DS-pipe hotspot at 839-919 TF; profiles/README r02 fp8 entry).
+  const int wm = (wid & 3) * 32;
+  const int wn = (wid >> 2) * 64;
   const int line = lane & 31;
-  const int khalf = (lane >> 5) * 8;
 
   floatx16 accm[2];
 #pragma unroll
@@ -117,8 +142,8 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
     };
     auto ld32 = [&](const char* base, int row, int kb) {
       frag32 u;
-      u.q[0] = *reinterpret_cast<const uint4*>(base + swz128(row * BK + kb));
-      u.q[1] = *reinterpret_cast<const uint4*>(base + swz128(row * BK + kb + 16));
+      u.q[0] = *reinterpret_cast<const uint4*>(base + swz256(row * BK + kb));
+      u.q[1] = *reinterpret_cast<const uint4*>(base + swz256(row * BK + kb + 16));
       return u.v;
     };
     const int kh32 = (lane >> 5) * 32;
@@ -126,23 +151,25 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
 #pragma unroll
     for (int g2 = 0; g2 < 2; ++g2) {  // two K=64 groups per BK=128 tile
       int kb = g2 * 64 + kh32;
-      intx8 bfrag = ld32(b_lds, wn + line, kb);
+      intx8 afrag = ld32(a_lds, wm + line, kb);  // B-operand: col = m
 #pragma unroll
       for (int i = 0; i < 2; ++i)
-        acc[i] = mfma_32x32x64_fp8(ld32(a_lds, wm + i * 32 + line, kb), bfrag,
+        acc[i] = mfma_32x32x64_fp8(ld32(b_lds, wn + i * 32 + line, kb), afrag,
                                    acc[i]);
     }
     __builtin_amdgcn_s_setprio(0);
-    // rescale local accumulator into master
+    // rescale local accumulator into master (per-lane scalar sa*sb)
     if constexpr (GROUPWISE) {
-      float sb = b_scales[((int64_t)seg * nk + kt) * ((N + 127) / 128) + (bn0 + wn) / 128];
+      float sb = b_scales[((int64_t)seg * nk + kt) * ((N + 127) / 128) + bn0 / 128];
+      float sab = ascale_s[cur][wm + line] * sb;
+      // packed f32 FMA halves the rescale VALU chain (v_pk_fma_f32)
+      typedef __attribute__((ext_vector_type(2))) float f32x2;
 #pragma unroll
       for (int i = 0; i < 2; ++i) {
+        f32x2* am = reinterpret_cast<f32x2*>(&accm[i]);
+        const f32x2* ac = reinterpret_cast<const f32x2*>(&acc[i]);
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          float sa = ascale_s[cur][wm + i * 32 + mfma32_cd_row(r, lane)];
-          accm[i][r] += acc[i][r] * (sa * sb);
-        }
+        for (int r = 0; r < 8; ++r) am[r] += ac[r] * sab;
       }
     } else {
 #pragma unroll
@@ -150,18 +177,23 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
 #pragma unroll
         for (int r = 0; r < 16; ++r) accm[i][r] += acc[i][r];
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
+    // barrier WITHOUT the compiler's forced vmcnt(0) drain: tile n+1's 4
+    // global_load_lds per thread stay in flight across it (tile n's landed
+    // once <=4 remain outstanding — in-order vmcnt). ascale_s ds_writes are
+    // covered by lgkmcnt. (guide: HIP emits vmcnt(0) before s_barrier,
+    // stalling ~20%; this is the AITER-style manual pipeline.)
+    asm volatile("s_waitcnt vmcnt(4) lgkmcnt(0)\n\ts_barrier" ::: "memory");
     cur ^= 1;
   }
 
   float fs = GROUPWISE ? 1.f : scalar_scale;
+  // C^T epilogue: lane column = m, accumulator rows = n
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
-      int m = m0 + wm + i * 32 + mfma32_cd_row(r, lane);
-      int n = bn0 + wn + mfma32_cd_col(lane);
+      int m = m0 + wm + mfma32_cd_col(lane);
+      int n = bn0 + wn + i * 32 + mfma32_cd_row(r, lane);
       if (m < m_end && n < N)
         C[(int64_t)m * ldc + n] = from_f32<bf16>(accm[i][r] * fs);
     }
@@ -177,20 +209,28 @@ extern "C" hipError_t fi_gemm_fp8_grouped(
     const int32_t* w_indices, int num_segments, int max_m_tiles, int N, int K,
     int64_t lda, int64_t ldw_n, int64_t ldw_seg, int64_t ldc, const float* a_scales,
     const float* b_scales, float scalar_scale, int64_t a_scale_stride,
-    hipStream_t stream) {
+    int flat_tiles, hipStream_t stream) {
   if (K % 128 != 0) return hipErrorInvalidValue;
-  dim3 grid((N + fi::f8gemm::BN - 1) / fi::f8gemm::BN, max_m_tiles, num_segments);
-  dim3 blk(fi::f8gemm::NTH);
+  // flat < 0 => legacy z-grid; flat >= 0 ignored sentinel handled by caller
+  dim3 grid, blk(fi::f8gemm::NTH);
+  int flat_segs = 0;
+  if (flat_tiles > 0) {
+    grid = dim3((N + fi::f8gemm::BN - 1) / fi::f8gemm::BN, flat_tiles, 1);
+    flat_segs = num_segments;
+  } else {
+    grid = dim3((N + fi::f8gemm::BN - 1) / fi::f8gemm::BN, max_m_tiles,
+                num_segments);
+  }
   if (a_scales && b_scales) {
     hipLaunchKernelGGL((fi::f8gemm::gemm_fp8_kernel<true>), grid, blk, 0, stream,
                        (const uint8_t*)A, (const uint8_t*)W, (fi::bf16*)C, m_indptr,
                        w_indices, N, K, lda, ldw_n, ldw_seg, ldc, a_scales, b_scales,
-                       scalar_scale, a_scale_stride);
+                       scalar_scale, a_scale_stride, flat_segs);
   } else {
     hipLaunchKernelGGL((fi::f8gemm::gemm_fp8_kernel<false>), grid, blk, 0, stream,
                        (const uint8_t*)A, (const uint8_t*)W, (fi::bf16*)C, m_indptr,
                        w_indices, N, K, lda, ldw_n, ldw_seg, ldc, a_scales, b_scales,
-                       scalar_scale, a_scale_stride);
+                       scalar_scale, a_scale_stride, flat_segs);
   }
   return hipGetLastError();
 }

@@ -172,15 +172,19 @@ __global__ void moe_hist_kernel(const int32_t* __restrict__ ids,
   if (i < n) atomicAdd(counts + ids[i], 1);
 }
 
-// one workgroup: exclusive scan of counts into m_indptr, zero the cursors
+// one workgroup: exclusive scan of counts into m_indptr (segment starts
+// 128-aligned so every GEMM M-tile belongs to exactly ONE expert — the
+// grouped GEMM runs a flat tile grid with a per-WG segment binary search
+// instead of a z-dim padded to the worst-case tile count), zero the cursors
 __global__ void moe_scan_kernel(const int32_t* __restrict__ counts,
                                 int32_t* __restrict__ m_indptr,
-                                int32_t* __restrict__ cursor, int E) {
+                                int32_t* __restrict__ cursor, int E, int align) {
   if (threadIdx.x == 0) {
     int acc = 0;
     m_indptr[0] = 0;
     for (int e = 0; e < E; ++e) {
       acc += counts[e];
+      if (align > 1) acc = (acc + align - 1) / align * align;
       m_indptr[e + 1] = acc;
     }
   }
@@ -299,11 +303,11 @@ hipError_t fi_dsv3_routing(const float* logits, const float* bias, float* weight
 hipError_t fi_moe_build_permute(const int32_t* ids, int32_t* counts,
                                 int32_t* m_indptr, int32_t* cursor,
                                 int32_t* token_of_copy, int32_t* inv, int n,
-                                int k, int E, hipStream_t stream) {
+                                int k, int E, int align, hipStream_t stream) {
   hipLaunchKernelGGL(fi::moe_hist_kernel, dim3((n + 255) / 256), dim3(256), 0,
                      stream, ids, counts, n);
   hipLaunchKernelGGL(fi::moe_scan_kernel, dim3(1), dim3(256), 0, stream, counts,
-                     m_indptr, cursor, E);
+                     m_indptr, cursor, E, align);
   hipLaunchKernelGGL(fi::moe_scatter_kernel, dim3((n + 255) / 256), dim3(256), 0,
                      stream, ids, m_indptr, cursor, token_of_copy, inv, n, k);
   return hipGetLastError();

@@ -65,7 +65,7 @@ def gemm_fp8_nt_groupwise(
     m_indptr = torch.tensor([0, M], dtype=torch.int32, device=a.device)
     get_ext().gemm_fp8_grouped(
         a.view(torch.uint8), b.view(torch.uint8).unsqueeze(0), out, m_indptr, None,
-        ceil_div(M, 128), a_scale.contiguous(), b_scale.unsqueeze(0).contiguous(), 1.0,
+        ceil_div(M, 128), a_scale.contiguous(), b_scale.unsqueeze(0).contiguous(), 1.0, 0,
     )
     return out
 
@@ -85,7 +85,7 @@ def group_gemm_fp8_nt_groupwise(
     get_ext().gemm_fp8_grouped(
         a.view(torch.uint8), b.view(torch.uint8), out,
         m_indptr.to(torch.int32), None, ceil_div(M, 128) + m_indptr.numel(),
-        a_scale.contiguous(), b_scale.contiguous(), 1.0,
+        a_scale.contiguous(), b_scale.contiguous(), 1.0, 0,
     )
     return out
 
@@ -108,7 +108,7 @@ def bmm_fp8(
     )
     get_ext().gemm_fp8_grouped(
         A.reshape(Bb * M, K).view(torch.uint8), b_nt.view(torch.uint8), out.view(Bb * M, N),
-        m_indptr, None, ceil_div(M, 128), None, None, scale,
+        m_indptr, None, ceil_div(M, 128), None, None, scale, 0,
     )
     return out
 

@@ -97,7 +97,7 @@ def _grouped_nt_bf16(a, w, m_indptr, max_m_tiles):
     return out
 
 
-def _build_permute(topk_ids: torch.Tensor, num_experts: int):
+def _build_permute(topk_ids: torch.Tensor, num_experts: int, align: int = 1):
     """hist -> scan -> atomic scatter kernels (replaces the torch
     argsort/bincount/cumsum chain — reference moe_kernels.h expand stage).
     Returns (m_indptr, token_of_copy, inv) where inv[t, j] is the permuted
@@ -107,13 +107,16 @@ def _build_permute(topk_ids: torch.Tensor, num_experts: int):
     T, k = topk_ids.shape
     dev = topk_ids.device
     ids = topk_ids.to(torch.int32).contiguous()
+    rmax = T * k + (num_experts * (align - 1) if align > 1 else 0)
     counts = torch.zeros(num_experts, dtype=torch.int32, device=dev)
     cursor = torch.empty(num_experts, dtype=torch.int32, device=dev)
     m_indptr = torch.empty(num_experts + 1, dtype=torch.int32, device=dev)
-    token_of_copy = torch.empty(T * k, dtype=torch.int32, device=dev)
+    # pad rows gather token 0 (valid data, outputs never read) — zeros, not
+    # empty, so no NaN can enter the pipeline
+    token_of_copy = torch.zeros(rmax, dtype=torch.int32, device=dev)
     inv = torch.empty(T * k, dtype=torch.int32, device=dev)
     get_ext().moe_build_permute_run(ids, counts, m_indptr, cursor,
-                                    token_of_copy, inv)
+                                    token_of_copy, inv, align)
     return m_indptr, token_of_copy, inv
 
 
@@ -121,7 +124,7 @@ def _permute(x: torch.Tensor, topk_ids: torch.Tensor, num_experts: int):
     m_indptr, token_of_copy, inv = _build_permute(topk_ids, num_experts)
     a_perm = torch.empty(topk_ids.numel(), x.shape[1], dtype=x.dtype,
                          device=x.device)
-    get_ext().gather_rows(x, a_perm, token_of_copy)
+    get_ext().gather_rows(x, a_perm, token_of_copy[: topk_ids.numel()])
     return a_perm, m_indptr, inv, token_of_copy
 
 
@@ -153,23 +156,32 @@ def fused_moe(
 
     if fp8:
         # fused data movers: gather+quant feeds GEMM1, silu_mul+quant feeds
-        # GEMM2 — the bf16 intermediate round-trips are gone
-        m_indptr, token_of_copy, inv = _build_permute(topk_ids, E)
-        a_q = torch.empty(R, H, dtype=torch.uint8, device=x.device)
-        a_s = torch.empty(H // 128, R, dtype=torch.float32, device=x.device)
+        # GEMM2 — the bf16 intermediate round-trips are gone. At high expert
+        # counts the segments are 128-aligned and the grouped GEMM runs a
+        # FLAT tile grid (the z-grid pads every expert to the worst-case
+        # tile count: E x total_tiles dead dispatches); at small E the pad
+        # rows cost more than the dead WGs (measured: E=8 1.455 -> 1.405
+        # M tok/s), so the z-grid stays.
+        use_flat = E >= 16
+        m_indptr, token_of_copy, inv = _build_permute(
+            topk_ids, E, align=128 if use_flat else 1)
+        Rp = token_of_copy.shape[0]
+        flat_tiles = Rp // 128 if use_flat else 0
+        a_q = torch.empty(Rp, H, dtype=torch.uint8, device=x.device)
+        a_s = torch.empty(H // 128, Rp, dtype=torch.float32, device=x.device)
         ext.gather_quant_run(x, token_of_copy, a_q, a_s)
-        h1 = torch.empty(R, I2, dtype=torch.bfloat16, device=x.device)
+        h1 = torch.empty(Rp, I2, dtype=torch.bfloat16, device=x.device)
         ext.gemm_fp8_grouped(a_q, w13.view(torch.uint8), h1,
                              m_indptr, None, max_m_tiles, a_s,
-                             w13_scale.contiguous(), 1.0)
-        act_q = torch.empty(R, inter, dtype=torch.uint8, device=x.device)
-        act_s = torch.empty(inter // 128, R, dtype=torch.float32,
+                             w13_scale.contiguous(), 1.0, flat_tiles)
+        act_q = torch.empty(Rp, inter, dtype=torch.uint8, device=x.device)
+        act_s = torch.empty(inter // 128, Rp, dtype=torch.float32,
                             device=x.device)
         ext.silu_mul_quant_run(h1, act_q, act_s, activation == "gelu")
-        h2 = torch.empty(R, H, dtype=torch.bfloat16, device=x.device)
+        h2 = torch.empty(Rp, H, dtype=torch.bfloat16, device=x.device)
         ext.gemm_fp8_grouped(act_q, w2.view(torch.uint8), h2,
                              m_indptr, None, max_m_tiles, act_s,
-                             w2_scale.contiguous(), 1.0)
+                             w2_scale.contiguous(), 1.0, flat_tiles)
     else:
         a_perm, m_indptr, inv, token_of_copy = _permute(x, topk_ids, E)
         h1 = _grouped_nt_bf16(a_perm, w13, m_indptr, max_m_tiles)

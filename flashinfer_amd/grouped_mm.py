@@ -67,6 +67,6 @@ def grouped_mm_fp8(
     get_ext().gemm_fp8_grouped(
         a.view(torch.uint8), b.view(torch.uint8), out,
         m_indptr.to(dev, torch.int32), None, ceil_div(cum_m, 128) + E,
-        a_scale, b_scale, 1.0,
+        a_scale, b_scale, 1.0, 0,
     )
     return out
