@@ -30,17 +30,31 @@ class CustomAllReduce:
         out = ar.all_reduce_rmsnorm(x, residual, w)  # fused epilogue
     """
 
+    # one-shot (each rank reads all peers' full buffers) wins while the
+    # message is latency-bound; above this the two-shot RS+AG (each rank
+    # pulls only its 1/world shard, then the reduced shards) wins on
+    # traffic (~2x numel vs world x numel). Default from the same-die
+    # probe; override with FI_AR_TWO_SHOT_BYTES, re-tune on real xGMI.
+    TWO_SHOT_BYTES = 512 * 1024
+
     def __init__(self, max_bytes: int, group=None, device: str = "cuda",
                  spin_limit: int = 1 << 26):
+        import os
+
         self.group = group
         self.world = dist.get_world_size(group)
         self.rank = dist.get_rank(group)
         if self.world > 8:
             raise ValueError("one node = up to 8 ranks")
         self.max_bytes = max_bytes
-        # two data slots (seq parity): a rank may run one call ahead of the
-        # slowest peer, so its next write must land in the other slot
-        self.bufs = create_shared_buffer(FLAG_BYTES + 2 * max_bytes, group)
+        self.two_shot_bytes = int(
+            os.environ.get("FI_AR_TWO_SHOT_BYTES", self.TWO_SHOT_BYTES))
+        # two data slots (seq parity) + two reduced-shard slots: a rank may
+        # run one call ahead of the slowest peer, so every write lands in
+        # the slot of the other parity
+        self._shard_bytes = max_bytes  # shard slot (worst case world=1)
+        self.bufs = create_shared_buffer(
+            FLAG_BYTES + 2 * max_bytes + 2 * self._shard_bytes, group)
         self.seq = 0
         self.spin_limit = spin_limit
         self.error_flag = torch.zeros(1, dtype=torch.int32, device=device)
@@ -71,13 +85,25 @@ class CustomAllReduce:
                 "one-shot allreduce spin timeout: a peer never arrived")
 
     def all_reduce(self, x: torch.Tensor,
-                   out: Optional[torch.Tensor] = None) -> torch.Tensor:
+                   out: Optional[torch.Tensor] = None,
+                   strategy: Optional[str] = None) -> torch.Tensor:
+        nbytes = x.numel() * x.element_size()
         off = self._push_input(x)
         if out is None:
             out = torch.empty_like(x)
-        get_ext().one_shot_all_reduce(out.view(-1), self.bufs, self.rank,
-                                      self.seq, self.error_flag,
-                                      self.spin_limit, off)
+        if strategy is None:
+            strategy = ("two_shot" if nbytes > self.two_shot_bytes
+                        and self.world > 1 else "one_shot")
+        if strategy == "two_shot":
+            shard_off = (FLAG_BYTES + 2 * self.max_bytes
+                         + (self.seq % 2) * self._shard_bytes)
+            get_ext().two_shot_all_reduce(out.view(-1), self.bufs, self.rank,
+                                          self.seq, self.error_flag,
+                                          self.spin_limit, off, shard_off)
+        else:
+            get_ext().one_shot_all_reduce(out.view(-1), self.bufs, self.rank,
+                                          self.seq, self.error_flag,
+                                          self.spin_limit, off)
         return out
 
     def all_reduce_rmsnorm(self, x: torch.Tensor, residual: Optional[torch.Tensor],

@@ -130,6 +130,8 @@ hipError_t fi_gdn_decode(int dtype, int state_f32, int per_channel_gate, void* s
                          hipStream_t stream);
 hipError_t fi_one_shot_ar(int dtype, fi::ARParams* p, void* out, int64_t numel,
                           hipStream_t stream);
+hipError_t fi_two_shot_ar(int dtype, fi_ext::ARParams* p, unsigned long long shard_off,
+                          void* out, int64_t numel, hipStream_t stream);
 hipError_t fi_one_shot_ar_rmsnorm(int dtype, fi::ARParams* p, void* out,
                                   void* residual, const void* weight, int rows,
                                   int d, float eps, hipStream_t stream);
@@ -1003,6 +1005,15 @@ void one_shot_all_reduce(at::Tensor out, std::vector<int64_t> bufs, int64_t rank
             "fi_one_shot_ar");
 }
 
+void two_shot_all_reduce(at::Tensor out, std::vector<int64_t> bufs, int64_t rank,
+                         int64_t seq, at::Tensor error_flag, int64_t spin_limit,
+                         int64_t data_off, int64_t shard_off) {
+  fi::ARParams p = make_ar_params(bufs, rank, seq, error_flag, spin_limit, data_off);
+  check_hip(fi_two_shot_ar(dtype_code(out), &p, (unsigned long long)shard_off,
+                           out.data_ptr(), out.numel(), cur_stream(out)),
+            "fi_two_shot_ar");
+}
+
 void one_shot_all_reduce_rmsnorm(at::Tensor out, c10::optional<at::Tensor> residual,
                                  at::Tensor weight, std::vector<int64_t> bufs,
                                  int64_t rank, int64_t seq, at::Tensor error_flag,
@@ -1227,6 +1238,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("concat_mla_k", &concat_mla_k);
   m.def("ipc_memcpy_to", &ipc_memcpy_to);
   m.def("one_shot_all_reduce", &one_shot_all_reduce);
+  m.def("two_shot_all_reduce", &two_shot_all_reduce);
   m.def("one_shot_all_reduce_rmsnorm", &one_shot_all_reduce_rmsnorm);
   m.def("debug_fastdiv", &debug_fastdiv);
 }

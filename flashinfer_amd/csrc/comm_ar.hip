@@ -139,3 +139,97 @@ extern "C" hipError_t fi_one_shot_ar_rmsnorm(int dtype, fi::ARParams* p, void* o
 #undef LAR
   return hipGetLastError();
 }
+
+// ------------------- two-shot RS + AG allreduce -------------------
+// Reference role: trtllm_allreduce.cuh twoShotAllReduceKernel:1279, but
+// shaped for xGMI point-to-point: phase 1 (reduce-scatter) has every rank
+// PULL its own 1/world shard from all 7 peers concurrently (all links busy,
+// total remote traffic ~ numel vs the one-shot's 7x numel); phase 2
+// (all-gather) pulls each peer's reduced shard. The phases are two kernel
+// launches — the kernel boundary is the device-wide sync that makes the
+// shard complete before its flag publishes. Shard slots are double-buffered
+// by seq parity like the data slots; flags[1] is the shard-ready sequence.
+
+namespace fi {
+
+// reduce own shard into own shard slot
+template <typename T>
+__global__ void two_shot_rs_kernel(ARParams p, unsigned long long shard_off,
+                                   int64_t numel) {
+  __shared__ int s_ok;
+  if (!ar_sync(p, &s_ok)) return;
+  int64_t per = (numel + p.world - 1) / p.world;
+  int64_t s0 = (int64_t)p.rank * per;
+  int64_t s1 = s0 + per < numel ? s0 + per : numel;
+  T* dst = (T*)(p.bufs[p.rank] + shard_off);
+  for (int64_t i = s0 + blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < s1;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float acc = 0.f;
+    for (int r = 0; r < p.world; ++r) acc += to_f32<T>(ar_data<T>(p, r)[i]);
+    dst[i - s0] = from_f32<T>(acc);
+  }
+}
+
+// publish shard-ready, wait peers, gather shards into out
+template <typename T>
+__global__ void two_shot_ag_kernel(ARParams p, unsigned long long shard_off,
+                                   T* __restrict__ out, int64_t numel) {
+  __shared__ int s_ok;
+  if (threadIdx.x == 0) s_ok = 1;
+  __syncthreads();
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    __threadfence_system();
+    atomicExch(ar_flag(p.bufs[p.rank]) + 1, p.seq);
+  }
+  if (threadIdx.x < kMaxRanks && (int)threadIdx.x < p.world) {
+    int r = threadIdx.x;
+    unsigned long long it = 0;
+    while (atomicAdd(ar_flag(p.bufs[r]) + 1, 0ull) < p.seq) {
+      if (++it > p.spin_limit) {
+        atomicExch(p.error_flag, 1);
+        s_ok = 0;
+        break;
+      }
+      __builtin_amdgcn_s_sleep(8);
+    }
+  }
+  __syncthreads();
+  __threadfence_system();
+  if (!s_ok) return;
+  int64_t per = (numel + p.world - 1) / p.world;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < numel;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int r = (int)(i / per);
+    const T* src = (const T*)(p.bufs[r] + shard_off);
+    out[i] = src[i - (int64_t)r * per];
+  }
+}
+
+}  // namespace fi
+
+extern "C" hipError_t fi_two_shot_ar(int dtype, fi::ARParams* p,
+                                     unsigned long long shard_off, void* out,
+                                     int64_t numel, hipStream_t stream) {
+  int grid = (int)((numel / (p->world > 0 ? p->world : 1) + 255) / 256);
+  if (grid > 256) grid = 256;
+  if (grid == 0) grid = 1;
+  dim3 g(grid), g2(grid * (p->world > 0 ? p->world : 1) > 256
+                       ? 256
+                       : grid * (p->world > 0 ? p->world : 1));
+  dim3 blk(256);
+#define LTS(T)                                                                   \
+  do {                                                                           \
+    hipLaunchKernelGGL((fi::two_shot_rs_kernel<T>), g, blk, 0, stream, *p,       \
+                       shard_off, numel);                                        \
+    hipLaunchKernelGGL((fi::two_shot_ag_kernel<T>), g2, blk, 0, stream, *p,      \
+                       shard_off, (T*)out, numel);                               \
+  } while (0)
+  switch (dtype) {
+    case 0: LTS(fi::bf16); break;
+    case 1: LTS(fi::fp16); break;
+    case 2: LTS(float); break;
+    default: return hipErrorInvalidValue;
+  }
+#undef LTS
+  return hipGetLastError();
+}

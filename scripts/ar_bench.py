@@ -20,19 +20,27 @@ def worker(rank, world, port):
     dist.init_process_group("gloo", rank=rank, world_size=world)
     torch.cuda.set_device(0)
     ar = CustomAllReduce(max_bytes=32 << 20, spin_limit=1 << 26)
-    for numel in (4096, 65536, 1 << 20, 4 << 20):
+    for numel in (4096, 65536, 1 << 20, 4 << 20, 16 << 20):
         x = torch.randn(numel, dtype=torch.bfloat16, device="cuda")
-        for _ in range(5):
-            ar.all_reduce(x)
-        torch.cuda.synchronize()
-        dist.barrier()
-        t0 = time.perf_counter()
-        for _ in range(20):
-            ar.all_reduce(x)
-        torch.cuda.synchronize()
-        dt = (time.perf_counter() - t0) / 20
-        if rank == 0:
-            print(f"one-shot AR world=2 {numel*2/1024:.0f} KiB: {dt*1e6:.1f} us")
+        ref = x.clone()
+        dist.all_reduce(ref)  # gloo reference (CPU path ok for check)
+        for strat in ("one_shot", "two_shot"):
+            y = ar.all_reduce(x, strategy=strat)
+            torch.cuda.synchronize()
+            err = (y.float() - world * x.float()).abs().max().item()
+            assert err < 0.1, f"{strat} numel={numel} err={err}"
+            for _ in range(5):
+                ar.all_reduce(x, strategy=strat)
+            torch.cuda.synchronize()
+            dist.barrier()
+            t0 = time.perf_counter()
+            for _ in range(20):
+                ar.all_reduce(x, strategy=strat)
+            torch.cuda.synchronize()
+            dt = (time.perf_counter() - t0) / 20
+            if rank == 0:
+                print(f"{strat} AR world={world} {numel*2/1024:.0f} KiB: "
+                      f"{dt*1e6:.1f} us  {numel*2/dt/1e9:.1f} GB/s")
     ar.close()
     dist.destroy_process_group()
 
