@@ -21,13 +21,14 @@ def worker(rank, world, port):
     torch.cuda.set_device(0)
     ar = CustomAllReduce(max_bytes=32 << 20, spin_limit=1 << 26)
     for numel in (4096, 65536, 1 << 20, 4 << 20, 16 << 20):
+        torch.manual_seed(7 + rank)
         x = torch.randn(numel, dtype=torch.bfloat16, device="cuda")
-        ref = x.clone()
-        dist.all_reduce(ref)  # gloo reference (CPU path ok for check)
+        ref_c = x.float().cpu()
+        dist.all_reduce(ref_c)  # gloo CPU reference
         for strat in ("one_shot", "two_shot"):
             y = ar.all_reduce(x, strategy=strat)
             torch.cuda.synchronize()
-            err = (y.float() - world * x.float()).abs().max().item()
+            err = (y.float().cpu() - ref_c).abs().max().item()
             assert err < 0.1, f"{strat} numel={numel} err={err}"
             for _ in range(5):
                 ar.all_reduce(x, strategy=strat)
