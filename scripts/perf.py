@@ -283,6 +283,30 @@ def bench_holistic(n_prefill=8, n_decode=120, s_prefill=1024, kv_decode=1024,
 
 
 
+def bench_mla_prefill(bs=16, s=1024, H=128, page=32):
+    """Full chunked MLA prefill (qo = kv = s) — reference mla.cuh:976 role."""
+    import math
+    torch.manual_seed(0)
+    pages_per = (s + page - 1) // page
+    kv_indptr = torch.arange(0, (bs + 1) * pages_per, pages_per, dtype=torch.int32, device="cuda")
+    npages = bs * pages_per
+    kv_indices = torch.randperm(npages, dtype=torch.int32, device="cuda")
+    kv_len_arr = torch.full((bs,), s, dtype=torch.int32, device="cuda")
+    qo_indptr = torch.arange(0, (bs + 1) * s, s, dtype=torch.int32, device="cuda")
+    ckv = torch.randn(npages, page, 512, dtype=torch.bfloat16, device="cuda") / 4
+    kpe = torch.randn(npages, page, 64, dtype=torch.bfloat16, device="cuda") / 4
+    q_nope = torch.randn(bs * s, H, 512, dtype=torch.bfloat16, device="cuda") / 4
+    q_pe = torch.randn(bs * s, H, 64, dtype=torch.bfloat16, device="cuda") / 4
+    ws = torch.empty(4096 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchMLAPagedAttentionWrapper(ws)
+    sm = 1.0 / math.sqrt(576)
+    w.plan(qo_indptr, kv_indptr, kv_indices, kv_len_arr, H, 512, 64, page, True, sm, torch.bfloat16)
+    out = torch.empty(bs * s, H, 512, dtype=torch.bfloat16, device="cuda")
+    t = timeit(lambda: w.run(q_nope, q_pe, ckv, kpe, out=out), iters=10, warmup=3)
+    fl = bs * H * s * s / 2 * 2 * (576 + 512)
+    print(f"mla prefill bs={bs} s={s} H={H}: {t*1e3:.2f} ms  {fl/t/1e12:.1f} TFLOPS")
+
+
 if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "all"
     if which in ("all", "prefill"):
@@ -310,6 +334,8 @@ if __name__ == "__main__":
     if which in ("all", "mla"):
         bench_mla()
         bench_mla(bs=64, kv=4096)
+        bench_mla_prefill(bs=1)
+        bench_mla_prefill(bs=16)
     if which in ("all", "holistic"):
         bench_holistic()
         bench_holistic(n_prefill=2, n_decode=200, kv_decode=2048)

@@ -26,6 +26,8 @@ def mla_ref(q_nope, q_pe, ckv, kpe, sm_scale, causal, qo_len, kv_len):
     ([64, 129, 1000], [1, 1, 1], True),
     ([2048], [1], True),
     ([33, 80], [2, 4], True),   # incremental prefill (speculative)
+    ([1024], [1024], True),     # full chunked MLA prefill
+    ([1024, 700], [1024, 300], True),  # mixed full + partial prefill
 ])
 @pytest.mark.parametrize("H", [16, 128])
 def test_mla_paged_decode(kv_lens, qo_lens, causal, H):
