@@ -591,7 +591,8 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                        int64_t kv_chunk = 0,
                        c10::optional<at::Tensor> req_slot_base = c10::nullopt,
                        c10::optional<at::Tensor> tmp_v = c10::nullopt,
-                       c10::optional<at::Tensor> tmp_s = c10::nullopt) {
+                       c10::optional<at::Tensor> tmp_s = c10::nullopt,
+                       int64_t custom_fn = 0) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 3, "q must be [nnz, Hq, D]");
   TORCH_CHECK(q.stride(2) == 1 && out.stride(2) == 1);
   fi_ext::PrefillParams p{};
@@ -665,6 +666,16 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
     p.req_slot_base = req_slot_base->data_ptr<int32_t>();
     p.tmp_v = (float*)tmp_v->data_ptr();
     p.tmp_s = tmp_s->data_ptr<float>();
+  }
+  if (custom_fn) {
+    // JIT attention variant: dispatch through the custom module's entry
+    // (same signature as fi_batch_prefill; fn resolved by ctypes in
+    // flashinfer_amd/jit/attention.py)
+    auto fn = (hipError_t (*)(int, fi_ext::PrefillParams*, int, hipStream_t))
+        (void*)custom_fn;
+    check_hip(fn(dtype_code(q), &p, paged ? 1 : 0, cur_stream(q)),
+              "fi_batch_prefill_custom");
+    return;
   }
   check_hip(fi_batch_prefill(dtype_code(q), dtype_code(k_cache), &p, paged ? 1 : 0,
                              cur_stream(q)),

@@ -14,6 +14,26 @@ namespace fi {
 
 constexpr int kVTileStride = 72;  // V subtile stride in elems (144 B)
 
+// Attention-variant hook (reference variants.cuh:31 DefaultAttention /
+// variant_helper.cuh:75 + the csrc/*_customize_config.jinja injection
+// mechanism): user-supplied device code compiled into the kernel via
+// flashinfer_amd.jit.gen_customize_batch_prefill_module. kActive=false
+// keeps the default instantiations codegen-identical (the interior
+// fast path stays).
+struct NoVariant {
+  static constexpr bool kActive = false;
+  __device__ static float logits_transform(float s, int /*qo_idx*/,
+                                           int64_t /*kv_idx*/, int /*head*/,
+                                           int /*qo_len*/, int64_t /*kv_len*/) {
+    return s;
+  }
+  __device__ static bool logits_mask(int /*qo_idx*/, int64_t /*kv_idx*/,
+                                     int /*head*/, int /*qo_len*/,
+                                     int64_t /*kv_len*/) {
+    return true;
+  }
+};
+
 constexpr int KVB = 64;      // prefill kv tile
 constexpr float kLog2e = 1.4426950408889634f;
 
@@ -53,7 +73,7 @@ constexpr int prefill_tile_smem_bytes() {
 // Shared by the standalone batch_prefill kernel and the persistent holistic
 // BatchAttention kernel (reference attention/persistent.cuh role).
 template <typename T, typename TKV, int HDQK, int HDVO, int CTAQ, bool PAGED,
-          bool CAUSAL, bool MASK = false>
+          bool CAUSAL, bool MASK = false, typename VARIANT = NoVariant>
 __device__ __forceinline__ void prefill_tile_body(const PrefillParams& p, int req,
                                                   int qstart, int kv_head, int chunk_in,
                                                   char* smem) {
@@ -341,7 +361,8 @@ __device__ __forceinline__ void prefill_tile_body(const PrefillParams& p, int re
       // per-element predicate work (the VALU hotspot) is skipped.
       float pr[16];
       const int64_t kvt0 = kv0 + kt * 32;
-      bool tile_full = (kvt0 + 32 <= kv_valid_hi) && !MASK && cap <= 0.f && !p.alibi;
+      bool tile_full = (kvt0 + 32 <= kv_valid_hi) && !MASK && cap <= 0.f &&
+                       !p.alibi && !VARIANT::kActive;
       if constexpr (CAUSAL) {
         int wave_min_qpos = wq0 / (int)group;
         tile_full &= (kvt0 + 32 <= wave_min_qpos + diag + 1);
@@ -367,8 +388,13 @@ __device__ __forceinline__ void prefill_tile_body(const PrefillParams& p, int re
           float sv = acc_s[r] * p.sm_scale;
           if (cap > 0.f) sv = cap * tanhf(sv / cap);
           if (p.alibi) sv -= slope * (float)(my_qpos + diag - kv);
-          sv *= kLog2e;
           bool ok = kv < kv_valid_hi;
+          if constexpr (VARIANT::kActive) {
+            int head = (int)(kv_head * group + my_g_u);
+            sv = VARIANT::logits_transform(sv, my_qpos, kv, head, qo_len, kv_len);
+            ok &= VARIANT::logits_mask(my_qpos, kv, head, qo_len, kv_len);
+          }
+          sv *= kLog2e;
           if constexpr (CAUSAL) ok &= kv <= my_qpos + diag;
           if (wleft >= 0) ok &= kv >= my_qpos + diag - wleft;
           if constexpr (MASK) {

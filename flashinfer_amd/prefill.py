@@ -95,12 +95,15 @@ def _plan_tiles_split(qo_lens, kv_lens, group, causal, num_kv_heads,
 
 class _BatchPrefillBase:
     def __init__(self, float_workspace_buffer, kv_layout="NHD", use_cuda_graph=False,
-                 backend="fa2", jit_args=None, **kwargs):
+                 backend="fa2", jit_args=None, jit_module=None, **kwargs):
         self._float_workspace_buffer = float_workspace_buffer
         self.device = float_workspace_buffer.device
         self._kv_layout = kv_layout
         self._use_cuda_graph = use_cuda_graph
         self._plan_info = None
+        # JIT attention variant (reference customize-config mechanism):
+        # a module from jit.attention.gen_customize_batch_prefill_module
+        self._custom_fn = jit_module.run_ptr if jit_module is not None else 0
 
     @property
     def is_cuda_graph_enabled(self):
@@ -227,6 +230,7 @@ class _BatchPrefillBase:
             self._req_slot_base if split else None,
             self._tmp_v if split else None,
             self._tmp_s if split else None,
+            getattr(self, "_custom_fn", 0),
         )
         if split:
             # LSE merge of the per-chunk partials (cascade merge kernel)
@@ -372,7 +376,7 @@ def single_prefill_with_kv_cache(
         sm_scale if sm_scale is not None else default_sm_scale(D),
         float(logits_soft_cap or 0.0), window_left, causal, False, cta_q,
         mask_data, mask_indptr, pos_encoding_mode == "ALIBI", 1.0, 1.0, None,
-        None, 0, None, None, None,
+        None, 0, None, None, None, 0,
     )
     return (out, lse) if return_lse else out
 
