@@ -177,12 +177,14 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
 #pragma unroll
         for (int r = 0; r < 16; ++r) accm[i][r] += acc[i][r];
     }
-    // barrier WITHOUT the compiler's forced vmcnt(0) drain: tile n+1's 4
-    // global_load_lds per thread stay in flight across it (tile n's landed
-    // once <=4 remain outstanding — in-order vmcnt). ascale_s ds_writes are
-    // covered by lgkmcnt. (guide: HIP emits vmcnt(0) before s_barrier,
-    // stalling ~20%; this is the AITER-style manual pipeline.)
-    asm volatile("s_waitcnt vmcnt(4) lgkmcnt(0)\n\ts_barrier" ::: "memory");
+    // Full vmcnt drain before the barrier is REQUIRED here: the prefetch
+    // distance is one tile (double-buffered LDS), so the tile read next
+    // iteration is the one whose DMAs are still in flight at this barrier —
+    // a vmcnt(4) partial wait was tried (+46 TF) and raced (caught by the
+    // 64-expert MoE e2e test: nondeterministic 8-column stripes). Deeper
+    // pipelining needs a third LDS buffer, not a looser wait.
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
     cur ^= 1;
   }
 
