@@ -154,19 +154,30 @@ __device__ __forceinline__ void prefill_tile_body(const PrefillParams& p, int re
     if (chi < kv_valid_hi) kv_valid_hi = chi;
   }
 
-  // ---- load Q fragments (pre-fold nothing; scale applied post-MFMA) ----
+  // ---- load Q fragments, PRE-SCALED by sm_scale*log2e: S^T comes out of
+  // the MFMA already in the base-2 logit domain, removing 16 VALU multiplies
+  // per 32-kv subtile from the softmax chain (PMC r01: VALUBusy 42.7 vs
+  // MfmaUtil 16.4 — the logits pipeline is the bound). bf16 precision is
+  // relative, so pre-scaling costs no accuracy. The soft-cap/ALiBi/variant
+  // branch un-scales where it needs the natural domain. ----
   using frag = typename mfma_ab_frag<T>::type;
   frag qf[KCH];
   {
     const T* qptr = (const T*)p.q +
                     (int64_t)(qo_begin + my_qpos) * p.q_stride_n +
                     (int64_t)(kv_head * group + my_g_u) * p.q_stride_h;
+    const float qs = p.sm_scale * kLog2e;
 #pragma unroll
     for (int c = 0; c < KCH; ++c) {
       if (row_valid) {
         qf[c] = *reinterpret_cast<const frag*>(qptr + c * 16 + khalf);
       } else {
         qf[c] = frag{};
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        reinterpret_cast<T*>(&qf[c])[j] =
+            from_f32<T>(to_f32<T>(reinterpret_cast<T*>(&qf[c])[j]) * qs);
       }
     }
   }
@@ -178,7 +189,6 @@ __device__ __forceinline__ void prefill_tile_body(const PrefillParams& p, int re
 #pragma unroll
   for (int i = 0; i < DT; ++i) acc_o[i] = {};
 
-  const float scale2 = p.sm_scale * kLog2e;
   const float cap = p.logits_soft_cap;
   const int wleft = p.window_left;
 
@@ -372,9 +382,8 @@ __device__ __forceinline__ void prefill_tile_body(const PrefillParams& p, int re
         tile_full &= (kvt0 >= wave_max_qpos2 + diag - wleft);
       }
       if (tile_full) {
-        const float sc2 = p.sm_scale * kLog2e;
 #pragma unroll
-        for (int r = 0; r < 16; ++r) pr[r] = acc_s[r] * sc2;
+        for (int r = 0; r < 16; ++r) pr[r] = acc_s[r];  // already base-2 domain
       } else {
         // ALiBi slope for this lane's qo head: 2^(-8*(h+1)/Hq)
         float slope = 0.f;
@@ -382,19 +391,21 @@ __device__ __forceinline__ void prefill_tile_body(const PrefillParams& p, int re
           int qh = (int)(kv_head * group + my_g_u);
           slope = __builtin_exp2f(-8.f * (qh + 1) / p.num_qo_heads);
         }
+        constexpr float kInvLog2e = 0.6931471805599453f;  // 1/log2(e)
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           int64_t kv = kvt0 + mfma32_cd_row(r, lane);
-          float sv = acc_s[r] * p.sm_scale;
-          if (cap > 0.f) sv = cap * tanhf(sv / cap);
-          if (p.alibi) sv -= slope * (float)(my_qpos + diag - kv);
+          float sv = acc_s[r];  // base-2 domain (Q pre-scaled)
+          if (cap > 0.f)
+            sv = cap * tanhf(sv * kInvLog2e / cap) * kLog2e;
+          if (p.alibi) sv -= (slope * kLog2e) * (float)(my_qpos + diag - kv);
           bool ok = kv < kv_valid_hi;
           if constexpr (VARIANT::kActive) {
             int head = (int)(kv_head * group + my_g_u);
-            sv = VARIANT::logits_transform(sv, my_qpos, kv, head, qo_len, kv_len);
+            sv = VARIANT::logits_transform(sv * kInvLog2e, my_qpos, kv, head,
+                                           qo_len, kv_len) * kLog2e;
             ok &= VARIANT::logits_mask(my_qpos, kv, head, qo_len, kv_len);
           }
-          sv *= kLog2e;
           if constexpr (CAUSAL) ok &= kv <= my_qpos + diag;
           if (wleft >= 0) ok &= kv >= my_qpos + diag - wleft;
           if constexpr (MASK) {
