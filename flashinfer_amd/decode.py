@@ -192,6 +192,38 @@ class BatchDecodeWithPagedKVCacheWrapper:
         )
         self._fused = (fused_ok or mfma_ok) and not self._tc
         self._fused_mfma = mfma_ok and not self._tc
+        # autotuner hook (reference tactic system role): in fully-auto mode
+        # the route choice is a tactic keyed on the bucketed shape — a cached
+        # winner overrides the heuristic; under autotune() the first run()
+        # profiles every eligible route on the real inputs (see run()).
+        self._tune = None
+        if (self._use_tensor_cores is None and fixed_split_size is None
+                and not disable_split_kv):
+            from . import autotuner as _at
+
+            routes = []
+            if fused_ok:
+                routes.append("fused")
+            if mfma_ok:
+                routes.append("mfma")
+            if kv_dt == q_data_type:
+                routes.append("tc")
+            routes.append("split")
+            key = (f"decode_route:({_at._bucket(max(1, batch))},"
+                   f"{_at._bucket(max(1, max_len))},{group},{head_dim})")
+            cached = _at._cache.get(key)
+            if cached is not None and routes:
+                route = routes[cached % len(routes)]
+                self._fused = route in ("fused", "mfma")
+                self._fused_mfma = route == "mfma"
+                self._tc = route == "tc"
+            elif _at._tuning_enabled and len(routes) > 1:
+                self._tune = (key, routes,
+                              (indptr, indices, last_page_len, num_qo_heads,
+                               num_kv_heads, head_dim, page_size,
+                               pos_encoding_mode, window_left,
+                               logits_soft_cap, q_data_type, kv_data_type,
+                               sm_scale))
         if self._tc:
             from .prefill import BatchPrefillWithPagedKVCacheWrapper
 
@@ -321,6 +353,8 @@ class BatchDecodeWithPagedKVCacheWrapper:
             raise RuntimeError("must call plan() before run()")
         if q.dim() == 2:
             q = q.unsqueeze(1)  # [B, D] MQA convenience
+        if getattr(self, "_tune", None) is not None:
+            self._profile_routes(q, paged_kv_cache)
         if getattr(self, "_tc", False):
             need_lse = return_lse or sinks is not None
             # q_scale and k_scale both act on the logits: fold into the
@@ -417,6 +451,78 @@ class BatchDecodeWithPagedKVCacheWrapper:
         return (out, lse) if return_lse else out
 
     forward = run
+
+    def _profile_routes(self, q, paged_kv_cache):
+        """First run under autotune(): time every eligible route on the real
+        inputs, cache the winner under the bucketed shape key, and adopt it
+        (reference autotuner.py:98-190 tactic-profiling role)."""
+        import time as _time
+
+        import torch as _t
+
+        from . import autotuner as _at
+        from . import decode as _dec
+
+        key, routes, plan_args = self._tune
+        self._tune = None
+        (indptr, indices, last_page_len, Hq, Hkv, D, page, pem, wl, cap,
+         qdt, kdt, sm) = plan_args
+
+        def make(route):
+            saved = (_dec._FUSED_MAX_KV, _dec._MFMA_MAX_KV)
+            utc = None
+            try:
+                if route == "split":
+                    _dec._FUSED_MAX_KV = _dec._MFMA_MAX_KV = 0
+                    utc = False
+                elif route == "tc":
+                    utc = True
+                elif route == "mfma":
+                    _dec._FUSED_MAX_KV = 0
+                    _dec._MFMA_MAX_KV = 1 << 30
+                else:  # fused
+                    _dec._FUSED_MAX_KV = 1 << 30
+                w = BatchDecodeWithPagedKVCacheWrapper(
+                    self._float_workspace_buffer, self._kv_layout,
+                    use_tensor_cores=utc)
+                prev = _at._tuning_enabled
+                _at._tuning_enabled = False
+                _at._cache.pop(key, None)
+                try:
+                    w.plan(indptr, indices, last_page_len, Hq, Hkv, D, page,
+                           pos_encoding_mode=pem, window_left=wl,
+                           logits_soft_cap=cap, q_data_type=qdt,
+                           kv_data_type=kdt, sm_scale=sm)
+                finally:
+                    _at._tuning_enabled = prev
+            finally:
+                _dec._FUSED_MAX_KV, _dec._MFMA_MAX_KV = saved
+            return w
+
+        timings = []
+        for route in routes:
+            try:
+                w = make(route)
+                w.run(q, paged_kv_cache)
+                _t.cuda.synchronize()
+                t0 = _time.perf_counter()
+                for _ in range(3):
+                    w.run(q, paged_kv_cache)
+                _t.cuda.synchronize()
+                timings.append((_time.perf_counter() - t0, route))
+            except Exception:
+                continue
+        if timings:
+            best = min(timings)[1]
+            _at._cache[key] = routes.index(best)
+            mine = ("tc" if self._tc else
+                    "mfma" if getattr(self, "_fused_mfma", False) else
+                    "fused" if self._fused else "split")
+            if best != mine:
+                self.plan(indptr, indices, last_page_len, Hq, Hkv, D, page,
+                          pos_encoding_mode=pem, window_left=wl,
+                          logits_soft_cap=cap, q_data_type=qdt,
+                          kv_data_type=kdt, sm_scale=sm)
 
     def end_forward(self) -> None:
         pass
