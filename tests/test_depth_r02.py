@@ -230,6 +230,18 @@ def test_autotune_decode_route():
     q = torch.randn(bs, Hq, D, dtype=torch.bfloat16, device="cuda")
     ws = torch.empty(256 << 20, dtype=torch.uint8, device="cuda")
     autotuner._cache.clear()
+    try:
+        _run_autotune_body(fi, autotuner, ws, indptr, indices, lpl, Hq, Hkv,
+                           D, page, q, kc, vc)
+    finally:
+        # the route cache is global state: scrub it so later tests'
+        # routing assertions see the heuristic, not this test's winner
+        for k in [k for k in autotuner._cache if k.startswith("decode_route")]:
+            del autotuner._cache[k]
+
+
+def _run_autotune_body(fi, autotuner, ws, indptr, indices, lpl, Hq, Hkv, D,
+                       page, q, kc, vc):
     with autotuner.autotune():
         w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
         w.plan(indptr, indices, lpl, Hq, Hkv, D, page,
@@ -243,10 +255,4 @@ def test_autotune_decode_route():
             q_data_type=torch.bfloat16)
     out2 = w2.run(q, (kc, vc))
     # numerics unchanged regardless of the winning route
-    import math
-    for b in [0, 5]:
-        kk = kc.view(-1, Hkv, D)[b * kv:(b + 1) * kv]
-        vv = vc.view(-1, Hkv, D)[b * kv:(b + 1) * kv]
-        # pages are permuted; gather properly
-    g = Hq // Hkv
     torch.testing.assert_close(out.float(), out2.float(), atol=2e-2, rtol=2e-2)
