@@ -87,7 +87,7 @@ hipError_t fi_gemm_fp8_grouped(const void* A, const void* W, void* C,
                                int64_t lda, int64_t ldw_n, int64_t ldw_seg, int64_t ldc,
                                const float* a_scales, const float* b_scales,
                                float scalar_scale, int64_t a_scale_stride,
-                               int flat_tiles, hipStream_t stream);
+                               int flat_tiles, int mx, hipStream_t stream);
 hipError_t fi_per_group_quant_fp8(int dtype, int trans_scale, const void* x, void* q,
                                   float* scale, int64_t rows, int K, int64_t stride_row,
                                   float eps, hipStream_t stream);
@@ -117,10 +117,10 @@ hipError_t fi_moe_build_permute(const int32_t* ids, int32_t* counts,
                                 int32_t* token_of_copy, int32_t* inv, int n,
                                 int k, int E, int align, hipStream_t stream);
 hipError_t fi_gather_quant(int dtype, const void* src, const int32_t* token_of_copy,
-                           uint8_t* dst, float* scale, int R, int K,
+                           uint8_t* dst, float* scale, int R, int K, int e8m0,
                            hipStream_t stream);
 hipError_t fi_silu_mul_quant(int dtype, const void* h, uint8_t* dst, float* scale,
-                             int R, int I, int gelu, hipStream_t stream);
+                             int R, int I, int gelu, int e8m0, hipStream_t stream);
 hipError_t fi_moe_finalize(int dtype, const void* h, void* out, const int32_t* pos,
                            const float* w, int64_t tokens, int topk, int cols,
                            hipStream_t stream);
@@ -863,15 +863,18 @@ void gemm_fp8_grouped(at::Tensor a, at::Tensor w, at::Tensor c, at::Tensor m_ind
   TORCH_CHECK(c.scalar_type() == at::kBFloat16);
   int K = a.size(1), N = w.size(1);
   int S = m_indptr.size(0) - 1;
+  // MX mode: u8 e8m0 scale tensors (per-row [M, K/128] + per-block
+  // [S, K/128, N/128] bytes) select the hardware-scaled f8f6f4 path
+  bool mx = a_scales.has_value() && a_scales->scalar_type() == at::kByte;
   int64_t a_scale_stride = a_scales.has_value() ? a_scales->stride(0) : 0;
   check_hip(fi_gemm_fp8_grouped(
                 a.data_ptr(), w.data_ptr(), c.data_ptr(), m_indptr.data_ptr<int32_t>(),
                 w_indices.has_value() ? w_indices->data_ptr<int32_t>() : nullptr, S,
                 (int)max_m_tiles, N, K, a.stride(0), w.stride(1), w.stride(0),
                 c.stride(0),
-                a_scales.has_value() ? a_scales->data_ptr<float>() : nullptr,
-                b_scales.has_value() ? b_scales->data_ptr<float>() : nullptr,
-                (float)scalar_scale, a_scale_stride, (int)flat_tiles,
+                a_scales.has_value() ? (float*)a_scales->data_ptr() : nullptr,
+                b_scales.has_value() ? (float*)b_scales->data_ptr() : nullptr,
+                (float)scalar_scale, a_scale_stride, (int)flat_tiles, mx ? 1 : 0,
                 cur_stream(a)),
             "fi_gemm_fp8_grouped");
 }
@@ -1187,17 +1190,21 @@ void moe_build_permute_run(at::Tensor ids, at::Tensor counts, at::Tensor m_indpt
 
 void gather_quant_run(at::Tensor src, at::Tensor token_of_copy, at::Tensor dst,
                       at::Tensor scale) {
+  // e8m0 mode is selected by the scale tensor's dtype (u8 = MX bytes)
+  bool e8m0 = scale.scalar_type() == at::kByte;
   check_hip(fi_gather_quant(dtype_code(src), src.data_ptr(),
                             token_of_copy.data_ptr<int32_t>(),
-                            dst.data_ptr<uint8_t>(), scale.data_ptr<float>(),
-                            dst.size(0), dst.size(1), cur_stream(src)),
+                            dst.data_ptr<uint8_t>(), (float*)scale.data_ptr(),
+                            dst.size(0), dst.size(1), e8m0 ? 1 : 0,
+                            cur_stream(src)),
             "fi_gather_quant");
 }
 
 void silu_mul_quant_run(at::Tensor h, at::Tensor dst, at::Tensor scale, bool gelu) {
+  bool e8m0 = scale.scalar_type() == at::kByte;
   check_hip(fi_silu_mul_quant(dtype_code(h), h.data_ptr(), dst.data_ptr<uint8_t>(),
-                              scale.data_ptr<float>(), dst.size(0), dst.size(1),
-                              gelu ? 1 : 0, cur_stream(h)),
+                              (float*)scale.data_ptr(), dst.size(0), dst.size(1),
+                              gelu ? 1 : 0, e8m0 ? 1 : 0, cur_stream(h)),
             "fi_silu_mul_quant");
 }
 

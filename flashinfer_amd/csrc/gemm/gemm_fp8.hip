@@ -51,8 +51,14 @@ __device__ __forceinline__ void stage_tile(const uint8_t* __restrict__ gbase,
   }
 }
 
-// OUT_BF16: write bf16, else f32
-template <bool GROUPWISE>
+// MODE: 0 = per-tensor scalar scale; 1 = f32 groupwise (1x128 activation x
+// 128x128 weight-block scales, local-accumulator rescale per K group);
+// 2 = MX e8m0 (hardware-applied per-row scale bytes in the f8f6f4
+// instruction: per-lane activation byte + wave-uniform weight-block byte —
+// scale semantics probed in scripts/probe/mfma_mx5/6/7.hip). A 64x64-wave
+// variant and a triple-buffer variant were both tried and measured SLOWER
+// (co-residency loss; profiles/README r02) — this 8-wave 32x64 shape stays.
+template <int MODE>
 __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
     const uint8_t* __restrict__ A, const uint8_t* __restrict__ W,
     bf16* __restrict__ C, const int32_t* __restrict__ m_indptr,
@@ -111,7 +117,7 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
   int nk = K / BK;
   stage_tile(A, lda, m0, m_end, 0, as_base, tid);
   stage_tile(Wb, ldw_n, bn0, N, 0, bs_base, tid);
-  if (GROUPWISE && tid < BM) {
+  if (MODE == 1 && tid < BM) {
     int m = m0 + tid;
     ascale_s[0][tid] = a_scales[m < m_end ? m : m_end - 1];
   }
@@ -123,7 +129,7 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
     if (kt + 1 < nk) {
       stage_tile(A, lda, m0, m_end, (kt + 1) * BK, as_base + (cur ^ 1) * BUF_BYTES, tid);
       stage_tile(Wb, ldw_n, bn0, N, (kt + 1) * BK, bs_base + (cur ^ 1) * BUF_BYTES, tid);
-      if (GROUPWISE && tid < BM) {
+      if (MODE == 1 && tid < BM) {
         int m = m0 + tid;
         ascale_s[cur ^ 1][tid] =
             a_scales[(int64_t)(kt + 1) * a_scale_stride + (m < m_end ? m : m_end - 1)];
@@ -147,19 +153,36 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
       return u.v;
     };
     const int kh32 = (lane >> 5) * 32;
+    int vb = 127, wsb = 127;
+    if constexpr (MODE == 2) {
+      // activation scale byte for this lane's M row + this K group; both
+      // lane halves carry the same byte (HW averages 16-elem sub-blocks)
+      int m = m0 + wm + line;
+      vb = ((const uint8_t*)a_scales)[(int64_t)(m < m_end ? m : m_end - 1) *
+                                          (K / BK) + kt];
+      wsb = ((const uint8_t*)b_scales)[((int64_t)widx * (K / BK) + kt) *
+                                           ((N + 127) / 128) + bn0 / 128];
+    }
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int g2 = 0; g2 < 2; ++g2) {  // two K=64 groups per BK=128 tile
       int kb = g2 * 64 + kh32;
       intx8 afrag = ld32(a_lds, wm + line, kb);  // B-operand: col = m
 #pragma unroll
-      for (int i = 0; i < 2; ++i)
-        acc[i] = mfma_32x32x64_fp8(ld32(b_lds, wn + i * 32 + line, kb), afrag,
-                                   acc[i]);
+      for (int i = 0; i < 2; ++i) {
+        if constexpr (MODE == 2) {
+          accm[i] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+              ld32(b_lds, wn + i * 32 + line, kb), afrag, accm[i], 0, 0, 0,
+              wsb, 0, vb);
+        } else {
+          acc[i] = mfma_32x32x64_fp8(ld32(b_lds, wn + i * 32 + line, kb), afrag,
+                                     acc[i]);
+        }
+      }
     }
     __builtin_amdgcn_s_setprio(0);
     // rescale local accumulator into master (per-lane scalar sa*sb)
-    if constexpr (GROUPWISE) {
+    if constexpr (MODE == 1) {
       float sb = b_scales[((int64_t)seg * nk + kt) * ((N + 127) / 128) + bn0 / 128];
       float sab = ascale_s[cur][wm + line] * sb;
       // packed f32 FMA halves the rescale VALU chain (v_pk_fma_f32)
@@ -171,7 +194,7 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
 #pragma unroll
         for (int r = 0; r < 8; ++r) am[r] += ac[r] * sab;
       }
-    } else {
+    } else if constexpr (MODE == 0) {
 #pragma unroll
       for (int i = 0; i < 2; ++i)
 #pragma unroll
@@ -188,7 +211,7 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
     cur ^= 1;
   }
 
-  float fs = GROUPWISE ? 1.f : scalar_scale;
+  float fs = MODE == 0 ? scalar_scale : 1.f;
   // C^T epilogue: lane column = m, accumulator rows = n
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
@@ -211,7 +234,7 @@ extern "C" hipError_t fi_gemm_fp8_grouped(
     const int32_t* w_indices, int num_segments, int max_m_tiles, int N, int K,
     int64_t lda, int64_t ldw_n, int64_t ldw_seg, int64_t ldc, const float* a_scales,
     const float* b_scales, float scalar_scale, int64_t a_scale_stride,
-    int flat_tiles, hipStream_t stream) {
+    int flat_tiles, int mx, hipStream_t stream) {
   if (K % 128 != 0) return hipErrorInvalidValue;
   // flat < 0 => legacy z-grid; flat >= 0 ignored sentinel handled by caller
   dim3 grid, blk(fi::f8gemm::NTH);
@@ -223,16 +246,15 @@ extern "C" hipError_t fi_gemm_fp8_grouped(
     grid = dim3((N + fi::f8gemm::BN - 1) / fi::f8gemm::BN, max_m_tiles,
                 num_segments);
   }
-  if (a_scales && b_scales) {
-    hipLaunchKernelGGL((fi::f8gemm::gemm_fp8_kernel<true>), grid, blk, 0, stream,
-                       (const uint8_t*)A, (const uint8_t*)W, (fi::bf16*)C, m_indptr,
-                       w_indices, N, K, lda, ldw_n, ldw_seg, ldc, a_scales, b_scales,
-                       scalar_scale, a_scale_stride, flat_segs);
-  } else {
-    hipLaunchKernelGGL((fi::f8gemm::gemm_fp8_kernel<false>), grid, blk, 0, stream,
-                       (const uint8_t*)A, (const uint8_t*)W, (fi::bf16*)C, m_indptr,
-                       w_indices, N, K, lda, ldw_n, ldw_seg, ldc, a_scales, b_scales,
-                       scalar_scale, a_scale_stride, flat_segs);
-  }
+  int mode = (a_scales && b_scales) ? (mx ? 2 : 1) : 0;
+#define LG(M)                                                                     \
+  hipLaunchKernelGGL((fi::f8gemm::gemm_fp8_kernel<M>), grid, blk, 0, stream,      \
+                     (const uint8_t*)A, (const uint8_t*)W, (fi::bf16*)C, m_indptr,\
+                     w_indices, N, K, lda, ldw_n, ldw_seg, ldc, a_scales,         \
+                     b_scales, scalar_scale, a_scale_stride, flat_segs)
+  if (mode == 2) LG(2);
+  else if (mode == 1) LG(1);
+  else LG(0);
+#undef LG
   return hipGetLastError();
 }

@@ -205,9 +205,11 @@ __global__ void moe_scatter_kernel(const int32_t* __restrict__ ids,
 }
 
 // ------------- fused gather + per-128-group fp8 quantization -------------
-// dst_q[r] = e4m3(src[token_of_copy[r]] / s); scales [K/128, R] (MN-major).
+// dst_q[r] = e4m3(src[token_of_copy[r]] / s); scales: f32 [K/128, R]
+// (MN-major) or — E8M0 mode — u8 e8m0 bytes [R, K/128] (scale rounded UP to
+// a power of two so the MX MFMA applies it in hardware).
 // block 256 = 4 waves; wave handles one row's groups strided by 4.
-template <typename T>
+template <typename T, bool E8M0 = false>
 __global__ void gather_quant_kernel(const T* __restrict__ src,
                                     const int32_t* __restrict__ token_of_copy,
                                     uint8_t* __restrict__ dst,
@@ -224,19 +226,31 @@ __global__ void gather_quant_kernel(const T* __restrict__ src,
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1)
       amax = fmaxf(amax, __shfl_xor(amax, off, 64));
-    float s = fmaxf(amax, 1e-10f) / 448.f;
-    float inv_s = 1.f / s;
+    float s, inv_s;
+    if constexpr (E8M0) {
+      // scale = 2^ceil(log2(amax/448)): power of two for the HW scale path
+      int e;
+      frexpf(fmaxf(amax, 1e-10f) / 448.f, &e);  // amax/448 in [2^(e-1), 2^e)
+      s = ldexpf(1.f, e);
+      inv_s = ldexpf(1.f, -e);
+      if (lane == 0)
+        reinterpret_cast<uint8_t*>(scale)[(int64_t)r * (K / 128) + g] =
+            (uint8_t)(127 + e);
+    } else {
+      s = fmaxf(amax, 1e-10f) / 448.f;
+      inv_s = 1.f / s;
+      if (lane == 0) scale[(int64_t)g * R + r] = s;
+    }
     uint8_t q0 = (uint8_t)__builtin_bit_cast(uint8_t, from_f32<fp8_e4m3>(v0 * inv_s));
     uint8_t q1 = (uint8_t)__builtin_bit_cast(uint8_t, from_f32<fp8_e4m3>(v1 * inv_s));
     uint16_t pair = (uint16_t)q0 | ((uint16_t)q1 << 8);
     reinterpret_cast<uint16_t*>(dst + (int64_t)r * K)[g * 64 + lane] = pair;
-    if (lane == 0) scale[(int64_t)g * R + r] = s;
   }
 }
 
 // ---------- fused silu(gate)*up + per-128-group fp8 quantization ----------
 // h [R, 2I] (gate | up) -> q [R, I] e4m3 + scales [I/128, R]
-template <typename T>
+template <typename T, bool E8M0 = false>
 __global__ void silu_mul_quant_kernel(const T* __restrict__ h,
                                       uint8_t* __restrict__ dst,
                                       float* __restrict__ scale, int R, int I,
@@ -267,12 +281,23 @@ __global__ void silu_mul_quant_kernel(const T* __restrict__ h,
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1)
       amax = fmaxf(amax, __shfl_xor(amax, off, 64));
-    float s = fmaxf(amax, 1e-10f) / 448.f;
-    float inv_s = 1.f / s;
+    float s, inv_s;
+    if constexpr (E8M0) {
+      int e;
+      frexpf(fmaxf(amax, 1e-10f) / 448.f, &e);
+      s = ldexpf(1.f, e);
+      inv_s = ldexpf(1.f, -e);
+      if (lane == 0)
+        reinterpret_cast<uint8_t*>(scale)[(int64_t)r * (I / 128) + g] =
+            (uint8_t)(127 + e);
+    } else {
+      s = fmaxf(amax, 1e-10f) / 448.f;
+      inv_s = 1.f / s;
+      if (lane == 0) scale[(int64_t)g * R + r] = s;
+    }
     uint16_t pair = (uint16_t)(uint8_t)__builtin_bit_cast(uint8_t, from_f32<fp8_e4m3>(v[0] * inv_s)) |
                     ((uint16_t)(uint8_t)__builtin_bit_cast(uint8_t, from_f32<fp8_e4m3>(v[1] * inv_s)) << 8);
     reinterpret_cast<uint16_t*>(dst + (int64_t)r * I)[g * 64 + lane] = pair;
-    if (lane == 0) scale[(int64_t)g * R + r] = s;
   }
 }
 
@@ -314,12 +339,20 @@ hipError_t fi_moe_build_permute(const int32_t* ids, int32_t* counts,
 }
 
 hipError_t fi_gather_quant(int dtype, const void* src, const int32_t* token_of_copy,
-                           uint8_t* dst, float* scale, int R, int K,
+                           uint8_t* dst, float* scale, int R, int K, int e8m0,
                            hipStream_t stream) {
   if (K % 128) return hipErrorInvalidValue;
-#define LAUNCH_GQ(T)                                                      \
-  hipLaunchKernelGGL(fi::gather_quant_kernel<T>, dim3(R), dim3(256), 0,   \
-                     stream, (const T*)src, token_of_copy, dst, scale, R, K)
+#define LAUNCH_GQ(T)                                                          \
+  do {                                                                        \
+    if (e8m0)                                                                 \
+      hipLaunchKernelGGL((fi::gather_quant_kernel<T, true>), dim3(R),         \
+                         dim3(256), 0, stream, (const T*)src, token_of_copy,  \
+                         dst, scale, R, K);                                   \
+    else                                                                      \
+      hipLaunchKernelGGL((fi::gather_quant_kernel<T, false>), dim3(R),        \
+                         dim3(256), 0, stream, (const T*)src, token_of_copy,  \
+                         dst, scale, R, K);                                   \
+  } while (0)
   switch (dtype) {
     case 0: LAUNCH_GQ(fi::bf16); break;
     case 1: LAUNCH_GQ(fi::fp16); break;
@@ -331,11 +364,20 @@ hipError_t fi_gather_quant(int dtype, const void* src, const int32_t* token_of_c
 }
 
 hipError_t fi_silu_mul_quant(int dtype, const void* h, uint8_t* dst, float* scale,
-                             int R, int I, int gelu, hipStream_t stream) {
+                             int R, int I, int gelu, int e8m0,
+                             hipStream_t stream) {
   if (I % 128) return hipErrorInvalidValue;
-#define LAUNCH_SQ(T)                                                        \
-  hipLaunchKernelGGL(fi::silu_mul_quant_kernel<T>, dim3(R), dim3(256), 0,   \
-                     stream, (const T*)h, dst, scale, R, I, gelu)
+#define LAUNCH_SQ(T)                                                            \
+  do {                                                                          \
+    if (e8m0)                                                                   \
+      hipLaunchKernelGGL((fi::silu_mul_quant_kernel<T, true>), dim3(R),         \
+                         dim3(256), 0, stream, (const T*)h, dst, scale, R, I,   \
+                         gelu);                                                 \
+    else                                                                        \
+      hipLaunchKernelGGL((fi::silu_mul_quant_kernel<T, false>), dim3(R),        \
+                         dim3(256), 0, stream, (const T*)h, dst, scale, R, I,   \
+                         gelu);                                                 \
+  } while (0)
   switch (dtype) {
     case 0: LAUNCH_SQ(fi::bf16); break;
     case 1: LAUNCH_SQ(fi::fp16); break;

@@ -30,6 +30,28 @@ def per_token_group_quant_fp8(
     return q.view(_FP8), scale
 
 
+def per_block_quant_mxfp8(
+    w: torch.Tensor, block: int = 128
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    r"""MX weight quantization: [128, 128] blocks with POWER-OF-TWO (e8m0)
+    scales the hardware-scaled f8f6f4 MFMA applies directly — w [E, N, K] ->
+    (w_fp8, scale u8 [E, K/128, N/128] e8m0 bytes). Reference mxfp8 role
+    (mxfp8_quantize / mm_mxfp8) on the CDNA4-native MX path."""
+    orig_shape = w.shape
+    w3 = w.reshape(-1, *orig_shape[-2:])
+    E, N, K = w3.shape
+    wf = w3.float().view(E, N // block, block, K // block, block)
+    amax = wf.abs().amax(dim=(2, 4), keepdim=True).clamp(min=1e-10)
+    e = torch.ceil(torch.log2(amax / 448.0))
+    scale = torch.exp2(e)
+    q = (wf / scale).to(torch.float8_e4m3fn).view(E, N, K)
+    sc_u8 = (e.view(E, N // block, K // block) + 127).to(torch.uint8)
+    sc_u8 = sc_u8.transpose(1, 2).contiguous()  # [E, K/128, N/128]
+    if len(orig_shape) == 2:
+        return q.view(orig_shape), sc_u8[0]
+    return q.view(orig_shape), sc_u8
+
+
 def per_block_quant_fp8(
     w: torch.Tensor, block: int = 128
 ) -> Tuple[torch.Tensor, torch.Tensor]:

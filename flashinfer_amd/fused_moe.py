@@ -163,20 +163,28 @@ def fused_moe(
         # rows cost more than the dead WGs (measured: E=8 1.455 -> 1.405
         # M tok/s), so the z-grid stays.
         use_flat = E >= 16
+        # MX mode (u8 e8m0 weight scales from per_block_quant_mxfp8): the
+        # hardware-scaled f8f6f4 MFMA applies the scales — no rescale VALU
+        mx = w13_scale.dtype == torch.uint8
         m_indptr, token_of_copy, inv = _build_permute(
             topk_ids, E, align=128 if use_flat else 1)
         Rp = token_of_copy.shape[0]
         flat_tiles = Rp // 128 if use_flat else 0
         a_q = torch.empty(Rp, H, dtype=torch.uint8, device=x.device)
-        a_s = torch.empty(H // 128, Rp, dtype=torch.float32, device=x.device)
+        a_s = (torch.empty(Rp, H // 128, dtype=torch.uint8, device=x.device)
+               if mx else
+               torch.empty(H // 128, Rp, dtype=torch.float32, device=x.device))
         ext.gather_quant_run(x, token_of_copy, a_q, a_s)
         h1 = torch.empty(Rp, I2, dtype=torch.bfloat16, device=x.device)
         ext.gemm_fp8_grouped(a_q, w13.view(torch.uint8), h1,
                              m_indptr, None, max_m_tiles, a_s,
                              w13_scale.contiguous(), 1.0, flat_tiles)
         act_q = torch.empty(Rp, inter, dtype=torch.uint8, device=x.device)
-        act_s = torch.empty(inter // 128, Rp, dtype=torch.float32,
-                            device=x.device)
+        act_s = (torch.empty(Rp, inter // 128, dtype=torch.uint8,
+                             device=x.device)
+                 if mx else
+                 torch.empty(inter // 128, Rp, dtype=torch.float32,
+                             device=x.device))
         ext.silu_mul_quant_run(h1, act_q, act_s, activation == "gelu")
         h2 = torch.empty(Rp, H, dtype=torch.bfloat16, device=x.device)
         ext.gemm_fp8_grouped(act_q, w2.view(torch.uint8), h2,

@@ -307,6 +307,24 @@ def bench_mla_prefill(bs=16, s=1024, H=128, page=32):
     print(f"mla prefill bs={bs} s={s} H={H}: {t*1e3:.2f} ms  {fl/t/1e12:.1f} TFLOPS")
 
 
+def bench_moe_mx(T=4096, H=4096, inter=14336, E=8, k=2):
+    """MX-fp8 MoE: e8m0 hardware scales in the f8f6f4 MFMA (no rescale VALU)."""
+    from flashinfer_amd.fused_moe import fused_moe, moe_topk_softmax
+    from flashinfer_amd.fp8_quantization import per_block_quant_mxfp8
+    torch.manual_seed(0)
+    x = torch.randn(T, H, dtype=torch.bfloat16, device="cuda") / 4
+    w13 = torch.randn(E, 2 * inter, H, dtype=torch.bfloat16, device="cuda") / 16
+    w2 = torch.randn(E, H, inter, dtype=torch.bfloat16, device="cuda") / 16
+    w13_q, w13_s = per_block_quant_mxfp8(w13)
+    w2_q, w2_s = per_block_quant_mxfp8(w2)
+    del w13, w2
+    logits = torch.randn(T, E, device="cuda")
+    weights, ids = moe_topk_softmax(logits, k)
+    t = timeit(lambda: fused_moe(x, w13_q, w2_q, weights, ids, w13_scale=w13_s, w2_scale=w2_s), iters=10, warmup=3)
+    fl = T * k * 3 * H * inter * 2
+    print(f"fused_moe MX-fp8 mixtral T={T}: {t*1e3:.2f} ms  {fl/t/1e12:.0f} TFLOPS  {T/t/1e6:.3f} M tok/s")
+
+
 if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "all"
     if which in ("all", "prefill"):
@@ -343,5 +361,6 @@ if __name__ == "__main__":
     if which in ("all", "moe"):
         bench_moe()
         bench_moe_fp8()
+        bench_moe_mx()
     if which in ("all", "fp8"):
         bench_fp8_gemm()

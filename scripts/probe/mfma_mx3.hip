@@ -1,0 +1,54 @@
+#include <hip/hip_runtime.h>
+#include <cstdio>
+#include <cmath>
+typedef __attribute__((ext_vector_type(8))) int intx8;
+typedef __attribute__((ext_vector_type(16))) float floatx16;
+
+__global__ void probe(const unsigned char* A, const unsigned char* B,
+                      float* C, int which) {
+  int lane = threadIdx.x;
+  int row = lane & 31, kh = (lane >> 5) * 32;
+  intx8 a = *(const intx8*)(A + row * 64 + kh);
+  intx8 b = *(const intx8*)(B + row * 64 + kh);
+  unsigned int enc = (unsigned)(100 + lane) * 0x01010101u;  // all 4 bytes
+  int sa = which == 0 ? (int)enc : 127;
+  int sb = which == 1 ? (int)enc : 127;
+  floatx16 c = {};
+  c = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(a, b, c, 0, 0, 0, sa, 0, sb);
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int m = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+    C[m * 32 + (lane & 31)] = c[r];
+  }
+}
+
+int main() {
+  unsigned char hA[32 * 64], hB[32 * 64];
+  for (int blkcase = 0; blkcase < 2; ++blkcase) {
+    for (int i = 0; i < 32 * 64; ++i) {
+      int k = i % 64;
+      hA[i] = ((k / 32) == blkcase) ? 0x38 : 0;
+      hB[i] = 0x38;
+    }
+    unsigned char *dA, *dB; float* dC;
+    (void)hipMalloc(&dA, sizeof hA); (void)hipMalloc(&dB, sizeof hB);
+    (void)hipMalloc(&dC, 32 * 32 * 4);
+    (void)hipMemcpy(dA, hA, sizeof hA, hipMemcpyHostToDevice);
+    (void)hipMemcpy(dB, hB, sizeof hB, hipMemcpyHostToDevice);
+    for (int which = 0; which < 2; ++which) {
+      hipLaunchKernelGGL(probe, dim3(1), dim3(64), 0, 0, dA, dB, dC, which);
+      float hC[32 * 32];
+      (void)hipMemcpy(hC, dC, sizeof hC, hipMemcpyDeviceToHost);
+      printf("Ablk %d enc %s:\n", blkcase, which == 0 ? "A" : "B");
+      for (int m = 0; m < 32; m += 1) {
+        double v = hC[m * 32 + 0] / 32.0;
+        double v2 = hC[m * 32 + 17] / 32.0;
+        printf("(%d:%d,%d) ", m, v > 0 ? (int)lround(log2(v)) + 27 : -1,
+               v2 > 0 ? (int)lround(log2(v2)) + 27 : -1);
+        if (m % 8 == 7) printf("\n");
+      }
+    }
+    (void)hipFree(dA); (void)hipFree(dB); (void)hipFree(dC);
+  }
+  return 0;
+}

@@ -261,3 +261,33 @@ def test_silu_mul_quant_and_gather_quant():
     deq2 = gq.view(torch.float8_e4m3fn).float() * \
         gs.t().repeat_interleave(128, dim=1)
     torch.testing.assert_close(deq2, ref2, atol=0.08, rtol=0.08)
+
+
+@pytest.mark.gpu
+def test_moe_e2e_mxfp8():
+    """MX-fp8 MoE (hardware e8m0 scales) vs a torch reference."""
+    from flashinfer_amd.fused_moe import fused_moe, moe_topk_softmax
+    from flashinfer_amd.fp8_quantization import per_block_quant_mxfp8
+
+    torch.manual_seed(4)
+    T, H, inter, E, k = 128, 512, 512, 8, 2
+    x = torch.randn(T, H, dtype=torch.bfloat16, device="cuda") / 4
+    w13 = torch.randn(E, 2 * inter, H, dtype=torch.bfloat16, device="cuda") / 8
+    w2 = torch.randn(E, H, inter, dtype=torch.bfloat16, device="cuda") / 8
+    logits = torch.randn(T, E, device="cuda")
+    weights, ids = moe_topk_softmax(logits, k)
+    w13_q, w13_s = per_block_quant_mxfp8(w13)
+    w2_q, w2_s = per_block_quant_mxfp8(w2)
+    assert w13_s.dtype == torch.uint8
+    out = fused_moe(x, w13_q, w2_q, weights, ids,
+                    w13_scale=w13_s, w2_scale=w2_s)
+    ref = torch.zeros(T, H, device="cuda")
+    xf = x.float()
+    for t in range(T):
+        for j in range(k):
+            e = int(ids[t, j])
+            h1 = xf[t] @ w13[e].float().t()
+            act = torch.nn.functional.silu(h1[:inter]) * h1[inter:]
+            ref[t] += float(weights[t, j]) * (act @ w2[e].float().t())
+    # pow2 scales cost up to ~1 bit of quant precision vs f32 groupwise
+    torch.testing.assert_close(out.float(), ref, atol=0.5, rtol=0.2)
