@@ -116,8 +116,7 @@ class MoeEp:
                  enable_fault_tolerance: bool = False):
         import torch.distributed as dist
 
-        if algorithm == EpAlgorithm.MEGA:
-            raise MoEEpNotBuiltError("mega mode is a later drop on MI355X")
+        self._mega = algorithm == EpAlgorithm.MEGA
         self._a2a = MoeAlltoAll(group, num_experts=num_experts, top_k=top_k)
         self._group = group
         self.num_experts = num_experts
@@ -162,6 +161,39 @@ class MoeEp:
 
     def combine(self, y: torch.Tensor, topk_weights: torch.Tensor, state):
         return self._a2a.combine(y, topk_weights, state)
+
+    def forward_mega(self, x: torch.Tensor, router_logits: torch.Tensor,
+                     expert_fn, n_chunks: int = 4) -> torch.Tensor:
+        """Mega mode: CHUNKED dispatch -> expert-compute -> combine pipeline
+        (reference moe_ep mega "fused comm+compute" role, MI355X-shaped:
+        the a2a data plane is RCCL over xGMI, so the fusion is a software
+        pipeline — chunk c+1's alltoallv is enqueued before chunk c's local
+        grouped MoE runs, and RCCL's internal stream overlaps it with the
+        compute. expert_fn(recv_x, recv_expert_local) -> expert_out runs the
+        local experts (e.g. fused_moe's grouped GEMMs).
+
+        Known limitation vs a true fused kernel: the count exchange inside
+        dispatch synchronizes the host per chunk; the overlap won is the
+        payload alltoallv (the dominant bytes), not the count handshake."""
+        T = x.shape[0]
+        w, ids = self.route(router_logits)
+        if T == 0 or n_chunks <= 1 or T < n_chunks:
+            rx, rexp, st = self._a2a.dispatch(x, ids)
+            return self.combine(expert_fn(rx, rexp), w, st)
+        bounds = [T * c // n_chunks for c in range(n_chunks + 1)]
+        # enqueue every chunk's dispatch first: the payload a2a of later
+        # chunks flows while earlier chunks compute
+        pend = []
+        for c in range(n_chunks):
+            lo, hi = bounds[c], bounds[c + 1]
+            pend.append(self._a2a.dispatch(x[lo:hi], ids[lo:hi]))
+        outs = []
+        for c in range(n_chunks):
+            lo, hi = bounds[c], bounds[c + 1]
+            rx, rexp, st = pend[c]
+            y = expert_fn(rx, rexp)
+            outs.append(self._a2a.combine(y, w[lo:hi], st))
+        return torch.cat(outs, dim=0)
 
     # checkpoint hooks pass through to the data plane
     def checkpoint_prepare(self):
