@@ -419,10 +419,15 @@ __device__ __forceinline__ void prefill_tile_body(const PrefillParams& p, int re
         }
       }
 
-      // ---- online softmax update (per lane; exchange with lane^32) ----
-      float tmax = pr[0];
+      // ---- online softmax update (per lane; exchange with lane^32).
+      // packed-f32 max/sum trees (v_pk_max/add_f32) halve the reduction
+      // VALU (PMC r02: VALUBusy 47% — the softmax chain IS the bound) ----
+      typedef __attribute__((ext_vector_type(2))) float f32x2_;
+      const f32x2_* pr2 = reinterpret_cast<const f32x2_*>(pr);
+      f32x2_ mx2 = pr2[0];
 #pragma unroll
-      for (int r = 1; r < 16; ++r) tmax = fmaxf(tmax, pr[r]);
+      for (int r = 1; r < 8; ++r) mx2 = __builtin_elementwise_max(mx2, pr2[r]);
+      float tmax = fmaxf(mx2.x, mx2.y);
       tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
       // defer-max (guide T13): skip the O-wide rescale while the running max
       // grows by < 8 (base-2) — p values stay bounded by 2^8, f32 accum is
@@ -435,11 +440,14 @@ __device__ __forceinline__ void prefill_tile_body(const PrefillParams& p, int re
         for (int r = 0; r < 16; ++r) pr[r] = 0.f;
         f = 1.f;
       } else {
+        // no -inf select: v_exp_f32(-inf - finite) = 0 in hardware, and the
+        // m_new == -inf case is handled above (16 cndmask saved per subtile)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          pr[r] = (pr[r] == -INFINITY) ? 0.f : __builtin_exp2f(pr[r] - m_new);
-          psum += pr[r];
-        }
+        for (int r = 0; r < 16; ++r) pr[r] = __builtin_exp2f(pr[r] - m_new);
+        f32x2_ s2 = pr2[0];
+#pragma unroll
+        for (int r = 1; r < 8; ++r) s2 += pr2[r];
+        psum = s2.x + s2.y;
         f = defer ? 1.f : __builtin_exp2f(m_run - m_new);
       }
       d_run = d_run * f + psum;
