@@ -25,21 +25,21 @@
 
 namespace fi {
 
-template <typename T, int HEAD_DIM, int GROUP>
+template <typename T, int HEAD_DIM, int GROUP, typename TKV = T>
 __global__ __launch_bounds__(kDecWaves * 64, 1) void decode_mfma_kernel(DecodeParams p) {
   __shared__ char smem[decode_mfma_smem_bytes<T, HEAD_DIM, GROUP>()];
-  decode_mfma_item_body<T, HEAD_DIM, GROUP>(p, blockIdx.x, blockIdx.y,
-                                            blockIdx.z, smem);
+  decode_mfma_item_body<T, HEAD_DIM, GROUP, TKV>(p, blockIdx.x, blockIdx.y,
+                                                 blockIdx.z, smem);
 }
 
-template <typename T>
+template <typename T, typename TKV = T>
 hipError_t decode_mfma_dispatch(DecodeParams& p, hipStream_t stream) {
   int group = p.num_qo_heads / p.num_kv_heads;
   dim3 g((uint32_t)p.batch, (uint32_t)p.num_kv_heads,
          (uint32_t)(p.split > 1 ? p.split : 1));
   dim3 blk(kDecWaves * 64);
 #define LAUNCH_M(HD, G) \
-  hipLaunchKernelGGL((decode_mfma_kernel<T, HD, G>), g, blk, 0, stream, p)
+  hipLaunchKernelGGL((decode_mfma_kernel<T, HD, G, TKV>), g, blk, 0, stream, p)
   if (p.head_dim == 128) {
     switch (group) {
       case 8: LAUNCH_M(128, 8); break;
@@ -63,12 +63,18 @@ hipError_t decode_mfma_dispatch(DecodeParams& p, hipStream_t stream) {
 
 }  // namespace fi
 
-extern "C" hipError_t fi_decode_mfma(int dtype, fi::DecodeParams* p,
+extern "C" hipError_t fi_decode_mfma(int dtype, int kv_dtype, fi::DecodeParams* p,
                                      hipStream_t stream) {
   if (p->batch == 0) return hipSuccess;
-  switch (dtype) {
-    case 0: return fi::decode_mfma_dispatch<fi::bf16>(*p, stream);
-    case 1: return fi::decode_mfma_dispatch<fi::fp16>(*p, stream);
+  if (kv_dtype == dtype) {
+    switch (dtype) {
+      case 0: return fi::decode_mfma_dispatch<fi::bf16>(*p, stream);
+      case 1: return fi::decode_mfma_dispatch<fi::fp16>(*p, stream);
+    }
+  } else if (kv_dtype == 3 && dtype == 0) {
+    return fi::decode_mfma_dispatch<fi::bf16, fi::fp8_e4m3>(*p, stream);
+  } else if (kv_dtype == 3 && dtype == 1) {
+    return fi::decode_mfma_dispatch<fi::fp16, fi::fp8_e4m3>(*p, stream);
   }
   return hipErrorInvalidValue;
 }

@@ -47,7 +47,8 @@ hipError_t fi_batch_decode(int dtype, int kv_dtype, fi_ext::DecodeParams* p,
                            hipStream_t stream);
 hipError_t fi_batch_decode_fused(int dtype, int kv_dtype, fi_ext::DecodeParams* p,
                                  hipStream_t stream);
-hipError_t fi_decode_mfma(int dtype, fi_ext::DecodeParams* p, hipStream_t stream);
+hipError_t fi_decode_mfma(int dtype, int kv_dtype, fi_ext::DecodeParams* p,
+                          hipStream_t stream);
 namespace fi_ext2 {
 struct HolisticParams {
   fi_ext::PrefillParams pf;
@@ -487,7 +488,9 @@ void batch_decode_mfma_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
     p.tmp_v = (float*)tmp_v->data_ptr();
     p.tmp_s = tmp_s->data_ptr<float>();
   }
-  check_hip(fi_decode_mfma(dtype_code(q), &p, cur_stream(q)), "fi_decode_mfma");
+  check_hip(fi_decode_mfma(dtype_code(q), dtype_code(k_cache), &p,
+                           cur_stream(q)),
+            "fi_decode_mfma");
 }
 
 // Persistent holistic BatchAttention: one launch over a tagged work queue of

@@ -27,7 +27,11 @@ constexpr int decode_mfma_smem_bytes() {
 
 // One (req, kv_head[, z-chunk]) MFMA decode item. Shared by the standalone
 // decode_mfma kernel and the persistent holistic BatchAttention kernel.
-template <typename T, int HEAD_DIM, int GROUP>
+// TKV != T (fp8 e4m3 KV cache): tokens are dequantized to T during the LDS
+// staging write (reference prefill.cuh:1150 repack design) — HBM KV bytes
+// halve, the MFMA pipeline stays bf16/f16; k_scale folds into sm_scale and
+// v_scale is applied host-side (same contract as the vector kernel).
+template <typename T, int HEAD_DIM, int GROUP, typename TKV = T>
 __device__ __forceinline__ void decode_mfma_item_body(const DecodeParams& p, int req,
                                                       int kv_head, int zidx,
                                                       char* smem,
@@ -103,8 +107,8 @@ __device__ __forceinline__ void decode_mfma_item_body(const DecodeParams& p, int
   if (we_ > z_hi) we_ = z_hi;
 
   const int32_t* page_ids = p.kv_indices + p.kv_indptr[req];
-  const T* kbase = (const T*)p.k_data;
-  const T* vbase = (const T*)p.v_data;
+  const TKV* kbase = (const TKV*)p.k_data;
+  const TKV* vbase = (const TKV*)p.v_data;
 
   // ---- Q staged ONCE into LDS (B-operand fragments re-read per tile via
   // ds_read: lgkm-tracked, so they never wait on the in-flight HBM staging
@@ -147,7 +151,7 @@ __device__ __forceinline__ void decode_mfma_item_body(const DecodeParams& p, int
   // ---- register-staged K/V tile loads (one tile in regs while the staged
   // tile computes; 2 waves/SIMD co-residency fills the vmcnt gaps) ----
   constexpr int S_ITER = KVB * HEAD_DIM / 8 / 64;
-  vec_t<T, 8> kregA[S_ITER], vregA[S_ITER];
+  vec_t<TKV, 8> kregA[S_ITER], vregA[S_ITER];
   auto stage_load = [&](int64_t kv0) {
 #pragma unroll
     for (int it = 0; it < S_ITER; ++it) {
@@ -174,15 +178,25 @@ __device__ __forceinline__ void decode_mfma_item_body(const DecodeParams& p, int
       int u = lane + it * 64;
       int row = u / (HEAD_DIM / 8);
       int chunk8 = u % (HEAD_DIM / 8);
+      shortx8 kw, vw;
+      if constexpr (__is_same(T, TKV)) {
+        kw = *reinterpret_cast<const shortx8*>(kregA[it].data);
+        vw = *reinterpret_cast<const shortx8*>(vregA[it].data);
+      } else {
+        // fp8 -> T dequant on the staging write
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          reinterpret_cast<T*>(&kw)[j] = from_f32<T>(kregA[it].get(j));
+          reinterpret_cast<T*>(&vw)[j] = from_f32<T>(vregA[it].get(j));
+        }
+      }
       *reinterpret_cast<shortx8*>(reinterpret_cast<char*>(Ks(wave)) +
                                   (KROWB == 256 ? swz256(row * KROWB + chunk8 * 16)
-                                                : swz128(row * KROWB + chunk8 * 16))) =
-          *reinterpret_cast<const shortx8*>(kregA[it].data);
+                                                : swz128(row * KROWB + chunk8 * 16))) = kw;
       *reinterpret_cast<shortx8*>(
           reinterpret_cast<char*>(Vs(wave)) +
           ((row >> 2) * (HEAD_DIM / 16) + (chunk8 >> 1)) * (VTILE_STRIDE * 2) +
-          (row & 3) * 32 + (chunk8 & 1) * 16) =
-          *reinterpret_cast<const shortx8*>(vregA[it].data);
+          (row & 3) * 32 + (chunk8 & 1) * 16) = vw;
     }
   };
 

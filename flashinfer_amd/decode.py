@@ -174,8 +174,8 @@ class BatchDecodeWithPagedKVCacheWrapper:
         mfma_ok = (
             head_dim in (64, 128)
             and group in (8, 16, 32)
-            and kv_dt == q_data_type
             and q_data_type in (torch.bfloat16, torch.float16)
+            and kv_dt in (q_data_type, torch.float8_e4m3fn)
             and max_len <= _MFMA_MAX_KV
             and fixed_split_size is None
             and self._use_tensor_cores is not True

@@ -293,6 +293,31 @@ def test_decode_fused_fp8_kv():
         torch.testing.assert_close(out[b].float(), ref, atol=6e-2, rtol=6e-2)
 
 
+def test_decode_mfma_fp8_kv():
+    """fp8 e4m3 KV cache on the GQA-8 MFMA decode route (dequant happens in
+    the LDS staging write; k_scale folds into sm_scale, v_scale on the out)."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(5)
+    Hq, Hkv, D, page = 64, 8, 128, 16
+    kv_lens = [1024, 511, 77]
+    indptr, indices, lpl, kc, vc = _paged(3, kv_lens, Hkv, D, page)
+    kc8 = kc.clamp(-8, 8).to(torch.float8_e4m3fn)
+    vc8 = vc.clamp(-8, 8).to(torch.float8_e4m3fn)
+    q = torch.randn(3, Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(64 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(indptr, indices, lpl, Hq, Hkv, D, page,
+           q_data_type=torch.bfloat16, kv_data_type=torch.float8_e4m3fn)
+    assert w._fused_mfma, "GQA-8 fp8-KV must take the MFMA decode route"
+    out = w.run(q, (kc8, vc8), k_scale=0.5, v_scale=2.0)
+    for b in range(3):
+        kk, vv = _gather(indptr, indices, kv_lens,
+                         kc8.to(torch.bfloat16), vc8.to(torch.bfloat16), b, page)
+        ref = sdpa_ref(q[b], kk, vv, sm_scale=0.5 / D ** 0.5) * 2.0
+        torch.testing.assert_close(out[b].float(), ref, atol=6e-2, rtol=6e-2)
+
+
 def test_decode_tensor_cores_opt_out():
     """Explicit use_tensor_cores=False must never route to the prefill-MFMA
     path, and explicit True must (advisor r01 contract fix)."""
