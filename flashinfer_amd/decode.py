@@ -42,6 +42,12 @@ _FUSED_MAX_KV = 2048
 # where both the vector kernel (8 dot+shfl chains per K read) and the
 # prefill-reuse path (94%-padded 128-row tiles) lose.
 _MFMA_MAX_KV = 4096
+# fp8 e4m3 KV has no tensor-core fallback (the prefill-based tc path needs
+# kv_dt == q_dt) and the split-vector fp8 dequant is scalar (~0.39 TB/s at
+# bs=128/kv=4096 vs 2.2 TB/s on the MFMA route — profiles/README.md r02
+# addendum), so for fp8 KV the MFMA route applies at any length: its per-wave
+# KV slices + cross-WG split scale with kv_len with no structural cap.
+_MFMA_MAX_KV_F8 = 1 << 30
 
 
 def _plan_chunks(kv_lens, num_kv_heads: int, page_size: int,
@@ -176,7 +182,8 @@ class BatchDecodeWithPagedKVCacheWrapper:
             and group in (8, 16, 32)
             and q_data_type in (torch.bfloat16, torch.float16)
             and kv_dt in (q_data_type, torch.float8_e4m3fn)
-            and max_len <= _MFMA_MAX_KV
+            and max_len <= (_MFMA_MAX_KV if kv_dt == q_data_type
+                            else _MFMA_MAX_KV_F8)
             and fixed_split_size is None
             and self._use_tensor_cores is not True
             and self._use_tensor_cores is not False

@@ -6,9 +6,9 @@ import flashinfer_amd as fi
 
 def run(kv_dtype, B=16, L=1024, force=None):
     import flashinfer_amd.decode as dec
-    sv = (dec._FUSED_MAX_KV, dec._MFMA_MAX_KV)
+    sv = (dec._FUSED_MAX_KV, dec._MFMA_MAX_KV, dec._MFMA_MAX_KV_F8)
     if force == "vector":
-        dec._FUSED_MAX_KV = dec._MFMA_MAX_KV = 0
+        dec._FUSED_MAX_KV = dec._MFMA_MAX_KV = dec._MFMA_MAX_KV_F8 = 0
     torch.manual_seed(0)
     Hq, Hkv, D, page = 64, 8, 128, 16
     npages = B * (L // page)
@@ -27,7 +27,7 @@ def run(kv_dtype, B=16, L=1024, force=None):
         w.plan(indptr, indices, lpl, Hq, Hkv, D, page,
                q_data_type=torch.bfloat16, kv_data_type=kv_dtype)
     finally:
-        dec._FUSED_MAX_KV, dec._MFMA_MAX_KV = sv
+        dec._FUSED_MAX_KV, dec._MFMA_MAX_KV, dec._MFMA_MAX_KV_F8 = sv
     if force is None:
         assert w._fused_mfma, f"route not mfma for {kv_dtype}"
     for _ in range(20):
@@ -46,8 +46,7 @@ def run(kv_dtype, B=16, L=1024, force=None):
     return out
 
 
-for B, L in [(16, 1024), (128, 4096)]:
+for B, L in [(16, 1024), (128, 4096), (32, 16384), (8, 32768)]:
     run(torch.float8_e4m3fn, B, L)
-    run(torch.bfloat16, B, L)
-for dt in (torch.float8_e4m3fn, torch.bfloat16):
-    run(dt, 128, 4096, force="vector")
+for B, L in [(32, 16384), (8, 32768)]:
+    run(torch.float8_e4m3fn, B, L, force="vector")

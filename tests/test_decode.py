@@ -300,7 +300,9 @@ def test_decode_mfma_fp8_kv():
 
     torch.manual_seed(5)
     Hq, Hkv, D, page = 64, 8, 128, 16
-    kv_lens = [1024, 511, 77]
+    # 8192 > _MFMA_MAX_KV: fp8 KV stays on the MFMA route at any length
+    # (no tc fallback exists for fp8; see decode._MFMA_MAX_KV_F8)
+    kv_lens = [8192, 511, 77]
     indptr, indices, lpl, kc, vc = _paged(3, kv_lens, Hkv, D, page)
     kc8 = kc.clamp(-8, 8).to(torch.float8_e4m3fn)
     vc8 = vc.clamp(-8, 8).to(torch.float8_e4m3fn)
