@@ -586,6 +586,9 @@ hipError_t decode_dispatch(DecodeParams& p, hipStream_t stream) {
       case 1: LAUNCH_D(HD, 1, SC); break;               \
       case 2: LAUNCH_D(HD, 2, SC); break;               \
       case 4: LAUNCH_D(HD, 4, SC); break;               \
+      case 5: LAUNCH_D(HD, 5, SC); break;               \
+      case 6: LAUNCH_D(HD, 6, SC); break;               \
+      case 7: LAUNCH_D(HD, 7, SC); break;               \
       case 8: LAUNCH_D(HD, 8, SC); break;               \
       case 16: LAUNCH_D(HD, 16, SC); break;             \
       default: return hipErrorInvalidValue;             \

@@ -40,23 +40,31 @@ hipError_t decode_mfma_dispatch(DecodeParams& p, hipStream_t stream) {
   dim3 blk(kDecWaves * 64);
 #define LAUNCH_M(HD, G) \
   hipLaunchKernelGGL((decode_mfma_kernel<T, HD, G, TKV>), g, blk, 0, stream, p)
+  // every group <= 32 rides the same 32x32 tile: the q dim is padding-
+  // tolerant (QROWS rows live, rest zeroed), so the QK/PV cost per KV tile
+  // is identical for GROUP 1..32 — small groups (MHA, GQA-2/4, Mixtral's 6,
+  // Yi's 7) get the MFMA pipeline instead of the scalar vector fallback.
+#define LAUNCH_SW(HD)                                 \
+    switch (group) {                                  \
+      case 1: LAUNCH_M(HD, 1); break;                 \
+      case 2: LAUNCH_M(HD, 2); break;                 \
+      case 4: LAUNCH_M(HD, 4); break;                 \
+      case 5: LAUNCH_M(HD, 5); break;                 \
+      case 6: LAUNCH_M(HD, 6); break;                 \
+      case 7: LAUNCH_M(HD, 7); break;                 \
+      case 8: LAUNCH_M(HD, 8); break;                 \
+      case 16: LAUNCH_M(HD, 16); break;               \
+      case 32: LAUNCH_M(HD, 32); break;               \
+      default: return hipErrorInvalidValue;           \
+    }
   if (p.head_dim == 128) {
-    switch (group) {
-      case 8: LAUNCH_M(128, 8); break;
-      case 16: LAUNCH_M(128, 16); break;
-      case 32: LAUNCH_M(128, 32); break;
-      default: return hipErrorInvalidValue;
-    }
+    LAUNCH_SW(128);
   } else if (p.head_dim == 64) {
-    switch (group) {
-      case 8: LAUNCH_M(64, 8); break;
-      case 16: LAUNCH_M(64, 16); break;
-      case 32: LAUNCH_M(64, 32); break;
-      default: return hipErrorInvalidValue;
-    }
+    LAUNCH_SW(64);
   } else {
     return hipErrorInvalidValue;
   }
+#undef LAUNCH_SW
 #undef LAUNCH_M
   return hipGetLastError();
 }

@@ -177,13 +177,18 @@ class BatchDecodeWithPagedKVCacheWrapper:
             and fixed_split_size is None
             and self._use_tensor_cores is not True
         )
+        # any group <= 32 rides the 32x32 MFMA tile (q dim zero-padded, so
+        # per-KV-tile cost is group-independent); for bf16/f16 the fused
+        # vector kernel keeps precedence where it is eligible (group <= 4,
+        # short kv), for fp8 KV the MFMA route always wins (the vector
+        # kernels' fp8 dequant path is scalar — 5.6x slower measured).
         mfma_ok = (
             head_dim in (64, 128)
-            and group in (8, 16, 32)
+            and group in (1, 2, 4, 5, 6, 7, 8, 16, 32)
             and q_data_type in (torch.bfloat16, torch.float16)
             and kv_dt in (q_data_type, torch.float8_e4m3fn)
-            and max_len <= (_MFMA_MAX_KV if kv_dt == q_data_type
-                            else _MFMA_MAX_KV_F8)
+            and (max_len <= _MFMA_MAX_KV_F8 if kv_dt != q_data_type
+                 else (max_len <= _MFMA_MAX_KV and not fused_ok))
             and fixed_split_size is None
             and self._use_tensor_cores is not True
             and self._use_tensor_cores is not False

@@ -4,13 +4,15 @@ import torch, time
 import flashinfer_amd as fi
 
 
-def run(kv_dtype, B=16, L=1024, force=None):
+def run(kv_dtype, B=16, L=1024, force=None, Hq=64, Hkv=8):
     import flashinfer_amd.decode as dec
     sv = (dec._FUSED_MAX_KV, dec._MFMA_MAX_KV, dec._MFMA_MAX_KV_F8)
     if force == "vector":
         dec._FUSED_MAX_KV = dec._MFMA_MAX_KV = dec._MFMA_MAX_KV_F8 = 0
+    elif force == "fused":
+        dec._MFMA_MAX_KV = dec._MFMA_MAX_KV_F8 = 0
     torch.manual_seed(0)
-    Hq, Hkv, D, page = 64, 8, 128, 16
+    D, page = 128, 16
     npages = B * (L // page)
     kc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
     vc = torch.randn_like(kc)
@@ -28,7 +30,9 @@ def run(kv_dtype, B=16, L=1024, force=None):
                q_data_type=torch.bfloat16, kv_data_type=kv_dtype)
     finally:
         dec._FUSED_MAX_KV, dec._MFMA_MAX_KV, dec._MFMA_MAX_KV_F8 = sv
-    if force is None:
+    if force == "fused":
+        assert w._fused and not w._fused_mfma
+    elif force is None:
         assert w._fused_mfma, f"route not mfma for {kv_dtype}"
     for _ in range(20):
         out = w.run(q, (kc, vc))
@@ -40,7 +44,7 @@ def run(kv_dtype, B=16, L=1024, force=None):
     torch.cuda.synchronize()
     us = (time.perf_counter() - t0) / iters * 1e6
     kv_bytes = 2 * npages * page * Hkv * D * kc.element_size()
-    print(f"bs={B} kv={L} {str(kv_dtype):24s} {us:7.2f} us  "
+    print(f"bs={B} kv={L} G={Hq // Hkv} {str(kv_dtype):24s} {us:7.2f} us  "
           f"{kv_bytes / (us * 1e-6) / 1e12:5.2f} TB/s KV  "
           f"route={'mfma' if w._fused_mfma else ('fused' if w._fused else 'tc/vec')}")
     return out
@@ -48,5 +52,9 @@ def run(kv_dtype, B=16, L=1024, force=None):
 
 for B, L in [(16, 1024), (128, 4096), (32, 16384), (8, 32768)]:
     run(torch.float8_e4m3fn, B, L)
-for B, L in [(32, 16384), (8, 32768)]:
-    run(torch.float8_e4m3fn, B, L, force="vector")
+# small groups: GQA-6 bf16, GROUP-4 fp8 short (mfma vs fused A/B)
+run(torch.bfloat16, 16, 4096, Hq=48, Hkv=8)
+run(torch.bfloat16, 16, 4096, Hq=48, Hkv=8, force="vector")
+run(torch.float8_e4m3fn, 16, 1024, Hq=32, Hkv=8)
+run(torch.float8_e4m3fn, 16, 1024, Hq=32, Hkv=8, force="fused")
+run(torch.float8_e4m3fn, 16, 1024, Hq=32, Hkv=32)  # MHA fp8

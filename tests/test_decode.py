@@ -320,6 +320,39 @@ def test_decode_mfma_fp8_kv():
         torch.testing.assert_close(out[b].float(), ref, atol=6e-2, rtol=6e-2)
 
 
+@pytest.mark.parametrize("Hq,Hkv,kv_dtype", [
+    (48, 8, torch.bfloat16),    # GROUP 6 (Mixtral-8x22B shape)
+    (56, 8, torch.bfloat16),    # GROUP 7 (Yi-34B shape)
+    (32, 32, torch.float8_e4m3fn),   # MHA GROUP 1, fp8 KV
+    (32, 8, torch.float8_e4m3fn),    # GROUP 4, fp8 KV (mfma beats fused fp8)
+])
+def test_decode_mfma_small_groups(Hq, Hkv, kv_dtype):
+    """Non-power-of-2 / small GQA groups on the MFMA route: the 32x32 tile's
+    q dim is zero-padded, so any group <= 32 rides the same kernel."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(7)
+    D, page = 128, 16
+    kv_lens = [3000, 511, 64]
+    indptr, indices, lpl, kc, vc = _paged(3, kv_lens, Hkv, D, page)
+    if kv_dtype != torch.bfloat16:
+        kc = kc.clamp(-8, 8).to(kv_dtype)
+        vc = vc.clamp(-8, 8).to(kv_dtype)
+    q = torch.randn(3, Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(64 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(indptr, indices, lpl, Hq, Hkv, D, page,
+           q_data_type=torch.bfloat16, kv_data_type=kv_dtype)
+    assert w._fused_mfma, f"group {Hq // Hkv} must take the MFMA route here"
+    out = w.run(q, (kc, vc))
+    kvb = kc.to(torch.bfloat16) if kv_dtype != torch.bfloat16 else kc
+    vvb = vc.to(torch.bfloat16) if kv_dtype != torch.bfloat16 else vc
+    for b in range(3):
+        kk, vv = _gather(indptr, indices, kv_lens, kvb, vvb, b, page)
+        ref = sdpa_ref(q[b], kk, vv)
+        torch.testing.assert_close(out[b].float(), ref, atol=6e-2, rtol=6e-2)
+
+
 def test_decode_tensor_cores_opt_out():
     """Explicit use_tensor_cores=False must never route to the prefill-MFMA
     path, and explicit True must (advisor r01 contract fix)."""
