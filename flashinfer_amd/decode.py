@@ -193,13 +193,20 @@ class BatchDecodeWithPagedKVCacheWrapper:
             and self._use_tensor_cores is not True
             and self._use_tensor_cores is not False
         )
+        # the split-vector kernel instantiates groups {1,2,4,5,6,7,8,16}
+        # only; any other group (3, 12, 24, 64, ...) must take the
+        # group-agnostic prefill-based tensor-core path rather than fail
+        # at dispatch
+        vector_group_ok = group in (1, 2, 4, 5, 6, 7, 8, 16)
         self._tc = (
             not fused_ok and not mfma_ok
             and kv_dt == q_data_type
             and (
                 self._use_tensor_cores is True
-                or (self._use_tensor_cores is None and group >= 8
+                or (self._use_tensor_cores is None
+                    and (group >= 8 or not vector_group_ok)
                     and fixed_split_size is None and not disable_split_kv)
+                or not vector_group_ok
             )
         )
         self._fused = (fused_ok or mfma_ok) and not self._tc
