@@ -190,7 +190,10 @@ class BatchDecodeWithPagedKVCacheWrapper:
             and (max_len <= _MFMA_MAX_KV_F8 if kv_dt != q_data_type
                  else (max_len <= _MFMA_MAX_KV and not fused_ok))
             and fixed_split_size is None
-            and self._use_tensor_cores is not True
+            # use_tensor_cores=True with an fp8 KV cache means THIS kernel:
+            # the prefill-based tc path needs kv_dt == q_dt, and the MFMA
+            # decode kernel is the matrix-core path for fp8
+            and (self._use_tensor_cores is not True or kv_dt != q_data_type)
             and self._use_tensor_cores is not False
         )
         # the split-vector kernel instantiates groups {1,2,4,5,6,7,8,16}

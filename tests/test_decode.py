@@ -325,6 +325,7 @@ def test_decode_mfma_fp8_kv():
     (56, 8, torch.bfloat16),    # GROUP 7 (Yi-34B shape)
     (32, 32, torch.float8_e4m3fn),   # MHA GROUP 1, fp8 KV
     (32, 8, torch.float8_e4m3fn),    # GROUP 4, fp8 KV (mfma beats fused fp8)
+    (32, 8, torch.bfloat16),         # GROUP 4 bf16, kv>2048: past fused range
 ])
 def test_decode_mfma_small_groups(Hq, Hkv, kv_dtype):
     """Non-power-of-2 / small GQA groups on the MFMA route: the 32x32 tile's
@@ -351,6 +352,28 @@ def test_decode_mfma_small_groups(Hq, Hkv, kv_dtype):
         kk, vv = _gather(indptr, indices, kv_lens, kvb, vvb, b, page)
         ref = sdpa_ref(q[b], kk, vv)
         torch.testing.assert_close(out[b].float(), ref, atol=6e-2, rtol=6e-2)
+
+
+def test_decode_odd_group_tc_fallback():
+    """Groups outside the instantiated vector/mfma sets (e.g. 3) must route
+    to the group-agnostic prefill-based tc path, not fail at dispatch."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(9)
+    Hq, Hkv, D, page = 24, 8, 128, 16   # group 3
+    kv_lens = [700, 33]
+    indptr, indices, lpl, kc, vc = _paged(2, kv_lens, Hkv, D, page)
+    q = torch.randn(2, Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(64 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(indptr, indices, lpl, Hq, Hkv, D, page,
+           q_data_type=torch.bfloat16)
+    assert w._tc, "group 3 must fall back to the tensor-core path"
+    out = w.run(q, (kc, vc))
+    for b in range(2):
+        kk, vv = _gather(indptr, indices, kv_lens, kc, vc, b, page)
+        torch.testing.assert_close(out[b].float(), sdpa_ref(q[b], kk, vv),
+                                   atol=3e-2, rtol=3e-2)
 
 
 def test_decode_tensor_cores_opt_out():
