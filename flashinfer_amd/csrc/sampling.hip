@@ -114,9 +114,14 @@ __global__ void softmax_kernel(SamplingParams p) {
 // ---------------- generic rejection sampler ----------------
 // MODE: 0 = plain (no constraint), 1 = top-k, 2 = top-p, 3 = top-k AND top-p,
 //       4 = min-p (single round: mask p < min_p * max_p)
+// vocab cap for the LDS chunk-sum hierarchy (512 chunks x 2048 = 1M ids;
+// largest real vocabs are ~256k). The host dispatch guards this.
+constexpr int kMaxSampChunks = 512;
+
 template <int MODE, bool FROM_LOGITS>
 __global__ void sampling_kernel(SamplingParams p) {
   __shared__ float smem[SW + 1];
+  __shared__ float s_csum[kMaxSampChunks];
   __shared__ float s_scalar;
   __shared__ int s_cand;
   for (int b = blockIdx.x; b < p.rows; b += gridDim.x) {
@@ -161,64 +166,94 @@ __global__ void sampling_kernel(SamplingParams p) {
       if (pivot <= 0.f) pivot = -1.f;
     }
 
+    // ---- per-chunk partial-sum hierarchy (r02 roadmap item 7) ----
+    // One full-vocab pass builds s_csum[c] = sum of probs > pivot in chunk
+    // c; the inverse-CDF walk then touches ONE chunk (selected by a serial
+    // deterministic scan of s_csum). Each rejection round fuses the next
+    // pivot's rebuild into the acceptance pass, so a round costs ~1 vocab
+    // pass instead of the previous ~2.5 (pass1 + half-vocab walk + accept).
+    const int nchunks = (p.vocab + CHUNK - 1) / CHUNK;
+    const int wv = threadIdx.x / kWaveSize, lane = threadIdx.x % kWaveSize;
+    {
+      for (int c = wv; c < nchunks; c += SW) {
+        float sch = 0.f;
+        const int base = c * CHUNK;
+        const int lim = base + CHUNK < p.vocab ? base + CHUNK : p.vocab;
+        for (int i = base + lane; i < lim; i += kWaveSize) {
+          float v = P(i);
+          if (v > pivot) sch += v;
+        }
+        sch = wave_reduce_sum<kWaveSize>(sch);
+        if (lane == 0) s_csum[c] = sch;
+      }
+      __syncthreads();
+    }
+
     int cand = -1;
     for (int round = 0; round < p.rounds; ++round) {
       float u = p.uniforms[(uint64_t)b * p.rounds + round];
-      // pass 1: total mass above pivot
-      float tsum = 0.f;
-      for (int i = threadIdx.x; i < p.vocab; i += SB) {
-        float v = P(i);
-        if (v > pivot) tsum += v;
-      }
-      float total = block_sum(tsum, smem);
-      if (total <= 0.f) break;  // nothing above pivot (keep last cand)
-      float target = u * total;
-      // pass 2: ordered CDF walk
-      float running = 0.f;
-      int found = INT_MAX, last_valid = -1;
-      for (int base = 0; base < p.vocab; base += CHUNK) {
-        float vals[SVEC];
-        int i0 = base + threadIdx.x * SVEC;
-        float th_sum = 0.f;
-#pragma unroll
-        for (int j = 0; j < SVEC; ++j) {
-          int i = i0 + j;
-          float v = (i < p.vocab) ? P(i) : 0.f;
-          vals[j] = (v > pivot) ? v : 0.f;
-          th_sum += vals[j];
-        }
-        float ctot;
-        float prefix = block_exscan(th_sum, smem, &ctot);
-        if (running + ctot > target) {
-          // candidate is in this chunk
-          float loc = running + prefix;
-          int my = INT_MAX;
-#pragma unroll
-          for (int j = 0; j < SVEC; ++j) {
-            if (vals[j] > 0.f) {
-              if (my == INT_MAX && loc + vals[j] > target) my = i0 + j;
-              loc += vals[j];
-            }
+      // chunk select: serial scan of the per-chunk sums (deterministic,
+      // <= kMaxSampChunks dependent adds — negligible next to a pass)
+      if (threadIdx.x == 0) {
+        float total = 0.f;
+        for (int c = 0; c < nchunks; ++c) total += s_csum[c];
+        int cstar = -1;
+        float run = 0.f;
+        if (total > 0.f) {
+          float target = u * total;
+          for (int c = 0; c < nchunks; ++c) {
+            float cs = s_csum[c];
+            if (run + cs > target && cs > 0.f) { cstar = c; break; }
+            run += cs;
           }
-          // block-min to pick the first satisfying element
-          if (threadIdx.x == 0) s_cand = INT_MAX;
-          __syncthreads();
-          if (my != INT_MAX) atomicMin(&s_cand, my);
-          __syncthreads();
-          found = s_cand;
-          if (found != INT_MAX) break;
-          // numerical edge: fall through to next chunk
-          running += ctot;
-        } else {
-          running += ctot;
+          if (cstar < 0) {
+            // rounding tail: last chunk with mass
+            run = 0.f;
+            for (int c = 0; c < nchunks; ++c) {
+              if (s_csum[c] > 0.f) cstar = c;
+            }
+            for (int c = 0; c < cstar; ++c) run += s_csum[c];
+            target = run + s_csum[cstar];  // clamp into the chunk
+          }
+          s_scalar = target - run;  // in-chunk target mass
         }
+        s_cand = cstar;
+      }
+      __syncthreads();
+      const int cstar = s_cand;
+      if (cstar < 0) break;  // nothing above pivot (keep last cand)
+      const float trem = s_scalar;
+      __syncthreads();
+      // in-chunk inverse-CDF walk (one chunk == one CHUNK-wide block scan)
+      const int i0 = cstar * CHUNK + threadIdx.x * SVEC;
+      float vals[SVEC];
+      float th_sum = 0.f;
 #pragma unroll
-        for (int j = 0; j < SVEC; ++j) {
-          if (vals[j] > 0.f) last_valid = i0 + j;
+      for (int j = 0; j < SVEC; ++j) {
+        int i = i0 + j;
+        float v = (i < p.vocab) ? P(i) : 0.f;
+        vals[j] = (v > pivot) ? v : 0.f;
+        th_sum += vals[j];
+      }
+      float ctot;
+      float loc = block_exscan(th_sum, smem, &ctot);
+      int my = INT_MAX, last_valid = -1;
+#pragma unroll
+      for (int j = 0; j < SVEC; ++j) {
+        if (vals[j] > 0.f) {
+          if (my == INT_MAX && loc + vals[j] > trem) my = i0 + j;
+          loc += vals[j];
+          last_valid = i0 + j;
         }
       }
+      if (threadIdx.x == 0) s_cand = INT_MAX;
+      __syncthreads();
+      if (my != INT_MAX) atomicMin(&s_cand, my);
+      __syncthreads();
+      int found = s_cand;
       if (found == INT_MAX) {
-        // fallback: last valid element (block max of last_valid)
+        // rounding edge: target beyond the chunk's walked mass — take the
+        // chunk's last valid element
         if (threadIdx.x == 0) s_cand = -1;
         __syncthreads();
         if (last_valid >= 0) atomicMax(&s_cand, last_valid);
@@ -228,24 +263,33 @@ __global__ void sampling_kernel(SamplingParams p) {
       }
       cand = found;
       if constexpr (MODE == 0 || MODE == 4) break;  // no rejection test
-      // acceptance test
+      // acceptance test FUSED with the csum rebuild for pivot=pc: if this
+      // candidate is rejected the next round's hierarchy is already built
       float pc = P(cand);
       float gt_sum = 0.f;
-      int gt_cnt = 0;
-      for (int i = threadIdx.x; i < p.vocab; i += SB) {
-        float v = P(i);
-        if (v > pc) {
-          gt_sum += v;
-          gt_cnt++;
+      float gt_cnt = 0.f;
+      for (int c = wv; c < nchunks; c += SW) {
+        float sch = 0.f;
+        const int base = c * CHUNK;
+        const int lim = base + CHUNK < p.vocab ? base + CHUNK : p.vocab;
+        for (int i = base + lane; i < lim; i += kWaveSize) {
+          float v = P(i);
+          if (v > pc) {
+            sch += v;
+            gt_cnt += 1.f;
+          }
         }
+        gt_sum += sch;
+        sch = wave_reduce_sum<kWaveSize>(sch);
+        if (lane == 0) s_csum[c] = sch;
       }
       float g_sum = block_sum(gt_sum, smem);
-      float g_cnt = block_sum((float)gt_cnt, smem);
+      float g_cnt = block_sum(gt_cnt, smem);
       bool ok = true;
       if constexpr (MODE == 1 || MODE == 3) ok &= (g_cnt < k);
       if constexpr (MODE == 2 || MODE == 3) ok &= (g_sum < pp);
       if (ok) break;
-      pivot = pc;  // reject: tighten
+      pivot = pc;  // reject: tighten (s_csum already rebuilt for pc)
     }
     if (threadIdx.x == 0) p.out_ids[b] = cand < 0 ? 0 : cand;
     __syncthreads();
@@ -436,6 +480,7 @@ extern "C" hipError_t fi_softmax(fi::SamplingParams* p, hipStream_t stream) {
 // mode: 0 plain, 1 topk, 2 topp, 3 topk+topp, 4 minp; from_logits bool
 extern "C" hipError_t fi_sampling(int mode, int from_logits, fi::SamplingParams* p,
                                   hipStream_t stream) {
+  if (p->vocab > fi::kMaxSampChunks * fi::CHUNK) return hipErrorInvalidValue;
   int grid = p->rows < 1024 ? p->rows : 1024;
   dim3 g(grid), blk(fi::SB);
 #define LAUNCH_S(M, L) \
