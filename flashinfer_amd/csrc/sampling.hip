@@ -307,9 +307,20 @@ __device__ __forceinline__ float u2f(uint32_t u) {
 }
 
 // WHICH: 0 = top_k_renorm_probs, 1 = top_p_renorm_probs, 2 = top_k_mask_logits
+// The k-th/top-p threshold tau is found EXACTLY (full 32-bit float ordering)
+// by a 3-level radix histogram over the monotonic-uint space (11+11+10
+// bits): 3 vocab passes instead of the former 24-iteration binary search
+// (~24 passes). Counts use integer LDS atomics (order-independent, so
+// top-k results stay bitwise deterministic); per-bin float sums (top-p bin
+// selection only) use float atomics — the final renorm sum is recomputed
+// in the fixed block_sum order.
 template <int WHICH>
 __global__ void renorm_kernel(SamplingParams p) {
   __shared__ float smem[SW + 1];
+  __shared__ uint32_t s_hcnt[2048];
+  __shared__ float s_hsum[2048];
+  __shared__ uint32_t s_sel;
+  __shared__ float s_carry[2];
   for (int b = blockIdx.x; b < p.rows; b += gridDim.x) {
     const float* in = p.probs + (uint64_t)b * p.stride_row;
     float* out = p.out_probs + (uint64_t)b * p.vocab;
@@ -321,28 +332,64 @@ __global__ void renorm_kernel(SamplingParams p) {
       __syncthreads();
       continue;
     }
-    // binary search threshold tau over the float ordering: keep v >= tau.
-    // top_k: |{v >= tau}| >= k, maximize tau. top_p: sum_{v >= tau} >= p.
-    uint32_t lo = 0, hi = 0xFFFFFFFFu;  // monotonic-u space
-    for (int it = 0; it < 24; ++it) {
-      uint32_t mid = lo + ((hi - lo) >> 1);
-      float tau = u2f(mid);
-      float cnt = 0.f, sum = 0.f;
+    uint32_t prefix = 0;      // matched high bits of tau's monotonic uint
+    float cum_cnt = 0.f, cum_sum = 0.f;  // mass strictly above prefix region
+    constexpr int kShift[3] = {21, 10, 0};
+    constexpr int kBits[3] = {11, 11, 10};
+    for (int lvl = 0; lvl < 3; ++lvl) {
+      const int nb = 1 << kBits[lvl];
+      const int hi_shift = kShift[lvl] + kBits[lvl];  // bits already matched
+      for (int i = threadIdx.x; i < nb; i += SB) {
+        s_hcnt[i] = 0;
+        if constexpr (WHICH == 1) s_hsum[i] = 0.f;
+      }
+      __syncthreads();
       for (int i = threadIdx.x; i < p.vocab; i += SB) {
         float v = in[i];
-        if (v >= tau) {
-          cnt += 1.f;
-          sum += v;
+        uint32_t u = f2u(v);
+        if (lvl == 0 || (u >> hi_shift) == prefix) {
+          int bin = (u >> kShift[lvl]) & (nb - 1);
+          atomicAdd(&s_hcnt[bin], 1u);
+          if constexpr (WHICH == 1) atomicAdd(&s_hsum[bin], v);
         }
       }
-      float g_cnt = block_sum(cnt, smem);
-      float g_sum = block_sum(sum, smem);
-      bool enough = (WHICH == 1) ? (g_sum >= pp) : (g_cnt >= (float)k);
-      if (enough) lo = mid;  // can raise tau
-      else hi = mid - 1;
-      if (lo >= hi) break;
+      __syncthreads();
+      if (threadIdx.x == 0) {
+        int chosen = -1;
+        float cc = cum_cnt, cs = cum_sum;
+        for (int bin = nb - 1; bin >= 0; --bin) {
+          float bc = (float)s_hcnt[bin];
+          if (bc > 0.f) {
+            bool crossed = (WHICH == 1) ? (cs + s_hsum[bin] >= pp)
+                                        : (cc + bc >= (float)k);
+            if (crossed) { chosen = bin; break; }
+          }
+          cc += bc;
+          if constexpr (WHICH == 1) cs += s_hsum[bin];
+        }
+        if (chosen < 0) {
+          // never crossed (rounding tail / pp > total): lowest present bin
+          cc = cum_cnt; cs = cum_sum;
+          for (int bin = nb - 1; bin >= 0; --bin) {
+            if (s_hcnt[bin] > 0) chosen = bin;
+          }
+          if (chosen < 0) chosen = 0;
+          for (int bin = nb - 1; bin > chosen; --bin) {
+            cc += (float)s_hcnt[bin];
+            if constexpr (WHICH == 1) cs += s_hsum[bin];
+          }
+        }
+        s_sel = (uint32_t)chosen;
+        s_carry[0] = cc;
+        s_carry[1] = cs;
+      }
+      __syncthreads();
+      prefix = (prefix << kBits[lvl]) | s_sel;
+      cum_cnt = s_carry[0];
+      cum_sum = s_carry[1];
+      __syncthreads();
     }
-    float tau = u2f(lo);
+    float tau = u2f(prefix);  // exact k-th / top-p boundary value
     // renormalize / mask
     float ssum = 0.f;
     if (WHICH != 2) {

@@ -30,3 +30,8 @@ print(f"min_p(0.05): {timeit(lambda: fi.min_p_sampling_from_probs(probs, 0.05)):
 print(f"plain      : {timeit(lambda: fi.sampling_from_probs(probs)):7.3f} ms")
 print(f"logits topk50: {timeit(lambda: fi.top_k_top_p_sampling_from_logits(logits, 50, 1.0)):7.3f} ms")
 print(f"torch.multinomial: {timeit(lambda: torch.multinomial(probs, 1)):7.3f} ms")
+
+from flashinfer_amd import sampling
+print(f"top_k_renorm(50) : {timeit(lambda: sampling.top_k_renorm_probs(probs, 50)):7.3f} ms")
+print(f"top_p_renorm(0.9): {timeit(lambda: sampling.top_p_renorm_probs(probs, 0.9)):7.3f} ms")
+print(f"top_k_mask_logits: {timeit(lambda: sampling.top_k_mask_logits(logits, 50)):7.3f} ms")
