@@ -349,8 +349,9 @@ __global__ void renorm_kernel(SamplingParams p) {
         uint32_t u = f2u(v);
         if (lvl == 0 || (u >> hi_shift) == prefix) {
           int bin = (u >> kShift[lvl]) & (nb - 1);
-          atomicAdd(&s_hcnt[bin], 1u);
+          // top-p selects on mass only; skip the count atomic there
           if constexpr (WHICH == 1) atomicAdd(&s_hsum[bin], v);
+          else atomicAdd(&s_hcnt[bin], 1u);
         }
       }
       __syncthreads();
@@ -358,25 +359,26 @@ __global__ void renorm_kernel(SamplingParams p) {
         int chosen = -1;
         float cc = cum_cnt, cs = cum_sum;
         for (int bin = nb - 1; bin >= 0; --bin) {
-          float bc = (float)s_hcnt[bin];
-          if (bc > 0.f) {
+          bool present = (WHICH == 1) ? (s_hsum[bin] != 0.f) : (s_hcnt[bin] > 0);
+          if (present) {
             bool crossed = (WHICH == 1) ? (cs + s_hsum[bin] >= pp)
-                                        : (cc + bc >= (float)k);
+                                        : (cc + (float)s_hcnt[bin] >= (float)k);
             if (crossed) { chosen = bin; break; }
           }
-          cc += bc;
           if constexpr (WHICH == 1) cs += s_hsum[bin];
+          else cc += (float)s_hcnt[bin];
         }
         if (chosen < 0) {
           // never crossed (rounding tail / pp > total): lowest present bin
           cc = cum_cnt; cs = cum_sum;
           for (int bin = nb - 1; bin >= 0; --bin) {
-            if (s_hcnt[bin] > 0) chosen = bin;
+            bool present = (WHICH == 1) ? (s_hsum[bin] != 0.f) : (s_hcnt[bin] > 0);
+            if (present) chosen = bin;
           }
           if (chosen < 0) chosen = 0;
           for (int bin = nb - 1; bin > chosen; --bin) {
-            cc += (float)s_hcnt[bin];
             if constexpr (WHICH == 1) cs += s_hsum[bin];
+            else cc += (float)s_hcnt[bin];
           }
         }
         s_sel = (uint32_t)chosen;
