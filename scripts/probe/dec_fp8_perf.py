@@ -50,6 +50,8 @@ def run(kv_dtype, B=16, L=1024, force=None, Hq=64, Hkv=8):
     return out
 
 
+for B, L in [(16, 1024), (64, 1024), (128, 4096)]:
+    run(torch.bfloat16, B, L)
 for B, L in [(16, 1024), (128, 4096), (32, 16384), (8, 32768)]:
     run(torch.float8_e4m3fn, B, L)
 # small groups: GQA-6 bf16, GROUP-4 fp8 short (mfma vs fused A/B)
