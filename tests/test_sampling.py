@@ -165,3 +165,52 @@ def test_chain_speculative_sampling():
     )
     assert out[0, 0] == 7 and out[0, 1] == -1
     assert acc[0] == 0
+
+
+def test_renorm_large_vocab_exact():
+    """128k vocab exercises all three radix-histogram levels; the threshold
+    must match torch.topk exactly (bit-exact tau)."""
+    from flashinfer_amd import sampling
+
+    torch.manual_seed(11)
+    B, V = 8, 128256
+    logits = torch.randn(B, V, device="cuda") * 6
+    probs = torch.softmax(logits, -1)
+    for k in (1, 50, 1000):
+        out = sampling.top_k_renorm_probs(probs, k)
+        kth = probs.topk(k, -1).values[:, -1:]
+        ref = torch.where(probs >= kth, probs, torch.zeros_like(probs))
+        ref = ref / ref.sum(-1, keepdim=True)
+        torch.testing.assert_close(out, ref, atol=1e-6, rtol=1e-5)
+        mask = sampling.top_k_mask_logits(logits, k)
+        kth_l = logits.topk(k, -1).values[:, -1:]
+        assert ((mask == float("-inf")) == (logits < kth_l)).all(), k
+    # per-row k tensor
+    ks = torch.tensor([1, 7, 50, 333, 1000, 5, 2, 64], dtype=torch.int32,
+                      device="cuda")
+    out = sampling.top_k_renorm_probs(probs, ks)
+    for b in range(B):
+        kth = probs[b].topk(int(ks[b]), -1).values[-1]
+        ref = torch.where(probs[b] >= kth, probs[b],
+                          torch.zeros_like(probs[b]))
+        torch.testing.assert_close(out[b], ref / ref.sum(), atol=1e-6,
+                                   rtol=1e-5)
+
+
+def test_top_p_renorm_large_vocab():
+    from flashinfer_amd import sampling
+
+    torch.manual_seed(12)
+    B, V = 8, 128256
+    probs = torch.softmax(torch.randn(B, V, device="cuda") * 4, -1)
+    pth = 0.9
+    out = sampling.top_p_renorm_probs(probs, pth)
+    # kept set = minimal top mass >= pth: verify via sort
+    sp, si = probs.sort(-1, descending=True)
+    cum = sp.cumsum(-1)
+    nkeep = (cum < pth).sum(-1) + 1
+    for b in range(B):
+        kept = out[b] > 0
+        assert kept.sum() == nkeep[b], (b, int(kept.sum()), int(nkeep[b]))
+    torch.testing.assert_close(out.sum(-1), torch.ones(B, device="cuda"),
+                               atol=1e-5, rtol=1e-5)
