@@ -291,3 +291,19 @@ def test_moe_e2e_mxfp8():
             ref[t] += float(weights[t, j]) * (act @ w2[e].float().t())
     # pow2 scales cost up to ~1 bit of quant precision vs f32 groupwise
     torch.testing.assert_close(out.float(), ref, atol=0.5, rtol=0.2)
+
+
+@pytest.mark.gpu
+def test_dsv3_router_gemms():
+    """DSv3 skinny router GEMMs (dsv3_ops re-exports) vs torch matmul."""
+    from flashinfer_amd import dsv3_ops
+
+    torch.manual_seed(0)
+    for M in (1, 16):
+        a = torch.randn(M, 7168, dtype=torch.bfloat16, device="cuda")
+        for N, fn in ((128, dsv3_ops.mm_M1_16_K7168_N128),
+                      (256, dsv3_ops.mm_M1_16_K7168_N256)):
+            b = torch.randn(7168, N, dtype=torch.bfloat16, device="cuda")
+            out = fn(a, b)
+            ref = (a.float() @ b.float())
+            torch.testing.assert_close(out.float(), ref, atol=2e-1, rtol=2e-2)
