@@ -487,6 +487,10 @@ void batch_decode_mfma_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                 "split mfma decode needs tmp_v/tmp_s");
     p.tmp_v = (float*)tmp_v->data_ptr();
     p.tmp_s = tmp_s->data_ptr<float>();
+    // same-XCD in-kernel merge (host gates on (batch*kv_heads)%8==0)
+    p.counters = counters.has_value()
+                     ? (uint32_t*)counters->data_ptr<int32_t>()
+                     : nullptr;
   }
   check_hip(fi_decode_mfma(dtype_code(q), dtype_code(k_cache), &p,
                            cur_stream(q)),

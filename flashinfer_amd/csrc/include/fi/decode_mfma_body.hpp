@@ -356,10 +356,12 @@ __device__ __forceinline__ void decode_mfma_item_body(const DecodeParams& p, int
   __syncthreads();
 
   // ---- cross-wave merge; split==1 stores directly, else the per-z
-  // NORMALIZED partial + base-2 lse goes to the merge-kernel workspace
-  // (an in-kernel atomic last-WG merge was tried: the device-scope
-  // threadfence it needs write-backs L2 across the 8 XCDs and measured 4x
-  // slower than a second tiny launch — profiles/README r02) ----
+  // NORMALIZED partial + base-2 lse goes to the merge workspace, merged
+  // either by the merge_states launch or (when the host passes counters)
+  // by the last-arriving z-WG below. (A DEVICE-scope fence variant
+  // measured 4x slower than a second launch — profiles/README r02; the
+  // counter variant needs no fence because the host only enables it when
+  // all z-WGs of a (req, kv_head) share one XCD.) ----
   T* obase = (T*)p.o + (int64_t)qo_row * p.o_stride_n;
   for (int idx = tid; idx < QROWS * HEAD_DIM; idx += WAVES * 64) {
     int g = idx / HEAD_DIM;
@@ -390,6 +392,76 @@ __device__ __forceinline__ void decode_mfma_item_body(const DecodeParams& p, int
       if (d == 0)
         p.tmp_s[item * p.num_qo_heads + qh] =
             d_sum > 0.f ? m_star + __builtin_log2f(d_sum) : -INFINITY;
+    }
+  }
+
+  // ---- same-XCD in-kernel split merge. Precondition (host-gated):
+  // (batch * num_kv_heads) % 8 == 0, so the dispatcher's strict
+  // round-robin (measured: linear_workgroup_id % 8 == XCC_ID exactly —
+  // scripts/probe/xcd_probe.hip) places every z-WG of one (req, kv_head)
+  // on the SAME XCD. Partials are then coherent in that XCD's L2 and the
+  // last-arriving WG merges them with plain loads + an agent-scope
+  // counter — no device-scope fence (the fenced variant measured 4x
+  // slower). The winner resets its counter, so plan-time zeroing
+  // suffices and the path is hipGraph-safe. ----
+  if (split > 1 && p.counters) {
+    __shared__ int s_lastwg;
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // partials in L2
+    __syncthreads();
+    if (tid == 0) {
+      uint32_t* ctr = p.counters + (int64_t)req * p.num_kv_heads + kv_head;
+      // RELAXED on purpose: acq_rel at agent scope emits buffer_wbl2 +
+      // buffer_inv (cross-XCD L2 maintenance — the 4x-slower mechanism).
+      // Within one XCD the vmcnt(0) drain above IS the release (stores are
+      // at the shared L2 before the atomic issues, and the atomic lands at
+      // that same L2 in order); the winner's merge loads hit the same L2.
+      uint32_t prev = __hip_atomic_fetch_add(ctr, 1u, __ATOMIC_RELAXED,
+                                             __HIP_MEMORY_SCOPE_AGENT);
+      s_lastwg = (prev == (uint32_t)(split - 1));
+      if (s_lastwg)
+        __hip_atomic_store(ctr, 0u, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+      asm volatile("" ::: "memory");  // compile-time order: loads after win
+    }
+    __syncthreads();
+    if (!s_lastwg) return;
+    const int64_t base_item = (int64_t)qo_row * split;
+    // per-g weights computed once (not per output element): one thread per
+    // g reads the split lse values, exp2-weights them into LDS, writes lse
+    __shared__ float s_mw[32 * 8];  // [g][z] normalized weight
+    if (tid < QROWS) {
+      const int g = tid;
+      const int qh = kv_head * GROUP + g;
+      float m_star = -INFINITY, sz[8];
+      for (int z = 0; z < split; ++z) {
+        sz[z] = p.tmp_s[(base_item + z) * p.num_qo_heads + qh];
+        m_star = fmaxf(m_star, sz[z]);
+      }
+      float wsum = 0.f;
+      for (int z = 0; z < split; ++z) {
+        float wz = (m_star == -INFINITY || sz[z] == -INFINITY)
+                       ? 0.f
+                       : __builtin_exp2f(sz[z] - m_star);
+        s_mw[g * 8 + z] = wz;
+        wsum += wz;
+      }
+      float inv = wsum > 0.f ? 1.f / wsum : 0.f;
+      for (int z = 0; z < split; ++z) s_mw[g * 8 + z] *= inv;
+      if (p.lse)
+        p.lse[(int64_t)qo_row * p.num_qo_heads + qh] =
+            wsum > 0.f ? m_star + __builtin_log2f(wsum) : -INFINITY;
+    }
+    __syncthreads();
+    for (int idx = tid; idx < QROWS * HEAD_DIM; idx += WAVES * 64) {
+      int g = idx / HEAD_DIM;
+      int d = idx % HEAD_DIM;
+      int qh = kv_head * GROUP + g;
+      float osum = 0.f;
+      for (int z = 0; z < split; ++z)
+        osum += p.tmp_v[((base_item + z) * p.num_qo_heads + qh) * HEAD_DIM +
+                        d] *
+                s_mw[g * 8 + z];
+      obase[(int64_t)qh * p.o_stride_h + d] = from_f32<T>(osum);
     }
   }
 }

@@ -300,6 +300,16 @@ class BatchDecodeWithPagedKVCacheWrapper:
                     self._mfma_merge_indptr = torch.arange(
                         0, (batch + 1) * self._mfma_split, self._mfma_split,
                         dtype=torch.int32).to(dev, non_blocking=non_blocking)
+                    # same-XCD in-kernel merge: the dispatcher round-robins
+                    # linear workgroup ids across the 8 XCDs, so when
+                    # batch*kv_heads % 8 == 0 every z-WG of a (req, kv_head)
+                    # lands on one XCD and the last arriver merges in-kernel
+                    # (no merge_states launch). Counters self-reset; zeroed
+                    # once here (hipGraph-safe).
+                    self._mfma_counters = None
+                    if units % 8 == 0:
+                        self._mfma_counters = torch.zeros(
+                            units, dtype=torch.int32, device=dev)
             self._plan_info = dict(
                 batch=batch, num_qo_heads=num_qo_heads,
                 num_kv_heads=num_kv_heads, head_dim=head_dim,
@@ -419,9 +429,9 @@ class BatchDecodeWithPagedKVCacheWrapper:
                     pi["alibi"], sp,
                     self._mfma_tmp_v if sp > 1 else None,
                     self._mfma_tmp_s if sp > 1 else None,
-                    None,
+                    getattr(self, "_mfma_counters", None) if sp > 1 else None,
                 )
-                if sp > 1:
+                if sp > 1 and getattr(self, "_mfma_counters", None) is None:
                     get_ext().merge_states(
                         self._mfma_tmp_v, self._mfma_tmp_s, out, lse,
                         self._mfma_merge_indptr, 0, pi["batch"])
