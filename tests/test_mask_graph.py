@@ -246,3 +246,39 @@ def test_fused_decode_graph_capture():
     g.replay()
     torch.cuda.synchronize()
     torch.testing.assert_close(out, ref)
+
+
+def test_mfma_split_inkernel_merge_graph_capture():
+    """MFMA decode with cross-WG split + same-XCD in-kernel merge captures
+    into a hipGraph and replays bitwise — the arrival counters self-reset,
+    so no per-replay zeroing is needed."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(6)
+    Hq, Hkv, D, page = 64, 8, 128, 16
+    bs, L = 16, 1024
+    npages = bs * (L // page)
+    kc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn_like(kc)
+    indptr = torch.arange(0, bs + 1, dtype=torch.int32, device="cuda") * (L // page)
+    indices = torch.arange(npages, dtype=torch.int32, device="cuda")
+    lpl = torch.full((bs,), page, dtype=torch.int32, device="cuda")
+    q = torch.randn(bs, Hq, D, dtype=torch.bfloat16, device="cuda")
+    out = torch.empty_like(q)
+    ws = torch.empty(128 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(indptr, indices, lpl, Hq, Hkv, D, page, q_data_type=torch.bfloat16)
+    assert w._fused_mfma and w._mfma_split > 1
+    assert w._mfma_counters is not None, "in-kernel merge must be active"
+    w.run(q, (kc, vc), out=out)
+    torch.cuda.synchronize()
+    ref = out.clone()
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        w.run(q, (kc, vc), out=out)
+    for _ in range(3):
+        out.zero_()
+        g.replay()
+        torch.cuda.synchronize()
+        torch.testing.assert_close(out, ref)
+    assert int(w._mfma_counters.abs().sum().item()) == 0, "counters must self-reset"
