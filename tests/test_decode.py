@@ -55,7 +55,9 @@ def test_single_decode_soft_cap_window():
 
 @pytest.mark.parametrize("page_size", [1, 16])
 @pytest.mark.parametrize("kv_layout", ["NHD", "HND"])
-@pytest.mark.parametrize("Hq,Hkv,D", [(32, 8, 128), (4, 4, 64)])
+# (64, 8, 128) exercises the MFMA split route with the same-XCD in-kernel
+# merge (batch 5 x 8 kv_heads % 8 == 0) including its LSE epilogue
+@pytest.mark.parametrize("Hq,Hkv,D", [(32, 8, 128), (4, 4, 64), (64, 8, 128)])
 def test_batch_decode(page_size, kv_layout, Hq, Hkv, D):
     import flashinfer_amd as fi
 
