@@ -282,11 +282,16 @@ class BatchDecodeWithPagedKVCacheWrapper:
             self._last_page_len_d = _to_dev_f(
                 last_page_len, self._fixed_last_page_len_buf)
             # cross-WG split (mfma path): fill the 256-CU chip when
-            # batch x kv_heads alone cannot (bs=16 x 8 = 128 WGs = half idle)
+            # batch x kv_heads alone cannot. Target ONE WG per CU: the
+            # kernel is 8 waves at 256 VGPRs (2 waves/SIMD), so a CU hosts
+            # exactly one WG — more than 256 WGs means a second dispatch
+            # round. Measured at bs=16/kv=1024 GQA-8 (128 units): split=2
+            # (256 WGs, 1 round) 22.1 us; split=4 (512 WGs, 2 rounds)
+            # 27.7 us; split=8 (4 rounds) 43 us — profiles/README r02.
             self._mfma_split = 1
             if self._fused_mfma and not disable_split_kv:
                 units = batch * num_kv_heads
-                want = min(-(-512 // max(1, units)), max(1, max_len // 256), 8)
+                want = min(max(1, 256 // units), max(1, max_len // 256), 8)
                 if want > 1:
                     self._mfma_split = int(want)
                     alloc = WorkspaceAllocator(self._float_workspace_buffer)
