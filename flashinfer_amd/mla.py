@@ -54,10 +54,13 @@ class BatchMLAPagedAttentionWrapper:
         qo_lens = (qi[1:] - qi[:-1]).tolist()
         nnz = int(qi[-1])
 
-        # split-KV chunking: target enough workgroups to fill 256 CUs
+        # split-KV chunking: target ONE workgroup per CU (the kernel is 8
+        # waves at launch_bounds(512,1) — a CU hosts exactly one WG, so
+        # >256 items dispatch in extra rounds; the decode-route sweep
+        # measured 2 rounds ~25% slower — profiles/README r02 addendum 6)
         total_rows = nnz * num_heads
         row_tiles = sum(ceil_div(L * num_heads, 64) for L in qo_lens)
-        target_items = 512
+        target_items = 256
         chunks_per = max(1, target_items // max(1, row_tiles))
         max_kv = max(kv_lens) if kv_lens else 1
         chunk = max(page_size, 128, math.ceil(max_kv / chunks_per))
