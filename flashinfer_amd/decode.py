@@ -187,8 +187,17 @@ class BatchDecodeWithPagedKVCacheWrapper:
             and group in (1, 2, 4, 5, 6, 7, 8, 16, 32)
             and q_data_type in (torch.bfloat16, torch.float16)
             and kv_dt in (q_data_type, torch.float8_e4m3fn)
+            # bf16/f16 precedence vs the fused vector kernel (measured,
+            # profiles/README r02 addendum 6): GROUP >= 4 prefers MFMA up
+            # to ~384 (req, kv_head) units (20.9 vs 31.1 us at bs=16/
+            # kv=1024 G=4; crossover between 384 and 512 units where the
+            # 8-wave WGs exceed one dispatch round); GROUP 1-2 stay on the
+            # fused kernel. fp8 KV always prefers MFMA.
             and (max_len <= _MFMA_MAX_KV_F8 if kv_dt != q_data_type
-                 else (max_len <= _MFMA_MAX_KV and not fused_ok))
+                 else (max_len <= _MFMA_MAX_KV
+                       and (not fused_ok
+                            or (group >= 4
+                                and batch * num_kv_heads <= 384))))
             and fixed_split_size is None
             # use_tensor_cores=True with an fp8 KV cache means THIS kernel:
             # the prefill-based tc path needs kv_dt == q_dt, and the MFMA

@@ -49,10 +49,49 @@ def run(B, L, dt, inkernel, Hq=64, Hkv=8, force_split=None):
     return out
 
 
-o4 = run(16, 1024, torch.bfloat16, True)
-for sp in (2, 8):
-    o = run(16, 1024, torch.bfloat16, True, force_split=sp)
-    torch.testing.assert_close(o, o4, atol=3e-2, rtol=3e-2)
-    o = run(16, 1024, torch.bfloat16, False, force_split=sp)
-    torch.testing.assert_close(o, o4, atol=3e-2, rtol=3e-2)
-run(16, 1024, torch.bfloat16, True)
+import flashinfer_amd.decode as dec
+
+def run_route(B, L, Hq, Hkv, route):
+    sv = (dec._FUSED_MAX_KV, dec._MFMA_MAX_KV)
+    if route == "mfma":
+        dec._FUSED_MAX_KV = 0
+    torch.manual_seed(0)
+    D, page = 128, 16
+    npages = B * max(1, L // page)
+    kc = torch.randn(npages, page, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn_like(kc)
+    indptr = torch.arange(0, B + 1, dtype=torch.int32, device="cuda") * (L // page)
+    indices = torch.arange(npages, dtype=torch.int32, device="cuda")
+    lpl = torch.full((B,), page, dtype=torch.int32, device="cuda")
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device="cuda")
+    ws = torch.empty(128 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    try:
+        w.plan(indptr, indices, lpl, Hq, Hkv, D, page,
+               q_data_type=torch.bfloat16)
+    finally:
+        dec._FUSED_MAX_KV, dec._MFMA_MAX_KV = sv
+    if route == "mfma":
+        assert w._fused_mfma
+    for _ in range(30):
+        out = w.run(q, (kc, vc))
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(500):
+        out = w.run(q, (kc, vc))
+    torch.cuda.synchronize()
+    us = (time.perf_counter() - t0) / 500 * 1e6
+    rt = ("mfma" if w._fused_mfma else
+          ("fused" if w._fused else ("tc" if w._tc else "vec")))
+    print(f"bs={B} kv={L} G={Hq // Hkv}: route={rt:6s} {us:7.2f} us")
+
+
+for B, L in [(16, 1024), (32, 512), (64, 2048), (8, 256)]:
+    run_route(B, L, 32, 8, "auto")
+    run_route(B, L, 32, 8, "mfma")
+for B, L in [(16, 1024), (32, 512)]:
+    for Hq in (16, 8):
+        run_route(B, L, Hq, 8, "auto")
+        run_route(B, L, Hq, 8, "mfma")
+run_route(48, 1024, 32, 8, "auto")   # 384 units: fused vs
+run_route(48, 1024, 32, 8, "mfma")
