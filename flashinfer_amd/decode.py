@@ -232,14 +232,23 @@ class BatchDecodeWithPagedKVCacheWrapper:
                 and not disable_split_kv):
             from . import autotuner as _at
 
+            # candidates reflect STRUCTURAL capability (what each kernel
+            # family can run), not the heuristic's preference — autotune()
+            # exists to discover the crossovers the heuristic approximates
+            mfma_possible = (
+                head_dim in (64, 128)
+                and group in (1, 2, 4, 5, 6, 7, 8, 16, 32)
+                and q_data_type in (torch.bfloat16, torch.float16)
+                and kv_dt in (q_data_type, torch.float8_e4m3fn))
             routes = []
             if fused_ok:
                 routes.append("fused")
-            if mfma_ok:
+            if mfma_possible:
                 routes.append("mfma")
             if kv_dt == q_data_type:
                 routes.append("tc")
-            routes.append("split")
+            if vector_group_ok:
+                routes.append("split")
             key = (f"decode_route:({_at._bucket(max(1, batch))},"
                    f"{_at._bucket(max(1, max_len))},{group},{head_dim})")
             cached = _at._cache.get(key)
@@ -515,19 +524,25 @@ class BatchDecodeWithPagedKVCacheWrapper:
          qdt, kdt, sm) = plan_args
 
         def make(route):
-            saved = (_dec._FUSED_MAX_KV, _dec._MFMA_MAX_KV)
+            saved = (_dec._FUSED_MAX_KV, _dec._MFMA_MAX_KV,
+                     _dec._MFMA_MAX_KV_F8)
             utc = None
             try:
                 if route == "split":
                     _dec._FUSED_MAX_KV = _dec._MFMA_MAX_KV = 0
+                    _dec._MFMA_MAX_KV_F8 = 0
                     utc = False
                 elif route == "tc":
                     utc = True
                 elif route == "mfma":
                     _dec._FUSED_MAX_KV = 0
                     _dec._MFMA_MAX_KV = 1 << 30
-                else:  # fused
+                else:  # fused (also pin mfma off: the mfma gate yields to
+                    # fused only below 384 units, and forcing fused must
+                    # actually run the fused kernel)
                     _dec._FUSED_MAX_KV = 1 << 30
+                    _dec._MFMA_MAX_KV = 0
+                    _dec._MFMA_MAX_KV_F8 = 0
                 w = BatchDecodeWithPagedKVCacheWrapper(
                     self._float_workspace_buffer, self._kv_layout,
                     use_tensor_cores=utc)
@@ -542,7 +557,8 @@ class BatchDecodeWithPagedKVCacheWrapper:
                 finally:
                     _at._tuning_enabled = prev
             finally:
-                _dec._FUSED_MAX_KV, _dec._MFMA_MAX_KV = saved
+                (_dec._FUSED_MAX_KV, _dec._MFMA_MAX_KV,
+                 _dec._MFMA_MAX_KV_F8) = saved
             return w
 
         timings = []
