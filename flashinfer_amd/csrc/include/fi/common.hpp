@@ -21,6 +21,32 @@ constexpr int kLdsBytes = 160 * 1024;
 using bf16 = __hip_bfloat16;
 using fp16 = __half;
 using fp8_e4m3 = __hip_fp8_e4m3;     // OCP e4m3fn (gfx950), NOT fnuz
+
+// ---- CDNA4 v_permlane32_swap_b32 (guide T12): one VALU op exchanges
+// register halves across lane<32 / lane>=32 — replaces a ds_bpermute
+// (__shfl_xor(x, 32)) round trip through LDS. Probed semantics
+// (scripts/probe/plprobe.hip): returns {concat(a.lo32, b.lo32),
+// concat(a.hi32, b.hi32)}. Must run in uniform control flow. ----
+__device__ __forceinline__ void permlane32_pair(uint32_t a, uint32_t b,
+                                                uint32_t& lo, uint32_t& hi) {
+  auto r = __builtin_amdgcn_permlane32_swap((int)a, (int)b, false, false);
+  lo = (uint32_t)r[0];
+  hi = (uint32_t)r[1];
+}
+
+// fmaxf(x, __shfl_xor(x, 32, 64)) without LDS
+__device__ __forceinline__ float xhalf_max(float x) {
+  auto r = __builtin_amdgcn_permlane32_swap(__float_as_int(x),
+                                            __float_as_int(x), false, false);
+  return fmaxf(__int_as_float(r[0]), __int_as_float(r[1]));
+}
+
+// x + __shfl_xor(x, 32, 64) without LDS
+__device__ __forceinline__ float xhalf_sum(float x) {
+  auto r = __builtin_amdgcn_permlane32_swap(__float_as_int(x),
+                                            __float_as_int(x), false, false);
+  return __int_as_float(r[0]) + __int_as_float(r[1]);
+}
 using fp8_e5m2 = __hip_fp8_e5m2;
 
 // Ext-vector typedefs (register-resident fragments and vector loads).

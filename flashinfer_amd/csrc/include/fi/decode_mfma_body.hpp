@@ -240,7 +240,7 @@ __device__ __forceinline__ void decode_mfma_item_body(const DecodeParams& p, int
     float tmax = pr[0];
 #pragma unroll
     for (int r = 1; r < 16; ++r) tmax = fmaxf(tmax, pr[r]);
-    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    tmax = xhalf_max(tmax);
     bool defer = m_run != -INFINITY && __all(tmax - m_run <= 8.f);
     float m_new = defer ? m_run : fmaxf(m_run, tmax);
     float f, psum = 0.f;
@@ -274,19 +274,14 @@ __device__ __forceinline__ void decode_mfma_item_body(const DecodeParams& p, int
           : "=v"(W[j])
           : "v"(pr[2 * j]), "v"(pr[2 * j + 1]));
     }
-    uint32_t X[8];
-#pragma unroll
-    for (int j = 0; j < 8; ++j) X[j] = (uint32_t)__shfl_xor((int)W[j], 32, 64);
-    const bool hiH = (lane >> 5) != 0;
+    // half-exchange via v_permlane32_swap (guide T12): 4 VALU ops replace
+    // 8 ds_bpermute + 8 selects (b0[0]=concat(W0.lo,W2.lo),
+    // b0[2]=concat(W0.hi,W2.hi), ...)
     uint32_t b0[4], b1[4];
-    b0[0] = hiH ? X[2] : W[0];
-    b0[1] = hiH ? X[3] : W[1];
-    b0[2] = hiH ? W[2] : X[0];
-    b0[3] = hiH ? W[3] : X[1];
-    b1[0] = hiH ? X[6] : W[4];
-    b1[1] = hiH ? X[7] : W[5];
-    b1[2] = hiH ? W[6] : X[4];
-    b1[3] = hiH ? W[7] : X[5];
+    permlane32_pair(W[0], W[2], b0[0], b0[2]);
+    permlane32_pair(W[1], W[3], b0[1], b0[3]);
+    permlane32_pair(W[4], W[6], b1[0], b1[2]);
+    permlane32_pair(W[5], W[7], b1[1], b1[3]);
 
     // ---- O^T += V^T * P^T via HW transpose reads ----
     {
@@ -337,7 +332,7 @@ __device__ __forceinline__ void decode_mfma_item_body(const DecodeParams& p, int
   // ---- per-wave state -> merge rows (aliased into this wave's stage LDS;
   // every wave has finished reading its stage before writing here, and the
   // barrier below orders the cross-wave reads) ----
-  float d_full = d_run + __shfl_xor(d_run, 32, 64);
+  float d_full = xhalf_sum(d_run);
   if (lq < QROWS) {
     float* orow = merge_row(wave, lq);
 #pragma unroll

@@ -428,7 +428,7 @@ __device__ __forceinline__ void prefill_tile_body(const PrefillParams& p, int re
 #pragma unroll
       for (int r = 1; r < 8; ++r) mx2 = __builtin_elementwise_max(mx2, pr2[r]);
       float tmax = fmaxf(mx2.x, mx2.y);
-      tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+      tmax = xhalf_max(tmax);
       // defer-max (guide T13): skip the O-wide rescale while the running max
       // grows by < 8 (base-2) — p values stay bounded by 2^8, f32 accum is
       // fine. __all keeps the wave branch-uniform.
@@ -470,19 +470,15 @@ __device__ __forceinline__ void prefill_tile_body(const PrefillParams& p, int re
             : "v"(pr[2 * j]), "v"(pr[2 * j + 1]));
       }
       // NOTE: shfl must run on ALL lanes (uniform control flow) — select after.
-      uint32_t X[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) X[j] = (uint32_t)__shfl_xor((int)W[j], 32, 64);
-      const bool hiH = (lane >> 5) != 0;
+      // half-exchange via v_permlane32_swap (guide T12): each call yields
+      // BOTH fragments of a (Wlo, Whi) pair — 4 VALU ops replace the
+      // former 8 ds_bpermute + 8 selects (verified equal to the select
+      // table: b0[0]=concat(W0.lo,W2.lo), b0[2]=concat(W0.hi,W2.hi), ...)
       uint32_t b0[4], b1[4];
-      b0[0] = hiH ? X[2] : W[0];
-      b0[1] = hiH ? X[3] : W[1];
-      b0[2] = hiH ? W[2] : X[0];
-      b0[3] = hiH ? W[3] : X[1];
-      b1[0] = hiH ? X[6] : W[4];
-      b1[1] = hiH ? X[7] : W[5];
-      b1[2] = hiH ? W[6] : X[4];
-      b1[3] = hiH ? W[7] : X[5];
+      permlane32_pair(W[0], W[2], b0[0], b0[2]);
+      permlane32_pair(W[1], W[3], b0[1], b0[3]);
+      permlane32_pair(W[4], W[6], b1[0], b1[2]);
+      permlane32_pair(W[5], W[7], b1[1], b1[3]);
 
       // ---- O^T += V^T * P^T: V^T fragments via ds_read_b64_tr_b16 (HW
       // 4x16 transpose read — guide T10); 4 reads + one wait per d-tile ----
@@ -525,7 +521,7 @@ __device__ __forceinline__ void prefill_tile_body(const PrefillParams& p, int re
   prof_event(p.prof_buf, 1, ProfType::kEnd);
 
   // ---- epilogue: normalize and write O (transpose from O^T frags) ----
-  float d_full = d_run + __shfl_xor(d_run, 32, 64);
+  float d_full = xhalf_sum(d_run);
   float inv_d = d_full > 0.f ? 1.f / d_full : 0.f;
   if (row_valid) {
     const int qh = (int)(kv_head * group + my_g_u);
