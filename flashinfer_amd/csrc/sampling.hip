@@ -314,13 +314,21 @@ __device__ __forceinline__ float u2f(uint32_t u) {
 // top-k results stay bitwise deterministic); per-bin float sums (top-p bin
 // selection only) use float atomics — the final renorm sum is recomputed
 // in the fixed block_sum order.
+// top-p bin masses use FIXED-POINT u64 LDS atomics (probs in [0,1] scaled
+// by 2^40): integer adds are order-free, so the selected threshold — and
+// therefore the whole output — is bitwise deterministic run to run, which
+// float atomicAdd is not (quantization error <= vocab * 2^-40 ~ 1e-7, the
+// same class as the float rounding it replaces).
+constexpr float kPScale = 1099511627776.f;  // 2^40
+
 template <int WHICH>
 __global__ void renorm_kernel(SamplingParams p) {
   __shared__ float smem[SW + 1];
   __shared__ uint32_t s_hcnt[2048];
-  __shared__ float s_hsum[2048];
+  __shared__ uint64_t s_hsum[2048];
   __shared__ uint32_t s_sel;
-  __shared__ float s_carry[2];
+  __shared__ float s_carry;
+  __shared__ uint64_t s_carry64;
   for (int b = blockIdx.x; b < p.rows; b += gridDim.x) {
     const float* in = p.probs + (uint64_t)b * p.stride_row;
     float* out = p.out_probs + (uint64_t)b * p.vocab;
@@ -332,8 +340,10 @@ __global__ void renorm_kernel(SamplingParams p) {
       __syncthreads();
       continue;
     }
+    const uint64_t ppq = (uint64_t)((double)pp * (double)kPScale);
     uint32_t prefix = 0;      // matched high bits of tau's monotonic uint
-    float cum_cnt = 0.f, cum_sum = 0.f;  // mass strictly above prefix region
+    float cum_cnt = 0.f;      // count strictly above prefix region
+    uint64_t cum_q = 0;       // fixed-point mass strictly above
     constexpr int kShift[3] = {21, 10, 0};
     constexpr int kBits[3] = {11, 11, 10};
     for (int lvl = 0; lvl < 3; ++lvl) {
@@ -341,7 +351,7 @@ __global__ void renorm_kernel(SamplingParams p) {
       const int hi_shift = kShift[lvl] + kBits[lvl];  // bits already matched
       for (int i = threadIdx.x; i < nb; i += SB) {
         s_hcnt[i] = 0;
-        if constexpr (WHICH == 1) s_hsum[i] = 0.f;
+        if constexpr (WHICH == 1) s_hsum[i] = 0;
       }
       __syncthreads();
       for (int i = threadIdx.x; i < p.vocab; i += SB) {
@@ -350,45 +360,51 @@ __global__ void renorm_kernel(SamplingParams p) {
         if (lvl == 0 || (u >> hi_shift) == prefix) {
           int bin = (u >> kShift[lvl]) & (nb - 1);
           // top-p selects on mass only; skip the count atomic there
-          if constexpr (WHICH == 1) atomicAdd(&s_hsum[bin], v);
-          else atomicAdd(&s_hcnt[bin], 1u);
+          if constexpr (WHICH == 1) {
+            atomicAdd((unsigned long long*)&s_hsum[bin],
+                      (unsigned long long)(fmaxf(v, 0.f) * kPScale));
+          } else {
+            atomicAdd(&s_hcnt[bin], 1u);
+          }
         }
       }
       __syncthreads();
       if (threadIdx.x == 0) {
         int chosen = -1;
-        float cc = cum_cnt, cs = cum_sum;
+        float cc = cum_cnt;
+        uint64_t cq = cum_q;
         for (int bin = nb - 1; bin >= 0; --bin) {
-          bool present = (WHICH == 1) ? (s_hsum[bin] != 0.f) : (s_hcnt[bin] > 0);
+          bool present = (WHICH == 1) ? (s_hsum[bin] != 0) : (s_hcnt[bin] > 0);
           if (present) {
-            bool crossed = (WHICH == 1) ? (cs + s_hsum[bin] >= pp)
+            bool crossed = (WHICH == 1) ? (cq + s_hsum[bin] >= ppq)
                                         : (cc + (float)s_hcnt[bin] >= (float)k);
             if (crossed) { chosen = bin; break; }
           }
-          if constexpr (WHICH == 1) cs += s_hsum[bin];
+          if constexpr (WHICH == 1) cq += s_hsum[bin];
           else cc += (float)s_hcnt[bin];
         }
         if (chosen < 0) {
           // never crossed (rounding tail / pp > total): lowest present bin
-          cc = cum_cnt; cs = cum_sum;
+          cc = cum_cnt;
+          cq = cum_q;
           for (int bin = nb - 1; bin >= 0; --bin) {
-            bool present = (WHICH == 1) ? (s_hsum[bin] != 0.f) : (s_hcnt[bin] > 0);
+            bool present = (WHICH == 1) ? (s_hsum[bin] != 0) : (s_hcnt[bin] > 0);
             if (present) chosen = bin;
           }
           if (chosen < 0) chosen = 0;
           for (int bin = nb - 1; bin > chosen; --bin) {
-            if constexpr (WHICH == 1) cs += s_hsum[bin];
+            if constexpr (WHICH == 1) cq += s_hsum[bin];
             else cc += (float)s_hcnt[bin];
           }
         }
         s_sel = (uint32_t)chosen;
-        s_carry[0] = cc;
-        s_carry[1] = cs;
+        s_carry = cc;
+        s_carry64 = cq;
       }
       __syncthreads();
       prefix = (prefix << kBits[lvl]) | s_sel;
-      cum_cnt = s_carry[0];
-      cum_sum = s_carry[1];
+      cum_cnt = s_carry;
+      cum_q = s_carry64;
       __syncthreads();
     }
     float tau = u2f(prefix);  // exact k-th / top-p boundary value
