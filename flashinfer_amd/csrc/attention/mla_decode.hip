@@ -194,7 +194,7 @@ __global__ __launch_bounds__(NTH, 1) void mla_decode_kernel(MlaParams p) {
     float tmax = pr[0];
 #pragma unroll
     for (int r = 1; r < 16; ++r) tmax = fmaxf(tmax, pr[r]);
-    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    tmax = xhalf_max(tmax);
     float m_new = fmaxf(m_run, tmax);
     float f, psum = 0.f;
     if (m_new == -INFINITY) {
@@ -225,19 +225,13 @@ __global__ __launch_bounds__(NTH, 1) void mla_decode_kernel(MlaParams p) {
           : "=v"(W[j])
           : "v"(pr[2 * j]), "v"(pr[2 * j + 1]));
     }
-    uint32_t X[8];
-#pragma unroll
-    for (int j = 0; j < 8; ++j) X[j] = (uint32_t)__shfl_xor((int)W[j], 32, 64);
-    const bool hiH = (lane >> 5) != 0;
+    // half-exchange via v_permlane32_swap (guide T12): 4 VALU ops replace
+    // 8 ds_bpermute + 8 selects
     uint32_t b0[4], b1[4];
-    b0[0] = hiH ? X[2] : W[0];
-    b0[1] = hiH ? X[3] : W[1];
-    b0[2] = hiH ? W[2] : X[0];
-    b0[3] = hiH ? W[3] : X[1];
-    b1[0] = hiH ? X[6] : W[4];
-    b1[1] = hiH ? X[7] : W[5];
-    b1[2] = hiH ? W[6] : X[4];
-    b1[3] = hiH ? W[7] : X[5];
+    permlane32_pair(W[0], W[2], b0[0], b0[2]);
+    permlane32_pair(W[1], W[3], b0[1], b0[3]);
+    permlane32_pair(W[4], W[6], b1[0], b1[2]);
+    permlane32_pair(W[5], W[7], b1[1], b1[3]);
 
     // ---- O^T += ckv^T * P^T over my 128-d output slice (tr reads) ----
     {
@@ -272,7 +266,7 @@ __global__ __launch_bounds__(NTH, 1) void mla_decode_kernel(MlaParams p) {
   }
 
   // ---- epilogue: normalized partial + base-2 lse into tmp slot ----
-  float d_full = d_run + __shfl_xor(d_run, 32, 64);
+  float d_full = xhalf_sum(d_run);
   float inv_d = d_full > 0.f ? 1.f / d_full : 0.f;
   if (row_valid) {
     int64_t pos = (int64_t)(qo_begin + qpos_u) * H + head_u;

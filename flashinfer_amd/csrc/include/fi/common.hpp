@@ -106,13 +106,21 @@ __host__ __device__ __forceinline__ constexpr int64_t ceil_div64(int64_t a, int6
 template <int width>
 __device__ __forceinline__ float wave_reduce_sum(float x) {
 #pragma unroll
-  for (int off = width / 2; off > 0; off >>= 1) x += __shfl_xor(x, off, 64);
+  for (int off = width / 2; off > 0; off >>= 1) {
+    // the lane^32 exchange has a pure-VALU instruction on CDNA4; smaller
+    // offsets lower to ds_swizzle and stay as-is
+    if (off == 32) x = xhalf_sum(x);
+    else x += __shfl_xor(x, off, 64);
+  }
   return x;
 }
 template <int width>
 __device__ __forceinline__ float wave_reduce_max(float x) {
 #pragma unroll
-  for (int off = width / 2; off > 0; off >>= 1) x = fmaxf(x, __shfl_xor(x, off, 64));
+  for (int off = width / 2; off > 0; off >>= 1) {
+    if (off == 32) x = xhalf_max(x);
+    else x = fmaxf(x, __shfl_xor(x, off, 64));
+  }
   return x;
 }
 
