@@ -178,6 +178,42 @@ class _BatchPrefillBase:
             self._tmp_s = alloc.alloc(
                 ns * num_qo_heads * 4, torch.float32, (ns, num_qo_heads))
         else:
+            # causal short-diagonal retiling: tiles whose causal KV reach is
+            # < 512 tokens are dominated by the fixed CTAQ-256 Q-stage /
+            # epilogue cost (measured: causal 374 TF vs 514 non-causal at
+            # bs=16 s=1024 with only ~6% diagonal MFMA waste — profiles/
+            # README r02 addendum 7). They are re-emitted as CTAQ-128
+            # tiles and run as a second launch, halving their fixed cost
+            # and diagonal waste.
+            self._tile_req2 = None
+            if causal and cta_q == 256 and kv_lens is not None:
+                big_r, big_s, sm_r, sm_s = [], [], [], []
+                for b, st in zip(tile_req, tile_qstart):
+                    pk = qo_lens[b] * group
+                    kvh = kv_lens[b] - qo_lens[b] + (min(st + cta_q, pk)
+                                                     + group - 1) // group
+                    if kvh < 512:
+                        sm_r.append(b)
+                        sm_s.append(st)
+                        if st + 128 < pk:
+                            sm_r.append(b)
+                            sm_s.append(st + 128)
+                    else:
+                        big_r.append(b)
+                        big_s.append(st)
+                # only worth it when short tiles are a MINORITY: when most
+                # tiles are short (s ~ 1024 at group 4, or group-1 ragged)
+                # doubling their count costs more in tail rounds than the
+                # halved fixed cost saves (measured: bs=1/s=8192 439->493
+                # TF, bs=4/s=4096 551 TF; bs=16/s=1024 374->368 and
+                # ragged-192 442->418 without this gate)
+                if sm_r and big_r and len(sm_r) <= len(big_r) // 2:
+                    tile_req, tile_qstart = big_r, big_s
+                    n_tiles = len(tile_req)
+                    meta2 = torch.tensor(sm_r + sm_s, dtype=torch.int32).to(
+                        self.device, non_blocking=non_blocking)
+                    self._tile_req2 = meta2[: len(sm_r)]
+                    self._tile_qstart2 = meta2[len(sm_r):]
             meta = torch.tensor(tile_req + tile_qstart, dtype=torch.int32).to(
                 self.device, non_blocking=non_blocking
             )
@@ -232,6 +268,19 @@ class _BatchPrefillBase:
             self._tmp_s if split else None,
             getattr(self, "_custom_fn", 0),
         )
+        if getattr(self, "_tile_req2", None) is not None and not split:
+            # second launch: the causal short-diagonal tiles at CTAQ 128
+            get_ext().batch_prefill_run(
+                q, k_cache, v_cache, self._qo_indptr_d, kv_indices, kv_indptr,
+                kv_last_page_len, layout_code(self._kv_layout),
+                self._tile_req2, self._tile_qstart2, out,
+                lse if return_lse else None, sm_scale,
+                pi["logits_soft_cap"], pi["window_left"], pi["causal"], paged,
+                128, getattr(self, "_mask_data", None),
+                getattr(self, "_mask_byte_indptr", None),
+                pi.get("alibi", False), k_descale, v_descale, profiler_buffer,
+                None, 0, None, None, None, getattr(self, "_custom_fn", 0),
+            )
         if split:
             # LSE merge of the per-chunk partials (cascade merge kernel)
             get_ext().merge_states(
