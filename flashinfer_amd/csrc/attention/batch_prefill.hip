@@ -31,6 +31,13 @@ __global__ __launch_bounds__(CTAQ * 2, ((HDQK >= 192 || HDVO >= 256) ? 1 : 2)) v
   prefill_tile_body<T, TKV, HDQK, HDVO, CTAQ, PAGED, CAUSAL, MASK>(
       p, p.tile_req[tile], p.tile_qstart[tile], blockIdx.y,
       p.tile_kv_chunk ? (int)p.tile_kv_chunk[tile] : -1, smem);
+  // chained second short tile (same WG, fresh state; the barrier orders the
+  // first item's epilogue LDS reads against the second's Q staging)
+  if (p.tile_req_b && p.tile_req_b[tile] >= 0) {
+    __syncthreads();
+    prefill_tile_body<T, TKV, HDQK, HDVO, CTAQ, PAGED, CAUSAL, MASK>(
+        p, p.tile_req_b[tile], p.tile_qstart_b[tile], blockIdx.y, -1, smem);
+  }
 }
 
 template <typename T, typename TKV>

@@ -599,7 +599,9 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                        c10::optional<at::Tensor> req_slot_base = c10::nullopt,
                        c10::optional<at::Tensor> tmp_v = c10::nullopt,
                        c10::optional<at::Tensor> tmp_s = c10::nullopt,
-                       int64_t custom_fn = 0) {
+                       int64_t custom_fn = 0,
+                       c10::optional<at::Tensor> tile_req_b = c10::nullopt,
+                       c10::optional<at::Tensor> tile_qstart_b = c10::nullopt) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 3, "q must be [nnz, Hq, D]");
   TORCH_CHECK(q.stride(2) == 1 && out.stride(2) == 1);
   fi_ext::PrefillParams p{};
@@ -640,6 +642,11 @@ void batch_prefill_run(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   p.page_size = fi::uint_fastdiv((uint32_t)page_size);
   p.tile_req = tile_req.data_ptr<int32_t>();
   p.tile_qstart = tile_qstart.data_ptr<int32_t>();
+  p.tile_req_b = tile_req_b.has_value() ? tile_req_b->data_ptr<int32_t>()
+                                        : nullptr;
+  p.tile_qstart_b = tile_qstart_b.has_value()
+                        ? tile_qstart_b->data_ptr<int32_t>()
+                        : nullptr;
   p.n_tiles = tile_req.size(0);
   p.num_qo_heads = q.size(1);
   p.num_kv_heads = num_kv_heads;

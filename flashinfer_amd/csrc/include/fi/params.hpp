@@ -71,6 +71,10 @@ struct PrefillParams {
   int64_t kv_stride_page, kv_stride_n, kv_stride_h;
   const int32_t* tile_req;     // [n_tiles]
   const int32_t* tile_qstart;  // [n_tiles] packed-row offset in request
+  // optional second item per WG (causal short-tile chaining; -1 = none):
+  // halves the dispatch-round count when most tiles are short-diagonal
+  const int32_t* tile_req_b;
+  const int32_t* tile_qstart_b;
   int n_tiles;
   int num_qo_heads, num_kv_heads, head_dim;
   uint_fastdiv group;  // Hq / Hkv

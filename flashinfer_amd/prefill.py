@@ -186,21 +186,26 @@ class _BatchPrefillBase:
             # tiles and run as a second launch, halving their fixed cost
             # and diagonal waste.
             self._tile_req2 = None
+            self._tile_chain_req = None
             if causal and cta_q == 256 and kv_lens is not None:
-                big_r, big_s, sm_r, sm_s = [], [], [], []
+                big_r, big_s, big_kv = [], [], []
+                sm_r, sm_s = [], []
+                shorts = []   # ORIGINAL 256-row short tiles (for chaining)
                 for b, st in zip(tile_req, tile_qstart):
                     pk = qo_lens[b] * group
                     kvh = kv_lens[b] - qo_lens[b] + (min(st + cta_q, pk)
                                                      + group - 1) // group
                     if kvh < 512:
-                        sm_r.append(b)
-                        sm_s.append(st)
+                        shorts.append((b, st, kvh))
+                        sm_r.append(b)          # 128-row retiling for the
+                        sm_s.append(st)         # minority second-launch path
                         if st + 128 < pk:
                             sm_r.append(b)
                             sm_s.append(st + 128)
                     else:
                         big_r.append(b)
                         big_s.append(st)
+                        big_kv.append(kvh)
                 # only worth it when short tiles are a MINORITY: when most
                 # tiles are short (s ~ 1024 at group 4, or group-1 ragged)
                 # doubling their count costs more in tail rounds than the
@@ -214,6 +219,40 @@ class _BatchPrefillBase:
                         self.device, non_blocking=non_blocking)
                     self._tile_req2 = meta2[: len(sm_r)]
                     self._tile_qstart2 = meta2[len(sm_r):]
+                elif (len(shorts) >= 4 and not getattr(self, "_custom_fn", 0)):
+                    # shorts-majority (e.g. bs=16 s=1024): CHAIN pairs of
+                    # short 256-row tiles into one WG instead — halves the
+                    # WG count for the short population (fewer dispatch
+                    # rounds) with no extra fixed cost. Pair smallest-with-
+                    # largest so chained costs balance.
+                    order = sorted(range(len(shorts)), key=lambda i: shorts[i][2])
+                    pr_r, pr_s, ch_r, ch_s, costs = [], [], [], [], []
+                    lo, hi = 0, len(order) - 1
+                    while lo < hi:
+                        a, z = order[hi], order[lo]   # big + small
+                        pr_r.append(shorts[a][0]); pr_s.append(shorts[a][1])
+                        ch_r.append(shorts[z][0]); ch_s.append(shorts[z][1])
+                        costs.append(shorts[a][2] + shorts[z][2])
+                        lo += 1; hi -= 1
+                    if lo == hi:
+                        a = order[lo]
+                        pr_r.append(shorts[a][0]); pr_s.append(shorts[a][1])
+                        ch_r.append(-1); ch_s.append(-1)
+                        costs.append(shorts[a][2])
+                    items = sorted(
+                        list(zip(big_r, big_s, [-1] * len(big_r),
+                                 [-1] * len(big_r), big_kv))
+                        + list(zip(pr_r, pr_s, ch_r, ch_s, costs)),
+                        key=lambda t: -t[4])
+                    tile_req = [t[0] for t in items]
+                    tile_qstart = [t[1] for t in items]
+                    n_tiles = len(tile_req)
+                    metac = torch.tensor(
+                        [t[2] for t in items] + [t[3] for t in items],
+                        dtype=torch.int32).to(self.device,
+                                              non_blocking=non_blocking)
+                    self._tile_chain_req = metac[:n_tiles]
+                    self._tile_chain_qstart = metac[n_tiles:]
             meta = torch.tensor(tile_req + tile_qstart, dtype=torch.int32).to(
                 self.device, non_blocking=non_blocking
             )
@@ -267,6 +306,8 @@ class _BatchPrefillBase:
             self._tmp_v if split else None,
             self._tmp_s if split else None,
             getattr(self, "_custom_fn", 0),
+            getattr(self, "_tile_chain_req", None) if not split else None,
+            getattr(self, "_tile_chain_qstart", None) if not split else None,
         )
         if getattr(self, "_tile_req2", None) is not None and not split:
             # second launch: the causal short-diagonal tiles at CTAQ 128
@@ -280,6 +321,7 @@ class _BatchPrefillBase:
                 getattr(self, "_mask_byte_indptr", None),
                 pi.get("alibi", False), k_descale, v_descale, profiler_buffer,
                 None, 0, None, None, None, getattr(self, "_custom_fn", 0),
+                None, None,
             )
         if split:
             # LSE merge of the per-chunk partials (cascade merge kernel)
@@ -425,7 +467,7 @@ def single_prefill_with_kv_cache(
         sm_scale if sm_scale is not None else default_sm_scale(D),
         float(logits_soft_cap or 0.0), window_left, causal, False, cta_q,
         mask_data, mask_indptr, pos_encoding_mode == "ALIBI", 1.0, 1.0, None,
-        None, 0, None, None, None, 0,
+        None, 0, None, None, None, 0, None, None,
     )
     return (out, lse) if return_lse else out
 
