@@ -74,9 +74,25 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
   // 128-aligned segments; binary-search the owning segment. Legacy grid
   // (num_segments == 0): blockIdx.z is the segment, y the tile within —
   // kept for unaligned segment GEMM callers (LoRA / plain groupwise).
+  // XCD-aware block swizzle (guide T1): the dispatcher round-robins
+  // linear block ids across the 8 XCD L2s, so consecutive ids — which
+  // share an A panel (same m-tile, neighboring n-tiles) — would each
+  // refetch A from HBM. Remapping gives each XCD a CONTIGUOUS run of the
+  // grid, so the shared A panel stays in that XCD's L2. Bijective only
+  // when nwg % 8 == 0 (guide errata #11) — identity otherwise.
+  int bx = blockIdx.x, by = blockIdx.y;
+  if (gridDim.z == 1) {
+    const int nwg = gridDim.x * gridDim.y;
+    if ((nwg & 7) == 0) {
+      int bid = bx + gridDim.x * by;
+      int sw = (bid & 7) * (nwg >> 3) + (bid >> 3);
+      bx = sw % gridDim.x;
+      by = sw / gridDim.x;
+    }
+  }
   int seg, m0, m_end;
   if (flat_segments > 0) {
-    m0 = blockIdx.y * BM;
+    m0 = by * BM;
     int lo = 0, hi = flat_segments - 1;
     while (lo < hi) {
       int mid = (lo + hi + 1) >> 1;
@@ -87,11 +103,11 @@ __global__ __launch_bounds__(NTH, 2) void gemm_fp8_kernel(
     m_end = m_indptr[seg + 1];
   } else {
     seg = blockIdx.z;
-    m0 = m_indptr[seg] + blockIdx.y * BM;
+    m0 = m_indptr[seg] + by * BM;
     m_end = m_indptr[seg + 1];
   }
   if (m0 >= m_end) return;
-  const int bn0 = blockIdx.x * BN;
+  const int bn0 = bx * BN;
   const int widx = w_indices ? w_indices[seg] : seg;
   const uint8_t* Wb = W + (int64_t)widx * ldw_seg;
 
