@@ -325,8 +325,21 @@ def bench_moe_mx(T=4096, H=4096, inter=14336, E=8, k=2):
     print(f"fused_moe MX-fp8 mixtral T={T}: {t*1e3:.2f} ms  {fl/t/1e12:.0f} TFLOPS  {T/t/1e6:.3f} M tok/s")
 
 
+def _clock_warm(seconds=0.7):
+    """DVFS reaches steady clocks only after ~0.5-1 s of sustained load;
+    without this the first benches in a sweep read ~10% low."""
+    import time as _t
+
+    a = torch.randn(4096, 4096, dtype=torch.bfloat16, device="cuda")
+    t0 = _t.perf_counter()
+    while _t.perf_counter() - t0 < seconds:
+        a = a @ a * 1e-3
+    torch.cuda.synchronize()
+
+
 if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "all"
+    _clock_warm()
     if which in ("all", "prefill"):
         bench_prefill()
         bench_prefill(bs=1, s=8192)
