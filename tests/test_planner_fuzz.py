@@ -93,3 +93,28 @@ def test_causal_planner_paths_reachable():
     w = plan([8192] * 8, [8192] * 8, 32, 8)
     if not getattr(w, "_split", False):
         assert getattr(w, "_tile_req2", None) is not None
+
+
+@pytest.mark.parametrize("seed", list(range(8)))
+def test_decode_chunk_planner_coverage(seed):
+    """Split-vector decode planning: work items cover each request's chunks
+    exactly, and merge_indptr allots n_chunks slots per request."""
+    from flashinfer_amd.decode import _plan_chunks
+
+    rng = random.Random(100 + seed)
+    n = rng.randint(1, 40)
+    kv_lens = [rng.choice([1, 15, 16, 17, 511, 4096, 32768, 100000])
+               for _ in range(n)]
+    page = rng.choice([1, 16, 32])
+    chunk, work_req, work_chunk, merge_indptr = _plan_chunks(
+        kv_lens, num_kv_heads=8, page_size=page)
+    assert chunk % page == 0 and chunk > 0
+    import math
+    per_req = {}
+    for b, c in zip(work_req, work_chunk):
+        per_req.setdefault(b, []).append(c)
+    for b, L in enumerate(kv_lens):
+        nc = max(1, math.ceil(L / chunk))
+        assert sorted(per_req[b]) == list(range(nc)), (b, L, chunk)
+        assert merge_indptr[b + 1] - merge_indptr[b] == nc
+    assert merge_indptr[0] == 0 and len(merge_indptr) == n + 1
