@@ -12,35 +12,71 @@ __device__ __forceinline__ uint8_t to_fp8(float x) {
   return __hip_cvt_float_to_fp8(x, __HIP_SATFINITE, __HIP_E4M3);
 }
 
-// One wave per (row, group of 128): amax -> scale -> quantize.
+// Per-128-group quantize: a wave covers FOUR groups (512 elems; lane holds
+// 8 contiguous values via a 16B load, group amax is a segmented 16-lane
+// shfl tree, 8 fp8 bytes store as one u64). The 2-elem-per-lane version
+// measured 2.5 TB/s in the MoE loop (profiles/README r02 addendum 10).
+// Rows with K % 512 fall back to one-wave-per-group for the tail groups.
 // scale layout: MN-major [K/128, M] (transposed=true) or [M, K/128].
 template <typename T, bool TRANS_SCALE>
 __global__ void per_group_quant_kernel(const T* __restrict__ x, uint8_t* __restrict__ q,
                                        float* __restrict__ scale, int64_t rows, int K,
                                        int64_t stride_row, float eps) {
-  int groups = K / 128;
-  int64_t total = rows * groups;
+  const int groups = K / 128;
+  const int c4 = K / 512;                 // full 4-group wave clusters
+  const int tail = groups - c4 * 4;       // leftover 128-groups
+  const int64_t total = rows * (c4 + tail);
   int64_t wave = (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) >> 6;
   int lane = threadIdx.x & 63;
   if (wave >= total) return;
-  int g = (int)(wave % groups);
-  int64_t row = wave / groups;
-  const T* src = x + row * stride_row + g * 128;
-  float v[2];
-  vec_t<T, 2> lv;
-  lv.load(src + lane * 2);
-  v[0] = lv.get(0);
-  v[1] = lv.get(1);
-  float amax = fmaxf(fabsf(v[0]), fabsf(v[1]));
-  amax = wave_reduce_max<kWaveSize>(amax);
-  float s = fmaxf(amax, eps) / kF8Max;
-  float inv = 1.f / s;
-  uint8_t* dst = q + row * K + g * 128 + lane * 2;
-  dst[0] = to_fp8(v[0] * inv);
-  dst[1] = to_fp8(v[1] * inv);
-  if (lane == 0) {
-    if constexpr (TRANS_SCALE) scale[(int64_t)g * rows + row] = s;
-    else scale[row * groups + g] = s;
+  int item = (int)(wave % (c4 + tail));
+  int64_t row = wave / (c4 + tail);
+  if (item < c4) {
+    const T* src = x + row * stride_row + item * 512;
+    vec_t<T, 8> lv;
+    lv.load(src + lane * 8);
+    float v[8];
+    float amax = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      v[j] = lv.get(j);
+      amax = fmaxf(amax, fabsf(v[j]));
+    }
+#pragma unroll
+    for (int off = 8; off > 0; off >>= 1)
+      amax = fmaxf(amax, __shfl_xor(amax, off, 16));
+    float s = fmaxf(amax, eps) / kF8Max;
+    float inv = 1.f / s;
+    uint64_t packed = 0;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      packed |= (uint64_t)__builtin_bit_cast(uint8_t, to_fp8(v[j] * inv))
+                << (8 * j);
+    *reinterpret_cast<uint64_t*>(q + row * K + item * 512 + lane * 8) = packed;
+    if ((lane & 15) == 0) {
+      int g = item * 4 + (lane >> 4);
+      if constexpr (TRANS_SCALE) scale[(int64_t)g * rows + row] = s;
+      else scale[row * groups + g] = s;
+    }
+  } else {
+    int g = c4 * 4 + (item - c4);
+    const T* src = x + row * stride_row + g * 128;
+    float v[2];
+    vec_t<T, 2> lv;
+    lv.load(src + lane * 2);
+    v[0] = lv.get(0);
+    v[1] = lv.get(1);
+    float amax = fmaxf(fabsf(v[0]), fabsf(v[1]));
+    amax = wave_reduce_max<kWaveSize>(amax);
+    float s = fmaxf(amax, eps) / kF8Max;
+    float inv = 1.f / s;
+    uint8_t* dst = q + row * K + g * 128 + lane * 2;
+    dst[0] = to_fp8(v[0] * inv);
+    dst[1] = to_fp8(v[1] * inv);
+    if (lane == 0) {
+      if constexpr (TRANS_SCALE) scale[(int64_t)g * rows + row] = s;
+      else scale[row * groups + g] = s;
+    }
   }
 }
 
@@ -61,7 +97,8 @@ extern "C" hipError_t fi_per_group_quant_fp8(int dtype, int trans_scale, const v
                                              int64_t stride_row, float eps,
                                              hipStream_t stream) {
   if (K % 128 != 0) return hipErrorInvalidValue;
-  int64_t waves = rows * (K / 128);
+  // work item = one 4-group cluster (512 elems) or one tail 128-group
+  int64_t waves = rows * ((K / 512) + (K / 128 - (K / 512) * 4));
   int64_t blocks = (waves * 64 + 255) / 256;
   dim3 g((uint32_t)(blocks < 0x7FFFFFFF ? blocks : 0x7FFFFFFF)), blk(256);
 #define LQ(T, TS)                                                                  \
