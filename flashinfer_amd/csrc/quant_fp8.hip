@@ -84,8 +84,21 @@ template <typename T>
 __global__ void scale_quant_kernel(const T* __restrict__ x, uint8_t* __restrict__ q,
                                    const float* __restrict__ inv_scale, int64_t n) {
   float is = *inv_scale;
-  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+  // vectorized main body: 16B loads / 8B stores per thread
+  int64_t n8 = n / 8;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n8;
        i += (int64_t)gridDim.x * blockDim.x) {
+    vec_t<T, 8> lv;
+    lv.load(x + i * 8);
+    uint64_t packed = 0;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      packed |= (uint64_t)__builtin_bit_cast(uint8_t, to_fp8(lv.get(j) * is))
+                << (8 * j);
+    *reinterpret_cast<uint64_t*>(q + i * 8) = packed;
+  }
+  for (int64_t i = n8 * 8 + blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       i < n; i += (int64_t)gridDim.x * blockDim.x) {
     q[i] = to_fp8(to_f32<T>(x[i]) * is);
   }
 }
