@@ -95,3 +95,8 @@ for B, L in [(16, 1024), (32, 512)]:
         run_route(B, L, Hq, 8, "mfma")
 run_route(48, 1024, 32, 8, "auto")   # 384 units: fused vs
 run_route(48, 1024, 32, 8, "mfma")
+
+
+# fixed-vs-per-tile decomposition: bs=16 GQA-8, split fixed at 2
+for L in (256, 512, 1024, 2048, 4096):
+    run_route(16, L, 64, 8, "auto")
