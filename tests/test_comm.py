@@ -434,3 +434,55 @@ def test_moe_ep_mega_mode_gloo():
         p.join(180)
     assert all(p.exitcode == 0 for p in ps)
     assert fail.empty()
+
+
+def _dcp_body(rank, world):
+    """DCP q-scatter / o-gather vs a single-rank reference: each rank holds
+    partial attention states for the FULL batch over its KV shard; the
+    LSE merge must equal full-KV attention."""
+    from flashinfer_amd.comm.dcp import dcp_gather_o, dcp_scatter_q
+
+    B, H, D, L = 2, 4, 16, 32
+    g = torch.Generator().manual_seed(7)
+    q_full = torch.randn(world * B, H, D, generator=g)
+    k = torch.randn(L * world, H, D, generator=g)
+    v = torch.randn(L * world, H, D, generator=g)
+    # scatter: local batch shard -> full batch everywhere
+    q_local = q_full[rank * B : (rank + 1) * B]
+    q_all = dcp_scatter_q(q_local)
+    assert torch.equal(q_all, q_full)
+    # partial attention on this rank's KV shard (fp32 reference math)
+    ks = k[rank * L : (rank + 1) * L]
+    vs = v[rank * L : (rank + 1) * L]
+    logits = torch.einsum("bhd,lhd->bhl", q_full.float(), ks.float())
+    m = logits.max(-1, keepdim=True).values
+    p = torch.exp(logits - m)
+    o_part = torch.einsum("bhl,lhd->bhd", p, vs.float()) / p.sum(-1)[..., None]
+    lse_part = (m[..., 0] + p.sum(-1).log()) / torch.tensor(2.0).log()
+    sl = slice(rank * B, (rank + 1) * B)
+    o, lse = dcp_gather_o(o_part, lse_part, sl)
+    # reference: full-KV attention for this rank's batch rows
+    logits_f = torch.einsum("bhd,lhd->bhl", q_full[sl].float(), k.float())
+    ref = torch.softmax(logits_f, -1) @ v.float().permute(1, 0, 2).reshape(
+        H, world * L, D)[0] if False else torch.einsum(
+        "bhl,lhd->bhd", torch.softmax(logits_f, -1), v.float())
+    torch.testing.assert_close(o, ref, atol=1e-4, rtol=1e-4)
+
+
+def test_dcp_gloo():
+    _run_mp("_dcp_body", port=29517)
+
+
+def _backend_body(rank, world):
+    from flashinfer_amd.comm.comm_backend import TorchDistBackend
+
+    be = TorchDistBackend()
+    assert be.rank == rank and be.world_size == world
+    objs = be.allgather_object({"r": rank})
+    assert [o["r"] for o in objs] == list(range(world))
+    assert be.broadcast_object("x" if rank == 0 else None, src=0) == "x"
+    be.barrier()
+
+
+def test_comm_backend_gloo():
+    _run_mp("_backend_body", port=29518)
