@@ -150,7 +150,15 @@ def apply_llama31_rope_pos_ids_inplace(
 
 
 def _csc_reshape(x: torch.Tensor, head_size: int):
-    # vLLM convention: [..., num_heads*head_size] flat; reshape to 3-D
+    # vLLM convention: [..., num_heads*head_size] flat; reshape to 3-D.
+    # Guard: a 3-D [nnz, H, D] input here would silently become nnz*H
+    # "tokens" and index pos_ids out of bounds (native crash).
+    if x.dim() == 3 and x.shape[-1] == head_size:
+        return x
+    if x.shape[-1] % head_size:
+        raise ValueError(
+            f"last dim {x.shape[-1]} is not a multiple of head_size "
+            f"{head_size} (expected flat [tokens, num_heads*head_size])")
     nnz = x.numel() // x.shape[-1]
     return x.view(nnz, x.shape[-1] // head_size, head_size)
 

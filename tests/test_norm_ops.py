@@ -199,3 +199,20 @@ def test_fused_qk_rmsnorm_rope_3d():
     ref_q[..., 1::2] = x1 * sin[None, :, None, 1::2] + x2 * cos[None, :, None, 0::2]
     torch.testing.assert_close(q.float(), ref_q, atol=3e-2, rtol=3e-2)
     torch.testing.assert_close(v.float(), qkv3[:, :, 2].float())
+
+
+def test_gemma_fused_add_rmsnorm():
+    """Gemma variant: norm weight is (1 + w) — export previously untested."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(5)
+    x = torch.randn(32, 512, dtype=torch.bfloat16, device="cuda")
+    res = torch.randn(32, 512, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(512, dtype=torch.bfloat16, device="cuda") * 0.1
+    x2, r2 = x.clone(), res.clone()
+    fi.gemma_fused_add_rmsnorm(x2, r2, w)
+    s = x.float() + res.float()
+    ref = (s * torch.rsqrt(s.pow(2).mean(-1, keepdim=True) + 1e-6)
+           * (1.0 + w.float()))
+    torch.testing.assert_close(r2.float(), s, atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(x2.float(), ref, atol=3e-2, rtol=3e-2)

@@ -98,3 +98,30 @@ def test_merge_states_uniform(ns):
     torch.testing.assert_close(v_m, ref, atol=1e-3, rtol=1e-3)
     ref_s = (m.squeeze(1) + torch.log(w.sum(1))) / math.log(2)
     torch.testing.assert_close(s_m, ref_s, atol=1e-3, rtol=1e-3)
+
+
+def test_merge_state_in_place():
+    """In-place LSE merge (cascade primitive) vs the functional merge_state,
+    including the optional per-position mask."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(6)
+    n, h, d = 16, 8, 64
+    va = torch.randn(n, h, d, dtype=torch.bfloat16, device="cuda")
+    sa = torch.randn(n, h, device="cuda")
+    vb = torch.randn(n, h, d, dtype=torch.bfloat16, device="cuda")
+    sb = torch.randn(n, h, device="cuda")
+    ref_v, ref_s = fi.merge_state(va.clone(), sa.clone(), vb, sb)
+    v, s = va.clone(), sa.clone()
+    fi.merge_state_in_place(v, s, vb, sb)
+    torch.testing.assert_close(v.float(), ref_v.float(), atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(s, ref_s, atol=1e-4, rtol=1e-4)
+    # masked positions keep their original state
+    mask = torch.zeros(n, dtype=torch.bool, device="cuda")
+    mask[: n // 2] = True
+    v2, s2 = va.clone(), sa.clone()
+    fi.merge_state_in_place(v2, s2, vb, sb, mask=mask)
+    torch.testing.assert_close(v2[: n // 2].float(), ref_v[: n // 2].float(),
+                               atol=2e-2, rtol=2e-2)
+    assert torch.equal(v2[n // 2:], va[n // 2:])
+    assert torch.equal(s2[n // 2:], sa[n // 2:])

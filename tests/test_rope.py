@@ -122,3 +122,79 @@ def test_llama31_rope_matches_hf_formula():
     ref[..., :half] = xf[..., :half] * cos - xf[..., half:] * sin
     ref[..., half:] = xf[..., half:] * cos + xf[..., :half] * sin
     torch.testing.assert_close(q_out.float(), ref, atol=2e-2, rtol=2e-2)
+
+
+def test_inplace_rope_variants_match_out_of_place():
+    """The inplace/pos-ids/llama-3.1 export variants share the same kernel —
+    each must match its out-of-place sibling exactly."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(3)
+    nnz, Hq, Hkv, D = 64, 8, 2, 128
+    indptr = torch.tensor([0, 40, nnz], dtype=torch.int32, device="cuda")
+    offsets = torch.tensor([5, 17], dtype=torch.int32, device="cuda")
+
+    def fresh():
+        g = torch.Generator("cuda").manual_seed(11)
+        q = torch.randn(nnz, Hq, D, dtype=torch.bfloat16, device="cuda",
+                        generator=g)
+        k = torch.randn(nnz, Hkv, D, dtype=torch.bfloat16, device="cuda",
+                        generator=g)
+        return q, k
+
+    # plain rope: inplace == out-of-place
+    q, k = fresh()
+    qo, ko = fi.apply_rope(q, k, indptr, offsets)
+    qi, ki = fresh()
+    fi.apply_rope_inplace(qi, ki, indptr, offsets)
+    assert torch.equal(qo, qi) and torch.equal(ko, ki)
+
+    # pos-ids inplace == pos-ids out-of-place
+    pos_ids = torch.arange(nnz, dtype=torch.int32, device="cuda") % 40
+    q, k = fresh()
+    qo, ko = fi.apply_rope_pos_ids(q, k, pos_ids)
+    qi, ki = fresh()
+    fi.apply_rope_pos_ids_inplace(qi, ki, pos_ids)
+    assert torch.equal(qo, qi) and torch.equal(ko, ki)
+
+    # llama-3.1 wavelength-dependent scaling: inplace == out-of-place,
+    # and differs from plain rope (the frequency remap must be active)
+    q, k = fresh()
+    qo, ko = fi.apply_llama31_rope(q, k, indptr, offsets)
+    qi, ki = fresh()
+    fi.apply_llama31_rope_inplace(qi, ki, indptr, offsets)
+    assert torch.equal(qo, qi) and torch.equal(ko, ki)
+    qp, _ = fresh()
+    qplain, _ = fi.apply_rope(qp, k.clone(), indptr, offsets)
+    assert not torch.equal(qo, qplain)
+    qi, ki = fresh()
+    fi.apply_llama31_rope_pos_ids_inplace(
+        qi, ki, torch.cat([torch.arange(40, device="cuda"),
+                           torch.arange(nnz - 40, device="cuda")]).int() +
+        torch.cat([torch.full((40,), 5, device="cuda"),
+                   torch.full((nnz - 40,), 17, device="cuda")]).int())
+    assert torch.equal(qi, qo) and torch.equal(ki, ko)
+
+
+def test_rope_cos_sin_cache_inplace():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(4)
+    nnz, H, D = 32, 4, 64
+    pos_ids = torch.arange(nnz, dtype=torch.int32, device="cuda")
+    inv = 1.0 / (1e4 ** (torch.arange(0, D // 2, device="cuda").float()
+                         / (D // 2)))
+    ang = pos_ids.float()[:, None] * inv[None, :]
+    cos_sin = torch.cat([ang.cos(), ang.sin()], -1)
+    q3 = torch.randn(nnz, H, D, dtype=torch.bfloat16, device="cuda")
+    k3 = torch.randn(nnz, H, D, dtype=torch.bfloat16, device="cuda")
+    # vLLM convention: FLAT [tokens, heads*head_size]
+    q = q3.reshape(nnz, H * D).clone()
+    k = k3.reshape(nnz, H * D).clone()
+    fi.apply_rope_with_cos_sin_cache_inplace(pos_ids, q, k, D, cos_sin,
+                                             is_neox=True)
+    ref_q, ref_k = fi.apply_rope_pos_ids(q3, k3, pos_ids)
+    torch.testing.assert_close(q.view(nnz, H, D).float(), ref_q.float(),
+                               atol=5e-2, rtol=5e-2)
+    torch.testing.assert_close(k.view(nnz, H, D).float(), ref_k.float(),
+                               atol=5e-2, rtol=5e-2)
