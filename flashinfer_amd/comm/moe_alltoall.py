@@ -83,7 +83,8 @@ class MoeAlltoAll:
         out_l = out_splits.tolist()
         recv_x = _a2a_v(send_x, out_l, in_l, self.group)
         recv_exp = _a2a_v(send_exp.unsqueeze(1), out_l, in_l, self.group).squeeze(1)
-        state = dict(order=order, in_splits=in_l, out_splits=out_l, T=T, K=K)
+        state = dict(order=order, in_splits=in_l, out_splits=out_l, T=T, K=K,
+                     hidden=x.shape[-1])
         return recv_x, recv_exp - self.rank * self.experts_per_rank, state
 
     def combine(
@@ -96,6 +97,12 @@ class MoeAlltoAll:
         """
         back = _a2a_v(expert_out, state["in_splits"], state["out_splits"], self.group)
         T, K = state["T"], state["K"]
+        if T == 0:
+            # zero-token idle rank (reference repro_ikr_zero_token_idle
+            # contract): the alltoallv above MUST still run — peers'
+            # splits reference this rank — but there is nothing to reduce
+            # locally and view(0, K, -1) is ambiguous.
+            return expert_out.new_zeros((0, state["hidden"]))
         inv = torch.empty_like(state["order"])
         inv[state["order"]] = torch.arange(len(state["order"]),
                                            device=state["order"].device)

@@ -486,3 +486,33 @@ def _backend_body(rank, world):
 
 def test_comm_backend_gloo():
     _run_mp("_backend_body", port=29518)
+
+
+def _ep_zero_token_body(rank, world):
+    """Reference regression class (repro_ikr_zero_token_idle): an idle DP
+    rank forwards a ZERO-TOKEN batch while peers have work. The rank must
+    still issue its alltoallv (skipping would deadlock the peers), and
+    the tokens routed TO its experts must come back reduced correctly."""
+    from flashinfer_amd.moe_ep import MoeEp
+
+    E, K, H = 4, 2, 8
+    ep = MoeEp(E, K)
+    T = 6 if rank == 0 else 0   # rank 1 idle this step
+    logits = torch.full((T, E), -10.0)
+    if T:
+        logits[:, E // world:] = 10.0  # route everything to rank 1's experts
+    w, ids = ep.route(logits)
+    x = torch.randn(T, H)
+    recv_x, local_exp, state = ep.dispatch(x, ids)
+    if rank == 1:
+        # the idle rank received rank 0's tokens for its experts
+        assert recv_x.shape[0] == 6 * K
+    y = recv_x * 2.0
+    out = ep.combine(y, w, state)
+    assert out.shape == (T, H)
+    if T:
+        torch.testing.assert_close(out, 2.0 * x, atol=1e-5, rtol=1e-5)
+
+
+def test_moe_ep_zero_token_rank_gloo():
+    _run_mp("_ep_zero_token_body", port=29519)
