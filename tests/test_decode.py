@@ -414,3 +414,35 @@ def test_decode_tensor_cores_opt_out():
     o1 = w.run(q, (kc, vc))
     o2 = w2.run(q, (kc, vc))
     torch.testing.assert_close(o1.float(), o2.float(), atol=2e-2, rtol=2e-2)
+
+
+def test_decode_strided_q_from_fused_qkv():
+    """q sliced out of a fused QKV projection buffer (non-contiguous row
+    stride) must work on every decode route — the kernels take
+    q_stride_n/h, not an implicit contiguity assumption."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(8)
+    Hq, Hkv, D, page = 64, 8, 128, 16
+    kv_lens = [1024, 333]
+    indptr, indices, lpl, kc, vc = _paged(2, kv_lens, Hkv, D, page)
+    qkv = torch.randn(2, (Hq + 2 * Hkv) * D, dtype=torch.bfloat16,
+                      device="cuda")
+    q = qkv[:, : Hq * D].view(2, Hq, D)   # row stride = (Hq+2*Hkv)*D
+    assert not q.is_contiguous()
+    ws = torch.empty(64 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD")
+    w.plan(indptr, indices, lpl, Hq, Hkv, D, page,
+           q_data_type=torch.bfloat16)
+    assert w._fused_mfma
+    out = w.run(q, (kc, vc))
+    ref_out = w.run(q.contiguous(), (kc, vc))
+    assert torch.equal(out, ref_out)
+    # vector route too
+    w2 = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD",
+                                               use_tensor_cores=False)
+    w2.plan(indptr, indices, lpl, Hq, Hkv, D, page,
+            q_data_type=torch.bfloat16)
+    out2 = w2.run(q, (kc, vc))
+    torch.testing.assert_close(out2.float(), ref_out.float(), atol=3e-2,
+                               rtol=3e-2)

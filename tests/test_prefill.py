@@ -303,3 +303,26 @@ def test_prefill_split_kv_paged():
         vv = torch.cat(toks_v, 0)
         ref = ref_attn(q[qs:qe], kk, vv, causal=True)
         torch.testing.assert_close(out[qs:qe].float(), ref, atol=3e-2, rtol=3e-2)
+
+
+def test_prefill_strided_q_from_fused_qkv():
+    """Ragged prefill with q sliced from a fused QKV buffer (non-contiguous
+    row stride)."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(9)
+    nnz, Hq, Hkv, D = 512, 32, 8, 128
+    qkv = torch.randn(nnz, (Hq + 2 * Hkv) * D, dtype=torch.bfloat16,
+                      device="cuda")
+    q = qkv[:, : Hq * D].view(nnz, Hq, D)
+    assert not q.is_contiguous()
+    k = torch.randn(nnz, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(nnz, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    qo_indptr = torch.tensor([0, 256, nnz], dtype=torch.int32, device="cuda")
+    ws = torch.empty(128 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchPrefillWithRaggedKVCacheWrapper(ws, "NHD")
+    w.plan(qo_indptr, qo_indptr, Hq, Hkv, D, causal=True,
+           q_data_type=torch.bfloat16)
+    out = w.run(q, k, v)
+    ref = w.run(q.contiguous(), k, v)
+    assert torch.equal(out, ref)
