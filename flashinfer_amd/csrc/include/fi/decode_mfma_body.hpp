@@ -270,9 +270,17 @@ __device__ __forceinline__ void decode_mfma_item_body(const DecodeParams& p, int
     uint32_t W[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      asm("v_cvt_pk_bf16_f32 %0, %1, %2"
-          : "=v"(W[j])
-          : "v"(pr[2 * j]), "v"(pr[2 * j + 1]));
+      // dtype-correct packed convert (f16 kernels pack f16 fragments —
+      // was unconditionally bf16, a latent fp16 numerics bug)
+      if constexpr (__is_same(T, fp16)) {
+        asm("v_cvt_pkrtz_f16_f32 %0, %1, %2"
+            : "=v"(W[j])
+            : "v"(pr[2 * j]), "v"(pr[2 * j + 1]));
+      } else {
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2"
+            : "=v"(W[j])
+            : "v"(pr[2 * j]), "v"(pr[2 * j + 1]));
+      }
     }
     // half-exchange via v_permlane32_swap (guide T12): 4 VALU ops replace
     // 8 ds_bpermute + 8 selects (b0[0]=concat(W0.lo,W2.lo),

@@ -464,10 +464,19 @@ __device__ __forceinline__ void prefill_tile_body(const PrefillParams& p, int re
       uint32_t W[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        // hardware packed convert (no builtin on gfx950 — inline asm)
-        asm("v_cvt_pk_bf16_f32 %0, %1, %2"
-            : "=v"(W[j])
-            : "v"(pr[2 * j]), "v"(pr[2 * j + 1]));
+        // hardware packed convert (no builtin on gfx950 — inline asm);
+        // dtype-correct: f16 kernels must pack f16 fragments (this was
+        // v_cvt_pk_bf16_f32 unconditionally — latent fp16 numerics bug
+        // caught by test_prefill_fp16)
+        if constexpr (__is_same(T, fp16)) {
+          asm("v_cvt_pkrtz_f16_f32 %0, %1, %2"
+              : "=v"(W[j])
+              : "v"(pr[2 * j]), "v"(pr[2 * j + 1]));
+        } else {
+          asm("v_cvt_pk_bf16_f32 %0, %1, %2"
+              : "=v"(W[j])
+              : "v"(pr[2 * j]), "v"(pr[2 * j + 1]));
+        }
       }
       // NOTE: shfl must run on ALL lanes (uniform control flow) — select after.
       // half-exchange via v_permlane32_swap (guide T12): each call yields

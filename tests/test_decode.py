@@ -446,3 +446,32 @@ def test_decode_strided_q_from_fused_qkv():
     out2 = w2.run(q, (kc, vc))
     torch.testing.assert_close(out2.float(), ref_out.float(), atol=3e-2,
                                rtol=3e-2)
+
+
+def test_decode_fp16_routes():
+    """fp16 instantiations of the MFMA and fused decode kernels (everything
+    else in the suite runs bf16)."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(10)
+    D, page = 128, 16
+    kv_lens = [777, 128]
+    for Hq, Hkv, utc, want_mfma in ((64, 8, None, True),
+                                    (32, 8, None, True),
+                                    (32, 8, False, False)):
+        indptr, indices, lpl, kc, vc = _paged(2, kv_lens, Hkv, D, page)
+        kc = kc.to(torch.float16)
+        vc = vc.to(torch.float16)
+        q = torch.randn(2, Hq, D, dtype=torch.float16, device="cuda")
+        ws = torch.empty(64 << 20, dtype=torch.uint8, device="cuda")
+        w = fi.BatchDecodeWithPagedKVCacheWrapper(ws, "NHD",
+                                                  use_tensor_cores=utc)
+        w.plan(indptr, indices, lpl, Hq, Hkv, D, page,
+               q_data_type=torch.float16, kv_data_type=torch.float16)
+        assert w._fused_mfma == want_mfma
+        out = w.run(q, (kc, vc))
+        for b in range(2):
+            kk, vv = _gather(indptr, indices, kv_lens, kc, vc, b, page)
+            ref = sdpa_ref(q[b], kk, vv)
+            torch.testing.assert_close(out[b].float(), ref, atol=3e-2,
+                                       rtol=3e-2)

@@ -326,3 +326,21 @@ def test_prefill_strided_q_from_fused_qkv():
     out = w.run(q, k, v)
     ref = w.run(q.contiguous(), k, v)
     assert torch.equal(out, ref)
+
+
+def test_prefill_fp16():
+    import flashinfer_amd as fi
+
+    torch.manual_seed(11)
+    nnz, Hq, Hkv, D = 384, 32, 8, 128
+    q = torch.randn(nnz, Hq, D, dtype=torch.float16, device="cuda")
+    k = torch.randn(nnz, Hkv, D, dtype=torch.float16, device="cuda")
+    v = torch.randn(nnz, Hkv, D, dtype=torch.float16, device="cuda")
+    qo_indptr = torch.tensor([0, nnz], dtype=torch.int32, device="cuda")
+    ws = torch.empty(128 << 20, dtype=torch.uint8, device="cuda")
+    w = fi.BatchPrefillWithRaggedKVCacheWrapper(ws, "NHD")
+    w.plan(qo_indptr, qo_indptr, Hq, Hkv, D, causal=True,
+           q_data_type=torch.float16)
+    out = w.run(q, k, v)
+    ref = ref_attn(q, k, v, causal=True)
+    torch.testing.assert_close(out.float(), ref, atol=3e-2, rtol=3e-2)
