@@ -92,3 +92,37 @@ def test_holistic_deterministic():
     o1, _ = w.run(q, (kc, vc))
     o2, _ = w.run(q, (kc, vc))
     assert torch.equal(o1, o2)
+
+
+def test_holistic_fp16():
+    """fp16 instantiation of the persistent holistic kernel (exercises the
+    dtype-conditional P-pack through the shared prefill/decode bodies)."""
+    import flashinfer_amd as fi
+
+    torch.manual_seed(1)
+    Hq, Hkv, D, page = 64, 8, 128, 16
+    qo_lens = [256, 1, 1, 128]
+    kv_lens = [256, 1024, 512, 128]
+    (qo_indptr, kv_indptr, kv_indices, kv_len_arr, kc, vc,
+     pages_per) = _mixed_batch(qo_lens, kv_lens, Hkv, D, page)
+    kc = kc.to(torch.float16)
+    vc = vc.to(torch.float16)
+    nnz = sum(qo_lens)
+    q = torch.randn(nnz, Hq, D, dtype=torch.float16, device="cuda")
+    w = fi.BatchAttention("NHD")
+    w.plan(qo_indptr, kv_indptr, kv_indices, kv_len_arr, Hq, Hkv, D, D, page,
+           causal=True, q_data_type=torch.float16)
+    out, lse = w.run(q, (kc, vc))
+    for b in range(len(qo_lens)):
+        qs, qe = int(qo_indptr[b]), int(qo_indptr[b + 1])
+        base = int(kv_indptr[b])
+        toks_k, toks_v = [], []
+        for p_ in range(pages_per[b]):
+            pg = int(kv_indices[base + p_])
+            n = min(page, kv_lens[b] - p_ * page)
+            toks_k.append(kc[pg, :n])
+            toks_v.append(vc[pg, :n])
+        ref = _ref(q[qs:qe], torch.cat(toks_k, 0), torch.cat(toks_v, 0),
+                   True, qo_lens[b], kv_lens[b])
+        torch.testing.assert_close(out[qs:qe].float(), ref, atol=3e-2,
+                                   rtol=3e-2, msg=lambda m: f"req {b}: {m}")
