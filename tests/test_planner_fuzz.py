@@ -118,3 +118,31 @@ def test_decode_chunk_planner_coverage(seed):
         assert sorted(per_req[b]) == list(range(nc)), (b, L, chunk)
         assert merge_indptr[b + 1] - merge_indptr[b] == nc
     assert merge_indptr[0] == 0 and len(merge_indptr) == n + 1
+
+
+def test_workspace_allocator():
+    """Alignment, disjointness, overflow of the workspace carve-out used by
+    every split-KV path's partial buffers."""
+    from flashinfer_amd.utils import WorkspaceAllocator
+
+    buf = torch.zeros(4096, dtype=torch.uint8)
+    a = WorkspaceAllocator(buf)
+    t1 = a.alloc(100, torch.uint8, (100,))
+    t2 = a.alloc(256, torch.float32, (64,))
+    t3 = a.alloc(16, torch.int32, (4,))
+    # alignment: every allocation starts on a 256-byte boundary
+    base = buf.data_ptr()
+    for t in (t1, t2, t3):
+        assert (t.data_ptr() - base) % 256 == 0
+    # disjointness: writes don't alias
+    t1.fill_(1)
+    t2.fill_(2.0)
+    t3.fill_(3)
+    assert (t1 == 1).all() and (t2 == 2.0).all() and (t3 == 3).all()
+    # overflow raises with a helpful message
+    with pytest.raises(RuntimeError, match="workspace too small"):
+        a.alloc(100000, torch.uint8, (100000,))
+    # fresh allocator reuses from the start (wrappers re-plan this way)
+    b = WorkspaceAllocator(buf)
+    t4 = b.alloc(8, torch.uint8, (8,))
+    assert t4.data_ptr() == base
